@@ -1,0 +1,301 @@
+// Paged-attention decode (single query token per sequence), MI355X-native.
+//
+// Flash-decode structure: phase 1 fans (seq, kv_head, context-chunk) over
+// workgroups, each computing an online-softmax partial (m, l, acc) for the
+// GQA group of query heads sharing that kv head; phase 2 merges chunks.
+// Decode attention is HBM-bound (streaming the KV cache once); the kernel is
+// laid out for coalesced 32 B/lane KV reads: a wave covers 8 tokens x 8
+// dim-slices (16 dims each), so consecutive lanes read consecutive 16 B
+// chunks of a page row.
+//
+// Capability parity: the reference (ai-dynamo/dynamo) delegates paged
+// attention to vLLM/TRT-LLM; this is the native CDNA4 engine kernel.
+#include "common.h"
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+
+namespace {
+
+constexpr int kBlock = 256;        // 4 waves
+constexpr int kChunk = 512;        // context tokens per workgroup
+constexpr int kSlab = kChunk / 4;  // tokens per wave (128)
+constexpr float kNegInf = -1e30f;
+
+// G = GQA group size (Hq / Hkv), templated so the accumulator unrolls.
+template <int G>
+__global__ __launch_bounds__(kBlock) void paged_decode_phase1(
+    float* __restrict__ partial,        // [B, Hq, C, hd] fp32
+    float* __restrict__ ml,             // [B, Hq, C, 2] fp32 (m, l)
+    short* __restrict__ out,            // [B, Hq, hd] bf16 (used when C==1)
+    const short* __restrict__ q,        // [B, Hq, hd]
+    const short* __restrict__ kcache,   // [P, Hkv, ps, hd]
+    const short* __restrict__ vcache,
+    const int32_t* __restrict__ page_table,  // [B, max_pages]
+    const int32_t* __restrict__ ctx_lens,    // [B]
+    float scale, int B, int Hkv, int C, int max_pages, int log2_ps, int hd) {
+  const int b = blockIdx.x;
+  const int h = blockIdx.y;   // kv head
+  const int c = blockIdx.z;   // context chunk
+  const int Hq = Hkv * G;
+  const int ctx = ctx_lens[b];
+  const int chunk_start = c * kChunk;
+  const int lane = threadIdx.x & 63;
+  const int wid = threadIdx.x >> 6;
+  const int ts = lane >> 3;       // token slot within 8-token step
+  const int dp = lane & 7;        // dim slice: dims [dp*16, dp*16+16)
+  const int ps = 1 << log2_ps;
+
+  // LDS: q tile [G][hd] fp32 (pre-scaled), + merge scratch [4][G][hd+2]
+  extern __shared__ float lds[];
+  float* q_lds = lds;                       // G * hd
+  float* merge = lds + G * hd;              // 4 * G * (hd + 2)
+
+  if (chunk_start >= ctx) {
+    if (C > 1) {
+      // mark empty chunk for phase 2
+      for (int i = threadIdx.x; i < G; i += kBlock) {
+        const int qh = h * G + i;
+        float* mlp = ml + (((int64_t)b * Hq + qh) * C + c) * 2;
+        mlp[0] = kNegInf; mlp[1] = 0.f;
+      }
+    }
+    return;
+  }
+
+  for (int i = threadIdx.x; i < G * hd; i += kBlock) {
+    const int g = i / hd;
+    q_lds[i] = bf16_to_f32(((const short*)q)[((int64_t)b * Hq + h * G + g) * hd + i % hd]) * scale;
+  }
+  __syncthreads();
+
+  // preload this lane's q slice for each group head: q[g][dp*16 .. +16)
+  float qreg[G][16];
+#pragma unroll
+  for (int g = 0; g < G; g++)
+#pragma unroll
+    for (int i = 0; i < 16; i++) qreg[g][i] = q_lds[g * hd + dp * 16 + i];
+
+  float m[G], l[G], acc[G][16];
+#pragma unroll
+  for (int g = 0; g < G; g++) {
+    m[g] = kNegInf; l[g] = 0.f;
+#pragma unroll
+    for (int i = 0; i < 16; i++) acc[g][i] = 0.f;
+  }
+
+  const int slab_start = chunk_start + wid * kSlab;
+  const int slab_end = min(slab_start + kSlab, ctx);
+  const int32_t* pt = page_table + (int64_t)b * max_pages;
+
+  for (int step = slab_start; step < slab_end; step += 8) {
+    const int t = step + ts;
+    const bool valid = t < ctx;
+    // k row address via page table
+    const short* krow = nullptr;
+    const short* vrow = nullptr;
+    if (valid) {
+      const int64_t page = pt[t >> log2_ps];
+      const int row = t & (ps - 1);
+      const int64_t base = (((page * Hkv + h) * ps + row) * hd);
+      krow = kcache + base;
+      vrow = vcache + base;
+    }
+    short8 k0 = valid ? *reinterpret_cast<const short8*>(krow + dp * 16) : short8{};
+    short8 k1 = valid ? *reinterpret_cast<const short8*>(krow + dp * 16 + 8) : short8{};
+
+    float s[G];
+#pragma unroll
+    for (int g = 0; g < G; g++) {
+      float d = 0.f;
+#pragma unroll
+      for (int i = 0; i < 8; i++) {
+        d += bf16_to_f32(k0[i]) * qreg[g][i];
+        d += bf16_to_f32(k1[i]) * qreg[g][8 + i];
+      }
+      // butterfly over the 8 dim-slice lanes -> full dot product on all lanes
+      d += __shfl_xor(d, 1, WAVE_SIZE);
+      d += __shfl_xor(d, 2, WAVE_SIZE);
+      d += __shfl_xor(d, 4, WAVE_SIZE);
+      s[g] = valid ? d : kNegInf;
+    }
+
+    // load v slice once per token
+    short8 v0 = valid ? *reinterpret_cast<const short8*>(vrow + dp * 16) : short8{};
+    short8 v1 = valid ? *reinterpret_cast<const short8*>(vrow + dp * 16 + 8) : short8{};
+    float vf[16];
+#pragma unroll
+    for (int i = 0; i < 8; i++) { vf[i] = bf16_to_f32(v0[i]); vf[8 + i] = bf16_to_f32(v1[i]); }
+
+#pragma unroll
+    for (int g = 0; g < G; g++) {
+      // step max over the 8 tokens (s[g] is uniform within a token's lanes)
+      float ms = s[g];
+      ms = fmaxf(ms, __shfl_xor(ms, 8, WAVE_SIZE));
+      ms = fmaxf(ms, __shfl_xor(ms, 16, WAVE_SIZE));
+      ms = fmaxf(ms, __shfl_xor(ms, 32, WAVE_SIZE));
+      if (ms > m[g]) {
+        const float corr = __expf(m[g] - ms);
+        l[g] *= corr;
+#pragma unroll
+        for (int i = 0; i < 16; i++) acc[g][i] *= corr;
+        m[g] = ms;
+      }
+      const float p = (s[g] > kNegInf * 0.5f) ? __expf(s[g] - m[g]) : 0.f;
+      // sum of p over the 8 tokens of this step
+      float psum = p;
+      psum += __shfl_xor(psum, 8, WAVE_SIZE);
+      psum += __shfl_xor(psum, 16, WAVE_SIZE);
+      psum += __shfl_xor(psum, 32, WAVE_SIZE);
+      l[g] += psum;
+#pragma unroll
+      for (int i = 0; i < 16; i++) acc[g][i] = fmaf(p, vf[i], acc[g][i]);
+    }
+  }
+
+  // fold the 8 token-slots: acc currently holds per-(ts) partial sums
+#pragma unroll
+  for (int g = 0; g < G; g++)
+#pragma unroll
+    for (int i = 0; i < 16; i++) {
+      float a = acc[g][i];
+      a += __shfl_xor(a, 8, WAVE_SIZE);
+      a += __shfl_xor(a, 16, WAVE_SIZE);
+      a += __shfl_xor(a, 32, WAVE_SIZE);
+      acc[g][i] = a;
+    }
+
+  // cross-wave merge via LDS. Wave w writes [G][hd] acc + m,l.
+  __syncthreads();  // q_lds no longer needed
+  float* my = merge + wid * G * (hd + 2);
+  if (ts == 0) {  // lanes 0..7 cover the 8 dim slices exactly once
+#pragma unroll
+    for (int g = 0; g < G; g++) {
+#pragma unroll
+      for (int i = 0; i < 16; i++) my[g * (hd + 2) + dp * 16 + i] = acc[g][i];
+      if (dp == 0) { my[g * (hd + 2) + hd] = m[g]; my[g * (hd + 2) + hd + 1] = l[g]; }
+    }
+  }
+  __syncthreads();
+
+  // threads [0, G*hd) each merge one (g, d) across the 4 waves
+  for (int i = threadIdx.x; i < G * hd; i += kBlock) {
+    const int g = i / hd;
+    const int d = i % hd;
+    float mw[4], lw[4];
+    float mstar = kNegInf;
+#pragma unroll
+    for (int w = 0; w < 4; w++) {
+      mw[w] = merge[w * G * (hd + 2) + g * (hd + 2) + hd];
+      lw[w] = merge[w * G * (hd + 2) + g * (hd + 2) + hd + 1];
+      mstar = fmaxf(mstar, mw[w]);
+    }
+    float lsum = 0.f, asum = 0.f;
+#pragma unroll
+    for (int w = 0; w < 4; w++) {
+      const float corr = (lw[w] > 0.f) ? __expf(mw[w] - mstar) : 0.f;
+      lsum += lw[w] * corr;
+      asum += merge[w * G * (hd + 2) + g * (hd + 2) + d] * corr;
+    }
+    const int qh = h * G + g;
+    if (C == 1) {
+      out[((int64_t)b * Hq + qh) * hd + d] =
+          f32_to_bf16(lsum > 0.f ? asum / lsum : 0.f);
+    } else {
+      partial[(((int64_t)b * Hq + qh) * C + c) * hd + d] = asum;
+      if (d == 0) {
+        float* mlp = ml + (((int64_t)b * Hq + qh) * C + c) * 2;
+        mlp[0] = mstar; mlp[1] = lsum;
+      }
+    }
+  }
+}
+
+// Phase 2: merge chunk partials. grid (B, Hq), block = 128.
+__global__ void paged_decode_phase2(short* __restrict__ out,  // [B, Hq, hd]
+                                    const float* __restrict__ partial,
+                                    const float* __restrict__ ml,
+                                    const int32_t* __restrict__ ctx_lens,
+                                    int Hq, int C, int hd) {
+  const int b = blockIdx.x;
+  const int qh = blockIdx.y;
+  const int nc = min(C, (ctx_lens[b] + kChunk - 1) / kChunk);
+  const float* mlp = ml + (((int64_t)b * Hq + qh) * C) * 2;
+
+  __shared__ float smax[1];
+  float mstar = kNegInf;
+  for (int c = threadIdx.x; c < nc; c += blockDim.x) mstar = fmaxf(mstar, mlp[2 * c]);
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) mstar = fmaxf(mstar, __shfl_xor(mstar, off, WAVE_SIZE));
+  if (threadIdx.x == 0) smax[0] = mstar;
+  __syncthreads();
+  if (threadIdx.x == 64) smax[0] = fmaxf(smax[0], mstar);
+  __syncthreads();
+  mstar = smax[0];
+
+  // each thread owns output dims stride-wise
+  for (int d = threadIdx.x; d < hd; d += blockDim.x) {
+    float asum = 0.f, lsum = 0.f;
+    for (int c = 0; c < nc; c++) {
+      const float lc = mlp[2 * c + 1];
+      if (lc <= 0.f) continue;
+      const float corr = __expf(mlp[2 * c] - mstar);
+      asum += partial[(((int64_t)b * Hq + qh) * C + c) * hd + d] * corr;
+      lsum += lc * corr;
+    }
+    out[((int64_t)b * Hq + qh) * hd + d] = f32_to_bf16(lsum > 0.f ? asum / lsum : 0.f);
+  }
+}
+
+}  // namespace
+
+int64_t paged_decode_num_chunks(int64_t max_ctx) {
+  return std::max<int64_t>(1, (max_ctx + kChunk - 1) / kChunk);
+}
+
+void paged_attention_decode(torch::Tensor out, torch::Tensor q,
+                            torch::Tensor kcache, torch::Tensor vcache,
+                            torch::Tensor page_table, torch::Tensor ctx_lens,
+                            torch::Tensor partial, torch::Tensor ml,
+                            double scale) {
+  TORCH_CHECK(q.is_cuda() && q.dtype() == torch::kBFloat16);
+  TORCH_CHECK(page_table.dtype() == torch::kInt32 && ctx_lens.dtype() == torch::kInt32);
+  const int B = q.size(0);
+  const int Hq = q.size(1);
+  const int hd = q.size(2);
+  const int Hkv = kcache.size(1);
+  const int ps = kcache.size(2);
+  const int G = Hq / Hkv;
+  const int max_pages = page_table.size(1);
+  const int C = ml.size(2);  // chunk capacity allocated by caller
+  TORCH_CHECK(hd == 128, "only head_dim=128 supported natively");
+  TORCH_CHECK((ps & (ps - 1)) == 0, "page_size must be a power of 2");
+  int log2_ps = 0; while ((1 << log2_ps) < ps) log2_ps++;
+  if (B == 0) return;
+  auto stream = at::cuda::getCurrentHIPStream();
+
+  const int lds_bytes = (G * hd + 4 * G * (hd + 2)) * sizeof(float);
+  dim3 grid(B, Hkv, C);
+#define LAUNCH_G(GG)                                                          \
+  paged_decode_phase1<GG><<<grid, kBlock, lds_bytes, stream>>>(               \
+      partial.data_ptr<float>(), ml.data_ptr<float>(), (short*)out.data_ptr(),\
+      (const short*)q.data_ptr(), (const short*)kcache.data_ptr(),            \
+      (const short*)vcache.data_ptr(), page_table.data_ptr<int32_t>(),        \
+      ctx_lens.data_ptr<int32_t>(), (float)scale, B, Hkv, C, max_pages,       \
+      log2_ps, hd)
+  switch (G) {
+    case 1: LAUNCH_G(1); break;
+    case 2: LAUNCH_G(2); break;
+    case 4: LAUNCH_G(4); break;
+    case 8: LAUNCH_G(8); break;
+    default: TORCH_CHECK(false, "unsupported GQA group size ", G);
+  }
+#undef LAUNCH_G
+  HIP_CHECK_KERNEL();
+  if (C > 1) {
+    dim3 grid2(B, Hq);
+    paged_decode_phase2<<<grid2, 128, 0, stream>>>(
+        (short*)out.data_ptr(), partial.data_ptr<float>(), ml.data_ptr<float>(),
+        ctx_lens.data_ptr<int32_t>(), Hq, C, hd);
+    HIP_CHECK_KERNEL();
+  }
+}

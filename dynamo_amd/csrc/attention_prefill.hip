@@ -1,0 +1,263 @@
+// Paged flash-attention prefill (causal, GQA, chunked-prefill-aware),
+// MI355X-native, MFMA mfma_f32_16x16x32_bf16 with LDS-staged KV tiles.
+//
+// Structure (guide §5/§B): workgroup = 4 waves; each workgroup owns a 64-row
+// query tile of one query head; waves own 16 rows each. K/V tiles of 64
+// tokens are gathered from the paged cache into LDS (XOR-swizzled layouts,
+// guide T2/G4: row-major [*][128] bf16 is otherwise a 16-32-way bank
+// conflict on ds_read_b128). Online softmax with per-row running (m, l).
+// The P tile round-trips through LDS to re-shape the S-layout (C/D frag)
+// into the PV A-operand layout.
+//
+// MFMA fragment mappings used (verified on HW by tests/test_gpu_mfma.py):
+//   A: lane l holds A[row = l%16][k = 8*(l/16) + i]      (i = 0..7)
+//   B: lane l holds B[k = 8*(l/16) + i][col = l%16]
+//   C/D: lane l, reg r holds D[row = (l/16)*4 + r][col = l%16]
+//
+// Capability parity: the reference (ai-dynamo/dynamo) delegates prefill
+// attention to vLLM/TRT-LLM; this is the native CDNA4 engine kernel.
+#include "common.h"
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+
+namespace {
+
+typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8_t;
+
+constexpr int kBlock = 256;  // 4 waves
+constexpr int kQTile = 64;   // query rows per workgroup
+constexpr int kKvTile = 64;  // kv tokens per LDS tile
+constexpr float kNegInf = -1e30f;
+
+DEVINL int swz(int byte_in_row, int row) { return byte_in_row ^ ((row & 7) << 4); }
+
+__global__ __launch_bounds__(kBlock) void prefill_kernel(
+    short* __restrict__ out,            // [Tq, Hq, 128]
+    const short* __restrict__ q,        // [Tq, Hq, 128]
+    const short* __restrict__ kcache,   // [P, Hkv, ps, 128]
+    const short* __restrict__ vcache,
+    const int32_t* __restrict__ page_table,   // [nseq, max_pages]
+    const int32_t* __restrict__ tile_seq,     // [ntiles]
+    const int32_t* __restrict__ tile_q0,      // [ntiles] local q row of tile
+    const int32_t* __restrict__ seq_q_start,  // [nseq] offset into Tq
+    const int32_t* __restrict__ seq_q_len,    // [nseq]
+    const int32_t* __restrict__ seq_ctx_len,  // [nseq] total kv len
+    float scale, int Hq, int Hkv, int max_pages, int log2_ps) {
+  constexpr int HD = 128;
+  const int tile = blockIdx.x;
+  const int qh = blockIdx.y;
+  const int kvh = qh / (Hq / Hkv);
+  const int seq = tile_seq[tile];
+  const int q0 = tile_q0[tile];
+  const int qlen = seq_q_len[seq];
+  const int ctx = seq_ctx_len[seq];
+  const int qstart = seq_q_start[seq];
+  const int ps = 1 << log2_ps;
+  const int32_t* pt = page_table + (int64_t)seq * max_pages;
+
+  const int lane = threadIdx.x & 63;
+  const int wid = threadIdx.x >> 6;
+  const int lr = lane & 15;   // row-or-col lane index
+  const int lg = lane >> 4;   // 4 k-groups
+
+  // LDS: K [64][128] (16 KB) + V^T [128][64] (16 KB) + P [4][16][64] (8 KB)
+  __shared__ short k_lds[kKvTile * HD];
+  __shared__ short vt_lds[HD * kKvTile];
+  __shared__ short p_lds[4][16 * kKvTile];
+
+  // ---- load Q fragments (once; reused across all kv tiles) ----
+  // wave's rows: q0 + wid*16 + lr ; A-frag kc covers dims [kc*32, kc*32+32)
+  bf16x8_t q_frag[4];
+  const int my_qrow = q0 + wid * 16 + lr;
+  const bool row_valid = my_qrow < qlen;
+  {
+    const short* qrow_ptr = q + ((int64_t)(qstart + (row_valid ? my_qrow : 0)) * Hq + qh) * HD;
+#pragma unroll
+    for (int kc = 0; kc < 4; kc++) {
+      short8 v = row_valid ? *reinterpret_cast<const short8*>(qrow_ptr + kc * 32 + lg * 8)
+                           : short8{};
+      q_frag[kc] = *reinterpret_cast<bf16x8_t*>(&v);
+    }
+  }
+  const int my_qpos = ctx - qlen + my_qrow;  // absolute kv position of this row
+
+  float m[4], lsum[4];
+  f32x4 acc_o[8];
+#pragma unroll
+  for (int r = 0; r < 4; r++) { m[r] = kNegInf; lsum[r] = 0.f; }
+#pragma unroll
+  for (int d = 0; d < 8; d++) acc_o[d] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+  // last kv position any row of this tile may attend to
+  const int tile_qpos_max = ctx - qlen + min(q0 + kQTile - 1, qlen - 1);
+  const int kv_end = min(ctx, tile_qpos_max + 1);
+
+  for (int t0 = 0; t0 < kv_end; t0 += kKvTile) {
+    __syncthreads();
+    // ---- stage K tile + V^T tile ----
+#pragma unroll
+    for (int i = 0; i < 4; i++) {
+      const int slot = i * kBlock + threadIdx.x;  // 1024 short8 slots
+      const int row = slot >> 4;
+      const int col8 = slot & 15;
+      const int t = t0 + row;
+      short8 kv_k{}, kv_v{};
+      if (t < ctx) {
+        const int64_t page = pt[t >> log2_ps];
+        const int64_t base = ((page * Hkv + kvh) * ps + (t & (ps - 1))) * HD + col8 * 8;
+        kv_k = *reinterpret_cast<const short8*>(kcache + base);
+        kv_v = *reinterpret_cast<const short8*>(vcache + base);
+      }
+      *reinterpret_cast<short8*>((char*)k_lds + row * 256 + swz(col8 * 16, row)) = kv_k;
+#pragma unroll
+      for (int e = 0; e < 8; e++) {
+        const int dim = col8 * 8 + e;
+        *(short*)((char*)vt_lds + dim * 128 + swz(row * 2, dim)) = kv_v[e];
+      }
+    }
+    __syncthreads();
+
+    // ---- S = Q K^T for the wave's 16 rows x 64 tokens ----
+    f32x4 s[4];
+#pragma unroll
+    for (int n = 0; n < 4; n++) {
+      f32x4 accs{0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+      for (int kc = 0; kc < 4; kc++) {
+        // B-frag: K[tok = n*16 + lr][d = kc*32 + lg*8 + j]
+        const int tok = n * 16 + lr;
+        short8 kv = *reinterpret_cast<const short8*>(
+            (char*)k_lds + tok * 256 + swz(kc * 64 + lg * 16, tok));
+        accs = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            q_frag[kc], *reinterpret_cast<bf16x8_t*>(&kv), accs, 0, 0, 0);
+      }
+      s[n] = accs;
+    }
+
+    // ---- online softmax (per row; rows live in (lg, r); cols in lr) ----
+    float mt[4];
+#pragma unroll
+    for (int r = 0; r < 4; r++) {
+      const int qrow = q0 + wid * 16 + lg * 4 + r;
+      const int qpos = ctx - qlen + qrow;
+      float mx = kNegInf;
+#pragma unroll
+      for (int n = 0; n < 4; n++) {
+        const int kvpos = t0 + n * 16 + lr;
+        float sv = s[n][r] * scale;
+        sv = (qrow < qlen && kvpos <= qpos && kvpos < ctx) ? sv : kNegInf;
+        s[n][r] = sv;
+        mx = fmaxf(mx, sv);
+      }
+      mx = fmaxf(mx, __shfl_xor(mx, 1, WAVE_SIZE));
+      mx = fmaxf(mx, __shfl_xor(mx, 2, WAVE_SIZE));
+      mx = fmaxf(mx, __shfl_xor(mx, 4, WAVE_SIZE));
+      mx = fmaxf(mx, __shfl_xor(mx, 8, WAVE_SIZE));
+      mt[r] = mx;
+    }
+
+    // NOTE: rows of S (and acc_o) map to (lg, r): row = lg*4 + r. The running
+    // m/l state for a given physical row is therefore kept by all 16 lanes
+    // with that lg, redundantly — shuffles above keep them consistent.
+    float p[4][4];  // [n][r]
+#pragma unroll
+    for (int r = 0; r < 4; r++) {
+      float mnew = m[r];
+      if (mt[r] > mnew) {
+        const float corr = (mnew <= kNegInf * 0.5f) ? 0.f : __expf(mnew - mt[r]);
+        lsum[r] *= corr;
+#pragma unroll
+        for (int d = 0; d < 8; d++) acc_o[d][r] *= corr;
+        mnew = mt[r];
+        m[r] = mnew;
+      }
+      float rowsum = 0.f;
+#pragma unroll
+      for (int n = 0; n < 4; n++) {
+        const float pv = (s[n][r] <= kNegInf * 0.5f || mnew <= kNegInf * 0.5f)
+                             ? 0.f : __expf(s[n][r] - mnew);
+        p[n][r] = pv;
+        rowsum += pv;
+      }
+      rowsum += __shfl_xor(rowsum, 1, WAVE_SIZE);
+      rowsum += __shfl_xor(rowsum, 2, WAVE_SIZE);
+      rowsum += __shfl_xor(rowsum, 4, WAVE_SIZE);
+      rowsum += __shfl_xor(rowsum, 8, WAVE_SIZE);
+      lsum[r] += rowsum;
+    }
+
+    // ---- write P tile to LDS (re-layout for the PV A-operand) ----
+    __syncthreads();  // all waves done reading k_lds-dependent S
+#pragma unroll
+    for (int r = 0; r < 4; r++) {
+#pragma unroll
+      for (int n = 0; n < 4; n++) {
+        const int row = lg * 4 + r;       // q row within wave tile
+        const int tok = n * 16 + lr;
+        *(short*)((char*)p_lds[wid] + row * 128 + swz(tok * 2, row)) =
+            f32_to_bf16(p[n][r]);
+      }
+    }
+    __syncthreads();
+
+    // ---- O += P V ----
+#pragma unroll
+    for (int kt = 0; kt < 2; kt++) {
+      // A-frag: P[row = lr][tok = kt*32 + lg*8 + j]
+      short8 pa = *reinterpret_cast<const short8*>(
+          (char*)p_lds[wid] + lr * 128 + swz(kt * 64 + lg * 16, lr));
+#pragma unroll
+      for (int d = 0; d < 8; d++) {
+        // B-frag: V[tok = kt*32 + lg*8 + j][dim = d*16 + lr] = V^T[dim][tok]
+        const int dim = d * 16 + lr;
+        short8 vb = *reinterpret_cast<const short8*>(
+            (char*)vt_lds + dim * 128 + swz(kt * 64 + lg * 16, dim));
+        acc_o[d] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            *reinterpret_cast<bf16x8_t*>(&pa), *reinterpret_cast<bf16x8_t*>(&vb),
+            acc_o[d], 0, 0, 0);
+      }
+    }
+  }
+
+  // ---- epilogue: normalize and store ----
+#pragma unroll
+  for (int r = 0; r < 4; r++) {
+    const int qrow = q0 + wid * 16 + lg * 4 + r;
+    if (qrow >= qlen) continue;
+    const float inv = lsum[r] > 0.f ? 1.f / lsum[r] : 0.f;
+    short* orow = out + ((int64_t)(qstart + qrow) * Hq + qh) * HD;
+#pragma unroll
+    for (int d = 0; d < 8; d++) orow[d * 16 + lr] = f32_to_bf16(acc_o[d][r] * inv);
+  }
+}
+
+}  // namespace
+
+void attention_prefill_paged(torch::Tensor out, torch::Tensor q,
+                             torch::Tensor kcache, torch::Tensor vcache,
+                             torch::Tensor page_table, torch::Tensor tile_seq,
+                             torch::Tensor tile_q0, torch::Tensor seq_q_start,
+                             torch::Tensor seq_q_len, torch::Tensor seq_ctx_len,
+                             double scale) {
+  TORCH_CHECK(q.is_cuda() && q.dtype() == torch::kBFloat16);
+  TORCH_CHECK(q.size(-1) == 128, "only head_dim=128 supported natively");
+  TORCH_CHECK(page_table.dtype() == torch::kInt32);
+  const int Hq = q.size(1);
+  const int Hkv = kcache.size(1);
+  const int ps = kcache.size(2);
+  TORCH_CHECK((ps & (ps - 1)) == 0);
+  int log2_ps = 0; while ((1 << log2_ps) < ps) log2_ps++;
+  const int ntiles = tile_seq.size(0);
+  const int max_pages = page_table.size(1);
+  if (ntiles == 0) return;
+  auto stream = at::cuda::getCurrentHIPStream();
+  dim3 grid(ntiles, Hq);
+  prefill_kernel<<<grid, kBlock, 0, stream>>>(
+      (short*)out.data_ptr(), (const short*)q.data_ptr(),
+      (const short*)kcache.data_ptr(), (const short*)vcache.data_ptr(),
+      page_table.data_ptr<int32_t>(), tile_seq.data_ptr<int32_t>(),
+      tile_q0.data_ptr<int32_t>(), seq_q_start.data_ptr<int32_t>(),
+      seq_q_len.data_ptr<int32_t>(), seq_ctx_len.data_ptr<int32_t>(),
+      (float)scale, Hq, Hkv, max_pages, log2_ps);
+  HIP_CHECK_KERNEL();
+}

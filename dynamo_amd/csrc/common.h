@@ -1,0 +1,97 @@
+// Common device helpers for dynamo_amd HIP kernels (gfx950 / CDNA4 only).
+//
+// Design notes (see /opt/skills/guides/cdna_hip_programming.md):
+//  - wavefront = 64 lanes; block sizes are multiples of 64
+//  - bf16 traffic is always vectorized as short4/short8 (scalar bf16 loads
+//    are ~2-2.5x slower; guide Common-mistake #2)
+//  - fp32 accumulation everywhere
+#pragma once
+
+#include <hip/hip_runtime.h>
+#include <hip/hip_bf16.h>
+#include <hip/hip_fp16.h>
+#include <stdint.h>
+
+#define WAVE_SIZE 64
+
+#define DEVINL __device__ __forceinline__
+
+// ---- vector types -------------------------------------------------------
+// 8 bf16 = 16 B, the coalescing sweet spot.
+typedef __attribute__((ext_vector_type(8))) short short8;
+typedef __attribute__((ext_vector_type(4))) short short4v;
+typedef __attribute__((ext_vector_type(4))) float float4v;
+typedef __attribute__((ext_vector_type(2))) float float2v;
+
+// MFMA fragment types for mfma_f32_16x16x32_bf16 (gfx950).
+typedef __attribute__((ext_vector_type(8))) short bf16x8;   // A/B operand: 8 bf16
+typedef __attribute__((ext_vector_type(4))) float f32x4;    // C/D accum: 4 fp32
+// MFMA fragment types for mfma_f32_32x32x16_bf16.
+typedef __attribute__((ext_vector_type(16))) float f32x16;  // C/D accum: 16 fp32
+
+DEVINL float bf16_to_f32(short u) {
+  union { float f; uint32_t i; } v;
+  v.i = ((uint32_t)(uint16_t)u) << 16;
+  return v.f;
+}
+
+DEVINL short f32_to_bf16(float f) {
+  // round-to-nearest-even
+  union { float f; uint32_t i; } v;
+  v.f = f;
+  uint32_t lsb = (v.i >> 16) & 1;
+  uint32_t rounded = v.i + 0x7fff + lsb;
+  return (short)(rounded >> 16);
+}
+
+// ---- wave reductions ----------------------------------------------------
+DEVINL float wave_reduce_sum(float x) {
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) x += __shfl_xor(x, off, WAVE_SIZE);
+  return x;
+}
+
+DEVINL float wave_reduce_max(float x) {
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) x = fmaxf(x, __shfl_xor(x, off, WAVE_SIZE));
+  return x;
+}
+
+// Reduce across a contiguous group of `width` lanes (width power of 2).
+template <int WIDTH>
+DEVINL float group_reduce_sum(float x) {
+#pragma unroll
+  for (int off = WIDTH / 2; off > 0; off >>= 1) x += __shfl_xor(x, off, WAVE_SIZE);
+  return x;
+}
+
+// ---- block reduction (needs extern LDS scratch of >= num_waves floats) --
+// Block-level sum using a caller-provided LDS buffer (size: block/64 floats).
+DEVINL float block_reduce_sum(float x, float* lds_scratch) {
+  const int lane = threadIdx.x & (WAVE_SIZE - 1);
+  const int wid = threadIdx.x / WAVE_SIZE;
+  const int nwaves = blockDim.x / WAVE_SIZE;
+  x = wave_reduce_sum(x);
+  if (lane == 0) lds_scratch[wid] = x;
+  __syncthreads();
+  float total = 0.f;
+  if (threadIdx.x < (unsigned)nwaves) total = lds_scratch[threadIdx.x];
+  if (wid == 0) {
+#pragma unroll
+    for (int off = 8; off > 0; off >>= 1) total += __shfl_xor(total, off, WAVE_SIZE);
+    if (lane == 0) lds_scratch[0] = total;
+  }
+  __syncthreads();
+  return lds_scratch[0];
+}
+
+// ---- misc ---------------------------------------------------------------
+DEVINL int64_t cdiv64(int64_t a, int64_t b) { return (a + b - 1) / b; }
+
+#define HIP_CHECK_KERNEL()                                                \
+  do {                                                                    \
+    hipError_t _e = hipGetLastError();                                    \
+    if (_e != hipSuccess) {                                               \
+      TORCH_CHECK(false, "HIP kernel launch failed: ", hipGetErrorString(_e)); \
+    }                                                                     \
+  } while (0)

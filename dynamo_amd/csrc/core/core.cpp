@@ -1,0 +1,165 @@
+// dynamo_amd._core — native C++ control-plane components.
+//
+// KV-aware routing index + canonical block hashing. This is the MI355X
+// build's equivalent of the reference's Rust routing library
+// (ai-dynamo/dynamo lib/kv-router/src/indexer/radix_tree.rs:49 RadixTree,
+// find_matches :225, apply_event :229) and of the canonical block-hash
+// chain (lib/kv-hashing/src/compute.rs:15-35, lib/tokens/src/blocks.rs).
+//
+// Because sequence hashes are chained (each block hash mixes its parent's
+// hash), prefix matching over a radix tree degenerates to walking the hash
+// chain and probing a flat hash map — same asymptotics, better constants,
+// and the structure the router actually needs (hash -> worker set).
+#include <pybind11/pybind11.h>
+#include <pybind11/stl.h>
+#include <cstdint>
+#include <mutex>
+#include <unordered_map>
+#include <unordered_set>
+#include <vector>
+
+namespace py = pybind11;
+
+namespace {
+
+// splitmix64-based mixing; stable across the framework (engine + router
+// must agree — hash parity is load-bearing for KV-aware routing).
+inline uint64_t mix64(uint64_t x) {
+  x += 0x9e3779b97f4a7c15ull;
+  x = (x ^ (x >> 30)) * 0xbf58476d1ce4e5b9ull;
+  x = (x ^ (x >> 27)) * 0x94d049bb133111ebull;
+  return x ^ (x >> 31);
+}
+
+inline uint64_t hash_tokens_impl(uint64_t parent, const int32_t* tokens, size_t n) {
+  uint64_t h = mix64(parent ^ 0xd6e8feb86659fd93ull);
+  for (size_t i = 0; i < n; i++) h = mix64(h ^ (uint64_t)(uint32_t)tokens[i]);
+  return h;
+}
+
+}  // namespace
+
+// Chain hashes for a token sequence: block i covers tokens
+// [i*bs, (i+1)*bs); only full blocks are hashed. salt seeds the chain
+// (LoRA-name/model-aware salting like the reference's SaltHash).
+static std::vector<uint64_t> chain_hashes(const std::vector<int32_t>& tokens,
+                                          int64_t block_size, uint64_t salt) {
+  std::vector<uint64_t> out;
+  const size_t nb = tokens.size() / (size_t)block_size;
+  out.reserve(nb);
+  uint64_t parent = mix64(salt ^ 0xa0761d6478bd642full);
+  for (size_t b = 0; b < nb; b++) {
+    parent = hash_tokens_impl(parent, tokens.data() + b * block_size, block_size);
+    out.push_back(parent);
+  }
+  return out;
+}
+
+static uint64_t hash_block(uint64_t parent, const std::vector<int32_t>& tokens) {
+  return hash_tokens_impl(parent, tokens.data(), tokens.size());
+}
+
+// ---------------------------------------------------------------------------
+// KvIndexer: sequence-hash -> {workers that hold the block}, plus per-worker
+// block counts. Thread-safe (event ingestion and routing run on different
+// threads).
+class KvIndexer {
+ public:
+  void apply_stored(int64_t worker, const std::vector<uint64_t>& hashes) {
+    std::lock_guard<std::mutex> g(mu_);
+    auto& wset = worker_blocks_[worker];
+    for (uint64_t h : hashes) {
+      if (wset.insert(h).second) index_[h].insert(worker);
+    }
+  }
+
+  void apply_removed(int64_t worker, const std::vector<uint64_t>& hashes) {
+    std::lock_guard<std::mutex> g(mu_);
+    auto wit = worker_blocks_.find(worker);
+    if (wit == worker_blocks_.end()) return;
+    for (uint64_t h : hashes) {
+      if (wit->second.erase(h)) {
+        auto it = index_.find(h);
+        if (it != index_.end()) {
+          it->second.erase(worker);
+          if (it->second.empty()) index_.erase(it);
+        }
+      }
+    }
+  }
+
+  void remove_worker(int64_t worker) {
+    std::lock_guard<std::mutex> g(mu_);
+    auto wit = worker_blocks_.find(worker);
+    if (wit == worker_blocks_.end()) return;
+    for (uint64_t h : wit->second) {
+      auto it = index_.find(h);
+      if (it != index_.end()) {
+        it->second.erase(worker);
+        if (it->second.empty()) index_.erase(it);
+      }
+    }
+    worker_blocks_.erase(wit);
+  }
+
+  void clear_worker(int64_t worker) { remove_worker(worker); }
+
+  // Longest matched *prefix* (in blocks) per worker for a chained hash
+  // sequence. Mirrors RadixTree::find_matches (radix_tree.rs:225): a worker
+  // only counts while it holds every block so far.
+  std::unordered_map<int64_t, int64_t> find_matches(
+      const std::vector<uint64_t>& seq_hashes) const {
+    std::lock_guard<std::mutex> g(mu_);
+    std::unordered_map<int64_t, int64_t> scores;
+    std::unordered_set<int64_t> alive;
+    bool first = true;
+    for (uint64_t h : seq_hashes) {
+      auto it = index_.find(h);
+      if (it == index_.end()) break;
+      if (first) {
+        for (int64_t w : it->second) { alive.insert(w); }
+        first = false;
+      } else {
+        for (auto ai = alive.begin(); ai != alive.end();) {
+          if (!it->second.count(*ai)) ai = alive.erase(ai);
+          else ++ai;
+        }
+      }
+      if (alive.empty()) break;
+      for (int64_t w : alive) scores[w] += 1;
+    }
+    return scores;
+  }
+
+  int64_t worker_block_count(int64_t worker) const {
+    std::lock_guard<std::mutex> g(mu_);
+    auto it = worker_blocks_.find(worker);
+    return it == worker_blocks_.end() ? 0 : (int64_t)it->second.size();
+  }
+
+  int64_t size() const {
+    std::lock_guard<std::mutex> g(mu_);
+    return (int64_t)index_.size();
+  }
+
+ private:
+  mutable std::mutex mu_;
+  std::unordered_map<uint64_t, std::unordered_set<int64_t>> index_;
+  std::unordered_map<int64_t, std::unordered_set<uint64_t>> worker_blocks_;
+};
+
+PYBIND11_MODULE(_core, m) {
+  m.doc() = "dynamo_amd native C++ control-plane components";
+  m.def("chain_hashes", &chain_hashes, py::arg("tokens"), py::arg("block_size"),
+        py::arg("salt") = 0);
+  m.def("hash_block", &hash_block, py::arg("parent"), py::arg("tokens"));
+  py::class_<KvIndexer>(m, "KvIndexer")
+      .def(py::init<>())
+      .def("apply_stored", &KvIndexer::apply_stored)
+      .def("apply_removed", &KvIndexer::apply_removed)
+      .def("remove_worker", &KvIndexer::remove_worker)
+      .def("clear_worker", &KvIndexer::clear_worker)
+      .def("find_matches", &KvIndexer::find_matches)
+      .def("worker_block_count", &KvIndexer::worker_block_count)
+      .def("size", &KvIndexer::size);
+}

@@ -1,0 +1,47 @@
+#include "hip/hip_runtime.h"
+// MFMA layout probe: computes one 16x16x32 bf16 MFMA with the fragment
+// mappings the engine kernels assume, so a GPU test can verify them against
+// a torch reference (guide §3 G9: asymmetric-input check — a transposed
+// mapping is invisible to symmetric tests).
+#include "common.h"
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+
+namespace {
+
+typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8_t;
+
+__global__ void mfma_probe_kernel(float* __restrict__ D,      // [16,16]
+                                  const short* __restrict__ A,  // [16,32] bf16
+                                  const short* __restrict__ B)  // [32,16] bf16
+{
+  const int l = threadIdx.x;
+  bf16x8_t a, b;
+#pragma unroll
+  for (int i = 0; i < 8; i++) {
+    union { short s; __bf16 h; } ua, ub;
+    ua.s = A[(l % 16) * 32 + 8 * (l / 16) + i];
+    ub.s = B[(8 * (l / 16) + i) * 16 + (l % 16)];
+    a[i] = ua.h;
+    b[i] = ub.h;
+  }
+  f32x4 c{0.f, 0.f, 0.f, 0.f};
+  c = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, c, 0, 0, 0);
+#pragma unroll
+  for (int r = 0; r < 4; r++) D[((l / 16) * 4 + r) * 16 + (l % 16)] = c[r];
+}
+
+}  // namespace
+
+torch::Tensor mfma_probe(torch::Tensor A, torch::Tensor B) {
+  TORCH_CHECK(A.is_cuda() && A.dtype() == torch::kBFloat16);
+  TORCH_CHECK(A.sizes() == torch::IntArrayRef({16, 32}));
+  TORCH_CHECK(B.sizes() == torch::IntArrayRef({32, 16}));
+  auto D = torch::empty({16, 16}, A.options().dtype(torch::kFloat32));
+  auto stream = at::cuda::getCurrentHIPStream();
+ hipLaunchKernelGGL(( mfma_probe_kernel), dim3(1), dim3(64), 0, stream, 
+      D.data_ptr<float>(), (const short*)A.contiguous().data_ptr(),
+      (const short*)B.contiguous().data_ptr());
+  HIP_CHECK_KERNEL();
+  return D;
+}

@@ -1,0 +1,195 @@
+"""Op dispatch: native HIP kernels on GPU, torch reference on CPU.
+
+On a GPU box the HIP extension is REQUIRED — a missing/failed extension
+raises instead of silently falling back to eager PyTorch, so GPU tests can
+never pass on a non-native path.
+"""
+from __future__ import annotations
+
+import torch
+
+from . import torch_ref
+from .torch_ref import make_cos_sin_cache  # re-export
+
+_hip_mod = None
+_hip_err: Exception | None = None
+
+
+def hip():
+    """The native kernel extension; raises loudly if unavailable."""
+    global _hip_mod, _hip_err
+    if _hip_mod is None:
+        try:
+            from dynamo_amd import _hip as m  # built in-tree by setup.py
+            _hip_mod = m
+        except Exception as e:  # pragma: no cover
+            _hip_err = e
+            raise RuntimeError(
+                "dynamo_amd._hip native extension is not available on a GPU "
+                "device path. Build it with `python setup.py build_ext "
+                f"--inplace` (PYTORCH_ROCM_ARCH=gfx950). Original error: {e}"
+            ) from e
+    return _hip_mod
+
+
+def hip_available() -> bool:
+    try:
+        hip()
+        return True
+    except RuntimeError:
+        return False
+
+
+# --------------------------------------------------------------------------
+def rmsnorm(x: torch.Tensor, weight: torch.Tensor, eps: float) -> torch.Tensor:
+    if x.is_cuda:
+        out = torch.empty_like(x)
+        hip().rmsnorm(out, x.contiguous(), weight, eps)
+        return out
+    return torch_ref.rmsnorm(x, weight, eps)
+
+
+def fused_add_rmsnorm(x, residual, weight, eps):
+    """residual <- x + residual (in place); returns normalized residual."""
+    if x.is_cuda:
+        hip().fused_add_rmsnorm(x, residual, weight, eps)
+        return x
+    return torch_ref.fused_add_rmsnorm(x, residual, weight, eps)
+
+
+def rope_inplace(q, k, positions, cos_sin, num_q_heads, num_k_heads, head_dim):
+    if q.is_cuda:
+        hip().rope_inplace(q, k, positions, cos_sin, num_q_heads, num_k_heads,
+                           head_dim)
+        return q, k
+    return torch_ref.rope(q, k, positions, cos_sin, num_q_heads, num_k_heads,
+                          head_dim)
+
+
+def silu_mul(gate_up: torch.Tensor) -> torch.Tensor:
+    if gate_up.is_cuda:
+        d = gate_up.shape[-1] // 2
+        out = torch.empty(gate_up.shape[:-1] + (d,), dtype=gate_up.dtype,
+                          device=gate_up.device)
+        hip().silu_mul(out, gate_up.contiguous())
+        return out
+    return torch_ref.silu_mul(gate_up)
+
+
+def gelu(x: torch.Tensor) -> torch.Tensor:
+    if x.is_cuda:
+        out = torch.empty_like(x)
+        hip().gelu(out, x.contiguous())
+        return out
+    return torch_ref.gelu(x)
+
+
+def kv_cache_append(kcache, vcache, k, v, slot_mapping):
+    if kcache.is_cuda:
+        hip().kv_cache_append(kcache, vcache, k.contiguous(), v.contiguous(),
+                              slot_mapping)
+    else:
+        torch_ref.kv_cache_append(kcache, vcache, k, v, slot_mapping)
+
+
+class DecodeScratch:
+    """Reusable scratch for two-phase flash-decode (sized once per engine)."""
+
+    def __init__(self, max_batch: int, num_q_heads: int, head_dim: int,
+                 max_ctx: int, device):
+        self.chunks = int(hip().paged_decode_num_chunks(max_ctx))
+        self.partial = torch.empty(max_batch, num_q_heads, self.chunks, head_dim,
+                                   dtype=torch.float32, device=device)
+        self.ml = torch.empty(max_batch, num_q_heads, self.chunks, 2,
+                              dtype=torch.float32, device=device)
+
+    def view(self, batch: int):
+        return self.partial[:batch], self.ml[:batch]
+
+
+def paged_attention_decode(q, kcache, vcache, page_table, ctx_lens, scale,
+                           scratch: "DecodeScratch | None" = None):
+    if q.is_cuda:
+        out = torch.empty_like(q)
+        assert scratch is not None, "GPU decode needs a DecodeScratch"
+        partial, ml = scratch.view(q.shape[0])
+        hip().paged_attention_decode(out, q, kcache, vcache, page_table,
+                                     ctx_lens, partial, ml, scale)
+        return out
+    return torch_ref.paged_attention_decode(q, kcache, vcache, page_table,
+                                            ctx_lens, scale)
+
+
+def build_prefill_tiles(seq_q_lens, device):
+    """Host-side tile descriptors for the prefill kernel (64-row q tiles)."""
+    tile_seq, tile_q0 = [], []
+    for s, ql in enumerate(seq_q_lens):
+        for q0 in range(0, int(ql), 64):
+            tile_seq.append(s)
+            tile_q0.append(q0)
+    return (torch.tensor(tile_seq, dtype=torch.int32, device=device),
+            torch.tensor(tile_q0, dtype=torch.int32, device=device))
+
+
+def attention_prefill_paged(q, kcache, vcache, page_table, seq_q_start,
+                            seq_q_len, seq_ctx_len, scale, tiles=None):
+    if q.is_cuda:
+        out = torch.empty_like(q)
+        if tiles is None:
+            tiles = build_prefill_tiles(seq_q_len.tolist(), q.device)
+        tile_seq, tile_q0 = tiles
+        hip().attention_prefill_paged(out, q, kcache, vcache, page_table,
+                                      tile_seq, tile_q0, seq_q_start,
+                                      seq_q_len, seq_ctx_len, scale)
+        return out
+    return torch_ref.attention_prefill_paged(q, kcache, vcache, page_table,
+                                             seq_q_start, seq_q_len,
+                                             seq_ctx_len, scale)
+
+
+def greedy_sample(logits: torch.Tensor) -> torch.Tensor:
+    if logits.is_cuda:
+        out = torch.empty(logits.shape[0], dtype=torch.int32,
+                          device=logits.device)
+        hip().greedy_sample(out, logits.float().contiguous())
+        return out
+    return torch_ref.greedy_sample(logits)
+
+
+def gumbel_sample(logits: torch.Tensor, inv_temp: torch.Tensor,
+                  seed: int) -> torch.Tensor:
+    if logits.is_cuda:
+        out = torch.empty(logits.shape[0], dtype=torch.int32,
+                          device=logits.device)
+        hip().gumbel_sample(out, logits.float().contiguous(), inv_temp, seed)
+        return out
+    # CPU: exact same semantics via torch gumbel noise (not bit-identical)
+    g = -torch.log(-torch.log(torch.rand_like(logits.float())))
+    return (logits.float() * inv_temp.unsqueeze(-1) + g).argmax(-1).to(torch.int32)
+
+
+def gather_pages(staging, cache, page_ids):
+    if cache.is_cuda:
+        hip().gather_pages(staging, cache, page_ids)
+    else:
+        n = page_ids.shape[0]
+        pe = cache[0].numel()
+        staging.view(-1)[: n * pe] = cache[page_ids.long()].reshape(-1)
+
+
+def scatter_pages(staging, cache, page_ids):
+    if cache.is_cuda:
+        hip().scatter_pages(staging, cache, page_ids)
+    else:
+        n = page_ids.shape[0]
+        pe = cache[0].numel()
+        cache[page_ids.long()] = staging.view(-1)[: n * pe].view(n, *cache.shape[1:])
+
+
+def copy_pages(dst_cache, src_cache, pairs):
+    if src_cache.is_cuda:
+        hip().copy_pages(dst_cache, src_cache, pairs)
+    else:
+        src = pairs[:, 0].long()
+        dst = pairs[:, 1].long()
+        dst_cache[dst] = src_cache[src]
