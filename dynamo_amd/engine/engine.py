@@ -1,0 +1,168 @@
+"""LLMEngine: the native continuous-batching engine loop.
+
+One engine per GPU (TP>1: one engine per rank, rank 0 drives scheduling and
+broadcasts — see workers/). Produces per-step outputs, KV events for the
+router (stored/removed block hashes) and ForwardPassMetrics for the planner.
+"""
+from __future__ import annotations
+
+import time
+from dataclasses import dataclass, field
+from typing import Dict, List, Optional
+
+import torch
+
+from dynamo_amd.models.layers import TPContext
+from .config import EngineConfig
+from .kv_cache import KvEvent, PageAllocator
+from .model_runner import ModelRunner
+from .scheduler import Request, ReqState, SamplingParams, Scheduler
+
+
+@dataclass
+class StepOutput:
+    req_id: str
+    new_token: Optional[int]
+    finished: bool
+    finish_reason: Optional[str] = None
+    num_output_tokens: int = 0
+
+
+@dataclass
+class ForwardPassMetrics:
+    """Per-iteration scheduler/engine metrics for the planner (mirrors the
+    reference's ForwardPassMetrics schema,
+    components/src/dynamo/common/forward_pass_metrics.py:153)."""
+    worker_id: str = ""
+    step: int = 0
+    num_running: int = 0
+    num_waiting: int = 0
+    kv_usage: float = 0.0
+    num_tokens_step: int = 0
+    prefill_tokens_step: int = 0
+    decode_tokens_step: int = 0
+    step_time_ms: float = 0.0
+    total_kv_pages: int = 0
+
+
+class LLMEngine:
+    def __init__(self, cfg: EngineConfig, tp: Optional[TPContext] = None,
+                 seed: int = 0):
+        self.cfg = cfg
+        self.runner = ModelRunner(cfg, tp, seed)
+        self.alloc = PageAllocator(self.runner.num_pages, cfg.page_size,
+                                   cfg.enable_prefix_caching)
+        self.scheduler = Scheduler(cfg, self.alloc)
+        self.requests: Dict[str, Request] = {}
+        self.step_count = 0
+        self.kv_events: List[KvEvent] = []
+        self.last_metrics = ForwardPassMetrics()
+        self._held: Dict[str, Request] = {}  # finished but KV retained (disagg)
+
+    # ------------------------------------------------------------------
+    def add_request(self, req_id: str, prompt_tokens: List[int],
+                    sampling: SamplingParams | None = None) -> Request:
+        req = Request(req_id, prompt_tokens, sampling or SamplingParams())
+        self.requests[req_id] = req
+        self.scheduler.add_request(req)
+        return req
+
+    def abort(self, req_id: str):
+        self.scheduler.abort(req_id)
+        self.requests.pop(req_id, None)
+
+    def has_work(self) -> bool:
+        return self.scheduler.has_work()
+
+    # ------------------------------------------------------------------
+    def _check_finish(self, req: Request) -> Optional[str]:
+        sp = req.sampling
+        if len(req.output_tokens) >= sp.max_tokens:
+            return "length"
+        if req.total_len >= self.cfg.max_model_len:
+            return "length"
+        if (not sp.ignore_eos and sp.stop_token_ids
+                and req.output_tokens
+                and req.output_tokens[-1] in sp.stop_token_ids):
+            return "stop"
+        return None
+
+    def step(self) -> List[StepOutput]:
+        t0 = time.monotonic()
+        sched = self.scheduler.schedule()
+        if sched.is_empty:
+            return []
+        self.step_count += 1
+        sampled, sample_reqs = self.runner.execute(sched, self.step_count)
+        sampled = sampled.cpu().tolist() if len(sample_reqs) else []
+
+        # advance computed counts
+        for ss in sched.seqs:
+            ss.req.num_computed += ss.n_new
+
+        outputs: List[StepOutput] = []
+        for tok, req in zip(sampled, sample_reqs):
+            req.output_tokens.append(int(tok))
+            if req.first_token_time is None:
+                req.first_token_time = time.monotonic()
+            reason = self._check_finish(req)
+            finished = reason is not None
+            if finished:
+                hold = req.hold_kv if hasattr(req, "hold_kv") else False
+                if hold:
+                    # keep pages alive for disagg transfer
+                    req.state = ReqState.FINISHED
+                    req.finish_reason = reason
+                    req.finish_time = time.monotonic()
+                    self.scheduler.running.remove(req)
+                    self._held[req.req_id] = req
+                else:
+                    self._finish(req, reason)
+            outputs.append(StepOutput(req.req_id, int(tok), finished, reason,
+                                      len(req.output_tokens)))
+
+        # prefix-cache hash registration + KV events
+        if self.cfg.kv_events or self.cfg.enable_prefix_caching:
+            for ss in sched.seqs:
+                r = ss.req
+                if r.kv is not None:
+                    r.kv.commit_full_pages(r.all_tokens, r.num_computed)
+        self.kv_events.extend(self.alloc.drain_events())
+
+        dt = (time.monotonic() - t0) * 1000
+        self.last_metrics = ForwardPassMetrics(
+            step=self.step_count,
+            num_running=self.scheduler.num_running(),
+            num_waiting=self.scheduler.num_waiting(),
+            kv_usage=self.alloc.usage,
+            num_tokens_step=sched.num_tokens,
+            prefill_tokens_step=sum(s.n_new for s in sched.prefills),
+            decode_tokens_step=len(sched.decodes),
+            step_time_ms=dt,
+            total_kv_pages=self.alloc.num_pages,
+        )
+        return outputs
+
+    def _finish(self, req: Request, reason: str):
+        # commit full pages before release so prefix cache retains them
+        if req.kv is not None:
+            req.kv.commit_full_pages(req.all_tokens, req.num_computed)
+        self.scheduler.finish(req, reason)
+
+    def release_held(self, req_id: str):
+        """Release KV of a finished-but-held request (disagg prefill side)."""
+        req = self._held.pop(req_id, None)
+        if req is not None and req.kv is not None:
+            req.kv.commit_full_pages(req.all_tokens, req.num_computed)
+            req.kv.release()
+            req.kv = None
+
+    def drain_kv_events(self) -> List[KvEvent]:
+        ev, self.kv_events = self.kv_events, []
+        return ev
+
+    def clear_kv(self):
+        """clear_kv_blocks endpoint parity (reference: vllm/main.py
+        clear_kv_blocks)."""
+        self.alloc.clear()
+        self.kv_events.extend(self.alloc.drain_events())

@@ -1,0 +1,225 @@
+"""Paged KV cache: page allocator with prefix caching + the device pool.
+
+MI355X-native design decisions:
+  - pages are 64 tokens by default (one page = one LDS tile of the prefill
+    kernel and a natural xGMI transfer granule);
+  - the whole pool is ONE hipMalloc allocation ([L, 2, P, Hkv, ps, hd]) so a
+    single hipIpc handle exports every layer to a peer decode worker;
+  - 288 GB HBM3E means the pool is sized generously (gpu_mem_fraction of
+    whatever is left after weights).
+
+Roles mirrored from the reference: block pool + prefix reuse (vLLM-side
+behavior the reference orchestrates), KV event emission for the KV-aware
+router (ai-dynamo/dynamo lib/kv-router/src/protocols.rs:1000-1355
+KvCacheEvent Stored/Removed/Cleared), hash chain parity via dynamo_amd._core
+(lib/kv-hashing/src/compute.rs:15-35).
+"""
+from __future__ import annotations
+
+from collections import OrderedDict, deque
+from dataclasses import dataclass, field
+from typing import Dict, List, Optional
+
+import torch
+
+from dynamo_amd import _core
+
+
+@dataclass
+class KvEvent:
+    kind: str                  # "stored" | "removed" | "cleared"
+    hashes: List[int] = field(default_factory=list)
+    parent: Optional[int] = None
+
+
+class PageAllocator:
+    """Refcounted page allocator with hash-addressed prefix cache.
+
+    Pages with refcount 0 that carry a block hash are kept in an LRU of
+    evictable cached pages; allocation evicts from it when the free list is
+    empty (emitting a `removed` KV event)."""
+
+    def __init__(self, num_pages: int, page_size: int, enable_prefix: bool = True):
+        self.num_pages = num_pages
+        self.page_size = page_size
+        self.enable_prefix = enable_prefix
+        self.free: deque[int] = deque(range(num_pages))
+        self.ref = [0] * num_pages
+        self.page_hash: List[Optional[int]] = [None] * num_pages
+        self.hash_to_page: Dict[int, int] = {}
+        self.evictable: "OrderedDict[int, None]" = OrderedDict()  # page -> None
+        self.events: List[KvEvent] = []
+
+    # -- stats ------------------------------------------------------------
+    @property
+    def num_free(self) -> int:
+        return len(self.free) + len(self.evictable)
+
+    @property
+    def usage(self) -> float:
+        return 1.0 - self.num_free / max(1, self.num_pages)
+
+    # -- allocation -------------------------------------------------------
+    def alloc(self) -> int:
+        if self.free:
+            pid = self.free.popleft()
+        elif self.evictable:
+            pid, _ = self.evictable.popitem(last=False)  # LRU evict
+            h = self.page_hash[pid]
+            if h is not None:
+                del self.hash_to_page[h]
+                self.page_hash[pid] = None
+                self.events.append(KvEvent("removed", [h]))
+        else:
+            raise MemoryError("KV pool exhausted")
+        self.ref[pid] = 1
+        return pid
+
+    def incref(self, pid: int):
+        if self.ref[pid] == 0 and pid in self.evictable:
+            del self.evictable[pid]
+        self.ref[pid] += 1
+
+    def decref(self, pid: int):
+        self.ref[pid] -= 1
+        assert self.ref[pid] >= 0
+        if self.ref[pid] == 0:
+            if self.page_hash[pid] is not None and self.enable_prefix:
+                self.evictable[pid] = None  # retain for prefix reuse
+            else:
+                h = self.page_hash[pid]
+                if h is not None:
+                    del self.hash_to_page[h]
+                    self.page_hash[pid] = None
+                    self.events.append(KvEvent("removed", [h]))
+                self.free.append(pid)
+
+    # -- prefix cache -----------------------------------------------------
+    def lookup(self, h: int) -> Optional[int]:
+        """Find a cached page by hash and take a reference."""
+        if not self.enable_prefix:
+            return None
+        pid = self.hash_to_page.get(h)
+        if pid is None:
+            return None
+        self.incref(pid)
+        return pid
+
+    def register_hash(self, pid: int, h: int, parent: Optional[int]):
+        """Mark a now-full page as carrying block hash `h` (emits `stored`)."""
+        if self.page_hash[pid] is not None:
+            return
+        if h in self.hash_to_page:
+            # another page already holds this content; keep this one unhashed
+            return
+        self.page_hash[pid] = h
+        self.hash_to_page[h] = pid
+        self.events.append(KvEvent("stored", [h], parent))
+
+    def drain_events(self) -> List[KvEvent]:
+        ev, self.events = self.events, []
+        return ev
+
+    def clear(self):
+        self.__init__(self.num_pages, self.page_size, self.enable_prefix)
+        self.events.append(KvEvent("cleared"))
+
+
+class SequenceKV:
+    """Per-sequence page table + hash chain bookkeeping."""
+
+    def __init__(self, alloc: PageAllocator, salt: int = 0):
+        self.alloc = alloc
+        self.salt = salt
+        self.pages: List[int] = []
+        self.num_cached_tokens = 0  # tokens whose KV was reused from cache
+
+    def match_prefix(self, tokens: List[int]) -> int:
+        """Reuse cached full pages covering a prefix of `tokens`.
+
+        Returns the number of reused tokens. Never reuses the *entire*
+        prompt (at least one token must be recomputed to produce logits)."""
+        ps = self.alloc.page_size
+        hashes = _core.chain_hashes(tokens, ps, self.salt)
+        reused = 0
+        for i, h in enumerate(hashes):
+            # keep at least one token to compute
+            if (i + 1) * ps >= len(tokens):
+                break
+            pid = self.alloc.lookup(h)
+            if pid is None:
+                break
+            self.pages.append(pid)
+            reused = (i + 1) * ps
+        self.num_cached_tokens = reused
+        return reused
+
+    def ensure_capacity(self, num_tokens: int) -> int:
+        """Allocate pages so `num_tokens` tokens fit. Returns pages added."""
+        ps = self.alloc.page_size
+        need = (num_tokens + ps - 1) // ps
+        added = 0
+        while len(self.pages) < need:
+            self.pages.append(self.alloc.alloc())
+            added += 1
+        return added
+
+    def commit_full_pages(self, tokens: List[int], num_computed: int):
+        """Register hashes for pages that just became full (KV written for
+        the first `num_computed` tokens of `tokens`)."""
+        ps = self.alloc.page_size
+        full = num_computed // ps
+        hashes = _core.chain_hashes(tokens[: full * ps], ps, self.salt)
+        parent = None
+        for i in range(full):
+            self.alloc.register_hash(self.pages[i], hashes[i], parent)
+            parent = hashes[i]
+
+    def release(self):
+        for pid in self.pages:
+            self.alloc.decref(pid)
+        self.pages = []
+
+
+class KVCachePool:
+    """Device memory for the paged KV cache of all layers.
+
+    GPU: one hipMalloc allocation (ipc-exportable); CPU: a torch tensor.
+    Layout [L, 2, P, Hkv_local, ps, hd] — [l, 0] is K, [l, 1] is V.
+    """
+
+    def __init__(self, num_layers: int, num_pages: int, num_kv_heads: int,
+                 page_size: int, head_dim: int, device: str,
+                 dtype=torch.bfloat16):
+        self.shape = (num_layers, 2, num_pages, num_kv_heads, page_size, head_dim)
+        self.num_pages = num_pages
+        self.page_size = page_size
+        self.device = torch.device(device)
+        self.dtype = dtype
+        numel = 1
+        for s in self.shape:
+            numel *= s
+        nbytes = numel * torch.tensor([], dtype=dtype).element_size()
+        if self.device.type == "cuda":
+            from dynamo_amd import ops
+            self._raw = ops.hip().ipc_alloc(nbytes, self.device.index or 0)
+            self.buffer = self._raw.view(dtype).view(self.shape)
+        else:
+            self.buffer = torch.zeros(self.shape, dtype=dtype, device=device)
+        self.nbytes = nbytes
+
+    def kcache(self, layer: int) -> torch.Tensor:
+        return self.buffer[layer, 0]
+
+    def vcache(self, layer: int) -> torch.Tensor:
+        return self.buffer[layer, 1]
+
+    def ipc_export(self) -> bytes:
+        from dynamo_amd import ops
+        return ops.hip().ipc_export(self._raw)
+
+    # page_bytes of ONE layer's K (or V) page — transfer granularity
+    @property
+    def layer_page_numel(self) -> int:
+        _, _, _, hkv, ps, hd = self.shape
+        return hkv * ps * hd

@@ -1,0 +1,147 @@
+"""Model runner: turns a SchedulerOutput into one forward pass + sampling.
+
+Builds the flattened token batch (decode tokens first, then prefill chunks),
+the attention metadata (page tables, slot mappings, varlen descriptors) and
+runs the model. hipGraph capture for decode-only steps is handled by
+GraphRunner (engine/graphs.py) layered on top of this.
+"""
+from __future__ import annotations
+
+import time
+from typing import List, Optional, Tuple
+
+import torch
+
+from dynamo_amd import ops
+from dynamo_amd.models.layers import AttnMetadata, TPContext
+from dynamo_amd.models.registry import build_model
+from .config import EngineConfig
+from .kv_cache import KVCachePool
+from .scheduler import SchedulerOutput, ScheduledSeq
+
+
+class ModelRunner:
+    def __init__(self, cfg: EngineConfig, tp: Optional[TPContext] = None,
+                 seed: int = 0):
+        self.cfg = cfg
+        self.device = torch.device(cfg.device)
+        self.dtype = cfg.torch_dtype
+        self.tp = tp or TPContext(cfg.tp_size, cfg.tp_rank)
+        m = cfg.model
+        self.model = build_model(m, self.device, self.dtype, self.tp, seed)
+        self.hkv_local = max(1, m.num_kv_heads // self.tp.size)
+        self.hq_local = m.num_q_heads // self.tp.size
+
+        num_pages = cfg.kv_pool_pages or self._auto_pages()
+        self.kv_pool = KVCachePool(m.num_layers, num_pages, self.hkv_local,
+                                   cfg.page_size, m.head_dim, cfg.device,
+                                   self.dtype)
+        self.num_pages = num_pages
+        self.max_pages_per_seq = (cfg.max_model_len + cfg.page_size - 1) // cfg.page_size
+        self.decode_scratch = None
+        if self.device.type == "cuda":
+            self.decode_scratch = ops.DecodeScratch(
+                cfg.max_num_seqs, self.hq_local, m.head_dim,
+                cfg.max_model_len, self.device)
+
+    def _auto_pages(self) -> int:
+        m = self.cfg.model
+        page_bytes = (m.num_layers * 2 * self.hkv_local * self.cfg.page_size
+                      * m.head_dim * 2)
+        if self.device.type == "cuda":
+            free, _total = torch.cuda.mem_get_info(self.device)
+            budget = int(free * self.cfg.gpu_mem_fraction) - (2 << 30)
+        else:
+            budget = 1 << 30  # 1 GiB for CPU tests
+        return max(16, budget // page_bytes)
+
+    # ------------------------------------------------------------------
+    def prepare(self, sched: SchedulerOutput) -> Tuple[torch.Tensor, AttnMetadata]:
+        cfg = self.cfg
+        ps = cfg.page_size
+        dev = self.device
+        tokens: List[int] = []
+        positions: List[int] = []
+        slots: List[int] = []
+        logits_rows: List[int] = []
+
+        # ---- decode part ----
+        nd = len(sched.decodes)
+        dec_tables, dec_ctx = [], []
+        for i, ss in enumerate(sched.decodes):
+            r = ss.req
+            pos = r.num_computed
+            tokens.append(r.all_tokens[pos])
+            positions.append(pos)
+            pages = r.kv.pages
+            slots.append(pages[pos // ps] * ps + pos % ps)
+            dec_tables.append(pages)
+            dec_ctx.append(pos + 1)
+            logits_rows.append(i)
+
+        # ---- prefill part ----
+        pf_tables, q_start, q_len, ctx_len = [], [], [], []
+        qpos = 0
+        for ss in sched.prefills:
+            r = ss.req
+            nc, n = r.num_computed, ss.n_new
+            seq_tokens = r.all_tokens
+            for j in range(nc, nc + n):
+                tokens.append(seq_tokens[j])
+                positions.append(j)
+                slots.append(r.kv.pages[j // ps] * ps + j % ps)
+            pf_tables.append(r.kv.pages)
+            q_start.append(qpos)
+            q_len.append(n)
+            ctx_len.append(nc + n)
+            if ss.sample:
+                logits_rows.append(nd + qpos + n - 1)
+            qpos += n
+
+        def table_tensor(tables):
+            if not tables:
+                return None
+            maxp = max(len(t) for t in tables)
+            out = torch.zeros(len(tables), max(1, maxp), dtype=torch.int32)
+            for i, t in enumerate(tables):
+                out[i, :len(t)] = torch.tensor(t, dtype=torch.int32)
+            return out.to(dev, non_blocking=True)
+
+        meta = AttnMetadata(
+            slot_mapping=torch.tensor(slots, dtype=torch.int64).to(dev, non_blocking=True),
+            positions=torch.tensor(positions, dtype=torch.int32).to(dev, non_blocking=True),
+            num_decode=nd,
+            decode_page_table=table_tensor(dec_tables),
+            decode_ctx_lens=(torch.tensor(dec_ctx, dtype=torch.int32).to(dev, non_blocking=True)
+                             if dec_ctx else None),
+            decode_scratch=self.decode_scratch,
+            num_prefill_tokens=qpos,
+            prefill_page_table=table_tensor(pf_tables),
+            seq_q_start=(torch.tensor(q_start, dtype=torch.int32).to(dev, non_blocking=True)
+                         if q_start else None),
+            seq_q_len=(torch.tensor(q_len, dtype=torch.int32).to(dev, non_blocking=True)
+                       if q_len else None),
+            seq_ctx_len=(torch.tensor(ctx_len, dtype=torch.int32).to(dev, non_blocking=True)
+                         if ctx_len else None),
+            logits_rows=torch.tensor(logits_rows, dtype=torch.int64).to(dev, non_blocking=True),
+        )
+        if qpos and dev.type == "cuda":
+            meta.prefill_tiles = ops.build_prefill_tiles(q_len, dev)
+        input_ids = torch.tensor(tokens, dtype=torch.int32).to(dev, non_blocking=True)
+        return input_ids, meta
+
+    # ------------------------------------------------------------------
+    @torch.inference_mode()
+    def execute(self, sched: SchedulerOutput, step_seed: int = 0):
+        """Returns (sampled_tokens int32 [n_sample] on device, sample_reqs)."""
+        input_ids, meta = self.prepare(sched)
+        hidden = self.model.forward(input_ids, self.kv_pool, meta)
+        sample_seqs = [s for s in sched.decodes if s.sample] + \
+                      [s for s in sched.prefills if s.sample]
+        if not sample_seqs:
+            return torch.empty(0, dtype=torch.int32), []
+        rows = hidden[meta.logits_rows]
+        logits = self.model.compute_logits(rows)
+        from .sampling import sample_tokens
+        sampled = sample_tokens(logits, [s.req for s in sample_seqs], step_seed)
+        return sampled, [s.req for s in sample_seqs]

@@ -1,0 +1,54 @@
+"""Batch sampling: greedy (native HIP argmax) and temperature sampling via
+the native Gumbel-max kernel; top-k/top-p apply a torch-side logit filter
+first (filtered Gumbel-max samples the renormalized distribution exactly)."""
+from __future__ import annotations
+
+from typing import List
+
+import torch
+
+from dynamo_amd import ops
+from .scheduler import Request
+
+
+def _filter_topk_topp(logits: torch.Tensor, top_k: int, top_p: float):
+    if top_k > 0 and top_k < logits.shape[-1]:
+        kth = torch.topk(logits, top_k, dim=-1).values[..., -1:]
+        logits = logits.masked_fill(logits < kth, float("-inf"))
+    if top_p < 1.0:
+        sorted_logits, idx = torch.sort(logits, descending=True, dim=-1)
+        probs = torch.softmax(sorted_logits, dim=-1)
+        cum = probs.cumsum(-1)
+        # keep tokens until cumulative prob exceeds top_p (always keep first)
+        drop = cum - probs > top_p
+        sorted_logits = sorted_logits.masked_fill(drop, float("-inf"))
+        logits = torch.full_like(logits, float("-inf")).scatter(
+            -1, idx, sorted_logits)
+    return logits
+
+
+def sample_tokens(logits: torch.Tensor, reqs: List[Request],
+                  step_seed: int) -> torch.Tensor:
+    """logits [n, V] fp32 -> token ids [n] int32."""
+    n = logits.shape[0]
+    assert n == len(reqs)
+    temps = [r.sampling.temperature for r in reqs]
+    if all(t == 0.0 for t in temps):
+        return ops.greedy_sample(logits)
+
+    greedy_mask = torch.tensor([t == 0.0 for t in temps], device=logits.device)
+    inv_t = torch.tensor([1.0 / t if t > 0 else 1.0 for t in temps],
+                         dtype=torch.float32, device=logits.device)
+    filt = logits
+    if any(r.sampling.top_k > 0 or r.sampling.top_p < 1.0 for r in reqs):
+        # per-request filters; batch-apply with worst-case params per row
+        rows = []
+        for i, r in enumerate(reqs):
+            rows.append(_filter_topk_topp(logits[i:i + 1], r.sampling.top_k,
+                                          r.sampling.top_p))
+        filt = torch.cat(rows, 0)
+    sampled = ops.gumbel_sample(filt, inv_t, step_seed)
+    if greedy_mask.any():
+        greedy = ops.greedy_sample(logits)
+        sampled = torch.where(greedy_mask, greedy, sampled)
+    return sampled

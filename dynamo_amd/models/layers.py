@@ -1,0 +1,131 @@
+"""Building blocks shared by the native models.
+
+GEMMs go through torch.nn.functional.linear (hipBLASLt on ROCm — the
+library path for plain GEMMs per the MI355X design rules); everything fused
+(norms, rope, attention, activation, sampling) is our HIP kernels via
+dynamo_amd.ops.
+"""
+from __future__ import annotations
+
+from dataclasses import dataclass, field
+from typing import List, Optional
+
+import torch
+import torch.nn.functional as F
+
+from dynamo_amd import ops
+
+
+@dataclass
+class AttnMetadata:
+    """Everything the attention layers need for one engine step.
+
+    Token order in the flattened batch: decode tokens first (one per decode
+    sequence), then prefill chunks back to back.
+    """
+    slot_mapping: torch.Tensor          # [T] int64
+    positions: torch.Tensor             # [T] int32
+    # decode part
+    num_decode: int = 0
+    decode_page_table: Optional[torch.Tensor] = None   # [Bd, maxp] int32
+    decode_ctx_lens: Optional[torch.Tensor] = None     # [Bd] int32
+    decode_scratch: Optional["ops.DecodeScratch"] = None
+    # prefill part
+    num_prefill_tokens: int = 0
+    prefill_page_table: Optional[torch.Tensor] = None  # [Bp, maxp] int32
+    seq_q_start: Optional[torch.Tensor] = None         # [Bp] int32
+    seq_q_len: Optional[torch.Tensor] = None
+    seq_ctx_len: Optional[torch.Tensor] = None
+    prefill_tiles: Optional[tuple] = None
+    # sampling
+    logits_rows: Optional[torch.Tensor] = None         # [Bs] int64 rows to sample
+
+
+def linear(x: torch.Tensor, w: torch.Tensor) -> torch.Tensor:
+    return F.linear(x, w)
+
+
+class TPContext:
+    """Tensor-parallel context: rank/world + the RCCL process group."""
+
+    def __init__(self, tp_size: int = 1, tp_rank: int = 0, group=None):
+        self.size = tp_size
+        self.rank = tp_rank
+        self.group = group
+
+    def all_reduce(self, x: torch.Tensor) -> torch.Tensor:
+        if self.size > 1:
+            torch.distributed.all_reduce(x, group=self.group)
+        return x
+
+
+_DEFAULT_TP = TPContext()
+
+
+def init_weight(shape, device, dtype, std=0.02, generator=None):
+    w = torch.empty(shape, device=device, dtype=dtype)
+    with torch.no_grad():
+        w.normal_(0.0, std, generator=generator)
+    return w
+
+
+class Attention(torch.nn.Module):
+    """GQA attention over the paged KV cache (native HIP kernels)."""
+
+    def __init__(self, cfg, layer_idx: int, tp: TPContext, device, dtype):
+        super().__init__()
+        self.layer_idx = layer_idx
+        self.tp = tp
+        self.hq = cfg.num_q_heads // tp.size
+        self.hkv = max(1, cfg.num_kv_heads // tp.size)
+        self.hd = cfg.head_dim
+        self.scale = self.hd ** -0.5
+        D = cfg.hidden_size
+        qkv_out = (self.hq + 2 * self.hkv) * self.hd
+        self.wqkv = init_weight((qkv_out, D), device, dtype)
+        self.wo = init_weight((D, self.hq * self.hd), device, dtype)
+
+    def forward(self, x, cos_sin, kcache, vcache, meta: AttnMetadata):
+        T = x.shape[0]
+        qkv = linear(x, self.wqkv)
+        q, k, v = qkv.split([self.hq * self.hd, self.hkv * self.hd,
+                             self.hkv * self.hd], dim=-1)
+        q = q.contiguous()
+        k = k.contiguous()
+        q, k = ops.rope_inplace(q, k, meta.positions, cos_sin, self.hq,
+                                self.hkv, self.hd)
+        kh = k.view(T, self.hkv, self.hd)
+        vh = v.view(T, self.hkv, self.hd)
+        ops.kv_cache_append(kcache, vcache, kh, vh, meta.slot_mapping)
+
+        qh = q.view(T, self.hq, self.hd)
+        out = torch.empty_like(qh)
+        nd = meta.num_decode
+        if nd:
+            out[:nd] = ops.paged_attention_decode(
+                qh[:nd], kcache, vcache, meta.decode_page_table,
+                meta.decode_ctx_lens, self.scale, meta.decode_scratch)
+        if meta.num_prefill_tokens:
+            out[nd:] = ops.attention_prefill_paged(
+                qh[nd:].contiguous(), kcache, vcache, meta.prefill_page_table,
+                meta.seq_q_start, meta.seq_q_len, meta.seq_ctx_len, self.scale,
+                meta.prefill_tiles)
+        o = linear(out.view(T, self.hq * self.hd), self.wo)
+        return self.tp.all_reduce(o)
+
+
+class SwiGLUMLP(torch.nn.Module):
+    def __init__(self, cfg, tp: TPContext, device, dtype):
+        super().__init__()
+        self.tp = tp
+        D = cfg.hidden_size
+        I = cfg.intermediate_size // tp.size
+        self.I = I
+        self.w_gate_up = init_weight((2 * I, D), device, dtype)
+        self.w_down = init_weight((D, I), device, dtype)
+
+    def forward(self, x):
+        gu = linear(x, self.w_gate_up)
+        # silu_mul expects [., 2I] with gate then up
+        act = ops.silu_mul(gu)
+        return self.tp.all_reduce(linear(act, self.w_down))

@@ -1,0 +1,113 @@
+"""Mixtral-style MoE causal LM (config #5).
+
+Top-k gating + expert MLPs. The expert compute is segment-batched per
+expert (sorted token dispatch); expert-parallel all-to-all over RCCL is
+layered in the parallel package. A HIP grouped-GEMM fast path replaces the
+per-expert loop when available (dynamo_amd.ops.moe)."""
+from __future__ import annotations
+
+import torch
+import torch.nn.functional as F
+
+from dynamo_amd import ops
+from .layers import AttnMetadata, Attention, TPContext, init_weight, linear
+
+
+class MoEMLP(torch.nn.Module):
+    def __init__(self, cfg, tp: TPContext, device, dtype):
+        super().__init__()
+        self.tp = tp
+        self.E = cfg.num_experts
+        self.topk = cfg.num_experts_per_tok
+        D = cfg.hidden_size
+        I = cfg.intermediate_size // tp.size
+        self.I = I
+        self.router = init_weight((self.E, D), device, dtype)
+        # fused per-expert weights: [E, 2I, D] and [E, D, I]
+        self.w_gate_up = init_weight((self.E, 2 * I, D), device, dtype)
+        self.w_down = init_weight((self.E, D, I), device, dtype)
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        T, D = x.shape
+        logits = linear(x, self.router).float()            # [T, E]
+        weights = torch.softmax(logits, dim=-1)
+        topw, topi = torch.topk(weights, self.topk, dim=-1)  # [T, k]
+        topw = (topw / topw.sum(-1, keepdim=True)).to(x.dtype)
+
+        out = torch.zeros_like(x)
+        flat_expert = topi.reshape(-1)                     # [T*k]
+        flat_token = (torch.arange(T, device=x.device)
+                      .repeat_interleave(self.topk))       # [T*k]
+        # sort by expert -> contiguous segments
+        order = torch.argsort(flat_expert, stable=True)
+        seg_expert = flat_expert[order]
+        seg_token = flat_token[order]
+        counts = torch.bincount(seg_expert, minlength=self.E)
+        starts = torch.cumsum(counts, 0) - counts
+        xg = x[seg_token]                                  # [T*k, D]
+        yg = torch.empty_like(xg)
+        counts_l = counts.tolist()
+        starts_l = starts.tolist()
+        for e in range(self.E):
+            n = counts_l[e]
+            if n == 0:
+                continue
+            s = starts_l[e]
+            gu = linear(xg[s:s + n], self.w_gate_up[e])
+            yg[s:s + n] = linear(ops.silu_mul(gu), self.w_down[e])
+        w = topw.reshape(-1)[order].unsqueeze(-1)
+        out.index_add_(0, seg_token, (yg * w).to(x.dtype))
+        return self.tp.all_reduce(out)
+
+
+class MixtralDecoderLayer(torch.nn.Module):
+    def __init__(self, cfg, layer_idx, tp, device, dtype):
+        super().__init__()
+        self.attn = Attention(cfg, layer_idx, tp, device, dtype)
+        self.moe = MoEMLP(cfg, tp, device, dtype)
+        self.input_norm_w = torch.ones(cfg.hidden_size, device=device, dtype=dtype)
+        self.post_norm_w = torch.ones(cfg.hidden_size, device=device, dtype=dtype)
+        self.eps = cfg.rms_eps
+
+    def forward(self, x, residual, cos_sin, kcache, vcache, meta):
+        if residual is None:
+            residual = x.clone()
+            x = ops.rmsnorm(x, self.input_norm_w, self.eps)
+        else:
+            x = ops.fused_add_rmsnorm(x, residual, self.input_norm_w, self.eps)
+        x = self.attn.forward(x, cos_sin, kcache, vcache, meta)
+        x = ops.fused_add_rmsnorm(x, residual, self.post_norm_w, self.eps)
+        x = self.moe.forward(x)
+        return x, residual
+
+
+class MixtralForCausalLM(torch.nn.Module):
+    def __init__(self, cfg, device="cpu", dtype=torch.bfloat16,
+                 tp: TPContext | None = None, seed: int = 0):
+        super().__init__()
+        self.cfg = cfg
+        self.tp = tp or TPContext()
+        torch.manual_seed(seed)
+        self.embed = init_weight((cfg.vocab_size, cfg.hidden_size), device, dtype)
+        self.layers = torch.nn.ModuleList([
+            MixtralDecoderLayer(cfg, i, self.tp, device, dtype)
+            for i in range(cfg.num_layers)
+        ])
+        self.final_norm_w = torch.ones(cfg.hidden_size, device=device, dtype=dtype)
+        self.lm_head = init_weight((cfg.vocab_size, cfg.hidden_size), device, dtype)
+        from dynamo_amd.ops import torch_ref
+        self.cos_sin = torch_ref.make_cos_sin_cache(
+            cfg.max_position, cfg.head_dim, cfg.rope_theta, device=device)
+
+    def forward(self, input_ids, kv_pool, meta: AttnMetadata):
+        x = F.embedding(input_ids.long(), self.embed)
+        residual = None
+        for i, layer in enumerate(self.layers):
+            x, residual = layer.forward(x, residual, self.cos_sin,
+                                        kv_pool.kcache(i), kv_pool.vcache(i),
+                                        meta)
+        x = ops.fused_add_rmsnorm(x, residual, self.final_norm_w, self.cfg.rms_eps)
+        return x
+
+    def compute_logits(self, hidden):
+        return linear(hidden, self.lm_head).float()
