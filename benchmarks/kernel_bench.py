@@ -3,11 +3,14 @@
 Prints per-kernel achieved bandwidth / TFLOPs vs the hardware ceilings
 (HBM ~6.3 TB/s achievable, bf16 MFMA 2.5 PF dense)."""
 import argparse
+import os
+import sys
 import time
 
 import torch
 
-from dynamo_amd import ops
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+from dynamo_amd import ops  # noqa: E402
 
 
 def timeit(fn, iters=20, warmup=5):
