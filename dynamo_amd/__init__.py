@@ -7,4 +7,6 @@ HIP kernels (paged attention, RMSNorm, RoPE, sampling, MoE), SLA planner,
 and a tiered KV-block manager. See SURVEY.md for the blueprint.
 """
 
+import torch as _torch  # noqa: F401  (extensions link against torch libs)
+
 __version__ = "0.1.0"

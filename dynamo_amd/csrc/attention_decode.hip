@@ -4,9 +4,9 @@
 // workgroups, each computing an online-softmax partial (m, l, acc) for the
 // GQA group of query heads sharing that kv head; phase 2 merges chunks.
 // Decode attention is HBM-bound (streaming the KV cache once); the kernel is
-// laid out for coalesced 32 B/lane KV reads: a wave covers 8 tokens x 8
-// dim-slices (16 dims each), so consecutive lanes read consecutive 16 B
-// chunks of a page row.
+// laid out for coalesced KV reads: a wave covers (64/DP) tokens x DP
+// dim-slices, so consecutive lanes read consecutive 16 B chunks of a page
+// row. DP=8 for GQA group <= 4; DP=16 for group 8 (register pressure).
 //
 // Capability parity: the reference (ai-dynamo/dynamo) delegates paged
 // attention to vLLM/TRT-LLM; this is the native CDNA4 engine kernel.
@@ -21,8 +21,9 @@ constexpr int kChunk = 512;        // context tokens per workgroup
 constexpr int kSlab = kChunk / 4;  // tokens per wave (128)
 constexpr float kNegInf = -1e30f;
 
-// G = GQA group size (Hq / Hkv), templated so the accumulator unrolls.
-template <int G>
+// G = GQA group size (Hq / Hkv); DP = lanes per token. G*hd/DP accumulator
+// floats per lane — keep <= 64 to stay under the occupancy cliff.
+template <int G, int DP>
 __global__ __launch_bounds__(kBlock) void paged_decode_phase1(
     float* __restrict__ partial,        // [B, Hq, C, hd] fp32
     float* __restrict__ ml,             // [B, Hq, C, 2] fp32 (m, l)
@@ -33,6 +34,9 @@ __global__ __launch_bounds__(kBlock) void paged_decode_phase1(
     const int32_t* __restrict__ page_table,  // [B, max_pages]
     const int32_t* __restrict__ ctx_lens,    // [B]
     float scale, int B, int Hkv, int C, int max_pages, int log2_ps, int hd) {
+  constexpr int ND = 128 / DP;       // dims per lane
+  constexpr int NV8 = ND / 8;        // short8 loads per row slice
+  constexpr int TS = 64 / DP;        // tokens per wave step
   const int b = blockIdx.x;
   const int h = blockIdx.y;   // kv head
   const int c = blockIdx.z;   // context chunk
@@ -41,18 +45,17 @@ __global__ __launch_bounds__(kBlock) void paged_decode_phase1(
   const int chunk_start = c * kChunk;
   const int lane = threadIdx.x & 63;
   const int wid = threadIdx.x >> 6;
-  const int ts = lane >> 3;       // token slot within 8-token step
-  const int dp = lane & 7;        // dim slice: dims [dp*16, dp*16+16)
+  const int ts = lane / DP;       // token slot within step
+  const int dp = lane % DP;       // dim slice: dims [dp*ND, dp*ND+ND)
   const int ps = 1 << log2_ps;
 
-  // LDS: q tile [G][hd] fp32 (pre-scaled), + merge scratch [4][G][hd+2]
+  // LDS: q tile [G][hd] fp32 (pre-scaled) + merge scratch [4][G][hd+2]
   extern __shared__ float lds[];
   float* q_lds = lds;                       // G * hd
   float* merge = lds + G * hd;              // 4 * G * (hd + 2)
 
   if (chunk_start >= ctx) {
     if (C > 1) {
-      // mark empty chunk for phase 2
       for (int i = threadIdx.x; i < G; i += kBlock) {
         const int qh = h * G + i;
         float* mlp = ml + (((int64_t)b * Hq + qh) * C + c) * 2;
@@ -68,29 +71,28 @@ __global__ __launch_bounds__(kBlock) void paged_decode_phase1(
   }
   __syncthreads();
 
-  // preload this lane's q slice for each group head: q[g][dp*16 .. +16)
-  float qreg[G][16];
+  // preload this lane's q slice for each group head: q[g][dp*ND .. +ND)
+  float qreg[G][ND];
 #pragma unroll
   for (int g = 0; g < G; g++)
 #pragma unroll
-    for (int i = 0; i < 16; i++) qreg[g][i] = q_lds[g * hd + dp * 16 + i];
+    for (int i = 0; i < ND; i++) qreg[g][i] = q_lds[g * hd + dp * ND + i];
 
-  float m[G], l[G], acc[G][16];
+  float m[G], l[G], acc[G][ND];
 #pragma unroll
   for (int g = 0; g < G; g++) {
     m[g] = kNegInf; l[g] = 0.f;
 #pragma unroll
-    for (int i = 0; i < 16; i++) acc[g][i] = 0.f;
+    for (int i = 0; i < ND; i++) acc[g][i] = 0.f;
   }
 
   const int slab_start = chunk_start + wid * kSlab;
   const int slab_end = min(slab_start + kSlab, ctx);
   const int32_t* pt = page_table + (int64_t)b * max_pages;
 
-  for (int step = slab_start; step < slab_end; step += 8) {
+  for (int step = slab_start; step < slab_end; step += TS) {
     const int t = step + ts;
     const bool valid = t < ctx;
-    // k row address via page table
     const short* krow = nullptr;
     const short* vrow = nullptr;
     if (valid) {
@@ -100,78 +102,82 @@ __global__ __launch_bounds__(kBlock) void paged_decode_phase1(
       krow = kcache + base;
       vrow = vcache + base;
     }
-    short8 k0 = valid ? *reinterpret_cast<const short8*>(krow + dp * 16) : short8{};
-    short8 k1 = valid ? *reinterpret_cast<const short8*>(krow + dp * 16 + 8) : short8{};
+    short8 kv8[NV8];
+#pragma unroll
+    for (int i = 0; i < NV8; i++)
+      kv8[i] = valid ? *reinterpret_cast<const short8*>(krow + dp * ND + i * 8)
+                     : short8{};
 
     float s[G];
 #pragma unroll
     for (int g = 0; g < G; g++) {
       float d = 0.f;
 #pragma unroll
-      for (int i = 0; i < 8; i++) {
-        d += bf16_to_f32(k0[i]) * qreg[g][i];
-        d += bf16_to_f32(k1[i]) * qreg[g][8 + i];
-      }
-      // butterfly over the 8 dim-slice lanes -> full dot product on all lanes
-      d += __shfl_xor(d, 1, WAVE_SIZE);
-      d += __shfl_xor(d, 2, WAVE_SIZE);
-      d += __shfl_xor(d, 4, WAVE_SIZE);
+      for (int i = 0; i < NV8; i++)
+#pragma unroll
+        for (int e = 0; e < 8; e++)
+          d += bf16_to_f32(kv8[i][e]) * qreg[g][i * 8 + e];
+      // butterfly over the DP dim-slice lanes -> full dot on all lanes
+#pragma unroll
+      for (int off = 1; off < DP; off <<= 1) d += __shfl_xor(d, off, WAVE_SIZE);
       s[g] = valid ? d : kNegInf;
     }
 
     // load v slice once per token
-    short8 v0 = valid ? *reinterpret_cast<const short8*>(vrow + dp * 16) : short8{};
-    short8 v1 = valid ? *reinterpret_cast<const short8*>(vrow + dp * 16 + 8) : short8{};
-    float vf[16];
+    float vf[ND];
 #pragma unroll
-    for (int i = 0; i < 8; i++) { vf[i] = bf16_to_f32(v0[i]); vf[8 + i] = bf16_to_f32(v1[i]); }
+    for (int i = 0; i < NV8; i++) {
+      short8 vv = valid ? *reinterpret_cast<const short8*>(vrow + dp * ND + i * 8)
+                        : short8{};
+#pragma unroll
+      for (int e = 0; e < 8; e++) vf[i * 8 + e] = bf16_to_f32(vv[e]);
+    }
 
 #pragma unroll
     for (int g = 0; g < G; g++) {
-      // step max over the 8 tokens (s[g] is uniform within a token's lanes)
+      // step max over the TS tokens (s[g] is uniform within a token's lanes)
       float ms = s[g];
-      ms = fmaxf(ms, __shfl_xor(ms, 8, WAVE_SIZE));
-      ms = fmaxf(ms, __shfl_xor(ms, 16, WAVE_SIZE));
-      ms = fmaxf(ms, __shfl_xor(ms, 32, WAVE_SIZE));
+#pragma unroll
+      for (int off = DP; off < 64; off <<= 1)
+        ms = fmaxf(ms, __shfl_xor(ms, off, WAVE_SIZE));
       if (ms > m[g]) {
         const float corr = __expf(m[g] - ms);
         l[g] *= corr;
 #pragma unroll
-        for (int i = 0; i < 16; i++) acc[g][i] *= corr;
+        for (int i = 0; i < ND; i++) acc[g][i] *= corr;
         m[g] = ms;
       }
       const float p = (s[g] > kNegInf * 0.5f) ? __expf(s[g] - m[g]) : 0.f;
-      // sum of p over the 8 tokens of this step
       float psum = p;
-      psum += __shfl_xor(psum, 8, WAVE_SIZE);
-      psum += __shfl_xor(psum, 16, WAVE_SIZE);
-      psum += __shfl_xor(psum, 32, WAVE_SIZE);
+#pragma unroll
+      for (int off = DP; off < 64; off <<= 1)
+        psum += __shfl_xor(psum, off, WAVE_SIZE);
       l[g] += psum;
 #pragma unroll
-      for (int i = 0; i < 16; i++) acc[g][i] = fmaf(p, vf[i], acc[g][i]);
+      for (int i = 0; i < ND; i++) acc[g][i] = fmaf(p, vf[i], acc[g][i]);
     }
   }
 
-  // fold the 8 token-slots: acc currently holds per-(ts) partial sums
+  // fold the token-slot partials: acc holds per-(ts) sums
 #pragma unroll
   for (int g = 0; g < G; g++)
 #pragma unroll
-    for (int i = 0; i < 16; i++) {
+    for (int i = 0; i < ND; i++) {
       float a = acc[g][i];
-      a += __shfl_xor(a, 8, WAVE_SIZE);
-      a += __shfl_xor(a, 16, WAVE_SIZE);
-      a += __shfl_xor(a, 32, WAVE_SIZE);
+#pragma unroll
+      for (int off = DP; off < 64; off <<= 1)
+        a += __shfl_xor(a, off, WAVE_SIZE);
       acc[g][i] = a;
     }
 
   // cross-wave merge via LDS. Wave w writes [G][hd] acc + m,l.
   __syncthreads();  // q_lds no longer needed
   float* my = merge + wid * G * (hd + 2);
-  if (ts == 0) {  // lanes 0..7 cover the 8 dim slices exactly once
+  if (ts == 0) {  // lanes 0..DP-1 cover the DP dim slices exactly once
 #pragma unroll
     for (int g = 0; g < G; g++) {
 #pragma unroll
-      for (int i = 0; i < 16; i++) my[g * (hd + 2) + dp * 16 + i] = acc[g][i];
+      for (int i = 0; i < ND; i++) my[g * (hd + 2) + dp * ND + i] = acc[g][i];
       if (dp == 0) { my[g * (hd + 2) + hd] = m[g]; my[g * (hd + 2) + hd + 1] = l[g]; }
     }
   }
@@ -232,7 +238,6 @@ __global__ void paged_decode_phase2(short* __restrict__ out,  // [B, Hq, hd]
   __syncthreads();
   mstar = smax[0];
 
-  // each thread owns output dims stride-wise
   for (int d = threadIdx.x; d < hd; d += blockDim.x) {
     float asum = 0.f, lsum = 0.f;
     for (int c = 0; c < nc; c++) {
@@ -275,18 +280,18 @@ void paged_attention_decode(torch::Tensor out, torch::Tensor q,
 
   const int lds_bytes = (G * hd + 4 * G * (hd + 2)) * sizeof(float);
   dim3 grid(B, Hkv, C);
-#define LAUNCH_G(GG)                                                          \
-  paged_decode_phase1<GG><<<grid, kBlock, lds_bytes, stream>>>(               \
+#define LAUNCH_G(GG, DP)                                                      \
+  paged_decode_phase1<GG, DP><<<grid, kBlock, lds_bytes, stream>>>(           \
       partial.data_ptr<float>(), ml.data_ptr<float>(), (short*)out.data_ptr(),\
       (const short*)q.data_ptr(), (const short*)kcache.data_ptr(),            \
       (const short*)vcache.data_ptr(), page_table.data_ptr<int32_t>(),        \
       ctx_lens.data_ptr<int32_t>(), (float)scale, B, Hkv, C, max_pages,       \
       log2_ps, hd)
   switch (G) {
-    case 1: LAUNCH_G(1); break;
-    case 2: LAUNCH_G(2); break;
-    case 4: LAUNCH_G(4); break;
-    case 8: LAUNCH_G(8); break;
+    case 1: LAUNCH_G(1, 8); break;
+    case 2: LAUNCH_G(2, 8); break;
+    case 4: LAUNCH_G(4, 8); break;
+    case 8: LAUNCH_G(8, 16); break;
     default: TORCH_CHECK(false, "unsupported GQA group size ", G);
   }
 #undef LAUNCH_G

@@ -59,6 +59,12 @@ static uint64_t hash_block(uint64_t parent, const std::vector<int32_t>& tokens) 
   return hash_tokens_impl(parent, tokens.data(), tokens.size());
 }
 
+// Root parent of a chain (so incremental hashing via hash_block matches
+// chain_hashes exactly).
+static uint64_t chain_root(uint64_t salt) {
+  return mix64(salt ^ 0xa0761d6478bd642full);
+}
+
 // ---------------------------------------------------------------------------
 // KvIndexer: sequence-hash -> {workers that hold the block}, plus per-worker
 // block counts. Thread-safe (event ingestion and routing run on different
@@ -153,6 +159,7 @@ PYBIND11_MODULE(_core, m) {
   m.def("chain_hashes", &chain_hashes, py::arg("tokens"), py::arg("block_size"),
         py::arg("salt") = 0);
   m.def("hash_block", &hash_block, py::arg("parent"), py::arg("tokens"));
+  m.def("chain_root", &chain_root, py::arg("salt") = 0);
   py::class_<KvIndexer>(m, "KvIndexer")
       .def(py::init<>())
       .def("apply_stored", &KvIndexer::apply_stored)

@@ -58,6 +58,13 @@ class LLMEngine:
         self.kv_events: List[KvEvent] = []
         self.last_metrics = ForwardPassMetrics()
         self._held: Dict[str, Request] = {}  # finished but KV retained (disagg)
+        # hipGraph decode fast path
+        self.graph_runner = None
+        if (cfg.enable_hip_graphs and self.runner.device.type == "cuda"
+                and cfg.tp_size == 1):
+            from .graphs import GraphRunner
+            self.graph_runner = GraphRunner(self.runner, cfg.max_num_seqs)
+        self._last_sampled = None
 
     # ------------------------------------------------------------------
     def add_request(self, req_id: str, prompt_tokens: List[int],
@@ -87,7 +94,85 @@ class LLMEngine:
             return "stop"
         return None
 
+    # ------------------------------------------------------------------
+    def _fast_decode_eligible(self) -> bool:
+        if self.graph_runner is None or self.scheduler.waiting:
+            return False
+        running = self.scheduler.running
+        if not running or len(running) > self.graph_runner.max_batch:
+            return False
+        return all(r.is_decode and r.total_len - r.num_computed == 1
+                   for r in running)
+
+    def _fast_decode_step(self) -> List[StepOutput]:
+        gr = self.graph_runner
+        running = self.scheduler.running
+        rebuilt = gr.dirty or gr.reqs != running
+        # host-side page growth (every page_size steps per seq)
+        for i, r in enumerate(running):
+            pos = r.num_computed
+            need = pos // self.cfg.page_size + 1
+            if len(r.kv.pages) < need:
+                try:
+                    r.kv.ensure_capacity(pos + 1)
+                except MemoryError:
+                    return None  # let the eager path preempt
+                if not rebuilt:
+                    gr.patch_new_page(i, len(r.kv.pages) - 1, r.kv.pages[-1])
+        if rebuilt:
+            gr.rebuild(running)
+        self.step_count += 1
+        t0 = time.monotonic()
+        logits = gr.step(self._last_sampled)
+        from .sampling import sample_tokens
+        sampled = sample_tokens(logits, running, self.step_count)
+        self._last_sampled = sampled
+        toks = sampled.cpu().tolist()
+
+        outputs: List[StepOutput] = []
+        finished_any = False
+        for tok, req in zip(toks, list(running)):
+            req.num_computed += 1
+            req.output_tokens.append(int(tok))
+            if req.first_token_time is None:
+                req.first_token_time = time.monotonic()
+            reason = self._check_finish(req)
+            finished = reason is not None
+            if finished:
+                finished_any = True
+                if req.hold_kv:
+                    req.state = ReqState.FINISHED
+                    req.finish_reason = reason
+                    req.finish_time = time.monotonic()
+                    self.scheduler.running.remove(req)
+                    self._held[req.req_id] = req
+                else:
+                    self._finish(req, reason)
+            elif (self.cfg.kv_events or self.cfg.enable_prefix_caching) and \
+                    req.num_computed % self.cfg.page_size == 0:
+                req.kv.commit_full_pages(req.all_tokens, req.num_computed)
+            outputs.append(StepOutput(req.req_id, int(tok), finished, reason,
+                                      len(req.output_tokens)))
+        if finished_any:
+            gr.dirty = True
+        self.kv_events.extend(self.alloc.drain_events())
+        n = len(outputs)
+        self.last_metrics = ForwardPassMetrics(
+            step=self.step_count, num_running=self.scheduler.num_running(),
+            num_waiting=0, kv_usage=self.alloc.usage, num_tokens_step=n,
+            decode_tokens_step=n,
+            step_time_ms=(time.monotonic() - t0) * 1000,
+            total_kv_pages=self.alloc.num_pages)
+        return outputs
+
     def step(self) -> List[StepOutput]:
+        if self._fast_decode_eligible():
+            out = self._fast_decode_step()
+            if out is not None:
+                return out
+        if self.graph_runner is not None:
+            self.graph_runner.dirty = True
+            self._last_sampled = None
         t0 = time.monotonic()
         sched = self.scheduler.schedule()
         if sched.is_empty:

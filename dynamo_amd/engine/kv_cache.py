@@ -133,6 +133,10 @@ class SequenceKV:
         self.salt = salt
         self.pages: List[int] = []
         self.num_cached_tokens = 0  # tokens whose KV was reused from cache
+        # incremental hash-chain state: hashes registered so far
+        self._num_hashed = 0
+        self._parent = _core.chain_root(salt)
+        self._parent_hash = None
 
     def match_prefix(self, tokens: List[int]) -> int:
         """Reuse cached full pages covering a prefix of `tokens`.
@@ -151,6 +155,10 @@ class SequenceKV:
                 break
             self.pages.append(pid)
             reused = (i + 1) * ps
+            # advance incremental hash state past reused pages
+            self._parent = h
+            self._parent_hash = h
+            self._num_hashed = i + 1
         self.num_cached_tokens = reused
         return reused
 
@@ -166,14 +174,17 @@ class SequenceKV:
 
     def commit_full_pages(self, tokens: List[int], num_computed: int):
         """Register hashes for pages that just became full (KV written for
-        the first `num_computed` tokens of `tokens`)."""
+        the first `num_computed` tokens of `tokens`). Incremental: only new
+        full pages are hashed."""
         ps = self.alloc.page_size
         full = num_computed // ps
-        hashes = _core.chain_hashes(tokens[: full * ps], ps, self.salt)
-        parent = None
-        for i in range(full):
-            self.alloc.register_hash(self.pages[i], hashes[i], parent)
-            parent = hashes[i]
+        for i in range(self._num_hashed, full):
+            h = _core.hash_block(self._parent, tokens[i * ps:(i + 1) * ps])
+            prev = None if i == 0 else self._parent_hash
+            self.alloc.register_hash(self.pages[i], h, prev)
+            self._parent_hash = h
+            self._parent = h
+            self._num_hashed = i + 1
 
     def release(self):
         for pid in self.pages:
