@@ -66,18 +66,21 @@ __global__ __launch_bounds__(kBlock) void paged_decode_phase1(
     return;
   }
 
+  // stage q as raw bf16 (scale is applied to the dot result)
+  short* q_lds_s = reinterpret_cast<short*>(q_lds);
   for (int i = threadIdx.x; i < G * hd; i += kBlock) {
     const int g = i / hd;
-    q_lds[i] = bf16_to_f32(((const short*)q)[((int64_t)b * Hq + h * G + g) * hd + i % hd]) * scale;
+    q_lds_s[i] = ((const short*)q)[((int64_t)b * Hq + h * G + g) * hd + i % hd];
   }
   __syncthreads();
 
-  // preload this lane's q slice for each group head: q[g][dp*ND .. +ND)
-  float qreg[G][ND];
+  // this lane's q slice per group head, kept as packed bf16 (VGPR budget)
+  short8 qreg[G][NV8];
 #pragma unroll
   for (int g = 0; g < G; g++)
 #pragma unroll
-    for (int i = 0; i < ND; i++) qreg[g][i] = q_lds[g * hd + dp * ND + i];
+    for (int i = 0; i < NV8; i++)
+      qreg[g][i] = *reinterpret_cast<const short8*>(q_lds_s + g * hd + dp * ND + i * 8);
 
   float m[G], l[G], acc[G][ND];
 #pragma unroll
@@ -91,71 +94,84 @@ __global__ __launch_bounds__(kBlock) void paged_decode_phase1(
   const int slab_end = min(slab_start + kSlab, ctx);
   const int32_t* pt = page_table + (int64_t)b * max_pages;
 
-  for (int step = slab_start; step < slab_end; step += TS) {
-    const int t = step + ts;
-    const bool valid = t < ctx;
-    const short* krow = nullptr;
-    const short* vrow = nullptr;
-    if (valid) {
-      const int64_t page = pt[t >> log2_ps];
-      const int row = t & (ps - 1);
-      const int64_t base = (((page * Hkv + h) * ps + row) * hd);
-      krow = kcache + base;
-      vrow = vcache + base;
+  // iterate page-aligned windows (page base lookup is wave-uniform and
+  // hoisted); within a window, K/V loads for step+TS are issued before
+  // processing step (register double-buffer) so HBM latency overlaps the
+  // softmax VALU work.
+  for (int w0 = slab_start; w0 < slab_end; w0 += ps) {
+    const int wend = min(w0 + ps, slab_end);
+    const int64_t pbase = (((int64_t)pt[w0 >> log2_ps] * Hkv + h) * ps) * hd;
+    short8 kcur[NV8], vcur[NV8], knxt[NV8], vnxt[NV8];
+    {
+      const int t = w0 + ts;
+      const bool v = t < wend;
+      const short* kp = kcache + pbase + (int64_t)(t & (ps - 1)) * hd + dp * ND;
+      const short* vp = vcache + pbase + (int64_t)(t & (ps - 1)) * hd + dp * ND;
+#pragma unroll
+      for (int i = 0; i < NV8; i++) {
+        kcur[i] = v ? *reinterpret_cast<const short8*>(kp + i * 8) : short8{};
+        vcur[i] = v ? *reinterpret_cast<const short8*>(vp + i * 8) : short8{};
+      }
     }
-    short8 kv8[NV8];
+    for (int step = w0; step < wend; step += TS) {
+      // prefetch next token group
+      if (step + TS < wend) {
+        const int t = step + TS + ts;
+        const bool v = t < wend;
+        const short* kp = kcache + pbase + (int64_t)(t & (ps - 1)) * hd + dp * ND;
+        const short* vp = vcache + pbase + (int64_t)(t & (ps - 1)) * hd + dp * ND;
 #pragma unroll
-    for (int i = 0; i < NV8; i++)
-      kv8[i] = valid ? *reinterpret_cast<const short8*>(krow + dp * ND + i * 8)
-                     : short8{};
+        for (int i = 0; i < NV8; i++) {
+          knxt[i] = v ? *reinterpret_cast<const short8*>(kp + i * 8) : short8{};
+          vnxt[i] = v ? *reinterpret_cast<const short8*>(vp + i * 8) : short8{};
+        }
+      }
+      const bool valid = (step + ts) < wend;
 
-    float s[G];
+      float s[G];
 #pragma unroll
-    for (int g = 0; g < G; g++) {
-      float d = 0.f;
+      for (int g = 0; g < G; g++) {
+        float d = 0.f;
+#pragma unroll
+        for (int i = 0; i < NV8; i++)
+#pragma unroll
+          for (int e = 0; e < 8; e++)
+            d += bf16_to_f32(kcur[i][e]) * bf16_to_f32(qreg[g][i][e]);
+#pragma unroll
+        for (int off = 1; off < DP; off <<= 1) d += __shfl_xor(d, off, WAVE_SIZE);
+        s[g] = valid ? d * scale : kNegInf;
+      }
+
+      float vf[ND];
 #pragma unroll
       for (int i = 0; i < NV8; i++)
 #pragma unroll
-        for (int e = 0; e < 8; e++)
-          d += bf16_to_f32(kv8[i][e]) * qreg[g][i * 8 + e];
-      // butterfly over the DP dim-slice lanes -> full dot on all lanes
-#pragma unroll
-      for (int off = 1; off < DP; off <<= 1) d += __shfl_xor(d, off, WAVE_SIZE);
-      s[g] = valid ? d : kNegInf;
-    }
-
-    // load v slice once per token
-    float vf[ND];
-#pragma unroll
-    for (int i = 0; i < NV8; i++) {
-      short8 vv = valid ? *reinterpret_cast<const short8*>(vrow + dp * ND + i * 8)
-                        : short8{};
-#pragma unroll
-      for (int e = 0; e < 8; e++) vf[i * 8 + e] = bf16_to_f32(vv[e]);
-    }
+        for (int e = 0; e < 8; e++) vf[i * 8 + e] = bf16_to_f32(vcur[i][e]);
 
 #pragma unroll
-    for (int g = 0; g < G; g++) {
-      // step max over the TS tokens (s[g] is uniform within a token's lanes)
-      float ms = s[g];
+      for (int g = 0; g < G; g++) {
+        float ms = s[g];
 #pragma unroll
-      for (int off = DP; off < 64; off <<= 1)
-        ms = fmaxf(ms, __shfl_xor(ms, off, WAVE_SIZE));
-      if (ms > m[g]) {
-        const float corr = __expf(m[g] - ms);
-        l[g] *= corr;
+        for (int off = DP; off < 64; off <<= 1)
+          ms = fmaxf(ms, __shfl_xor(ms, off, WAVE_SIZE));
+        if (ms > m[g]) {
+          const float corr = __expf(m[g] - ms);
+          l[g] *= corr;
 #pragma unroll
-        for (int i = 0; i < ND; i++) acc[g][i] *= corr;
-        m[g] = ms;
+          for (int i = 0; i < ND; i++) acc[g][i] *= corr;
+          m[g] = ms;
+        }
+        const float p = (s[g] > kNegInf * 0.5f) ? __expf(s[g] - m[g]) : 0.f;
+        float psum = p;
+#pragma unroll
+        for (int off = DP; off < 64; off <<= 1)
+          psum += __shfl_xor(psum, off, WAVE_SIZE);
+        l[g] += psum;
+#pragma unroll
+        for (int i = 0; i < ND; i++) acc[g][i] = fmaf(p, vf[i], acc[g][i]);
       }
-      const float p = (s[g] > kNegInf * 0.5f) ? __expf(s[g] - m[g]) : 0.f;
-      float psum = p;
 #pragma unroll
-      for (int off = DP; off < 64; off <<= 1)
-        psum += __shfl_xor(psum, off, WAVE_SIZE);
-      l[g] += psum;
-#pragma unroll
-      for (int i = 0; i < ND; i++) acc[g][i] = fmaf(p, vf[i], acc[g][i]);
+      for (int i = 0; i < NV8; i++) { kcur[i] = knxt[i]; vcur[i] = vnxt[i]; }
     }
   }
 
@@ -291,7 +307,7 @@ void paged_attention_decode(torch::Tensor out, torch::Tensor q,
   switch (G) {
     case 1: LAUNCH_G(1, 8); break;
     case 2: LAUNCH_G(2, 8); break;
-    case 4: LAUNCH_G(4, 8); break;
+    case 4: LAUNCH_G(4, 16); break;
     case 8: LAUNCH_G(8, 16); break;
     default: TORCH_CHECK(false, "unsupported GQA group size ", G);
   }
