@@ -22,9 +22,14 @@ constexpr int kChunk = 512;        // context tokens per workgroup
 constexpr int kSlab = kChunk / 4;  // tokens per wave (128)
 constexpr float kNegInf = -1e30f;
 
-// G = GQA group size (Hq / Hkv); DP = lanes per token. G*hd/DP accumulator
-// floats per lane — keep <= 64 to stay under the occupancy cliff.
-template <int G, int DP>
+// G = GQA group size (Hq / Hkv); DP = lanes per token.
+// HEADSPLIT (G >= 4): the 4 waves split the GQA group (G/4 heads each) and
+// every wave streams the whole chunk — shrinks the per-lane accumulator 4x
+// (occupancy: 2 -> 6+ waves/SIMD for G=8) and removes the LDS merge; the 4
+// waves read the same KV stream, so 3 of 4 passes hit L1/L2.
+// !HEADSPLIT (G < 4): waves split the chunk into 128-token slabs and merge
+// partials through LDS.
+template <int G, int DP, bool HEADSPLIT>
 __global__ __launch_bounds__(kBlock) void paged_decode_phase1(
     float* __restrict__ partial,        // [B, Hq, C, hd] fp32
     float* __restrict__ ml,             // [B, Hq, C, 2] fp32 (m, l)
@@ -74,24 +79,30 @@ __global__ __launch_bounds__(kBlock) void paged_decode_phase1(
   }
   __syncthreads();
 
+  // heads this wave computes
+  constexpr int GW = HEADSPLIT ? G / 4 : G;
+  const int hoff = HEADSPLIT ? wid * GW : 0;
+
   // this lane's q slice per group head, kept as packed bf16 (VGPR budget)
-  short8 qreg[G][NV8];
+  short8 qreg[GW][NV8];
 #pragma unroll
-  for (int g = 0; g < G; g++)
+  for (int g = 0; g < GW; g++)
 #pragma unroll
     for (int i = 0; i < NV8; i++)
-      qreg[g][i] = *reinterpret_cast<const short8*>(q_lds_s + g * hd + dp * ND + i * 8);
+      qreg[g][i] = *reinterpret_cast<const short8*>(
+          q_lds_s + (hoff + g) * hd + dp * ND + i * 8);
 
-  float m[G], l[G], acc[G][ND];
+  float m[GW], l[GW], acc[GW][ND];
 #pragma unroll
-  for (int g = 0; g < G; g++) {
+  for (int g = 0; g < GW; g++) {
     m[g] = kNegInf; l[g] = 0.f;
 #pragma unroll
     for (int i = 0; i < ND; i++) acc[g][i] = 0.f;
   }
 
-  const int slab_start = chunk_start + wid * kSlab;
-  const int slab_end = min(slab_start + kSlab, ctx);
+  const int slab_start = HEADSPLIT ? chunk_start : chunk_start + wid * kSlab;
+  const int slab_end = min(HEADSPLIT ? chunk_start + kChunk : slab_start + kSlab,
+                           ctx);
   const int32_t* pt = page_table + (int64_t)b * max_pages;
 
   // iterate page-aligned windows (page base lookup is wave-uniform and
@@ -128,9 +139,9 @@ __global__ __launch_bounds__(kBlock) void paged_decode_phase1(
       }
       const bool valid = (step + ts) < wend;
 
-      float s[G];
+      float s[GW];
 #pragma unroll
-      for (int g = 0; g < G; g++) {
+      for (int g = 0; g < GW; g++) {
         float d = 0.f;
 #pragma unroll
         for (int i = 0; i < NV8; i++)
@@ -149,7 +160,7 @@ __global__ __launch_bounds__(kBlock) void paged_decode_phase1(
         for (int e = 0; e < 8; e++) vf[i * 8 + e] = bf16_to_f32(vcur[i][e]);
 
 #pragma unroll
-      for (int g = 0; g < G; g++) {
+      for (int g = 0; g < GW; g++) {
         float ms = s[g];
 #pragma unroll
         for (int off = DP; off < 64; off <<= 1)
@@ -177,7 +188,7 @@ __global__ __launch_bounds__(kBlock) void paged_decode_phase1(
 
   // fold the token-slot partials: acc holds per-(ts) sums
 #pragma unroll
-  for (int g = 0; g < G; g++)
+  for (int g = 0; g < GW; g++)
 #pragma unroll
     for (int i = 0; i < ND; i++) {
       float a = acc[g][i];
@@ -187,12 +198,37 @@ __global__ __launch_bounds__(kBlock) void paged_decode_phase1(
       acc[g][i] = a;
     }
 
+  if (HEADSPLIT) {
+    // each wave covered the whole chunk for its own heads: write directly
+    if (ts == 0) {  // lanes 0..DP-1 cover the DP dim slices exactly once
+#pragma unroll
+      for (int g = 0; g < GW; g++) {
+        const int qh = h * G + hoff + g;
+#pragma unroll
+        for (int i = 0; i < ND; i++) {
+          const int d = dp * ND + i;
+          if (C == 1) {
+            out[((int64_t)b * Hq + qh) * hd + d] =
+                f32_to_bf16(l[g] > 0.f ? acc[g][i] / l[g] : 0.f);
+          } else {
+            partial[(((int64_t)b * Hq + qh) * C + c) * hd + d] = acc[g][i];
+          }
+        }
+        if (C > 1 && dp == 0) {
+          float* mlp = ml + (((int64_t)b * Hq + qh) * C + c) * 2;
+          mlp[0] = m[g]; mlp[1] = l[g];
+        }
+      }
+    }
+    return;
+  }
+
   // cross-wave merge via LDS. Wave w writes [G][hd] acc + m,l.
   __syncthreads();  // q_lds no longer needed
   float* my = merge + wid * G * (hd + 2);
   if (ts == 0) {  // lanes 0..DP-1 cover the DP dim slices exactly once
 #pragma unroll
-    for (int g = 0; g < G; g++) {
+    for (int g = 0; g < GW; g++) {
 #pragma unroll
       for (int i = 0; i < ND; i++) my[g * (hd + 2) + dp * ND + i] = acc[g][i];
       if (dp == 0) { my[g * (hd + 2) + hd] = m[g]; my[g * (hd + 2) + hd + 1] = l[g]; }
@@ -297,18 +333,18 @@ void paged_attention_decode(torch::Tensor out, torch::Tensor q,
 
   const int lds_bytes = (G * hd + 4 * G * (hd + 2)) * sizeof(float);
   dim3 grid(B, Hkv, C);
-#define LAUNCH_G(GG, DP)                                                      \
- hipLaunchKernelGGL(( paged_decode_phase1<GG, DP>), dim3(grid), dim3(kBlock), lds_bytes, stream,            \
+#define LAUNCH_G(GG, DP, HS)                                                  \
+ hipLaunchKernelGGL(( paged_decode_phase1<GG, DP, HS>), dim3(grid), dim3(kBlock), lds_bytes, stream,        \
       partial.data_ptr<float>(), ml.data_ptr<float>(), (short*)out.data_ptr(),\
       (const short*)q.data_ptr(), (const short*)kcache.data_ptr(),            \
       (const short*)vcache.data_ptr(), page_table.data_ptr<int32_t>(),        \
       ctx_lens.data_ptr<int32_t>(), (float)scale, B, Hkv, C, max_pages,       \
       log2_ps, hd)
   switch (G) {
-    case 1: LAUNCH_G(1, 8); break;
-    case 2: LAUNCH_G(2, 8); break;
-    case 4: LAUNCH_G(4, 16); break;
-    case 8: LAUNCH_G(8, 16); break;
+    case 1: LAUNCH_G(1, 8, false); break;
+    case 2: LAUNCH_G(2, 8, false); break;
+    case 4: LAUNCH_G(4, 16, true); break;
+    case 8: LAUNCH_G(8, 16, true); break;
     default: TORCH_CHECK(false, "unsupported GQA group size ", G);
   }
 #undef LAUNCH_G
