@@ -343,7 +343,7 @@ void paged_attention_decode(torch::Tensor out, torch::Tensor q,
   switch (G) {
     case 1: LAUNCH_G(1, 8, false); break;
     case 2: LAUNCH_G(2, 8, false); break;
-    case 4: LAUNCH_G(4, 16, true); break;
+    case 4: LAUNCH_G(4, 16, false); break;
     case 8: LAUNCH_G(8, 16, true); break;
     default: TORCH_CHECK(false, "unsupported GQA group size ", G);
   }
