@@ -1,0 +1,304 @@
+// Paged-attention decode kernels (torch-free; included by the extension and
+// by the standalone sweep tool benchmarks/decode_sweep.hip).
+//
+// Template space:
+//   G     : GQA group size (Hq / Hkv)
+//   DP    : lanes per token (8 or 16); dims per lane = 128/DP
+//   HS    : head-split factor (1, 2 or 4): the 4 waves form HS head-groups
+//           x (4/HS) token-groups. More head-split => smaller accumulators
+//           (occupancy) but each KV token is streamed by HS waves.
+//   DEPTH : software-pipeline depth (token groups in flight per wave).
+//           Static buffer indices only (runtime-indexed register arrays
+//           spill to scratch).
+#pragma once
+#include "common.h"
+
+namespace decode_attn {
+
+constexpr int kBlock = 256;        // 4 waves
+constexpr int kChunk = 512;        // context tokens per workgroup
+constexpr float kNegInf = -1e30f;
+
+template <int G, int DP, int HS, int DEPTH>
+__global__ __launch_bounds__(kBlock) void paged_decode_phase1(
+    float* __restrict__ partial,        // [B, Hq, C, hd] fp32
+    float* __restrict__ ml,             // [B, Hq, C, 2] fp32 (m, l)
+    short* __restrict__ out,            // [B, Hq, hd] bf16 (used when C==1)
+    const short* __restrict__ q,        // [B, Hq, hd]
+    const short* __restrict__ kcache,   // [P, Hkv, ps, hd]
+    const short* __restrict__ vcache,
+    const int32_t* __restrict__ page_table,  // [B, max_pages]
+    const int32_t* __restrict__ ctx_lens,    // [B]
+    float scale, int B, int Hkv, int C, int max_pages, int log2_ps, int hd) {
+  constexpr int ND = 128 / DP;       // dims per lane
+  constexpr int NV8 = ND / 8;        // short8 loads per row slice
+  constexpr int TS = 64 / DP;        // tokens per wave step
+  constexpr int GW = G / HS;         // heads per wave
+  constexpr int TG = 4 / HS;         // token-groups (waves splitting tokens)
+  constexpr int kSlab = kChunk / TG;
+  const int b = blockIdx.x;
+  const int h = blockIdx.y;   // kv head
+  const int c = blockIdx.z;   // context chunk
+  const int Hq = Hkv * G;
+  const int ctx = ctx_lens[b];
+  const int chunk_start = c * kChunk;
+  const int lane = threadIdx.x & 63;
+  const int wid = threadIdx.x >> 6;
+  const int tg = wid / HS;        // token-group index
+  const int hg = wid % HS;        // head-group index
+  const int ts = lane / DP;       // token slot within step
+  const int dp = lane % DP;       // dim slice: dims [dp*ND, dp*ND+ND)
+  const int ps = 1 << log2_ps;
+  const int hoff = hg * GW;
+
+  // LDS: q tile [G][hd] bf16 + merge scratch [TG][G][hd+2] fp32
+  extern __shared__ float lds[];
+  float* merge = lds;                          // TG * G * (hd+2)
+  short* q_lds_s = reinterpret_cast<short*>(lds + (TG > 1 ? TG * G * (hd + 2) : 0));
+
+  if (chunk_start >= ctx) {
+    if (C > 1) {
+      for (int i = threadIdx.x; i < G; i += kBlock) {
+        const int qh = h * G + i;
+        float* mlp = ml + (((int64_t)b * Hq + qh) * C + c) * 2;
+        mlp[0] = kNegInf; mlp[1] = 0.f;
+      }
+    }
+    return;
+  }
+
+  for (int i = threadIdx.x; i < G * hd; i += kBlock) {
+    const int g = i / hd;
+    q_lds_s[i] = q[((int64_t)b * Hq + h * G + g) * hd + i % hd];
+  }
+  __syncthreads();
+
+  float qf[GW][ND];
+#pragma unroll
+  for (int g = 0; g < GW; g++)
+#pragma unroll
+    for (int i = 0; i < ND; i++)
+      qf[g][i] = bf16_to_f32(q_lds_s[(hoff + g) * hd + dp * ND + i]);
+
+  float m[GW], l[GW], acc[GW][ND];
+#pragma unroll
+  for (int g = 0; g < GW; g++) {
+    m[g] = kNegInf; l[g] = 0.f;
+#pragma unroll
+    for (int i = 0; i < ND; i++) acc[g][i] = 0.f;
+  }
+
+  const int slab_start = chunk_start + tg * kSlab;
+  const int slab_end = min(slab_start + kSlab, ctx);
+  const int32_t* pt = page_table + (int64_t)b * max_pages;
+
+  for (int w0 = slab_start; w0 < slab_end; w0 += ps) {
+    const int wend = min(w0 + ps, slab_end);
+    const int64_t pbase = (((int64_t)pt[w0 >> log2_ps] * Hkv + h) * ps) * hd;
+    const int ngroups = (wend - w0 + TS - 1) / TS;
+
+    short8 kbuf[DEPTH][NV8], vbuf[DEPTH][NV8];
+
+    auto load_group = [&](int gi, short8 (&kb)[NV8], short8 (&vb)[NV8]) {
+      const int t = w0 + gi * TS + ts;
+      const bool v = t < wend;
+      const short* kp = kcache + pbase + (int64_t)(t & (ps - 1)) * hd + dp * ND;
+      const short* vp = vcache + pbase + (int64_t)(t & (ps - 1)) * hd + dp * ND;
+#pragma unroll
+      for (int i = 0; i < NV8; i++) {
+        kb[i] = v ? *reinterpret_cast<const short8*>(kp + i * 8) : short8{};
+        vb[i] = v ? *reinterpret_cast<const short8*>(vp + i * 8) : short8{};
+      }
+    };
+
+    auto process_group = [&](int gi, short8 (&kb)[NV8], short8 (&vb)[NV8]) {
+      const bool valid = (w0 + gi * TS + ts) < wend;
+      float s[GW];
+#pragma unroll
+      for (int g = 0; g < GW; g++) {
+        float d = 0.f;
+#pragma unroll
+        for (int i = 0; i < NV8; i++)
+#pragma unroll
+          for (int e = 0; e < 8; e++)
+            d += bf16_to_f32(kb[i][e]) * qf[g][i * 8 + e];
+#pragma unroll
+        for (int off = 1; off < DP; off <<= 1) d += __shfl_xor(d, off, WAVE_SIZE);
+        s[g] = valid ? d * scale : kNegInf;
+      }
+      float vf[ND];
+#pragma unroll
+      for (int i = 0; i < NV8; i++)
+#pragma unroll
+        for (int e = 0; e < 8; e++) vf[i * 8 + e] = bf16_to_f32(vb[i][e]);
+#pragma unroll
+      for (int g = 0; g < GW; g++) {
+        float ms = s[g];
+#pragma unroll
+        for (int off = DP; off < 64; off <<= 1)
+          ms = fmaxf(ms, __shfl_xor(ms, off, WAVE_SIZE));
+        if (ms > m[g]) {
+          const float corr = __expf(m[g] - ms);
+          l[g] *= corr;
+#pragma unroll
+          for (int i = 0; i < ND; i++) acc[g][i] *= corr;
+          m[g] = ms;
+        }
+        const float p = (s[g] > kNegInf * 0.5f) ? __expf(s[g] - m[g]) : 0.f;
+        float psum = p;
+#pragma unroll
+        for (int off = DP; off < 64; off <<= 1)
+          psum += __shfl_xor(psum, off, WAVE_SIZE);
+        l[g] += psum;
+#pragma unroll
+        for (int i = 0; i < ND; i++) acc[g][i] = fmaf(p, vf[i], acc[g][i]);
+      }
+    };
+
+#pragma unroll
+    for (int j = 0; j < DEPTH; j++)
+      if (j < ngroups) load_group(j, kbuf[j], vbuf[j]);
+
+    int gi = 0;
+    while (gi + DEPTH <= ngroups) {
+#pragma unroll
+      for (int j = 0; j < DEPTH; j++) {
+        process_group(gi + j, kbuf[j], vbuf[j]);
+        if (gi + j + DEPTH < ngroups)
+          load_group(gi + j + DEPTH, kbuf[j], vbuf[j]);
+      }
+      gi += DEPTH;
+    }
+#pragma unroll
+    for (int j = 0; j < DEPTH; j++)
+      if (gi + j < ngroups) process_group(gi + j, kbuf[j], vbuf[j]);
+  }
+
+  // fold the token-slot partials: acc holds per-(ts) sums
+#pragma unroll
+  for (int g = 0; g < GW; g++)
+#pragma unroll
+    for (int i = 0; i < ND; i++) {
+      float a = acc[g][i];
+#pragma unroll
+      for (int off = DP; off < 64; off <<= 1)
+        a += __shfl_xor(a, off, WAVE_SIZE);
+      acc[g][i] = a;
+    }
+
+  if (TG == 1) {
+    // each wave covered the whole chunk for its heads: write directly
+    if (ts == 0) {
+#pragma unroll
+      for (int g = 0; g < GW; g++) {
+        const int qh = h * G + hoff + g;
+#pragma unroll
+        for (int i = 0; i < ND; i++) {
+          const int d = dp * ND + i;
+          if (C == 1) {
+            out[((int64_t)b * Hq + qh) * hd + d] =
+                f32_to_bf16(l[g] > 0.f ? acc[g][i] / l[g] : 0.f);
+          } else {
+            partial[(((int64_t)b * Hq + qh) * C + c) * hd + d] = acc[g][i];
+          }
+        }
+        if (C > 1 && dp == 0) {
+          float* mlp = ml + (((int64_t)b * Hq + qh) * C + c) * 2;
+          mlp[0] = m[g]; mlp[1] = l[g];
+        }
+      }
+    }
+    return;
+  }
+
+  // cross-token-group merge via LDS: wave (tg, hg) writes its heads' slice.
+  __syncthreads();
+  float* my = merge + tg * G * (hd + 2);
+  if (ts == 0) {
+#pragma unroll
+    for (int g = 0; g < GW; g++) {
+      const int gq = hoff + g;
+#pragma unroll
+      for (int i = 0; i < ND; i++) my[gq * (hd + 2) + dp * ND + i] = acc[g][i];
+      if (dp == 0) { my[gq * (hd + 2) + hd] = m[g]; my[gq * (hd + 2) + hd + 1] = l[g]; }
+    }
+  }
+  __syncthreads();
+
+  for (int i = threadIdx.x; i < G * hd; i += kBlock) {
+    const int g = i / hd;
+    const int d = i % hd;
+    float mw[TG], lw[TG];
+    float mstar = kNegInf;
+#pragma unroll
+    for (int w = 0; w < TG; w++) {
+      mw[w] = merge[w * G * (hd + 2) + g * (hd + 2) + hd];
+      lw[w] = merge[w * G * (hd + 2) + g * (hd + 2) + hd + 1];
+      mstar = fmaxf(mstar, mw[w]);
+    }
+    float lsum = 0.f, asum = 0.f;
+#pragma unroll
+    for (int w = 0; w < TG; w++) {
+      const float corr = (lw[w] > 0.f) ? __expf(mw[w] - mstar) : 0.f;
+      lsum += lw[w] * corr;
+      asum += merge[w * G * (hd + 2) + g * (hd + 2) + d] * corr;
+    }
+    const int qh = h * G + g;
+    if (C == 1) {
+      out[((int64_t)b * Hq + qh) * hd + d] =
+          f32_to_bf16(lsum > 0.f ? asum / lsum : 0.f);
+    } else {
+      partial[(((int64_t)b * Hq + qh) * C + c) * hd + d] = asum;
+      if (d == 0) {
+        float* mlp = ml + (((int64_t)b * Hq + qh) * C + c) * 2;
+        mlp[0] = mstar; mlp[1] = lsum;
+      }
+    }
+  }
+}
+
+// Phase 2: merge chunk partials. grid (B, Hq), block = 128.
+__global__ inline void paged_decode_phase2(short* __restrict__ out,
+                                           const float* __restrict__ partial,
+                                           const float* __restrict__ ml,
+                                           const int32_t* __restrict__ ctx_lens,
+                                           int Hq, int C, int hd) {
+  const int b = blockIdx.x;
+  const int qh = blockIdx.y;
+  const int nc = min(C, (ctx_lens[b] + kChunk - 1) / kChunk);
+  const float* mlp = ml + (((int64_t)b * Hq + qh) * C) * 2;
+
+  __shared__ float smax[1];
+  float mstar = kNegInf;
+  for (int c = threadIdx.x; c < nc; c += blockDim.x) mstar = fmaxf(mstar, mlp[2 * c]);
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1)
+    mstar = fmaxf(mstar, __shfl_xor(mstar, off, WAVE_SIZE));
+  if (threadIdx.x == 0) smax[0] = mstar;
+  __syncthreads();
+  if (threadIdx.x == 64) smax[0] = fmaxf(smax[0], mstar);
+  __syncthreads();
+  mstar = smax[0];
+
+  for (int d = threadIdx.x; d < hd; d += blockDim.x) {
+    float asum = 0.f, lsum = 0.f;
+    for (int c = 0; c < nc; c++) {
+      const float lc = mlp[2 * c + 1];
+      if (lc <= 0.f) continue;
+      const float corr = __expf(mlp[2 * c] - mstar);
+      asum += partial[(((int64_t)b * Hq + qh) * C + c) * hd + d] * corr;
+      lsum += lc * corr;
+    }
+    out[((int64_t)b * Hq + qh) * hd + d] =
+        f32_to_bf16(lsum > 0.f ? asum / lsum : 0.f);
+  }
+}
+
+// LDS bytes for phase 1
+inline int phase1_lds_bytes(int G, int HS, int hd) {
+  const int TG = 4 / HS;
+  const int merge_f = (TG > 1) ? TG * G * (hd + 2) : 0;
+  return merge_f * 4 + G * hd * 2;
+}
+
+}  // namespace decode_attn
