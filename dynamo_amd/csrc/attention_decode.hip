@@ -42,11 +42,11 @@ void paged_attention_decode(torch::Tensor out, torch::Tensor q,
       (const short*)vcache.data_ptr(), page_table.data_ptr<int32_t>(),        \
       ctx_lens.data_ptr<int32_t>(), (float)scale, B, Hkv, C, max_pages,       \
       log2_ps, hd)
-  switch (G) {
+  switch (G) {  // combos picked by benchmarks/decode_sweep on MI355X
     case 1: LAUNCH_G(1, 8, 1, 2); break;
     case 2: LAUNCH_G(2, 8, 2, 2); break;
-    case 4: LAUNCH_G(4, 16, 2, 2); break;
-    case 8: LAUNCH_G(8, 16, 4, 2); break;
+    case 4: LAUNCH_G(4, 16, 2, 4); break;
+    case 8: LAUNCH_G(8, 8, 2, 2); break;
     default: TORCH_CHECK(false, "unsupported GQA group size ", G);
   }
 #undef LAUNCH_G
