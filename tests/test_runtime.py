@@ -1,0 +1,174 @@
+"""Runtime-layer tests: codec, request plane streaming RPC, discovery,
+endpoint registration + push-client routing, cancellation, worker death."""
+import asyncio
+import time
+
+import pytest
+
+from dynamo_amd.runtime import (DistributedRuntime, EndpointError,
+                                FileDiscovery, Instance, MemoryDiscovery,
+                                NoInstancesError)
+from dynamo_amd.runtime.codec import decode_frame, decode_prefix, encode_frame
+
+
+def test_codec_roundtrip():
+    h = {"type": "req", "rid": 3, "endpoint": "gen"}
+    b = {"tokens": list(range(100)), "text": "héllo", "b": b"\x00\x01"}
+    frame = encode_frame(h, b)
+    hlen, blen, csum = decode_prefix(frame[:24])
+    h2, b2 = decode_frame(frame[24:24 + hlen], frame[24 + hlen:], csum)
+    assert h2 == h
+    assert b2["tokens"] == b["tokens"]
+    assert b2["b"] == b["b"]
+    with pytest.raises(ValueError):
+        bad = frame[:30] + bytes([frame[30] ^ 1]) + frame[31:]
+        decode_frame(bad[24:24 + hlen], bad[24 + hlen:], csum)
+
+
+def run(coro):
+    return asyncio.new_event_loop().run_until_complete(coro)
+
+
+def test_request_plane_streaming():
+    async def main():
+        shared = MemoryDiscovery()
+        rt = DistributedRuntime(shared)
+        comp = rt.namespace("ns").component("backend")
+
+        async def gen(payload, ctx):
+            for i in range(payload["n"]):
+                yield {"i": i}
+        comp.serve_endpoint("generate", gen)
+        await comp.register()
+
+        client = rt.namespace("ns").component("backend").endpoint("generate").client()
+        client.runtime = rt  # same-process client
+        chunks = [c async for c in client.generate({"n": 5})]
+        assert chunks == [{"i": i} for i in range(5)]
+        await rt.shutdown()
+    run(main())
+
+
+def test_request_plane_error_propagation():
+    async def main():
+        rt = DistributedRuntime(MemoryDiscovery())
+        comp = rt.namespace("ns").component("backend")
+
+        async def bad(payload, ctx):
+            yield {"ok": 1}
+            raise RuntimeError("boom")
+        comp.serve_endpoint("generate", bad)
+        await comp.register()
+        client = comp.endpoint("generate").client()
+        with pytest.raises(EndpointError, match="boom"):
+            async for _ in client.generate({}):
+                pass
+        await rt.shutdown()
+    run(main())
+
+
+def test_cancellation_propagates():
+    async def main():
+        rt = DistributedRuntime(MemoryDiscovery())
+        comp = rt.namespace("ns").component("backend")
+        seen = {"max": 0, "cancelled": False}
+
+        async def slow(payload, ctx):
+            for i in range(1000):
+                if ctx.cancelled:
+                    seen["cancelled"] = True
+                    return
+                seen["max"] = i
+                yield {"i": i}
+                await asyncio.sleep(0.001)
+        comp.serve_endpoint("generate", slow)
+        await comp.register()
+        client = comp.endpoint("generate").client()
+
+        async def consume():
+            async for c in client.generate({}):
+                if c["i"] >= 3:
+                    raise asyncio.CancelledError
+        with pytest.raises(asyncio.CancelledError):
+            await consume()
+        await asyncio.sleep(0.1)
+        assert seen["max"] < 999  # handler stopped early
+        await rt.shutdown()
+    run(main())
+
+
+def test_round_robin_across_instances():
+    async def main():
+        shared = MemoryDiscovery()
+        rts = []
+        for w in range(3):
+            rt = DistributedRuntime(shared)
+            comp = rt.namespace("ns").component("backend")
+
+            async def gen(payload, ctx, w=w):
+                yield {"worker": w}
+            comp.serve_endpoint("generate", gen)
+            await comp.register()
+            rts.append(rt)
+        client_rt = DistributedRuntime(shared)
+        client = client_rt.namespace("ns").component("backend").endpoint(
+            "generate").client()
+        seen = set()
+        for _ in range(9):
+            r = await client.call({})
+            seen.add(r["worker"])
+        assert seen == {0, 1, 2}
+        for rt in rts:
+            await rt.shutdown()
+        await client_rt.shutdown()
+    run(main())
+
+
+def test_dead_worker_inhibited_and_failover():
+    async def main():
+        shared = MemoryDiscovery()
+        rt1 = DistributedRuntime(shared)
+        c1 = rt1.namespace("ns").component("backend")
+
+        async def gen1(payload, ctx):
+            yield {"worker": 1}
+        c1.serve_endpoint("generate", gen1)
+        await c1.register()
+
+        rt2 = DistributedRuntime(shared)
+        c2 = rt2.namespace("ns").component("backend")
+
+        async def gen2(payload, ctx):
+            yield {"worker": 2}
+        c2.serve_endpoint("generate", gen2)
+        await c2.register()
+
+        # kill worker 1's server but leave it in discovery (stale instance)
+        await rt1.server.stop(drain=False)
+
+        client_rt = DistributedRuntime(shared)
+        client = client_rt.namespace("ns").component("backend").endpoint(
+            "generate").client()
+        results = []
+        for _ in range(4):
+            try:
+                results.append((await client.call({}))["worker"])
+            except EndpointError:
+                pass  # first hit on the dead worker inhibits it
+        assert results and all(r == 2 for r in results)
+        await rt2.shutdown()
+        await client_rt.shutdown()
+    run(main())
+
+
+def test_file_discovery_lease(tmp_path):
+    d = FileDiscovery(str(tmp_path), ttl=0.2)
+    inst = Instance("ns", "backend", "abc", "127.0.0.1:1", ["generate"])
+    d.register(inst)
+    assert len(d.list("ns", "backend")) == 1
+    time.sleep(0.3)
+    assert d.list("ns", "backend") == []  # lease expired
+    d.refresh(inst)
+    assert len(d.list("ns", "backend")) == 1
+    d.deregister(inst)
+    assert d.list("ns", "backend") == []
