@@ -47,9 +47,9 @@ class ForwardPassMetrics:
 
 class LLMEngine:
     def __init__(self, cfg: EngineConfig, tp: Optional[TPContext] = None,
-                 seed: int = 0):
+                 seed: int = 0, runner=None):
         self.cfg = cfg
-        self.runner = ModelRunner(cfg, tp, seed)
+        self.runner = runner if runner is not None else ModelRunner(cfg, tp, seed)
         self.alloc = PageAllocator(self.runner.num_pages, cfg.page_size,
                                    cfg.enable_prefix_caching)
         self.scheduler = Scheduler(cfg, self.alloc)

@@ -1,0 +1,207 @@
+"""OpenAI-compatible HTTP frontend (FastAPI).
+
+Re-creates the serving surface of the reference's Axum frontend
+(ai-dynamo/dynamo lib/llm/src/http/service/openai.rs:3875-4165):
+/v1/models, /v1/completions, /v1/chat/completions (SSE streaming and
+unary), /health, /metrics (Prometheus). Client disconnects propagate as
+stream cancellation (http/service/disconnect.rs parity).
+"""
+from __future__ import annotations
+
+import asyncio
+import json
+import time
+import uuid
+from typing import AsyncIterator, List, Optional, Union
+
+from fastapi import FastAPI, HTTPException, Request
+from fastapi.responses import JSONResponse, StreamingResponse
+from pydantic import BaseModel, Field
+
+from prometheus_client import (CONTENT_TYPE_LATEST, Counter, Histogram,
+                               generate_latest)
+
+from .service import ModelManager
+
+REQS = Counter("dynamo_amd_requests_total", "requests", ["model", "route"])
+TTFT = Histogram("dynamo_amd_ttft_seconds", "time to first token", ["model"])
+LATENCY = Histogram("dynamo_amd_request_seconds", "request latency", ["model"])
+
+
+class CompletionRequest(BaseModel):
+    model: str = ""
+    prompt: Union[str, List[int]] = ""
+    max_tokens: int = 128
+    temperature: float = 0.0
+    top_p: float = 1.0
+    top_k: int = 0
+    stream: bool = False
+    stop: Optional[Union[str, List[str]]] = None
+    seed: int = 0
+    ignore_eos: bool = False
+
+
+class ChatMessage(BaseModel):
+    role: str
+    content: str
+
+
+class ChatRequest(BaseModel):
+    model: str = ""
+    messages: List[ChatMessage] = Field(default_factory=list)
+    max_tokens: int = 128
+    temperature: float = 0.0
+    top_p: float = 1.0
+    top_k: int = 0
+    stream: bool = False
+    seed: int = 0
+    ignore_eos: bool = False
+
+
+def build_app(manager: ModelManager) -> FastAPI:
+    app = FastAPI(title="dynamo_amd", version="0.1.0")
+    app.state.manager = manager
+
+    @app.get("/health")
+    async def health():
+        return {"status": "ok", "models": list(manager.models)}
+
+    @app.get("/metrics")
+    async def metrics():
+        from fastapi import Response
+        return Response(generate_latest(), media_type=CONTENT_TYPE_LATEST)
+
+    @app.get("/v1/models")
+    async def models():
+        return {"object": "list", "data": [
+            {"id": name, "object": "model", "owned_by": "dynamo_amd",
+             "created": int(e.card.get("registered_at", time.time()))
+             if isinstance(e.card, dict) else int(time.time())}
+            for name, e in manager.models.items()]}
+
+    async def _run(entry, token_ids, req, rid) -> AsyncIterator[dict]:
+        sampling = {"temperature": req.temperature, "top_p": req.top_p,
+                    "top_k": req.top_k, "seed": req.seed}
+        eos = getattr(entry.tokenizer, "eos_id", None)
+        stop = {"max_tokens": req.max_tokens,
+                "ignore_eos": req.ignore_eos,
+                "stop_token_ids": [eos] if eos is not None else []}
+        async for chunk in manager.generate_tokens(entry, token_ids, sampling,
+                                                   stop, request_id=rid):
+            yield chunk
+
+    def _entry_or_404(model):
+        try:
+            return manager.get(model)
+        except KeyError as e:
+            raise HTTPException(404, str(e))
+
+    @app.post("/v1/completions")
+    async def completions(req: CompletionRequest, raw: Request):
+        entry = _entry_or_404(req.model)
+        REQS.labels(entry.name, "completions").inc()
+        if isinstance(req.prompt, list):
+            token_ids = list(req.prompt)
+        else:
+            token_ids = entry.tokenizer.encode(req.prompt)
+        rid = f"cmpl-{uuid.uuid4().hex[:24]}"
+        t0 = time.time()
+
+        if req.stream:
+            async def sse():
+                produced: List[int] = []
+                first = True
+                try:
+                    async for chunk in _run(entry, token_ids, req, rid):
+                        if await raw.is_disconnected():
+                            break
+                        if first:
+                            TTFT.labels(entry.name).observe(time.time() - t0)
+                            first = False
+                        new = chunk.get("token_ids", [])
+                        prev = len(produced)
+                        produced.extend(new)
+                        text = entry.tokenizer.decode_incremental(produced, prev)
+                        data = {"id": rid, "object": "text_completion",
+                                "model": entry.name, "choices": [{
+                                    "index": 0, "text": text,
+                                    "finish_reason": chunk.get("finish_reason")}]}
+                        yield f"data: {json.dumps(data)}\n\n"
+                    yield "data: [DONE]\n\n"
+                finally:
+                    LATENCY.labels(entry.name).observe(time.time() - t0)
+            return StreamingResponse(sse(), media_type="text/event-stream")
+
+        produced: List[int] = []
+        finish = None
+        async for chunk in _run(entry, token_ids, req, rid):
+            produced.extend(chunk.get("token_ids", []))
+            finish = chunk.get("finish_reason") or finish
+        LATENCY.labels(entry.name).observe(time.time() - t0)
+        return {
+            "id": rid, "object": "text_completion", "created": int(t0),
+            "model": entry.name,
+            "choices": [{"index": 0,
+                         "text": entry.tokenizer.decode(produced),
+                         "finish_reason": finish or "stop",
+                         "token_ids": produced}],
+            "usage": {"prompt_tokens": len(token_ids),
+                      "completion_tokens": len(produced),
+                      "total_tokens": len(token_ids) + len(produced)},
+        }
+
+    @app.post("/v1/chat/completions")
+    async def chat(req: ChatRequest, raw: Request):
+        entry = _entry_or_404(req.model)
+        REQS.labels(entry.name, "chat").inc()
+        prompt = entry.templater.render([m.model_dump() for m in req.messages])
+        token_ids = entry.tokenizer.encode(prompt)
+        rid = f"chatcmpl-{uuid.uuid4().hex[:24]}"
+        t0 = time.time()
+
+        if req.stream:
+            async def sse():
+                produced: List[int] = []
+                first = True
+                try:
+                    async for chunk in _run(entry, token_ids, req, rid):
+                        if await raw.is_disconnected():
+                            break
+                        delta = {}
+                        if first:
+                            TTFT.labels(entry.name).observe(time.time() - t0)
+                            delta["role"] = "assistant"
+                            first = False
+                        prev = len(produced)
+                        produced.extend(chunk.get("token_ids", []))
+                        delta["content"] = entry.tokenizer.decode_incremental(
+                            produced, prev)
+                        data = {"id": rid, "object": "chat.completion.chunk",
+                                "model": entry.name, "choices": [{
+                                    "index": 0, "delta": delta,
+                                    "finish_reason": chunk.get("finish_reason")}]}
+                        yield f"data: {json.dumps(data)}\n\n"
+                    yield "data: [DONE]\n\n"
+                finally:
+                    LATENCY.labels(entry.name).observe(time.time() - t0)
+            return StreamingResponse(sse(), media_type="text/event-stream")
+
+        produced: List[int] = []
+        finish = None
+        async for chunk in _run(entry, token_ids, req, rid):
+            produced.extend(chunk.get("token_ids", []))
+            finish = chunk.get("finish_reason") or finish
+        LATENCY.labels(entry.name).observe(time.time() - t0)
+        return {
+            "id": rid, "object": "chat.completion", "created": int(t0),
+            "model": entry.name,
+            "choices": [{"index": 0, "message": {
+                "role": "assistant",
+                "content": entry.tokenizer.decode(produced)},
+                "finish_reason": finish or "stop"}],
+            "usage": {"prompt_tokens": len(token_ids),
+                      "completion_tokens": len(produced),
+                      "total_tokens": len(token_ids) + len(produced)},
+        }
+
+    return app

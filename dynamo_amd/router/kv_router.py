@@ -1,0 +1,213 @@
+"""KV-aware router: prefix-overlap + load cost routing.
+
+Native re-creation of the reference's kv-router
+(ai-dynamo/dynamo lib/kv-router): the C++ indexer (dynamo_amd._core,
+RadixTree::find_matches parity) is fed by per-worker KV-event streams; the
+worker selector implements the DefaultWorkerSelector cost
+(kv-router/src/scheduling/selector/default.rs:93,217):
+
+  logit = prefill_load_scale * max(0, prefill_blocks - overlap_credit)
+        + decode_cost_blocks
+        + decode_active_request_weight * active_requests
+
+with softmax temperature sampling (default.rs:23-69) or argmin at T=0.
+Load state comes from get_perf_metrics polling + local in-flight tracking.
+"""
+from __future__ import annotations
+
+import asyncio
+import logging
+import math
+import random
+import time
+from dataclasses import dataclass, field
+from typing import Dict, List, Optional
+
+from dynamo_amd import _core
+from dynamo_amd.runtime import DistributedRuntime, PushClient
+
+log = logging.getLogger("dynamo_amd.router")
+
+
+@dataclass
+class RouterConfig:
+    """KvRouterConfig parity (kv-router/src/scheduling/config.rs:660)."""
+    mode: str = "kv"               # kv|round_robin|random|least_loaded|p2c
+    block_size: int = 64
+    block_salt: int = 0
+    overlap_score_weight: float = 1.0
+    decode_active_request_weight: float = 1.0
+    prefill_load_scale: float = 1.0
+    router_temperature: float = 0.0
+    metrics_poll_interval: float = 1.0
+
+
+@dataclass
+class WorkerState:
+    instance_id: str
+    active_requests: int = 0        # local in-flight tracking
+    active_blocks: int = 0
+    kv_usage: float = 0.0
+    num_waiting: int = 0
+    total_kv_pages: int = 1
+    last_metrics_ts: float = 0.0
+    event_task: Optional[asyncio.Task] = None
+
+
+class KvRouter:
+    """Routes token sequences to worker instances of one component."""
+
+    def __init__(self, runtime: DistributedRuntime, namespace: str,
+                 component: str, cfg: RouterConfig | None = None,
+                 endpoint: str = "generate"):
+        self.runtime = runtime
+        self.namespace = namespace
+        self.component = component
+        self.cfg = cfg or RouterConfig()
+        self.client = PushClient(runtime, namespace, component, endpoint)
+        self.indexer = _core.KvIndexer()
+        self.workers: Dict[str, WorkerState] = {}
+        self._rr = 0
+        self._tasks: List[asyncio.Task] = []
+        self._started = False
+
+    async def start(self):
+        if not self._started:
+            self._started = True
+            self._tasks.append(asyncio.create_task(self._watch_loop()))
+        return self
+
+    async def stop(self):
+        for t in self._tasks:
+            t.cancel()
+        for w in self.workers.values():
+            if w.event_task:
+                w.event_task.cancel()
+
+    # -- discovery + event/metrics ingestion ---------------------------
+    async def _watch_loop(self):
+        while True:
+            try:
+                self._sync_workers()
+            except Exception:
+                log.exception("router watch failed")
+            await asyncio.sleep(self.cfg.metrics_poll_interval)
+
+    def _sync_workers(self):
+        live = {i.instance_id: i for i in self.client.instances()}
+        for iid, inst in live.items():
+            ws = self.workers.setdefault(iid, WorkerState(iid))
+            if ws.event_task is None:  # not yet subscribed (may have been
+                # pre-created by begin_request accounting)
+                ws.event_task = asyncio.create_task(
+                    self._consume_events(inst.address, iid))
+                self._tasks.append(asyncio.create_task(
+                    self._poll_metrics(inst.address, iid)))
+        for iid in list(self.workers):
+            if iid not in live:
+                ws = self.workers.pop(iid)
+                if ws.event_task:
+                    ws.event_task.cancel()
+                self.indexer.remove_worker(self._wid(iid))
+
+    @staticmethod
+    def _wid(instance_id: str) -> int:
+        return int(instance_id[:12], 16)
+
+    async def _consume_events(self, address: str, iid: str):
+        wid = self._wid(iid)
+        try:
+            async for batch in self.runtime.client.call_stream(
+                    address, f"{self.component}.kv_events", {}):
+                for ev in batch:
+                    if ev["kind"] == "stored":
+                        self.indexer.apply_stored(wid, ev["hashes"])
+                    elif ev["kind"] == "removed":
+                        self.indexer.apply_removed(wid, ev["hashes"])
+                    elif ev["kind"] == "cleared":
+                        self.indexer.clear_worker(wid)
+        except Exception:
+            pass  # worker died; watch loop will clean up
+
+    async def _poll_metrics(self, address: str, iid: str):
+        while iid in self.workers:
+            try:
+                m = await self.runtime.client.call(
+                    address, f"{self.component}.get_perf_metrics", {})
+                ws = self.workers.get(iid)
+                if ws and m:
+                    ws.kv_usage = m.get("kv_usage", 0.0)
+                    ws.num_waiting = m.get("num_waiting", 0)
+                    ws.total_kv_pages = max(1, m.get("total_kv_pages", 1))
+                    ws.last_metrics_ts = time.time()
+            except Exception:
+                pass
+            await asyncio.sleep(self.cfg.metrics_poll_interval)
+
+    # -- selection ------------------------------------------------------
+    def select(self, token_ids: List[int]) -> Optional[str]:
+        """Pick a worker instance_id for this token sequence."""
+        insts = self.client.instances()
+        if not insts:
+            return None
+        mode = self.cfg.mode
+        if mode == "round_robin":
+            self._rr += 1
+            return insts[self._rr % len(insts)].instance_id
+        if mode == "random":
+            return random.choice(insts).instance_id
+        if mode in ("least_loaded", "p2c"):
+            cand = insts if mode == "least_loaded" else random.sample(
+                insts, min(2, len(insts)))
+            return min(cand, key=lambda i: self._load(i.instance_id)).instance_id
+
+        # kv mode
+        bs = self.cfg.block_size
+        hashes = _core.chain_hashes(token_ids, bs, self.cfg.block_salt)
+        matches = self.indexer.find_matches(hashes)
+        prefill_blocks = (len(token_ids) + bs - 1) // bs
+        logits = []
+        for inst in insts:
+            ws = self.workers.get(inst.instance_id) or WorkerState(
+                inst.instance_id)
+            overlap = matches.get(self._wid(inst.instance_id), 0)
+            cost = (self.cfg.prefill_load_scale
+                    * max(0.0, prefill_blocks
+                          - self.cfg.overlap_score_weight * overlap)
+                    + ws.active_blocks
+                    + self.cfg.decode_active_request_weight * ws.active_requests)
+            logits.append(cost)
+        T = self.cfg.router_temperature
+        if T <= 0:
+            best = min(range(len(insts)), key=lambda i: logits[i])
+            return insts[best].instance_id
+        # softmax sampling over -cost/T
+        ws_ = [math.exp(-(c - min(logits)) / T) for c in logits]
+        total = sum(ws_)
+        r = random.random() * total
+        acc = 0.0
+        for i, w in enumerate(ws_):
+            acc += w
+            if r <= acc:
+                return insts[i].instance_id
+        return insts[-1].instance_id
+
+    def _load(self, iid: str) -> float:
+        ws = self.workers.get(iid)
+        if not ws:
+            return 0.0
+        return ws.active_requests + ws.kv_usage
+
+    # -- request accounting ---------------------------------------------
+    def begin_request(self, iid: str, token_ids: List[int]):
+        ws = self.workers.setdefault(iid, WorkerState(iid))
+        ws.active_requests += 1
+        ws.active_blocks += (len(token_ids) + self.cfg.block_size - 1) // self.cfg.block_size
+
+    def end_request(self, iid: str, token_ids: List[int]):
+        ws = self.workers.get(iid)
+        if ws:
+            ws.active_requests = max(0, ws.active_requests - 1)
+            ws.active_blocks = max(
+                0, ws.active_blocks
+                - (len(token_ids) + self.cfg.block_size - 1) // self.cfg.block_size)

@@ -49,6 +49,7 @@ class RequestPlaneServer:
         self._inflight: Dict[tuple, asyncio.Task] = {}
         self._contexts: Dict[tuple, RequestContext] = {}
         self._conn_counter = itertools.count()
+        self._writers: set = set()
 
     def add_endpoint(self, name: str, handler: Handler):
         self.handlers[name] = handler
@@ -72,11 +73,18 @@ class RequestPlaneServer:
                                  return_exceptions=True)
         for t in self._inflight.values():
             t.cancel()
+        # close established connections (clients see EOF and fail over)
+        for w in list(self._writers):
+            try:
+                w.close()
+            except Exception:
+                pass
 
     async def _on_conn(self, reader: asyncio.StreamReader,
                        writer: asyncio.StreamWriter):
         cid = next(self._conn_counter)
         wlock = asyncio.Lock()
+        self._writers.add(writer)
         try:
             while True:
                 frame = await read_frame(reader)
@@ -105,6 +113,7 @@ class RequestPlaneServer:
                 ctx = self._contexts.get(key)
                 if ctx:
                     ctx.cancel()
+            self._writers.discard(writer)
             writer.close()
 
     async def _run_handler(self, endpoint, body, ctx, rid, writer, wlock, key):

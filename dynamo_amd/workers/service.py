@@ -1,0 +1,270 @@
+"""WorkerService: serves an engine (native or mock) on the request plane.
+
+Fulfills the worker contract the reference defines for its engine adapters
+(components/src/dynamo/vllm/main.py:149 worker()):
+  - registers a model card + runtime config in discovery
+  - serves endpoints: generate, clear_kv_blocks, get_perf_metrics,
+    kv_events (stream), release_kv (disagg), lora stubs
+  - prefill workers finish at 1 token, hold KV and return
+    disaggregated_params; decode workers pull KV over xGMI before decoding
+
+Request payload (PreprocessedRequest parity,
+lib/llm/src/protocols/common/preprocessor.rs:243):
+  {request_id, token_ids, sampling_options{temperature, top_p, top_k, seed},
+   stop_conditions{max_tokens, stop_token_ids, ignore_eos},
+   routing{worker_instance_id?}, prefill_result?, annotations?}
+Response chunks (LLMEngineOutput parity, llm_backend.rs:163):
+  {token_ids: [t], finish_reason?, disaggregated_params?}
+"""
+from __future__ import annotations
+
+import asyncio
+import logging
+import time
+from typing import Any, AsyncIterator, Dict, Optional
+
+from dynamo_amd.engine.engine import LLMEngine
+from dynamo_amd.engine.scheduler import Request, SamplingParams
+from dynamo_amd.engine.kv_cache import SequenceKV
+from dynamo_amd.runtime import DistributedRuntime, RequestContext
+from dynamo_amd.disagg.transfer import KvPuller, pool_transfer_metadata
+
+log = logging.getLogger("dynamo_amd.worker")
+
+
+def make_sampling(payload: dict) -> SamplingParams:
+    so = payload.get("sampling_options") or {}
+    sc = payload.get("stop_conditions") or {}
+    return SamplingParams(
+        max_tokens=int(sc.get("max_tokens", 128)),
+        temperature=float(so.get("temperature", 0.0)),
+        top_p=float(so.get("top_p", 1.0)),
+        top_k=int(so.get("top_k", 0)),
+        stop_token_ids=list(sc.get("stop_token_ids", [])),
+        ignore_eos=bool(sc.get("ignore_eos", False)),
+        seed=int(so.get("seed", 0)),
+    )
+
+
+class WorkerService:
+    def __init__(self, engine: LLMEngine, runtime: DistributedRuntime,
+                 namespace: str = "dynamo", component: str = "backend",
+                 model_name: Optional[str] = None):
+        self.engine = engine
+        self.runtime = runtime
+        self.namespace = namespace
+        self.component_name = component
+        self.model_name = model_name or engine.cfg.model.name
+        self.comp = runtime.namespace(namespace).component(component)
+        self.queues: Dict[str, asyncio.Queue] = {}
+        self.kv_event_subs: list = []
+        self._work = asyncio.Event()
+        self._loop_task: Optional[asyncio.Task] = None
+        self._engine_lock = asyncio.Lock()
+        self._req_counter = 0
+        self._puller: Optional[KvPuller] = None
+        self.worker_type = engine.cfg.worker_type
+
+    # ------------------------------------------------------------------
+    def model_card(self) -> dict:
+        cfg = self.engine.cfg
+        return {
+            "name": self.model_name,
+            "model_config": cfg.model.to_dict(),
+            "context_length": cfg.max_model_len,
+            "kv_cache_block_size": cfg.page_size,
+            "block_salt": cfg.block_salt,
+            "runtime_config": {
+                "total_kv_blocks": self.engine.alloc.num_pages,
+                "max_num_seqs": cfg.max_num_seqs,
+                "max_num_batched_tokens": cfg.max_batched_tokens,
+                "worker_type": self.worker_type,
+            },
+        }
+
+    async def start(self):
+        self.comp.serve_endpoint("generate", self.generate)
+        self.comp.serve_endpoint("clear_kv_blocks", self.clear_kv_blocks)
+        self.comp.serve_endpoint("get_perf_metrics", self.get_perf_metrics)
+        self.comp.serve_endpoint("kv_events", self.kv_events)
+        self.comp.serve_endpoint("release_kv", self.release_kv)
+        metadata = {"worker_type": self.worker_type}
+        if self.engine.runner.kv_pool is not None:
+            metadata["kv_transfer"] = pool_transfer_metadata(
+                self.comp.instance_id, self.engine.runner.kv_pool)
+        await self.comp.register(model_card=self.model_card(),
+                                 metadata=metadata)
+        self._loop_task = asyncio.create_task(self._engine_loop())
+        return self
+
+    async def stop(self):
+        if self._loop_task:
+            self._loop_task.cancel()
+        self.comp.deregister()
+
+    @property
+    def instance_id(self) -> str:
+        return self.comp.instance_id
+
+    # ------------------------------------------------------------------
+    async def _engine_loop(self):
+        while True:
+            if not self.engine.has_work():
+                self._work.clear()
+                await self._work.wait()
+            async with self._engine_lock:
+                outputs = await asyncio.to_thread(self.engine.step)
+            for so in outputs:
+                q = self.queues.get(so.req_id)
+                if q is not None:
+                    q.put_nowait(so)
+            # fan KV events to subscribers (drain regardless, to bound memory)
+            events = self.engine.drain_kv_events()
+            if events and self.kv_event_subs:
+                batch = [{"kind": e.kind, "hashes": list(e.hashes),
+                          "parent": e.parent} for e in events]
+                for q in self.kv_event_subs:
+                    q.put_nowait(batch)
+            await asyncio.sleep(0)
+
+    # ------------------------------------------------------------------
+    async def generate(self, payload: dict, ctx: RequestContext
+                       ) -> AsyncIterator[dict]:
+        self._req_counter += 1
+        req_id = payload.get("request_id") or f"req-{self._req_counter}"
+        tokens = list(payload["token_ids"])
+        sp = make_sampling(payload)
+        is_prefill_role = self.worker_type == "prefill"
+        if is_prefill_role:
+            sp = SamplingParams(max_tokens=1, temperature=sp.temperature,
+                                top_p=sp.top_p, top_k=sp.top_k,
+                                seed=sp.seed, ignore_eos=True)
+
+        q: asyncio.Queue = asyncio.Queue()
+        self.queues[req_id] = q
+        try:
+            async with self._engine_lock:
+                req = self.engine.add_request(req_id, tokens, sp)
+                if is_prefill_role:
+                    req.hold_kv = True
+                pr = payload.get("prefill_result")
+                if pr is not None:
+                    await self._attach_remote_kv(req, pr)
+            self._work.set()
+
+            while True:
+                get = asyncio.create_task(q.get())
+                done, _ = await asyncio.wait({get}, timeout=0.05)
+                if not done:
+                    get.cancel()
+                    if ctx.cancelled:
+                        async with self._engine_lock:
+                            self.engine.abort(req_id)
+                        return
+                    continue
+                so = get.result()
+                chunk: dict = {"token_ids": [so.new_token]}
+                if so.finished:
+                    chunk["finish_reason"] = so.finish_reason
+                    if is_prefill_role:
+                        chunk["disaggregated_params"] = \
+                            self._disagg_params(req)
+                    yield chunk
+                    return
+                yield chunk
+        finally:
+            self.queues.pop(req_id, None)
+
+    def _disagg_params(self, req: Request) -> dict:
+        return {
+            "prefill_instance_id": self.instance_id,
+            "page_ids": list(req.kv.pages),
+            "num_tokens": req.num_computed,
+            "first_token": req.output_tokens[0] if req.output_tokens else None,
+        }
+
+    async def _attach_remote_kv(self, req: Request, pr: dict):
+        """Decode side of the disagg handoff: pull KV pages over xGMI."""
+        insts = self.runtime.discovery.list(self.namespace)
+        src_meta = None
+        src_addr = None
+        for inst in insts:
+            if inst.instance_id == pr["prefill_instance_id"]:
+                src_meta = inst.metadata.get("kv_transfer")
+                src_addr = inst.address
+                break
+        if src_meta is None:
+            raise RuntimeError(
+                f"prefill instance {pr['prefill_instance_id']} not found")
+        if self._puller is None:
+            self._puller = KvPuller(self.engine.runner.kv_pool)
+        num_tokens = int(pr["num_tokens"])
+        kv = SequenceKV(self.engine.alloc, self.engine.cfg.block_salt)
+        kv.ensure_capacity(num_tokens)
+        src_pages = [int(p) for p in pr["page_ids"]]
+        npages = (num_tokens + self.engine.cfg.page_size - 1) // self.engine.cfg.page_size
+        await asyncio.to_thread(self._puller.pull, src_meta,
+                                src_pages[:npages], kv.pages[:npages])
+        req.kv = kv
+        req.num_computed = num_tokens
+        if pr.get("first_token") is not None:
+            req.output_tokens.append(int(pr["first_token"]))
+        # release the prefill side's hold
+        comp = pr["prefill_instance_id"]
+        try:
+            await self.runtime.client.call(
+                src_addr, f"{self.component_name}.release_kv",
+                {"request_id": req.req_id})
+        except Exception:
+            log.warning("release_kv to %s failed", comp)
+
+    # ------------------------------------------------------------------
+    async def clear_kv_blocks(self, payload, ctx):
+        async with self._engine_lock:
+            self.engine.clear_kv()
+        yield {"status": "ok"}
+
+    async def get_perf_metrics(self, payload, ctx):
+        m = self.engine.last_metrics
+        yield {
+            "worker_id": self.instance_id,
+            "worker_type": self.worker_type,
+            "step": m.step,
+            "num_running": self.engine.scheduler.num_running(),
+            "num_waiting": self.engine.scheduler.num_waiting(),
+            "kv_usage": self.engine.alloc.usage,
+            "total_kv_pages": self.engine.alloc.num_pages,
+            "num_tokens_step": m.num_tokens_step,
+            "prefill_tokens_step": m.prefill_tokens_step,
+            "decode_tokens_step": m.decode_tokens_step,
+            "step_time_ms": m.step_time_ms,
+            "ts": time.time(),
+        }
+
+    async def kv_events(self, payload, ctx):
+        """Streaming subscription: batches of KV events (router feed).
+
+        On subscribe, the current cached-block state is replayed as one
+        `stored` batch so late subscribers converge (the reference's router
+        rebuilds state from the event stream + dedup refcounts)."""
+        q: asyncio.Queue = asyncio.Queue()
+        async with self._engine_lock:
+            snapshot = list(self.engine.alloc.hash_to_page.keys())
+            self.kv_event_subs.append(q)
+        if snapshot:
+            yield [{"kind": "stored", "hashes": snapshot, "parent": None}]
+        try:
+            while not ctx.cancelled:
+                get = asyncio.create_task(q.get())
+                done, _ = await asyncio.wait({get}, timeout=0.1)
+                if not done:
+                    get.cancel()
+                    continue
+                yield get.result()
+        finally:
+            self.kv_event_subs.remove(q)
+
+    async def release_kv(self, payload, ctx):
+        async with self._engine_lock:
+            self.engine.release_held(payload["request_id"])
+        yield {"status": "ok"}
