@@ -1,0 +1,57 @@
+"""Frontend CLI: `python -m dynamo_amd.frontend --discovery file:/run/dyn`
+
+The analog of `python -m dynamo.frontend`
+(components/src/dynamo/frontend/main.py:33-40): watches discovery for
+model cards and serves the OpenAI HTTP API.
+"""
+from __future__ import annotations
+
+import argparse
+import asyncio
+import logging
+
+import uvicorn
+
+from dynamo_amd.router import RouterConfig
+from dynamo_amd.runtime import DistributedRuntime
+from .openai import build_app
+from .service import ModelManager
+
+
+def build_parser():
+    p = argparse.ArgumentParser("dynamo_amd.frontend")
+    p.add_argument("--discovery", default="memory")
+    p.add_argument("--namespace", default="dynamo")
+    p.add_argument("--host", default="127.0.0.1")
+    p.add_argument("--port", type=int, default=8000)
+    p.add_argument("--router-mode", default="kv",
+                   choices=["kv", "round_robin", "random", "least_loaded",
+                            "p2c"])
+    p.add_argument("--router-temperature", type=float, default=0.0)
+    return p
+
+
+async def async_main(args):
+    logging.basicConfig(level=logging.INFO)
+    rt = DistributedRuntime(args.discovery, host=args.host)
+    mgr = ModelManager(rt, namespace=args.namespace,
+                       router_cfg=RouterConfig(
+                           mode=args.router_mode,
+                           router_temperature=args.router_temperature))
+    await mgr.start()
+    app = build_app(mgr)
+    config = uvicorn.Config(app, host=args.host, port=args.port,
+                            log_level="warning")
+    server = uvicorn.Server(config)
+    print(f"FRONTEND_READY http://{args.host}:{args.port}", flush=True)
+    await server.serve()
+    await mgr.stop()
+    await rt.shutdown()
+
+
+def main():
+    asyncio.run(async_main(build_parser().parse_args()))
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,100 @@
+"""Worker CLI: `python -m dynamo_amd.workers [--mock] --model ...`
+
+The native analog of the reference's engine-worker entrypoints
+(`python -m dynamo.vllm`, components/src/dynamo/vllm/main.py:149): builds
+the engine (native CDNA4 or mock), registers the model card in discovery,
+and serves the worker endpoint contract on the request plane.
+"""
+from __future__ import annotations
+
+import argparse
+import asyncio
+import logging
+import signal
+
+from dynamo_amd.engine import EngineConfig, LLMEngine
+from dynamo_amd.models.registry import resolve_model_config
+from dynamo_amd.runtime import DistributedRuntime
+from .service import WorkerService
+
+
+def build_parser():
+    p = argparse.ArgumentParser("dynamo_amd.workers")
+    p.add_argument("--model", default="tiny-llama")
+    p.add_argument("--mock", action="store_true", help="GPU-free mock engine")
+    p.add_argument("--device", default=None, help="cuda:0 | cpu (default auto)")
+    p.add_argument("--discovery", default="memory",
+                   help="memory | file:/path (DYN_DISCOVERY_BACKEND analog)")
+    p.add_argument("--namespace", default="dynamo")
+    p.add_argument("--component", default=None,
+                   help="default: backend, or prefill for prefill workers")
+    p.add_argument("--worker-type", default="aggregated",
+                   choices=["aggregated", "prefill", "decode"])
+    p.add_argument("--page-size", type=int, default=64)
+    p.add_argument("--max-num-seqs", type=int, default=64)
+    p.add_argument("--max-batched-tokens", type=int, default=8192)
+    p.add_argument("--max-model-len", type=int, default=16384)
+    p.add_argument("--kv-pool-pages", type=int, default=0)
+    p.add_argument("--gpu-mem-fraction", type=float, default=0.9)
+    p.add_argument("--no-prefix-caching", action="store_true")
+    p.add_argument("--no-hip-graphs", action="store_true")
+    p.add_argument("--seed", type=int, default=0)
+    p.add_argument("--host", default="127.0.0.1")
+    return p
+
+
+def make_engine_from_args(args) -> LLMEngine:
+    mc = resolve_model_config(args.model)
+    if args.mock:
+        from dynamo_amd.mocker import make_mock_engine
+        return make_mock_engine(
+            model=mc, page_size=args.page_size, max_num_seqs=args.max_num_seqs,
+            max_batched_tokens=args.max_batched_tokens,
+            max_model_len=args.max_model_len,
+            num_pages=args.kv_pool_pages or 1024,
+            worker_type=args.worker_type)
+    import torch
+    device = args.device or ("cuda:0" if torch.cuda.is_available() else "cpu")
+    cfg = EngineConfig(
+        model=mc, device=device, page_size=args.page_size,
+        max_num_seqs=args.max_num_seqs,
+        max_batched_tokens=args.max_batched_tokens,
+        max_model_len=args.max_model_len,
+        kv_pool_pages=args.kv_pool_pages,
+        gpu_mem_fraction=args.gpu_mem_fraction,
+        enable_prefix_caching=not args.no_prefix_caching,
+        enable_hip_graphs=not args.no_hip_graphs,
+        worker_type=args.worker_type)
+    return LLMEngine(cfg, seed=args.seed)
+
+
+async def async_main(args):
+    logging.basicConfig(level=logging.INFO,
+                        format="%(asctime)s %(name)s %(levelname)s %(message)s")
+    engine = make_engine_from_args(args)
+    rt = DistributedRuntime(args.discovery, host=args.host)
+    component = args.component or (
+        "prefill" if args.worker_type == "prefill" else "backend")
+    ws = WorkerService(engine, rt, namespace=args.namespace,
+                       component=component)
+    await ws.start()
+    logging.info("worker %s (%s) serving %s on %s", ws.instance_id,
+                 args.worker_type, args.model, rt.server.address)
+    print(f"WORKER_READY {ws.instance_id} {rt.server.address}", flush=True)
+
+    stop = asyncio.Event()
+    loop = asyncio.get_running_loop()
+    for sig in (signal.SIGINT, signal.SIGTERM):
+        loop.add_signal_handler(sig, stop.set)
+    await stop.wait()
+    await ws.stop()
+    await rt.shutdown()
+
+
+def main():
+    args = build_parser().parse_args()
+    asyncio.run(async_main(args))
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,57 @@
+"""Multi-process E2E on CPU: subprocess mock workers + frontend subprocess
+over file discovery, exercised through real HTTP."""
+import sys
+import time
+
+import httpx
+import pytest
+
+from tests.proc_utils import ManagedProcess, worker_cmd
+
+
+@pytest.mark.timeout(180)
+def test_multiprocess_workers_and_frontend(tmp_path):
+    disc = f"file:{tmp_path}/disc"
+    workers = []
+    front = None
+    try:
+        for _ in range(2):
+            workers.append(ManagedProcess(
+                worker_cmd(mock=True, model="tiny-llama", discovery=disc,
+                           page_size=16),
+                ready_marker="WORKER_READY").start())
+        front = ManagedProcess(
+            [sys.executable, "-m", "dynamo_amd.frontend", "--discovery", disc,
+             "--port", "18231"],
+            ready_marker="FRONTEND_READY").start()
+        base = "http://127.0.0.1:18231"
+        # wait for model registration to reach the frontend
+        deadline = time.time() + 60
+        with httpx.Client(timeout=30) as client:
+            while time.time() < deadline:
+                try:
+                    r = client.get(base + "/health")
+                    if r.status_code == 200 and r.json()["models"]:
+                        break
+                except httpx.TransportError:
+                    pass
+                time.sleep(0.3)
+            else:
+                raise TimeoutError("frontend never saw the model")
+            r = client.post(base + "/v1/completions", json={
+                "model": "tiny-llama", "prompt": "multi process hello",
+                "max_tokens": 6})
+            assert r.status_code == 200, r.text
+            assert r.json()["usage"]["completion_tokens"] == 6
+            # streaming
+            with client.stream("POST", base + "/v1/completions", json={
+                    "model": "tiny-llama", "prompt": "stream", "max_tokens": 4,
+                    "stream": True}) as rs:
+                lines = [ln for ln in rs.iter_lines()
+                         if ln.startswith("data: ")]
+            assert lines[-1] == "data: [DONE]"
+    finally:
+        if front:
+            front.stop()
+        for w in workers:
+            w.stop()
