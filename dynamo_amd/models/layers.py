@@ -69,6 +69,21 @@ def init_weight(shape, device, dtype, std=0.02, generator=None):
     return w
 
 
+def init_sharded(shape, device, dtype, tp: "TPContext", dim: int, std=0.02):
+    """Generate the FULL weight deterministically, return this rank's shard.
+
+    Every rank consumes the same RNG stream, so shards are consistent and a
+    TP-N model computes exactly the same function as TP-1 (up to reduction
+    order) — which keeps disagg/TP determinism tests meaningful."""
+    full = init_weight(shape, device, dtype, std)
+    if tp.size == 1:
+        return full
+    n = shape[dim] // tp.size
+    shard = full.narrow(dim, tp.rank * n, n).contiguous()
+    del full
+    return shard
+
+
 class Attention(torch.nn.Module):
     """GQA attention over the paged KV cache (native HIP kernels)."""
 
@@ -81,9 +96,12 @@ class Attention(torch.nn.Module):
         self.hd = cfg.head_dim
         self.scale = self.hd ** -0.5
         D = cfg.hidden_size
-        qkv_out = (self.hq + 2 * self.hkv) * self.hd
-        self.wqkv = init_weight((qkv_out, D), device, dtype)
-        self.wo = init_weight((D, self.hq * self.hd), device, dtype)
+        wq = init_sharded((cfg.num_q_heads * self.hd, D), device, dtype, tp, 0)
+        wk = init_sharded((cfg.num_kv_heads * self.hd, D), device, dtype, tp, 0)
+        wv = init_sharded((cfg.num_kv_heads * self.hd, D), device, dtype, tp, 0)
+        self.wqkv = torch.cat([wq, wk, wv], dim=0)
+        self.wo = init_sharded((D, cfg.num_q_heads * self.hd), device, dtype,
+                               tp, 1)
 
     def forward(self, x, cos_sin, kcache, vcache, meta: AttnMetadata):
         T = x.shape[0]
@@ -121,8 +139,11 @@ class SwiGLUMLP(torch.nn.Module):
         D = cfg.hidden_size
         I = cfg.intermediate_size // tp.size
         self.I = I
-        self.w_gate_up = init_weight((2 * I, D), device, dtype)
-        self.w_down = init_weight((D, I), device, dtype)
+        wg = init_sharded((cfg.intermediate_size, D), device, dtype, tp, 0)
+        wu = init_sharded((cfg.intermediate_size, D), device, dtype, tp, 0)
+        self.w_gate_up = torch.cat([wg, wu], dim=0)
+        self.w_down = init_sharded((D, cfg.intermediate_size), device, dtype,
+                                   tp, 1)
 
     def forward(self, x):
         gu = linear(x, self.w_gate_up)

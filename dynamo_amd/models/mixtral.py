@@ -22,10 +22,14 @@ class MoEMLP(torch.nn.Module):
         D = cfg.hidden_size
         I = cfg.intermediate_size // tp.size
         self.I = I
+        from .layers import init_sharded
         self.router = init_weight((self.E, D), device, dtype)
-        # fused per-expert weights: [E, 2I, D] and [E, D, I]
-        self.w_gate_up = init_weight((self.E, 2 * I, D), device, dtype)
-        self.w_down = init_weight((self.E, D, I), device, dtype)
+        # fused per-expert weights: [E, 2I_local, D] and [E, D, I_local]
+        wg = init_sharded((self.E, cfg.intermediate_size, D), device, dtype, tp, 1)
+        wu = init_sharded((self.E, cfg.intermediate_size, D), device, dtype, tp, 1)
+        self.w_gate_up = torch.cat([wg, wu], dim=1)
+        self.w_down = init_sharded((self.E, D, cfg.intermediate_size), device,
+                                   dtype, tp, 2)
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         T, D = x.shape
