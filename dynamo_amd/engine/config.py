@@ -100,6 +100,7 @@ class EngineConfig:
     gpu_mem_fraction: float = 0.90
     enable_prefix_caching: bool = True
     enable_hip_graphs: bool = True
+    host_cache_pages: int = 0           # KVBM G2 tier size (0 = disabled)
     kv_events: bool = True              # emit stored/removed block events
     block_salt: int = 0
     # disaggregation

@@ -52,6 +52,12 @@ class LLMEngine:
         self.runner = runner if runner is not None else ModelRunner(cfg, tp, seed)
         self.alloc = PageAllocator(self.runner.num_pages, cfg.page_size,
                                    cfg.enable_prefix_caching)
+        self.host_tier = None
+        if cfg.host_cache_pages > 0 and self.runner.kv_pool is not None:
+            from dynamo_amd.kvbm.host_tier import HostKVTier
+            self.host_tier = HostKVTier(self.runner.kv_pool,
+                                        cfg.host_cache_pages)
+            self.alloc.host_tier = self.host_tier
         self.scheduler = Scheduler(cfg, self.alloc)
         self.requests: Dict[str, Request] = {}
         self.step_count = 0
@@ -123,6 +129,8 @@ class LLMEngine:
             gr.rebuild(running)
         self.step_count += 1
         t0 = time.monotonic()
+        if self.host_tier is not None:
+            self.host_tier.fence()
         logits = gr.step(self._last_sampled)
         from .sampling import sample_tokens
         sampled = sample_tokens(logits, running, self.step_count)
@@ -178,6 +186,8 @@ class LLMEngine:
         if sched.is_empty:
             return []
         self.step_count += 1
+        if self.host_tier is not None:
+            self.host_tier.fence()
         sampled, sample_reqs = self.runner.execute(sched, self.step_count)
         sampled = sampled.cpu().tolist() if len(sample_reqs) else []
 
@@ -243,6 +253,9 @@ class LLMEngine:
             req.kv = None
 
     def drain_kv_events(self) -> List[KvEvent]:
+        if self.host_tier is not None:
+            for kind, h in self.host_tier.drain_events():
+                self.kv_events.append(KvEvent(kind, [h]))
         ev, self.kv_events = self.kv_events, []
         return ev
 

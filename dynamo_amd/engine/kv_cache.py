@@ -49,6 +49,7 @@ class PageAllocator:
         self.hash_to_page: Dict[int, int] = {}
         self.evictable: "OrderedDict[int, None]" = OrderedDict()  # page -> None
         self.events: List[KvEvent] = []
+        self.host_tier = None  # KVBM G2 (kvbm.host_tier.HostKVTier)
 
     # -- stats ------------------------------------------------------------
     @property
@@ -67,6 +68,8 @@ class PageAllocator:
             pid, _ = self.evictable.popitem(last=False)  # LRU evict
             h = self.page_hash[pid]
             if h is not None:
+                if self.host_tier is not None:
+                    self.host_tier.offload(pid, h)  # demote to G2
                 del self.hash_to_page[h]
                 self.page_hash[pid] = None
                 self.events.append(KvEvent("removed", [h]))
@@ -96,14 +99,29 @@ class PageAllocator:
 
     # -- prefix cache -----------------------------------------------------
     def lookup(self, h: int) -> Optional[int]:
-        """Find a cached page by hash and take a reference."""
+        """Find a cached page by hash and take a reference. Falls back to
+        the host tier (onboard = H2D copy into a fresh device page)."""
         if not self.enable_prefix:
             return None
         pid = self.hash_to_page.get(h)
-        if pid is None:
-            return None
-        self.incref(pid)
-        return pid
+        if pid is not None:
+            self.incref(pid)
+            return pid
+        if self.host_tier is not None and self.host_tier.contains(h):
+            try:
+                pid = self.alloc()
+            except MemoryError:
+                return None
+            if self.host_tier.onboard(h, pid):
+                # register without re-emitting stored (content unchanged on
+                # this worker from the router's perspective)
+                if h not in self.hash_to_page:
+                    self.page_hash[pid] = h
+                    self.hash_to_page[h] = pid
+                    self.events.append(KvEvent("stored", [h], None))
+                return pid
+            self.decref(pid)
+        return None
 
     def register_hash(self, pid: int, h: int, parent: Optional[int]):
         """Mark a now-full page as carrying block hash `h` (emits `stored`)."""

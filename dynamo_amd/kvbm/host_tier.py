@@ -1,0 +1,120 @@
+"""KVBM G2 tier: pinned-host KV page pool with offload/onboard.
+
+The MI355X-native core of the reference's tiered KV block manager
+(ai-dynamo/dynamo lib/llm/src/block_manager: G1 HBM / G2 pinned host,
+offload.rs offload manager, TransferStrategy CudaAsyncH2D/D2H
+block/transfer.rs:97): pages evicted from the device prefix cache are
+offloaded to a pinned host pool (async on a dedicated HIP stream, DMA over
+PCIe) and onboarded back on a prefix-cache hit, instead of being
+recomputed.
+
+Host layout is page-contiguous [Ph, L, 2, hkv, ps, hd] so one D2H/H2D
+memcpy moves a whole page; the device side gathers/scatters the (L, 2)
+planes of a page through the page-copy kernels via a staging buffer.
+"""
+from __future__ import annotations
+
+import logging
+from collections import OrderedDict
+from typing import Dict, Optional
+
+import torch
+
+log = logging.getLogger("dynamo_amd.kvbm")
+
+
+class HostKVTier:
+    def __init__(self, kv_pool, num_host_pages: int):
+        self.pool = kv_pool
+        self.device = kv_pool.device
+        L, two, P, hkv, ps, hd = kv_pool.shape
+        self.planes = L * two
+        self.plane_elems = hkv * ps * hd
+        self.page_elems = self.planes * self.plane_elems
+        self.num_host_pages = num_host_pages
+        pin = self.device.type == "cuda"
+        self.host = torch.empty(num_host_pages, self.page_elems,
+                                dtype=kv_pool.dtype, pin_memory=pin)
+        self.flat = kv_pool.buffer.reshape(self.planes * P, self.plane_elems)
+        self.P = P
+        # hash -> host page (insertion-ordered for LRU)
+        self.map: "OrderedDict[int, int]" = OrderedDict()
+        self.free = list(range(num_host_pages))
+        if self.device.type == "cuda":
+            self.stream = torch.cuda.Stream(device=self.device)
+            self.staging = torch.empty(self.page_elems, dtype=kv_pool.dtype,
+                                       device=self.device)
+            self._fence_event = torch.cuda.Event()
+        self.stats = {"offloaded": 0, "onboarded": 0, "evicted_host": 0,
+                      "hits": 0}
+        self.events = []  # (kind, hash) host-tier events
+
+    def _plane_ids(self, pid: int) -> torch.Tensor:
+        ids = [k * self.P + pid for k in range(self.planes)]
+        return torch.tensor(ids, dtype=torch.int32, device=self.device)
+
+    def contains(self, h: int) -> bool:
+        return h in self.map
+
+    def _alloc_host(self) -> Optional[int]:
+        if self.free:
+            return self.free.pop()
+        if self.map:
+            old_h, hp = self.map.popitem(last=False)  # LRU
+            self.stats["evicted_host"] += 1
+            self.events.append(("removed_host", old_h))
+            return hp
+        return None
+
+    # -- device -> host (called from PageAllocator eviction hook) --------
+    def offload(self, pid: int, h: int):
+        if h in self.map:
+            return
+        hp = self._alloc_host()
+        if hp is None:
+            return
+        if self.device.type == "cuda":
+            from dynamo_amd import ops
+            with torch.cuda.stream(self.stream):
+                ops.hip().gather_pages(self.staging, self.flat,
+                                       self._plane_ids(pid))
+                self.host[hp].copy_(self.staging, non_blocking=True)
+                self._fence_event.record(self.stream)
+        else:
+            ids = [k * self.P + pid for k in range(self.planes)]
+            self.host[hp].copy_(self.flat[ids].reshape(-1))
+        self.map[h] = hp
+        self.map.move_to_end(h)
+        self.stats["offloaded"] += 1
+        self.events.append(("stored_host", h))
+
+    # -- host -> device (prefix-cache onboard) ---------------------------
+    def onboard(self, h: int, pid: int) -> bool:
+        hp = self.map.get(h)
+        if hp is None:
+            return False
+        if self.device.type == "cuda":
+            from dynamo_amd import ops
+            with torch.cuda.stream(self.stream):
+                self.staging.copy_(self.host[hp], non_blocking=True)
+                ops.hip().scatter_pages(self.staging, self.flat,
+                                        self._plane_ids(pid))
+                self._fence_event.record(self.stream)
+        else:
+            ids = [k * self.P + pid for k in range(self.planes)]
+            self.flat[ids] = self.host[hp].reshape(self.planes,
+                                                   self.plane_elems)
+        self.map.move_to_end(h)
+        self.stats["onboarded"] += 1
+        self.stats["hits"] += 1
+        return True
+
+    def fence(self):
+        """Make the current compute stream wait for in-flight transfers.
+        Call once before each forward pass."""
+        if self.device.type == "cuda":
+            torch.cuda.current_stream(self.device).wait_event(self._fence_event)
+
+    def drain_events(self):
+        ev, self.events = self.events, []
+        return ev
