@@ -416,6 +416,8 @@ def main():
                              if torch.cuda.is_available() else "cpu")
     if device.startswith("cuda"):
         torch.cuda.set_device(local_rank)
+        from dynamo_amd.utils import enable_tunableop
+        enable_tunableop(tuning=False)
     if world > 1:
         dist.init_process_group("nccl" if device.startswith("cuda") else "gloo")
 

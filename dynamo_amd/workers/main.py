@@ -55,6 +55,9 @@ def make_engine_from_args(args) -> LLMEngine:
             worker_type=args.worker_type)
     import torch
     device = args.device or ("cuda:0" if torch.cuda.is_available() else "cpu")
+    if device.startswith("cuda"):
+        from dynamo_amd.utils import enable_tunableop
+        enable_tunableop(tuning=False)
     cfg = EngineConfig(
         model=mc, device=device, page_size=args.page_size,
         max_num_seqs=args.max_num_seqs,
