@@ -1,0 +1,146 @@
+"""AIPerf-style load generator for the OpenAI frontend.
+
+Parity with the reference's benchmarking discipline
+(ai-dynamo/dynamo docs/.../llama-3-3-70b-topology.mdx:55-63 AIPerf: fixed
+ISL/OSL via min/max tokens, concurrency-driven closed loop;
+benchmarks/router prefix-ratio sweeps): generates synthetic prompts of
+exact ISL (optionally with a shared prefix ratio to exercise KV-aware
+routing), drives fixed concurrency or Poisson arrivals, and reports
+output tok/s, TTFT and ITL percentiles.
+
+Usage:
+  python benchmarks/loadgen.py --url http://127.0.0.1:8000 \\
+      --model mock-model --isl 8192 --osl 1024 --concurrency 64 \\
+      --requests 160 [--prefix-ratio 0.5] [--rate 5.0]
+"""
+from __future__ import annotations
+
+import argparse
+import asyncio
+import json
+import random
+import statistics
+import sys
+import time
+
+import httpx
+
+
+def pct(xs, p):
+    if not xs:
+        return None
+    xs = sorted(xs)
+    i = min(len(xs) - 1, int(p / 100 * len(xs)))
+    return xs[i]
+
+
+async def one_request(client, args, prompt_tokens, stats):
+    t0 = time.monotonic()
+    first = None
+    ntok = 0
+    last = t0
+    itls = []
+    try:
+        async with client.stream("POST", f"{args.url}/v1/completions", json={
+                "model": args.model, "prompt": prompt_tokens,
+                "max_tokens": args.osl, "stream": True,
+                "ignore_eos": True}) as r:
+            if r.status_code != 200:
+                stats["errors"] += 1
+                return
+            async for line in r.aiter_lines():
+                if not line.startswith("data: ") or line == "data: [DONE]":
+                    continue
+                now = time.monotonic()
+                if first is None:
+                    first = now
+                else:
+                    itls.append((now - last) * 1000)
+                last = now
+                ntok += 1
+    except httpx.HTTPError:
+        stats["errors"] += 1
+        return
+    stats["ttft"].append(first - t0 if first else None)
+    stats["itl"].extend(itls)
+    stats["tokens"] += ntok
+    stats["latency"].append(time.monotonic() - t0)
+
+
+async def run(args):
+    rng = random.Random(args.seed)
+    vocab = args.vocab
+    shared_prefix = [rng.randrange(vocab)
+                     for _ in range(int(args.isl * args.prefix_ratio))]
+
+    def make_prompt():
+        own = [rng.randrange(vocab) for _ in range(args.isl - len(shared_prefix))]
+        return shared_prefix + own
+
+    stats = {"ttft": [], "itl": [], "latency": [], "tokens": 0, "errors": 0}
+    limits = httpx.Limits(max_connections=args.concurrency + 8)
+    async with httpx.AsyncClient(timeout=None, limits=limits) as client:
+        sem = asyncio.Semaphore(args.concurrency)
+
+        async def bounded(i):
+            if args.rate > 0:  # open loop: Poisson arrivals
+                await asyncio.sleep(rng.expovariate(args.rate) * i / max(i, 1))
+            async with sem:
+                await one_request(client, args, make_prompt(), stats)
+
+        t0 = time.monotonic()
+        if args.rate > 0:
+            tasks = []
+            t = 0.0
+            for i in range(args.requests):
+                t += rng.expovariate(args.rate)
+
+                async def delayed(d=t, i=i):
+                    await asyncio.sleep(d)
+                    async with sem:
+                        await one_request(client, args, make_prompt(), stats)
+                tasks.append(asyncio.create_task(delayed()))
+            await asyncio.gather(*tasks)
+        else:
+            await asyncio.gather(*[bounded(i) for i in range(args.requests)])
+        wall = time.monotonic() - t0
+
+    ttfts = [t for t in stats["ttft"] if t is not None]
+    out = {
+        "requests": args.requests,
+        "errors": stats["errors"],
+        "wall_s": round(wall, 3),
+        "output_tok_s": round(stats["tokens"] / wall, 2),
+        "ttft_p50_s": round(pct(ttfts, 50), 4) if ttfts else None,
+        "ttft_p95_s": round(pct(ttfts, 95), 4) if ttfts else None,
+        "itl_p50_ms": round(pct(stats["itl"], 50), 3) if stats["itl"] else None,
+        "itl_p95_ms": round(pct(stats["itl"], 95), 3) if stats["itl"] else None,
+        "latency_p50_s": round(pct(stats["latency"], 50), 3)
+        if stats["latency"] else None,
+        "config": {"isl": args.isl, "osl": args.osl,
+                   "concurrency": args.concurrency,
+                   "prefix_ratio": args.prefix_ratio, "rate": args.rate},
+    }
+    print(json.dumps(out))
+    return out
+
+
+def main():
+    p = argparse.ArgumentParser()
+    p.add_argument("--url", default="http://127.0.0.1:8000")
+    p.add_argument("--model", default="")
+    p.add_argument("--isl", type=int, default=8192)
+    p.add_argument("--osl", type=int, default=1024)
+    p.add_argument("--concurrency", type=int, default=16)
+    p.add_argument("--requests", type=int, default=32)
+    p.add_argument("--prefix-ratio", type=float, default=0.0)
+    p.add_argument("--rate", type=float, default=0.0,
+                   help=">0: open-loop Poisson req/s; 0: closed loop")
+    p.add_argument("--vocab", type=int, default=512)
+    p.add_argument("--seed", type=int, default=0)
+    args = p.parse_args()
+    asyncio.run(run(args))
+
+
+if __name__ == "__main__":
+    main()

@@ -18,6 +18,7 @@ import uuid
 from dataclasses import dataclass, field
 from typing import AsyncIterator, Dict, List, Optional
 
+from dynamo_amd.observability import span, trace_event
 from dynamo_amd.runtime import DistributedRuntime, EndpointError, NoInstancesError
 from dynamo_amd.router import KvRouter, PrefillRouter, RouterConfig
 from .tokenizer import ChatTemplater, make_tokenizer
@@ -148,6 +149,8 @@ class ModelManager:
         rid = request_id or f"{uuid.uuid4().hex[:16]}"
         delivered: List[int] = []
         attempts = 0
+        trace_event("request_start", request_id=rid, model=entry.name,
+                    prompt_tokens=len(token_ids))
         while True:
             payload = {
                 "request_id": rid if not attempts else f"{rid}-m{attempts}",
@@ -156,6 +159,7 @@ class ModelManager:
                 "stop_conditions": dict(
                     stop, max_tokens=max(1, stop.get("max_tokens", 128)
                                          - len(delivered))),
+                "annotations": {"trace_id": rid},
             }
             try:
                 if entry.prefill_router is not None and \
@@ -171,6 +175,8 @@ class ModelManager:
                     for t in chunk.get("token_ids", []):
                         delivered.append(t)
                     yield chunk
+                trace_event("request_end", request_id=rid,
+                            output_tokens=len(delivered), attempts=attempts)
                 return
             except (EndpointError, ConnectionError, OSError) as e:
                 attempts += 1

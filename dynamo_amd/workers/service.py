@@ -26,6 +26,7 @@ from typing import Any, AsyncIterator, Dict, Optional
 from dynamo_amd.engine.engine import LLMEngine
 from dynamo_amd.engine.scheduler import Request, SamplingParams
 from dynamo_amd.engine.kv_cache import SequenceKV
+from dynamo_amd.observability import trace_event
 from dynamo_amd.runtime import DistributedRuntime, RequestContext
 from dynamo_amd.disagg.transfer import KvPuller, pool_transfer_metadata
 
@@ -140,6 +141,9 @@ class WorkerService:
                                 top_p=sp.top_p, top_k=sp.top_k,
                                 seed=sp.seed, ignore_eos=True)
 
+        trace_event("handle_payload", request_id=req_id,
+                    worker_id=self.instance_id, worker_type=self.worker_type,
+                    prompt_tokens=len(tokens))
         q: asyncio.Queue = asyncio.Queue()
         self.queues[req_id] = q
         try:
