@@ -48,6 +48,36 @@ static Bufs make(int G) {
   return bf;
 }
 
+template <int G>
+static void run_mfma(const Bufs& bf) {
+  dim3 grid(B, Hkv, bf.C);
+  const int lds = mfma_lds_bytes(G, HD);
+  const int iters = 30;
+  for (int i = 0; i < 5; i++)
+    paged_decode_mfma<G><<<grid, kBlock, lds>>>(
+        bf.partial, bf.ml, bf.out, bf.q, bf.kc, bf.vc, bf.pt, bf.ctx,
+        0.0883883f, B, Hkv, bf.C, CTX / PS, 6, HD);
+  CK(hipDeviceSynchronize());
+  hipEvent_t e0, e1;
+  CK(hipEventCreate(&e0)); CK(hipEventCreate(&e1));
+  CK(hipEventRecord(e0));
+  for (int i = 0; i < iters; i++) {
+    paged_decode_mfma<G><<<grid, kBlock, lds>>>(
+        bf.partial, bf.ml, bf.out, bf.q, bf.kc, bf.vc, bf.pt, bf.ctx,
+        0.0883883f, B, Hkv, bf.C, CTX / PS, 6, HD);
+    paged_decode_phase2<<<dim3(B, G * Hkv), 128>>>(
+        bf.out, bf.partial, bf.ml, bf.ctx, G * Hkv, bf.C, HD);
+  }
+  CK(hipEventRecord(e1));
+  CK(hipEventSynchronize(e1));
+  float ms;
+  CK(hipEventElapsedTime(&ms, e0, e1));
+  double t = ms / 1000.0 / iters;
+  double gb = 2.0 * B * CTX * Hkv * HD * 2 / 1e9;
+  printf("G%d MFMA            %8.1f us  %7.0f GB/s\n", G, t * 1e6, gb / t);
+  fflush(stdout);
+}
+
 template <int G, int DP, int HS, int DEPTH>
 static void run(const Bufs& bf, const char* tag) {
   dim3 grid(B, Hkv, bf.C);
@@ -92,6 +122,7 @@ int main() {
     run<8, 8, 4, 3>(bf, "");
     run<8, 16, 1, 2>(bf, "");
     run<8, 8, 2, 2>(bf, "");
+    run_mfma<8>(bf);
   }
   {
     Bufs bf = make(4);
@@ -103,6 +134,7 @@ int main() {
     run<4, 16, 4, 2>(bf, "");
     run<4, 8, 2, 2>(bf, "");
     run<4, 8, 1, 2>(bf, "");
+    run_mfma<4>(bf);
   }
   {
     Bufs bf = make(1);

@@ -42,14 +42,35 @@ void paged_attention_decode(torch::Tensor out, torch::Tensor q,
       (const short*)vcache.data_ptr(), page_table.data_ptr<int32_t>(),        \
       ctx_lens.data_ptr<int32_t>(), (float)scale, B, Hkv, C, max_pages,       \
       log2_ps, hd)
+  static const bool use_mfma = [] {
+    const char* e = getenv("DYNAMO_DECODE_MFMA");
+    return e == nullptr || e[0] != '0';  // default ON (sweep-verified)
+  }();
+#define LAUNCH_MFMA(GG)                                                       \
+  paged_decode_mfma<GG><<<grid, kBlock, mfma_lds_bytes(GG, hd), stream>>>(    \
+      partial.data_ptr<float>(), ml.data_ptr<float>(), (short*)out.data_ptr(),\
+      (const short*)q.data_ptr(), (const short*)kcache.data_ptr(),            \
+      (const short*)vcache.data_ptr(), page_table.data_ptr<int32_t>(),        \
+      ctx_lens.data_ptr<int32_t>(), (float)scale, B, Hkv, C, max_pages,       \
+      log2_ps, hd)
+  const bool mfma_ok = use_mfma && ps % 32 == 0 && hd == 128;
   switch (G) {  // combos picked by benchmarks/decode_sweep on MI355X
     case 1: LAUNCH_G(1, 8, 1, 2); break;
     case 2: LAUNCH_G(2, 8, 2, 2); break;
-    case 4: LAUNCH_G(4, 16, 2, 4); break;
-    case 8: LAUNCH_G(8, 8, 2, 2); break;
+    case 4:
+      if (mfma_ok) { LAUNCH_MFMA(4); } else { LAUNCH_G(4, 16, 2, 4); }
+      break;
+    case 8:
+      if (mfma_ok) { LAUNCH_MFMA(8); } else { LAUNCH_G(8, 8, 2, 2); }
+      break;
+    case 16:
+      TORCH_CHECK(mfma_ok, "G=16 requires the MFMA decode path");
+      LAUNCH_MFMA(16);
+      break;
     default: TORCH_CHECK(false, "unsupported GQA group size ", G);
   }
 #undef LAUNCH_G
+#undef LAUNCH_MFMA
   HIP_CHECK_KERNEL();
   if (C > 1) {
     dim3 grid2(B, Hq);

@@ -264,6 +264,253 @@ __global__ __launch_bounds__(kBlock) void paged_decode_phase1(
   }
 }
 
+// ---------------------------------------------------------------------------
+// MFMA decode variant: the GQA group rides the MFMA M dimension, so the
+// per-head VALU cost of the dot/accumulate path disappears (measured: the
+// VALU kernel scales 4978 -> 2467 GB/s from G=1 to G=8; this variant keeps
+// the G=1 rate for G=8).
+//   S   = Q[G<=16 pad, 128] x K^T[128, 16 toks]   (K B-frags read DIRECTLY
+//         from the paged cache — no LDS staging for K)
+//   P   row-major via a small per-wave LDS tile (C/D -> A-frag re-layout)
+//   PV  = P[16, 32 toks] x V[32 toks, 128]; V staged per-wave in an LDS
+//         layout matching ds_read_b64_tr_b16's native transpose pattern
+//         (slab-permuted 4x16 blocks), so B-frags cost 2 tr-reads each.
+// Softmax runs on the C/D layout: row = head (4 per lane), col = token
+// (16 lanes), reductions are 4 shfl per row per 16 tokens.
+// Requires page_size % 32 == 0 and hd == 128.
+template <int G>
+__global__ __launch_bounds__(kBlock) void paged_decode_mfma(
+    float* __restrict__ partial, float* __restrict__ ml,
+    short* __restrict__ out, const short* __restrict__ q,
+    const short* __restrict__ kcache, const short* __restrict__ vcache,
+    const int32_t* __restrict__ page_table, const int32_t* __restrict__ ctx_lens,
+    float scale, int B, int Hkv, int C, int max_pages, int log2_ps, int hd) {
+  typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8_t;
+  constexpr int kSlab = kChunk / 4;   // 128 tokens per wave
+  const int b = blockIdx.x;
+  const int h = blockIdx.y;
+  const int c = blockIdx.z;
+  const int Hq = Hkv * G;
+  const int ctx = ctx_lens[b];
+  const int chunk_start = c * kChunk;
+  const int lane = threadIdx.x & 63;
+  const int wid = threadIdx.x >> 6;
+  const int lr = lane & 15;       // col lane (token / dim-subtile index)
+  const int lg = lane >> 4;       // 4 k-groups
+  const int ps = 1 << log2_ps;
+
+  // LDS: [4 waves][V tile 8KB] + [4 waves][P 16x32 bf16 1KB] + q G*128 bf16
+  // + merge scratch (4*G*(hd+2) fp32)
+  extern __shared__ float lds[];
+  float* merge = lds;                                   // 4*G*(hd+2)
+  short* q_lds_s = reinterpret_cast<short*>(merge + 4 * G * (hd + 2));
+  short* v_lds = q_lds_s + G * hd + wid * 4096;         // per-wave 8KB
+  short* p_lds = q_lds_s + G * hd + 4 * 4096 + wid * 512;  // per-wave 1KB
+
+  if (chunk_start >= ctx) {
+    if (C > 1) {
+      for (int i = threadIdx.x; i < G; i += kBlock) {
+        const int qh = h * G + i;
+        float* mlp = ml + (((int64_t)b * Hq + qh) * C + c) * 2;
+        mlp[0] = kNegInf; mlp[1] = 0.f;
+      }
+    }
+    return;
+  }
+
+  for (int i = threadIdx.x; i < G * hd; i += kBlock) {
+    const int g = i / hd;
+    q_lds_s[i] = q[((int64_t)b * Hq + h * G + g) * hd + i % hd];
+  }
+  __syncthreads();
+
+  // A-frags: row = lr (head, zero-padded beyond G), k = lg*8+j (+32*kc)
+  bf16x8_t q_frag[4];
+#pragma unroll
+  for (int kc = 0; kc < 4; kc++) {
+    short8 v{};
+    if (lr < G)
+      v = *reinterpret_cast<const short8*>(q_lds_s + lr * hd + kc * 32 + lg * 8);
+    q_frag[kc] = *reinterpret_cast<bf16x8_t*>(&v);
+  }
+
+  float m[4], l[4];
+  f32x4 acc[8];
+#pragma unroll
+  for (int r = 0; r < 4; r++) { m[r] = kNegInf; l[r] = 0.f; }
+#pragma unroll
+  for (int d = 0; d < 8; d++) acc[d] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+  const int slab_start = chunk_start + wid * kSlab;
+  const int slab_end = min(slab_start + kSlab, ctx);
+  const int32_t* pt = page_table + (int64_t)b * max_pages;
+
+  const unsigned v_base = (unsigned)(unsigned long long)(void*)v_lds;
+  const unsigned tr_lane_addr = v_base + (lane & 15) * 2 + (lane >> 4) * 128;
+
+  for (int t0 = slab_start; t0 < slab_end; t0 += 32) {
+    const int64_t pbase = (((int64_t)pt[t0 >> log2_ps] * Hkv + h) * ps) * hd;
+    // ---- K fragments (direct global) + S MFMA for two 16-token groups ----
+    f32x4 sA{0.f, 0.f, 0.f, 0.f}, sB{0.f, 0.f, 0.f, 0.f};
+    {
+      const int tA = t0 + lr, tB = t0 + 16 + lr;
+      const short* krA = kcache + pbase + (int64_t)(tA & (ps - 1)) * hd;
+      const short* krB = kcache + pbase + (int64_t)(tB & (ps - 1)) * hd;
+      const bool vA = tA < slab_end, vB = tB < slab_end;
+#pragma unroll
+      for (int kc = 0; kc < 4; kc++) {
+        short8 ka = vA ? *reinterpret_cast<const short8*>(krA + kc * 32 + lg * 8)
+                       : short8{};
+        sA = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            q_frag[kc], *reinterpret_cast<bf16x8_t*>(&ka), sA, 0, 0, 0);
+      }
+      if (t0 + 16 < slab_end) {
+#pragma unroll
+        for (int kc = 0; kc < 4; kc++) {
+          short8 kb2 = vB ? *reinterpret_cast<const short8*>(krB + kc * 32 + lg * 8)
+                          : short8{};
+          sB = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              q_frag[kc], *reinterpret_cast<bf16x8_t*>(&kb2), sB, 0, 0, 0);
+        }
+      }
+    }
+    // ---- stage V tile (32 toks x 128 dims) into the tr-ready layout ----
+#pragma unroll
+    for (int it = 0; it < 8; it++) {
+      const int slot = lane + it * 64;
+      const int d8 = slot & 7;          // 16B chunk of the row
+      const int tl = slot >> 3;         // token 0..31
+      const int t = t0 + tl;
+      short8 vv = (t < slab_end)
+          ? *reinterpret_cast<const short8*>(
+                vcache + pbase + (int64_t)(t & (ps - 1)) * hd + d8 * 8)
+          : short8{};
+      const int s4 = tl >> 2;                       // 4-token slab 0..7
+      const int p4 = (s4 >> 1) + (s4 & 1) * 4;      // tr-order permutation
+      short* dst = v_lds + (d8 >> 1) * 512 + p4 * 64 + (tl & 3) * 16 + (d8 & 1) * 8;
+      *reinterpret_cast<short8*>(dst) = vv;
+    }
+
+    // ---- online softmax on the C/D layout ----
+    const bool vA = (t0 + lr) < slab_end, vB = (t0 + 16 + lr) < slab_end;
+    float pA[4], pB[4];
+#pragma unroll
+    for (int r = 0; r < 4; r++) {
+      float a = vA ? sA[r] * scale : kNegInf;
+      float bb = vB ? sB[r] * scale : kNegInf;
+      float mt = fmaxf(a, bb);
+      mt = fmaxf(mt, __shfl_xor(mt, 1, WAVE_SIZE));
+      mt = fmaxf(mt, __shfl_xor(mt, 2, WAVE_SIZE));
+      mt = fmaxf(mt, __shfl_xor(mt, 4, WAVE_SIZE));
+      mt = fmaxf(mt, __shfl_xor(mt, 8, WAVE_SIZE));
+      if (mt > m[r]) {
+        const float corr = (m[r] <= kNegInf * 0.5f) ? 0.f : __expf(m[r] - mt);
+        l[r] *= corr;
+#pragma unroll
+        for (int d = 0; d < 8; d++) acc[d][r] *= corr;
+        m[r] = mt;
+      }
+      pA[r] = (a > kNegInf * 0.5f) ? __expf(a - m[r]) : 0.f;
+      pB[r] = (bb > kNegInf * 0.5f) ? __expf(bb - m[r]) : 0.f;
+      float rs = pA[r] + pB[r];
+      rs += __shfl_xor(rs, 1, WAVE_SIZE);
+      rs += __shfl_xor(rs, 2, WAVE_SIZE);
+      rs += __shfl_xor(rs, 4, WAVE_SIZE);
+      rs += __shfl_xor(rs, 8, WAVE_SIZE);
+      l[r] += rs;
+    }
+    // write P tile [16 heads][32 toks] (bank-spread via row XOR)
+#pragma unroll
+    for (int r = 0; r < 4; r++) {
+      const int row = lg * 4 + r;
+      const int x = (row & 3) << 4;
+      *(short*)((char*)p_lds + row * 64 + ((lr * 2) ^ x)) = f32_to_bf16(pA[r]);
+      *(short*)((char*)p_lds + row * 64 + (((16 + lr) * 2) ^ x)) = f32_to_bf16(pB[r]);
+    }
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    __builtin_amdgcn_sched_barrier(0);
+
+    // ---- PV: A = P (re-layout via LDS), B = V via hardware tr-reads ----
+    short8 pa_s;
+    {
+      const int row = lr;
+      const int x = (row & 3) << 4;
+      pa_s = *reinterpret_cast<const short8*>(
+          (char*)p_lds + row * 64 + ((lg * 16) ^ x));
+    }
+    bf16x8_t pa = *reinterpret_cast<bf16x8_t*>(&pa_s);
+#pragma unroll
+    for (int db = 0; db < 8; db++) {
+      unsigned long long v0, v1;
+      asm volatile("ds_read_b64_tr_b16 %0, %2 offset:%3\n\t"
+                   "ds_read_b64_tr_b16 %1, %2 offset:%4"
+                   : "=v"(v0), "=v"(v1)
+                   : "v"(tr_lane_addr), "i"(db * 1024), "i"(db * 1024 + 512)
+                   : "memory");
+      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+      __builtin_amdgcn_sched_barrier(0);
+      union { unsigned long long u[2]; bf16x8_t f; } vb;
+      vb.u[0] = v0; vb.u[1] = v1;
+      acc[db] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pa, vb.f, acc[db],
+                                                        0, 0, 0);
+    }
+    // v_lds/p_lds are per-wave: no cross-wave barrier needed per tile
+  }
+
+  // ---- cross-wave merge (same scheme as the VALU kernel) ----
+  __syncthreads();
+  float* my = merge + wid * G * (hd + 2);
+#pragma unroll
+  for (int r = 0; r < 4; r++) {
+    const int row = lg * 4 + r;
+    if (row < G) {
+#pragma unroll
+      for (int db = 0; db < 8; db++)
+        my[row * (hd + 2) + db * 16 + lr] = acc[db][r];
+      if (lr == 0) {
+        my[row * (hd + 2) + hd] = m[r];
+        my[row * (hd + 2) + hd + 1] = l[r];
+      }
+    }
+  }
+  __syncthreads();
+
+  for (int i = threadIdx.x; i < G * hd; i += kBlock) {
+    const int g = i / hd;
+    const int d = i % hd;
+    float mw[4], lw[4];
+    float mstar = kNegInf;
+#pragma unroll
+    for (int w = 0; w < 4; w++) {
+      mw[w] = merge[w * G * (hd + 2) + g * (hd + 2) + hd];
+      lw[w] = merge[w * G * (hd + 2) + g * (hd + 2) + hd + 1];
+      mstar = fmaxf(mstar, mw[w]);
+    }
+    float lsum = 0.f, asum = 0.f;
+#pragma unroll
+    for (int w = 0; w < 4; w++) {
+      const float corr = (lw[w] > 0.f) ? __expf(mw[w] - mstar) : 0.f;
+      lsum += lw[w] * corr;
+      asum += merge[w * G * (hd + 2) + g * (hd + 2) + d] * corr;
+    }
+    const int qh = h * G + g;
+    if (C == 1) {
+      out[((int64_t)b * Hq + qh) * hd + d] =
+          f32_to_bf16(lsum > 0.f ? asum / lsum : 0.f);
+    } else {
+      partial[(((int64_t)b * Hq + qh) * C + c) * hd + d] = asum;
+      if (d == 0) {
+        float* mlp = ml + (((int64_t)b * Hq + qh) * C + c) * 2;
+        mlp[0] = mstar; mlp[1] = lsum;
+      }
+    }
+  }
+}
+
+inline int mfma_lds_bytes(int G, int hd) {
+  return 4 * G * (hd + 2) * 4 + G * hd * 2 + 4 * 4096 * 2 + 4 * 512 * 2;
+}
+
 // Phase 2: merge chunk partials. grid (B, Hq), block = 128.
 __global__ inline void paged_decode_phase2(short* __restrict__ out,
                                            const float* __restrict__ partial,

@@ -45,6 +45,7 @@ torch::Tensor ipc_open(py::bytes handle_bytes, int64_t nbytes, int64_t device);
 void enable_peer_access(int64_t device, int64_t peer);
 // probe.hip
 torch::Tensor mfma_probe(torch::Tensor A, torch::Tensor B);
+torch::Tensor tr16_probe();
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.doc() = "dynamo_amd native MI355X (gfx950) kernels";
@@ -67,4 +68,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("ipc_open", &ipc_open);
   m.def("enable_peer_access", &enable_peer_access);
   m.def("mfma_probe", &mfma_probe);
+  m.def("tr16_probe", &tr16_probe);
 }

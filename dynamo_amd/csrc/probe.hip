@@ -30,7 +30,42 @@ __global__ void mfma_probe_kernel(float* __restrict__ D,      // [16,16]
   for (int r = 0; r < 4; r++) D[((l / 16) * 4 + r) * 16 + (l % 16)] = c[r];
 }
 
+__global__ void tr16_probe_kernel(int32_t* __restrict__ out) {  // [64, 8]
+  __shared__ short lds[512];
+  for (int i = threadIdx.x; i < 512; i += blockDim.x) lds[i] = (short)i;
+  __syncthreads();
+  const int l = threadIdx.x & 63;
+  const unsigned base = (unsigned)(unsigned long long)(void*)&lds[0];
+  const unsigned a = base + (l & 15) * 2 + (l >> 4) * 128;
+  unsigned long long v0, v1;
+  asm volatile("ds_read_b64_tr_b16 %0, %2 offset:0\n\t"
+               "ds_read_b64_tr_b16 %1, %2 offset:512"
+               : "=v"(v0), "=v"(v1) : "v"(a) : "memory");
+  asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+  __builtin_amdgcn_sched_barrier(0);
+  if (threadIdx.x < 64) {
+#pragma unroll
+    for (int j = 0; j < 4; j++) {
+      out[l * 8 + j] = (int32_t)((v0 >> (16 * j)) & 0xffff);
+      out[l * 8 + 4 + j] = (int32_t)((v1 >> (16 * j)) & 0xffff);
+    }
+  }
+}
+
 }  // namespace
+
+// Returns [64, 8]: the LDS bf16 element indices each lane's two
+// ds_read_b64_tr_b16 reads delivered (lds was filled with iota), so tests
+// can verify the transpose mapping the PV path assumes.
+torch::Tensor tr16_probe() {
+  auto out = torch::empty({64, 8}, torch::TensorOptions()
+                                       .dtype(torch::kInt32)
+                                       .device(torch::kCUDA, 0));
+  auto stream = at::cuda::getCurrentHIPStream();
+  tr16_probe_kernel<<<1, 64, 0, stream>>>(out.data_ptr<int32_t>());
+  HIP_CHECK_KERNEL();
+  return out;
+}
 
 torch::Tensor mfma_probe(torch::Tensor A, torch::Tensor B) {
   TORCH_CHECK(A.is_cuda() && A.dtype() == torch::kBFloat16);

@@ -161,6 +161,34 @@ def test_attention_prefill(G, spec):
     assert_close(out, ref)
 
 
+def test_tr16_probe_mapping():
+    """Verify the ds_read_b64_tr_b16 lane mapping the MFMA decode PV path
+    assumes: lane l, elem j reads lds[(l&15) + j*16 + (l>>4)*64] (+256 for
+    the second read at offset 512 B)."""
+    got = ops.hip().tr16_probe().cpu()
+    for l in range(64):
+        for j in range(4):
+            expect = (l & 15) + j * 16 + (l >> 4) * 64
+            assert got[l, j].item() == expect, (l, j, got[l, j].item(), expect)
+            assert got[l, 4 + j].item() == expect + 256, (l, j)
+
+
+def test_decode_valu_path_subprocess():
+    """The non-MFMA (VALU) decode variant stays correct (it serves G=1/2
+    and non-32-multiple page sizes)."""
+    import os
+    import subprocess
+    import sys
+    env = dict(os.environ, DYNAMO_DECODE_MFMA="0")
+    out = subprocess.run(
+        [sys.executable, "-m", "pytest", "tests/test_gpu_kernels.py", "-q",
+         "-k", "test_paged_attention_decode", "-m", "gpu", "-p",
+         "no:cacheprovider"],
+        capture_output=True, text=True, env=env, timeout=600,
+        cwd=os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+    assert out.returncode == 0, out.stdout[-2000:]
+
+
 def test_greedy_sample():
     torch.manual_seed(0)
     logits = torch.randn(9, 32000, device=DEV)
