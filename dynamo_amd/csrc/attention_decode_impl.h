@@ -345,8 +345,15 @@ __global__ __launch_bounds__(kBlock) void paged_decode_mfma(
   const int slab_end = min(slab_start + kSlab, ctx);
   const int32_t* pt = page_table + (int64_t)b * max_pages;
 
+  // ds_read_b64_tr_b16 true semantics (HW-probed, tests/test_gpu_kernels
+  // test_tr16_probe_mapping): lane reads 4 bf16 at 8-byte stride from its
+  // own address: v[j] = lds16[addr/2 + 4j]. The V layout below places
+  // V[slab_tok j][dim c] at slab_base + ((c>>2)<<4) + (j<<2) + (c&3), so a
+  // lane addressed at ((c>>2)<<4)+(c&3) gathers column c of the 4-token
+  // slab — a hardware 4x16 transpose per 16-lane group.
   const unsigned v_base = (unsigned)(unsigned long long)(void*)v_lds;
-  const unsigned tr_lane_addr = v_base + (lane & 15) * 2 + (lane >> 4) * 128;
+  const unsigned tr_lane_addr = v_base + (lane >> 4) * 128 +
+      ((lane & 15) >> 2) * 32 + (lane & 3) * 2;
 
   for (int t0 = slab_start; t0 < slab_end; t0 += 32) {
     const int64_t pbase = (((int64_t)pt[t0 >> log2_ps] * Hkv + h) * ps) * hd;
@@ -387,8 +394,14 @@ __global__ __launch_bounds__(kBlock) void paged_decode_mfma(
           : short8{};
       const int s4 = tl >> 2;                       // 4-token slab 0..7
       const int p4 = (s4 >> 1) + (s4 & 1) * 4;      // tr-order permutation
-      short* dst = v_lds + (d8 >> 1) * 512 + p4 * 64 + (tl & 3) * 16 + (d8 & 1) * 8;
-      *reinterpret_cast<short8*>(dst) = vv;
+      // element (j=tl&3, c) of the slab sits at ((c>>2)<<4)+(j<<2)+(c&3);
+      // the 8 dims of vv split into two contiguous 4-element runs
+      short* slab = v_lds + (d8 >> 1) * 512 + p4 * 64;
+      const int c0 = (d8 & 1) * 8;
+      short4v lo{vv[0], vv[1], vv[2], vv[3]};
+      short4v hi{vv[4], vv[5], vv[6], vv[7]};
+      *reinterpret_cast<short4v*>(slab + ((c0 >> 2) << 4) + ((tl & 3) << 2)) = lo;
+      *reinterpret_cast<short4v*>(slab + (((c0 + 4) >> 2) << 4) + ((tl & 3) << 2)) = hi;
     }
 
     // ---- online softmax on the C/D layout ----

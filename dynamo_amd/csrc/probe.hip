@@ -36,7 +36,8 @@ __global__ void tr16_probe_kernel(int32_t* __restrict__ out) {  // [64, 8]
   __syncthreads();
   const int l = threadIdx.x & 63;
   const unsigned base = (unsigned)(unsigned long long)(void*)&lds[0];
-  const unsigned a = base + (l & 15) * 2 + (l >> 4) * 128;
+  // production address pattern (attention_decode_impl.h MFMA PV path)
+  const unsigned a = base + (l >> 4) * 128 + ((l & 15) >> 2) * 32 + (l & 3) * 2;
   unsigned long long v0, v1;
   asm volatile("ds_read_b64_tr_b16 %0, %2 offset:0\n\t"
                "ds_read_b64_tr_b16 %1, %2 offset:512"

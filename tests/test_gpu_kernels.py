@@ -162,15 +162,16 @@ def test_attention_prefill(G, spec):
 
 
 def test_tr16_probe_mapping():
-    """Verify the ds_read_b64_tr_b16 lane mapping the MFMA decode PV path
-    assumes: lane l, elem j reads lds[(l&15) + j*16 + (l>>4)*64] (+256 for
-    the second read at offset 512 B)."""
+    """Verify the ds_read_b64_tr_b16 semantics the MFMA decode PV path
+    assumes (HW-probed): with lane address a = (l&15)*2 + (l>>4)*128 bytes,
+    elem j = lds16[a/2 + 4j]; the offset:512 read adds 256 elements."""
     got = ops.hip().tr16_probe().cpu()
     for l in range(64):
+        c = l & 15
+        base = (l >> 4) * 64 + (c >> 2) * 16 + (c & 3)
         for j in range(4):
-            expect = (l & 15) + j * 16 + (l >> 4) * 64
-            assert got[l, j].item() == expect, (l, j, got[l, j].item(), expect)
-            assert got[l, 4 + j].item() == expect + 256, (l, j)
+            assert got[l, j].item() == base + 4 * j, (l, j, got[l, j].item())
+            assert got[l, 4 + j].item() == base + 256 + 4 * j, (l, j)
 
 
 def test_decode_valu_path_subprocess():
