@@ -38,6 +38,10 @@ void attention_prefill_paged(torch::Tensor out, torch::Tensor q,
 void greedy_sample(torch::Tensor out, torch::Tensor logits);
 void gumbel_sample(torch::Tensor out, torch::Tensor logits, torch::Tensor inv_temp,
                    int64_t seed);
+// moe.hip
+void moe_grouped_gemm(torch::Tensor y, torch::Tensor x, torch::Tensor w,
+                      torch::Tensor tiles);
+void topk_gating(torch::Tensor topw, torch::Tensor topi, torch::Tensor logits);
 // ipc.hip
 torch::Tensor ipc_alloc(int64_t nbytes, int64_t device);
 py::bytes ipc_export(torch::Tensor t);
@@ -63,6 +67,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("attention_prefill_paged", &attention_prefill_paged);
   m.def("greedy_sample", &greedy_sample);
   m.def("gumbel_sample", &gumbel_sample);
+  m.def("moe_grouped_gemm", &moe_grouped_gemm);
+  m.def("topk_gating", &topk_gating);
   m.def("ipc_alloc", &ipc_alloc);
   m.def("ipc_export", &ipc_export);
   m.def("ipc_open", &ipc_open);

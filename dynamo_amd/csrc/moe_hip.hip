@@ -1,0 +1,162 @@
+#include "hip/hip_runtime.h"
+// MoE kernels: top-k gating + grouped GEMM for expert FFNs.
+//
+// Decode-regime grouped GEMM (few tokens per expert, e.g. conc 16 x top-2
+// over 8 experts) is WEIGHTS-bandwidth bound: every live expert's weights
+// stream from HBM once regardless of token count. The kernel assigns one
+// N-row of the current expert's weight to each lane and streams D in
+// 128-element chunks, dotting against <=16 segment tokens staged in LDS
+// (broadcast reads). Arithmetic intensity M*2 flops / 2 B keeps VALU far
+// from the ceiling, so the kernel runs at the HBM roofline like a dense
+// skinny GEMM but with ONE launch for all experts (vs E x hipBLASLt
+// launches). Large (prefill) segments go through hipBLASLt per expert at
+// the Python layer instead.
+//
+// Capability parity: the reference delegates MoE math to its engines
+// (SURVEY.md §2.4 "engine kernels the reference outsources").
+#include "common.h"
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+
+namespace {
+
+constexpr int kBlock = 256;
+constexpr int kMaxM = 16;   // tokens per tile (host splits larger segments)
+constexpr int kDC = 128;    // D chunk staged per iteration
+
+// tiles: [ntiles, 3] int32 = (expert, row0_in_gathered_x, m_count)
+__global__ __launch_bounds__(kBlock) void moe_gemm_kernel(
+    short* __restrict__ y,        // [T, N] bf16 (gathered order)
+    const short* __restrict__ x,  // [T, D] bf16 (gathered by expert)
+    const short* __restrict__ w,  // [E, N, D] bf16
+    const int32_t* __restrict__ tiles,
+    int D, int N, int ntiles) {
+  typedef __attribute__((ext_vector_type(2))) __bf16 bf16x2_t;
+  const int tile = blockIdx.x;
+  if (tile >= ntiles) return;
+  const int e = tiles[3 * tile];
+  const int r0 = tiles[3 * tile + 1];
+  const int m = tiles[3 * tile + 2];
+  const int n = blockIdx.y * kBlock + threadIdx.x;  // this lane's W row
+  const short* wrow = w + ((int64_t)e * N + n) * D;
+
+  __shared__ short x_lds[kMaxM * kDC];
+
+  float acc[kMaxM];
+#pragma unroll
+  for (int i = 0; i < kMaxM; i++) acc[i] = 0.f;
+
+  for (int dc = 0; dc < D; dc += kDC) {
+    __syncthreads();
+    // stage X chunk [m][kDC]
+    for (int i = threadIdx.x; i < m * (kDC / 8); i += kBlock) {
+      const int mi = i / (kDC / 8);
+      const int d8 = i % (kDC / 8);
+      *reinterpret_cast<short8*>(x_lds + mi * kDC + d8 * 8) =
+          *reinterpret_cast<const short8*>(x + ((int64_t)(r0 + mi)) * D + dc + d8 * 8);
+    }
+    __syncthreads();
+    if (n < N) {
+      // stream this lane's weight chunk once; dot against every token
+      short8 wv[kDC / 8];
+#pragma unroll
+      for (int i = 0; i < kDC / 8; i++)
+        wv[i] = *reinterpret_cast<const short8*>(wrow + dc + i * 8);
+      for (int mi = 0; mi < m; mi++) {
+        float d = acc[mi];
+#pragma unroll
+        for (int i = 0; i < kDC / 8; i++) {
+          const bf16x2_t* w2 = reinterpret_cast<const bf16x2_t*>(&wv[i]);
+          const bf16x2_t* x2 = reinterpret_cast<const bf16x2_t*>(
+              x_lds + mi * kDC + i * 8);
+#pragma unroll
+          for (int p = 0; p < 4; p++)
+            d = __builtin_amdgcn_fdot2_f32_bf16(w2[p], x2[p], d, false);
+        }
+        acc[mi] = d;
+      }
+    }
+  }
+  if (n < N) {
+    for (int mi = 0; mi < m; mi++)
+      y[((int64_t)(r0 + mi)) * N + n] = f32_to_bf16(acc[mi]);
+  }
+}
+
+// top-k gating: softmax over E experts, renormalized top-k weights.
+// one wave per token; E <= 64.
+__global__ void topk_gating_kernel(float* __restrict__ topw,   // [T, k]
+                                   int32_t* __restrict__ topi, // [T, k]
+                                   const float* __restrict__ logits,  // [T,E]
+                                   int T, int E, int K) {
+  const int t = blockIdx.x * (blockDim.x / 64) + (threadIdx.x / 64);
+  if (t >= T) return;
+  const int lane = threadIdx.x & 63;
+  float x = (lane < E) ? logits[(int64_t)t * E + lane] : -1e30f;
+  // softmax denominator over E
+  float mx = wave_reduce_max(x);
+  float ex = (lane < E) ? __expf(x - mx) : 0.f;
+  float denom = wave_reduce_sum(ex);
+  float p = ex / denom;
+  // iterative top-k (K <= 8; static indices to keep arrays in registers)
+  float mine = p;
+  float wsum = 0.f;
+  float myw[8];
+  int myi[8];
+#pragma unroll
+  for (int k = 0; k < 8; k++) {
+    if (k >= K) break;
+    float best = wave_reduce_max(mine);
+    // first lane holding `best` wins
+    unsigned long long mask = __ballot(mine == best && lane < E);
+    int win = __ffsll((long long)mask) - 1;
+    myw[k] = best;
+    myi[k] = win;
+    wsum += best;
+    if (lane == win) mine = -1.f;  // remove from candidates
+  }
+#pragma unroll
+  for (int k = 0; k < 8; k++) {
+    if (k >= K) break;
+    if (lane == k) {
+      topw[(int64_t)t * K + k] = myw[k] / wsum;
+      topi[(int64_t)t * K + k] = myi[k];
+    }
+  }
+}
+
+}  // namespace
+
+void moe_grouped_gemm(torch::Tensor y, torch::Tensor x, torch::Tensor w,
+                      torch::Tensor tiles) {
+  TORCH_CHECK(x.is_cuda() && x.dtype() == torch::kBFloat16);
+  TORCH_CHECK(tiles.dtype() == torch::kInt32 && tiles.size(1) == 3);
+  const int D = x.size(1);
+  const int N = w.size(1);
+  TORCH_CHECK(w.size(2) == D && y.size(1) == N);
+  TORCH_CHECK(D % kDC == 0, "hidden size must be a multiple of 128");
+  const int ntiles = tiles.size(0);
+  if (ntiles == 0) return;
+  auto stream = at::cuda::getCurrentHIPStream();
+  dim3 grid(ntiles, (N + kBlock - 1) / kBlock);
+ hipLaunchKernelGGL(( moe_gemm_kernel), dim3(grid), dim3(kBlock), 0, stream, 
+      (short*)y.data_ptr(), (const short*)x.data_ptr(),
+      (const short*)w.data_ptr(), tiles.data_ptr<int32_t>(), D, N, ntiles);
+  HIP_CHECK_KERNEL();
+}
+
+void topk_gating(torch::Tensor topw, torch::Tensor topi, torch::Tensor logits) {
+  TORCH_CHECK(logits.is_cuda() && logits.dtype() == torch::kFloat32);
+  const int T = logits.size(0);
+  const int E = logits.size(1);
+  const int K = topw.size(1);
+  TORCH_CHECK(E <= 64 && K <= 8);
+  if (T == 0) return;
+  auto stream = at::cuda::getCurrentHIPStream();
+  const int waves_per_block = 4;
+  const int grid = (T + waves_per_block - 1) / waves_per_block;
+ hipLaunchKernelGGL(( topk_gating_kernel), dim3(grid), dim3(waves_per_block * 64), 0, stream, 
+      topw.data_ptr<float>(), topi.data_ptr<int32_t>(),
+      logits.data_ptr<float>(), T, E, K);
+  HIP_CHECK_KERNEL();
+}

@@ -74,6 +74,11 @@ PRESETS = {
         name="tiny-llama-gpu", arch="llama", hidden_size=512,
         intermediate_size=1024, num_layers=2, num_q_heads=4, num_kv_heads=2,
         head_dim=128, vocab_size=1024, max_position=8192, rope_theta=10000.0),
+    "tiny-mixtral-gpu": ModelConfig(
+        name="tiny-mixtral-gpu", arch="mixtral", hidden_size=512,
+        intermediate_size=1024, num_layers=2, num_q_heads=4, num_kv_heads=2,
+        head_dim=128, vocab_size=1024, max_position=8192, rope_theta=10000.0,
+        num_experts=4, num_experts_per_tok=2),
     "tiny-opt": ModelConfig(
         name="tiny-opt", arch="opt", hidden_size=128, intermediate_size=256,
         num_layers=2, num_q_heads=4, num_kv_heads=4, head_dim=32,

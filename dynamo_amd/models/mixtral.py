@@ -34,12 +34,11 @@ class MoEMLP(torch.nn.Module):
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         T, D = x.shape
         logits = linear(x, self.router).float()            # [T, E]
-        weights = torch.softmax(logits, dim=-1)
-        topw, topi = torch.topk(weights, self.topk, dim=-1)  # [T, k]
-        topw = (topw / topw.sum(-1, keepdim=True)).to(x.dtype)
+        topw, topi = ops.topk_gating(logits, self.topk)    # [T, k]
+        topw = topw.to(x.dtype)
 
         out = torch.zeros_like(x)
-        flat_expert = topi.reshape(-1)                     # [T*k]
+        flat_expert = topi.reshape(-1).long()              # [T*k]
         flat_token = (torch.arange(T, device=x.device)
                       .repeat_interleave(self.topk))       # [T*k]
         # sort by expert -> contiguous segments
@@ -47,18 +46,27 @@ class MoEMLP(torch.nn.Module):
         seg_expert = flat_expert[order]
         seg_token = flat_token[order]
         counts = torch.bincount(seg_expert, minlength=self.E)
-        starts = torch.cumsum(counts, 0) - counts
         xg = x[seg_token]                                  # [T*k, D]
-        yg = torch.empty_like(xg)
         counts_l = counts.tolist()
-        starts_l = starts.tolist()
-        for e in range(self.E):
-            n = counts_l[e]
-            if n == 0:
-                continue
-            s = starts_l[e]
-            gu = linear(xg[s:s + n], self.w_gate_up[e])
-            yg[s:s + n] = linear(ops.silu_mul(gu), self.w_down[e])
+        if x.is_cuda and T <= 256:
+            # decode regime: one fused weights-streaming kernel per matmul
+            tiles = ops.build_moe_tiles(counts_l)
+            tiles_t = torch.tensor(tiles, dtype=torch.int32,
+                                   device=x.device).view(-1, 3)
+            gu = ops.moe_grouped_gemm(xg, self.w_gate_up, tiles_t)
+            act = ops.silu_mul(gu)
+            yg = ops.moe_grouped_gemm(act, self.w_down, tiles_t)
+        else:
+            # prefill regime: large segments -> hipBLASLt per expert
+            yg = torch.empty_like(xg)
+            s = 0
+            for e in range(self.E):
+                n = counts_l[e]
+                if n == 0:
+                    continue
+                gu = linear(xg[s:s + n], self.w_gate_up[e])
+                yg[s:s + n] = linear(ops.silu_mul(gu), self.w_down[e])
+                s += n
         w = topw.reshape(-1)[order].unsqueeze(-1)
         out.index_add_(0, seg_token, (yg * w).to(x.dtype))
         return self.tp.all_reduce(out)

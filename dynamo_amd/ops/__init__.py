@@ -168,6 +168,41 @@ def gumbel_sample(logits: torch.Tensor, inv_temp: torch.Tensor,
     return (logits.float() * inv_temp.unsqueeze(-1) + g).argmax(-1).to(torch.int32)
 
 
+def topk_gating(logits: torch.Tensor, k: int):
+    """softmax + renormalized top-k over experts. logits [T, E] fp32."""
+    if logits.is_cuda:
+        T = logits.shape[0]
+        topw = torch.empty(T, k, dtype=torch.float32, device=logits.device)
+        topi = torch.empty(T, k, dtype=torch.int32, device=logits.device)
+        hip().topk_gating(topw, topi, logits.float().contiguous())
+        return topw, topi
+    p = torch.softmax(logits.float(), dim=-1)
+    topw, topi = torch.topk(p, k, dim=-1)
+    return topw / topw.sum(-1, keepdim=True), topi.to(torch.int32)
+
+
+def build_moe_tiles(counts, max_m: int = 16):
+    """Host-side tile list [(expert, row0, m)] for the grouped GEMM;
+    segments larger than max_m split into multiple tiles."""
+    tiles = []
+    r0 = 0
+    for e, n in enumerate(counts):
+        off = 0
+        while off < n:
+            m = min(max_m, n - off)
+            tiles.append((e, r0 + off, m))
+            off += m
+        r0 += n
+    return tiles
+
+
+def moe_grouped_gemm(x: torch.Tensor, w: torch.Tensor, tiles_t: torch.Tensor):
+    """x [T, D] gathered by expert; w [E, N, D]; -> y [T, N]."""
+    y = torch.empty(x.shape[0], w.shape[1], dtype=x.dtype, device=x.device)
+    hip().moe_grouped_gemm(y, x.contiguous(), w, tiles_t)
+    return y
+
+
 def gather_pages(staging, cache, page_ids):
     if cache.is_cuda:
         hip().gather_pages(staging, cache, page_ids)

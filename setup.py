@@ -26,6 +26,7 @@ hip_sources = [
         "attention_decode.hip",
         "attention_prefill.hip",
         "sampling.hip",
+        "moe.hip",
         "ipc.hip",
         "probe.hip",
     )
