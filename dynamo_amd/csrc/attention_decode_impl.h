@@ -345,15 +345,13 @@ __global__ __launch_bounds__(kBlock) void paged_decode_mfma(
   const int slab_end = min(slab_start + kSlab, ctx);
   const int32_t* pt = page_table + (int64_t)b * max_pages;
 
-  // ds_read_b64_tr_b16 true semantics (HW-probed, tests/test_gpu_kernels
-  // test_tr16_probe_mapping): lane reads 4 bf16 at 8-byte stride from its
-  // own address: v[j] = lds16[addr/2 + 4j]. The V layout below places
-  // V[slab_tok j][dim c] at slab_base + ((c>>2)<<4) + (j<<2) + (c&3), so a
-  // lane addressed at ((c>>2)<<4)+(c&3) gathers column c of the 4-token
-  // slab — a hardware 4x16 transpose per 16-lane group.
-  const unsigned v_base = (unsigned)(unsigned long long)(void*)v_lds;
-  const unsigned tr_lane_addr = v_base + (lane >> 4) * 128 +
-      ((lane & 15) >> 2) * 32 + (lane & 3) * 2;
+  // NOTE on ds_read_b64_tr_b16 (HW-probed, tests test_tr16_probe_mapping):
+  // the instruction is CROSS-LANE cooperative — per 16-lane group it reads
+  // four 64-bit rows at the addresses of subgroup-leader lanes {0,4,8,12}
+  // and hands every lane column (lane&3) of that 4x4 tile. Only 4 distinct
+  // columns reach a 16-lane group, so it cannot feed a 16-column MFMA
+  // B-fragment; the PV path therefore reads the transposed V fragments as
+  // swizzled scalar LDS loads instead.
 
   for (int t0 = slab_start; t0 < slab_end; t0 += 32) {
     const int64_t pbase = (((int64_t)pt[t0 >> log2_ps] * Hkv + h) * ps) * hd;
@@ -392,16 +390,9 @@ __global__ __launch_bounds__(kBlock) void paged_decode_mfma(
           ? *reinterpret_cast<const short8*>(
                 vcache + pbase + (int64_t)(t & (ps - 1)) * hd + d8 * 8)
           : short8{};
-      const int s4 = tl >> 2;                       // 4-token slab 0..7
-      const int p4 = (s4 >> 1) + (s4 & 1) * 4;      // tr-order permutation
-      // element (j=tl&3, c) of the slab sits at ((c>>2)<<4)+(j<<2)+(c&3);
-      // the 8 dims of vv split into two contiguous 4-element runs
-      short* slab = v_lds + (d8 >> 1) * 512 + p4 * 64;
-      const int c0 = (d8 & 1) * 8;
-      short4v lo{vv[0], vv[1], vv[2], vv[3]};
-      short4v hi{vv[4], vv[5], vv[6], vv[7]};
-      *reinterpret_cast<short4v*>(slab + ((c0 >> 2) << 4) + ((tl & 3) << 2)) = lo;
-      *reinterpret_cast<short4v*>(slab + (((c0 + 4) >> 2) << 4) + ((tl & 3) << 2)) = hi;
+      // row-major [32][128] with an XOR bank swizzle on the row offset
+      *reinterpret_cast<short8*>(
+          (char*)v_lds + tl * 256 + ((d8 * 16) ^ ((tl & 7) << 4))) = vv;
     }
 
     // ---- online softmax on the C/D layout ----
@@ -454,17 +445,18 @@ __global__ __launch_bounds__(kBlock) void paged_decode_mfma(
     bf16x8_t pa = *reinterpret_cast<bf16x8_t*>(&pa_s);
 #pragma unroll
     for (int db = 0; db < 8; db++) {
-      unsigned long long v0, v1;
-      asm volatile("ds_read_b64_tr_b16 %0, %2 offset:%3\n\t"
-                   "ds_read_b64_tr_b16 %1, %2 offset:%4"
-                   : "=v"(v0), "=v"(v1)
-                   : "v"(tr_lane_addr), "i"(db * 1024), "i"(db * 1024 + 512)
-                   : "memory");
-      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-      __builtin_amdgcn_sched_barrier(0);
-      union { unsigned long long u[2]; bf16x8_t f; } vb;
-      vb.u[0] = v0; vb.u[1] = v1;
-      acc[db] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pa, vb.f, acc[db],
+      // B-frag: lane holds V[tok = 8*lg + j][dim = db*16 + lr], j = 0..7 —
+      // transposed scalar gathers from the swizzled row-major tile
+      short8 vb_s;
+#pragma unroll
+      for (int j = 0; j < 8; j++) {
+        const int tok = 8 * lg + j;
+        vb_s[j] = *(const short*)(
+            (const char*)v_lds + tok * 256 +
+            (((db * 16 + lr) * 2) ^ ((tok & 7) << 4)));
+      }
+      bf16x8_t vbf = *reinterpret_cast<bf16x8_t*>(&vb_s);
+      acc[db] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pa, vbf, acc[db],
                                                         0, 0, 0);
     }
     // v_lds/p_lds are per-wave: no cross-wave barrier needed per tile

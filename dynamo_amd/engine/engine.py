@@ -66,8 +66,9 @@ class LLMEngine:
         self._held: Dict[str, Request] = {}  # finished but KV retained (disagg)
         # hipGraph decode fast path
         self.graph_runner = None
+        # MoE routing builds per-step host-side tile lists -> not capturable
         if (cfg.enable_hip_graphs and self.runner.device.type == "cuda"
-                and cfg.tp_size == 1):
+                and cfg.tp_size == 1 and cfg.model.num_experts == 0):
             from .graphs import GraphRunner
             self.graph_runner = GraphRunner(self.runner, cfg.max_num_seqs)
         self._last_sampled = None

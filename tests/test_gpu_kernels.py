@@ -162,16 +162,21 @@ def test_attention_prefill(G, spec):
 
 
 def test_tr16_probe_mapping():
-    """Verify the ds_read_b64_tr_b16 semantics the MFMA decode PV path
-    assumes (HW-probed): with lane address a = (l&15)*2 + (l>>4)*128 bytes,
-    elem j = lds16[a/2 + 4j]; the offset:512 read adds 256 elements."""
+    """Document the HW-probed ds_read_b64_tr_b16 semantics: per 16-lane
+    group, four 64-bit rows are read at the (8B-aligned) addresses of
+    subgroup-leader lanes {0,4,8,12}; every lane receives column (lane&3)
+    of that 4x4 bf16 tile. (Cross-lane cooperative — only 4 distinct
+    columns per group, which is why the decode PV path does NOT use it.)
+
+    Probe addresses: lane l -> (l>>4)*128 + ((l&15)>>2)*32 + (l&3)*2 bytes,
+    so leader 4s of group g points at element g*64 + s*16."""
     got = ops.hip().tr16_probe().cpu()
     for l in range(64):
-        c = l & 15
-        base = (l >> 4) * 64 + (c >> 2) * 16 + (c & 3)
+        g, c = l >> 4, l & 15
         for j in range(4):
-            assert got[l, j].item() == base + 4 * j, (l, j, got[l, j].item())
-            assert got[l, 4 + j].item() == base + 256 + 4 * j, (l, j)
+            expect = g * 64 + j * 16 + (c & 3)
+            assert got[l, j].item() == expect, (l, j, got[l, j].item(), expect)
+            assert got[l, 4 + j].item() == expect + 256, (l, j)
 
 
 def test_decode_valu_path_subprocess():
