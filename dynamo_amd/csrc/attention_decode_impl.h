@@ -382,17 +382,17 @@ __global__ __launch_bounds__(kBlock) void paged_decode_mfma(
     // ---- stage V tile (32 toks x 128 dims) into the tr-ready layout ----
 #pragma unroll
     for (int it = 0; it < 8; it++) {
-      const int slot = lane + it * 64;
-      const int d8 = slot & 7;          // 16B chunk of the row
-      const int tl = slot >> 3;         // token 0..31
+      const int slot = lane + it * 64;   // 512 slots = 32 toks x 16 chunks
+      const int d16 = slot & 15;         // 16B chunk of the 256B row
+      const int tl = slot >> 4;          // token 0..31
       const int t = t0 + tl;
       short8 vv = (t < slab_end)
           ? *reinterpret_cast<const short8*>(
-                vcache + pbase + (int64_t)(t & (ps - 1)) * hd + d8 * 8)
+                vcache + pbase + (int64_t)(t & (ps - 1)) * hd + d16 * 8)
           : short8{};
       // row-major [32][128] with an XOR bank swizzle on the row offset
       *reinterpret_cast<short8*>(
-          (char*)v_lds + tl * 256 + ((d8 * 16) ^ ((tl & 7) << 4))) = vv;
+          (char*)v_lds + tl * 256 + ((d16 * 16) ^ ((tl & 7) << 4))) = vv;
     }
 
     // ---- online softmax on the C/D layout ----
