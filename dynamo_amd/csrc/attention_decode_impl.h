@@ -353,11 +353,16 @@ __global__ __launch_bounds__(kBlock) void paged_decode_mfma(
   // B-fragment; the PV path therefore reads the transposed V fragments as
   // swizzled scalar LDS loads instead.
 
-  for (int t0 = slab_start; t0 < slab_end; t0 += 32) {
-    const int64_t pbase = (((int64_t)pt[t0 >> log2_ps] * Hkv + h) * ps) * hd;
-    // ---- K fragments (direct global) + S MFMA for two 16-token groups ----
+  // Uniform tile count across ALL waves (inactive waves still hit the
+  // barriers — __syncthreads inside a loop with per-wave iteration counts
+  // is barrier divergence / UB).
+  for (int ti = 0; ti < kSlab / 32; ti++) {
+    const int t0 = slab_start + ti * 32;
+    const bool active = t0 < slab_end;
+    const int64_t pbase = active
+        ? (((int64_t)pt[t0 >> log2_ps] * Hkv + h) * ps) * hd : 0;
     f32x4 sA{0.f, 0.f, 0.f, 0.f}, sB{0.f, 0.f, 0.f, 0.f};
-    {
+    if (active) {
       const int tA = t0 + lr, tB = t0 + 16 + lr;
       const short* krA = kcache + pbase + (int64_t)(tA & (ps - 1)) * hd;
       const short* krB = kcache + pbase + (int64_t)(tB & (ps - 1)) * hd;
@@ -379,7 +384,8 @@ __global__ __launch_bounds__(kBlock) void paged_decode_mfma(
         }
       }
     }
-    // ---- stage V tile (32 toks x 128 dims) into the tr-ready layout ----
+    // ---- stage V tile (32 toks x 128 dims) ----
+    if (active)
 #pragma unroll
     for (int it = 0; it < 8; it++) {
       const int slot = lane + it * 64;   // 512 slots = 32 toks x 16 chunks
@@ -398,6 +404,7 @@ __global__ __launch_bounds__(kBlock) void paged_decode_mfma(
     // ---- online softmax on the C/D layout ----
     const bool vA = (t0 + lr) < slab_end, vB = (t0 + 16 + lr) < slab_end;
     float pA[4], pB[4];
+    if (active)
 #pragma unroll
     for (int r = 0; r < 4; r++) {
       float a = vA ? sA[r] * scale : kNegInf;
@@ -424,6 +431,7 @@ __global__ __launch_bounds__(kBlock) void paged_decode_mfma(
       l[r] += rs;
     }
     // write P tile [16 heads][32 toks] (bank-spread via row XOR)
+    if (active)
 #pragma unroll
     for (int r = 0; r < 4; r++) {
       const int row = lg * 4 + r;
@@ -431,10 +439,15 @@ __global__ __launch_bounds__(kBlock) void paged_decode_mfma(
       *(short*)((char*)p_lds + row * 64 + ((lr * 2) ^ x)) = f32_to_bf16(pA[r]);
       *(short*)((char*)p_lds + row * 64 + (((16 + lr) * 2) ^ x)) = f32_to_bf16(pB[r]);
     }
+    // per-wave buffers: a wave-level scheduling fence + in-order LDS
+    // suffice (ds ops of one wave complete in issue order); sweep-verified
+    // against the __syncthreads variant for correctness and speed
+    __builtin_amdgcn_wave_barrier();
     asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
     __builtin_amdgcn_sched_barrier(0);
 
-    // ---- PV: A = P (re-layout via LDS), B = V via hardware tr-reads ----
+    // ---- PV: A = P (re-layout via LDS), B = V scalar-transposed ----
+    if (active) {
     short8 pa_s;
     {
       const int row = lr;
@@ -451,15 +464,23 @@ __global__ __launch_bounds__(kBlock) void paged_decode_mfma(
 #pragma unroll
       for (int j = 0; j < 8; j++) {
         const int tok = 8 * lg + j;
+#ifdef DECODE_MFMA_V_DIRECT   // debug: bypass LDS staging
+        const int t = t0 + tok;
+        vb_s[j] = (t < slab_end)
+            ? vcache[pbase + (int64_t)(t & (ps - 1)) * hd + db * 16 + lr]
+            : (short)0;
+#else
         vb_s[j] = *(const short*)(
             (const char*)v_lds + tok * 256 +
             (((db * 16 + lr) * 2) ^ ((tok & 7) << 4)));
+#endif
       }
       bf16x8_t vbf = *reinterpret_cast<bf16x8_t*>(&vb_s);
       acc[db] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pa, vbf, acc[db],
                                                         0, 0, 0);
     }
-    // v_lds/p_lds are per-wave: no cross-wave barrier needed per tile
+    }  // active
+    __builtin_amdgcn_wave_barrier();
   }
 
   // ---- cross-wave merge (same scheme as the VALU kernel) ----

@@ -50,6 +50,8 @@ void enable_peer_access(int64_t device, int64_t peer);
 // probe.hip
 torch::Tensor mfma_probe(torch::Tensor A, torch::Tensor B);
 torch::Tensor tr16_probe();
+torch::Tensor vstage_probe();
+torch::Tensor pv_probe(int64_t hot);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.doc() = "dynamo_amd native MI355X (gfx950) kernels";
@@ -75,4 +77,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("enable_peer_access", &enable_peer_access);
   m.def("mfma_probe", &mfma_probe);
   m.def("tr16_probe", &tr16_probe);
+  m.def("vstage_probe", &vstage_probe);
+  m.def("pv_probe", &pv_probe);
 }
