@@ -1,18 +1,16 @@
 #include "hip/hip_runtime.h"
 // Paged flash-attention prefill (causal, GQA, chunked-prefill-aware),
-// MI355X-native, MFMA mfma_f32_16x16x32_bf16 with LDS-swizzled KV tiles.
+// MI355X-native, MFMA mfma_f32_16x16x32_bf16 with LDS-staged KV tiles.
 //
 // Structure (guide §5/§B): workgroup = 4 waves; each workgroup owns a 64-row
 // query tile of one query head; waves own 16 rows each. K/V tiles of 64
-// tokens are gathered from the paged cache into double-buffered LDS tiles
-// (XOR-swizzled, guide T2/G4: row-major [*][128] bf16 is otherwise a
-// 16-32-way bank conflict on ds_read_b128). Async staging split (guide T14):
-// the NEXT tile's global loads are issued into registers before computing
-// the current tile, and written to the back LDS buffer afterwards — HBM
-// latency hides under the MFMA phase. Online softmax with per-row (m, l).
+// tokens are gathered from the paged cache into LDS (XOR-swizzled layouts,
+// guide T2/G4: row-major [*][128] bf16 is otherwise a 16-32-way bank
+// conflict on ds_read_b128). Online softmax with per-row running (m, l).
+// The P tile round-trips through LDS to re-shape the S-layout (C/D frag)
+// into the PV A-operand layout.
 //
-// MFMA fragment mappings (verified on HW by tests/test_gpu_mfma layout
-// probes):
+// MFMA fragment mappings used (verified on HW by tests/test_gpu_mfma.py):
 //   A: lane l holds A[row = l%16][k = 8*(l/16) + i]      (i = 0..7)
 //   B: lane l holds B[k = 8*(l/16) + i][col = l%16]
 //   C/D: lane l, reg r holds D[row = (l/16)*4 + r][col = l%16]
@@ -63,12 +61,13 @@ __global__ __launch_bounds__(kBlock) void prefill_kernel(
   const int lr = lane & 15;   // row-or-col lane index
   const int lg = lane >> 4;   // 4 k-groups
 
-  // LDS: 2x K [64][128] (32 KB) + 2x V^T [128][64] (32 KB) + P [4][16][64]
-  __shared__ short k_lds[2][kKvTile * HD];
-  __shared__ short vt_lds[2][HD * kKvTile];
+  // LDS: K [64][128] (16 KB) + V^T [128][64] (16 KB) + P [4][16][64] (8 KB)
+  __shared__ short k_lds[kKvTile * HD];
+  __shared__ short vt_lds[HD * kKvTile];
   __shared__ short p_lds[4][16 * kKvTile];
 
   // ---- load Q fragments (once; reused across all kv tiles) ----
+  // wave's rows: q0 + wid*16 + lr ; A-frag kc covers dims [kc*32, kc*32+32)
   bf16x8_t q_frag[4];
   const int my_qrow = q0 + wid * 16 + lr;
   const bool row_valid = my_qrow < qlen;
@@ -81,6 +80,7 @@ __global__ __launch_bounds__(kBlock) void prefill_kernel(
       q_frag[kc] = *reinterpret_cast<bf16x8_t*>(&v);
     }
   }
+  const int my_qpos = ctx - qlen + my_qrow;  // absolute kv position of this row
 
   float m[4], lsum[4];
   f32x4 acc_o[8];
@@ -93,51 +93,30 @@ __global__ __launch_bounds__(kBlock) void prefill_kernel(
   const int tile_qpos_max = ctx - qlen + min(q0 + kQTile - 1, qlen - 1);
   const int kv_end = min(ctx, tile_qpos_max + 1);
 
-  // per-thread staging slice: 4 slots of (K, V) short8
-  short8 kreg[4], vreg[4];
-
-  auto load_tile = [&](int t0) {
+  for (int t0 = 0; t0 < kv_end; t0 += kKvTile) {
+    __syncthreads();
+    // ---- stage K tile + V^T tile ----
 #pragma unroll
     for (int i = 0; i < 4; i++) {
       const int slot = i * kBlock + threadIdx.x;  // 1024 short8 slots
       const int row = slot >> 4;
+      const int col8 = slot & 15;
       const int t = t0 + row;
+      short8 kv_k{}, kv_v{};
       if (t < ctx) {
-        const int col8 = slot & 15;
         const int64_t page = pt[t >> log2_ps];
         const int64_t base = ((page * Hkv + kvh) * ps + (t & (ps - 1))) * HD + col8 * 8;
-        kreg[i] = *reinterpret_cast<const short8*>(kcache + base);
-        vreg[i] = *reinterpret_cast<const short8*>(vcache + base);
-      } else {
-        kreg[i] = short8{};
-        vreg[i] = short8{};
+        kv_k = *reinterpret_cast<const short8*>(kcache + base);
+        kv_v = *reinterpret_cast<const short8*>(vcache + base);
       }
-    }
-  };
-
-  auto write_tile = [&](int buf) {
-#pragma unroll
-    for (int i = 0; i < 4; i++) {
-      const int slot = i * kBlock + threadIdx.x;
-      const int row = slot >> 4;
-      const int col8 = slot & 15;
-      *reinterpret_cast<short8*>((char*)k_lds[buf] + row * 256 + swz(col8 * 16, row)) = kreg[i];
+      *reinterpret_cast<short8*>((char*)k_lds + row * 256 + swz(col8 * 16, row)) = kv_k;
 #pragma unroll
       for (int e = 0; e < 8; e++) {
         const int dim = col8 * 8 + e;
-        *(short*)((char*)vt_lds[buf] + dim * 128 + swz(row * 2, dim)) = vreg[i][e];
+        *(short*)((char*)vt_lds + dim * 128 + swz(row * 2, dim)) = kv_v[e];
       }
     }
-  };
-
-  load_tile(0);
-  write_tile(0);
-  int cur = 0;
-
-  for (int t0 = 0; t0 < kv_end; t0 += kKvTile) {
-    const bool has_next = t0 + kKvTile < kv_end;
-    if (has_next) load_tile(t0 + kKvTile);  // in flight during compute
-    __syncthreads();  // buf[cur] fully written (all waves)
+    __syncthreads();
 
     // ---- S = Q K^T for the wave's 16 rows x 64 tokens ----
     f32x4 s[4];
@@ -149,7 +128,7 @@ __global__ __launch_bounds__(kBlock) void prefill_kernel(
         // B-frag: K[tok = n*16 + lr][d = kc*32 + lg*8 + j]
         const int tok = n * 16 + lr;
         short8 kv = *reinterpret_cast<const short8*>(
-            (char*)k_lds[cur] + tok * 256 + swz(kc * 64 + lg * 16, tok));
+            (char*)k_lds + tok * 256 + swz(kc * 64 + lg * 16, tok));
         accs = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
             q_frag[kc], *reinterpret_cast<bf16x8_t*>(&kv), accs, 0, 0, 0);
       }
@@ -179,8 +158,8 @@ __global__ __launch_bounds__(kBlock) void prefill_kernel(
     }
 
     // NOTE: rows of S (and acc_o) map to (lg, r): row = lg*4 + r. The running
-    // m/l state for a given physical row is kept by all 16 lanes with that
-    // lg, redundantly — shuffles above keep them consistent.
+    // m/l state for a given physical row is therefore kept by all 16 lanes
+    // with that lg, redundantly — shuffles above keep them consistent.
     float p[4][4];  // [n][r]
 #pragma unroll
     for (int r = 0; r < 4; r++) {
@@ -209,7 +188,7 @@ __global__ __launch_bounds__(kBlock) void prefill_kernel(
     }
 
     // ---- write P tile to LDS (re-layout for the PV A-operand) ----
-    // p_lds is per-wave: wave-local fence + in-order LDS ordering suffice
+    __syncthreads();  // all waves done reading k_lds-dependent S
 #pragma unroll
     for (int r = 0; r < 4; r++) {
 #pragma unroll
@@ -220,9 +199,7 @@ __global__ __launch_bounds__(kBlock) void prefill_kernel(
             f32_to_bf16(p[n][r]);
       }
     }
-    __builtin_amdgcn_wave_barrier();
-    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-    __builtin_amdgcn_sched_barrier(0);
+    __syncthreads();
 
     // ---- O += P V ----
 #pragma unroll
@@ -235,16 +212,12 @@ __global__ __launch_bounds__(kBlock) void prefill_kernel(
         // B-frag: V[tok = kt*32 + lg*8 + j][dim = d*16 + lr] = V^T[dim][tok]
         const int dim = d * 16 + lr;
         short8 vb = *reinterpret_cast<const short8*>(
-            (char*)vt_lds[cur] + dim * 128 + swz(kt * 64 + lg * 16, dim));
+            (char*)vt_lds + dim * 128 + swz(kt * 64 + lg * 16, dim));
         acc_o[d] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
             *reinterpret_cast<bf16x8_t*>(&pa), *reinterpret_cast<bf16x8_t*>(&vb),
             acc_o[d], 0, 0, 0);
       }
     }
-    // stage the prefetched tile into the back buffer (vmcnt drains here,
-    // after the MFMA phase — not before it)
-    if (has_next) write_tile(cur ^ 1);
-    cur ^= 1;
   }
 
   // ---- epilogue: normalize and store ----
