@@ -47,9 +47,10 @@ class ForwardPassMetrics:
 
 class LLMEngine:
     def __init__(self, cfg: EngineConfig, tp: Optional[TPContext] = None,
-                 seed: int = 0, runner=None):
+                 seed: int = 0, runner=None, weight_pool=None):
         self.cfg = cfg
-        self.runner = runner if runner is not None else ModelRunner(cfg, tp, seed)
+        self.runner = (runner if runner is not None
+                       else ModelRunner(cfg, tp, seed, weight_pool=weight_pool))
         self.alloc = PageAllocator(self.runner.num_pages, cfg.page_size,
                                    cfg.enable_prefix_caching)
         self.host_tier = None

@@ -14,7 +14,6 @@ import torch
 
 from dynamo_amd import ops
 from dynamo_amd.models.layers import AttnMetadata, TPContext
-from dynamo_amd.models.registry import build_model
 from .config import EngineConfig
 from .kv_cache import KVCachePool
 from .scheduler import SchedulerOutput, ScheduledSeq
@@ -22,13 +21,20 @@ from .scheduler import SchedulerOutput, ScheduledSeq
 
 class ModelRunner:
     def __init__(self, cfg: EngineConfig, tp: Optional[TPContext] = None,
-                 seed: int = 0):
+                 seed: int = 0, weight_pool=None):
         self.cfg = cfg
         self.device = torch.device(cfg.device)
         self.dtype = cfg.torch_dtype
         self.tp = tp or TPContext(cfg.tp_size, cfg.tp_rank)
         m = cfg.model
-        self.model = build_model(m, self.device, self.dtype, self.tp, seed)
+        from dynamo_amd.models.registry import build_model
+        if weight_pool is not None:
+            from dynamo_amd.gms import weight_allocator
+            with weight_allocator(weight_pool):
+                self.model = build_model(m, self.device, self.dtype, self.tp,
+                                         seed)
+        else:
+            self.model = build_model(m, self.device, self.dtype, self.tp, seed)
         self.hkv_local = max(1, m.num_kv_heads // self.tp.size)
         self.hq_local = m.num_q_heads // self.tp.size
 

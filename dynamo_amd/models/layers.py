@@ -62,10 +62,30 @@ class TPContext:
 _DEFAULT_TP = TPContext()
 
 
+def _alloc(shape, device, dtype):
+    """Allocate a weight tensor — from the active GMS weight pool when one
+    is set (zero-copy shared weights; dynamo_amd.gms), else a fresh tensor.
+    Returns (tensor, needs_init)."""
+    from dynamo_amd.gms import current_allocator
+    pool = current_allocator()
+    if pool is not None:
+        return pool.allocate(shape, dtype), pool.needs_init
+    return torch.empty(shape, device=device, dtype=dtype), True
+
+
 def init_weight(shape, device, dtype, std=0.02, generator=None):
-    w = torch.empty(shape, device=device, dtype=dtype)
-    with torch.no_grad():
-        w.normal_(0.0, std, generator=generator)
+    w, needs = _alloc(shape, device, dtype)
+    if needs:
+        with torch.no_grad():
+            w.normal_(0.0, std, generator=generator)
+    return w
+
+
+def init_const(shape, device, dtype, val=1.0):
+    w, needs = _alloc(shape, device, dtype)
+    if needs:
+        with torch.no_grad():
+            w.fill_(val)
     return w
 
 
@@ -75,12 +95,18 @@ def init_sharded(shape, device, dtype, tp: "TPContext", dim: int, std=0.02):
     Every rank consumes the same RNG stream, so shards are consistent and a
     TP-N model computes exactly the same function as TP-1 (up to reduction
     order) — which keeps disagg/TP determinism tests meaningful."""
-    full = init_weight(shape, device, dtype, std)
     if tp.size == 1:
-        return full
+        return init_weight(shape, device, dtype, std)
     n = shape[dim] // tp.size
-    shard = full.narrow(dim, tp.rank * n, n).contiguous()
-    del full
+    shard_shape = list(shape)
+    shard_shape[dim] = n
+    shard, needs = _alloc(tuple(shard_shape), device, dtype)
+    if needs:
+        full = torch.empty(shape, device=device, dtype=dtype)
+        with torch.no_grad():
+            full.normal_(0.0, std)
+        shard.copy_(full.narrow(dim, tp.rank * n, n))
+        del full
     return shard
 
 
@@ -96,10 +122,24 @@ class Attention(torch.nn.Module):
         self.hd = cfg.head_dim
         self.scale = self.hd ** -0.5
         D = cfg.hidden_size
-        wq = init_sharded((cfg.num_q_heads * self.hd, D), device, dtype, tp, 0)
-        wk = init_sharded((cfg.num_kv_heads * self.hd, D), device, dtype, tp, 0)
-        wv = init_sharded((cfg.num_kv_heads * self.hd, D), device, dtype, tp, 0)
-        self.wqkv = torch.cat([wq, wk, wv], dim=0)
+        # fused qkv allocated as ONE tensor (pool-friendly: stays zero-copy
+        # under a GMS weight pool); init replays the same RNG stream as
+        # separate full q/k/v generation + sharding
+        qkv_rows = (self.hq + 2 * self.hkv) * self.hd
+        self.wqkv, needs = _alloc((qkv_rows, D), device, dtype)
+        if needs:
+            with torch.no_grad():
+                off = 0
+                for hf, hl in ((cfg.num_q_heads, self.hq),
+                               (cfg.num_kv_heads, self.hkv),
+                               (cfg.num_kv_heads, self.hkv)):
+                    full = torch.empty(hf * self.hd, D, device=device,
+                                       dtype=dtype).normal_(0.0, 0.02)
+                    rows = hl * self.hd
+                    self.wqkv[off:off + rows].copy_(
+                        full[tp.rank * rows:(tp.rank + 1) * rows]
+                        if tp.size > 1 else full)
+                    off += rows
         self.wo = init_sharded((D, cfg.num_q_heads * self.hd), device, dtype,
                                tp, 1)
 
@@ -139,9 +179,15 @@ class SwiGLUMLP(torch.nn.Module):
         D = cfg.hidden_size
         I = cfg.intermediate_size // tp.size
         self.I = I
-        wg = init_sharded((cfg.intermediate_size, D), device, dtype, tp, 0)
-        wu = init_sharded((cfg.intermediate_size, D), device, dtype, tp, 0)
-        self.w_gate_up = torch.cat([wg, wu], dim=0)
+        self.w_gate_up, needs = _alloc((2 * I, D), device, dtype)
+        if needs:
+            with torch.no_grad():
+                for sec in range(2):  # gate rows then up rows
+                    full = torch.empty(cfg.intermediate_size, D, device=device,
+                                       dtype=dtype).normal_(0.0, 0.02)
+                    self.w_gate_up[sec * I:(sec + 1) * I].copy_(
+                        full[tp.rank * I:(tp.rank + 1) * I]
+                        if tp.size > 1 else full)
         self.w_down = init_sharded((D, cfg.intermediate_size), device, dtype,
                                    tp, 1)
 

@@ -11,7 +11,7 @@ import torch
 from dynamo_amd import ops
 from dynamo_amd.ops import torch_ref
 from .layers import (AttnMetadata, Attention, SwiGLUMLP, TPContext,
-                     init_weight, linear)
+                     init_const, init_weight, linear)
 
 
 class LlamaDecoderLayer(torch.nn.Module):
@@ -19,8 +19,8 @@ class LlamaDecoderLayer(torch.nn.Module):
         super().__init__()
         self.attn = Attention(cfg, layer_idx, tp, device, dtype)
         self.mlp = SwiGLUMLP(cfg, tp, device, dtype)
-        self.input_norm_w = torch.ones(cfg.hidden_size, device=device, dtype=dtype)
-        self.post_norm_w = torch.ones(cfg.hidden_size, device=device, dtype=dtype)
+        self.input_norm_w = init_const((cfg.hidden_size,), device, dtype, 1.0)
+        self.post_norm_w = init_const((cfg.hidden_size,), device, dtype, 1.0)
         self.eps = cfg.rms_eps
 
     def forward(self, x, residual, cos_sin, kcache, vcache, meta):
@@ -47,7 +47,7 @@ class LlamaForCausalLM(torch.nn.Module):
             LlamaDecoderLayer(cfg, i, self.tp, device, dtype)
             for i in range(cfg.num_layers)
         ])
-        self.final_norm_w = torch.ones(cfg.hidden_size, device=device, dtype=dtype)
+        self.final_norm_w = init_const((cfg.hidden_size,), device, dtype, 1.0)
         self.lm_head = (self.embed if cfg.tie_embeddings
                         else init_weight((cfg.vocab_size, cfg.hidden_size),
                                          device, dtype))

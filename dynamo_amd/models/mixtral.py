@@ -10,7 +10,8 @@ import torch
 import torch.nn.functional as F
 
 from dynamo_amd import ops
-from .layers import AttnMetadata, Attention, TPContext, init_weight, linear
+from .layers import (AttnMetadata, Attention, TPContext, init_const,
+                     init_weight, linear)
 
 
 class MoEMLP(torch.nn.Module):
@@ -22,12 +23,18 @@ class MoEMLP(torch.nn.Module):
         D = cfg.hidden_size
         I = cfg.intermediate_size // tp.size
         self.I = I
-        from .layers import init_sharded
+        from .layers import _alloc, init_sharded
         self.router = init_weight((self.E, D), device, dtype)
-        # fused per-expert weights: [E, 2I_local, D] and [E, D, I_local]
-        wg = init_sharded((self.E, cfg.intermediate_size, D), device, dtype, tp, 1)
-        wu = init_sharded((self.E, cfg.intermediate_size, D), device, dtype, tp, 1)
-        self.w_gate_up = torch.cat([wg, wu], dim=1)
+        # fused per-expert weights [E, 2I_local, D], one pool allocation
+        self.w_gate_up, needs = _alloc((self.E, 2 * I, D), device, dtype)
+        if needs:
+            with torch.no_grad():
+                for sec in range(2):  # gate then up
+                    full = torch.empty(self.E, cfg.intermediate_size, D,
+                                       device=device, dtype=dtype).normal_(0.0, 0.02)
+                    self.w_gate_up[:, sec * I:(sec + 1) * I].copy_(
+                        full[:, tp.rank * I:(tp.rank + 1) * I]
+                        if tp.size > 1 else full)
         self.w_down = init_sharded((self.E, D, cfg.intermediate_size), device,
                                    dtype, tp, 2)
 
@@ -77,8 +84,8 @@ class MixtralDecoderLayer(torch.nn.Module):
         super().__init__()
         self.attn = Attention(cfg, layer_idx, tp, device, dtype)
         self.moe = MoEMLP(cfg, tp, device, dtype)
-        self.input_norm_w = torch.ones(cfg.hidden_size, device=device, dtype=dtype)
-        self.post_norm_w = torch.ones(cfg.hidden_size, device=device, dtype=dtype)
+        self.input_norm_w = init_const((cfg.hidden_size,), device, dtype, 1.0)
+        self.post_norm_w = init_const((cfg.hidden_size,), device, dtype, 1.0)
         self.eps = cfg.rms_eps
 
     def forward(self, x, residual, cos_sin, kcache, vcache, meta):
@@ -105,7 +112,7 @@ class MixtralForCausalLM(torch.nn.Module):
             MixtralDecoderLayer(cfg, i, self.tp, device, dtype)
             for i in range(cfg.num_layers)
         ])
-        self.final_norm_w = torch.ones(cfg.hidden_size, device=device, dtype=dtype)
+        self.final_norm_w = init_const((cfg.hidden_size,), device, dtype, 1.0)
         self.lm_head = init_weight((cfg.vocab_size, cfg.hidden_size), device, dtype)
         from dynamo_amd.ops import torch_ref
         self.cos_sin = torch_ref.make_cos_sin_cache(

@@ -10,7 +10,8 @@ import torch
 import torch.nn.functional as F
 
 from dynamo_amd import ops
-from .layers import AttnMetadata, Attention, TPContext, init_weight, linear
+from .layers import (AttnMetadata, Attention, TPContext, init_const,
+                     init_weight, linear)
 
 
 class OPTDecoderLayer(torch.nn.Module):
@@ -18,10 +19,10 @@ class OPTDecoderLayer(torch.nn.Module):
         super().__init__()
         D = cfg.hidden_size
         self.attn = Attention(cfg, layer_idx, tp, device, dtype)
-        self.ln1_w = torch.ones(D, device=device, dtype=dtype)
-        self.ln1_b = torch.zeros(D, device=device, dtype=dtype)
-        self.ln2_w = torch.ones(D, device=device, dtype=dtype)
-        self.ln2_b = torch.zeros(D, device=device, dtype=dtype)
+        self.ln1_w = init_const((D,), device, dtype, 1.0)
+        self.ln1_b = init_const((D,), device, dtype, 0.0)
+        self.ln2_w = init_const((D,), device, dtype, 1.0)
+        self.ln2_b = init_const((D,), device, dtype, 0.0)
         self.fc1 = init_weight((cfg.intermediate_size, D), device, dtype)
         self.fc2 = init_weight((D, cfg.intermediate_size), device, dtype)
 
@@ -51,8 +52,8 @@ class OPTForCausalLM(torch.nn.Module):
             OPTDecoderLayer(cfg, i, self.tp, device, dtype)
             for i in range(cfg.num_layers)
         ])
-        self.final_ln_w = torch.ones(cfg.hidden_size, device=device, dtype=dtype)
-        self.final_ln_b = torch.zeros(cfg.hidden_size, device=device, dtype=dtype)
+        self.final_ln_w = init_const((cfg.hidden_size,), device, dtype, 1.0)
+        self.final_ln_b = init_const((cfg.hidden_size,), device, dtype, 0.0)
         self.lm_head = self.embed  # tied
         # OPT has no rotary cache; Attention.forward still expects one.
         # A zero-rotation table (cos=1, sin=0) makes rope a no-op.

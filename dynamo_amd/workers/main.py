@@ -40,6 +40,8 @@ def build_parser():
     p.add_argument("--no-hip-graphs", action="store_true")
     p.add_argument("--seed", type=int, default=0)
     p.add_argument("--host", default="127.0.0.1")
+    p.add_argument("--gms", action="store_true",
+                   help="import weights zero-copy from a GMS weight server")
     return p
 
 
@@ -58,6 +60,18 @@ def make_engine_from_args(args) -> LLMEngine:
     if device.startswith("cuda"):
         from dynamo_amd.utils import enable_tunableop
         enable_tunableop(tuning=False)
+    weight_pool = None
+    if args.gms:
+        from dynamo_amd.gms import WeightPool
+        from dynamo_amd.runtime import make_discovery
+        disc = make_discovery(args.discovery)
+        metas = [i for i in disc.list(args.namespace, "gms")
+                 if i.metadata.get("model") == args.model]
+        if not metas:
+            raise RuntimeError(f"no GMS server found for {args.model}")
+        weight_pool = WeightPool.open(metas[0].metadata["gms"], device)
+        logging.info("imported %d weight bytes zero-copy from GMS %s",
+                     weight_pool.buffer.numel(), metas[0].instance_id)
     cfg = EngineConfig(
         model=mc, device=device, page_size=args.page_size,
         max_num_seqs=args.max_num_seqs,
@@ -68,7 +82,7 @@ def make_engine_from_args(args) -> LLMEngine:
         enable_prefix_caching=not args.no_prefix_caching,
         enable_hip_graphs=not args.no_hip_graphs,
         worker_type=args.worker_type)
-    return LLMEngine(cfg, seed=args.seed)
+    return LLMEngine(cfg, seed=args.seed, weight_pool=weight_pool)
 
 
 async def async_main(args):
