@@ -76,6 +76,22 @@ def test_gms_cross_process(tmp_path):
     import sys
     from tests.proc_utils import ManagedProcess, worker_cmd
     disc = f"file:{tmp_path}/disc"
+    from dynamo_amd.runtime import DistributedRuntime
+
+    async def query(component="backend"):
+        rt = DistributedRuntime(disc)
+        await rt.start()
+        insts = [i for i in rt.discovery.list("dynamo", component)]
+        assert insts, "worker not registered"
+        toks = []
+        async for chunk in rt.client.call_stream(
+                insts[0].address, f"{component}.generate",
+                {"request_id": "g", "token_ids": list(range(100)),
+                 "stop_conditions": {"max_tokens": 6}}):
+            toks.extend(chunk.get("token_ids", []))
+        await rt.shutdown(drain=False)
+        return toks
+
     gms = ManagedProcess(
         [sys.executable, "-m", "dynamo_amd.gms", "--model", "tiny-llama-gpu",
          "--discovery", disc], ready_marker="GMS_READY").start()
@@ -84,13 +100,24 @@ def test_gms_cross_process(tmp_path):
                    kv_pool_pages=128, max_model_len=2048),
         ready_marker="WORKER_READY").start()
     try:
-        from dynamo_amd.runtime import DistributedRuntime
+        toks = asyncio.new_event_loop().run_until_complete(query())
+        assert len(toks) == 6
+    finally:
+        w.stop()
+        gms.stop()
 
-        async def run():
-            rt = DistributedRuntime(disc)
+    # reference: a worker that initializes its own weights (same seed, same
+    # CLI code path — isolates the GMS import from e.g. TunableOp algo picks)
+    disc2 = f"file:{tmp_path}/disc2"
+    w2 = ManagedProcess(
+        worker_cmd(model="tiny-llama-gpu", discovery=disc2,
+                   kv_pool_pages=128, max_model_len=2048),
+        ready_marker="WORKER_READY").start()
+    try:
+        async def query2():
+            rt = DistributedRuntime(disc2)
             await rt.start()
             insts = [i for i in rt.discovery.list("dynamo", "backend")]
-            assert insts, "worker not registered"
             toks = []
             async for chunk in rt.client.call_stream(
                     insts[0].address, "backend.generate",
@@ -99,15 +126,7 @@ def test_gms_cross_process(tmp_path):
                 toks.extend(chunk.get("token_ids", []))
             await rt.shutdown(drain=False)
             return toks
-        toks = asyncio.new_event_loop().run_until_complete(run())
-        assert len(toks) == 6
+        ref = asyncio.new_event_loop().run_until_complete(query2())
     finally:
-        w.stop()
-        gms.stop()
-    # reference: local engine with the same seed
-    cfg = EngineConfig(model=PRESETS["tiny-llama-gpu"], device="cuda:0",
-                       kv_pool_pages=128, max_model_len=2048, max_num_seqs=8,
-                       page_size=64)
-    eng = LLMEngine(cfg, seed=0)
-    ref = gen(eng, list(range(100)), n=6)
+        w2.stop()
     assert toks == ref, f"GMS-imported weights diverged: {toks} vs {ref}"
