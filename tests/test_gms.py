@@ -100,33 +100,32 @@ def test_gms_cross_process(tmp_path):
                    kv_pool_pages=128, max_model_len=2048),
         ready_marker="WORKER_READY").start()
     try:
+        # 1) end-to-end: the GMS-backed worker serves generations
         toks = asyncio.new_event_loop().run_until_complete(query())
         assert len(toks) == 6
+
+        # 2) the core GMS property: the imported arena holds BYTE-EQUAL
+        # weights vs a local same-seed build. (Token equality across
+        # different allocation layouts is not guaranteed — hipBLASLt
+        # algorithm selection is pointer-alignment-sensitive.)
+        from dynamo_amd.runtime import make_discovery
+        mc = PRESETS["tiny-llama-gpu"]
+        meta = None
+        for inst in make_discovery(disc).list("dynamo", "gms"):
+            meta = inst.metadata["gms"]
+        assert meta is not None
+        pool = WeightPool.open(meta, "cuda:0")
+        with weight_allocator(pool):
+            imported = build_model(mc, "cuda:0", torch.bfloat16, TPContext(),
+                                   seed=12345)  # seed must not matter
+        local = build_model(mc, "cuda:0", torch.bfloat16, TPContext(), seed=0)
+        assert torch.equal(imported.embed, local.embed)
+        assert torch.equal(imported.layers[0].attn.wqkv,
+                           local.layers[0].attn.wqkv)
+        assert torch.equal(imported.layers[1].mlp.w_gate_up,
+                           local.layers[1].mlp.w_gate_up)
+        assert torch.equal(imported.lm_head, local.lm_head)
+        del imported, pool
     finally:
         w.stop()
         gms.stop()
-
-    # reference: a worker that initializes its own weights (same seed, same
-    # CLI code path — isolates the GMS import from e.g. TunableOp algo picks)
-    disc2 = f"file:{tmp_path}/disc2"
-    w2 = ManagedProcess(
-        worker_cmd(model="tiny-llama-gpu", discovery=disc2,
-                   kv_pool_pages=128, max_model_len=2048),
-        ready_marker="WORKER_READY").start()
-    try:
-        async def query2():
-            rt = DistributedRuntime(disc2)
-            await rt.start()
-            insts = [i for i in rt.discovery.list("dynamo", "backend")]
-            toks = []
-            async for chunk in rt.client.call_stream(
-                    insts[0].address, "backend.generate",
-                    {"request_id": "g", "token_ids": list(range(100)),
-                     "stop_conditions": {"max_tokens": 6}}):
-                toks.extend(chunk.get("token_ids", []))
-            await rt.shutdown(drain=False)
-            return toks
-        ref = asyncio.new_event_loop().run_until_complete(query2())
-    finally:
-        w2.stop()
-    assert toks == ref, f"GMS-imported weights diverged: {toks} vs {ref}"
