@@ -74,10 +74,21 @@ def _alloc(shape, device, dtype):
 
 
 def init_weight(shape, device, dtype, std=0.02, generator=None):
+    from dynamo_amd.gms import current_allocator
+    pooled = current_allocator() is not None
     w, needs = _alloc(shape, device, dtype)
     if needs:
         with torch.no_grad():
-            w.normal_(0.0, std, generator=generator)
+            if pooled:
+                # pool-backed view: generate into a fresh tensor first —
+                # normal_ directly into a reinterpreted-uint8 view draws a
+                # DIFFERENT sequence on ROCm (HW-observed), which would
+                # break GMS build-vs-plain determinism
+                tmp = torch.empty(shape, device=device, dtype=dtype)
+                tmp.normal_(0.0, std, generator=generator)
+                w.copy_(tmp)
+            else:
+                w.normal_(0.0, std, generator=generator)
     return w
 
 
