@@ -73,6 +73,34 @@ class LLMEngine:
             from .graphs import GraphRunner
             self.graph_runner = GraphRunner(self.runner, cfg.max_num_seqs)
         self._last_sampled = None
+        self._lora = None
+
+    # -- LoRA (engine-level activation; see dynamo_amd/lora) -----------
+    @property
+    def lora(self):
+        if self._lora is None and hasattr(self.runner, "model"):
+            from dynamo_amd.lora import LoRAManager
+            self._lora = LoRAManager(self.runner.model)
+        return self._lora
+
+    def _invalidate_graphs(self):
+        if self.graph_runner is not None:
+            self.graph_runner.graphs.clear()
+            self.graph_runner.dirty = True
+
+    def load_lora(self, name, path=None, rank=8, alpha=16.0, seed=0,
+                  activate=True):
+        self.lora.load(name, path=path, rank=rank, alpha=alpha, seed=seed)
+        if activate:
+            self.lora.activate(name)
+            self._invalidate_graphs()
+
+    def unload_lora(self, name):
+        self.lora.unload(name)
+        self._invalidate_graphs()
+
+    def list_loras(self):
+        return [] if self._lora is None else self._lora.list()
 
     # ------------------------------------------------------------------
     def add_request(self, req_id: str, prompt_tokens: List[int],

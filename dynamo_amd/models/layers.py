@@ -45,6 +45,15 @@ def linear(x: torch.Tensor, w: torch.Tensor) -> torch.Tensor:
     return F.linear(x, w)
 
 
+def linear_lora(x: torch.Tensor, w: torch.Tensor, lora, key: str) -> torch.Tensor:
+    """F.linear plus the active LoRA low-rank delta, if any."""
+    y = F.linear(x, w)
+    if lora is not None and key in lora:
+        A, B, scale = lora[key]
+        y = y + F.linear(F.linear(x, A), B) * scale
+    return y
+
+
 class TPContext:
     """Tensor-parallel context: rank/world + the RCCL process group."""
 
@@ -153,10 +162,11 @@ class Attention(torch.nn.Module):
                     off += rows
         self.wo = init_sharded((D, cfg.num_q_heads * self.hd), device, dtype,
                                tp, 1)
+        self.lora = None  # set by dynamo_amd.lora.LoRAManager
 
     def forward(self, x, cos_sin, kcache, vcache, meta: AttnMetadata):
         T = x.shape[0]
-        qkv = linear(x, self.wqkv)
+        qkv = linear_lora(x, self.wqkv, self.lora, "qkv")
         q, k, v = qkv.split([self.hq * self.hd, self.hkv * self.hd,
                              self.hkv * self.hd], dim=-1)
         q = q.contiguous()
@@ -179,7 +189,7 @@ class Attention(torch.nn.Module):
                 qh[nd:].contiguous(), kcache, vcache, meta.prefill_page_table,
                 meta.seq_q_start, meta.seq_q_len, meta.seq_ctx_len, self.scale,
                 meta.prefill_tiles)
-        o = linear(out.view(T, self.hq * self.hd), self.wo)
+        o = linear_lora(out.view(T, self.hq * self.hd), self.wo, self.lora, "o")
         return self.tp.all_reduce(o)
 
 
@@ -201,9 +211,11 @@ class SwiGLUMLP(torch.nn.Module):
                         if tp.size > 1 else full)
         self.w_down = init_sharded((D, cfg.intermediate_size), device, dtype,
                                    tp, 1)
+        self.lora = None  # set by dynamo_amd.lora.LoRAManager
 
     def forward(self, x):
-        gu = linear(x, self.w_gate_up)
+        gu = linear_lora(x, self.w_gate_up, self.lora, "gate_up")
         # silu_mul expects [., 2I] with gate then up
         act = ops.silu_mul(gu)
-        return self.tp.all_reduce(linear(act, self.w_down))
+        return self.tp.all_reduce(
+            linear_lora(act, self.w_down, self.lora, "down"))

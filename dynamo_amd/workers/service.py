@@ -89,6 +89,9 @@ class WorkerService:
         self.comp.serve_endpoint("get_perf_metrics", self.get_perf_metrics)
         self.comp.serve_endpoint("kv_events", self.kv_events)
         self.comp.serve_endpoint("release_kv", self.release_kv)
+        self.comp.serve_endpoint("load_lora", self.load_lora)
+        self.comp.serve_endpoint("unload_lora", self.unload_lora)
+        self.comp.serve_endpoint("list_loras", self.list_loras)
         metadata = {"worker_type": self.worker_type}
         if self.engine.runner.kv_pool is not None:
             metadata["kv_transfer"] = pool_transfer_metadata(
@@ -272,3 +275,22 @@ class WorkerService:
         async with self._engine_lock:
             self.engine.release_held(payload["request_id"])
         yield {"status": "ok"}
+
+    # -- LoRA endpoints (reference: vllm/worker_factory.py:1378-1413) ----
+    async def load_lora(self, payload, ctx):
+        async with self._engine_lock:
+            await asyncio.to_thread(
+                self.engine.load_lora, payload["name"],
+                payload.get("path"), int(payload.get("rank", 8)),
+                float(payload.get("alpha", 16.0)), int(payload.get("seed", 0)))
+            self.comp.update_metadata(loras=self.engine.list_loras())
+        yield {"status": "ok", "loras": self.engine.list_loras()}
+
+    async def unload_lora(self, payload, ctx):
+        async with self._engine_lock:
+            self.engine.unload_lora(payload["name"])
+            self.comp.update_metadata(loras=self.engine.list_loras())
+        yield {"status": "ok", "loras": self.engine.list_loras()}
+
+    async def list_loras(self, payload, ctx):
+        yield {"loras": self.engine.list_loras()}
