@@ -65,6 +65,8 @@ class WorkerService:
         self._req_counter = 0
         self._puller: Optional[KvPuller] = None
         self.worker_type = engine.cfg.worker_type
+        self._paused = asyncio.Event()
+        self._paused.set()  # set = running
 
     # ------------------------------------------------------------------
     def model_card(self) -> dict:
@@ -89,6 +91,8 @@ class WorkerService:
         self.comp.serve_endpoint("get_perf_metrics", self.get_perf_metrics)
         self.comp.serve_endpoint("kv_events", self.kv_events)
         self.comp.serve_endpoint("release_kv", self.release_kv)
+        self.comp.serve_endpoint("pause", self.pause)
+        self.comp.serve_endpoint("resume", self.resume)
         self.comp.serve_endpoint("load_lora", self.load_lora)
         self.comp.serve_endpoint("unload_lora", self.unload_lora)
         self.comp.serve_endpoint("list_loras", self.list_loras)
@@ -113,9 +117,11 @@ class WorkerService:
     # ------------------------------------------------------------------
     async def _engine_loop(self):
         while True:
+            await self._paused.wait()  # snapshot lifecycle pause
             if not self.engine.has_work():
                 self._work.clear()
                 await self._work.wait()
+                await self._paused.wait()
             async with self._engine_lock:
                 outputs = await asyncio.to_thread(self.engine.step)
             for so in outputs:
@@ -275,6 +281,22 @@ class WorkerService:
         async with self._engine_lock:
             self.engine.release_held(payload["request_id"])
         yield {"status": "ok"}
+
+    # -- snapshot lifecycle: pause -> (snapshot externally) -> resume ----
+    # (reference: common/snapshot/lifecycle.py:32-108 + engine pause
+    # controller handlers.py:339; with GMS the weights already live in an
+    # external process, so pause/resume is the whole engine-side story)
+    async def pause(self, payload, ctx):
+        self._paused.clear()
+        async with self._engine_lock:
+            pass  # wait for the in-flight step to finish
+        yield {"status": "paused",
+               "in_flight": self.engine.scheduler.num_running()}
+
+    async def resume(self, payload, ctx):
+        self._paused.set()
+        self._work.set()
+        yield {"status": "running"}
 
     # -- LoRA endpoints (reference: vllm/worker_factory.py:1378-1413) ----
     async def load_lora(self, payload, ctx):

@@ -208,3 +208,37 @@ def test_migration_on_worker_death():
         assert total == 30, f"expected 30 tokens, got {total}"
         await teardown(services[1:], mgr, client)
     run(main())
+
+
+def test_pause_resume_lifecycle():
+    async def main():
+        shared, services, mgr, client = await with_stack(nworkers=1)
+        ws, rt = services[0]
+        for w, _ in services:
+            w.engine.runner.decode_step_ms = 5
+        addr = rt.server.address
+        # start a long generation, pause mid-flight, verify it stalls,
+        # resume, verify completion
+        import asyncio as aio
+        entry = mgr.get("mock-model")
+        toks = []
+
+        async def consume():
+            async for c in mgr.generate_tokens(
+                    entry, entry.tokenizer.encode("pause me"),
+                    {"temperature": 0.0}, {"max_tokens": 40}):
+                toks.extend(c.get("token_ids", []))
+        task = aio.create_task(consume())
+        while len(toks) < 3:
+            await aio.sleep(0.01)
+        r = await rt.client.call(addr, "backend.pause", {})
+        assert r["status"] == "paused"
+        n_at_pause = len(toks)
+        await aio.sleep(0.3)
+        assert len(toks) <= n_at_pause + 2, "engine kept stepping while paused"
+        r = await rt.client.call(addr, "backend.resume", {})
+        assert r["status"] == "running"
+        await aio.wait_for(task, 30)
+        assert len(toks) == 40
+        await teardown(services, mgr, client)
+    run(main())
