@@ -67,11 +67,14 @@ class LLMEngine:
         self._held: Dict[str, Request] = {}  # finished but KV retained (disagg)
         # hipGraph decode fast path
         self.graph_runner = None
-        # MoE routing builds per-step host-side tile lists -> not capturable
-        if (cfg.enable_hip_graphs and self.runner.device.type == "cuda"
-                and cfg.tp_size == 1 and cfg.model.num_experts == 0):
+        if cfg.enable_hip_graphs and self.runner.device.type == "cuda":
             from .graphs import GraphRunner
-            self.graph_runner = GraphRunner(self.runner, cfg.max_num_seqs)
+            # full hipGraph capture only for TP=1 dense models; TP ranks and
+            # MoE use the persistent-buffer eager fast path (static decode
+            # buffers, no per-step tensor rebuilds)
+            use_graphs = cfg.tp_size == 1 and cfg.model.num_experts == 0
+            self.graph_runner = GraphRunner(self.runner, cfg.max_num_seqs,
+                                            use_graphs=use_graphs)
         self._last_sampled = None
         self._lora = None
 

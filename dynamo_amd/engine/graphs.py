@@ -24,8 +24,9 @@ BUCKETS = [1, 2, 4, 8, 16, 24, 32, 48, 64, 96, 128, 192, 256]
 
 
 class GraphRunner:
-    def __init__(self, runner, max_batch: int):
+    def __init__(self, runner, max_batch: int, use_graphs: bool = True):
         self.runner = runner
+        self.use_graphs = use_graphs
         cfg = runner.cfg
         self.cfg = cfg
         self.device = runner.device
@@ -141,6 +142,14 @@ class GraphRunner:
             pages = self.page_table[:n].gather(1, page_idx).squeeze(1)
             self.slot_mapping[:n] = pages.to(torch.int64) * self.ps + pos % self.ps
         self._first_after_rebuild = False
+        if not self.use_graphs:
+            # persistent-buffer eager path (TP ranks: RCCL-in-graph capture
+            # is not exercised; MoE: host-side tile lists aren't capturable)
+            meta = self._meta(bc)
+            h = self.runner.model.forward(self.input_ids[:bc],
+                                          self.runner.kv_pool, meta)
+            self.logits[:bc] = self.runner.model.compute_logits(h)
+            return self.logits[:n]
         if bc not in self.graphs:
             self._capture(bc)
         self.graphs[bc].replay()

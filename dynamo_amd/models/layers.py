@@ -146,18 +146,23 @@ class Attention(torch.nn.Module):
         # under a GMS weight pool); init replays the same RNG stream as
         # separate full q/k/v generation + sharding
         qkv_rows = (self.hq + 2 * self.hkv) * self.hd
+        # when tp.size > num_kv_heads, KV heads are REPLICATED: rank r uses
+        # kv-head shard (r * num_kv_heads) // tp.size (matches the GQA
+        # mapping of the rank's q heads)
+        kv_idx = (tp.rank if tp.size <= cfg.num_kv_heads
+                  else (tp.rank * cfg.num_kv_heads) // tp.size)
         self.wqkv, needs = _alloc((qkv_rows, D), device, dtype)
         if needs:
             with torch.no_grad():
                 off = 0
-                for hf, hl in ((cfg.num_q_heads, self.hq),
-                               (cfg.num_kv_heads, self.hkv),
-                               (cfg.num_kv_heads, self.hkv)):
+                for hf, hl, idx in ((cfg.num_q_heads, self.hq, tp.rank),
+                                    (cfg.num_kv_heads, self.hkv, kv_idx),
+                                    (cfg.num_kv_heads, self.hkv, kv_idx)):
                     full = torch.empty(hf * self.hd, D, device=device,
                                        dtype=dtype).normal_(0.0, 0.02)
                     rows = hl * self.hd
                     self.wqkv[off:off + rows].copy_(
-                        full[tp.rank * rows:(tp.rank + 1) * rows]
+                        full[idx * rows:(idx + 1) * rows]
                         if tp.size > 1 else full)
                     off += rows
         self.wo = init_sharded((D, cfg.num_q_heads * self.hd), device, dtype,
