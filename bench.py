@@ -46,12 +46,17 @@ def parse_args():
     p.add_argument("--kv-pool-pages", type=int, default=0)
     p.add_argument("--device", default=None, help="cpu for gloo testing")
     p.add_argument("--max-batched-tokens", type=int, default=0)
+    p.add_argument("--moe-ep", action="store_true",
+                   help="expert parallelism for MoE models (auto for N>1)")
     return p.parse_args()
 
 
 def make_cfg(args, mc, device, world=1, rank=0, worker_type="aggregated",
              max_seqs=None):
     from dynamo_amd.engine import EngineConfig
+    if mc.num_experts and world > 1 and (args.moe_ep or True):
+        import dataclasses
+        mc = dataclasses.replace(mc, moe_ep=True)
     return EngineConfig(
         model=mc, device=device, page_size=args.page_size,
         max_num_seqs=max_seqs or args.conc_per_gpu,

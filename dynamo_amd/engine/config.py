@@ -32,6 +32,7 @@ class ModelConfig:
     # MoE (mixtral)
     num_experts: int = 0
     num_experts_per_tok: int = 2
+    moe_ep: bool = False   # expert parallelism over the TP group (vs MoE-TP)
     # OPT-style extras
     activation: str = "silu"       # silu | gelu
     norm: str = "rmsnorm"          # rmsnorm | layernorm

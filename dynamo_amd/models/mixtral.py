@@ -15,16 +15,49 @@ from .layers import (AttnMetadata, Attention, TPContext, init_const,
 
 
 class MoEMLP(torch.nn.Module):
+    """Two parallelism modes over the TP group (reference parity:
+    planner/config/parallelization.py moe_tp_size / moe_ep_size):
+      MoE-TP (default): every rank holds all experts with the intermediate
+        dim sharded; all-reduce after down-proj.
+      EP (cfg.model.moe_ep): experts are SHARDED across ranks at full
+        intermediate width; every rank computes its local experts for the
+        whole batch and the same all-reduce sums the partial outputs —
+        each (token, expert) pair is computed on exactly one rank."""
+
     def __init__(self, cfg, tp: TPContext, device, dtype):
         super().__init__()
         self.tp = tp
         self.E = cfg.num_experts
         self.topk = cfg.num_experts_per_tok
+        self.ep = cfg.moe_ep and tp.size > 1
         D = cfg.hidden_size
-        I = cfg.intermediate_size // tp.size
-        self.I = I
         from .layers import _alloc, init_sharded
         self.router = init_weight((self.E, D), device, dtype)
+        if self.ep:
+            assert self.E % tp.size == 0, "num_experts % ep_size != 0"
+            El = self.E // tp.size
+            self.e0 = tp.rank * El
+            self.El = El
+            I = cfg.intermediate_size
+            self.I = I
+            self.w_gate_up, needs = _alloc((El, 2 * I, D), device, dtype)
+            self.w_down, needs2 = _alloc((El, D, I), device, dtype)
+            if needs:
+                with torch.no_grad():
+                    for sec in range(2):  # gate then up (same RNG order)
+                        full = torch.empty(self.E, I, D, device=device,
+                                           dtype=dtype).normal_(0.0, 0.02)
+                        self.w_gate_up[:, sec * I:(sec + 1) * I].copy_(
+                            full[self.e0:self.e0 + El])
+            if needs2:
+                with torch.no_grad():
+                    full = torch.empty(self.E, D, I, device=device,
+                                       dtype=dtype).normal_(0.0, 0.02)
+                    self.w_down.copy_(full[self.e0:self.e0 + El])
+            return
+        self.e0, self.El = 0, self.E
+        I = cfg.intermediate_size // tp.size
+        self.I = I
         # fused per-expert weights [E, 2I_local, D], one pool allocation
         self.w_gate_up, needs = _alloc((self.E, 2 * I, D), device, dtype)
         if needs:
@@ -53,8 +86,15 @@ class MoEMLP(torch.nn.Module):
         seg_expert = flat_expert[order]
         seg_token = flat_token[order]
         counts = torch.bincount(seg_expert, minlength=self.E)
-        xg = x[seg_token]                                  # [T*k, D]
         counts_l = counts.tolist()
+        if self.ep:
+            # keep only this rank's expert segments (sorted -> contiguous)
+            lo = sum(counts_l[:self.e0])
+            hi = lo + sum(counts_l[self.e0:self.e0 + self.El])
+            order = order[lo:hi]
+            seg_token = seg_token[lo:hi]
+            counts_l = counts_l[self.e0:self.e0 + self.El]
+        xg = x[seg_token]                                  # [local, D]
         if x.is_cuda and T <= 256:
             # decode regime: one fused weights-streaming kernel per matmul
             tiles = ops.build_moe_tiles(counts_l)
@@ -67,7 +107,7 @@ class MoEMLP(torch.nn.Module):
             # prefill regime: large segments -> hipBLASLt per expert
             yg = torch.empty_like(xg)
             s = 0
-            for e in range(self.E):
+            for e in range(len(counts_l)):
                 n = counts_l[e]
                 if n == 0:
                     continue
