@@ -41,6 +41,8 @@ void gumbel_sample(torch::Tensor out, torch::Tensor logits, torch::Tensor inv_te
 // moe.hip
 void moe_grouped_gemm(torch::Tensor y, torch::Tensor x, torch::Tensor w,
                       torch::Tensor tiles);
+void moe_grouped_gemm_seg(torch::Tensor y, torch::Tensor x, torch::Tensor w,
+                          torch::Tensor seg_start, int64_t max_tokens);
 void topk_gating(torch::Tensor topw, torch::Tensor topi, torch::Tensor logits);
 // ipc.hip
 torch::Tensor ipc_alloc(int64_t nbytes, int64_t device);
@@ -70,6 +72,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("greedy_sample", &greedy_sample);
   m.def("gumbel_sample", &gumbel_sample);
   m.def("moe_grouped_gemm", &moe_grouped_gemm);
+  m.def("moe_grouped_gemm_seg", &moe_grouped_gemm_seg);
   m.def("topk_gating", &topk_gating);
   m.def("ipc_alloc", &ipc_alloc);
   m.def("ipc_export", &ipc_export);

@@ -24,19 +24,34 @@ constexpr int kBlock = 256;
 constexpr int kMaxM = 16;   // tokens per tile (host splits larger segments)
 constexpr int kDC = 128;    // D chunk staged per iteration
 
-// tiles: [ntiles, 3] int32 = (expert, row0_in_gathered_x, m_count)
+// Segment offsets live on DEVICE (seg_start [E+1], from a cumsum of the
+// routing counts) so the launch is sync-free and hipGraph-capturable: the
+// grid covers worst-case m-tiles per expert and blocks with no tokens
+// exit. tiles mode (tiles != nullptr) is kept for host-built tile lists.
 __global__ __launch_bounds__(kBlock) void moe_gemm_kernel(
     short* __restrict__ y,        // [T, N] bf16 (gathered order)
     const short* __restrict__ x,  // [T, D] bf16 (gathered by expert)
     const short* __restrict__ w,  // [E, N, D] bf16
-    const int32_t* __restrict__ tiles,
+    const int32_t* __restrict__ tiles,      // [ntiles,3] or null
+    const int32_t* __restrict__ seg_start,  // [E+1] or null
     int D, int N, int ntiles) {
   typedef __attribute__((ext_vector_type(2))) __bf16 bf16x2_t;
-  const int tile = blockIdx.x;
-  if (tile >= ntiles) return;
-  const int e = tiles[3 * tile];
-  const int r0 = tiles[3 * tile + 1];
-  const int m = tiles[3 * tile + 2];
+  int e, r0, m;
+  if (tiles != nullptr) {
+    const int tile = blockIdx.x;
+    if (tile >= ntiles) return;
+    e = tiles[3 * tile];
+    r0 = tiles[3 * tile + 1];
+    m = tiles[3 * tile + 2];
+  } else {
+    // blockIdx.x = expert * max_m_tiles + m_tile
+    e = blockIdx.x / ntiles;            // ntiles = max m-tiles per expert
+    const int mt = blockIdx.x % ntiles;
+    const int s = seg_start[e], cnt = seg_start[e + 1] - s;
+    if (mt * kMaxM >= cnt) return;
+    r0 = s + mt * kMaxM;
+    m = min(kMaxM, cnt - mt * kMaxM);
+  }
   const int n = blockIdx.y * kBlock + threadIdx.x;  // this lane's W row
   const short* wrow = w + ((int64_t)e * N + n) * D;
 
@@ -141,7 +156,29 @@ void moe_grouped_gemm(torch::Tensor y, torch::Tensor x, torch::Tensor w,
   dim3 grid(ntiles, (N + kBlock - 1) / kBlock);
  hipLaunchKernelGGL(( moe_gemm_kernel), dim3(grid), dim3(kBlock), 0, stream, 
       (short*)y.data_ptr(), (const short*)x.data_ptr(),
-      (const short*)w.data_ptr(), tiles.data_ptr<int32_t>(), D, N, ntiles);
+      (const short*)w.data_ptr(), tiles.data_ptr<int32_t>(), nullptr,
+      D, N, ntiles);
+  HIP_CHECK_KERNEL();
+}
+
+// sync-free variant: segment offsets on device (hipGraph-capturable).
+void moe_grouped_gemm_seg(torch::Tensor y, torch::Tensor x, torch::Tensor w,
+                          torch::Tensor seg_start, int64_t max_tokens) {
+  TORCH_CHECK(x.is_cuda() && x.dtype() == torch::kBFloat16);
+  TORCH_CHECK(seg_start.dtype() == torch::kInt32);
+  const int D = x.size(1);
+  const int N = w.size(1);
+  const int E = w.size(0);
+  TORCH_CHECK(seg_start.numel() == E + 1);
+  TORCH_CHECK(D % kDC == 0, "hidden size must be a multiple of 128");
+  const int max_mt = (int)((max_tokens + kMaxM - 1) / kMaxM);
+  if (max_mt == 0) return;
+  auto stream = at::cuda::getCurrentHIPStream();
+  dim3 grid(E * max_mt, (N + kBlock - 1) / kBlock);
+ hipLaunchKernelGGL(( moe_gemm_kernel), dim3(grid), dim3(kBlock), 0, stream, 
+      (short*)y.data_ptr(), (const short*)x.data_ptr(),
+      (const short*)w.data_ptr(), nullptr, seg_start.data_ptr<int32_t>(),
+      D, N, max_mt);
   HIP_CHECK_KERNEL();
 }
 

@@ -69,10 +69,10 @@ class LLMEngine:
         self.graph_runner = None
         if cfg.enable_hip_graphs and self.runner.device.type == "cuda":
             from .graphs import GraphRunner
-            # full hipGraph capture only for TP=1 dense models; TP ranks and
-            # MoE use the persistent-buffer eager fast path (static decode
-            # buffers, no per-step tensor rebuilds)
-            use_graphs = cfg.tp_size == 1 and cfg.model.num_experts == 0
+            # full hipGraph capture for TP=1 (MoE decode routing is fully
+            # device-side, so it captures too); TP ranks use the
+            # persistent-buffer eager fast path (RCCL-in-graph untested)
+            use_graphs = cfg.tp_size == 1
             self.graph_runner = GraphRunner(self.runner, cfg.max_num_seqs,
                                             use_graphs=use_graphs)
         self._last_sampled = None

@@ -86,9 +86,28 @@ class MoEMLP(torch.nn.Module):
         seg_expert = flat_expert[order]
         seg_token = flat_token[order]
         counts = torch.bincount(seg_expert, minlength=self.E)
+        if x.is_cuda and T <= 256:
+            # decode regime: fully device-side segment offsets — sync-free
+            # and hipGraph-capturable. EP ranks pass their expert window's
+            # offsets; the gathered rows outside the window are never
+            # touched by the kernel (their yg rows stay undefined and are
+            # excluded from the index_add below via the host EP slice).
+            seg_start_full = torch.zeros(self.E + 1, dtype=torch.int32,
+                                         device=x.device)
+            seg_start_full[1:] = torch.cumsum(counts, 0).to(torch.int32)
+            if not self.ep:
+                xg = x[seg_token]
+                gu = ops.moe_grouped_gemm_seg(xg, self.w_gate_up,
+                                              seg_start_full, T * self.topk)
+                act = ops.silu_mul(gu)
+                yg = ops.moe_grouped_gemm_seg(act, self.w_down,
+                                              seg_start_full, T * self.topk)
+                w = topw.reshape(-1)[order].unsqueeze(-1)
+                out.index_add_(0, seg_token, (yg * w).to(x.dtype))
+                return self.tp.all_reduce(out)
+        # host-side paths: EP slicing and/or large prefill segments
         counts_l = counts.tolist()
         if self.ep:
-            # keep only this rank's expert segments (sorted -> contiguous)
             lo = sum(counts_l[:self.e0])
             hi = lo + sum(counts_l[self.e0:self.e0 + self.El])
             order = order[lo:hi]
@@ -96,7 +115,6 @@ class MoEMLP(torch.nn.Module):
             counts_l = counts_l[self.e0:self.e0 + self.El]
         xg = x[seg_token]                                  # [local, D]
         if x.is_cuda and T <= 256:
-            # decode regime: one fused weights-streaming kernel per matmul
             tiles = ops.build_moe_tiles(counts_l)
             tiles_t = torch.tensor(tiles, dtype=torch.int32,
                                    device=x.device).view(-1, 3)

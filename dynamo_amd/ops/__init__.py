@@ -203,6 +203,13 @@ def moe_grouped_gemm(x: torch.Tensor, w: torch.Tensor, tiles_t: torch.Tensor):
     return y
 
 
+def moe_grouped_gemm_seg(x, w, seg_start, max_tokens):
+    """Sync-free grouped GEMM: seg_start [E+1] int32 on DEVICE."""
+    y = torch.empty(x.shape[0], w.shape[1], dtype=x.dtype, device=x.device)
+    hip().moe_grouped_gemm_seg(y, x.contiguous(), w, seg_start, max_tokens)
+    return y
+
+
 def gather_pages(staging, cache, page_ids):
     if cache.is_cuda:
         hip().gather_pages(staging, cache, page_ids)
