@@ -85,7 +85,10 @@ class MoEMLP(torch.nn.Module):
         order = torch.argsort(flat_expert, stable=True)
         seg_expert = flat_expert[order]
         seg_token = flat_token[order]
-        counts = torch.bincount(seg_expert, minlength=self.E)
+        # scatter_add instead of bincount (bincount is not hipGraph-capturable)
+        counts = torch.zeros(self.E, dtype=torch.long, device=x.device)
+        counts.scatter_add_(0, seg_expert,
+                            torch.ones_like(seg_expert, dtype=torch.long))
         if x.is_cuda and T <= 256:
             # decode regime: fully device-side segment offsets — sync-free
             # and hipGraph-capturable. EP ranks pass their expert window's
