@@ -72,7 +72,10 @@ class LLMEngine:
             # full hipGraph capture for TP=1 (MoE decode routing is fully
             # device-side, so it captures too); TP ranks use the
             # persistent-buffer eager fast path (RCCL-in-graph untested)
-            use_graphs = cfg.tp_size == 1
+            import os as _os
+            moe_graphs = _os.environ.get("DYNAMO_MOE_GRAPHS", "1") != "0"
+            use_graphs = cfg.tp_size == 1 and (
+                cfg.model.num_experts == 0 or moe_graphs)
             self.graph_runner = GraphRunner(self.runner, cfg.max_num_seqs,
                                             use_graphs=use_graphs)
         self._last_sampled = None
