@@ -27,6 +27,14 @@ constexpr int kDC = 128;    // D chunk staged per iteration
 // routing counts) so the launch is sync-free and hipGraph-capturable: the
 // grid covers worst-case m-tiles per expert and blocks with no tokens
 // exit. tiles mode (tiles != nullptr) is kept for host-built tile lists.
+//
+// Layout: WAVE-per-W-row (4 rows per block) with lanes splitting D —
+// the wave's 64 lanes read 4 consecutive elements each (256-elem chunks),
+// fully coalesced weight streaming; X chunks are staged in LDS and read
+// broadcast. A lane-per-row variant measured ~2.0 TB/s effective
+// (uncoalesced stride-D streams); this layout targets the HBM roofline.
+constexpr int kDC2 = 256;  // D chunk per pass (lane covers 4 elems)
+
 __global__ __launch_bounds__(kBlock) void moe_gemm_kernel(
     short* __restrict__ y,        // [T, N] bf16 (gathered order)
     const short* __restrict__ x,  // [T, D] bf16 (gathered by expert)
@@ -35,6 +43,7 @@ __global__ __launch_bounds__(kBlock) void moe_gemm_kernel(
     const int32_t* __restrict__ seg_start,  // [E+1] or null
     int D, int N, int ntiles) {
   typedef __attribute__((ext_vector_type(2))) __bf16 bf16x2_t;
+  typedef __attribute__((ext_vector_type(4))) short short4_t;
   int e, r0, m;
   if (tiles != nullptr) {
     const int tile = blockIdx.x;
@@ -51,49 +60,48 @@ __global__ __launch_bounds__(kBlock) void moe_gemm_kernel(
     r0 = s + mt * kMaxM;
     m = min(kMaxM, cnt - mt * kMaxM);
   }
-  const int n = blockIdx.y * kBlock + threadIdx.x;  // this lane's W row
+  const int lane = threadIdx.x & 63;
+  const int wid = threadIdx.x >> 6;
+  const int n = blockIdx.y * 4 + wid;   // this WAVE's W row
   const short* wrow = w + ((int64_t)e * N + n) * D;
 
-  __shared__ short x_lds[kMaxM * kDC];
+  __shared__ short x_lds[kMaxM * kDC2];
 
   float acc[kMaxM];
 #pragma unroll
   for (int i = 0; i < kMaxM; i++) acc[i] = 0.f;
 
-  for (int dc = 0; dc < D; dc += kDC) {
+  for (int dc = 0; dc < D; dc += kDC2) {
     __syncthreads();
-    // stage X chunk [m][kDC]
-    for (int i = threadIdx.x; i < m * (kDC / 8); i += kBlock) {
-      const int mi = i / (kDC / 8);
-      const int d8 = i % (kDC / 8);
-      *reinterpret_cast<short8*>(x_lds + mi * kDC + d8 * 8) =
-          *reinterpret_cast<const short8*>(x + ((int64_t)(r0 + mi)) * D + dc + d8 * 8);
+    // stage X chunk [m][kDC2] (coalesced: consecutive threads, consecutive
+    // 8-byte spans)
+    for (int i = threadIdx.x; i < m * (kDC2 / 4); i += kBlock) {
+      const int mi = i / (kDC2 / 4);
+      const int d4 = i % (kDC2 / 4);
+      *reinterpret_cast<short4_t*>(x_lds + mi * kDC2 + d4 * 4) =
+          *reinterpret_cast<const short4_t*>(
+              x + ((int64_t)(r0 + mi)) * D + dc + d4 * 4);
     }
     __syncthreads();
     if (n < N) {
-      // stream this lane's weight chunk once; dot against every token
-      short8 wv[kDC / 8];
-#pragma unroll
-      for (int i = 0; i < kDC / 8; i++)
-        wv[i] = *reinterpret_cast<const short8*>(wrow + dc + i * 8);
+      // lanes split the chunk: lane reads elems [lane*4, lane*4+4)
+      short4_t wv = *reinterpret_cast<const short4_t*>(wrow + dc + lane * 4);
+      const bf16x2_t* w2 = reinterpret_cast<const bf16x2_t*>(&wv);
       for (int mi = 0; mi < m; mi++) {
-        float d = acc[mi];
-#pragma unroll
-        for (int i = 0; i < kDC / 8; i++) {
-          const bf16x2_t* w2 = reinterpret_cast<const bf16x2_t*>(&wv[i]);
-          const bf16x2_t* x2 = reinterpret_cast<const bf16x2_t*>(
-              x_lds + mi * kDC + i * 8);
-#pragma unroll
-          for (int p = 0; p < 4; p++)
-            d = __builtin_amdgcn_fdot2_f32_bf16(w2[p], x2[p], d, false);
-        }
-        acc[mi] = d;
+        const bf16x2_t* x2 = reinterpret_cast<const bf16x2_t*>(
+            x_lds + mi * kDC2 + lane * 4);
+        float d = __builtin_amdgcn_fdot2_f32_bf16(w2[0], x2[0], 0.f, false);
+        d = __builtin_amdgcn_fdot2_f32_bf16(w2[1], x2[1], d, false);
+        acc[mi] += d;
       }
     }
   }
   if (n < N) {
-    for (int mi = 0; mi < m; mi++)
-      y[((int64_t)(r0 + mi)) * N + n] = f32_to_bf16(acc[mi]);
+    // fold lane partials per token
+    for (int mi = 0; mi < m; mi++) {
+      float a = wave_reduce_sum(acc[mi]);
+      if (lane == 0) y[((int64_t)(r0 + mi)) * N + n] = f32_to_bf16(a);
+    }
   }
 }
 
@@ -148,11 +156,11 @@ void moe_grouped_gemm(torch::Tensor y, torch::Tensor x, torch::Tensor w,
   const int D = x.size(1);
   const int N = w.size(1);
   TORCH_CHECK(w.size(2) == D && y.size(1) == N);
-  TORCH_CHECK(D % kDC == 0, "hidden size must be a multiple of 128");
+  TORCH_CHECK(D % kDC2 == 0, "in-features must be a multiple of 256");
   const int ntiles = tiles.size(0);
   if (ntiles == 0) return;
   auto stream = at::cuda::getCurrentHIPStream();
-  dim3 grid(ntiles, (N + kBlock - 1) / kBlock);
+  dim3 grid(ntiles, (N + 3) / 4);
   moe_gemm_kernel<<<grid, kBlock, 0, stream>>>(
       (short*)y.data_ptr(), (const short*)x.data_ptr(),
       (const short*)w.data_ptr(), tiles.data_ptr<int32_t>(), nullptr,
@@ -169,11 +177,11 @@ void moe_grouped_gemm_seg(torch::Tensor y, torch::Tensor x, torch::Tensor w,
   const int N = w.size(1);
   const int E = w.size(0);
   TORCH_CHECK(seg_start.numel() == E + 1);
-  TORCH_CHECK(D % kDC == 0, "hidden size must be a multiple of 128");
+  TORCH_CHECK(D % kDC2 == 0, "in-features must be a multiple of 256");
   const int max_mt = (int)((max_tokens + kMaxM - 1) / kMaxM);
   if (max_mt == 0) return;
   auto stream = at::cuda::getCurrentHIPStream();
-  dim3 grid(E * max_mt, (N + kBlock - 1) / kBlock);
+  dim3 grid(E * max_mt, (N + 3) / 4);
   moe_gemm_kernel<<<grid, kBlock, 0, stream>>>(
       (short*)y.data_ptr(), (const short*)x.data_ptr(),
       (const short*)w.data_ptr(), nullptr, seg_start.data_ptr<int32_t>(),
