@@ -64,45 +64,33 @@ __global__ __launch_bounds__(kBlock) void moe_gemm_kernel(
   const int lane = threadIdx.x & 63;
   const int wid = threadIdx.x >> 6;
   const int n = blockIdx.y * 4 + wid;   // this WAVE's W row
+  if (n >= N) return;
   const short* wrow = w + ((int64_t)e * N + n) * D;
 
-  __shared__ short x_lds[kMaxM * kDC2];
-
+  // No LDS, no barriers: X is tiny (<=16 rows) and L2-resident after the
+  // first y-blocks touch it; both W and X chunk loads are coalesced
+  // (lane covers 4 consecutive elements), and the m-loop issues m+1
+  // independent loads per chunk to keep HBM latency covered.
   float acc[kMaxM];
 #pragma unroll
   for (int i = 0; i < kMaxM; i++) acc[i] = 0.f;
 
   for (int dc = 0; dc < D; dc += kDC2) {
-    __syncthreads();
-    // stage X chunk [m][kDC2] (coalesced: consecutive threads, consecutive
-    // 8-byte spans)
-    for (int i = threadIdx.x; i < m * (kDC2 / 4); i += kBlock) {
-      const int mi = i / (kDC2 / 4);
-      const int d4 = i % (kDC2 / 4);
-      *reinterpret_cast<short4_t*>(x_lds + mi * kDC2 + d4 * 4) =
-          *reinterpret_cast<const short4_t*>(
-              x + ((int64_t)(r0 + mi)) * D + dc + d4 * 4);
-    }
-    __syncthreads();
-    if (n < N) {
-      // lanes split the chunk: lane reads elems [lane*4, lane*4+4)
-      short4_t wv = *reinterpret_cast<const short4_t*>(wrow + dc + lane * 4);
-      const bf16x2_t* w2 = reinterpret_cast<const bf16x2_t*>(&wv);
-      for (int mi = 0; mi < m; mi++) {
-        const bf16x2_t* x2 = reinterpret_cast<const bf16x2_t*>(
-            x_lds + mi * kDC2 + lane * 4);
-        float d = __builtin_amdgcn_fdot2_f32_bf16(w2[0], x2[0], 0.f, false);
-        d = __builtin_amdgcn_fdot2_f32_bf16(w2[1], x2[1], d, false);
-        acc[mi] += d;
-      }
+    short4_t wv = *reinterpret_cast<const short4_t*>(wrow + dc + lane * 4);
+    const bf16x2_t* w2 = reinterpret_cast<const bf16x2_t*>(&wv);
+    for (int mi = 0; mi < m; mi++) {
+      short4_t xv = *reinterpret_cast<const short4_t*>(
+          x + ((int64_t)(r0 + mi)) * D + dc + lane * 4);
+      const bf16x2_t* x2 = reinterpret_cast<const bf16x2_t*>(&xv);
+      float d = __builtin_amdgcn_fdot2_f32_bf16(w2[0], x2[0], 0.f, false);
+      d = __builtin_amdgcn_fdot2_f32_bf16(w2[1], x2[1], d, false);
+      acc[mi] += d;
     }
   }
-  if (n < N) {
-    // fold lane partials per token
-    for (int mi = 0; mi < m; mi++) {
-      float a = wave_reduce_sum(acc[mi]);
-      if (lane == 0) y[((int64_t)(r0 + mi)) * N + n] = f32_to_bf16(a);
-    }
+  // fold lane partials per token
+  for (int mi = 0; mi < m; mi++) {
+    float a = wave_reduce_sum(acc[mi]);
+    if (lane == 0) y[((int64_t)(r0 + mi)) * N + n] = f32_to_bf16(a);
   }
 }
 
