@@ -27,13 +27,6 @@ constexpr int kDC = 128;    // D chunk staged per iteration
 // routing counts) so the launch is sync-free and hipGraph-capturable: the
 // grid covers worst-case m-tiles per expert and blocks with no tokens
 // exit. tiles mode (tiles != nullptr) is kept for host-built tile lists.
-//
-// Layout: WAVE-per-W-row (4 rows per block) with lanes splitting D —
-// the wave's 64 lanes read 4 consecutive elements each (256-elem chunks),
-// fully coalesced weight streaming; X chunks are staged in LDS and read
-// broadcast. A lane-per-row variant measured ~2.0 TB/s effective
-// (uncoalesced stride-D streams); this layout targets the HBM roofline.
-constexpr int kDC2 = 256;  // D chunk per pass (lane covers 4 elems)
 
 __global__ __launch_bounds__(kBlock) void moe_gemm_kernel(
     short* __restrict__ y,        // [T, N] bf16 (gathered order)
