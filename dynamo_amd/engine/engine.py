@@ -26,6 +26,7 @@ class StepOutput:
     finished: bool
     finish_reason: Optional[str] = None
     num_output_tokens: int = 0
+    embedding: Optional[List[float]] = None
 
 
 @dataclass
@@ -251,6 +252,18 @@ class LLMEngine:
                     self._finish(req, reason)
             outputs.append(StepOutput(req.req_id, int(tok), finished, reason,
                                       len(req.output_tokens)))
+
+        # embedding requests: finish when the (chunked) prefill completes
+        for ss in sched.prefills:
+            r = ss.req
+            if (r.sampling.embed and r.num_computed >= r.total_len
+                    and r.state == ReqState.RUNNING):
+                pooled = (r._embed_sum / float(r.total_len)).cpu().tolist()
+                r.embedding = pooled
+                r._embed_sum = None
+                self._finish(r, "embed")
+                outputs.append(StepOutput(r.req_id, None, True, "embed", 0,
+                                          embedding=pooled))
 
         # prefix-cache hash registration + KV events
         if self.cfg.kv_events or self.cfg.enable_prefix_caching:

@@ -100,7 +100,7 @@ class ModelRunner:
             q_start.append(qpos)
             q_len.append(n)
             ctx_len.append(nc + n)
-            if ss.sample:
+            if ss.sample and not r.sampling.embed:
                 logits_rows.append(nd + qpos + n - 1)
             qpos += n
 
@@ -142,8 +142,20 @@ class ModelRunner:
         """Returns (sampled_tokens int32 [n_sample] on device, sample_reqs)."""
         input_ids, meta = self.prepare(sched)
         hidden = self.model.forward(input_ids, self.kv_pool, meta)
+        # embedding requests: accumulate fp32 hidden-state sums per chunk
+        # (mean pool finalized by the engine when the prefill completes)
+        qpos = 0
+        for ss in sched.prefills:
+            r = ss.req
+            if r.sampling.embed:
+                seg = hidden[meta.num_decode + qpos:
+                             meta.num_decode + qpos + ss.n_new].float().sum(0)
+                r._embed_sum = (seg if r._embed_sum is None
+                                else r._embed_sum + seg)
+            qpos += ss.n_new
         sample_seqs = [s for s in sched.decodes if s.sample] + \
-                      [s for s in sched.prefills if s.sample]
+                      [s for s in sched.prefills
+                       if s.sample and not s.req.sampling.embed]
         if not sample_seqs:
             return torch.empty(0, dtype=torch.int32), []
         rows = hidden[meta.logits_rows]

@@ -27,6 +27,10 @@ class SamplingParams:
     stop_token_ids: List[int] = field(default_factory=list)
     ignore_eos: bool = False
     seed: int = 0
+    # embedding request: prefill-only, mean-pooled hidden state instead of
+    # sampled tokens (reference parity: /v1/embeddings route,
+    # lib/llm/src/http service embeddings handler)
+    embed: bool = False
 
 
 class ReqState(Enum):
@@ -54,6 +58,9 @@ class Request:
         self.prefill_result: Optional[dict] = None
         # disaggregation: prefill-side — keep KV pages alive after finish
         self.hold_kv = False
+        # embedding request accumulators (mean pool over prompt tokens)
+        self.embedding: Optional[List[float]] = None
+        self._embed_sum = None
 
     @property
     def all_tokens(self) -> List[int]:

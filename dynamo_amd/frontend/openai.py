@@ -46,6 +46,12 @@ class ChatMessage(BaseModel):
     content: str
 
 
+class EmbeddingRequest(BaseModel):
+    model: str = ""
+    input: Union[str, List[str], List[int], List[List[int]]] = ""
+    encoding_format: str = "float"
+
+
 class ChatRequest(BaseModel):
     model: str = ""
     messages: List[ChatMessage] = Field(default_factory=list)
@@ -149,6 +155,39 @@ def build_app(manager: ModelManager) -> FastAPI:
                       "completion_tokens": len(produced),
                       "total_tokens": len(token_ids) + len(produced)},
         }
+
+    @app.post("/v1/embeddings")
+    async def embeddings(req: EmbeddingRequest):
+        """Mean-pooled last-hidden-state embeddings (reference parity:
+        lib/llm/src/http embeddings route)."""
+        entry = _entry_or_404(req.model)
+        REQS.labels(entry.name, "embeddings").inc()
+        raw_inputs = req.input
+        if isinstance(raw_inputs, str):
+            raw_inputs = [raw_inputs]
+        elif raw_inputs and isinstance(raw_inputs[0], int):
+            raw_inputs = [raw_inputs]      # single pre-tokenized prompt
+        data, total_tokens = [], 0
+        for idx, item in enumerate(raw_inputs):
+            token_ids = (list(item) if isinstance(item, list)
+                         else entry.tokenizer.encode(item))
+            total_tokens += len(token_ids)
+            rid = f"embd-{uuid.uuid4().hex[:24]}"
+            sampling = {"embed": True}
+            stop = {"max_tokens": 1, "ignore_eos": True}
+            vec = None
+            async for chunk in manager.generate_tokens(entry, token_ids,
+                                                       sampling, stop,
+                                                       request_id=rid):
+                if chunk.get("embedding") is not None:
+                    vec = chunk["embedding"]
+            if vec is None:
+                raise HTTPException(500, "worker returned no embedding")
+            data.append({"object": "embedding", "index": idx,
+                         "embedding": vec})
+        return {"object": "list", "model": entry.name, "data": data,
+                "usage": {"prompt_tokens": total_tokens,
+                          "total_tokens": total_tokens}}
 
     @app.post("/v1/chat/completions")
     async def chat(req: ChatRequest, raw: Request):

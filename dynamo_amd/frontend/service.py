@@ -162,8 +162,10 @@ class ModelManager:
                 "annotations": {"trace_id": rid},
             }
             try:
-                if entry.prefill_router is not None and \
-                        entry.prefill_router.has_prefill_pool():
+                # embedding requests are prefill-only: no P/D split
+                if (not sampling.get("embed")
+                        and entry.prefill_router is not None
+                        and entry.prefill_router.has_prefill_pool()):
                     gen = entry.prefill_router.generate(payload)
                 else:
                     iid = entry.router.select(payload["token_ids"])

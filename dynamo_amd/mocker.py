@@ -53,8 +53,15 @@ class MockRunner:
             delay += pf_tokens / self.prefill_tps
         if delay:
             time.sleep(delay)
+        for s in sched.prefills:
+            if s.req.sampling.embed:   # deterministic fake pooled hidden
+                seg = torch.full((self.cfg.model.hidden_size,),
+                                 float(s.n_new), dtype=torch.float32)
+                s.req._embed_sum = (seg if s.req._embed_sum is None
+                                    else s.req._embed_sum + seg)
         sample_seqs = [s for s in sched.decodes if s.sample] + \
-                      [s for s in sched.prefills if s.sample]
+                      [s for s in sched.prefills
+                       if s.sample and not s.req.sampling.embed]
         toks = [_token_for(s.req.req_id, s.req.total_len,
                            self.cfg.model.vocab_size) for s in sample_seqs]
         return (torch.tensor(toks, dtype=torch.int32),

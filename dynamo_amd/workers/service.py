@@ -44,6 +44,7 @@ def make_sampling(payload: dict) -> SamplingParams:
         stop_token_ids=list(sc.get("stop_token_ids", [])),
         ignore_eos=bool(sc.get("ignore_eos", False)),
         seed=int(so.get("seed", 0)),
+        embed=bool(so.get("embed", False)),
     )
 
 
@@ -176,7 +177,11 @@ class WorkerService:
                         return
                     continue
                 so = get.result()
-                chunk: dict = {"token_ids": [so.new_token]}
+                chunk: dict = {"token_ids": ([so.new_token]
+                                             if so.new_token is not None
+                                             else [])}
+                if so.embedding is not None:
+                    chunk["embedding"] = so.embedding
                 if so.finished:
                     chunk["finish_reason"] = so.finish_reason
                     if is_prefill_role:

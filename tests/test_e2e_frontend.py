@@ -242,3 +242,58 @@ def test_pause_resume_lifecycle():
         assert len(toks) == 40
         await teardown(services, mgr, client)
     run(main())
+
+
+def test_embeddings_route():
+    async def main():
+        shared, services, mgr, client = await with_stack(nworkers=1)
+        r = await client.post("/v1/embeddings", json={
+            "model": "mock-model", "input": ["hello world", "second doc"]})
+        assert r.status_code == 200, r.text
+        body = r.json()
+        assert body["object"] == "list" and len(body["data"]) == 2
+        vec = body["data"][0]["embedding"]
+        # mock runner pools to exactly 1.0 per dim (sum(n_new)/total_len)
+        assert len(vec) == MODEL.hidden_size
+        assert all(abs(v - 1.0) < 1e-6 for v in vec)
+        assert body["data"][1]["index"] == 1
+        assert body["usage"]["prompt_tokens"] > 0
+        await teardown(services, mgr, client)
+    run(main())
+
+
+def test_multi_model_frontend():
+    """Two model families served behind ONE frontend (reference parity:
+    ModelWatcher adds every discovered model card)."""
+    async def main():
+        shared = MemoryDiscovery()
+        model_b = ModelConfig(name="mock-model-b", vocab_size=256,
+                              hidden_size=64)
+        services = [await start_worker(shared)]
+        rt_b = DistributedRuntime(shared)
+        eng_b = make_mock_engine(model=model_b)
+        ws_b = WorkerService(eng_b, rt_b, component="backend-b")
+        await ws_b.start()
+        services.append((ws_b, rt_b))
+        mgr_rt = DistributedRuntime(shared)
+        mgr = ModelManager(mgr_rt)
+        await mgr.start(watch_interval=0.2)
+        app = build_app(mgr)
+        transport = httpx.ASGITransport(app=app)
+        client = httpx.AsyncClient(transport=transport, base_url="http://t")
+
+        r = await client.get("/v1/models")
+        names = sorted(m["id"] for m in r.json()["data"])
+        assert names == ["mock-model", "mock-model-b"]
+        for name in names:
+            r = await client.post("/v1/completions", json={
+                "model": name, "prompt": "hi there", "max_tokens": 4})
+            assert r.status_code == 200, r.text
+            assert r.json()["model"] == name
+            assert r.json()["usage"]["completion_tokens"] == 4
+        # unknown model -> 404 (with >1 model there is no fallback)
+        r = await client.post("/v1/completions", json={
+            "model": "nope", "prompt": "x", "max_tokens": 1})
+        assert r.status_code == 404
+        await teardown(services, mgr, client)
+    run(main())

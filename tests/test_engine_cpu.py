@@ -155,3 +155,53 @@ def test_metrics_populated():
     m = e.last_metrics
     assert m.num_tokens_step > 0
     assert 0 <= m.kv_usage <= 1
+
+
+def test_embedding_request():
+    """Embed requests finish prefill-only with a mean-pooled hidden state,
+    and chunked prefill produces the same pooled vector."""
+    import dataclasses
+    prompt = list(range(50, 120))  # 70 tokens
+
+    def embed_with(batched):
+        cfg = EngineConfig(model=PRESETS["tiny-llama"], device="cpu",
+                           dtype="float32", max_num_seqs=8,
+                           max_batched_tokens=batched, max_model_len=512,
+                           kv_pool_pages=64, page_size=16)
+        eng = LLMEngine(cfg, seed=7)
+        eng.add_request("e0", prompt, SamplingParams(embed=True, max_tokens=1))
+        vec = None
+        steps = 0
+        while eng.has_work():
+            for so in eng.step():
+                if so.finish_reason == "embed":
+                    vec = so.embedding
+            steps += 1
+            assert steps < 100
+        assert vec is not None and len(vec) == 256
+        return vec
+
+    v_full = embed_with(128)
+    v_chunk = embed_with(32)   # forces 3 prefill chunks
+    assert v_full == pytest.approx(v_chunk, rel=1e-4, abs=1e-5)
+
+
+def test_embedding_and_generation_batched():
+    """An embed request and a normal generation coexist in one batch and the
+    generation output is unchanged by the presence of the embed request."""
+    base = generate(make_engine(), [list(range(40, 80))], max_tokens=6)[0]
+    eng = make_engine()
+    eng.add_request("g0", list(range(40, 80)), SamplingParams(max_tokens=6))
+    eng.add_request("e0", list(range(10, 60)), SamplingParams(embed=True))
+    toks, vec = [], None
+    steps = 0
+    while eng.has_work():
+        for so in eng.step():
+            if so.req_id == "g0" and so.new_token is not None:
+                toks.append(so.new_token)
+            if so.finish_reason == "embed":
+                vec = so.embedding
+        steps += 1
+        assert steps < 100
+    assert toks == base
+    assert vec is not None and len(vec) == 256
