@@ -107,6 +107,8 @@ class EngineConfig:
     enable_prefix_caching: bool = True
     enable_hip_graphs: bool = True
     host_cache_pages: int = 0           # KVBM G2 tier size (0 = disabled)
+    disk_cache_pages: int = 0           # KVBM G3 tier size (0 = disabled)
+    disk_cache_path: str = ""           # G3 backing file (required if G3 on)
     kv_events: bool = True              # emit stored/removed block events
     block_salt: int = 0
     # disaggregation

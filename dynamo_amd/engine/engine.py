@@ -58,7 +58,9 @@ class LLMEngine:
         if cfg.host_cache_pages > 0 and self.runner.kv_pool is not None:
             from dynamo_amd.kvbm.host_tier import HostKVTier
             self.host_tier = HostKVTier(self.runner.kv_pool,
-                                        cfg.host_cache_pages)
+                                        cfg.host_cache_pages,
+                                        disk_path=cfg.disk_cache_path,
+                                        num_disk_pages=cfg.disk_cache_pages)
             self.alloc.host_tier = self.host_tier
         self.scheduler = Scheduler(cfg, self.alloc)
         self.requests: Dict[str, Request] = {}

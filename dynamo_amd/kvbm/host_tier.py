@@ -24,7 +24,8 @@ log = logging.getLogger("dynamo_amd.kvbm")
 
 
 class HostKVTier:
-    def __init__(self, kv_pool, num_host_pages: int):
+    def __init__(self, kv_pool, num_host_pages: int,
+                 disk_path: str = "", num_disk_pages: int = 0):
         self.pool = kv_pool
         self.device = kv_pool.device
         L, two, P, hkv, ps, hd = kv_pool.shape
@@ -46,15 +47,29 @@ class HostKVTier:
                                        device=self.device)
             self._fence_event = torch.cuda.Event()
         self.stats = {"offloaded": 0, "onboarded": 0, "evicted_host": 0,
-                      "hits": 0}
+                      "hits": 0, "spilled_disk": 0, "onboarded_disk": 0}
         self.events = []  # (kind, hash) host-tier events
+        # G3: disk tier under G2 — host LRU evictions spill there
+        self.disk = None
+        if num_disk_pages > 0 and disk_path:
+            from .disk_tier import DiskKVTier
+            elem_bytes = torch.empty(0, dtype=kv_pool.dtype).element_size()
+            self.disk = DiskKVTier(disk_path, num_disk_pages,
+                                   self.page_elems * elem_bytes)
 
     def _plane_ids(self, pid: int) -> torch.Tensor:
         ids = [k * self.P + pid for k in range(self.planes)]
         return torch.tensor(ids, dtype=torch.int32, device=self.device)
 
     def contains(self, h: int) -> bool:
-        return h in self.map
+        return h in self.map or (self.disk is not None
+                                 and self.disk.contains(h))
+
+    def _page_bytes(self, hp: int) -> bytes:
+        if self.device.type == "cuda":
+            # the page may have an in-flight D2H copy on our stream
+            self.stream.synchronize()
+        return self.host[hp].view(torch.uint8).numpy().tobytes()
 
     def _alloc_host(self) -> Optional[int]:
         if self.free:
@@ -62,9 +77,28 @@ class HostKVTier:
         if self.map:
             old_h, hp = self.map.popitem(last=False)  # LRU
             self.stats["evicted_host"] += 1
-            self.events.append(("removed_host", old_h))
+            if self.disk is not None and self.disk.put(old_h,
+                                                       self._page_bytes(hp)):
+                self.stats["spilled_disk"] += 1
+            else:
+                self.events.append(("removed_host", old_h))
             return hp
         return None
+
+    def _onboard_from_disk(self, h: int) -> Optional[int]:
+        """Promote a disk page back into a host slot; returns host page."""
+        data = self.disk.get(h) if self.disk is not None else None
+        if data is None:
+            return None
+        hp = self._alloc_host()
+        if hp is None:
+            return None
+        self.host[hp].view(torch.uint8).copy_(
+            torch.frombuffer(bytearray(data), dtype=torch.uint8))
+        self.map[h] = hp
+        self.map.move_to_end(h)
+        self.stats["onboarded_disk"] += 1
+        return hp
 
     # -- device -> host (called from PageAllocator eviction hook) --------
     def offload(self, pid: int, h: int):
@@ -91,6 +125,8 @@ class HostKVTier:
     # -- host -> device (prefix-cache onboard) ---------------------------
     def onboard(self, h: int, pid: int) -> bool:
         hp = self.map.get(h)
+        if hp is None:
+            hp = self._onboard_from_disk(h)   # G3 -> G2 promote
         if hp is None:
             return False
         if self.device.type == "cuda":

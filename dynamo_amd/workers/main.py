@@ -10,7 +10,9 @@ from __future__ import annotations
 import argparse
 import asyncio
 import logging
+import os
 import signal
+import tempfile
 
 from dynamo_amd.engine import EngineConfig, LLMEngine
 from dynamo_amd.models.registry import resolve_model_config
@@ -40,6 +42,12 @@ def build_parser():
     p.add_argument("--no-hip-graphs", action="store_true")
     p.add_argument("--seed", type=int, default=0)
     p.add_argument("--host", default="127.0.0.1")
+    p.add_argument("--host-cache-pages", type=int, default=0,
+                   help="KVBM G2 pinned-host tier size in pages")
+    p.add_argument("--disk-cache-pages", type=int, default=0,
+                   help="KVBM G3 disk tier size in pages")
+    p.add_argument("--disk-cache-path", default="",
+                   help="G3 backing file (default <tmp>/dynamo_kv_g3.bin)")
     p.add_argument("--gms", action="store_true",
                    help="import weights zero-copy from a GMS weight server")
     return p
@@ -81,7 +89,13 @@ def make_engine_from_args(args) -> LLMEngine:
         gpu_mem_fraction=args.gpu_mem_fraction,
         enable_prefix_caching=not args.no_prefix_caching,
         enable_hip_graphs=not args.no_hip_graphs,
-        worker_type=args.worker_type)
+        worker_type=args.worker_type,
+        host_cache_pages=args.host_cache_pages,
+        disk_cache_pages=args.disk_cache_pages,
+        disk_cache_path=(args.disk_cache_path or
+                         (os.path.join(tempfile.gettempdir(),
+                                       f"dynamo_kv_g3_{os.getpid()}.bin")
+                          if args.disk_cache_pages else "")))
     return LLMEngine(cfg, seed=args.seed, weight_pool=weight_pool)
 
 

@@ -55,3 +55,38 @@ def test_host_events_emitted():
         generate(eng, f"c{i}", [(300 + 80 * i + j) % 500 for j in range(80)])
     kinds = {e.kind for e in eng.drain_kv_events()}
     assert "stored_host" in kinds
+
+
+def test_disk_tier_spill_and_promote(tmp_path):
+    """G3: host LRU evictions spill to disk; disk pages promote back through
+    G2 on a prefix hit with bit-identical outputs."""
+    cfg = EngineConfig(model=PRESETS["tiny-llama"], device="cpu",
+                       max_num_seqs=4, max_batched_tokens=512,
+                       max_model_len=512, kv_pool_pages=24, page_size=16,
+                       host_cache_pages=4, disk_cache_pages=64,
+                       disk_cache_path=str(tmp_path / "g3.bin"))
+    eng = LLMEngine(cfg, seed=7)
+    p1 = list(range(64))
+    o1 = generate(eng, "a", p1)
+    # churn past BOTH the device pool (24 pages) and host pool (4 pages)
+    for i in range(8):
+        generate(eng, f"c{i}", [(100 + 70 * i + j) % 500 for j in range(64)])
+    st = eng.host_tier.stats
+    assert st["spilled_disk"] > 0, "nothing spilled to G3"
+    o2 = generate(eng, "a2", p1)
+    assert eng.host_tier.stats["onboarded_disk"] > 0, "no G3 promote"
+    assert o2 == o1, "outputs diverged after disk spill/promote cycle"
+
+
+def test_disk_tier_lru_and_capacity(tmp_path):
+    from dynamo_amd.kvbm.disk_tier import DiskKVTier
+    t = DiskKVTier(str(tmp_path / "d.bin"), num_pages=2, page_bytes=8)
+    assert t.put(1, b"a" * 8) and t.put(2, b"b" * 8)
+    assert t.get(1) == b"a" * 8          # touch 1 -> 2 becomes LRU
+    assert t.put(3, b"c" * 8)            # evicts 2
+    assert t.get(2) is None and t.stats["evicted"] == 1
+    assert t.get(1) == b"a" * 8 and t.get(3) == b"c" * 8
+    t.remove(1)
+    assert t.get(1) is None
+    assert t.put(4, b"d" * 8)            # reuses freed slot
+    t.close()
