@@ -20,11 +20,15 @@ def test_multiprocess_workers_and_frontend(tmp_path):
                 worker_cmd(mock=True, model="tiny-llama", discovery=disc,
                            page_size=16),
                 ready_marker="WORKER_READY").start())
+        import socket
+        with socket.socket() as s:          # pick a free port (avoids
+            s.bind(("127.0.0.1", 0))        # collisions across runs)
+            port = s.getsockname()[1]
         front = ManagedProcess(
             [sys.executable, "-m", "dynamo_amd.frontend", "--discovery", disc,
-             "--port", "18231"],
+             "--port", str(port)],
             ready_marker="FRONTEND_READY").start()
-        base = "http://127.0.0.1:18231"
+        base = f"http://127.0.0.1:{port}"
         # wait for model registration to reach the frontend
         deadline = time.time() + 60
         with httpx.Client(timeout=30) as client:
