@@ -36,6 +36,9 @@ void attention_prefill_paged(torch::Tensor out, torch::Tensor q,
                              double scale);
 // sampling.hip
 void greedy_sample(torch::Tensor out, torch::Tensor logits);
+void topkp_sample(torch::Tensor out, torch::Tensor logits,
+                  torch::Tensor inv_temp, torch::Tensor top_k,
+                  torch::Tensor top_p, int64_t seed);
 void gumbel_sample(torch::Tensor out, torch::Tensor logits, torch::Tensor inv_temp,
                    int64_t seed);
 // moe.hip
@@ -71,6 +74,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("attention_prefill_paged", &attention_prefill_paged);
   m.def("greedy_sample", &greedy_sample);
   m.def("gumbel_sample", &gumbel_sample);
+  m.def("topkp_sample", &topkp_sample);
   m.def("moe_grouped_gemm", &moe_grouped_gemm);
   m.def("moe_grouped_gemm_seg", &moe_grouped_gemm_seg);
   m.def("topk_gating", &topk_gating);

@@ -66,6 +66,130 @@ __global__ void sample_kernel(int32_t* __restrict__ out,          // [B]
   }
 }
 
+// Fused top-k / top-p (nucleus) sampling: one block per row, no sort.
+//
+// The k-th-largest-prob threshold (top-k) and the nucleus mass threshold
+// (top-p) are found by a FUSED bisection on the probability scale: each
+// iteration makes one pass over the vocab computing count(p_i >= t_k) and
+// mass(p_i >= t_p) for the two candidate thresholds, 24 iterations
+// (resolution 6e-8 on [0,1] — tokens below that probability are sampling
+// noise). The final pass Gumbel-argmaxes logits/T over the surviving set,
+// which samples the renormalized filtered distribution exactly. Nucleus
+// semantics match the torch reference in engine/sampling.py: the keep-set
+// is defined on the T=1 softmax; temperature only shapes sampling inside
+// the set. Single launch, no host sync, hipGraph-capturable.
+__global__ __launch_bounds__(kBlock) void topkp_sample_kernel(
+    int32_t* __restrict__ out,          // [B]
+    const float* __restrict__ logits,   // [B, V]
+    const float* __restrict__ inv_temp, // [B]
+    const int32_t* __restrict__ top_k,  // [B] (<=0: off)
+    const float* __restrict__ top_p,    // [B] (>=1: off)
+    uint64_t seed, int V) {
+  const int b = blockIdx.x;
+  const float* row = logits + (int64_t)b * V;
+  const int k = top_k[b];
+  const float p = top_p[b];
+  const bool use_k = (k > 0 && k < V);
+  const bool use_p = (p < 1.0f);
+  constexpr int kWaves = kBlock / WAVE_SIZE;
+  __shared__ float red[kWaves];
+  __shared__ float red2[kWaves];
+  __shared__ float bcast[2];
+
+  // pass 1: row max (softmax stability)
+  float m = -1e38f;
+  for (int v = threadIdx.x; v < V; v += kBlock) m = fmaxf(m, row[v]);
+  m = wave_reduce_max(m);
+  if ((threadIdx.x & 63) == 0) red[threadIdx.x / WAVE_SIZE] = m;
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    for (int w = 1; w < kWaves; w++) m = fmaxf(m, red[w]);
+    bcast[0] = m;
+  }
+  __syncthreads();
+  m = bcast[0];
+
+  // pass 2: partition sum Z
+  float z = 0.f;
+  for (int v = threadIdx.x; v < V; v += kBlock) z += __expf(row[v] - m);
+  z = wave_reduce_sum(z);
+  if ((threadIdx.x & 63) == 0) red[threadIdx.x / WAVE_SIZE] = z;
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    for (int w = 1; w < kWaves; w++) z += red[w];
+    bcast[0] = z;
+  }
+  __syncthreads();
+  const float inv_z = 1.f / bcast[0];
+
+  // fused bisection: t_k (count >= k) and t_p (mass >= p), both on [0,1]
+  float lo_k = 0.f, hi_k = 1.f, lo_p = 0.f, hi_p = 1.f;
+  if (use_k || use_p) {
+    for (int it = 0; it < 24; it++) {
+      const float mid_k = 0.5f * (lo_k + hi_k);
+      const float mid_p = 0.5f * (lo_p + hi_p);
+      float cnt = 0.f, mass = 0.f;
+      for (int v = threadIdx.x; v < V; v += kBlock) {
+        const float pv = __expf(row[v] - m) * inv_z;
+        if (use_k && pv >= mid_k) cnt += 1.f;
+        if (use_p && pv >= mid_p) mass += pv;
+      }
+      cnt = wave_reduce_sum(cnt);
+      mass = wave_reduce_sum(mass);
+      const int wid = threadIdx.x / WAVE_SIZE;
+      if ((threadIdx.x & 63) == 0) { red[wid] = cnt; red2[wid] = mass; }
+      __syncthreads();
+      if (threadIdx.x == 0) {
+        for (int w = 1; w < kWaves; w++) { cnt += red[w]; mass += red2[w]; }
+        bcast[0] = cnt;
+        bcast[1] = mass;
+      }
+      __syncthreads();
+      cnt = bcast[0];
+      mass = bcast[1];
+      if (use_k) { if (cnt >= (float)k) lo_k = mid_k; else hi_k = mid_k; }
+      if (use_p) { if (mass >= p) lo_p = mid_p; else hi_p = mid_p; }
+      __syncthreads();
+    }
+  }
+  const float thr = fmaxf(use_k ? lo_k : 0.f, use_p ? lo_p : 0.f);
+
+  // final pass: Gumbel-argmax of logits/T over {p_i >= thr}
+  const float it_ = inv_temp[b];
+  float best = -1e38f;
+  int besti = -1;
+  for (int v = threadIdx.x; v < V; v += kBlock) {
+    const float pv = __expf(row[v] - m) * inv_z;
+    if (pv < thr) continue;
+    const uint64_t h = hash_u64(seed ^ ((uint64_t)b << 32) ^ (uint64_t)v);
+    const float u = (float)((h >> 11) + 1) * 4.8828125e-4f * 2.2737367544323206e-13f;
+    const float x = row[v] * it_ + -__logf(-__logf(u));
+    if (x > best || (x == best && v < besti)) { best = x; besti = v; }
+  }
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) {
+    const float ob = __shfl_xor(best, off, WAVE_SIZE);
+    const int oi = __shfl_xor(besti, off, WAVE_SIZE);
+    if (ob > best || (ob == best && oi != -1 && (besti == -1 || oi < besti))) {
+      best = ob; besti = oi;
+    }
+  }
+  __shared__ float sb2[kWaves];
+  __shared__ int si2[kWaves];
+  const int wid = threadIdx.x / WAVE_SIZE;
+  if ((threadIdx.x & 63) == 0) { sb2[wid] = best; si2[wid] = besti; }
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    for (int w = 1; w < kWaves; w++) {
+      if (sb2[w] > best || (sb2[w] == best && si2[w] != -1 &&
+                            (besti == -1 || si2[w] < besti))) {
+        best = sb2[w]; besti = si2[w];
+      }
+    }
+    out[b] = besti;
+  }
+}
+
 }  // namespace
 
 void greedy_sample(torch::Tensor out, torch::Tensor logits) {
@@ -91,5 +215,24 @@ void gumbel_sample(torch::Tensor out, torch::Tensor logits, torch::Tensor inv_te
  hipLaunchKernelGGL(( sample_kernel<true>), dim3(B), dim3(kBlock), 0, stream, 
       out.data_ptr<int32_t>(), logits.data_ptr<float>(),
       inv_temp.data_ptr<float>(), (uint64_t)seed, V);
+  HIP_CHECK_KERNEL();
+}
+
+void topkp_sample(torch::Tensor out, torch::Tensor logits,
+                  torch::Tensor inv_temp, torch::Tensor top_k,
+                  torch::Tensor top_p, int64_t seed) {
+  TORCH_CHECK(logits.is_cuda() && logits.dtype() == torch::kFloat32);
+  TORCH_CHECK(inv_temp.dtype() == torch::kFloat32);
+  TORCH_CHECK(top_k.dtype() == torch::kInt32);
+  TORCH_CHECK(top_p.dtype() == torch::kFloat32);
+  const int B = logits.size(0);
+  const int V = logits.size(1);
+  TORCH_CHECK(top_k.numel() == B && top_p.numel() == B && inv_temp.numel() == B);
+  if (B == 0) return;
+  auto stream = at::cuda::getCurrentHIPStream();
+ hipLaunchKernelGGL(( topkp_sample_kernel), dim3(B), dim3(kBlock), 0, stream, 
+      out.data_ptr<int32_t>(), logits.data_ptr<float>(),
+      inv_temp.data_ptr<float>(), top_k.data_ptr<int32_t>(),
+      top_p.data_ptr<float>(), (uint64_t)seed, V);
   HIP_CHECK_KERNEL();
 }

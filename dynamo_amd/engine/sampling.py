@@ -39,15 +39,26 @@ def sample_tokens(logits: torch.Tensor, reqs: List[Request],
     greedy_mask = torch.tensor([t == 0.0 for t in temps], device=logits.device)
     inv_t = torch.tensor([1.0 / t if t > 0 else 1.0 for t in temps],
                          dtype=torch.float32, device=logits.device)
-    filt = logits
-    if any(r.sampling.top_k > 0 or r.sampling.top_p < 1.0 for r in reqs):
-        # per-request filters; batch-apply with worst-case params per row
-        rows = []
-        for i, r in enumerate(reqs):
-            rows.append(_filter_topk_topp(logits[i:i + 1], r.sampling.top_k,
-                                          r.sampling.top_p))
-        filt = torch.cat(rows, 0)
-    sampled = ops.gumbel_sample(filt, inv_t, step_seed)
+    has_filter = any(r.sampling.top_k > 0 or r.sampling.top_p < 1.0
+                     for r in reqs)
+    if has_filter and logits.is_cuda:
+        # native fused kernel: per-row thresholds, no sort, no host sync
+        tk = torch.tensor([r.sampling.top_k for r in reqs],
+                          dtype=torch.int32, device=logits.device)
+        tp = torch.tensor([r.sampling.top_p for r in reqs],
+                          dtype=torch.float32, device=logits.device)
+        sampled = ops.topkp_sample(logits, inv_t, tk, tp, step_seed)
+    else:
+        filt = logits
+        if has_filter:
+            # CPU: per-request torch filters, then filtered Gumbel-max
+            rows = []
+            for i, r in enumerate(reqs):
+                rows.append(_filter_topk_topp(logits[i:i + 1],
+                                              r.sampling.top_k,
+                                              r.sampling.top_p))
+            filt = torch.cat(rows, 0)
+        sampled = ops.gumbel_sample(filt, inv_t, step_seed)
     if greedy_mask.any():
         greedy = ops.greedy_sample(logits)
         sampled = torch.where(greedy_mask, greedy, sampled)

@@ -168,6 +168,17 @@ def gumbel_sample(logits: torch.Tensor, inv_temp: torch.Tensor,
     return (logits.float() * inv_temp.unsqueeze(-1) + g).argmax(-1).to(torch.int32)
 
 
+def topkp_sample(logits: torch.Tensor, inv_temp: torch.Tensor,
+                 top_k: torch.Tensor, top_p: torch.Tensor,
+                 seed: int) -> torch.Tensor:
+    """Fused top-k/top-p Gumbel sampling (GPU only; per-row params)."""
+    out = torch.empty(logits.shape[0], dtype=torch.int32,
+                      device=logits.device)
+    hip().topkp_sample(out, logits.float().contiguous(), inv_temp,
+                       top_k, top_p, seed)
+    return out
+
+
 def topk_gating(logits: torch.Tensor, k: int):
     """softmax + renormalized top-k over experts. logits [T, E] fp32."""
     if logits.is_cuda:
