@@ -234,6 +234,268 @@ __global__ __launch_bounds__(kBlock) void prefill_kernel(
 
 }  // namespace
 
+// ---------------------------------------------------------------------------
+// 8-wave 32x32 variant (GQA group == 8, e.g. Llama-70B 64q/8kv): one block
+// covers ALL 8 q-heads of one kv head on a 32-row q tile, so each K/V tile
+// is staged once and read by 8 warps. Swapped QK^T (mfma(K,Q), 32x32x16)
+// keeps every lane's scores on ONE q-row -> softmax fully in-register (fmax
+// chain + one cross-half shfl); P converts to PV A-fragments by a symmetric
+// half-exchange, no LDS round-trip. K row-major + V TRANSPOSED in LDS, both
+// XOR-swizzled; double-buffered with register prefetch (one barrier/tile);
+// role-split staging (256 threads K b128 / 256 threads V^T b64); defer-max
+// rescale skip. Measured 260 TF vs 128 TF for the 4-wave 16x16 kernel above
+// (benchmarks/prefill32_sweep.hip ladder; guide "8-warp 32x32 ladder").
+//
+// 32x32x16 fragment mappings (verified by the sweep's probe32 on HW):
+//   A: lane l holds A[row=l%32][k=8*(l/32)+i]
+//   B: lane l holds B[k=8*(l/32)+i][col=l%32]
+//   D: lane l, reg r holds D[row=(r&3)+8*(r>>2)+4*(l/32)][col=l%32]
+namespace {
+
+typedef __attribute__((ext_vector_type(16))) float f32x16;
+typedef __attribute__((ext_vector_type(4))) short short4_t;
+
+constexpr int kQB32 = 32;   // q rows per block
+constexpr int kKB32 = 64;   // kv tokens per tile
+constexpr int kLdsHalf32 = kKB32 * 256 + 128 * 128;  // K + V^T = 32 KB
+
+__global__ __launch_bounds__(512) void prefill32_kernel(
+    short* __restrict__ out,            // [Tq, Hq, 128]
+    const short* __restrict__ q,        // [Tq, Hq, 128]
+    const short* __restrict__ kcache,   // [P, Hkv, ps, 128]
+    const short* __restrict__ vcache,
+    const int32_t* __restrict__ page_table,   // [nseq, max_pages]
+    const int32_t* __restrict__ tile_seq,     // [ntiles] (32-row tiles)
+    const int32_t* __restrict__ tile_q0,
+    const int32_t* __restrict__ seq_q_start,
+    const int32_t* __restrict__ seq_q_len,
+    const int32_t* __restrict__ seq_ctx_len,
+    float scale, int Hq, int Hkv, int max_pages, int log2_ps) {
+  constexpr int HD = 128;
+  const int tile = blockIdx.x;
+  const int kvh = blockIdx.y;
+  const int w = threadIdx.x >> 6;
+  const int l = threadIdx.x & 63;
+  const int h = kvh * 8 + w;
+  const int seq = tile_seq[tile];
+  const int q0 = tile_q0[tile];
+  const int qlen = seq_q_len[seq];
+  const int ctx = seq_ctx_len[seq];
+  const int qstart = seq_q_start[seq];
+  const int ps = 1 << log2_ps;
+  const int32_t* pt = page_table + (int64_t)seq * max_pages;
+  const int lo = l & 31, hi = l >> 5;
+
+  extern __shared__ char lds32[];
+
+  // Q B-fragments for this warp's head, q row q0+lo (clamped if past qlen)
+  const int my_qrow = q0 + lo;
+  const bool row_valid = my_qrow < qlen;
+  const short* qrow_p =
+      q + ((int64_t)(qstart + (row_valid ? my_qrow : qlen - 1)) * Hq + h) * HD;
+  bf16x8_t qreg[8];
+#pragma unroll
+  for (int ds = 0; ds < 8; ds++) {
+    short8 v = *reinterpret_cast<const short8*>(qrow_p + ds * 16 + 8 * hi);
+    qreg[ds] = *reinterpret_cast<bf16x8_t*>(&v);
+  }
+
+  f32x16 o[4] = {};
+  float m_run = kNegInf, l_run = 0.f;
+  const int my_qpos = ctx - qlen + my_qrow;
+  const int tile_qpos_max = ctx - qlen + min(q0 + kQB32 - 1, qlen - 1);
+  const int kv_end = min(ctx, tile_qpos_max + 1);
+
+  // role-split staging: threads 0-255 stage V^T (b64 writes, 4 toks/write),
+  // threads 256-511 stage K (b128 writes); register prefetch double-buffer
+  const int vrole = threadIdx.x < 256;
+  short8 sreg[4];
+  auto load_tile = [&](int t0) {
+    if (vrole) {
+      const int unit = threadIdx.x;
+      const int row0 = (unit >> 4) * 4, d0 = (unit & 15) * 8;
+#pragma unroll
+      for (int j = 0; j < 4; j++) {
+        const int t = min(t0 + row0 + j, ctx - 1);
+        const int64_t page = pt[t >> log2_ps];
+        sreg[j] = *reinterpret_cast<const short8*>(
+            vcache + ((page * Hkv + kvh) * ps + (t & (ps - 1))) * HD + d0);
+      }
+    } else {
+      const int idx = threadIdx.x - 256;
+#pragma unroll
+      for (int u = 0; u < 4; u++) {
+        const int c = idx + u * 256;
+        const int row = c >> 4, col8 = (c & 15) * 8;
+        const int t = min(t0 + row, ctx - 1);
+        const int64_t page = pt[t >> log2_ps];
+        sreg[u] = *reinterpret_cast<const short8*>(
+            kcache + ((page * Hkv + kvh) * ps + (t & (ps - 1))) * HD + col8);
+      }
+    }
+  };
+  auto store_tile = [&](int buf) {
+    char* kl = lds32 + buf * kLdsHalf32;
+    char* vl = kl + kKB32 * 256;
+    if (vrole) {
+      const int unit = threadIdx.x;
+      const int row0 = (unit >> 4) * 4, d0 = (unit & 15) * 8;
+#pragma unroll
+      for (int i = 0; i < 8; i++) {
+        const int d = d0 + i;
+        short4_t pk = {sreg[0][i], sreg[1][i], sreg[2][i], sreg[3][i]};
+        *(short4_t*)(vl + d * 128 + ((row0 * 2) ^ ((d & 7) << 4))) = pk;
+      }
+    } else {
+      const int idx = threadIdx.x - 256;
+#pragma unroll
+      for (int u = 0; u < 4; u++) {
+        const int c = idx + u * 256;
+        const int row = c >> 4, col8 = (c & 15) * 8;
+        *reinterpret_cast<short8*>(
+            kl + row * 256 + ((col8 * 2) ^ ((row & 7) << 4))) = sreg[u];
+      }
+    }
+  };
+
+  load_tile(0);
+  store_tile(0);
+  __syncthreads();
+  if (kKB32 < kv_end) load_tile(kKB32);
+
+  for (int t0 = 0; t0 < kv_end; t0 += kKB32) {
+    const int cur = (t0 / kKB32) & 1;
+    const char* kl = lds32 + cur * kLdsHalf32;
+    const char* vl = kl + kKB32 * 256;
+
+    // ---- QK^T: S^T tiles [32tok x 32q] via mfma(K, Q) ----
+    f32x16 s0 = {}, s1 = {};
+#pragma unroll
+    for (int ds = 0; ds < 8; ds++) {
+      const int koff = ds * 32 + hi * 16;
+      short8 a0 = *reinterpret_cast<const short8*>(
+          kl + lo * 256 + (koff ^ ((lo & 7) << 4)));
+      short8 a1 = *reinterpret_cast<const short8*>(
+          kl + (lo + 32) * 256 + (koff ^ ((lo & 7) << 4)));
+      s0 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+          *reinterpret_cast<bf16x8_t*>(&a0), qreg[ds], s0, 0, 0, 0);
+      s1 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+          *reinterpret_cast<bf16x8_t*>(&a1), qreg[ds], s1, 0, 0, 0);
+    }
+
+    // ---- mask + scale + in-register softmax (lane owns q-row lo) ----
+    float p[32];
+    float mt = kNegInf;
+#pragma unroll
+    for (int r = 0; r < 16; r++) {
+      const int trow = (r & 3) + 8 * (r >> 2) + 4 * hi;
+      const bool ok0 = row_valid && t0 + trow <= my_qpos;
+      const bool ok1 = row_valid && t0 + 32 + trow <= my_qpos;
+      p[r] = ok0 ? s0[r] * scale : kNegInf;
+      p[16 + r] = ok1 ? s1[r] * scale : kNegInf;
+      mt = fmaxf(mt, fmaxf(p[r], p[16 + r]));
+    }
+    mt = fmaxf(mt, __shfl_xor(mt, 32, WAVE_SIZE));
+    float m_new = fmaxf(m_run, mt);
+    bool skip_rescale = false;
+    // defer-max: skip the O-rescale while the tile max stays within 8
+    if (__all(mt - m_run <= 8.0f)) { m_new = m_run; skip_rescale = true; }
+    const float alpha =
+        (skip_rescale || m_run <= kNegInf * 0.5f) ? 1.f : __expf(m_run - m_new);
+    float ls = 0.f;
+#pragma unroll
+    for (int r = 0; r < 32; r++) {
+      p[r] = (p[r] <= kNegInf * 0.5f || m_new <= kNegInf * 0.5f)
+                 ? 0.f : __expf(p[r] - m_new);
+      ls += p[r];
+    }
+    ls += __shfl_xor(ls, 32, WAVE_SIZE);
+    l_run = l_run * alpha + ls;
+    m_run = m_new;
+    if (!skip_rescale) {
+      // O is in D-layout: reg r belongs to q-row (r&3)+8*(r>>2)+4*hi, so
+      // rescale with THAT row's alpha (alpha is half-replicated per q-row)
+      float arow[16];
+#pragma unroll
+      for (int r = 0; r < 16; r++)
+        arow[r] = __shfl(alpha, (r & 3) + 8 * (r >> 2) + 4 * hi, WAVE_SIZE);
+#pragma unroll
+      for (int dt = 0; dt < 4; dt++)
+#pragma unroll
+        for (int r = 0; r < 16; r++) o[dt][r] *= arow[r];
+    }
+
+    // ---- P -> PV A-fragments via symmetric half-exchange ----
+    bf16x8_t pa[4];
+#pragma unroll
+    for (int ks = 0; ks < 4; ks++) {
+      float recv[4];
+#pragma unroll
+      for (int j = 0; j < 4; j++) {
+        const int Ts = ks * 16 + (hi ? 4 + j : 8 + j);
+        const int T32 = Ts & 31;
+        const float send = p[(T32 & 3) + 4 * (T32 >> 3) + 16 * (Ts >> 5)];
+        recv[j] = __shfl_xor(send, 32, WAVE_SIZE);
+      }
+#pragma unroll
+      for (int i = 0; i < 8; i++) {
+        float val;
+        if (i < 4) {
+          const int T = ks * 16 + i;
+          const int T32 = T & 31;
+          val = hi ? recv[i]
+                   : p[(T32 & 3) + 4 * (T32 >> 3) + 16 * (T >> 5)];
+        } else {
+          const int T = ks * 16 + 8 + i;
+          const int T32 = T & 31;
+          val = hi ? p[(T32 & 3) + 4 * (T32 >> 3) + 16 * (T >> 5)]
+                   : recv[i - 4];
+        }
+        pa[ks][i] = (__bf16)val;
+      }
+    }
+
+    // ---- O += P V ----
+#pragma unroll
+    for (int dt = 0; dt < 4; dt++) {
+      const int drow = dt * 32 + lo;
+#pragma unroll
+      for (int ks = 0; ks < 4; ks++) {
+        const int toff = (ks * 16 + 8 * hi) * 2;
+        short8 vb = *reinterpret_cast<const short8*>(
+            vl + drow * 128 + (toff ^ ((drow & 7) << 4)));
+        o[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+            pa[ks], *reinterpret_cast<bf16x8_t*>(&vb), o[dt], 0, 0, 0);
+      }
+    }
+
+    // stage next tile into the other buffer; prefetch tile t+2
+    if (t0 + kKB32 < kv_end) {
+      store_tile(cur ^ 1);
+      if (t0 + 2 * kKB32 < kv_end) load_tile(t0 + 2 * kKB32);
+    }
+    __syncthreads();
+  }
+
+  // ---- epilogue: divide by the q-ROW's l (D-layout), store bf16 ----
+  float lrow[16];
+#pragma unroll
+  for (int r = 0; r < 16; r++)
+    lrow[r] = __shfl(l_run, (r & 3) + 8 * (r >> 2) + 4 * hi, WAVE_SIZE);
+#pragma unroll
+  for (int dt = 0; dt < 4; dt++)
+#pragma unroll
+    for (int r = 0; r < 16; r++) {
+      const int qrow_i = q0 + (r & 3) + 8 * (r >> 2) + 4 * hi;
+      if (qrow_i >= qlen) continue;
+      const float inv = lrow[r] > 0.f ? 1.f / lrow[r] : 0.f;
+      out[((int64_t)(qstart + qrow_i) * Hq + h) * HD + dt * 32 + lo] =
+          f32_to_bf16(o[dt][r] * inv);
+    }
+}
+
+}  // namespace
+
 void attention_prefill_paged(torch::Tensor out, torch::Tensor q,
                              torch::Tensor kcache, torch::Tensor vcache,
                              torch::Tensor page_table, torch::Tensor tile_seq,
@@ -252,6 +514,20 @@ void attention_prefill_paged(torch::Tensor out, torch::Tensor q,
   const int max_pages = page_table.size(1);
   if (ntiles == 0) return;
   auto stream = at::cuda::getCurrentHIPStream();
+  // GQA group of 8 -> 8-wave 32x32 kernel (32-row tiles, built by the
+  // Python layer via prefill_tile_rows — keep the predicates in sync)
+  if (Hq == 8 * Hkv) {
+    dim3 grid(ntiles, Hkv);
+   hipLaunchKernelGGL(( prefill32_kernel), dim3(grid), dim3(512), 2 * kLdsHalf32, stream, 
+        (short*)out.data_ptr(), (const short*)q.data_ptr(),
+        (const short*)kcache.data_ptr(), (const short*)vcache.data_ptr(),
+        page_table.data_ptr<int32_t>(), tile_seq.data_ptr<int32_t>(),
+        tile_q0.data_ptr<int32_t>(), seq_q_start.data_ptr<int32_t>(),
+        seq_q_len.data_ptr<int32_t>(), seq_ctx_len.data_ptr<int32_t>(),
+        (float)scale, Hq, Hkv, max_pages, log2_ps);
+    HIP_CHECK_KERNEL();
+    return;
+  }
   dim3 grid(ntiles, Hq);
  hipLaunchKernelGGL(( prefill_kernel), dim3(grid), dim3(kBlock), 0, stream, 
       (short*)out.data_ptr(), (const short*)q.data_ptr(),

@@ -132,7 +132,8 @@ class ModelRunner:
             logits_rows=torch.tensor(logits_rows, dtype=torch.int64).to(dev, non_blocking=True),
         )
         if qpos and dev.type == "cuda":
-            meta.prefill_tiles = ops.build_prefill_tiles(q_len, dev)
+            rows = ops.prefill_tile_rows(self.hq_local, self.hkv_local)
+            meta.prefill_tiles = ops.build_prefill_tiles(q_len, dev, rows)
         input_ids = torch.tensor(tokens, dtype=torch.int32).to(dev, non_blocking=True)
         return input_ids, meta
 

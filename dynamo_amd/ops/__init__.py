@@ -120,11 +120,18 @@ def paged_attention_decode(q, kcache, vcache, page_table, ctx_lens, scale,
                                             ctx_lens, scale)
 
 
-def build_prefill_tiles(seq_q_lens, device):
-    """Host-side tile descriptors for the prefill kernel (64-row q tiles)."""
+def prefill_tile_rows(num_q_heads: int, num_kv_heads: int) -> int:
+    """Q-tile granularity of the native prefill kernel: the 8-wave 32x32
+    kernel (GQA group == 8) takes 32-row tiles; the 4-wave 16x16 kernel
+    takes 64-row tiles. MUST match the dispatch in attention_prefill.hip."""
+    return 32 if num_q_heads == 8 * num_kv_heads else 64
+
+
+def build_prefill_tiles(seq_q_lens, device, rows: int = 64):
+    """Host-side tile descriptors for the prefill kernel."""
     tile_seq, tile_q0 = [], []
     for s, ql in enumerate(seq_q_lens):
-        for q0 in range(0, int(ql), 64):
+        for q0 in range(0, int(ql), rows):
             tile_seq.append(s)
             tile_q0.append(q0)
     return (torch.tensor(tile_seq, dtype=torch.int32, device=device),
@@ -136,7 +143,8 @@ def attention_prefill_paged(q, kcache, vcache, page_table, seq_q_start,
     if q.is_cuda:
         out = torch.empty_like(q)
         if tiles is None:
-            tiles = build_prefill_tiles(seq_q_len.tolist(), q.device)
+            rows = prefill_tile_rows(q.shape[1], kcache.shape[1])
+            tiles = build_prefill_tiles(seq_q_len.tolist(), q.device, rows)
         tile_seq, tile_q0 = tiles
         hip().attention_prefill_paged(out, q, kcache, vcache, page_table,
                                       tile_seq, tile_q0, seq_q_start,
