@@ -65,7 +65,8 @@ def bench_prefill_attn(S=8192, Hq=32, Hkv=8, ps=64):
     starts = torch.tensor([0], dtype=torch.int32, device="cuda")
     qlen = torch.tensor([S], dtype=torch.int32, device="cuda")
     ctxl = torch.tensor([S], dtype=torch.int32, device="cuda")
-    tiles = ops.build_prefill_tiles([S], "cuda")
+    tiles = ops.build_prefill_tiles([S], "cuda",
+                                    ops.prefill_tile_rows(Hq, Hkv))
     t = timeit(lambda: ops.attention_prefill_paged(
         q, kc, vc, pt, starts, qlen, ctxl, hd ** -0.5, tiles), iters=5)
     # causal flops: 2 gemms * 2*S*S/2*hd per head
@@ -119,6 +120,8 @@ if __name__ == "__main__":
     if w in ("all", "prefill"):
         bench_prefill_attn()
         bench_prefill_attn(S=2048)
+        bench_prefill_attn(Hq=64)          # GQA 8 (llama-70b TP1)
+        bench_prefill_attn(S=2048, Hq=64)
     if w in ("all", "append"):
         bench_kv_append()
     if w in ("all", "sample"):

@@ -258,26 +258,32 @@ constexpr int kQB32 = 32;   // q rows per block
 constexpr int kKB32 = 64;   // kv tokens per tile
 constexpr int kLdsHalf32 = kKB32 * 256 + 128 * 128;  // K + V^T = 32 KB
 
+// GSPLIT generalizes to GQA groups G = 8/GSPLIT: the 8 warps are G heads
+// x GSPLIT 32-row q-subtiles, all sharing one staged K/V tile (block tile
+// is GSPLIT*32 q rows). G=8 -> 1 subtile; G=4 -> 2; G=2 -> 4.
+template <int GSPLIT>
 __global__ __launch_bounds__(512) void prefill32_kernel(
     short* __restrict__ out,            // [Tq, Hq, 128]
     const short* __restrict__ q,        // [Tq, Hq, 128]
     const short* __restrict__ kcache,   // [P, Hkv, ps, 128]
     const short* __restrict__ vcache,
     const int32_t* __restrict__ page_table,   // [nseq, max_pages]
-    const int32_t* __restrict__ tile_seq,     // [ntiles] (32-row tiles)
+    const int32_t* __restrict__ tile_seq,     // [ntiles] (GSPLIT*32-row tiles)
     const int32_t* __restrict__ tile_q0,
     const int32_t* __restrict__ seq_q_start,
     const int32_t* __restrict__ seq_q_len,
     const int32_t* __restrict__ seq_ctx_len,
     float scale, int Hq, int Hkv, int max_pages, int log2_ps) {
   constexpr int HD = 128;
+  constexpr int G = 8 / GSPLIT;           // q heads per kv head
   const int tile = blockIdx.x;
   const int kvh = blockIdx.y;
   const int w = threadIdx.x >> 6;
   const int l = threadIdx.x & 63;
-  const int h = kvh * 8 + w;
+  const int h = kvh * G + (w % G);
   const int seq = tile_seq[tile];
-  const int q0 = tile_q0[tile];
+  const int q0blk = tile_q0[tile];
+  const int q0 = q0blk + (w / G) * kQB32; // this warp's q-subtile
   const int qlen = seq_q_len[seq];
   const int ctx = seq_ctx_len[seq];
   const int qstart = seq_q_start[seq];
@@ -302,7 +308,9 @@ __global__ __launch_bounds__(512) void prefill32_kernel(
   f32x16 o[4] = {};
   float m_run = kNegInf, l_run = 0.f;
   const int my_qpos = ctx - qlen + my_qrow;
-  const int tile_qpos_max = ctx - qlen + min(q0 + kQB32 - 1, qlen - 1);
+  // block-uniform kv range: last q row of the whole block tile
+  const int tile_qpos_max =
+      ctx - qlen + min(q0blk + GSPLIT * kQB32 - 1, qlen - 1);
   const int kv_end = min(ctx, tile_qpos_max + 1);
 
   // role-split staging: threads 0-255 stage V^T (b64 writes, 4 toks/write),
@@ -513,17 +521,24 @@ void attention_prefill_paged(torch::Tensor out, torch::Tensor q,
   const int max_pages = page_table.size(1);
   if (ntiles == 0) return;
   auto stream = at::cuda::getCurrentHIPStream();
-  // GQA group of 8 -> 8-wave 32x32 kernel (32-row tiles, built by the
-  // Python layer via prefill_tile_rows — keep the predicates in sync)
-  if (Hq == 8 * Hkv) {
+  // GQA group G in {2,4,8} -> 8-wave 32x32 kernel with GSPLIT=8/G q-subtiles
+  // (tile rows = (8/G)*32, built by the Python layer via prefill_tile_rows —
+  // keep the predicates in sync)
+  const int G = (Hq % Hkv == 0) ? Hq / Hkv : 0;
+  if (G == 8 || G == 4 || G == 2) {
     dim3 grid(ntiles, Hkv);
-    prefill32_kernel<<<grid, 512, 2 * kLdsHalf32, stream>>>(
-        (short*)out.data_ptr(), (const short*)q.data_ptr(),
-        (const short*)kcache.data_ptr(), (const short*)vcache.data_ptr(),
-        page_table.data_ptr<int32_t>(), tile_seq.data_ptr<int32_t>(),
-        tile_q0.data_ptr<int32_t>(), seq_q_start.data_ptr<int32_t>(),
-        seq_q_len.data_ptr<int32_t>(), seq_ctx_len.data_ptr<int32_t>(),
-        (float)scale, Hq, Hkv, max_pages, log2_ps);
+    auto launch = [&](auto kern) {
+      kern<<<grid, 512, 2 * kLdsHalf32, stream>>>(
+          (short*)out.data_ptr(), (const short*)q.data_ptr(),
+          (const short*)kcache.data_ptr(), (const short*)vcache.data_ptr(),
+          page_table.data_ptr<int32_t>(), tile_seq.data_ptr<int32_t>(),
+          tile_q0.data_ptr<int32_t>(), seq_q_start.data_ptr<int32_t>(),
+          seq_q_len.data_ptr<int32_t>(), seq_ctx_len.data_ptr<int32_t>(),
+          (float)scale, Hq, Hkv, max_pages, log2_ps);
+    };
+    if (G == 8) launch(prefill32_kernel<1>);
+    else if (G == 4) launch(prefill32_kernel<2>);
+    else launch(prefill32_kernel<4>);
     HIP_CHECK_KERNEL();
     return;
   }

@@ -122,9 +122,14 @@ def paged_attention_decode(q, kcache, vcache, page_table, ctx_lens, scale,
 
 def prefill_tile_rows(num_q_heads: int, num_kv_heads: int) -> int:
     """Q-tile granularity of the native prefill kernel: the 8-wave 32x32
-    kernel (GQA group == 8) takes 32-row tiles; the 4-wave 16x16 kernel
-    takes 64-row tiles. MUST match the dispatch in attention_prefill.hip."""
-    return 32 if num_q_heads == 8 * num_kv_heads else 64
+    kernel handles GQA groups G in {2,4,8} with (8/G)*32-row tiles; the
+    4-wave 16x16 kernel takes 64-row tiles. MUST match the dispatch in
+    attention_prefill.hip."""
+    if num_q_heads % num_kv_heads == 0:
+        g = num_q_heads // num_kv_heads
+        if g in (2, 4, 8):
+            return (8 // g) * 32
+    return 64
 
 
 def build_prefill_tiles(seq_q_lens, device, rows: int = 64):
