@@ -48,6 +48,13 @@ void paged_attention_decode(torch::Tensor out, torch::Tensor q,
     return e == nullptr || e[0] != '0';  // default ON (sweep-verified)
   }();
 #define LAUNCH_MFMA(GG)                                                       \
+  do {                                                                        \
+    if (mfma_lds_bytes(GG, hd) > 65536)                                       \
+      (void)hipFuncSetAttribute(                                              \
+          reinterpret_cast<const void*>(&paged_decode_mfma<GG>),              \
+          hipFuncAttributeMaxDynamicSharedMemorySize,                         \
+          mfma_lds_bytes(GG, hd));                                            \
+  } while (0);                                                                \
  hipLaunchKernelGGL(( paged_decode_mfma<GG>), dim3(grid), dim3(kBlock), mfma_lds_bytes(GG, hd), stream,     \
       partial.data_ptr<float>(), ml.data_ptr<float>(), (short*)out.data_ptr(),\
       (const short*)q.data_ptr(), (const short*)kcache.data_ptr(),            \

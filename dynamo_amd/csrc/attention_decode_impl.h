@@ -299,13 +299,16 @@ __global__ __launch_bounds__(kBlock) void paged_decode_mfma(
   const int lg = lane >> 4;       // 4 k-groups
   const int ps = 1 << log2_ps;
 
-  // LDS: [4 waves][V tile 8KB] + [4 waves][P 16x32 bf16 1KB] + q G*128 bf16
-  // + merge scratch (4*G*(hd+2) fp32)
+  // LDS: [4 waves][V^T tile 10KB] + [4 waves][P 16x32 bf16 1KB] + q G*128
+  // bf16 + merge scratch (4*G*(hd+2) fp32). V is staged TRANSPOSED
+  // [128 dims][32 toks] with 80-byte row stride (16B pad) so PV B-frags
+  // are single b128 row reads (the 80B stride spreads the 16 lanes' rows
+  // across banks; was 8 scalar gathers per frag in the row-major layout).
   extern __shared__ float lds[];
   float* merge = lds;                                   // 4*G*(hd+2)
   short* q_lds_s = reinterpret_cast<short*>(merge + 4 * G * (hd + 2));
-  short* v_lds = q_lds_s + G * hd + wid * 4096;         // per-wave 8KB
-  short* p_lds = q_lds_s + G * hd + 4 * 4096 + wid * 512;  // per-wave 1KB
+  short* v_lds = q_lds_s + G * hd + wid * 5120;         // per-wave 10KB
+  short* p_lds = q_lds_s + G * hd + 4 * 5120 + wid * 512;  // per-wave 1KB
 
   if (chunk_start >= ctx) {
     if (C > 1) {
@@ -356,6 +359,8 @@ __global__ __launch_bounds__(kBlock) void paged_decode_mfma(
   // Uniform tile count across ALL waves (inactive waves still hit the
   // barriers — __syncthreads inside a loop with per-wave iteration counts
   // is barrier divergence / UB).
+  // (A register K-prefetch across tiles was measured SLOWER — 2559 vs
+  // 3000 GB/s — VGPR pressure beats the latency win on this kernel.)
   for (int ti = 0; ti < kSlab / 32; ti++) {
     const int t0 = slab_start + ti * 32;
     const bool active = t0 < slab_end;
@@ -384,21 +389,33 @@ __global__ __launch_bounds__(kBlock) void paged_decode_mfma(
         }
       }
     }
-    // ---- stage V tile (32 toks x 128 dims) ----
-    if (active)
+    // ---- stage V^T tile (128 dims x 32 toks, 80B row stride) ----
+    // unit = (8-dim chunk, 4-token group): 4 b128 global loads, then 8
+    // b64 packed transpose writes (4 toks each) — 16 b64 writes/lane/tile
+    // and NO scalar LDS traffic on either side of the transpose.
+    if (active) {
+      typedef __attribute__((ext_vector_type(4))) short short4v;
 #pragma unroll
-    for (int it = 0; it < 8; it++) {
-      const int slot = lane + it * 64;   // 512 slots = 32 toks x 16 chunks
-      const int d16 = slot & 15;         // 16B chunk of the 256B row
-      const int tl = slot >> 4;          // token 0..31
-      const int t = t0 + tl;
-      short8 vv = (t < slab_end)
-          ? *reinterpret_cast<const short8*>(
-                vcache + pbase + (int64_t)(t & (ps - 1)) * hd + d16 * 8)
-          : short8{};
-      // row-major [32][128] with an XOR bank swizzle on the row offset
-      *reinterpret_cast<short8*>(
-          (char*)v_lds + tl * 256 + ((d16 * 16) ^ ((tl & 7) << 4))) = vv;
+      for (int u = 0; u < 2; u++) {
+        const int unit = lane + u * 64;  // 128 units = 16 d8 x 8 tokgroups
+        const int d8 = unit & 15;
+        const int tg = unit >> 4;
+        short8 rows[4];
+#pragma unroll
+        for (int j = 0; j < 4; j++) {
+          const int t = t0 + tg * 4 + j;
+          rows[j] = (t < slab_end)
+              ? *reinterpret_cast<const short8*>(
+                    vcache + pbase + (int64_t)(t & (ps - 1)) * hd + d8 * 8)
+              : short8{};
+        }
+#pragma unroll
+        for (int i = 0; i < 8; i++) {
+          const int d = d8 * 8 + i;
+          short4v pk = {rows[0][i], rows[1][i], rows[2][i], rows[3][i]};
+          *reinterpret_cast<short4v*>((char*)v_lds + d * 80 + tg * 8) = pk;
+        }
+      }
     }
 
     // ---- online softmax on the C/D layout ----
@@ -446,7 +463,7 @@ __global__ __launch_bounds__(kBlock) void paged_decode_mfma(
     asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
     __builtin_amdgcn_sched_barrier(0);
 
-    // ---- PV: A = P (re-layout via LDS), B = V scalar-transposed ----
+    // ---- PV: A = P (re-layout via LDS), B = V transposed in LDS ----
     if (active) {
     short8 pa_s;
     {
@@ -459,22 +476,9 @@ __global__ __launch_bounds__(kBlock) void paged_decode_mfma(
 #pragma unroll
     for (int db = 0; db < 8; db++) {
       // B-frag: lane holds V[tok = 8*lg + j][dim = db*16 + lr], j = 0..7 —
-      // transposed scalar gathers from the swizzled row-major tile
-      short8 vb_s;
-#pragma unroll
-      for (int j = 0; j < 8; j++) {
-        const int tok = 8 * lg + j;
-#ifdef DECODE_MFMA_V_DIRECT   // debug: bypass LDS staging
-        const int t = t0 + tok;
-        vb_s[j] = (t < slab_end)
-            ? vcache[pbase + (int64_t)(t & (ps - 1)) * hd + db * 16 + lr]
-            : (short)0;
-#else
-        vb_s[j] = *(const short*)(
-            (const char*)v_lds + tok * 256 +
-            (((db * 16 + lr) * 2) ^ ((tok & 7) << 4)));
-#endif
-      }
+      // ONE b128 read of the transposed V^T row (80B stride spreads banks)
+      short8 vb_s = *reinterpret_cast<const short8*>(
+          (const char*)v_lds + (db * 16 + lr) * 80 + lg * 16);
       bf16x8_t vbf = *reinterpret_cast<bf16x8_t*>(&vb_s);
       acc[db] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pa, vbf, acc[db],
                                                         0, 0, 0);
@@ -534,7 +538,7 @@ __global__ __launch_bounds__(kBlock) void paged_decode_mfma(
 }
 
 inline int mfma_lds_bytes(int G, int hd) {
-  return 4 * G * (hd + 2) * 4 + G * hd * 2 + 4 * 4096 * 2 + 4 * 512 * 2;
+  return 4 * G * (hd + 2) * 4 + G * hd * 2 + 4 * 5120 * 2 + 4 * 512 * 2;
 }
 
 // Phase 2: merge chunk partials. grid (B, Hq), block = 128.
