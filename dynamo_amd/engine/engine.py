@@ -150,6 +150,7 @@ class LLMEngine:
                    for r in running)
 
     def _fast_decode_step(self) -> List[StepOutput]:
+        from dynamo_amd.observability import roctx_range
         gr = self.graph_runner
         running = self.scheduler.running
         rebuilt = gr.dirty or gr.reqs != running
@@ -170,7 +171,8 @@ class LLMEngine:
         t0 = time.monotonic()
         if self.host_tier is not None:
             self.host_tier.fence()
-        logits = gr.step(self._last_sampled)
+        with roctx_range("decode_step"):
+            logits = gr.step(self._last_sampled)
         from .sampling import sample_tokens
         sampled = sample_tokens(logits, running, self.step_count)
         self._last_sampled = sampled
@@ -227,7 +229,10 @@ class LLMEngine:
         self.step_count += 1
         if self.host_tier is not None:
             self.host_tier.fence()
-        sampled, sample_reqs = self.runner.execute(sched, self.step_count)
+        from dynamo_amd.observability import roctx_range
+        with roctx_range(f"engine_step[{len(sched.decodes)}d+"
+                         f"{sum(s.n_new for s in sched.prefills)}p]"):
+            sampled, sample_reqs = self.runner.execute(sched, self.step_count)
         sampled = sampled.cpu().tolist() if len(sample_reqs) else []
 
         # advance computed counts

@@ -28,6 +28,12 @@ def build_parser():
                    choices=["kv", "round_robin", "random", "least_loaded",
                             "p2c"])
     p.add_argument("--router-temperature", type=float, default=0.0)
+    p.add_argument("--busy-threshold", type=float, default=0.0,
+                   help="reject with 503 when every worker's load exceeds "
+                        "this (0 = disabled)")
+    p.add_argument("--record", default=None, metavar="FILE",
+                   help="record requests + response chunks to a JSONL file "
+                        "(replay with python -m dynamo_amd.tools.replay)")
     return p
 
 
@@ -37,7 +43,9 @@ async def async_main(args):
     mgr = ModelManager(rt, namespace=args.namespace,
                        router_cfg=RouterConfig(
                            mode=args.router_mode,
-                           router_temperature=args.router_temperature))
+                           router_temperature=args.router_temperature,
+                           busy_threshold=args.busy_threshold),
+                       record_path=args.record)
     await mgr.start()
     app = build_app(mgr)
     config = uvicorn.Config(app, host=args.host, port=args.port,

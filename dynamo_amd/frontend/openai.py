@@ -39,6 +39,7 @@ class CompletionRequest(BaseModel):
     stop: Optional[Union[str, List[str]]] = None
     seed: int = 0
     ignore_eos: bool = False
+    user: Optional[str] = None      # sticky-session key
 
 
 class ChatMessage(BaseModel):
@@ -62,6 +63,11 @@ class ChatRequest(BaseModel):
     stream: bool = False
     seed: int = 0
     ignore_eos: bool = False
+    user: Optional[str] = None      # sticky-session key
+
+
+def _session_of(req, raw: Request) -> Optional[str]:
+    return raw.headers.get("x-session-id") or getattr(req, "user", None)
 
 
 def build_app(manager: ModelManager) -> FastAPI:
@@ -85,16 +91,37 @@ def build_app(manager: ModelManager) -> FastAPI:
              if isinstance(e.card, dict) else int(time.time())}
             for name, e in manager.models.items()]}
 
-    async def _run(entry, token_ids, req, rid) -> AsyncIterator[dict]:
+    async def _run(entry, token_ids, req, rid,
+                   session_id=None) -> AsyncIterator[dict]:
         sampling = {"temperature": req.temperature, "top_p": req.top_p,
                     "top_k": req.top_k, "seed": req.seed}
         eos = getattr(entry.tokenizer, "eos_id", None)
         stop = {"max_tokens": req.max_tokens,
                 "ignore_eos": req.ignore_eos,
                 "stop_token_ids": [eos] if eos is not None else []}
-        async for chunk in manager.generate_tokens(entry, token_ids, sampling,
-                                                   stop, request_id=rid):
-            yield chunk
+        from dynamo_amd.router.kv_router import AllWorkersBusy
+        try:
+            async for chunk in manager.generate_tokens(
+                    entry, token_ids, sampling, stop, request_id=rid,
+                    session_id=session_id):
+                yield chunk
+        except AllWorkersBusy as e:
+            raise HTTPException(503, str(e))
+
+    @app.get("/config")
+    async def config_dump():
+        """Reproducibility config dump (reference parity:
+        components common/config_dump)."""
+        import dataclasses
+        return {
+            "router": dataclasses.asdict(manager.router_cfg),
+            "namespace": manager.namespace,
+            "models": {name: {"card": e.card,
+                              "workers": [i.instance_id for i in
+                                          e.router.client.instances()]
+                              if e.router else []}
+                       for name, e in manager.models.items()},
+        }
 
     def _entry_or_404(model):
         try:
@@ -118,7 +145,7 @@ def build_app(manager: ModelManager) -> FastAPI:
                 produced: List[int] = []
                 first = True
                 try:
-                    async for chunk in _run(entry, token_ids, req, rid):
+                    async for chunk in _run(entry, token_ids, req, rid, _session_of(req, raw)):
                         if await raw.is_disconnected():
                             break
                         if first:
@@ -140,7 +167,7 @@ def build_app(manager: ModelManager) -> FastAPI:
 
         produced: List[int] = []
         finish = None
-        async for chunk in _run(entry, token_ids, req, rid):
+        async for chunk in _run(entry, token_ids, req, rid, _session_of(req, raw)):
             produced.extend(chunk.get("token_ids", []))
             finish = chunk.get("finish_reason") or finish
         LATENCY.labels(entry.name).observe(time.time() - t0)
@@ -203,7 +230,7 @@ def build_app(manager: ModelManager) -> FastAPI:
                 produced: List[int] = []
                 first = True
                 try:
-                    async for chunk in _run(entry, token_ids, req, rid):
+                    async for chunk in _run(entry, token_ids, req, rid, _session_of(req, raw)):
                         if await raw.is_disconnected():
                             break
                         delta = {}
@@ -227,7 +254,7 @@ def build_app(manager: ModelManager) -> FastAPI:
 
         produced: List[int] = []
         finish = None
-        async for chunk in _run(entry, token_ids, req, rid):
+        async for chunk in _run(entry, token_ids, req, rid, _session_of(req, raw)):
             produced.extend(chunk.get("token_ids", []))
             finish = chunk.get("finish_reason") or finish
         LATENCY.labels(entry.name).observe(time.time() - t0)

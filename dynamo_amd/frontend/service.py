@@ -42,13 +42,25 @@ class ModelManager:
     """Watches discovery; maintains per-model routers + tokenizers."""
 
     def __init__(self, runtime: DistributedRuntime, namespace: str = "dynamo",
-                 router_cfg: RouterConfig | None = None):
+                 router_cfg: RouterConfig | None = None,
+                 record_path: Optional[str] = None):
         self.runtime = runtime
         self.namespace = namespace
         self.router_cfg = router_cfg or RouterConfig()
         self.models: Dict[str, ModelEntry] = {}
         self._watch_task: Optional[asyncio.Task] = None
         self.request_count = 0
+        # request recording (reference parity: lib/llm recorder.rs +
+        # components replay): JSONL of {request, chunks, latency}
+        self.record_path = record_path
+        self._record_fh = None
+
+    def _record(self, rec: dict):
+        import json
+        if self._record_fh is None:
+            self._record_fh = open(self.record_path, "a")
+        self._record_fh.write(json.dumps(rec) + "\n")
+        self._record_fh.flush()
 
     async def start(self, watch_interval: float = 1.0):
         await self.runtime.start()
@@ -142,12 +154,15 @@ class ModelManager:
 
     async def generate_tokens(self, entry: ModelEntry, token_ids: List[int],
                               sampling: dict, stop: dict,
-                              request_id: Optional[str] = None
+                              request_id: Optional[str] = None,
+                              session_id: Optional[str] = None
                               ) -> AsyncIterator[dict]:
         """Route + stream with migration retry (replays delivered tokens)."""
         self.request_count += 1
         rid = request_id or f"{uuid.uuid4().hex[:16]}"
         delivered: List[int] = []
+        recorded: List[dict] = []
+        t_start = time.time()
         attempts = 0
         trace_event("request_start", request_id=rid, model=entry.name,
                     prompt_tokens=len(token_ids))
@@ -168,7 +183,8 @@ class ModelManager:
                         and entry.prefill_router.has_prefill_pool()):
                     gen = entry.prefill_router.generate(payload)
                 else:
-                    iid = entry.router.select(payload["token_ids"])
+                    iid = entry.router.select(payload["token_ids"],
+                                              session_id=session_id)
                     if iid is None:
                         raise NoInstancesError(f"no workers for {entry.name}")
                     entry.router.begin_request(iid, payload["token_ids"])
@@ -176,9 +192,18 @@ class ModelManager:
                 async for chunk in gen:
                     for t in chunk.get("token_ids", []):
                         delivered.append(t)
+                    if self.record_path:
+                        recorded.append(chunk)
                     yield chunk
                 trace_event("request_end", request_id=rid,
                             output_tokens=len(delivered), attempts=attempts)
+                if self.record_path:
+                    self._record({"ts": t_start, "request_id": rid,
+                                  "model": entry.name,
+                                  "token_ids": list(token_ids),
+                                  "sampling": sampling, "stop": stop,
+                                  "latency_s": time.time() - t_start,
+                                  "chunks": recorded})
                 return
             except (EndpointError, ConnectionError, OSError) as e:
                 attempts += 1

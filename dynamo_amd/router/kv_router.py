@@ -20,6 +20,7 @@ import logging
 import math
 import random
 import time
+from collections import OrderedDict
 from dataclasses import dataclass, field
 from typing import Dict, List, Optional
 
@@ -40,6 +41,18 @@ class RouterConfig:
     prefill_load_scale: float = 1.0
     router_temperature: float = 0.0
     metrics_poll_interval: float = 1.0
+    # overload rejection (reference parity: http/service/busy_threshold.rs,
+    # discovery/worker_monitor.rs load thresholds): when EVERY worker's
+    # load exceeds the threshold, select() raises AllWorkersBusy and the
+    # frontend returns 503 instead of queueing.
+    busy_threshold: float = 0.0     # 0 = disabled
+    # sticky sessions (reference parity: lib/llm session_affinity/): cap on
+    # remembered session -> worker pins (LRU evicted beyond this).
+    max_sessions: int = 4096
+
+
+class AllWorkersBusy(Exception):
+    """Raised by select() when every worker is above busy_threshold."""
 
 
 @dataclass
@@ -67,6 +80,7 @@ class KvRouter:
         self.client = PushClient(runtime, namespace, component, endpoint)
         self.indexer = _core.KvIndexer()
         self.workers: Dict[str, WorkerState] = {}
+        self._sessions: "OrderedDict[str, str]" = OrderedDict()
         self._rr = 0
         self._tasks: List[asyncio.Task] = []
         self._started = False
@@ -145,11 +159,37 @@ class KvRouter:
             await asyncio.sleep(self.cfg.metrics_poll_interval)
 
     # -- selection ------------------------------------------------------
-    def select(self, token_ids: List[int]) -> Optional[str]:
-        """Pick a worker instance_id for this token sequence."""
+    def select(self, token_ids: List[int],
+               session_id: Optional[str] = None) -> Optional[str]:
+        """Pick a worker instance_id for this token sequence.
+
+        session_id pins a session to its previous worker while that worker
+        is alive (sticky sessions); busy_threshold > 0 rejects with
+        AllWorkersBusy when every worker is overloaded."""
         insts = self.client.instances()
         if not insts:
             return None
+        if self.cfg.busy_threshold > 0:
+            if all(self._load(i.instance_id) > self.cfg.busy_threshold
+                   for i in insts):
+                raise AllWorkersBusy(
+                    f"all {len(insts)} workers above busy threshold "
+                    f"{self.cfg.busy_threshold}")
+        if session_id is not None:
+            alive = {i.instance_id for i in insts}
+            pinned = self._sessions.get(session_id)
+            if pinned in alive:
+                self._sessions.move_to_end(session_id)
+                return pinned
+            iid = self._select_inner(insts, token_ids)
+            self._sessions[session_id] = iid
+            self._sessions.move_to_end(session_id)
+            while len(self._sessions) > self.cfg.max_sessions:
+                self._sessions.popitem(last=False)
+            return iid
+        return self._select_inner(insts, token_ids)
+
+    def _select_inner(self, insts, token_ids: List[int]) -> Optional[str]:
         mode = self.cfg.mode
         if mode == "round_robin":
             self._rr += 1

@@ -59,3 +59,55 @@ def test_multiprocess_workers_and_frontend(tmp_path):
             front.stop()
         for w in workers:
             w.stop()
+
+
+@pytest.mark.timeout(180)
+def test_record_and_replay(tmp_path):
+    """Record real HTTP traffic, then replay it with the replay tool."""
+    import json
+    import socket
+    import subprocess
+    disc = f"file:{tmp_path}/disc"
+    rec = tmp_path / "rec.jsonl"
+    workers, front = [], None
+    try:
+        workers.append(ManagedProcess(
+            worker_cmd(mock=True, model="tiny-llama", discovery=disc,
+                       page_size=16),
+            ready_marker="WORKER_READY").start())
+        with socket.socket() as s:
+            s.bind(("127.0.0.1", 0))
+            port = s.getsockname()[1]
+        front = ManagedProcess(
+            [sys.executable, "-m", "dynamo_amd.frontend", "--discovery", disc,
+             "--port", str(port), "--record", str(rec)],
+            ready_marker="FRONTEND_READY").start()
+        base = f"http://127.0.0.1:{port}"
+        with httpx.Client(timeout=30) as client:
+            deadline = time.time() + 60
+            while time.time() < deadline:
+                try:
+                    if client.get(base + "/health").json()["models"]:
+                        break
+                except httpx.TransportError:
+                    pass
+                time.sleep(0.3)
+            for i in range(3):
+                r = client.post(base + "/v1/completions", json={
+                    "model": "tiny-llama", "prompt": [10 + i, 20, 30],
+                    "max_tokens": 4})
+                assert r.status_code == 200
+        assert len(rec.read_text().splitlines()) == 3
+        out = subprocess.run(
+            [sys.executable, "-m", "dynamo_amd.tools.replay", str(rec),
+             "--url", base, "--speed", "100"],
+            capture_output=True, text=True, timeout=60)
+        assert out.returncode == 0, out.stderr
+        summary = json.loads(out.stdout.strip().splitlines()[-1])
+        assert summary["requests"] == 3 and summary["ok"] == 3
+        assert summary["output_tokens"] == 12
+    finally:
+        if front:
+            front.stop()
+        for w in workers:
+            w.stop()
