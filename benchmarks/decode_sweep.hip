@@ -54,17 +54,17 @@ static void run_mfma(const Bufs& bf) {
   const int lds = mfma_lds_bytes(G, HD);
   const int iters = 30;
   for (int i = 0; i < 5; i++)
-    paged_decode_mfma<G><<<grid, kBlock, lds>>>(
+    paged_decode_mfma<<<grid, kBlock, lds>>>(
         bf.partial, bf.ml, bf.out, bf.q, bf.kc, bf.vc, bf.pt, bf.ctx,
-        0.0883883f, B, Hkv, bf.C, CTX / PS, 6, HD);
+        0.0883883f, G, B, Hkv, bf.C, CTX / PS, 6, HD);
   CK(hipDeviceSynchronize());
   hipEvent_t e0, e1;
   CK(hipEventCreate(&e0)); CK(hipEventCreate(&e1));
   CK(hipEventRecord(e0));
   for (int i = 0; i < iters; i++) {
-    paged_decode_mfma<G><<<grid, kBlock, lds>>>(
+    paged_decode_mfma<<<grid, kBlock, lds>>>(
         bf.partial, bf.ml, bf.out, bf.q, bf.kc, bf.vc, bf.pt, bf.ctx,
-        0.0883883f, B, Hkv, bf.C, CTX / PS, 6, HD);
+        0.0883883f, G, B, Hkv, bf.C, CTX / PS, 6, HD);
     paged_decode_phase2<<<dim3(B, G * Hkv), 128>>>(
         bf.out, bf.partial, bf.ml, bf.ctx, G * Hkv, bf.C, HD);
   }

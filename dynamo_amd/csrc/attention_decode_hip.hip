@@ -51,15 +51,15 @@ void paged_attention_decode(torch::Tensor out, torch::Tensor q,
   do {                                                                        \
     if (mfma_lds_bytes(GG, hd) > 65536)                                       \
       (void)hipFuncSetAttribute(                                              \
-          reinterpret_cast<const void*>(&paged_decode_mfma<GG>),              \
+          reinterpret_cast<const void*>(&paged_decode_mfma),                  \
           hipFuncAttributeMaxDynamicSharedMemorySize,                         \
           mfma_lds_bytes(GG, hd));                                            \
   } while (0);                                                                \
- hipLaunchKernelGGL(( paged_decode_mfma<GG>), dim3(grid), dim3(kBlock), mfma_lds_bytes(GG, hd), stream,     \
+ hipLaunchKernelGGL(( paged_decode_mfma), dim3(grid), dim3(kBlock), mfma_lds_bytes(GG, hd), stream,         \
       partial.data_ptr<float>(), ml.data_ptr<float>(), (short*)out.data_ptr(),\
       (const short*)q.data_ptr(), (const short*)kcache.data_ptr(),            \
       (const short*)vcache.data_ptr(), page_table.data_ptr<int32_t>(),        \
-      ctx_lens.data_ptr<int32_t>(), (float)scale, B, Hkv, C, max_pages,       \
+      ctx_lens.data_ptr<int32_t>(), (float)scale, GG, B, Hkv, C, max_pages,   \
       log2_ps, hd)
   const bool mfma_ok = use_mfma && ps % 32 == 0 && hd == 128;
   switch (G) {  // combos picked by benchmarks/decode_sweep on MI355X
@@ -75,7 +75,13 @@ void paged_attention_decode(torch::Tensor out, torch::Tensor q,
       TORCH_CHECK(mfma_ok, "G=16 requires the MFMA decode path");
       LAUNCH_MFMA(16);
       break;
-    default: TORCH_CHECK(false, "unsupported GQA group size ", G);
+    default:
+      // odd groups (e.g. qwen2 G=7): the MFMA kernel takes G at runtime
+      TORCH_CHECK(mfma_ok && G <= 16,
+                  "GQA group ", G, " needs the MFMA decode path "
+                  "(page_size%32==0, head_dim==128)");
+      LAUNCH_MFMA(G);
+      break;
   }
 #undef LAUNCH_G
 #undef LAUNCH_MFMA

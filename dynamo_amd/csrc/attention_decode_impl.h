@@ -277,14 +277,16 @@ __global__ __launch_bounds__(kBlock) void paged_decode_phase1(
 //         (slab-permuted 4x16 blocks), so B-frags cost 2 tr-reads each.
 // Softmax runs on the C/D layout: row = head (4 per lane), col = token
 // (16 lanes), reductions are 4 shfl per row per 16 tokens.
-// Requires page_size % 32 == 0 and hd == 128.
-template <int G>
+// Requires page_size % 32 == 0 and hd == 128. G (<= 16) is RUNTIME — the
+// group only appears in bounds/guards, never in register shapes, so one
+// kernel serves every GQA group (e.g. qwen2's G=7).
 __global__ __launch_bounds__(kBlock) void paged_decode_mfma(
     float* __restrict__ partial, float* __restrict__ ml,
     short* __restrict__ out, const short* __restrict__ q,
     const short* __restrict__ kcache, const short* __restrict__ vcache,
     const int32_t* __restrict__ page_table, const int32_t* __restrict__ ctx_lens,
-    float scale, int B, int Hkv, int C, int max_pages, int log2_ps, int hd) {
+    float scale, int G, int B, int Hkv, int C, int max_pages, int log2_ps,
+    int hd) {
   typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8_t;
   constexpr int kSlab = kChunk / 4;   // 128 tokens per wave
   const int b = blockIdx.x;

@@ -37,6 +37,8 @@ class ModelConfig:
     activation: str = "silu"       # silu | gelu
     norm: str = "rmsnorm"          # rmsnorm | layernorm
     learned_pos_emb: bool = False
+    # qwen2-style QKV bias
+    attn_bias: bool = False
 
     def to_dict(self):
         return dataclasses.asdict(self)
@@ -60,6 +62,11 @@ PRESETS = {
         name="llama-3-70b", arch="llama", hidden_size=8192,
         intermediate_size=28672, num_layers=80, num_q_heads=64, num_kv_heads=8,
         head_dim=128, vocab_size=128256, rope_theta=500000.0),
+    "qwen2-7b": ModelConfig(
+        name="qwen2-7b", arch="qwen2", hidden_size=3584,
+        intermediate_size=18944, num_layers=28, num_q_heads=28,
+        num_kv_heads=4, head_dim=128, vocab_size=152064,
+        max_position=131072, rope_theta=1000000.0, attn_bias=True),
     "mixtral-8x7b": ModelConfig(
         name="mixtral-8x7b", arch="mixtral", hidden_size=4096,
         intermediate_size=14336, num_layers=32, num_q_heads=32, num_kv_heads=8,
@@ -85,6 +92,16 @@ PRESETS = {
         num_layers=2, num_q_heads=4, num_kv_heads=4, head_dim=32,
         vocab_size=512, max_position=2048, activation="gelu",
         norm="layernorm", learned_pos_emb=True, tie_embeddings=True),
+    "tiny-qwen": ModelConfig(
+        name="tiny-qwen", arch="qwen2", hidden_size=224,
+        intermediate_size=448, num_layers=2, num_q_heads=7, num_kv_heads=1,
+        head_dim=32, vocab_size=512, max_position=2048, rope_theta=10000.0,
+        attn_bias=True),
+    "tiny-qwen-gpu": ModelConfig(
+        name="tiny-qwen-gpu", arch="qwen2", hidden_size=896,
+        intermediate_size=1024, num_layers=2, num_q_heads=7, num_kv_heads=1,
+        head_dim=128, vocab_size=1024, max_position=8192,
+        rope_theta=10000.0, attn_bias=True),
     "tiny-mixtral": ModelConfig(
         name="tiny-mixtral", arch="mixtral", hidden_size=128,
         intermediate_size=256, num_layers=2, num_q_heads=4, num_kv_heads=2,

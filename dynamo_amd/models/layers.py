@@ -167,11 +167,30 @@ class Attention(torch.nn.Module):
                     off += rows
         self.wo = init_sharded((D, cfg.num_q_heads * self.hd), device, dtype,
                                tp, 1)
+        # qwen2-style QKV bias (same RNG-replay sharding discipline)
+        self.bqkv = None
+        if getattr(cfg, "attn_bias", False):
+            self.bqkv, needs_b = _alloc((qkv_rows,), device, dtype)
+            if needs_b:
+                with torch.no_grad():
+                    off = 0
+                    for hf, hl, idx in ((cfg.num_q_heads, self.hq, tp.rank),
+                                        (cfg.num_kv_heads, self.hkv, kv_idx),
+                                        (cfg.num_kv_heads, self.hkv, kv_idx)):
+                        full = torch.empty(hf * self.hd, device=device,
+                                           dtype=dtype).normal_(0.0, 0.02)
+                        rows = hl * self.hd
+                        self.bqkv[off:off + rows].copy_(
+                            full[idx * rows:(idx + 1) * rows]
+                            if tp.size > 1 else full)
+                        off += rows
         self.lora = None  # set by dynamo_amd.lora.LoRAManager
 
     def forward(self, x, cos_sin, kcache, vcache, meta: AttnMetadata):
         T = x.shape[0]
         qkv = linear_lora(x, self.wqkv, self.lora, "qkv")
+        if self.bqkv is not None:
+            qkv = qkv + self.bqkv
         q, k, v = qkv.split([self.hq * self.hd, self.hkv * self.hd,
                              self.hkv * self.hd], dim=-1)
         q = q.contiguous()

@@ -205,3 +205,33 @@ def test_embedding_and_generation_batched():
         assert steps < 100
     assert toks == base
     assert vec is not None and len(vec) == 256
+
+
+def test_qwen2_cpu_path():
+    """qwen2 arch: llama structure + QKV bias, odd GQA group (7q/1kv)."""
+    cfg = EngineConfig(model=PRESETS["tiny-qwen"], device="cpu",
+                       max_num_seqs=4, max_batched_tokens=128,
+                       max_model_len=512, kv_pool_pages=64, page_size=16)
+    e1, e2 = LLMEngine(cfg, seed=3), LLMEngine(cfg, seed=3)
+    o1 = generate(e1, [list(range(30, 70))], max_tokens=6)
+    o2 = generate(e2, [list(range(30, 70))], max_tokens=6)
+    assert o1 == o2 and len(o1[0]) == 6
+    # bias actually participates: zeroing it moves the pooled hidden state
+    import torch
+
+    def embed(engine):
+        engine.add_request("emb", list(range(10, 50)),
+                           SamplingParams(embed=True))
+        vec = None
+        while engine.has_work():
+            for so in engine.step():
+                if so.finish_reason == "embed":
+                    vec = torch.tensor(so.embedding)
+        return vec
+
+    v_bias = embed(e1)
+    with torch.no_grad():
+        for layer in e2.runner.model.layers:
+            layer.attn.bqkv.zero_()
+    v_nobias = embed(e2)
+    assert not torch.allclose(v_bias, v_nobias, atol=1e-5)

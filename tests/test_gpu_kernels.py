@@ -96,6 +96,7 @@ def test_kv_cache_append():
     (8, [900, 1, 4096]),
     (1, [333]),
     (2, [63, 65]),
+    (7, [700, 45, 1025]),   # odd GQA group (qwen2) -> runtime-G MFMA path
 ])
 def test_paged_attention_decode(G, ctxs):
     torch.manual_seed(0)
