@@ -73,3 +73,18 @@ def test_gpu_long_context_decode():
     prompt = list(range(5)) * 300  # 1500 tokens
     out = generate(make_engine(), [prompt], max_tokens=4)
     assert len(out[0]) == 4
+
+
+def test_gpu_qwen2_odd_group():
+    """qwen2 on the native path: QKV bias + odd GQA group (7q/1kv) through
+    the runtime-G MFMA decode kernel and the 16x16 prefill kernel."""
+    cfg = EngineConfig(model=PRESETS["tiny-qwen-gpu"], device="cuda:0",
+                       max_num_seqs=8, max_batched_tokens=512,
+                       max_model_len=2048, page_size=64, kv_pool_pages=256)
+    e1 = LLMEngine(cfg, seed=5)
+    e2 = LLMEngine(cfg, seed=5)
+    prompts = [list(range(100, 200)), [7, 8, 9] * 30]
+    o1 = generate(e1, prompts, max_tokens=8)
+    o2 = generate(e2, prompts, max_tokens=8)
+    assert o1 == o2
+    assert all(len(o) == 8 for o in o1)
