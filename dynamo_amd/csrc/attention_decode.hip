@@ -75,8 +75,9 @@ void paged_attention_decode(torch::Tensor out, torch::Tensor q,
       LAUNCH_MFMA(16);
       break;
     default:
-      // odd groups (e.g. qwen2 G=7): the MFMA kernel takes G at runtime
-      TORCH_CHECK(mfma_ok && G <= 16,
+      // odd groups (e.g. qwen2 G=7): the runtime-G MFMA kernel is the ONLY
+      // handler, so the DYNAMO_DECODE_MFMA preference toggle is ignored
+      TORCH_CHECK(ps % 32 == 0 && hd == 128 && G <= 16,
                   "GQA group ", G, " needs the MFMA decode path "
                   "(page_size%32==0, head_dim==128)");
       LAUNCH_MFMA(G);
