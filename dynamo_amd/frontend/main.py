@@ -34,6 +34,9 @@ def build_parser():
     p.add_argument("--record", default=None, metavar="FILE",
                    help="record requests + response chunks to a JSONL file "
                         "(replay with python -m dynamo_amd.tools.replay)")
+    p.add_argument("--grpc-port", type=int, default=0,
+                   help="also serve the KServe v2 gRPC inference protocol "
+                        "on this port (0 = disabled)")
     return p
 
 
@@ -51,8 +54,16 @@ async def async_main(args):
     config = uvicorn.Config(app, host=args.host, port=args.port,
                             log_level="warning")
     server = uvicorn.Server(config)
+    grpc_server = None
+    if args.grpc_port:
+        from .kserve import make_grpc_server
+        grpc_server, bound = make_grpc_server(mgr, args.host, args.grpc_port)
+        await grpc_server.start()
+        print(f"GRPC_READY {args.host}:{bound}", flush=True)
     print(f"FRONTEND_READY http://{args.host}:{args.port}", flush=True)
     await server.serve()
+    if grpc_server is not None:
+        await grpc_server.stop(grace=1.0)
     await mgr.stop()
     await rt.shutdown()
 
