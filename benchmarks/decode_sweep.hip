@@ -56,7 +56,7 @@ static void run_mfma(const Bufs& bf) {
   for (int i = 0; i < 5; i++)
     paged_decode_mfma<<<grid, kBlock, lds>>>(
         bf.partial, bf.ml, bf.out, bf.q, bf.kc, bf.vc, bf.pt, bf.ctx,
-        0.0883883f, G, B, Hkv, bf.C, CTX / PS, 6, HD);
+        0.0883883f, kChunk, G, B, Hkv, bf.C, CTX / PS, 6, HD);
   CK(hipDeviceSynchronize());
   hipEvent_t e0, e1;
   CK(hipEventCreate(&e0)); CK(hipEventCreate(&e1));
@@ -64,9 +64,9 @@ static void run_mfma(const Bufs& bf) {
   for (int i = 0; i < iters; i++) {
     paged_decode_mfma<<<grid, kBlock, lds>>>(
         bf.partial, bf.ml, bf.out, bf.q, bf.kc, bf.vc, bf.pt, bf.ctx,
-        0.0883883f, G, B, Hkv, bf.C, CTX / PS, 6, HD);
+        0.0883883f, kChunk, G, B, Hkv, bf.C, CTX / PS, 6, HD);
     paged_decode_phase2<<<dim3(B, G * Hkv), 128>>>(
-        bf.out, bf.partial, bf.ml, bf.ctx, G * Hkv, bf.C, HD);
+        bf.out, bf.partial, bf.ml, bf.ctx, kChunk, G * Hkv, bf.C, HD);
   }
   CK(hipEventRecord(e1));
   CK(hipEventSynchronize(e1));
@@ -87,7 +87,7 @@ static void run(const Bufs& bf, const char* tag) {
   for (int i = 0; i < 5; i++)
     paged_decode_phase1<G, DP, HS, DEPTH><<<grid, kBlock, lds>>>(
         bf.partial, bf.ml, bf.out, bf.q, bf.kc, bf.vc, bf.pt, bf.ctx,
-        0.0883883f, B, Hkv, bf.C, CTX / PS, 6, HD);
+        0.0883883f, kChunk, B, Hkv, bf.C, CTX / PS, 6, HD);
   CK(hipDeviceSynchronize());
   hipEvent_t e0, e1;
   CK(hipEventCreate(&e0)); CK(hipEventCreate(&e1));
@@ -95,9 +95,9 @@ static void run(const Bufs& bf, const char* tag) {
   for (int i = 0; i < iters; i++) {
     paged_decode_phase1<G, DP, HS, DEPTH><<<grid, kBlock, lds>>>(
         bf.partial, bf.ml, bf.out, bf.q, bf.kc, bf.vc, bf.pt, bf.ctx,
-        0.0883883f, B, Hkv, bf.C, CTX / PS, 6, HD);
+        0.0883883f, kChunk, B, Hkv, bf.C, CTX / PS, 6, HD);
     paged_decode_phase2<<<dim3(B, G * Hkv), 128>>>(
-        bf.out, bf.partial, bf.ml, bf.ctx, G * Hkv, bf.C, HD);
+        bf.out, bf.partial, bf.ml, bf.ctx, kChunk, G * Hkv, bf.C, HD);
   }
   CK(hipEventRecord(e1));
   CK(hipEventSynchronize(e1));

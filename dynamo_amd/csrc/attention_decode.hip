@@ -9,14 +9,19 @@
 using namespace decode_attn;
 
 int64_t paged_decode_num_chunks(int64_t max_ctx) {
-  return std::max<int64_t>(1, (max_ctx + kChunk - 1) / kChunk);
+  const int chunk = decode_chunk_tokens((int)max_ctx);
+  return std::max<int64_t>(1, (max_ctx + chunk - 1) / chunk);
+}
+
+int64_t decode_chunk_tokens_py(int64_t max_ctx) {
+  return decode_chunk_tokens((int)max_ctx);
 }
 
 void paged_attention_decode(torch::Tensor out, torch::Tensor q,
                             torch::Tensor kcache, torch::Tensor vcache,
                             torch::Tensor page_table, torch::Tensor ctx_lens,
                             torch::Tensor partial, torch::Tensor ml,
-                            double scale) {
+                            double scale, int64_t chunk_tokens) {
   TORCH_CHECK(q.is_cuda() && q.dtype() == torch::kBFloat16);
   TORCH_CHECK(page_table.dtype() == torch::kInt32 && ctx_lens.dtype() == torch::kInt32);
   const int B = q.size(0);
@@ -31,6 +36,8 @@ void paged_attention_decode(torch::Tensor out, torch::Tensor q,
   TORCH_CHECK((ps & (ps - 1)) == 0, "page_size must be a power of 2");
   int log2_ps = 0; while ((1 << log2_ps) < ps) log2_ps++;
   if (B == 0) return;
+  const int chunk = (int)chunk_tokens;
+  TORCH_CHECK(chunk >= 128 && chunk % 128 == 0, "bad chunk_tokens ", chunk);
   auto stream = at::cuda::getCurrentHIPStream();
 
   dim3 grid(B, Hkv, C);
@@ -40,8 +47,8 @@ void paged_attention_decode(torch::Tensor out, torch::Tensor q,
       partial.data_ptr<float>(), ml.data_ptr<float>(), (short*)out.data_ptr(),\
       (const short*)q.data_ptr(), (const short*)kcache.data_ptr(),            \
       (const short*)vcache.data_ptr(), page_table.data_ptr<int32_t>(),        \
-      ctx_lens.data_ptr<int32_t>(), (float)scale, B, Hkv, C, max_pages,       \
-      log2_ps, hd)
+      ctx_lens.data_ptr<int32_t>(), (float)scale, chunk, B, Hkv, C,            \
+      max_pages, log2_ps, hd)
   static const bool use_mfma = [] {
     const char* e = getenv("DYNAMO_DECODE_MFMA");
     return e == nullptr || e[0] != '0';  // default ON (sweep-verified)
@@ -58,8 +65,8 @@ void paged_attention_decode(torch::Tensor out, torch::Tensor q,
       partial.data_ptr<float>(), ml.data_ptr<float>(), (short*)out.data_ptr(),\
       (const short*)q.data_ptr(), (const short*)kcache.data_ptr(),            \
       (const short*)vcache.data_ptr(), page_table.data_ptr<int32_t>(),        \
-      ctx_lens.data_ptr<int32_t>(), (float)scale, GG, B, Hkv, C, max_pages,   \
-      log2_ps, hd)
+      ctx_lens.data_ptr<int32_t>(), (float)scale, chunk, GG, B, Hkv, C,        \
+      max_pages, log2_ps, hd)
   const bool mfma_ok = use_mfma && ps % 32 == 0 && hd == 128;
   switch (G) {  // combos picked by benchmarks/decode_sweep on MI355X
     case 1: LAUNCH_G(1, 8, 1, 2); break;
@@ -90,7 +97,7 @@ void paged_attention_decode(torch::Tensor out, torch::Tensor q,
     dim3 grid2(B, Hq);
     paged_decode_phase2<<<grid2, 128, 0, stream>>>(
         (short*)out.data_ptr(), partial.data_ptr<float>(), ml.data_ptr<float>(),
-        ctx_lens.data_ptr<int32_t>(), Hq, C, hd);
+        ctx_lens.data_ptr<int32_t>(), chunk, Hq, C, hd);
     HIP_CHECK_KERNEL();
   }
 }

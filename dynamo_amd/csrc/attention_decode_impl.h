@@ -16,7 +16,21 @@
 namespace decode_attn {
 
 constexpr int kBlock = 256;        // 4 waves
-constexpr int kChunk = 512;        // context tokens per workgroup
+
+// Context tokens per workgroup ("chunk"). RUNTIME-chosen per engine from
+// the max context: big chunks amortize per-tile overhead and shrink the
+// phase-2 merge (sweep: G8 MFMA 2982 -> 3344 GB/s from 512 -> 2048 at
+// ctx 8192), but the grid is B*Hkv*C blocks, so short contexts need small
+// chunks to fill 256 CUs. Python mirrors this via the binding.
+inline int decode_chunk_tokens(int max_ctx) {
+  if (max_ctx >= 8192) return 2048;
+  if (max_ctx >= 2048) return 1024;
+  return 512;
+}
+#ifndef DECODE_KCHUNK
+#define DECODE_KCHUNK 512
+#endif
+constexpr int kChunk = DECODE_KCHUNK;  // sweep-tool default (see above)
 constexpr float kNegInf = -1e30f;
 
 template <int G, int DP, int HS, int DEPTH>
@@ -29,19 +43,20 @@ __global__ __launch_bounds__(kBlock) void paged_decode_phase1(
     const short* __restrict__ vcache,
     const int32_t* __restrict__ page_table,  // [B, max_pages]
     const int32_t* __restrict__ ctx_lens,    // [B]
-    float scale, int B, int Hkv, int C, int max_pages, int log2_ps, int hd) {
+    float scale, int chunk, int B, int Hkv, int C, int max_pages,
+    int log2_ps, int hd) {
   constexpr int ND = 128 / DP;       // dims per lane
   constexpr int NV8 = ND / 8;        // short8 loads per row slice
   constexpr int TS = 64 / DP;        // tokens per wave step
   constexpr int GW = G / HS;         // heads per wave
   constexpr int TG = 4 / HS;         // token-groups (waves splitting tokens)
-  constexpr int kSlab = kChunk / TG;
+  const int kSlab = chunk / TG;
   const int b = blockIdx.x;
   const int h = blockIdx.y;   // kv head
   const int c = blockIdx.z;   // context chunk
   const int Hq = Hkv * G;
   const int ctx = ctx_lens[b];
-  const int chunk_start = c * kChunk;
+  const int chunk_start = c * chunk;
   const int lane = threadIdx.x & 63;
   const int wid = threadIdx.x >> 6;
   const int tg = wid / HS;        // token-group index
@@ -285,16 +300,16 @@ __global__ __launch_bounds__(kBlock) void paged_decode_mfma(
     short* __restrict__ out, const short* __restrict__ q,
     const short* __restrict__ kcache, const short* __restrict__ vcache,
     const int32_t* __restrict__ page_table, const int32_t* __restrict__ ctx_lens,
-    float scale, int G, int B, int Hkv, int C, int max_pages, int log2_ps,
-    int hd) {
+    float scale, int chunk, int G, int B, int Hkv, int C, int max_pages,
+    int log2_ps, int hd) {
   typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8_t;
-  constexpr int kSlab = kChunk / 4;   // 128 tokens per wave
+  const int kSlab = chunk / 4;        // tokens per wave
   const int b = blockIdx.x;
   const int h = blockIdx.y;
   const int c = blockIdx.z;
   const int Hq = Hkv * G;
   const int ctx = ctx_lens[b];
-  const int chunk_start = c * kChunk;
+  const int chunk_start = c * chunk;
   const int lane = threadIdx.x & 63;
   const int wid = threadIdx.x >> 6;
   const int lr = lane & 15;       // col lane (token / dim-subtile index)
@@ -548,10 +563,10 @@ __global__ inline void paged_decode_phase2(short* __restrict__ out,
                                            const float* __restrict__ partial,
                                            const float* __restrict__ ml,
                                            const int32_t* __restrict__ ctx_lens,
-                                           int Hq, int C, int hd) {
+                                           int chunk, int Hq, int C, int hd) {
   const int b = blockIdx.x;
   const int qh = blockIdx.y;
-  const int nc = min(C, (ctx_lens[b] + kChunk - 1) / kChunk);
+  const int nc = min(C, (ctx_lens[b] + chunk - 1) / chunk);
   const float* mlp = ml + (((int64_t)b * Hq + qh) * C) * 2;
 
   __shared__ float smax[1];

@@ -23,10 +23,12 @@ void scatter_pages(torch::Tensor staging, torch::Tensor cache, torch::Tensor pag
 void copy_pages(torch::Tensor dst_cache, torch::Tensor src_cache, torch::Tensor pairs);
 // attention_decode.hip
 int64_t paged_decode_num_chunks(int64_t max_ctx);
+int64_t decode_chunk_tokens_py(int64_t max_ctx);
 void paged_attention_decode(torch::Tensor out, torch::Tensor q, torch::Tensor kcache,
                             torch::Tensor vcache, torch::Tensor page_table,
                             torch::Tensor ctx_lens, torch::Tensor partial,
-                            torch::Tensor ml, double scale);
+                            torch::Tensor ml, double scale,
+                            int64_t chunk_tokens);
 // attention_prefill.hip
 void attention_prefill_paged(torch::Tensor out, torch::Tensor q,
                              torch::Tensor kcache, torch::Tensor vcache,
@@ -70,6 +72,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("scatter_pages", &scatter_pages);
   m.def("copy_pages", &copy_pages);
   m.def("paged_decode_num_chunks", &paged_decode_num_chunks);
+  m.def("decode_chunk_tokens", &decode_chunk_tokens_py);
   m.def("paged_attention_decode", &paged_attention_decode);
   m.def("attention_prefill_paged", &attention_prefill_paged);
   m.def("greedy_sample", &greedy_sample);

@@ -98,6 +98,7 @@ class DecodeScratch:
     def __init__(self, max_batch: int, num_q_heads: int, head_dim: int,
                  max_ctx: int, device):
         self.chunks = int(hip().paged_decode_num_chunks(max_ctx))
+        self.chunk_tokens = int(hip().decode_chunk_tokens(max_ctx))
         self.partial = torch.empty(max_batch, num_q_heads, self.chunks, head_dim,
                                    dtype=torch.float32, device=device)
         self.ml = torch.empty(max_batch, num_q_heads, self.chunks, 2,
@@ -114,7 +115,8 @@ def paged_attention_decode(q, kcache, vcache, page_table, ctx_lens, scale,
         assert scratch is not None, "GPU decode needs a DecodeScratch"
         partial, ml = scratch.view(q.shape[0])
         hip().paged_attention_decode(out, q, kcache, vcache, page_table,
-                                     ctx_lens, partial, ml, scale)
+                                     ctx_lens, partial, ml, scale,
+                                     scratch.chunk_tokens)
         return out
     return torch_ref.paged_attention_decode(q, kcache, vcache, page_table,
                                             ctx_lens, scale)
