@@ -97,6 +97,7 @@ class WorkerService:
         self.comp.serve_endpoint("load_lora", self.load_lora)
         self.comp.serve_endpoint("unload_lora", self.unload_lora)
         self.comp.serve_endpoint("list_loras", self.list_loras)
+        self.comp.serve_endpoint("update_weights", self.update_weights)
         metadata = {"worker_type": self.worker_type}
         if self.engine.runner.kv_pool is not None:
             metadata["kv_transfer"] = pool_transfer_metadata(
@@ -302,6 +303,44 @@ class WorkerService:
         self._paused.set()
         self._work.set()
         yield {"status": "running"}
+
+    # -- RL weight-update surface (reference: lib/rl/src/lib.rs:4-16 —
+    # frontend discovers workers' rl admin endpoints for weight pushes).
+    # Applies a deterministic in-place delta (seeded noise scaled by
+    # `scale`, modeling an RL policy push without checkpoint files),
+    # invalidates captured graphs and flushes the KV/prefix cache (old
+    # cache entries were computed under the old weights).
+    async def update_weights(self, payload, ctx):
+        seed = int(payload.get("seed", 0))
+        scale = float(payload.get("scale", 0.01))
+        async with self._engine_lock:
+            n = await asyncio.to_thread(self._apply_weight_delta, seed, scale)
+            self.engine.clear_kv()
+            self.engine._invalidate_graphs()
+        yield {"status": "ok", "tensors_updated": n, "seed": seed}
+
+    def _apply_weight_delta(self, seed: int, scale: float) -> int:
+        import torch
+        model = getattr(self.engine.runner, "model", None)
+        if model is None:
+            return 0  # mock engine: nothing to update
+        count = 0
+        seen = set()
+        with torch.no_grad():
+            for mod in model.modules():
+                for name, val in sorted(vars(mod).items()):
+                    if (isinstance(val, torch.Tensor)
+                            and val.is_floating_point() and val.numel()
+                            and name != "cos_sin"       # rope table, not a weight
+                            and id(val) not in seen):   # tied tensors once
+                        seen.add(id(val))
+                        g = torch.Generator(device="cpu")
+                        g.manual_seed(seed * 1000003 + count)
+                        noise = torch.randn(val.shape, generator=g,
+                                            dtype=torch.float32)
+                        val.add_(noise.to(val.device, val.dtype), alpha=scale)
+                        count += 1
+        return count
 
     # -- LoRA endpoints (reference: vllm/worker_factory.py:1378-1413) ----
     async def load_lora(self, payload, ctx):
