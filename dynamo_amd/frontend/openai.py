@@ -66,6 +66,22 @@ class ChatRequest(BaseModel):
     user: Optional[str] = None      # sticky-session key
 
 
+class AnthropicMessagesRequest(BaseModel):
+    """Anthropic Messages API shape (reference parity:
+    lib/llm/src/http/service/anthropic.rs /v1/messages)."""
+    model: str = ""
+    max_tokens: int = 128
+    messages: List[ChatMessage] = Field(default_factory=list)
+    system: Optional[str] = None
+    temperature: float = 0.0
+    top_p: float = 1.0
+    top_k: int = 0
+    stream: bool = False
+    ignore_eos: bool = False
+    seed: int = 0
+    user: Optional[str] = None
+
+
 def _session_of(req, raw: Request) -> Optional[str]:
     return raw.headers.get("x-session-id") or getattr(req, "user", None)
 
@@ -181,6 +197,80 @@ def build_app(manager: ModelManager) -> FastAPI:
             "usage": {"prompt_tokens": len(token_ids),
                       "completion_tokens": len(produced),
                       "total_tokens": len(token_ids) + len(produced)},
+        }
+
+    @app.post("/v1/messages")
+    async def anthropic_messages(req: AnthropicMessagesRequest, raw: Request):
+        """Anthropic Messages API over the same engine pipeline."""
+        entry = _entry_or_404(req.model)
+        REQS.labels(entry.name, "messages").inc()
+        msgs = [m.model_dump() for m in req.messages]
+        if req.system:
+            msgs = [{"role": "system", "content": req.system}] + msgs
+        prompt = entry.templater.render(msgs)
+        token_ids = entry.tokenizer.encode(prompt)
+        rid = f"msg_{uuid.uuid4().hex[:24]}"
+        t0 = time.time()
+
+        if req.stream:
+            async def sse():
+                yield ("event: message_start\ndata: " + json.dumps(
+                    {"type": "message_start", "message": {
+                        "id": rid, "type": "message", "role": "assistant",
+                        "model": entry.name, "content": [],
+                        "usage": {"input_tokens": len(token_ids)}}}) + "\n\n")
+                yield ("event: content_block_start\ndata: " + json.dumps(
+                    {"type": "content_block_start", "index": 0,
+                     "content_block": {"type": "text", "text": ""}}) + "\n\n")
+                produced: List[int] = []
+                finish = None
+                try:
+                    async for chunk in _run(entry, token_ids, req, rid,
+                                            _session_of(req, raw)):
+                        if await raw.is_disconnected():
+                            break
+                        prev = len(produced)
+                        produced.extend(chunk.get("token_ids", []))
+                        finish = chunk.get("finish_reason") or finish
+                        text = entry.tokenizer.decode_incremental(produced,
+                                                                  prev)
+                        if text:
+                            yield ("event: content_block_delta\ndata: "
+                                   + json.dumps(
+                                       {"type": "content_block_delta",
+                                        "index": 0, "delta": {
+                                            "type": "text_delta",
+                                            "text": text}}) + "\n\n")
+                    yield ("event: content_block_stop\ndata: " + json.dumps(
+                        {"type": "content_block_stop", "index": 0}) + "\n\n")
+                    stop = ("max_tokens" if finish == "length"
+                            else "end_turn")
+                    yield ("event: message_delta\ndata: " + json.dumps(
+                        {"type": "message_delta",
+                         "delta": {"stop_reason": stop},
+                         "usage": {"output_tokens": len(produced)}}) + "\n\n")
+                    yield ("event: message_stop\ndata: " + json.dumps(
+                        {"type": "message_stop"}) + "\n\n")
+                finally:
+                    LATENCY.labels(entry.name).observe(time.time() - t0)
+            return StreamingResponse(sse(), media_type="text/event-stream")
+
+        produced: List[int] = []
+        finish = None
+        async for chunk in _run(entry, token_ids, req, rid,
+                                _session_of(req, raw)):
+            produced.extend(chunk.get("token_ids", []))
+            finish = chunk.get("finish_reason") or finish
+        LATENCY.labels(entry.name).observe(time.time() - t0)
+        return {
+            "id": rid, "type": "message", "role": "assistant",
+            "model": entry.name,
+            "content": [{"type": "text",
+                         "text": entry.tokenizer.decode(produced)}],
+            "stop_reason": ("max_tokens" if finish == "length"
+                            else "end_turn"),
+            "usage": {"input_tokens": len(token_ids),
+                      "output_tokens": len(produced)},
         }
 
     @app.post("/v1/embeddings")

@@ -21,8 +21,12 @@ from .engine.scheduler import SchedulerOutput
 
 
 def _token_for(req_id: str, pos: int, vocab: int) -> int:
+    # stay in [3, min(vocab,256)): plain byte ids that decode to text and
+    # can never collide with the byte tokenizer's specials (eos_id 257) —
+    # a mock stream must not trigger an accidental eos stop
+    hi = min(vocab, 256)
     h = hashlib.blake2b(f"{req_id}:{pos}".encode(), digest_size=8).digest()
-    return int.from_bytes(h, "little") % vocab
+    return 3 + int.from_bytes(h, "little") % (hi - 3)
 
 
 class MockRunner:

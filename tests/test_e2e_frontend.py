@@ -427,3 +427,32 @@ def test_rl_update_weights():
         await rt.shutdown(drain=False)
         await mgr_rt.shutdown(drain=False)
     run(main())
+
+
+def test_anthropic_messages_route():
+    async def main():
+        shared, services, mgr, client = await with_stack(nworkers=1)
+        r = await client.post("/v1/messages", json={
+            "model": "mock-model", "max_tokens": 5,
+            "system": "be brief",
+            "messages": [{"role": "user", "content": "hello"}]})
+        assert r.status_code == 200, r.text
+        body = r.json()
+        assert body["type"] == "message" and body["role"] == "assistant"
+        assert body["content"][0]["type"] == "text"
+        assert body["stop_reason"] == "max_tokens"
+        assert body["usage"]["output_tokens"] == 5
+        # streaming event protocol
+        events = []
+        async with client.stream("POST", "/v1/messages", json={
+                "model": "mock-model", "max_tokens": 3, "stream": True,
+                "messages": [{"role": "user", "content": "hi"}]}) as rs:
+            assert rs.status_code == 200
+            async for line in rs.aiter_lines():
+                if line.startswith("event: "):
+                    events.append(line[7:])
+        assert events[0] == "message_start"
+        assert "content_block_delta" in events
+        assert events[-1] == "message_stop"
+        await teardown(services, mgr, client)
+    run(main())
