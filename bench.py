@@ -75,9 +75,14 @@ def make_prompts(args, mc, n, seed):
 
 
 def emit(args, world, value, ms_per_step, ttft_p50, prefill_time, mode, conc):
+    # BASELINE.json names the Llama-3-70B config; non-default models keep
+    # the same metric shape but say what was actually run
+    names = {"llama-3-70b": "Llama-3-70B", "llama-3-8b": "Llama-3-8B",
+             "mixtral-8x7b": "Mixtral-8x7B", "qwen2-7b": "Qwen2-7B"}
     result = {
-        "metric": "output tok/s (node), Llama-3-70B serving, "
-                  "ISL8192/OSL1024, conc 16/GPU",
+        "metric": (f"output tok/s (node), "
+                   f"{names.get(args.model, args.model)} serving, "
+                   f"ISL{args.isl}/OSL{args.osl}, conc {conc // world}/GPU"),
         "value": round(value, 2),
         "unit": "tok/s",
         "n_gpus": world,
