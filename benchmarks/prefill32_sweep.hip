@@ -315,37 +315,29 @@ __global__ __launch_bounds__(GW * 64) void prefill32_kernel(
     }
 
     // ---- P -> A-fragments: pa[ks] = P[q=lo][16ks + 8hi + i] -------------
-    // tok T lives at reg (T32&3) + 4*(T32>>3) + 16*(T>>5) on lane-half
-    // (T32>>2)&1 (T32 = T&31). Each half needs 4 of its 8 slot-values from
-    // the partner; exchange symmetrically: lane sends exactly what its
-    // partner needs (lo sends toks 16ks+8..11, hi sends 16ks+4..7).
+    // cvt_pk + permlane32_swap (guide rung): pack pairs of P values to
+    // bf16x2 with v_cvt_pk_bf16_f32, then ONE permlane32_swap delivers
+    // word0 to this lane and word2's source from the partner (and the
+    // second swap words 1/3) — 16 cvt_pk + 8 permlane replaces 16 shfl +
+    // 32 scalar converts. Derivation: slot ks's 8 own values live at regs
+    // b..b+7 with b = 8*(ks&1)+16*(ks>>1); swap(pack(p[b],p[b+1]),
+    // pack(p[b+4],p[b+5])) returns (word0, word2) on BOTH halves.
+    auto cvtpk = [](float a, float b) {
+      unsigned int r;
+      asm("v_cvt_pk_bf16_f32 %0, %1, %2" : "=v"(r) : "v"(a), "v"(b));
+      return r;
+    };
     bf16x8 pa[4];
 #pragma unroll
     for (int ks = 0; ks < 4; ks++) {
-      float recv[4];
-#pragma unroll
-      for (int j = 0; j < 4; j++) {
-        const int Ts = ks * 16 + (hi ? 4 + j : 8 + j);
-        const int T32 = Ts & 31;
-        const float send = p[(T32 & 3) + 4 * (T32 >> 3) + 16 * (Ts >> 5)];
-        recv[j] = __shfl_xor(send, 32, 64);
-      }
-#pragma unroll
-      for (int i = 0; i < 8; i++) {
-        float val;
-        if (i < 4) {
-          const int T = ks * 16 + (hi ? 0 : 0) + i;     // lo own / hi recv
-          const int T32 = T & 31;
-          val = hi ? recv[i]
-                   : p[(T32 & 3) + 4 * (T32 >> 3) + 16 * (T >> 5)];
-        } else {
-          const int T = ks * 16 + 8 + i;                // hi own (12..15)
-          const int T32 = T & 31;
-          val = hi ? p[(T32 & 3) + 4 * (T32 >> 3) + 16 * (T >> 5)]
-                   : recv[i - 4];
-        }
-        pa[ks][i] = (__bf16)val;
-      }
+      const int b0 = 8 * (ks & 1) + 16 * (ks >> 1);
+      uint2_t rA = __builtin_amdgcn_permlane32_swap(
+          cvtpk(p[b0], p[b0 + 1]), cvtpk(p[b0 + 4], p[b0 + 5]), false, false);
+      uint2_t rB = __builtin_amdgcn_permlane32_swap(
+          cvtpk(p[b0 + 2], p[b0 + 3]), cvtpk(p[b0 + 6], p[b0 + 7]),
+          false, false);
+      unsigned int w[4] = {rA.x, rB.x, rA.y, rB.y};
+      pa[ks] = *reinterpret_cast<bf16x8*>(w);
     }
 
     // ---- PV: O[q][d] += P[q][k] V[k][d] ---------------------------------

@@ -433,34 +433,31 @@ __global__ __launch_bounds__(512) void prefill32_kernel(
         for (int r = 0; r < 16; r++) o[dt][r] *= arow[r];
     }
 
-    // ---- P -> PV A-fragments via symmetric half-exchange ----
+    // ---- P -> PV A-fragments via cvt_pk + permlane32_swap ----
+    // Pack P pairs to bf16x2 with v_cvt_pk_bf16_f32, then one
+    // permlane32_swap yields (word0, word2) on both lane halves and a
+    // second (word1, word3): 16 cvt_pk + 8 permlane replaces 16 shfl +
+    // 32 scalar converts. Slot ks's own values sit at regs b..b+7,
+    // b = 8*(ks&1) + 16*(ks>>1). Measured 260 -> 569 TF standalone (the
+    // shfl/ds_bpermute chain was the VALU critical path;
+    // benchmarks/prefill32_sweep.hip).
+    typedef __attribute__((ext_vector_type(2))) unsigned int uint2_t;
+    auto cvtpk = [](float a, float b) {
+      unsigned int r;
+      asm("v_cvt_pk_bf16_f32 %0, %1, %2" : "=v"(r) : "v"(a), "v"(b));
+      return r;
+    };
     bf16x8_t pa[4];
 #pragma unroll
     for (int ks = 0; ks < 4; ks++) {
-      float recv[4];
-#pragma unroll
-      for (int j = 0; j < 4; j++) {
-        const int Ts = ks * 16 + (hi ? 4 + j : 8 + j);
-        const int T32 = Ts & 31;
-        const float send = p[(T32 & 3) + 4 * (T32 >> 3) + 16 * (Ts >> 5)];
-        recv[j] = __shfl_xor(send, 32, WAVE_SIZE);
-      }
-#pragma unroll
-      for (int i = 0; i < 8; i++) {
-        float val;
-        if (i < 4) {
-          const int T = ks * 16 + i;
-          const int T32 = T & 31;
-          val = hi ? recv[i]
-                   : p[(T32 & 3) + 4 * (T32 >> 3) + 16 * (T >> 5)];
-        } else {
-          const int T = ks * 16 + 8 + i;
-          const int T32 = T & 31;
-          val = hi ? p[(T32 & 3) + 4 * (T32 >> 3) + 16 * (T >> 5)]
-                   : recv[i - 4];
-        }
-        pa[ks][i] = (__bf16)val;
-      }
+      const int b0 = 8 * (ks & 1) + 16 * (ks >> 1);
+      uint2_t rA = __builtin_amdgcn_permlane32_swap(
+          cvtpk(p[b0], p[b0 + 1]), cvtpk(p[b0 + 4], p[b0 + 5]), false, false);
+      uint2_t rB = __builtin_amdgcn_permlane32_swap(
+          cvtpk(p[b0 + 2], p[b0 + 3]), cvtpk(p[b0 + 6], p[b0 + 7]),
+          false, false);
+      unsigned int w[4] = {rA.x, rB.x, rA.y, rB.y};
+      pa[ks] = *reinterpret_cast<bf16x8_t*>(w);
     }
 
     // ---- O += P V ----
