@@ -151,6 +151,11 @@ def build_app(manager: ModelManager) -> FastAPI:
         REQS.labels(entry.name, "completions").inc()
         if isinstance(req.prompt, list):
             token_ids = list(req.prompt)
+        elif len(req.prompt) > 4096:
+            # long prompts: tokenize off the event loop (reference parity:
+            # runtime/src/compute rayon pool for CPU-bound tokenization)
+            token_ids = await asyncio.to_thread(entry.tokenizer.encode,
+                                                req.prompt)
         else:
             token_ids = entry.tokenizer.encode(req.prompt)
         rid = f"cmpl-{uuid.uuid4().hex[:24]}"
@@ -208,7 +213,9 @@ def build_app(manager: ModelManager) -> FastAPI:
         if req.system:
             msgs = [{"role": "system", "content": req.system}] + msgs
         prompt = entry.templater.render(msgs)
-        token_ids = entry.tokenizer.encode(prompt)
+        token_ids = (await asyncio.to_thread(entry.tokenizer.encode, prompt)
+                     if len(prompt) > 4096 else
+                     entry.tokenizer.encode(prompt))
         rid = f"msg_{uuid.uuid4().hex[:24]}"
         t0 = time.time()
 
@@ -311,7 +318,9 @@ def build_app(manager: ModelManager) -> FastAPI:
         entry = _entry_or_404(req.model)
         REQS.labels(entry.name, "chat").inc()
         prompt = entry.templater.render([m.model_dump() for m in req.messages])
-        token_ids = entry.tokenizer.encode(prompt)
+        token_ids = (await asyncio.to_thread(entry.tokenizer.encode, prompt)
+                     if len(prompt) > 4096 else
+                     entry.tokenizer.encode(prompt))
         rid = f"chatcmpl-{uuid.uuid4().hex[:24]}"
         t0 = time.time()
 
