@@ -76,6 +76,9 @@ class ModelManager:
                 await e.router.stop()
             if e.prefill_router:
                 await e.prefill_router.stop()
+        if self._record_fh is not None:
+            self._record_fh.close()
+            self._record_fh = None
 
     async def _watch(self, interval: float):
         while True:
