@@ -443,11 +443,7 @@ __global__ __launch_bounds__(kBlock) void paged_decode_mfma(
     for (int r = 0; r < 4; r++) {
       float a = vA ? sA[r] * scale : kNegInf;
       float bb = vB ? sB[r] * scale : kNegInf;
-      float mt = fmaxf(a, bb);
-      mt = fmaxf(mt, __shfl_xor(mt, 1, WAVE_SIZE));
-      mt = fmaxf(mt, __shfl_xor(mt, 2, WAVE_SIZE));
-      mt = fmaxf(mt, __shfl_xor(mt, 4, WAVE_SIZE));
-      mt = fmaxf(mt, __shfl_xor(mt, 8, WAVE_SIZE));
+      float mt = row16_reduce_max(fmaxf(a, bb));
       if (mt > m[r]) {
         const float corr = (m[r] <= kNegInf * 0.5f) ? 0.f : __expf(m[r] - mt);
         l[r] *= corr;
@@ -457,12 +453,7 @@ __global__ __launch_bounds__(kBlock) void paged_decode_mfma(
       }
       pA[r] = (a > kNegInf * 0.5f) ? __expf(a - m[r]) : 0.f;
       pB[r] = (bb > kNegInf * 0.5f) ? __expf(bb - m[r]) : 0.f;
-      float rs = pA[r] + pB[r];
-      rs += __shfl_xor(rs, 1, WAVE_SIZE);
-      rs += __shfl_xor(rs, 2, WAVE_SIZE);
-      rs += __shfl_xor(rs, 4, WAVE_SIZE);
-      rs += __shfl_xor(rs, 8, WAVE_SIZE);
-      l[r] += rs;
+      l[r] += row16_reduce_sum(pA[r] + pB[r]);
     }
     // write P tile [16 heads][32 toks] (bank-spread via row XOR)
     if (active)

@@ -88,6 +88,34 @@ DEVINL float block_reduce_sum(float x, float* lds_scratch) {
 // ---- misc ---------------------------------------------------------------
 DEVINL int64_t cdiv64(int64_t a, int64_t b) { return (a + b - 1) / b; }
 
+// 16-lane (DPP row) reductions via row_ror rotations: pure VALU, no
+// ds_bpermute — cross-lane LDS-unit shuffles have ~60-cycle latency and
+// serialize softmax dependency chains (measured: replacing them in the
+// prefill kernel's P exchange took it 260 -> 569 TF).
+template <int CTRL>
+DEVINL float dpp_movf(float x) {
+  int i = __builtin_bit_cast(int, x);
+  i = __builtin_amdgcn_update_dpp(0, i, CTRL, 0xf, 0xf, false);
+  return __builtin_bit_cast(float, i);
+}
+
+// max over the 16 lanes of a DPP row, broadcast to every lane of the row
+DEVINL float row16_reduce_max(float x) {
+  x = fmaxf(x, dpp_movf<0x128>(x));   // row_ror:8
+  x = fmaxf(x, dpp_movf<0x124>(x));   // row_ror:4
+  x = fmaxf(x, dpp_movf<0x122>(x));   // row_ror:2
+  x = fmaxf(x, dpp_movf<0x121>(x));   // row_ror:1
+  return x;
+}
+
+DEVINL float row16_reduce_sum(float x) {
+  x += dpp_movf<0x128>(x);
+  x += dpp_movf<0x124>(x);
+  x += dpp_movf<0x122>(x);
+  x += dpp_movf<0x121>(x);
+  return x;
+}
+
 #define HIP_CHECK_KERNEL()                                                \
   do {                                                                    \
     hipError_t _e = hipGetLastError();                                    \
