@@ -117,6 +117,17 @@ static void permprobe() {
 #define KVBLK 64
 #define QBLK 32
 
+// cross-half (lane^32) exchange via ONE v_permlane32_swap instead of a
+// ds_bpermute shfl_xor: swap(x,x) gives the partner's x in r.y (lo half)
+// / r.x (hi half)
+__device__ __forceinline__ float xor32_swap(float x, int hi) {
+  typedef __attribute__((ext_vector_type(2))) unsigned int uint2_sw;
+  unsigned int u = __builtin_bit_cast(unsigned int, x);
+  uint2_sw r = __builtin_amdgcn_permlane32_swap(u, u, false, false);
+  return __builtin_bit_cast(float, hi ? r.x : r.y);
+}
+
+
 // VSTAGE: role-split staging (requires GW=8): threads 256-511 stage K with
 //   b128 writes; threads 0-255 stage V^T with b64 writes (4 tokens packed
 //   per write) — 4x fewer LDS write ops than scalar V^T stores.
@@ -282,7 +293,7 @@ __global__ __launch_bounds__(GW * 64) void prefill32_kernel(
       p[16 + r] = (t0 + 32 + trow <= qpos) ? s1[r] * scale : -1e30f;
       mt = fmaxf(mt, fmaxf(p[r], p[16 + r]));
     }
-    mt = fmaxf(mt, __shfl_xor(mt, 32, 64));     // combine lane halves
+    mt = fmaxf(mt, xor32_swap(mt, hi));         // combine lane halves
     float m_new = fmaxf(m_run, mt);
     bool skip_rescale = false;
     if constexpr (DEFER) {
@@ -297,7 +308,7 @@ __global__ __launch_bounds__(GW * 64) void prefill32_kernel(
       p[r] = __expf(p[r] - m_new);
       ls += p[r];
     }
-    ls += __shfl_xor(ls, 32, 64);
+    ls += xor32_swap(ls, hi);
     l_run = l_run * alpha + ls;
     m_run = m_new;
     if (!skip_rescale) {

@@ -254,6 +254,17 @@ namespace {
 typedef __attribute__((ext_vector_type(16))) float f32x16;
 typedef __attribute__((ext_vector_type(4))) short short4_t;
 
+
+// cross-half (lane^32) exchange via ONE v_permlane32_swap instead of a
+// ds_bpermute shfl_xor: swap(x,x) gives the partner's x in r.y (lo half)
+// / r.x (hi half)
+__device__ __forceinline__ float xor32_swap(float x, int hi) {
+  typedef __attribute__((ext_vector_type(2))) unsigned int uint2_sw;
+  unsigned int u = __builtin_bit_cast(unsigned int, x);
+  uint2_sw r = __builtin_amdgcn_permlane32_swap(u, u, false, false);
+  return __builtin_bit_cast(float, hi ? r.x : r.y);
+}
+
 constexpr int kQB32 = 32;   // q rows per block
 constexpr int kKB32 = 64;   // kv tokens per tile
 constexpr int kLdsHalf32 = kKB32 * 256 + 128 * 128;  // K + V^T = 32 KB
@@ -402,7 +413,7 @@ __global__ __launch_bounds__(512) void prefill32_kernel(
       p[16 + r] = ok1 ? s1[r] * scale : kNegInf;
       mt = fmaxf(mt, fmaxf(p[r], p[16 + r]));
     }
-    mt = fmaxf(mt, __shfl_xor(mt, 32, WAVE_SIZE));
+    mt = fmaxf(mt, xor32_swap(mt, hi));
     float m_new = fmaxf(m_run, mt);
     bool skip_rescale = false;
     // defer-max: skip the O-rescale while the tile max stays within 8
@@ -416,7 +427,7 @@ __global__ __launch_bounds__(512) void prefill32_kernel(
                  ? 0.f : __expf(p[r] - m_new);
       ls += p[r];
     }
-    ls += __shfl_xor(ls, 32, WAVE_SIZE);
+    ls += xor32_swap(ls, hi);
     l_run = l_run * alpha + ls;
     m_run = m_new;
     if (!skip_rescale) {
