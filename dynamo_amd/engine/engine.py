@@ -27,6 +27,7 @@ class StepOutput:
     finish_reason: Optional[str] = None
     num_output_tokens: int = 0
     embedding: Optional[List[float]] = None
+    logprobs: Optional[dict] = None   # {"token_logprob", "top"} if requested
 
 
 @dataclass
@@ -173,8 +174,9 @@ class LLMEngine:
             self.host_tier.fence()
         with roctx_range("decode_step"):
             logits = gr.step(self._last_sampled)
-        from .sampling import sample_tokens
+        from .sampling import compute_logprobs, sample_tokens
         sampled = sample_tokens(logits, running, self.step_count)
+        lps = compute_logprobs(logits, sampled, running)
         self._last_sampled = sampled
         toks = sampled.cpu().tolist()
 
@@ -201,7 +203,8 @@ class LLMEngine:
                     req.num_computed % self.cfg.page_size == 0:
                 req.kv.commit_full_pages(req.all_tokens, req.num_computed)
             outputs.append(StepOutput(req.req_id, int(tok), finished, reason,
-                                      len(req.output_tokens)))
+                                      len(req.output_tokens),
+                                      logprobs=lps[len(outputs)]))
         if finished_any:
             gr.dirty = True
         self.kv_events.extend(self.alloc.drain_events())
@@ -258,7 +261,9 @@ class LLMEngine:
                 else:
                     self._finish(req, reason)
             outputs.append(StepOutput(req.req_id, int(tok), finished, reason,
-                                      len(req.output_tokens)))
+                                      len(req.output_tokens),
+                                      logprobs=getattr(req, "_logprobs", None)))
+            req._logprobs = None
 
         # embedding requests: finish when the (chunked) prefill completes
         for ss in sched.prefills:

@@ -161,6 +161,9 @@ class ModelRunner:
             return torch.empty(0, dtype=torch.int32), []
         rows = hidden[meta.logits_rows]
         logits = self.model.compute_logits(rows)
-        from .sampling import sample_tokens
-        sampled = sample_tokens(logits, [s.req for s in sample_seqs], step_seed)
-        return sampled, [s.req for s in sample_seqs]
+        from .sampling import compute_logprobs, sample_tokens
+        reqs = [s.req for s in sample_seqs]
+        sampled = sample_tokens(logits, reqs, step_seed)
+        for r, lp in zip(reqs, compute_logprobs(logits, sampled, reqs)):
+            r._logprobs = lp
+        return sampled, reqs

@@ -27,6 +27,32 @@ def _filter_topk_topp(logits: torch.Tensor, top_k: int, top_p: float):
     return logits
 
 
+def compute_logprobs(logits: torch.Tensor, sampled: torch.Tensor,
+                     reqs: List[Request]):
+    """Per-request top-N logprobs (None for requests with logprobs=0).
+
+    Returns a list aligned with reqs: each entry is
+    {"token_logprob": float, "top": [[token_id, logprob], ...]} computed
+    from log_softmax of the full-vocab logits row."""
+    want = [r.sampling.logprobs for r in reqs]
+    if not any(want):
+        return [None] * len(reqs)
+    lp = torch.log_softmax(logits.float(), dim=-1)
+    kmax = min(max(want), lp.shape[-1])
+    topv, topi = torch.topk(lp, kmax, dim=-1)
+    own = lp.gather(-1, sampled.long().unsqueeze(-1)).squeeze(-1)
+    topv_l, topi_l, own_l = topv.tolist(), topi.tolist(), own.tolist()
+    out = []
+    for i, n in enumerate(want):
+        if not n:
+            out.append(None)
+            continue
+        out.append({"token_logprob": own_l[i],
+                    "top": [[topi_l[i][j], topv_l[i][j]]
+                            for j in range(min(n, kmax))]})
+    return out
+
+
 def sample_tokens(logits: torch.Tensor, reqs: List[Request],
                   step_seed: int) -> torch.Tensor:
     """logits [n, V] fp32 -> token ids [n] int32."""

@@ -31,6 +31,9 @@ class SamplingParams:
     # sampled tokens (reference parity: /v1/embeddings route,
     # lib/llm/src/http service embeddings handler)
     embed: bool = False
+    # top-N logprobs per generated token (0 = off); reference parity:
+    # LLMEngineOutput log_probs/top_logprobs (protocols/common/llm_backend.rs)
+    logprobs: int = 0
 
 
 class ReqState(Enum):

@@ -39,6 +39,7 @@ class CompletionRequest(BaseModel):
     stop: Optional[Union[str, List[str]]] = None
     seed: int = 0
     ignore_eos: bool = False
+    logprobs: Optional[int] = None  # top-N logprobs per token
     user: Optional[str] = None      # sticky-session key
 
 
@@ -110,7 +111,8 @@ def build_app(manager: ModelManager) -> FastAPI:
     async def _run(entry, token_ids, req, rid,
                    session_id=None) -> AsyncIterator[dict]:
         sampling = {"temperature": req.temperature, "top_p": req.top_p,
-                    "top_k": req.top_k, "seed": req.seed}
+                    "top_k": req.top_k, "seed": req.seed,
+                    "logprobs": getattr(req, "logprobs", None) or 0}
         eos = getattr(entry.tokenizer, "eos_id", None)
         stop = {"max_tokens": req.max_tokens,
                 "ignore_eos": req.ignore_eos,
@@ -187,17 +189,29 @@ def build_app(manager: ModelManager) -> FastAPI:
             return StreamingResponse(sse(), media_type="text/event-stream")
 
         produced: List[int] = []
+        lps: List[dict] = []
         finish = None
         async for chunk in _run(entry, token_ids, req, rid, _session_of(req, raw)):
             produced.extend(chunk.get("token_ids", []))
+            lps.extend(chunk.get("logprobs", []))
             finish = chunk.get("finish_reason") or finish
         LATENCY.labels(entry.name).observe(time.time() - t0)
+        logprobs_out = None
+        if req.logprobs and lps:
+            logprobs_out = {
+                "tokens": [entry.tokenizer.decode([t]) for t in produced],
+                "token_logprobs": [d["token_logprob"] for d in lps],
+                "top_logprobs": [
+                    {entry.tokenizer.decode([t]): v for t, v in d["top"]}
+                    for d in lps],
+            }
         return {
             "id": rid, "object": "text_completion", "created": int(t0),
             "model": entry.name,
             "choices": [{"index": 0,
                          "text": entry.tokenizer.decode(produced),
                          "finish_reason": finish or "stop",
+                         "logprobs": logprobs_out,
                          "token_ids": produced}],
             "usage": {"prompt_tokens": len(token_ids),
                       "completion_tokens": len(produced),

@@ -45,6 +45,7 @@ def make_sampling(payload: dict) -> SamplingParams:
         ignore_eos=bool(sc.get("ignore_eos", False)),
         seed=int(so.get("seed", 0)),
         embed=bool(so.get("embed", False)),
+        logprobs=int(so.get("logprobs", 0)),
     )
 
 
@@ -183,6 +184,8 @@ class WorkerService:
                                              else [])}
                 if so.embedding is not None:
                     chunk["embedding"] = so.embedding
+                if so.logprobs is not None:
+                    chunk["logprobs"] = [so.logprobs]
                 if so.finished:
                     chunk["finish_reason"] = so.finish_reason
                     if is_prefill_role:

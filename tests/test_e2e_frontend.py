@@ -456,3 +456,40 @@ def test_anthropic_messages_route():
         assert events[-1] == "message_stop"
         await teardown(services, mgr, client)
     run(main())
+
+
+def test_completions_logprobs_route():
+    """OpenAI logprobs in /v1/completions against a REAL tiny engine."""
+    from dynamo_amd.engine import EngineConfig, LLMEngine
+    from dynamo_amd.engine.config import PRESETS
+
+    async def main():
+        shared = MemoryDiscovery()
+        rt = DistributedRuntime(shared)
+        cfg = EngineConfig(model=PRESETS["tiny-llama"], device="cpu",
+                           max_num_seqs=4, max_batched_tokens=256,
+                           max_model_len=512, kv_pool_pages=64, page_size=16,
+                           enable_hip_graphs=False)
+        ws = WorkerService(LLMEngine(cfg, seed=7), rt)
+        await ws.start()
+        mgr_rt = DistributedRuntime(shared)
+        mgr = ModelManager(mgr_rt)
+        await mgr.start(watch_interval=0.2)
+        app = build_app(mgr)
+        client = httpx.AsyncClient(transport=httpx.ASGITransport(app=app),
+                                   base_url="http://t")
+        r = await client.post("/v1/completions", json={
+            "model": "tiny-llama", "prompt": [40, 41, 42], "max_tokens": 3,
+            "logprobs": 2})
+        assert r.status_code == 200, r.text
+        lp = r.json()["choices"][0]["logprobs"]
+        assert lp is not None
+        assert len(lp["token_logprobs"]) == 3
+        assert len(lp["top_logprobs"]) == 3 and len(lp["top_logprobs"][0]) <= 2
+        assert all(v <= 0 for v in lp["token_logprobs"])
+        await client.aclose()
+        await mgr.stop()
+        await ws.stop()
+        await rt.shutdown(drain=False)
+        await mgr_rt.shutdown(drain=False)
+    run(main())

@@ -235,3 +235,29 @@ def test_qwen2_cpu_path():
             layer.attn.bqkv.zero_()
     v_nobias = embed(e2)
     assert not torch.allclose(v_bias, v_nobias, atol=1e-5)
+
+
+def test_logprobs():
+    """Requested logprobs come back per generated token; the sampled
+    (greedy) token's logprob is the max and matches its top-1 entry."""
+    import math
+    eng = make_engine()
+    eng.add_request("r0", list(range(40, 80)),
+                    SamplingParams(max_tokens=4, logprobs=3))
+    eng.add_request("r1", list(range(10, 50)),
+                    SamplingParams(max_tokens=4))   # no logprobs requested
+    got = {"r0": [], "r1": []}
+    while eng.has_work():
+        for so in eng.step():
+            got[so.req_id].append((so.new_token, so.logprobs))
+    assert all(lp is None for _, lp in got["r1"])
+    assert len(got["r0"]) == 4
+    for tok, lp in got["r0"]:
+        assert lp is not None and len(lp["top"]) == 3
+        # greedy: sampled token is the argmax -> first top entry
+        assert lp["top"][0][0] == tok
+        assert abs(lp["top"][0][1] - lp["token_logprob"]) < 1e-5
+        assert lp["token_logprob"] <= 0.0
+        # top list sorted descending
+        assert lp["top"][0][1] >= lp["top"][1][1] >= lp["top"][2][1]
+        assert math.isfinite(lp["token_logprob"])

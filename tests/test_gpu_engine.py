@@ -88,3 +88,19 @@ def test_gpu_qwen2_odd_group():
     o2 = generate(e2, prompts, max_tokens=8)
     assert o1 == o2
     assert all(len(o) == 8 for o in o1)
+
+
+def test_gpu_logprobs_fast_path():
+    """logprobs through the hipGraph decode fast path."""
+    eng = make_engine()
+    eng.add_request("r", list(range(100, 180)),
+                    SamplingParams(max_tokens=4, logprobs=2))
+    got = []
+    while eng.has_work():
+        for so in eng.step():
+            got.append((so.new_token, so.logprobs))
+    assert len(got) == 4
+    for tok, lp in got:
+        assert lp is not None and len(lp["top"]) == 2
+        assert lp["top"][0][0] == tok        # greedy argmax is top-1
+        assert lp["token_logprob"] <= 0.0
