@@ -34,17 +34,20 @@ def pct(xs, p):
     return xs[i]
 
 
-async def one_request(client, args, prompt_tokens, stats):
+async def one_request(client, args, prompt_tokens, stats, session=None,
+                      collect=None):
     t0 = time.monotonic()
     first = None
     ntok = 0
     last = t0
     itls = []
+    body = {"model": args.model, "prompt": prompt_tokens,
+            "max_tokens": args.osl, "stream": True, "ignore_eos": True}
+    if session:
+        body["user"] = session    # sticky-session pin
     try:
-        async with client.stream("POST", f"{args.url}/v1/completions", json={
-                "model": args.model, "prompt": prompt_tokens,
-                "max_tokens": args.osl, "stream": True,
-                "ignore_eos": True}) as r:
+        async with client.stream("POST", f"{args.url}/v1/completions",
+                                 json=body) as r:
             if r.status_code != 200:
                 stats["errors"] += 1
                 return
@@ -65,6 +68,54 @@ async def one_request(client, args, prompt_tokens, stats):
     stats["itl"].extend(itls)
     stats["tokens"] += ntok
     stats["latency"].append(time.monotonic() - t0)
+    if collect is not None and first is not None:
+        collect.append(first - t0)
+
+
+async def run_multiturn(args):
+    """Multiturn conversations (reference parity: lib/bench multiturn_bench
+    TTFT benchmark): each conversation appends its history every turn and
+    pins to one worker via sticky sessions, so turns > 1 hit the prefix
+    cache and TTFT collapses to the new-chunk prefill."""
+    rng = random.Random(args.seed)
+    stats = {"ttft": [], "itl": [], "latency": [], "tokens": 0, "errors": 0}
+    per_turn = [[] for _ in range(args.turns)]
+    chunk = max(1, args.isl // args.turns)
+    limits = httpx.Limits(max_connections=args.concurrency + 8)
+    async with httpx.AsyncClient(timeout=None, limits=limits) as client:
+        sem = asyncio.Semaphore(args.concurrency)
+
+        async def conversation(i):
+            crng = random.Random(args.seed * 1000 + i)
+            history = []
+            async with sem:
+                for t in range(args.turns):
+                    history = history + [crng.randrange(args.vocab)
+                                         for _ in range(chunk)]
+                    await one_request(client, args, list(history), stats,
+                                      session=f"conv-{i}",
+                                      collect=per_turn[t])
+                    # the assistant's reply becomes part of the context
+                    history = history + [crng.randrange(args.vocab)
+                                         for _ in range(args.osl)]
+
+        t0 = time.monotonic()
+        await asyncio.gather(*[conversation(i)
+                               for i in range(args.requests)])
+        wall = time.monotonic() - t0
+
+    out = {
+        "conversations": args.requests, "turns": args.turns,
+        "errors": stats["errors"], "wall_s": round(wall, 3),
+        "output_tok_s": round(stats["tokens"] / wall, 2),
+        "ttft_p50_by_turn_s": [round(pct(t, 50), 4) if t else None
+                               for t in per_turn],
+        "itl_p50_ms": round(pct(stats["itl"], 50), 3) if stats["itl"] else None,
+        "config": {"isl_chunk": chunk, "osl": args.osl,
+                   "concurrency": args.concurrency},
+    }
+    print(json.dumps(out))
+    return out
 
 
 async def run(args):
@@ -134,12 +185,14 @@ def main():
     p.add_argument("--concurrency", type=int, default=16)
     p.add_argument("--requests", type=int, default=32)
     p.add_argument("--prefix-ratio", type=float, default=0.0)
+    p.add_argument("--turns", type=int, default=0,
+                   help="multiturn mode: turns per conversation (0 = off)")
     p.add_argument("--rate", type=float, default=0.0,
                    help=">0: open-loop Poisson req/s; 0: closed loop")
     p.add_argument("--vocab", type=int, default=512)
     p.add_argument("--seed", type=int, default=0)
     args = p.parse_args()
-    asyncio.run(run(args))
+    asyncio.run(run_multiturn(args) if args.turns > 0 else run(args))
 
 
 if __name__ == "__main__":

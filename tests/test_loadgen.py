@@ -44,3 +44,49 @@ def test_loadgen_against_stack(tmp_path):
     finally:
         front.stop()
         w.stop()
+
+
+@pytest.mark.timeout(180)
+def test_loadgen_multiturn(tmp_path):
+    """Multiturn mode: per-turn TTFT report against the mock stack, with
+    sticky sessions (2 workers) so every turn lands on its pinned worker."""
+    import socket
+    disc = f"file:{tmp_path}/disc"
+    ws = [ManagedProcess(worker_cmd(mock=True, model="tiny-llama",
+                                    discovery=disc, page_size=16),
+                         ready_marker="WORKER_READY").start()
+          for _ in range(2)]
+    with socket.socket() as s:
+        s.bind(("127.0.0.1", 0))
+        port = s.getsockname()[1]
+    front = ManagedProcess(
+        [sys.executable, "-m", "dynamo_amd.frontend", "--discovery", disc,
+         "--port", str(port)], ready_marker="FRONTEND_READY").start()
+    try:
+        base = f"http://127.0.0.1:{port}"
+        deadline = time.time() + 60
+        with httpx.Client(timeout=10) as c:
+            while time.time() < deadline:
+                try:
+                    if c.get(base + "/health").json()["models"]:
+                        break
+                except httpx.TransportError:
+                    pass
+                time.sleep(0.3)
+        out = subprocess.run(
+            [sys.executable, "benchmarks/loadgen.py", "--url", base,
+             "--model", "tiny-llama", "--isl", "96", "--osl", "8",
+             "--concurrency", "4", "--requests", "4", "--turns", "3"],
+            capture_output=True, text=True, timeout=90)
+        assert out.returncode == 0, out.stderr
+        summary = json.loads(out.stdout.strip().splitlines()[-1])
+        assert summary["errors"] == 0
+        assert summary["conversations"] == 4 and summary["turns"] == 3
+        assert len(summary["ttft_p50_by_turn_s"]) == 3
+        assert all(v is not None for v in summary["ttft_p50_by_turn_s"])
+        # 4 convs x 3 turns x 8 output tokens
+        assert summary["output_tok_s"] > 0
+    finally:
+        front.stop()
+        for w in ws:
+            w.stop()
