@@ -90,3 +90,49 @@ def test_loadgen_multiturn(tmp_path):
         front.stop()
         for w in ws:
             w.stop()
+
+
+@pytest.mark.timeout(300)
+def test_profiler_sweep(tmp_path):
+    """SLA profiler: concurrency sweep -> planner PerfModel bootstrap."""
+    import socket
+    disc = f"file:{tmp_path}/disc"
+    w = ManagedProcess(worker_cmd(mock=True, model="tiny-llama",
+                                  discovery=disc, page_size=16),
+                       ready_marker="WORKER_READY").start()
+    with socket.socket() as s:
+        s.bind(("127.0.0.1", 0))
+        port = s.getsockname()[1]
+    front = ManagedProcess(
+        [sys.executable, "-m", "dynamo_amd.frontend", "--discovery", disc,
+         "--port", str(port)], ready_marker="FRONTEND_READY").start()
+    try:
+        base = f"http://127.0.0.1:{port}"
+        deadline = time.time() + 60
+        with httpx.Client(timeout=10) as c:
+            while time.time() < deadline:
+                try:
+                    if c.get(base + "/health").json()["models"]:
+                        break
+                except httpx.TransportError:
+                    pass
+                time.sleep(0.3)
+        from dynamo_amd.profiler import run_profile
+        res = run_profile(base, "tiny-llama", isl=64, osl=8,
+                          concurrencies=[1, 2], requests_per_level=4,
+                          itl_slo_ms=10_000.0, ttft_slo_s=60.0,
+                          out=str(tmp_path / "prof.json"))
+        assert len(res["sweep"]) == 2
+        assert res["meets_slo"] is True
+        pm = res["perf_model"]
+        assert pm["max_conc_at_itl"] == 2
+        assert pm["decode_tokens_per_s_at_itl"] > 0
+        assert (tmp_path / "prof.json").exists()
+        # the derived numbers plug straight into the planner's PerfModel
+        from dynamo_amd.planner.planner import PerfModel
+        PerfModel(prefill_tokens_per_s=pm["prefill_tokens_per_s"] or 1.0,
+                  decode_tokens_per_s_at_itl=pm["decode_tokens_per_s_at_itl"],
+                  max_conc_at_itl=pm["max_conc_at_itl"])
+    finally:
+        front.stop()
+        w.stop()
