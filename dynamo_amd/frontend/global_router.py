@@ -79,12 +79,39 @@ class GlobalRouter:
                  and (not model or model in p.models or not p.models)]
         return sorted(pools, key=lambda p: p.inflight)
 
+    # -- cross-pool KV awareness (kv_dc_relay parity-lite) ---------------
+    async def _kv_rank(self, pools: List[PoolState], model: str,
+                       token_ids) -> List[PoolState]:
+        """Re-rank candidate pools by prefix overlap: ask each pool's
+        /internal/kv_overlap digest (short timeout, best-effort) and sort
+        by (-overlap_blocks, inflight)."""
+        if not token_ids or len(pools) < 2:
+            return pools
+
+        async def ask(p: PoolState) -> int:
+            try:
+                r = await self.client.post(
+                    p.url + "/internal/kv_overlap",
+                    json={"model": model, "token_ids": list(token_ids)},
+                    timeout=0.25)
+                return int(r.json().get("overlap_blocks", 0))
+            except Exception:
+                return 0
+
+        overlaps = await asyncio.gather(*[ask(p) for p in pools])
+        ranked = sorted(zip(pools, overlaps),
+                        key=lambda po: (-po[1], po[0].inflight))
+        return [p for p, _ in ranked]
+
     # -- proxying --------------------------------------------------------
     async def proxy(self, path: str, payload: dict, stream: bool):
         model = payload.get("model", "")
         cands = self.candidates(model)
         if not cands:
             raise HTTPException(503, f"no healthy pool serves {model!r}")
+        prompt = payload.get("prompt")
+        if isinstance(prompt, list) and prompt and isinstance(prompt[0], int):
+            cands = await self._kv_rank(cands, model, prompt)
         last_err: Optional[Exception] = None
         for pool in cands:                       # failover across pools
             pool.inflight += 1

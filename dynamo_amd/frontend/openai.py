@@ -126,6 +126,30 @@ def build_app(manager: ModelManager) -> FastAPI:
         except AllWorkersBusy as e:
             raise HTTPException(503, str(e))
 
+    @app.post("/internal/kv_overlap")
+    async def kv_overlap(raw: Request):
+        """Prefix-overlap digest for cross-pool (global) routing —
+        kv_dc_relay parity-lite: a global router asks each pool how many
+        KV blocks of a token prefix it already holds and routes to the
+        warmest pool. Returns the best per-worker overlap in blocks."""
+        payload = await raw.json()
+        token_ids = payload.get("token_ids", [])
+        model = payload.get("model", "")
+        try:
+            entry = manager.get(model)
+        except KeyError:
+            return {"overlap_blocks": 0, "block_size": 0}
+        r = entry.router
+        if r is None or not token_ids:
+            return {"overlap_blocks": 0, "block_size": 0}
+        from dynamo_amd import _core
+        hashes = _core.chain_hashes(list(token_ids), r.cfg.block_size,
+                                    r.cfg.block_salt)
+        matches = r.indexer.find_matches(hashes)
+        return {"overlap_blocks": max(matches.values()) if matches else 0,
+                "block_size": r.cfg.block_size,
+                "total_blocks": len(hashes)}
+
     @app.get("/config")
     async def config_dump():
         """Reproducibility config dump (reference parity:

@@ -99,3 +99,59 @@ def test_global_router_routing_and_failover():
             await router.stop()
             await stop_pool(pb)
     run(main())
+
+
+@pytest.mark.timeout(120)
+def test_global_router_kv_aware():
+    """Cross-pool prefix awareness: the global router asks each pool's
+    /internal/kv_overlap digest and routes a warm prefix to the pool that
+    already holds its KV (kv_dc_relay parity-lite)."""
+    import socket
+
+    def free_port():
+        with socket.socket() as s:
+            s.bind(("127.0.0.1", 0))
+            return s.getsockname()[1]
+
+    async def main():
+        pa = await start_pool("mock-model", free_port())
+        pb = await start_pool("mock-model", free_port())
+        router = GlobalRouter([pa["url"], pb["url"]], check_interval=0.3)
+        await router.start()
+        app = build_global_app(router)
+        client = httpx.AsyncClient(transport=httpx.ASGITransport(app=app),
+                                   base_url="http://g")
+        prompt = list(range(3, 3 + 96))     # 6 full 16-token pages
+        try:
+            # prime pool B directly so only IT holds the prefix
+            async with httpx.AsyncClient(timeout=30) as direct:
+                r = await direct.post(pb["url"] + "/v1/completions", json={
+                    "model": "mock-model", "prompt": prompt, "max_tokens": 2})
+                assert r.status_code == 200
+                # wait for KV events to reach pool B's router indexer
+                for _ in range(50):
+                    r = await direct.post(
+                        pb["url"] + "/internal/kv_overlap",
+                        json={"model": "mock-model", "token_ids": prompt})
+                    if r.json()["overlap_blocks"] > 0:
+                        break
+                    await asyncio.sleep(0.1)
+                assert r.json()["overlap_blocks"] > 0
+                r = await direct.post(
+                    pa["url"] + "/internal/kv_overlap",
+                    json={"model": "mock-model", "token_ids": prompt})
+                assert r.json()["overlap_blocks"] == 0
+            # same prefix through the GLOBAL router -> must land on pool B
+            a_before, b_before = (pa["mgr"].request_count,
+                                  pb["mgr"].request_count)
+            r = await client.post("/v1/completions", json={
+                "model": "mock-model", "prompt": prompt, "max_tokens": 2})
+            assert r.status_code == 200
+            assert pb["mgr"].request_count == b_before + 1
+            assert pa["mgr"].request_count == a_before
+        finally:
+            await client.aclose()
+            await router.stop()
+            await stop_pool(pa)
+            await stop_pool(pb)
+    run(main())
