@@ -114,8 +114,12 @@ class LLMEngine:
 
     # ------------------------------------------------------------------
     def add_request(self, req_id: str, prompt_tokens: List[int],
-                    sampling: SamplingParams | None = None) -> Request:
+                    sampling: SamplingParams | None = None,
+                    prompt_embeds=None) -> Request:
+        if prompt_embeds is not None and not prompt_tokens:
+            prompt_tokens = [0] * prompt_embeds.shape[0]
         req = Request(req_id, prompt_tokens, sampling or SamplingParams())
+        req.prompt_embeds = prompt_embeds
         self.requests[req_id] = req
         self.scheduler.add_request(req)
         return req
@@ -200,6 +204,7 @@ class LLMEngine:
                 else:
                     self._finish(req, reason)
             elif (self.cfg.kv_events or self.cfg.enable_prefix_caching) and \
+                    req.prompt_embeds is None and \
                     req.num_computed % self.cfg.page_size == 0:
                 req.kv.commit_full_pages(req.all_tokens, req.num_computed)
             outputs.append(StepOutput(req.req_id, int(tok), finished, reason,
@@ -281,7 +286,7 @@ class LLMEngine:
         if self.cfg.kv_events or self.cfg.enable_prefix_caching:
             for ss in sched.seqs:
                 r = ss.req
-                if r.kv is not None:
+                if r.kv is not None and r.prompt_embeds is None:
                     r.kv.commit_full_pages(r.all_tokens, r.num_computed)
         self.kv_events.extend(self.alloc.drain_events())
 
@@ -301,7 +306,7 @@ class LLMEngine:
 
     def _finish(self, req: Request, reason: str):
         # commit full pages before release so prefix cache retains them
-        if req.kv is not None:
+        if req.kv is not None and req.prompt_embeds is None:
             req.kv.commit_full_pages(req.all_tokens, req.num_computed)
         self.scheduler.finish(req, reason)
 

@@ -87,6 +87,7 @@ class ModelRunner:
 
         # ---- prefill part ----
         pf_tables, q_start, q_len, ctx_len = [], [], [], []
+        embeds_rows, embeds_parts = [], []
         qpos = 0
         for ss in sched.prefills:
             r = ss.req
@@ -96,6 +97,14 @@ class ModelRunner:
                 tokens.append(seq_tokens[j])
                 positions.append(j)
                 slots.append(r.kv.pages[j // ps] * ps + j % ps)
+            if r.prompt_embeds is not None:
+                # rows of this chunk that fall inside the prompt get their
+                # embeddings from the request, not the token table
+                pe = min(nc + n, len(r.prompt_tokens)) - nc
+                if pe > 0:
+                    base = nd + qpos
+                    embeds_rows.extend(range(base, base + pe))
+                    embeds_parts.append(r.prompt_embeds[nc:nc + pe])
             pf_tables.append(r.kv.pages)
             q_start.append(qpos)
             q_len.append(n)
@@ -131,6 +140,10 @@ class ModelRunner:
                          if ctx_len else None),
             logits_rows=torch.tensor(logits_rows, dtype=torch.int64).to(dev, non_blocking=True),
         )
+        if embeds_rows:
+            meta.embeds_rows = torch.tensor(embeds_rows, dtype=torch.int64,
+                                            device=dev)
+            meta.inputs_embeds = torch.cat(embeds_parts, 0).to(dev)
         if qpos and dev.type == "cuda":
             rows = ops.prefill_tile_rows(self.hq_local, self.hkv_local)
             meta.prefill_tiles = ops.build_prefill_tiles(q_len, dev, rows)

@@ -64,6 +64,11 @@ class Request:
         # embedding request accumulators (mean pool over prompt tokens)
         self.embedding: Optional[List[float]] = None
         self._embed_sum = None
+        # prompt_embeds input (PreprocessedRequest parity): [T, hidden]
+        # tensor replacing the token-table embeddings of the prompt;
+        # prefix caching is disabled for these requests (placeholder ids
+        # must not produce hash hits)
+        self.prompt_embeds = None
 
     @property
     def all_tokens(self) -> List[int]:
@@ -196,7 +201,8 @@ class Scheduler:
             req = self.waiting[0]
             if req.kv is None:
                 req.kv = SequenceKV(self.alloc, self.cfg.block_salt)
-                if self.cfg.enable_prefix_caching and not req.prefill_result:
+                if (self.cfg.enable_prefix_caching and not req.prefill_result
+                        and req.prompt_embeds is None):
                     req.num_computed = req.kv.match_prefix(req.all_tokens)
             remaining = req.total_len - req.num_computed
             n = min(remaining, budget)

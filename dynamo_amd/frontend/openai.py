@@ -40,6 +40,8 @@ class CompletionRequest(BaseModel):
     seed: int = 0
     ignore_eos: bool = False
     logprobs: Optional[int] = None  # top-N logprobs per token
+    # PreprocessedRequest prompt_embeds parity: {"b64", "shape", "dtype"}
+    prompt_embeds: Optional[dict] = None
     user: Optional[str] = None      # sticky-session key
 
 
@@ -118,10 +120,14 @@ def build_app(manager: ModelManager) -> FastAPI:
                 "ignore_eos": req.ignore_eos,
                 "stop_token_ids": [eos] if eos is not None else []}
         from dynamo_amd.router.kv_router import AllWorkersBusy
+        extra = None
+        pe = getattr(req, "prompt_embeds", None)
+        if pe:
+            extra = {"prompt_embeds": pe}
         try:
             async for chunk in manager.generate_tokens(
                     entry, token_ids, sampling, stop, request_id=rid,
-                    session_id=session_id):
+                    session_id=session_id, extra=extra):
                 yield chunk
         except AllWorkersBusy as e:
             raise HTTPException(503, str(e))

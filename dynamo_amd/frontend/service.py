@@ -158,7 +158,8 @@ class ModelManager:
     async def generate_tokens(self, entry: ModelEntry, token_ids: List[int],
                               sampling: dict, stop: dict,
                               request_id: Optional[str] = None,
-                              session_id: Optional[str] = None
+                              session_id: Optional[str] = None,
+                              extra: Optional[dict] = None
                               ) -> AsyncIterator[dict]:
         """Route + stream with migration retry (replays delivered tokens)."""
         self.request_count += 1
@@ -179,6 +180,8 @@ class ModelManager:
                                          - len(delivered))),
                 "annotations": {"trace_id": rid},
             }
+            if extra:
+                payload.update(extra)
             try:
                 # embedding requests are prefill-only: no P/D split
                 if (not sampling.get("embed")

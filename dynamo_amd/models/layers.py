@@ -39,6 +39,12 @@ class AttnMetadata:
     prefill_tiles: Optional[tuple] = None
     # sampling
     logits_rows: Optional[torch.Tensor] = None         # [Bs] int64 rows to sample
+    # prompt_embeds injection (reference parity: PreprocessedRequest
+    # prompt_embeds, lib/llm/src/protocols/common/preprocessor.rs:243):
+    # rows of the flattened batch whose input embeddings come from the
+    # request instead of the embedding table
+    embeds_rows: Optional[torch.Tensor] = None         # [Te] int64
+    inputs_embeds: Optional[torch.Tensor] = None       # [Te, hidden]
 
 
 def linear(x: torch.Tensor, w: torch.Tensor) -> torch.Tensor:

@@ -57,6 +57,9 @@ class LlamaForCausalLM(torch.nn.Module):
 
     def forward(self, input_ids, kv_pool, meta: AttnMetadata):
         x = torch.nn.functional.embedding(input_ids.long(), self.embed)
+        if meta.inputs_embeds is not None:
+            x = x.index_copy(0, meta.embeds_rows,
+                             meta.inputs_embeds.to(x.dtype))
         residual = None
         for i, layer in enumerate(self.layers):
             x, residual = layer.forward(x, residual, self.cos_sin,

@@ -181,6 +181,9 @@ class MixtralForCausalLM(torch.nn.Module):
 
     def forward(self, input_ids, kv_pool, meta: AttnMetadata):
         x = F.embedding(input_ids.long(), self.embed)
+        if meta.inputs_embeds is not None:
+            x = x.index_copy(0, meta.embeds_rows,
+                             meta.inputs_embeds.to(x.dtype))
         residual = None
         for i, layer in enumerate(self.layers):
             x, residual = layer.forward(x, residual, self.cos_sin,

@@ -65,6 +65,9 @@ class OPTForCausalLM(torch.nn.Module):
 
     def forward(self, input_ids, kv_pool, meta: AttnMetadata):
         x = F.embedding(input_ids.long(), self.embed)
+        if meta.inputs_embeds is not None:
+            x = x.index_copy(0, meta.embeds_rows,
+                             meta.inputs_embeds.to(x.dtype))
         x = x + F.embedding(meta.positions.long(), self.pos_embed)
         for i, layer in enumerate(self.layers):
             x = layer.forward(x, self.cos_sin, kv_pool.kcache(i),

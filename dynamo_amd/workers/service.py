@@ -158,9 +158,21 @@ class WorkerService:
                     prompt_tokens=len(tokens))
         q: asyncio.Queue = asyncio.Queue()
         self.queues[req_id] = q
+        pe = payload.get("prompt_embeds")
+        prompt_embeds = None
+        if pe is not None:
+            # PreprocessedRequest prompt_embeds (b64) parity
+            import base64
+            import numpy as np
+            import torch
+            arr = np.frombuffer(base64.b64decode(pe["b64"]),
+                                dtype=np.dtype(pe.get("dtype", "float16")))
+            prompt_embeds = torch.from_numpy(
+                arr.reshape(pe["shape"]).copy()).float()
         try:
             async with self._engine_lock:
-                req = self.engine.add_request(req_id, tokens, sp)
+                req = self.engine.add_request(req_id, tokens, sp,
+                                              prompt_embeds=prompt_embeds)
                 if is_prefill_role:
                     req.hold_kv = True
                 pr = payload.get("prefill_result")

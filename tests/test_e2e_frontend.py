@@ -493,3 +493,56 @@ def test_completions_logprobs_route():
         await rt.shutdown(drain=False)
         await mgr_rt.shutdown(drain=False)
     run(main())
+
+
+def test_prompt_embeds_over_http():
+    """prompt_embeds (b64 fp16) through the FULL wire: HTTP -> frontend ->
+    request plane -> worker -> engine; reproduces the token-path output."""
+    import base64
+
+    import numpy as np
+    import torch
+
+    from dynamo_amd.engine import EngineConfig, LLMEngine
+    from dynamo_amd.engine.config import PRESETS
+
+    async def main():
+        shared = MemoryDiscovery()
+        rt = DistributedRuntime(shared)
+        cfg = EngineConfig(model=PRESETS["tiny-llama"], device="cpu",
+                           dtype="float32", max_num_seqs=4,
+                           max_batched_tokens=256, max_model_len=512,
+                           kv_pool_pages=64, page_size=16,
+                           enable_hip_graphs=False)
+        eng = LLMEngine(cfg, seed=7)
+        ws = WorkerService(eng, rt)
+        await ws.start()
+        mgr_rt = DistributedRuntime(shared)
+        mgr = ModelManager(mgr_rt)
+        await mgr.start(watch_interval=0.2)
+        app = build_app(mgr)
+        client = httpx.AsyncClient(transport=httpx.ASGITransport(app=app),
+                                   base_url="http://t")
+        prompt = list(range(40, 90))
+        r = await client.post("/v1/completions", json={
+            "model": "tiny-llama", "prompt": prompt, "max_tokens": 4})
+        ref = r.json()["choices"][0]["token_ids"]
+        with torch.no_grad():
+            pe = eng.runner.model.embed[torch.tensor(prompt)].float()
+        arr = pe.numpy().astype(np.float16)
+        r = await client.post("/v1/completions", json={
+            "model": "tiny-llama", "prompt": [],
+            "prompt_embeds": {"b64": base64.b64encode(arr.tobytes()).decode(),
+                              "shape": list(arr.shape), "dtype": "float16"},
+            "max_tokens": 4})
+        assert r.status_code == 200, r.text
+        got = r.json()["choices"][0]["token_ids"]
+        # fp16 wire quantization of fp32 embeddings: allow small drift but
+        # require the same greedy trajectory on this well-separated model
+        assert got == ref, (got, ref)
+        await client.aclose()
+        await mgr.stop()
+        await ws.stop()
+        await rt.shutdown(drain=False)
+        await mgr_rt.shutdown(drain=False)
+    run(main())

@@ -261,3 +261,58 @@ def test_logprobs():
         # top list sorted descending
         assert lp["top"][0][1] >= lp["top"][1][1] >= lp["top"][2][1]
         assert math.isfinite(lp["token_logprob"])
+
+
+def test_prompt_embeds_input():
+    """prompt_embeds replacing token-table lookups: feeding the model's OWN
+    embedding rows for a token prompt must reproduce the token-path output
+    exactly; prefix caching stays off for embeds prompts."""
+    prompt = list(range(60, 100))
+    e_tok = make_engine()
+    ref = generate(e_tok, [prompt], max_tokens=5)[0]
+
+    e_emb = make_engine()
+    import torch
+    with torch.no_grad():
+        pe = e_emb.runner.model.embed[torch.tensor(prompt)].clone().float()
+    e_emb.add_request("emb", [], SamplingParams(max_tokens=5),
+                      prompt_embeds=pe)
+    req = e_emb.requests["emb"]
+    assert len(req.prompt_tokens) == len(prompt)  # placeholders
+    out = []
+    while e_emb.has_work():
+        for so in e_emb.step():
+            out.append(so.new_token)
+    assert out == ref
+    # placeholder ids must NOT poison the prefix cache: a fresh embeds
+    # request with DIFFERENT embeddings gets no cached reuse
+    with torch.no_grad():
+        pe2 = e_emb.runner.model.embed[
+            torch.tensor(list(range(10, 50)))].clone().float()
+    e_emb.add_request("emb2", [], SamplingParams(max_tokens=3),
+                      prompt_embeds=pe2)
+    ref2 = generate(make_engine(), [list(range(10, 50))], max_tokens=3)[0]
+    out2 = []
+    while e_emb.has_work():
+        for so in e_emb.step():
+            out2.append(so.new_token)
+    assert out2 == ref2
+
+
+def test_prompt_embeds_chunked():
+    """Chunked prefill slices the provided embeddings correctly."""
+    import torch
+    prompt = list(range(5, 165))   # 160 tokens >> 32-token budget
+    ref_engine = make_engine()
+    ref = generate(ref_engine, [prompt], max_tokens=4)[0]
+    e = make_engine()
+    e.cfg.max_batched_tokens = 32
+    e.scheduler.cfg.max_batched_tokens = 32
+    with torch.no_grad():
+        pe = e.runner.model.embed[torch.tensor(prompt)].clone().float()
+    e.add_request("c", [], SamplingParams(max_tokens=4), prompt_embeds=pe)
+    out = []
+    while e.has_work():
+        for so in e.step():
+            out.append(so.new_token)
+    assert out == ref
