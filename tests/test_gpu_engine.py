@@ -104,3 +104,23 @@ def test_gpu_logprobs_fast_path():
         assert lp is not None and len(lp["top"]) == 2
         assert lp["top"][0][0] == tok        # greedy argmax is top-1
         assert lp["token_logprob"] <= 0.0
+
+
+def test_gpu_prompt_embeds():
+    """prompt_embeds table-bypass on the native GPU path: feeding the
+    model's own embedding rows reproduces the token-path output."""
+    import torch
+    prompt = list(range(100, 160))
+    e_tok = make_engine()
+    ref = generate(e_tok, [prompt], max_tokens=4)[0]
+    e_emb = make_engine()
+    with torch.no_grad():
+        pe = e_emb.runner.model.embed[
+            torch.tensor(prompt, device="cuda")].clone().float()
+    e_emb.add_request("emb", [], SamplingParams(max_tokens=4),
+                      prompt_embeds=pe)
+    out = []
+    while e_emb.has_work():
+        for so in e_emb.step():
+            out.append(so.new_token)
+    assert out == ref
