@@ -183,8 +183,21 @@ class ModelManager:
             if extra:
                 payload.update(extra)
             try:
+                # direct routing hint (reference parity: RoutingHints
+                # backend_instance_id / RouterMode::Direct): bypass the
+                # KV-aware selection and pin to the named instance
+                hint = ((extra or {}).get("routing") or {}).get(
+                    "backend_instance_id")
                 # embedding requests are prefill-only: no P/D split
-                if (not sampling.get("embed")
+                if hint is not None:
+                    alive = {i.instance_id
+                             for i in entry.router.client.instances()}
+                    if hint not in alive:
+                        raise NoInstancesError(
+                            f"pinned instance {hint} is not alive")
+                    entry.router.begin_request(hint, payload["token_ids"])
+                    gen = self._direct_gen(entry, payload, hint)
+                elif (not sampling.get("embed")
                         and entry.prefill_router is not None
                         and entry.prefill_router.has_prefill_pool()):
                     gen = entry.prefill_router.generate(payload)

@@ -546,3 +546,36 @@ def test_prompt_embeds_over_http():
         await rt.shutdown(drain=False)
         await mgr_rt.shutdown(drain=False)
     run(main())
+
+
+def test_direct_routing_hint():
+    """routing.backend_instance_id pins a request to a named worker
+    (RouterMode::Direct parity); a dead pin raises instead of re-routing."""
+    async def main():
+        shared, services, mgr, client = await with_stack(nworkers=2)
+        entry = mgr.get("mock-model")
+        workers = {ws.instance_id: ws for ws, _ in services}
+        target = sorted(workers)[1]
+        before = {iid: ws._req_counter for iid, ws in workers.items()}
+        toks = []
+        async for ch in mgr.generate_tokens(
+                entry, [5, 6, 7], {"temperature": 0.0}, {"max_tokens": 3},
+                extra={"routing": {"backend_instance_id": target}}):
+            toks.extend(ch.get("token_ids", []))
+        assert len(toks) == 3
+        after = {iid: ws._req_counter for iid, ws in workers.items()}
+        assert after[target] == before[target] + 1
+        other = sorted(workers)[0]
+        assert after[other] == before[other]
+        # dead pin -> NoInstancesError (after migration retries)
+        from dynamo_amd.runtime import NoInstancesError
+        try:
+            async for _ in mgr.generate_tokens(
+                    entry, [1], {}, {"max_tokens": 1},
+                    extra={"routing": {"backend_instance_id": "nope"}}):
+                pass
+            assert False, "expected NoInstancesError"
+        except NoInstancesError:
+            pass
+        await teardown(services, mgr, client)
+    run(main())
