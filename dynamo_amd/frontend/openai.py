@@ -64,6 +64,7 @@ class ChatRequest(BaseModel):
     top_p: float = 1.0
     top_k: int = 0
     stream: bool = False
+    stop: Optional[Union[str, List[str]]] = None
     seed: int = 0
     ignore_eos: bool = False
     user: Optional[str] = None      # sticky-session key
@@ -87,6 +88,26 @@ class AnthropicMessagesRequest(BaseModel):
 
 def _session_of(req, raw: Request) -> Optional[str]:
     return raw.headers.get("x-session-id") or getattr(req, "user", None)
+
+
+def _stop_list(req) -> List[str]:
+    stop = getattr(req, "stop", None)
+    if stop is None:
+        return []
+    return [stop] if isinstance(stop, str) else list(stop)
+
+
+def _find_stop(text: str, stops: List[str]) -> int:
+    """Earliest stop-string position in text, or -1 (Backend-operator
+    stop-condition parity: the reference trims the stop string from the
+    returned text)."""
+    best = -1
+    for st in stops:
+        if st:
+            i = text.find(st)
+            if i >= 0 and (best < 0 or i < best):
+                best = i
+    return best
 
 
 def build_app(manager: ModelManager) -> FastAPI:
@@ -194,25 +215,54 @@ def build_app(manager: ModelManager) -> FastAPI:
         t0 = time.time()
 
         if req.stream:
+            stops = _stop_list(req)
+
             async def sse():
                 produced: List[int] = []
+                acc = ""     # decoded text so far (stable incremental path)
+                sent = 0     # chars of acc already emitted
+                # a stop may span chunk boundaries: withhold its max length
+                # minus one trailing chars until they are cleared
+                hold = max((len(s) for s in stops), default=1) - 1
                 first = True
+
+                def event(text, finish):
+                    data = {"id": rid, "object": "text_completion",
+                            "model": entry.name, "choices": [{
+                                "index": 0, "text": text,
+                                "finish_reason": finish}]}
+                    return f"data: {json.dumps(data)}\n\n"
+
                 try:
+                    finish = None
                     async for chunk in _run(entry, token_ids, req, rid, _session_of(req, raw)):
                         if await raw.is_disconnected():
                             break
                         if first:
                             TTFT.labels(entry.name).observe(time.time() - t0)
                             first = False
-                        new = chunk.get("token_ids", [])
                         prev = len(produced)
-                        produced.extend(new)
-                        text = entry.tokenizer.decode_incremental(produced, prev)
-                        data = {"id": rid, "object": "text_completion",
-                                "model": entry.name, "choices": [{
-                                    "index": 0, "text": text,
-                                    "finish_reason": chunk.get("finish_reason")}]}
-                        yield f"data: {json.dumps(data)}\n\n"
+                        produced.extend(chunk.get("token_ids", []))
+                        acc += entry.tokenizer.decode_incremental(produced,
+                                                                  prev)
+                        finish = chunk.get("finish_reason")
+                        if stops:
+                            cut = _find_stop(acc, stops)
+                            if cut >= 0:
+                                yield event(acc[sent:cut], "stop")
+                                sent = cut
+                                finish = "stop"
+                                break
+                            safe = max(sent, len(acc) - hold)
+                            if safe > sent or finish:
+                                tail = len(acc) if finish else safe
+                                yield event(acc[sent:tail], finish)
+                                sent = tail
+                        else:
+                            yield event(acc[sent:], finish)
+                            sent = len(acc)
+                    if stops and finish is None and sent < len(acc):
+                        yield event(acc[sent:], None)   # disconnect flush
                     yield "data: [DONE]\n\n"
                 finally:
                     LATENCY.labels(entry.name).observe(time.time() - t0)
@@ -221,10 +271,21 @@ def build_app(manager: ModelManager) -> FastAPI:
         produced: List[int] = []
         lps: List[dict] = []
         finish = None
+        stops = _stop_list(req)
+        text_cut = None
+        acc = ""
         async for chunk in _run(entry, token_ids, req, rid, _session_of(req, raw)):
+            prev = len(produced)
             produced.extend(chunk.get("token_ids", []))
             lps.extend(chunk.get("logprobs", []))
             finish = chunk.get("finish_reason") or finish
+            if stops:
+                acc += entry.tokenizer.decode_incremental(produced, prev)
+                cut = _find_stop(acc, stops)
+                if cut >= 0:
+                    text_cut = acc[:cut]
+                    finish = "stop"
+                    break
         LATENCY.labels(entry.name).observe(time.time() - t0)
         logprobs_out = None
         if req.logprobs and lps:
@@ -239,7 +300,8 @@ def build_app(manager: ModelManager) -> FastAPI:
             "id": rid, "object": "text_completion", "created": int(t0),
             "model": entry.name,
             "choices": [{"index": 0,
-                         "text": entry.tokenizer.decode(produced),
+                         "text": (text_cut if text_cut is not None
+                                  else entry.tokenizer.decode(produced)),
                          "finish_reason": finish or "stop",
                          "logprobs": logprobs_out,
                          "token_ids": produced}],
@@ -397,16 +459,28 @@ def build_app(manager: ModelManager) -> FastAPI:
 
         produced: List[int] = []
         finish = None
+        stops = _stop_list(req)
+        text_cut = None
+        acc = ""
         async for chunk in _run(entry, token_ids, req, rid, _session_of(req, raw)):
+            prev = len(produced)
             produced.extend(chunk.get("token_ids", []))
             finish = chunk.get("finish_reason") or finish
+            if stops:
+                acc += entry.tokenizer.decode_incremental(produced, prev)
+                cut = _find_stop(acc, stops)
+                if cut >= 0:
+                    text_cut = acc[:cut]
+                    finish = "stop"
+                    break
         LATENCY.labels(entry.name).observe(time.time() - t0)
         return {
             "id": rid, "object": "chat.completion", "created": int(t0),
             "model": entry.name,
             "choices": [{"index": 0, "message": {
                 "role": "assistant",
-                "content": entry.tokenizer.decode(produced)},
+                "content": (text_cut if text_cut is not None
+                            else entry.tokenizer.decode(produced))},
                 "finish_reason": finish or "stop"}],
             "usage": {"prompt_tokens": len(token_ids),
                       "completion_tokens": len(produced),

@@ -37,7 +37,14 @@ class ByteTokenizer:
 
     def decode(self, ids: List[int]) -> str:
         bs = bytes(i for i in ids if 0 <= i < 256)
-        return bs.decode("utf-8", errors="replace")
+        try:
+            return bs.decode("utf-8")
+        except UnicodeDecodeError:
+            # non-UTF-8 byte streams (synthetic tokens): latin-1 maps every
+            # byte to a DISTINCT char, keeping decode stable and reversible
+            # (utf-8 replacement chars collapse distinct bytes and break
+            # incremental/full decode consistency)
+            return bs.decode("latin-1")
 
     def decode_incremental(self, ids: List[int], prev_len: int) -> str:
         """Decode the new suffix, robust to split UTF-8 sequences."""
