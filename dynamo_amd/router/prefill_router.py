@@ -28,7 +28,8 @@ class PrefillRouter:
                  decode_component: str = "backend",
                  cfg: RouterConfig | None = None,
                  bypass_token_threshold: int = 2048,
-                 bypass_overlap_ratio: float = 0.7):
+                 bypass_overlap_ratio: float = 0.7,
+                 bypass_decode_busy_waiting: int = 4):
         self.runtime = runtime
         self.namespace = namespace
         self.cfg = cfg or RouterConfig()
@@ -38,6 +39,9 @@ class PrefillRouter:
                                       self.cfg)
         self.bypass_token_threshold = bypass_token_threshold
         self.bypass_overlap_ratio = bypass_overlap_ratio
+        # busy gating (conditional_disagg.rs:111 parity): never bypass onto
+        # a decode pool whose queues are already deep
+        self.bypass_decode_busy_waiting = bypass_decode_busy_waiting
 
     async def start(self):
         await self.prefill_router.start()
@@ -51,9 +55,18 @@ class PrefillRouter:
     def has_prefill_pool(self) -> bool:
         return bool(self.prefill_router.client.instances())
 
+    def _decode_pool_busy(self) -> bool:
+        ws = [w for w in self.decode_router.workers.values()]
+        if not ws:
+            return False
+        return all(w.num_waiting >= self.bypass_decode_busy_waiting
+                   for w in ws)
+
     def _should_bypass(self, token_ids: List[int]) -> bool:
         if not self.has_prefill_pool():
             return True
+        if self._decode_pool_busy():
+            return False   # busy gating: keep long prefills off decode
         from dynamo_amd import _core
         bs = self.cfg.block_size
         hashes = _core.chain_hashes(token_ids, bs, self.cfg.block_salt)

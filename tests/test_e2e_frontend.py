@@ -645,3 +645,31 @@ def test_stop_strings():
         await rt.shutdown(drain=False)
         await mgr_rt.shutdown(drain=False)
     run(main())
+
+
+def test_conditional_bypass_busy_gating():
+    """Busy gating: a short prefill that would normally bypass to the
+    decode pool goes through the prefill pool when every decode worker
+    reports a deep waiting queue."""
+    async def main():
+        shared, services, mgr, client = await with_stack(nworkers=1,
+                                                         disagg=True)
+        entry = mgr.get("mock-model")
+        pr = entry.prefill_router
+        assert pr is not None and pr.has_prefill_pool()
+        short = list(range(30))           # << bypass_token_threshold
+        assert pr._should_bypass(short) is True
+        # simulate a saturated decode pool via the polled metrics state
+        from dynamo_amd.router.kv_router import WorkerState
+        for inst in pr.decode_router.client.instances():
+            st = pr.decode_router.workers.setdefault(
+                inst.instance_id, WorkerState(inst.instance_id))
+            st.num_waiting = 99
+        assert pr._should_bypass(short) is False
+        # and requests still complete through the prefill path
+        r = await client.post("/v1/completions", json={
+            "model": "mock-model", "prompt": short, "max_tokens": 3})
+        assert r.status_code == 200
+        assert r.json()["usage"]["completion_tokens"] == 3
+        await teardown(services, mgr, client)
+    run(main())
