@@ -202,8 +202,9 @@ class ModelManager:
                         and entry.prefill_router.has_prefill_pool()):
                     gen = entry.prefill_router.generate(payload)
                 else:
-                    iid = entry.router.select(payload["token_ids"],
-                                              session_id=session_id)
+                    iid = entry.router.select(
+                        payload["token_ids"], session_id=session_id,
+                        override=(extra or {}).get("router_config_override"))
                     if iid is None:
                         raise NoInstancesError(f"no workers for {entry.name}")
                     entry.router.begin_request(iid, payload["token_ids"])

@@ -160,12 +160,16 @@ class KvRouter:
 
     # -- selection ------------------------------------------------------
     def select(self, token_ids: List[int],
-               session_id: Optional[str] = None) -> Optional[str]:
+               session_id: Optional[str] = None,
+               override: Optional[dict] = None) -> Optional[str]:
         """Pick a worker instance_id for this token sequence.
 
         session_id pins a session to its previous worker while that worker
         is alive (sticky sessions); busy_threshold > 0 rejects with
-        AllWorkersBusy when every worker is overloaded."""
+        AllWorkersBusy when every worker is overloaded; `override` applies
+        per-request router-config overrides (router_config_override parity:
+        kv-router/src/scheduling/config.rs) for the cost-formula weights
+        and sampling temperature."""
         insts = self.client.instances()
         if not insts:
             return None
@@ -181,16 +185,25 @@ class KvRouter:
             if pinned in alive:
                 self._sessions.move_to_end(session_id)
                 return pinned
-            iid = self._select_inner(insts, token_ids)
+            iid = self._select_inner(insts, token_ids, override)
             self._sessions[session_id] = iid
             self._sessions.move_to_end(session_id)
             while len(self._sessions) > self.cfg.max_sessions:
                 self._sessions.popitem(last=False)
             return iid
-        return self._select_inner(insts, token_ids)
+        return self._select_inner(insts, token_ids, override)
 
-    def _select_inner(self, insts, token_ids: List[int]) -> Optional[str]:
-        mode = self.cfg.mode
+    def _select_inner(self, insts, token_ids: List[int],
+                      override: Optional[dict] = None) -> Optional[str]:
+        cfg = self.cfg
+        if override:
+            import dataclasses
+            allowed = {"mode", "overlap_score_weight",
+                       "decode_active_request_weight", "prefill_load_scale",
+                       "router_temperature"}
+            cfg = dataclasses.replace(
+                cfg, **{k: v for k, v in override.items() if k in allowed})
+        mode = cfg.mode
         if mode == "round_robin":
             self._rr += 1
             return insts[self._rr % len(insts)].instance_id
@@ -202,8 +215,8 @@ class KvRouter:
             return min(cand, key=lambda i: self._load(i.instance_id)).instance_id
 
         # kv mode
-        bs = self.cfg.block_size
-        hashes = _core.chain_hashes(token_ids, bs, self.cfg.block_salt)
+        bs = cfg.block_size
+        hashes = _core.chain_hashes(token_ids, bs, cfg.block_salt)
         matches = self.indexer.find_matches(hashes)
         prefill_blocks = (len(token_ids) + bs - 1) // bs
         logits = []
@@ -211,13 +224,13 @@ class KvRouter:
             ws = self.workers.get(inst.instance_id) or WorkerState(
                 inst.instance_id)
             overlap = matches.get(self._wid(inst.instance_id), 0)
-            cost = (self.cfg.prefill_load_scale
+            cost = (cfg.prefill_load_scale
                     * max(0.0, prefill_blocks
-                          - self.cfg.overlap_score_weight * overlap)
+                          - cfg.overlap_score_weight * overlap)
                     + ws.active_blocks
-                    + self.cfg.decode_active_request_weight * ws.active_requests)
+                    + cfg.decode_active_request_weight * ws.active_requests)
             logits.append(cost)
-        T = self.cfg.router_temperature
+        T = cfg.router_temperature
         if T <= 0:
             best = min(range(len(insts)), key=lambda i: logits[i])
             return insts[best].instance_id

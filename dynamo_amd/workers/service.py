@@ -143,6 +143,13 @@ class WorkerService:
     # ------------------------------------------------------------------
     async def generate(self, payload: dict, ctx: RequestContext
                        ) -> AsyncIterator[dict]:
+        if payload.get("_HEALTH_CHECK"):
+            # canary parity (PreprocessedRequest _HEALTH_CHECK flag): verify
+            # the engine loop is alive without generating anything
+            yield {"token_ids": [], "health": "ok",
+                   "worker_type": self.worker_type,
+                   "num_running": self.engine.scheduler.num_running()}
+            return
         self._req_counter += 1
         req_id = payload.get("request_id") or f"req-{self._req_counter}"
         tokens = list(payload["token_ids"])

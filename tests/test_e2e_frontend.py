@@ -673,3 +673,34 @@ def test_conditional_bypass_busy_gating():
         assert r.json()["usage"]["completion_tokens"] == 3
         await teardown(services, mgr, client)
     run(main())
+
+
+def test_router_override_and_health_canary():
+    async def main():
+        shared, services, mgr, client = await with_stack(nworkers=2)
+        entry = mgr.get("mock-model")
+        # override flips mode to round_robin for one call: consecutive
+        # selects alternate workers regardless of kv cost
+        a = entry.router.select([1, 2, 3],
+                                override={"mode": "round_robin"})
+        b = entry.router.select([1, 2, 3],
+                                override={"mode": "round_robin"})
+        assert a != b
+        # unknown keys are ignored, known ones applied
+        c = entry.router.select([1, 2, 3],
+                                override={"bogus": 1, "mode": "random"})
+        assert c is not None
+        # _HEALTH_CHECK canary short-circuits generation on the worker
+        ws, _ = services[0]
+        inst_addr = None
+        for inst in mgr.runtime.discovery.list("dynamo"):
+            if inst.instance_id == ws.instance_id:
+                inst_addr = inst.address
+        chunks = []
+        async for ch in mgr.runtime.client.call_stream(
+                inst_addr, "backend.generate", {"_HEALTH_CHECK": True}):
+            chunks.append(ch)
+        assert chunks and chunks[0]["health"] == "ok"
+        assert chunks[0]["token_ids"] == []
+        await teardown(services, mgr, client)
+    run(main())
