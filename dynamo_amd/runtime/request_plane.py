@@ -128,10 +128,18 @@ class RequestPlaneServer:
                 await send({"type": "rsp", "rid": rid, "final": True,
                             "error": f"no such endpoint {endpoint!r}"}, None)
                 return
-            async for chunk in handler(body, ctx):
-                if ctx.cancelled:
-                    break
-                await send({"type": "rsp", "rid": rid, "final": False}, chunk)
+            agen = handler(body, ctx)
+            try:
+                async for chunk in agen:
+                    if ctx.cancelled:
+                        break
+                    await send({"type": "rsp", "rid": rid, "final": False},
+                               chunk)
+            finally:
+                # close the handler promptly so its cleanup (e.g. engine
+                # abort on cancellation) runs NOW, not at GC time
+                if hasattr(agen, "aclose"):
+                    await agen.aclose()
             await send({"type": "rsp", "rid": rid, "final": True}, None)
         except (ConnectionResetError, BrokenPipeError):
             pass
@@ -220,7 +228,11 @@ class RequestPlaneClient:
                 if header.get("final"):
                     break
                 yield body
-        except asyncio.CancelledError:
+        except (asyncio.CancelledError, GeneratorExit):
+            # CancelledError: the consuming task was cancelled.
+            # GeneratorExit: the consumer broke out of its `async for` and
+            # this generator is being aclose()d — awaiting a last send here
+            # is legal as long as we don't yield again.
             try:
                 if not c.closed:
                     await c.send({"type": "cancel", "rid": rid}, None)

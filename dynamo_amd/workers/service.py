@@ -165,6 +165,7 @@ class WorkerService:
                     prompt_tokens=len(tokens))
         q: asyncio.Queue = asyncio.Queue()
         self.queues[req_id] = q
+        completed = False
         pe = payload.get("prompt_embeds")
         prompt_embeds = None
         if pe is not None:
@@ -188,14 +189,15 @@ class WorkerService:
             self._work.set()
 
             while True:
+                if ctx.cancelled:   # checked EVERY iteration — a steadily
+                    # producing stream must still notice a cancel frame
+                    async with self._engine_lock:
+                        self.engine.abort(req_id)
+                    return
                 get = asyncio.create_task(q.get())
                 done, _ = await asyncio.wait({get}, timeout=0.05)
                 if not done:
                     get.cancel()
-                    if ctx.cancelled:
-                        async with self._engine_lock:
-                            self.engine.abort(req_id)
-                        return
                     continue
                 so = get.result()
                 chunk: dict = {"token_ids": ([so.new_token]
@@ -210,11 +212,17 @@ class WorkerService:
                     if is_prefill_role:
                         chunk["disaggregated_params"] = \
                             self._disagg_params(req)
+                    completed = True
                     yield chunk
                     return
                 yield chunk
         finally:
             self.queues.pop(req_id, None)
+            if not completed:
+                # stream ended early (cancel frame, client disconnect,
+                # handler close): stop generating for this request
+                async with self._engine_lock:
+                    self.engine.abort(req_id)
 
     def _disagg_params(self, req: Request) -> dict:
         return {

@@ -704,3 +704,38 @@ def test_router_override_and_health_canary():
         assert chunks[0]["token_ids"] == []
         await teardown(services, mgr, client)
     run(main())
+
+
+def test_cancellation_propagates_to_worker():
+    """Dropping the client stream mid-generation cancels the request on the
+    worker (AsyncEngineContext::stop_generating parity): the engine's
+    request is aborted, not run to completion."""
+    async def main():
+        shared, services, mgr, client = await with_stack(nworkers=1)
+        ws, _ = services[0]
+        ws.engine.runner.decode_step_ms = 30   # slow decode
+        entry = mgr.get("mock-model")
+
+        got = []
+
+        async def consume():
+            async for chunk in mgr.generate_tokens(
+                    entry, [4, 5, 6], {"temperature": 0.0},
+                    {"max_tokens": 200}, request_id="cancel-me"):
+                got.append(chunk)
+                if len(got) >= 3:
+                    break    # closes the generator -> cancel frame
+
+        await consume()
+        # the worker should abort the request shortly after
+        for _ in range(100):
+            if ("cancel-me" not in ws.engine.requests
+                    and not ws.engine.scheduler.has_work()):
+                break
+            await asyncio.sleep(0.05)
+        assert "cancel-me" not in ws.engine.requests
+        assert not ws.engine.scheduler.has_work(), \
+            "engine still generating after client cancelled"
+        assert len(got) < 200
+        await teardown(services, mgr, client)
+    run(main())
