@@ -71,7 +71,20 @@ class WorkerService:
         self._paused.set()  # set = running
 
     # ------------------------------------------------------------------
+    def _mdc_sum(self, card: dict) -> str:
+        import hashlib
+        import json
+        return hashlib.blake2b(
+            json.dumps(card, sort_keys=True).encode(),
+            digest_size=8).hexdigest()
+
     def model_card(self) -> dict:
+        cfg = self.engine.cfg
+        card = self._card_body()
+        card["mdc_sum"] = self._mdc_sum(card)
+        return card
+
+    def _card_body(self) -> dict:
         cfg = self.engine.cfg
         return {
             "name": self.model_name,
@@ -143,6 +156,15 @@ class WorkerService:
     # ------------------------------------------------------------------
     async def generate(self, payload: dict, ctx: RequestContext
                        ) -> AsyncIterator[dict]:
+        want = payload.get("mdc_sum")
+        if want is not None:
+            # model-deployment-card checksum (PreprocessedRequest mdc_sum
+            # parity): reject requests preprocessed against a STALE card
+            have = self.model_card()["mdc_sum"]
+            if want != have:
+                raise RuntimeError(
+                    f"mdc_sum mismatch: request {want} != worker {have} "
+                    "(frontend holds a stale model card)")
         if payload.get("_HEALTH_CHECK"):
             # canary parity (PreprocessedRequest _HEALTH_CHECK flag): verify
             # the engine loop is alive without generating anything

@@ -739,3 +739,33 @@ def test_cancellation_propagates_to_worker():
         assert len(got) < 200
         await teardown(services, mgr, client)
     run(main())
+
+
+def test_mdc_sum_validation():
+    """mdc_sum parity: a request carrying the worker's card checksum is
+    served; a stale checksum is rejected with an error frame."""
+    async def main():
+        shared, services, mgr, client = await with_stack(nworkers=1)
+        ws, rt = services[0]
+        good = ws.model_card()["mdc_sum"]
+        inst = mgr.runtime.discovery.list("dynamo")[0]
+        chunks = []
+        async for ch in mgr.runtime.client.call_stream(
+                inst.address, "backend.generate",
+                {"request_id": "mdc-ok", "token_ids": [1, 2],
+                 "stop_conditions": {"max_tokens": 2}, "mdc_sum": good}):
+            chunks.append(ch)
+        assert sum(len(c.get("token_ids", [])) for c in chunks) == 2
+        from dynamo_amd.runtime import EndpointError
+        try:
+            async for ch in mgr.runtime.client.call_stream(
+                    inst.address, "backend.generate",
+                    {"request_id": "mdc-bad", "token_ids": [1],
+                     "stop_conditions": {"max_tokens": 1},
+                     "mdc_sum": "deadbeef"}):
+                pass
+            assert False, "expected EndpointError"
+        except EndpointError as e:
+            assert "mdc_sum mismatch" in str(e)
+        await teardown(services, mgr, client)
+    run(main())
