@@ -121,10 +121,19 @@ class WorkerService:
         self._loop_task = asyncio.create_task(self._engine_loop())
         return self
 
-    async def stop(self):
+    async def stop(self, drain: bool = False, drain_timeout: float = 30.0):
+        """Stop serving. drain=True: deregister from discovery first (no
+        new routing), then let in-flight requests finish before killing
+        the engine loop (reference parity: push_endpoint.rs:46-56 inflight
+        counter + graceful-shutdown-architecture.md)."""
+        self.comp.deregister()
+        if drain:
+            deadline = asyncio.get_event_loop().time() + drain_timeout
+            while (self.engine.has_work()
+                   and asyncio.get_event_loop().time() < deadline):
+                await asyncio.sleep(0.02)
         if self._loop_task:
             self._loop_task.cancel()
-        self.comp.deregister()
 
     @property
     def instance_id(self) -> str:

@@ -769,3 +769,37 @@ def test_mdc_sum_validation():
             assert "mdc_sum mismatch" in str(e)
         await teardown(services, mgr, client)
     run(main())
+
+
+def test_graceful_shutdown_drains():
+    """stop(drain=True) deregisters, then lets the in-flight stream finish
+    instead of cutting it off."""
+    async def main():
+        shared, services, mgr, client = await with_stack(nworkers=1)
+        ws, rt = services[0]
+        ws.engine.runner.decode_step_ms = 10
+        entry = mgr.get("mock-model")
+        got = []
+
+        async def consume():
+            async for ch in mgr.generate_tokens(
+                    entry, [9, 9, 9], {"temperature": 0.0},
+                    {"max_tokens": 20}, request_id="drain-me"):
+                got.extend(ch.get("token_ids", []))
+
+        task = asyncio.create_task(consume())
+        # wait until the request is running, then gracefully stop
+        for _ in range(100):
+            if "drain-me" in ws.engine.requests:
+                break
+            await asyncio.sleep(0.01)
+        await ws.stop(drain=True)
+        await asyncio.wait_for(task, timeout=30)
+        assert len(got) == 20, f"stream was cut at {len(got)} tokens"
+        # worker no longer discoverable
+        assert not [i for i in mgr.runtime.discovery.list("dynamo")
+                    if i.instance_id == ws.instance_id]
+        await mgr.stop()
+        await rt.shutdown(drain=False)
+        await client.aclose()
+    run(main())
