@@ -126,6 +126,7 @@ class EngineConfig:
     host_cache_pages: int = 0           # KVBM G2 tier size (0 = disabled)
     disk_cache_pages: int = 0           # KVBM G3 tier size (0 = disabled)
     disk_cache_path: str = ""           # G3 backing file (required if G3 on)
+    object_cache_dir: str = ""          # G4 shared object store (disabled="")
     kv_events: bool = True              # emit stored/removed block events
     block_salt: int = 0
     # disaggregation

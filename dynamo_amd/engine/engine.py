@@ -61,7 +61,8 @@ class LLMEngine:
             self.host_tier = HostKVTier(self.runner.kv_pool,
                                         cfg.host_cache_pages,
                                         disk_path=cfg.disk_cache_path,
-                                        num_disk_pages=cfg.disk_cache_pages)
+                                        num_disk_pages=cfg.disk_cache_pages,
+                                        object_dir=cfg.object_cache_dir)
             self.alloc.host_tier = self.host_tier
         self.scheduler = Scheduler(cfg, self.alloc)
         self.requests: Dict[str, Request] = {}

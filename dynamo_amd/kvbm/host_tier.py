@@ -25,7 +25,8 @@ log = logging.getLogger("dynamo_amd.kvbm")
 
 class HostKVTier:
     def __init__(self, kv_pool, num_host_pages: int,
-                 disk_path: str = "", num_disk_pages: int = 0):
+                 disk_path: str = "", num_disk_pages: int = 0,
+                 object_dir: str = ""):
         self.pool = kv_pool
         self.device = kv_pool.device
         L, two, P, hkv, ps, hd = kv_pool.shape
@@ -51,19 +52,32 @@ class HostKVTier:
         self.events = []  # (kind, hash) host-tier events
         # G3: disk tier under G2 — host LRU evictions spill there
         self.disk = None
+        elem_bytes = torch.empty(0, dtype=kv_pool.dtype).element_size()
         if num_disk_pages > 0 and disk_path:
             from .disk_tier import DiskKVTier
-            elem_bytes = torch.empty(0, dtype=kv_pool.dtype).element_size()
             self.disk = DiskKVTier(disk_path, num_disk_pages,
                                    self.page_elems * elem_bytes)
+        # G4: shared content-addressed object store (cross-WORKER reuse);
+        # publishes happen off-thread so evictions never stall the engine
+        self.objects = None
+        self._obj_pool = None
+        if object_dir:
+            from concurrent.futures import ThreadPoolExecutor
+            from .object_tier import ObjectKVTier
+            self.objects = ObjectKVTier(object_dir,
+                                        self.page_elems * elem_bytes)
+            self._obj_pool = ThreadPoolExecutor(max_workers=1)
+        self.stats["published_object"] = 0
+        self.stats["onboarded_object"] = 0
 
     def _plane_ids(self, pid: int) -> torch.Tensor:
         ids = [k * self.P + pid for k in range(self.planes)]
         return torch.tensor(ids, dtype=torch.int32, device=self.device)
 
     def contains(self, h: int) -> bool:
-        return h in self.map or (self.disk is not None
-                                 and self.disk.contains(h))
+        return (h in self.map
+                or (self.disk is not None and self.disk.contains(h))
+                or (self.objects is not None and self.objects.contains(h)))
 
     def _page_bytes(self, hp: int) -> bytes:
         if self.device.type == "cuda":
@@ -86,8 +100,12 @@ class HostKVTier:
         return None
 
     def _onboard_from_disk(self, h: int) -> Optional[int]:
-        """Promote a disk page back into a host slot; returns host page."""
+        """Promote a G3/G4 page back into a host slot; returns host page."""
         data = self.disk.get(h) if self.disk is not None else None
+        if data is None and self.objects is not None:
+            data = self.objects.get(h)
+            if data is not None:
+                self.stats["onboarded_object"] += 1
         if data is None:
             return None
         hp = self._alloc_host()
@@ -121,6 +139,10 @@ class HostKVTier:
         self.map.move_to_end(h)
         self.stats["offloaded"] += 1
         self.events.append(("stored_host", h))
+        if self.objects is not None and not self.objects.contains(h):
+            data = self._page_bytes(hp)
+            self._obj_pool.submit(self.objects.put, h, data)
+            self.stats["published_object"] += 1
 
     # -- host -> device (prefix-cache onboard) ---------------------------
     def onboard(self, h: int, pid: int) -> bool:

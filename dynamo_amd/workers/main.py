@@ -48,6 +48,9 @@ def build_parser():
                    help="KVBM G3 disk tier size in pages")
     p.add_argument("--disk-cache-path", default="",
                    help="G3 backing file (default <tmp>/dynamo_kv_g3.bin)")
+    p.add_argument("--object-cache-dir", default="",
+                   help="G4 shared object-store directory (cross-worker "
+                        "KV reuse; disabled when empty)")
     p.add_argument("--gms", action="store_true",
                    help="import weights zero-copy from a GMS weight server")
     return p
@@ -92,6 +95,7 @@ def make_engine_from_args(args) -> LLMEngine:
         worker_type=args.worker_type,
         host_cache_pages=args.host_cache_pages,
         disk_cache_pages=args.disk_cache_pages,
+        object_cache_dir=args.object_cache_dir,
         disk_cache_path=(args.disk_cache_path or
                          (os.path.join(tempfile.gettempdir(),
                                        f"dynamo_kv_g3_{os.getpid()}.bin")
