@@ -79,3 +79,33 @@ def test_gpu_embedding_request():
     v2 = embed_with(64)
     assert torch.allclose(v1, v2, rtol=3e-2, atol=3e-3), \
         (v1 - v2).abs().max().item()
+
+
+def test_gpu_object_tier_cross_engine(tmp_path):
+    """G4 on GPU: pages published by one engine onboard into a second
+    engine via the shared object store, output-exact."""
+    import time
+    store = str(tmp_path / "store")
+
+    def make(seed=7):
+        cfg = EngineConfig(model=PRESETS["tiny-llama-gpu"], device="cuda:0",
+                           max_num_seqs=4, max_batched_tokens=1024,
+                           max_model_len=2048, kv_pool_pages=12, page_size=64,
+                           host_cache_pages=8, object_cache_dir=store)
+        return LLMEngine(cfg, seed=seed)
+
+    p1 = list(range(256))
+    e_a = make()
+    o_a = generate(e_a, "a", p1)
+    for i in range(3):
+        generate(e_a, f"c{i}", [(300 + i * 320 + j) % 1000 for j in range(320)])
+    assert e_a.host_tier.stats["published_object"] > 0
+    for _ in range(100):
+        if e_a.host_tier.objects.stats["put"] >= \
+                e_a.host_tier.stats["published_object"]:
+            break
+        time.sleep(0.02)
+    e_b = make()
+    o_b = generate(e_b, "b", p1)
+    assert e_b.host_tier.stats["onboarded_object"] > 0
+    assert o_b == o_a
