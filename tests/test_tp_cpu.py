@@ -146,3 +146,20 @@ def test_moe_expert_parallel_matches_single():
         p.join(60)
         assert p.exitcode == 0
     assert ep == single, f"EP {ep} != single {single}"
+
+
+@pytest.mark.timeout(300)
+def test_tp4_matches_tp1():
+    """TP-4 lockstep == TP-1 (the 70B disagg pools run TP4 groups)."""
+    tp1 = _gen_tp1()
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_tp_rank, args=(r, 4, 29613, q))
+             for r in range(4)]
+    for p in procs:
+        p.start()
+    tp4 = q.get(timeout=240)
+    for p in procs:
+        p.join(60)
+        assert p.exitcode == 0
+    assert tp4 == tp1, f"TP4 {tp4} != TP1 {tp1}"
