@@ -144,8 +144,10 @@ __global__ __launch_bounds__(kBlock) void paged_decode_phase1(
           for (int e = 0; e < 4; e++)
             d = __builtin_amdgcn_fdot2_f32_bf16(k2[e], q2[e], d, false);
         }
-#pragma unroll
-        for (int off = 1; off < DP; off <<= 1) d += __shfl_xor(d, off, WAVE_SIZE);
+        // DP-lane dot fold: pure-DPP butterflies (ds_bpermute chains
+        // serialize on the LDS unit; measured lever on the MFMA variant)
+        if constexpr (DP == 8) d = group8_reduce_sum(d);
+        else d = row16_reduce_sum(d);
         s[g] = valid ? d * scale : kNegInf;
       }
       float vf[ND];
@@ -156,9 +158,10 @@ __global__ __launch_bounds__(kBlock) void paged_decode_phase1(
 #pragma unroll
       for (int g = 0; g < GW; g++) {
         float ms = s[g];
-#pragma unroll
-        for (int off = DP; off < 64; off <<= 1)
-          ms = fmaxf(ms, __shfl_xor(ms, off, WAVE_SIZE));
+        if constexpr (DP == 8)   // fold the two 8-groups of the row first
+          ms = fmaxf(ms, dpp_movf<0x128>(ms));
+        ms = fmaxf(ms, __shfl_xor(ms, 16, WAVE_SIZE));
+        ms = fmaxf(ms, __shfl_xor(ms, 32, WAVE_SIZE));
         if (ms > m[g]) {
           const float corr = __expf(m[g] - ms);
           l[g] *= corr;
@@ -168,9 +171,10 @@ __global__ __launch_bounds__(kBlock) void paged_decode_phase1(
         }
         const float p = (s[g] > kNegInf * 0.5f) ? __expf(s[g] - m[g]) : 0.f;
         float psum = p;
-#pragma unroll
-        for (int off = DP; off < 64; off <<= 1)
-          psum += __shfl_xor(psum, off, WAVE_SIZE);
+        if constexpr (DP == 8)
+          psum += dpp_movf<0x128>(psum);
+        psum += __shfl_xor(psum, 16, WAVE_SIZE);
+        psum += __shfl_xor(psum, 32, WAVE_SIZE);
         l[g] += psum;
 #pragma unroll
         for (int i = 0; i < ND; i++) acc[g][i] = fmaf(p, vf[i], acc[g][i]);

@@ -116,6 +116,24 @@ DEVINL float row16_reduce_sum(float x) {
   return x;
 }
 
+// 8-lane-group reductions via XOR-mask butterflies in DPP: quad_perm
+// xor1 (0xB1) + xor2 (0x4E) + row_half_mirror (= xor7 within the 8-lane
+// half, 0x141). Masks {1,2,7} span bits 0-2, so every lane ends with the
+// full 8-group value. Pure VALU — no ds_bpermute.
+DEVINL float group8_reduce_sum(float x) {
+  x += dpp_movf<0xB1>(x);
+  x += dpp_movf<0x4E>(x);
+  x += dpp_movf<0x141>(x);
+  return x;
+}
+
+DEVINL float group8_reduce_max(float x) {
+  x = fmaxf(x, dpp_movf<0xB1>(x));
+  x = fmaxf(x, dpp_movf<0x4E>(x));
+  x = fmaxf(x, dpp_movf<0x141>(x));
+  return x;
+}
+
 #define HIP_CHECK_KERNEL()                                                \
   do {                                                                    \
     hipError_t _e = hipGetLastError();                                    \
