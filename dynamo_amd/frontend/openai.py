@@ -59,6 +59,7 @@ class EmbeddingRequest(BaseModel):
 class ChatRequest(BaseModel):
     model: str = ""
     messages: List[ChatMessage] = Field(default_factory=list)
+    tools: Optional[List[dict]] = None   # enables tool-call parsing
     max_tokens: int = 128
     temperature: float = 0.0
     top_p: float = 1.0
@@ -474,14 +475,26 @@ def build_app(manager: ModelManager) -> FastAPI:
                     finish = "stop"
                     break
         LATENCY.labels(entry.name).observe(time.time() - t0)
+        text_out = (text_cut if text_cut is not None
+                    else entry.tokenizer.decode(produced))
+        # postprocessing parity (preprocessor.rs:3953/:4577): reasoning
+        # split always; tool-call extraction only when tools were offered
+        from .parsers import parse_reasoning, parse_tool_calls
+        text_out, reasoning = parse_reasoning(text_out)
+        tool_calls = []
+        if req.tools:
+            text_out, tool_calls = parse_tool_calls(text_out)
+        message = {"role": "assistant", "content": text_out or None}
+        if reasoning:
+            message["reasoning_content"] = reasoning
+        if tool_calls:
+            message["tool_calls"] = tool_calls
         return {
             "id": rid, "object": "chat.completion", "created": int(t0),
             "model": entry.name,
-            "choices": [{"index": 0, "message": {
-                "role": "assistant",
-                "content": (text_cut if text_cut is not None
-                            else entry.tokenizer.decode(produced))},
-                "finish_reason": finish or "stop"}],
+            "choices": [{"index": 0, "message": message,
+                "finish_reason": ("tool_calls" if tool_calls
+                                  else finish or "stop")}],
             "usage": {"prompt_tokens": len(token_ids),
                       "completion_tokens": len(produced),
                       "total_tokens": len(token_ids) + len(produced)},
