@@ -171,6 +171,18 @@ def greedy_sample(logits: torch.Tensor) -> torch.Tensor:
     return torch_ref.greedy_sample(logits)
 
 
+def _splitmix64(x: torch.Tensor) -> torch.Tensor:
+    """splitmix64 finalizer on int64 tensors (wrapping arithmetic; torch
+    >> is arithmetic, so logical shifts are emulated with masks). Bit-for-
+    bit the same hash as the HIP sampling kernel's hash_u64."""
+    def lsr(v, k):
+        return (v >> k) & ((1 << (64 - k)) - 1)
+    x = x + 0x9e3779b97f4a7c15
+    x = (x ^ lsr(x, 30)) * -0x40a7b892e31b1a47   # 0xbf58476d1ce4e5b9
+    x = (x ^ lsr(x, 27)) * -0x6b2fb644ecceee15   # 0x94d049bb133111eb
+    return x ^ lsr(x, 31)
+
+
 def gumbel_sample(logits: torch.Tensor, inv_temp: torch.Tensor,
                   seed: int) -> torch.Tensor:
     if logits.is_cuda:
@@ -178,8 +190,15 @@ def gumbel_sample(logits: torch.Tensor, inv_temp: torch.Tensor,
                           device=logits.device)
         hip().gumbel_sample(out, logits.float().contiguous(), inv_temp, seed)
         return out
-    # CPU: exact same semantics via torch gumbel noise (not bit-identical)
-    g = -torch.log(-torch.log(torch.rand_like(logits.float())))
+    # CPU: DETERMINISTIC counter-based Gumbel noise — the same
+    # splitmix64(seed ^ b<<32 ^ v) stream as the HIP kernel, so sampling
+    # is reproducible across engines/processes (TP lockstep needs this)
+    B, V = logits.shape
+    b = torch.arange(B, dtype=torch.int64).unsqueeze(1) << 32
+    v = torch.arange(V, dtype=torch.int64).unsqueeze(0)
+    h = _splitmix64(torch.tensor(seed, dtype=torch.int64) ^ b ^ v)
+    u = (((h >> 11) & ((1 << 53) - 1)) + 1).double() * (2.0 ** -53)
+    g = -torch.log(-torch.log(u.float()))
     return (logits.float() * inv_temp.unsqueeze(-1) + g).argmax(-1).to(torch.int32)
 
 
