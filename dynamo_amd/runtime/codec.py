@@ -16,6 +16,10 @@ import xxhash
 
 PREFIX = struct.Struct("<QQQ")
 MAX_FRAME = 1 << 30
+# thin native clients (capi/dynamo_client.hpp) may send this sentinel to
+# skip checksum verification — TCP already guarantees integrity; xxh3 is
+# only a corruption tripwire for file/relay transports
+UNCHECKED = 0xFFFFFFFFFFFFFFFF
 
 
 def encode_frame(header: dict, body: Any) -> bytes:
@@ -30,7 +34,7 @@ def decode_prefix(buf: bytes) -> Tuple[int, int, int]:
 
 
 def decode_frame(h: bytes, b: bytes, csum: int) -> Tuple[dict, Any]:
-    if xxhash.xxh3_64_intdigest(h + b) != csum:
+    if csum != UNCHECKED and xxhash.xxh3_64_intdigest(h + b) != csum:
         raise ValueError("frame checksum mismatch")
     return json.loads(h.decode()), msgpack.unpackb(b, raw=False)
 
