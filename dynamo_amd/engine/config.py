@@ -128,6 +128,11 @@ class EngineConfig:
     disk_cache_path: str = ""           # G3 backing file (required if G3 on)
     object_cache_dir: str = ""          # G4 shared object store (disabled="")
     kv_events: bool = True              # emit stored/removed block events
+    # waiting-queue admission policy (reference parity: kv-router
+    # scheduling/policy.rs SchedulingPolicy FCFS/LCFS/WSPT):
+    #   fcfs = arrival order; lcfs = newest first; wspt = shortest
+    #   remaining prompt first (weighted shortest processing time)
+    queue_policy: str = "fcfs"          # fcfs | lcfs | wspt
     block_salt: int = 0
     # disaggregation
     worker_type: str = "aggregated"     # aggregated | prefill | decode

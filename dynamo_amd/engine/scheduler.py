@@ -195,7 +195,8 @@ class Scheduler:
             (out.decodes if (req.is_decode and n == 1) else out.prefills).append(item)
             budget -= n
 
-        # 2) admit waiting requests
+        # 2) admit waiting requests (queue policy decides who goes first)
+        self._order_waiting()
         while (self.waiting and budget > 0
                and len(self.running) < self.cfg.max_num_seqs):
             req = self.waiting[0]
@@ -218,6 +219,17 @@ class Scheduler:
             budget -= n
 
         return out
+
+    def _order_waiting(self):
+        pol = getattr(self.cfg, "queue_policy", "fcfs")
+        if pol == "fcfs" or len(self.waiting) < 2:
+            return
+        if pol == "lcfs":
+            self.waiting.sort(key=lambda r: -r.arrival)
+        elif pol == "wspt":
+            # shortest remaining prefill first; arrival breaks ties
+            self.waiting.sort(
+                key=lambda r: (r.total_len - r.num_computed, r.arrival))
 
     # ------------------------------------------------------------------
     def finish(self, req: Request, reason: str):

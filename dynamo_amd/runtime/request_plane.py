@@ -42,8 +42,12 @@ class RequestPlaneServer:
     """Serves named endpoints; handlers are async generators of chunks."""
 
     def __init__(self, host: str = "127.0.0.1", port: int = 0):
+        # host "unix:/path" serves on a Unix domain socket (second
+        # transport next to TCP — reference has a pluggable transport
+        # matrix, runtime/src/transports)
         self.host = host
         self.port = port
+        self.uds_path = host[5:] if host.startswith("unix:") else None
         self.handlers: Dict[str, Handler] = {}
         self._server: Optional[asyncio.AbstractServer] = None
         self._inflight: Dict[tuple, asyncio.Task] = {}
@@ -56,9 +60,15 @@ class RequestPlaneServer:
 
     @property
     def address(self) -> str:
+        if self.uds_path is not None:
+            return f"unix:{self.uds_path}"
         return f"{self.host}:{self.port}"
 
     async def start(self):
+        if self.uds_path is not None:
+            self._server = await asyncio.start_unix_server(self._on_conn,
+                                                           self.uds_path)
+            return self.address
         self._server = await asyncio.start_server(self._on_conn, self.host,
                                                   self.port)
         self.port = self._server.sockets[0].getsockname()[1]
@@ -204,8 +214,13 @@ class RequestPlaneClient:
         async with self._lock:
             c = self._conns.get(address)
             if c is None or c.closed:
-                host, port = address.rsplit(":", 1)
-                reader, writer = await asyncio.open_connection(host, int(port))
+                if address.startswith("unix:"):
+                    reader, writer = await asyncio.open_unix_connection(
+                        address[5:])
+                else:
+                    host, port = address.rsplit(":", 1)
+                    reader, writer = await asyncio.open_connection(
+                        host, int(port))
                 c = _Conn(reader, writer)
                 self._conns[address] = c
             return c

@@ -319,3 +319,35 @@ def test_prompt_embeds_chunked():
         for so in e.step():
             out.append(so.new_token)
     assert out == ref
+
+
+def test_queue_policies():
+    """Admission-queue policies (SchedulingPolicy FCFS/LCFS/WSPT parity):
+    with room for one request at a time, the policy decides admission
+    order — observed via first-token emission order."""
+    def first_token_order(policy):
+        cfg = EngineConfig(model=PRESETS["tiny-llama"], device="cpu",
+                           max_num_seqs=1, max_batched_tokens=512,
+                           max_model_len=512, kv_pool_pages=64, page_size=16,
+                           queue_policy=policy, enable_prefix_caching=False)
+        e = LLMEngine(cfg, seed=7)
+        # arrival order: long, medium, short
+        e.add_request("long", list(range(20, 180)),
+                      SamplingParams(max_tokens=2))
+        e.add_request("mid", list(range(30, 110)),
+                      SamplingParams(max_tokens=2))
+        e.add_request("short", list(range(40, 60)),
+                      SamplingParams(max_tokens=2))
+        order = []
+        steps = 0
+        while e.has_work():
+            for so in e.step():
+                if so.num_output_tokens == 1:
+                    order.append(so.req_id)
+            steps += 1
+            assert steps < 200
+        return order
+
+    assert first_token_order("fcfs") == ["long", "mid", "short"]
+    assert first_token_order("lcfs") == ["short", "mid", "long"]
+    assert first_token_order("wspt") == ["short", "mid", "long"]

@@ -172,3 +172,31 @@ def test_file_discovery_lease(tmp_path):
     assert len(d.list("ns", "backend")) == 1
     d.deregister(inst)
     assert d.list("ns", "backend") == []
+
+
+def test_request_plane_unix_socket(tmp_path):
+    """UDS transport: the same codec/endpoints over a Unix domain socket."""
+    import asyncio
+
+    from dynamo_amd.runtime.request_plane import (RequestPlaneClient,
+                                                  RequestPlaneServer)
+
+    async def main():
+        server = RequestPlaneServer(host=f"unix:{tmp_path}/rp.sock")
+
+        async def echo(payload, ctx):
+            for i in range(3):
+                yield {"i": i, "got": payload}
+
+        server.add_endpoint("echo", echo)
+        addr = await server.start()
+        assert addr.startswith("unix:")
+        client = RequestPlaneClient()
+        chunks = []
+        async for ch in client.call_stream(addr, "echo", {"x": 42}):
+            chunks.append(ch)
+        assert [c["i"] for c in chunks] == [0, 1, 2]
+        assert chunks[0]["got"] == {"x": 42}
+        await client.close()
+        await server.stop(drain=False)
+    asyncio.new_event_loop().run_until_complete(main())
