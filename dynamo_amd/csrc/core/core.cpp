@@ -154,12 +154,106 @@ class KvIndexer {
   std::unordered_map<int64_t, std::unordered_set<uint64_t>> worker_blocks_;
 };
 
+// ---------------------------------------------------------------------------
+// Cuckoo filter: compact approximate-membership digest for KV block hashes
+// (reference parity: lib/kv-router cuckoo.rs + the kv_dc_relay digests that
+// summarize a pool's cached prefixes for cross-pool routing). 2 candidate
+// buckets x 4 slots, 16-bit fingerprints; standard partial-key cuckoo
+// relocation with a bounded kick chain.
+class CuckooFilter {
+ public:
+  explicit CuckooFilter(size_t capacity) {
+    size_t nb = 1;
+    while (nb * 4 < capacity * 2) nb <<= 1;   // ~50% target load headroom
+    buckets_.assign(nb * 4, 0);
+    nbuckets_ = nb;
+  }
+
+  static uint64_t mix(uint64_t x) {
+    x += 0x9e3779b97f4a7c15ull;
+    x = (x ^ (x >> 30)) * 0xbf58476d1ce4e5b9ull;
+    x = (x ^ (x >> 27)) * 0x94d049bb133111ebull;
+    return x ^ (x >> 31);
+  }
+
+  bool insert(uint64_t h) {
+    uint16_t fp = fingerprint(h);
+    size_t i1 = bucket1(h), i2 = alt(i1, fp);
+    if (place(i1, fp) || place(i2, fp)) {
+      ++count_;
+      return true;
+    }
+    size_t i = (mix(h ^ 0x1234) & 1) ? i2 : i1;
+    for (int kick = 0; kick < 512; ++kick) {
+      size_t slot = i * 4 + (mix(h + kick) & 3);
+      std::swap(fp, buckets_[slot]);
+      i = alt(i, fp);
+      if (place(i, fp)) {
+        ++count_;
+        return true;
+      }
+    }
+    return false;   // table effectively full
+  }
+
+  bool contains(uint64_t h) const {
+    uint16_t fp = fingerprint(h);
+    size_t i1 = bucket1(h), i2 = alt(i1, fp);
+    for (int s = 0; s < 4; ++s)
+      if (buckets_[i1 * 4 + s] == fp || buckets_[i2 * 4 + s] == fp)
+        return true;
+    return false;
+  }
+
+  size_t count() const { return count_; }
+  size_t memory_bytes() const { return buckets_.size() * sizeof(uint16_t); }
+
+  int64_t max_prefix(const std::vector<uint64_t>& chain) const {
+    int64_t n = 0;
+    for (uint64_t h : chain) {
+      if (!contains(h)) break;
+      ++n;
+    }
+    return n;
+  }
+
+ private:
+  uint16_t fingerprint(uint64_t h) const {
+    uint16_t fp = (uint16_t)(mix(h) >> 48);
+    return fp ? fp : 1;   // 0 means empty slot
+  }
+  size_t bucket1(uint64_t h) const { return mix(h ^ 0xABCD) & (nbuckets_ - 1); }
+  size_t alt(size_t i, uint16_t fp) const {
+    return (i ^ (mix(fp) & (nbuckets_ - 1))) & (nbuckets_ - 1);
+  }
+  bool place(size_t i, uint16_t fp) {
+    for (int s = 0; s < 4; ++s) {
+      if (buckets_[i * 4 + s] == 0) {
+        buckets_[i * 4 + s] = fp;
+        return true;
+      }
+    }
+    return false;
+  }
+
+  std::vector<uint16_t> buckets_;
+  size_t nbuckets_ = 0;
+  size_t count_ = 0;
+};
+
 PYBIND11_MODULE(_core, m) {
   m.doc() = "dynamo_amd native C++ control-plane components";
   m.def("chain_hashes", &chain_hashes, py::arg("tokens"), py::arg("block_size"),
         py::arg("salt") = 0);
   m.def("hash_block", &hash_block, py::arg("parent"), py::arg("tokens"));
   m.def("chain_root", &chain_root, py::arg("salt") = 0);
+  py::class_<CuckooFilter>(m, "CuckooFilter")
+      .def(py::init<size_t>(), py::arg("capacity"))
+      .def("insert", &CuckooFilter::insert)
+      .def("contains", &CuckooFilter::contains)
+      .def("count", &CuckooFilter::count)
+      .def("memory_bytes", &CuckooFilter::memory_bytes)
+      .def("max_prefix", &CuckooFilter::max_prefix);
   py::class_<KvIndexer>(m, "KvIndexer")
       .def(py::init<>())
       .def("apply_stored", &KvIndexer::apply_stored)

@@ -46,3 +46,33 @@ def test_indexer_no_match():
     idx.apply_stored(1, [123, 456])
     assert idx.find_matches([999]) == {}
     assert idx.find_matches([]) == {}
+
+
+def test_cuckoo_filter():
+    """C++ cuckoo digest (kv-router cuckoo.rs parity): no false negatives,
+    low false-positive rate, prefix-walk helper, compact memory."""
+    import random
+
+    from dynamo_amd import _core
+    f = _core.CuckooFilter(10_000)
+    rng = random.Random(0)
+    inserted = [rng.getrandbits(63) for _ in range(8_000)]
+    for h in inserted:
+        assert f.insert(h)
+    assert f.count() == 8_000
+    # no false negatives
+    assert all(f.contains(h) for h in inserted)
+    # false positives bounded (16-bit fingerprints -> ~0.1% expected)
+    probes = [rng.getrandbits(63) for _ in range(20_000)]
+    fp = sum(1 for h in probes if f.contains(h))
+    assert fp < 100, f"false-positive rate too high: {fp}/20000"
+    # compact: ~4 bytes/key at this capacity
+    assert f.memory_bytes() < 40 * 8_000
+    # prefix walk over a hash chain (the digest question the global router
+    # asks: how deep does this pool's cache cover the chain)
+    chain = _core.chain_hashes(list(range(640)), 64, 0)
+    for h in chain[:7]:
+        f.insert(h)
+    assert f.max_prefix(chain) >= 7
+    empty = _core.CuckooFilter(128)
+    assert empty.max_prefix(chain) == 0
