@@ -803,3 +803,43 @@ def test_graceful_shutdown_drains():
         await rt.shutdown(drain=False)
         await client.aclose()
     run(main())
+
+
+def test_disagg_prefill_worker_death_recovers():
+    """Killing the prefill pool mid-request: the migration loop retries
+    and the request completes via the decode pool (bypass), so disagg
+    adds no availability cliff."""
+    async def main():
+        shared, services, mgr, client = await with_stack(nworkers=1,
+                                                         disagg=True)
+        entry = mgr.get("mock-model")
+        for _ in range(50):
+            if entry.prefill_router is not None:
+                break
+            await asyncio.sleep(0.05)
+        pf_ws, pf_rt = services[-1]
+        # slow the prefill engine so we can kill it mid-prefill
+        pf_ws.engine.runner.prefill_tps = 2000.0
+        prompt = "z" * 3000           # forces the disagg path
+
+        async def kill_prefill():
+            await asyncio.sleep(0.3)  # while the prefill is in flight
+            await pf_ws.stop()
+            await pf_rt.shutdown(drain=False)
+
+        killer = asyncio.create_task(kill_prefill())
+        r = await client.post("/v1/completions", json={
+            "model": "mock-model", "prompt": prompt, "max_tokens": 5})
+        await killer
+        assert r.status_code == 200, r.text
+        assert r.json()["usage"]["completion_tokens"] == 5
+        # later requests keep working without the prefill pool
+        r = await client.post("/v1/completions", json={
+            "model": "mock-model", "prompt": prompt, "max_tokens": 3})
+        assert r.status_code == 200
+        await client.aclose()
+        await mgr.stop()
+        for ws, rt in services[:-1]:
+            await ws.stop()
+            await rt.shutdown(drain=False)
+    run(main())
