@@ -53,6 +53,9 @@ def build_parser():
                         "KV reuse; disabled when empty)")
     p.add_argument("--gms", action="store_true",
                    help="import weights zero-copy from a GMS weight server")
+    p.add_argument("--status-port", type=int, default=0,
+                   help="serve a JSON system-status HTTP endpoint on this "
+                        "port (reference DYN_SYSTEM_PORT parity; 0 = off)")
     return p
 
 
@@ -113,6 +116,13 @@ async def async_main(args):
     ws = WorkerService(engine, rt, namespace=args.namespace,
                        component=component)
     await ws.start()
+    status_runner = None
+    if args.status_port:
+        from .status import start_status_server
+        status_runner = await start_status_server(ws, args.host,
+                                                  args.status_port)
+        print(f"STATUS_READY http://{args.host}:{args.status_port}",
+              flush=True)
     logging.info("worker %s (%s) serving %s on %s", ws.instance_id,
                  args.worker_type, args.model, rt.server.address)
     print(f"WORKER_READY {ws.instance_id} {rt.server.address}", flush=True)
@@ -122,6 +132,8 @@ async def async_main(args):
     for sig in (signal.SIGINT, signal.SIGTERM):
         loop.add_signal_handler(sig, stop.set)
     await stop.wait()
+    if status_runner is not None:
+        await status_runner()
     await ws.stop()
     await rt.shutdown()
 

@@ -111,3 +111,35 @@ def test_record_and_replay(tmp_path):
             front.stop()
         for w in workers:
             w.stop()
+
+
+@pytest.mark.timeout(120)
+def test_worker_status_server(tmp_path):
+    """--status-port serves liveness + engine state JSON out-of-band."""
+    import socket
+    with socket.socket() as s:
+        s.bind(("127.0.0.1", 0))
+        port = s.getsockname()[1]
+    disc = f"file:{tmp_path}/disc"
+    w = ManagedProcess(
+        worker_cmd(mock=True, model="tiny-llama", discovery=disc,
+                   page_size=16) + ["--status-port", str(port)],
+        ready_marker="STATUS_READY").start()
+    try:
+        with httpx.Client(timeout=10) as c:
+            deadline = time.time() + 30
+            body = None
+            while time.time() < deadline:
+                try:
+                    body = c.get(f"http://127.0.0.1:{port}/status").json()
+                    break
+                except httpx.TransportError:
+                    time.sleep(0.2)
+            assert body is not None
+            assert body["status"] == "ok"
+            assert body["model"] == "tiny-llama"
+            assert body["worker_type"] == "aggregated"
+            assert body["paused"] is False
+            assert body["total_kv_pages"] > 0
+    finally:
+        w.stop()
