@@ -200,3 +200,71 @@ def test_request_plane_unix_socket(tmp_path):
         await client.close()
         await server.stop(drain=False)
     asyncio.new_event_loop().run_until_complete(main())
+
+
+def test_codec_fuzz_roundtrip():
+    """Codec survives arbitrary nested payloads + the UNCHECKED sentinel."""
+    import random
+
+    from dynamo_amd.runtime.codec import (UNCHECKED, decode_frame,
+                                          decode_prefix, encode_frame)
+
+    rng = random.Random(42)
+
+    def rand_value(depth=0):
+        k = rng.randrange(7 if depth < 3 else 4)
+        if k == 0:
+            return rng.randrange(-2**40, 2**40)
+        if k == 1:
+            return "".join(chr(rng.randrange(32, 0x2FF))
+                           for _ in range(rng.randrange(0, 40)))
+        if k == 2:
+            return rng.random()
+        if k == 3:
+            return rng.choice([None, True, False])
+        if k == 4:
+            return bytes(rng.randrange(256) for _ in range(rng.randrange(64)))
+        if k == 5:
+            return [rand_value(depth + 1) for _ in range(rng.randrange(5))]
+        return {f"k{i}": rand_value(depth + 1)
+                for i in range(rng.randrange(5))}
+
+    for _ in range(100):
+        header = {"type": "rsp", "rid": rng.randrange(1 << 30),
+                  "final": rng.random() < 0.5}
+        body = rand_value()
+        frame = encode_frame(header, body)
+        hlen, blen, csum = decode_prefix(frame[:24])
+        h2, b2 = decode_frame(frame[24:24 + hlen],
+                              frame[24 + hlen:24 + hlen + blen], csum)
+        assert h2 == header and b2 == body
+        # thin-client sentinel skips verification
+        h3, b3 = decode_frame(frame[24:24 + hlen],
+                              frame[24 + hlen:24 + hlen + blen], UNCHECKED)
+        assert b3 == body
+        # corrupted checksum is rejected
+        try:
+            decode_frame(frame[24:24 + hlen],
+                         frame[24 + hlen:24 + hlen + blen], csum ^ 1)
+            assert False, "corruption not detected"
+        except ValueError:
+            pass
+
+
+def test_file_discovery_ttl_expiry(tmp_path):
+    """Instances whose lease file goes stale past the TTL disappear."""
+    import os
+    import time
+
+    from dynamo_amd.runtime.discovery import FileDiscovery, Instance
+    d = FileDiscovery(str(tmp_path), ttl=0.3)
+    d.register(Instance(namespace="dynamo", component="backend",
+                        instance_id="abc123", address="127.0.0.1:1"))
+    assert [i.instance_id for i in d.list("dynamo")] == ["abc123"]
+    # age the lease file beyond the TTL without touching it
+    for root, _, files in os.walk(tmp_path):
+        for f in files:
+            p = os.path.join(root, f)
+            old = time.time() - 5
+            os.utime(p, (old, old))
+    assert d.list("dynamo") == []
