@@ -153,14 +153,20 @@ class WorkerService:
                 q = self.queues.get(so.req_id)
                 if q is not None:
                     q.put_nowait(so)
-            # fan KV events to subscribers (drain regardless, to bound memory)
-            events = self.engine.drain_kv_events()
-            if events and self.kv_event_subs:
-                batch = [{"kind": e.kind, "hashes": list(e.hashes),
-                          "parent": e.parent} for e in events]
-                for q in self.kv_event_subs:
-                    q.put_nowait(batch)
+            self._fan_kv_events()
             await asyncio.sleep(0)
+
+    def _fan_kv_events(self):
+        """Drain engine KV events to subscribers — called from the engine
+        loop AND from endpoints that mutate KV while the engine is idle
+        (clear_kv_blocks/release_kv), else their events would stall until
+        the next step."""
+        events = self.engine.drain_kv_events()
+        if events and self.kv_event_subs:
+            batch = [{"kind": e.kind, "hashes": list(e.hashes),
+                      "parent": e.parent} for e in events]
+            for q in self.kv_event_subs:
+                q.put_nowait(batch)
 
     # ------------------------------------------------------------------
     async def generate(self, payload: dict, ctx: RequestContext
@@ -302,6 +308,7 @@ class WorkerService:
     async def clear_kv_blocks(self, payload, ctx):
         async with self._engine_lock:
             self.engine.clear_kv()
+        self._fan_kv_events()
         yield {"status": "ok"}
 
     async def get_perf_metrics(self, payload, ctx):
@@ -347,6 +354,7 @@ class WorkerService:
     async def release_kv(self, payload, ctx):
         async with self._engine_lock:
             self.engine.release_held(payload["request_id"])
+        self._fan_kv_events()
         yield {"status": "ok"}
 
     # -- snapshot lifecycle: pause -> (snapshot externally) -> resume ----

@@ -843,3 +843,36 @@ def test_disagg_prefill_worker_death_recovers():
             await ws.stop()
             await rt.shutdown(drain=False)
     run(main())
+
+
+def test_clear_kv_blocks_endpoint():
+    """clear_kv_blocks: flushes the prefix cache and emits `cleared` so the
+    router drops this worker's index (vllm main.py clear_kv_blocks parity)."""
+    async def main():
+        shared, services, mgr, client = await with_stack(nworkers=1)
+        entry = mgr.get("mock-model")
+        # populate the prefix cache + router index
+        r = await client.post("/v1/completions", json={
+            "model": "mock-model", "prompt": "c" * 200, "max_tokens": 2})
+        assert r.status_code == 200
+        for _ in range(50):
+            if entry.router.indexer.size() > 0:
+                break
+            await asyncio.sleep(0.05)
+        assert entry.router.indexer.size() > 0
+        inst = mgr.runtime.discovery.list("dynamo")[0]
+        resp = await mgr.runtime.client.call(
+            inst.address, "backend.clear_kv_blocks", {})
+        assert resp["status"] == "ok"
+        # the cleared event reaches the router and empties its index
+        for _ in range(100):
+            if entry.router.indexer.size() == 0:
+                break
+            await asyncio.sleep(0.05)
+        assert entry.router.indexer.size() == 0
+        # serving continues after the flush
+        r = await client.post("/v1/completions", json={
+            "model": "mock-model", "prompt": "after", "max_tokens": 2})
+        assert r.status_code == 200
+        await teardown(services, mgr, client)
+    run(main())
