@@ -876,3 +876,29 @@ def test_clear_kv_blocks_endpoint():
         assert r.status_code == 200
         await teardown(services, mgr, client)
     run(main())
+
+
+def test_kv_events_snapshot_on_subscribe():
+    """A LATE subscriber to kv_events receives a snapshot of already-stored
+    blocks first (so a restarted router rebuilds its index without waiting
+    for new traffic)."""
+    async def main():
+        shared, services, mgr, client = await with_stack(nworkers=1)
+        r = await client.post("/v1/completions", json={
+            "model": "mock-model", "prompt": "s" * 200, "max_tokens": 2})
+        assert r.status_code == 200
+        await asyncio.sleep(0.2)   # let the engine loop commit pages
+        # fresh runtime subscribes AFTER the fact
+        late_rt = DistributedRuntime(shared)
+        await late_rt.start()
+        inst = late_rt.discovery.list("dynamo")[0]
+        agen = late_rt.client.call_stream(
+            inst.address, "backend.kv_events", {})
+        batch = await asyncio.wait_for(agen.__anext__(), timeout=10)
+        stored = [ev for ev in batch if ev["kind"] == "stored"]
+        assert stored, "late subscriber got no snapshot of stored blocks"
+        assert all(ev["hashes"] for ev in stored)
+        await agen.aclose()
+        await late_rt.shutdown(drain=False)
+        await teardown(services, mgr, client)
+    run(main())
