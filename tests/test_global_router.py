@@ -87,7 +87,10 @@ def test_global_router_routing_and_failover():
             # failover: kill pool A; requests for model-a now 503 (no other
             # pool serves it), model-b still fine
             await stop_pool(pa)
-            await asyncio.sleep(0.8)   # health check notices
+            for _ in range(100):       # poll until the health watch notices
+                if not router.pools[pa["url"]].healthy:
+                    break
+                await asyncio.sleep(0.1)
             r = await client.post("/v1/completions", json={
                 "model": "model-a", "prompt": [1], "max_tokens": 1})
             assert r.status_code in (502, 503)
@@ -129,7 +132,7 @@ def test_global_router_kv_aware():
                     "model": "mock-model", "prompt": prompt, "max_tokens": 2})
                 assert r.status_code == 200
                 # wait for KV events to reach pool B's router indexer
-                for _ in range(50):
+                for _ in range(150):    # generous under machine load
                     r = await direct.post(
                         pb["url"] + "/internal/kv_overlap",
                         json={"model": "mock-model", "token_ids": prompt})
