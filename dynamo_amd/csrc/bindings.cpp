@@ -40,9 +40,10 @@ void attention_prefill_paged(torch::Tensor out, torch::Tensor q,
 void greedy_sample(torch::Tensor out, torch::Tensor logits);
 void topkp_sample(torch::Tensor out, torch::Tensor logits,
                   torch::Tensor inv_temp, torch::Tensor top_k,
-                  torch::Tensor top_p, int64_t seed);
+                  torch::Tensor top_p, int64_t seed,
+                  c10::optional<torch::Tensor> row_seeds);
 void gumbel_sample(torch::Tensor out, torch::Tensor logits, torch::Tensor inv_temp,
-                   int64_t seed);
+                   int64_t seed, c10::optional<torch::Tensor> row_seeds);
 // moe.hip
 void moe_grouped_gemm(torch::Tensor y, torch::Tensor x, torch::Tensor w,
                       torch::Tensor tiles);
@@ -76,8 +77,12 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("paged_attention_decode", &paged_attention_decode);
   m.def("attention_prefill_paged", &attention_prefill_paged);
   m.def("greedy_sample", &greedy_sample);
-  m.def("gumbel_sample", &gumbel_sample);
-  m.def("topkp_sample", &topkp_sample);
+  m.def("gumbel_sample", &gumbel_sample, py::arg("out"), py::arg("logits"),
+        py::arg("inv_temp"), py::arg("seed"),
+        py::arg("row_seeds") = py::none());
+  m.def("topkp_sample", &topkp_sample, py::arg("out"), py::arg("logits"),
+        py::arg("inv_temp"), py::arg("top_k"), py::arg("top_p"),
+        py::arg("seed"), py::arg("row_seeds") = py::none());
   m.def("moe_grouped_gemm", &moe_grouped_gemm);
   m.def("moe_grouped_gemm_seg", &moe_grouped_gemm_seg);
   m.def("topk_gating", &topk_gating);

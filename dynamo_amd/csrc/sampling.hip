@@ -28,17 +28,22 @@ template <bool GUMBEL>
 __global__ void sample_kernel(int32_t* __restrict__ out,          // [B]
                               const float* __restrict__ logits,   // [B, V]
                               const float* __restrict__ inv_temp, // [B] or null
-                              uint64_t seed, int V) {
+                              uint64_t seed,
+                              const uint64_t* __restrict__ row_seeds,  // [B] or null
+                              int V) {
   const int b = blockIdx.x;
   const float* row = logits + (int64_t)b * V;
   const float it = GUMBEL ? inv_temp[b] : 1.f;
+  // per-row seed (client-supplied sampling seed, already host-mixed with
+  // the output position) or the scalar engine-step stream
+  const uint64_t rs = row_seeds ? row_seeds[b] : (seed ^ ((uint64_t)b << 32));
 
   float best = -1e38f;
   int besti = -1;
   for (int v = threadIdx.x; v < V; v += kBlock) {
     float x = row[v] * it;
     if constexpr (GUMBEL) {
-      const uint64_t h = hash_u64(seed ^ ((uint64_t)b << 32) ^ (uint64_t)v);
+      const uint64_t h = hash_u64(rs ^ (uint64_t)v);
       // uniform in (0,1): use top 53 bits
       const float u = (float)((h >> 11) + 1) * 4.8828125e-4f * 2.2737367544323206e-13f;
       x += -__logf(-__logf(u));
@@ -83,8 +88,11 @@ __global__ __launch_bounds__(kBlock) void topkp_sample_kernel(
     const float* __restrict__ inv_temp, // [B]
     const int32_t* __restrict__ top_k,  // [B] (<=0: off)
     const float* __restrict__ top_p,    // [B] (>=1: off)
-    uint64_t seed, int V) {
+    uint64_t seed,
+    const uint64_t* __restrict__ row_seeds,  // [B] or null
+    int V) {
   const int b = blockIdx.x;
+  const uint64_t rs = row_seeds ? row_seeds[b] : (seed ^ ((uint64_t)b << 32));
   const float* row = logits + (int64_t)b * V;
   const int k = top_k[b];
   const float p = top_p[b];
@@ -160,7 +168,7 @@ __global__ __launch_bounds__(kBlock) void topkp_sample_kernel(
   for (int v = threadIdx.x; v < V; v += kBlock) {
     const float pv = __expf(row[v] - m) * inv_z;
     if (pv < thr) continue;
-    const uint64_t h = hash_u64(seed ^ ((uint64_t)b << 32) ^ (uint64_t)v);
+    const uint64_t h = hash_u64(rs ^ (uint64_t)v);
     const float u = (float)((h >> 11) + 1) * 4.8828125e-4f * 2.2737367544323206e-13f;
     const float x = row[v] * it_ + -__logf(-__logf(u));
     if (x > best || (x == best && v < besti)) { best = x; besti = v; }
@@ -199,12 +207,21 @@ void greedy_sample(torch::Tensor out, torch::Tensor logits) {
   if (B == 0) return;
   auto stream = at::cuda::getCurrentHIPStream();
   sample_kernel<false><<<B, kBlock, 0, stream>>>(
-      out.data_ptr<int32_t>(), logits.data_ptr<float>(), nullptr, 0, V);
+      out.data_ptr<int32_t>(), logits.data_ptr<float>(), nullptr, 0, nullptr,
+      V);
   HIP_CHECK_KERNEL();
 }
 
+static const uint64_t* opt_seeds(const c10::optional<torch::Tensor>& t,
+                                 int B) {
+  if (!t.has_value()) return nullptr;
+  TORCH_CHECK(t->dtype() == torch::kInt64 && t->numel() == B &&
+              t->is_cuda() && t->is_contiguous());
+  return reinterpret_cast<const uint64_t*>(t->data_ptr<int64_t>());
+}
+
 void gumbel_sample(torch::Tensor out, torch::Tensor logits, torch::Tensor inv_temp,
-                   int64_t seed) {
+                   int64_t seed, c10::optional<torch::Tensor> row_seeds) {
   TORCH_CHECK(logits.is_cuda() && logits.dtype() == torch::kFloat32);
   TORCH_CHECK(inv_temp.dtype() == torch::kFloat32);
   const int B = logits.size(0);
@@ -213,13 +230,14 @@ void gumbel_sample(torch::Tensor out, torch::Tensor logits, torch::Tensor inv_te
   auto stream = at::cuda::getCurrentHIPStream();
   sample_kernel<true><<<B, kBlock, 0, stream>>>(
       out.data_ptr<int32_t>(), logits.data_ptr<float>(),
-      inv_temp.data_ptr<float>(), (uint64_t)seed, V);
+      inv_temp.data_ptr<float>(), (uint64_t)seed, opt_seeds(row_seeds, B), V);
   HIP_CHECK_KERNEL();
 }
 
 void topkp_sample(torch::Tensor out, torch::Tensor logits,
                   torch::Tensor inv_temp, torch::Tensor top_k,
-                  torch::Tensor top_p, int64_t seed) {
+                  torch::Tensor top_p, int64_t seed,
+                  c10::optional<torch::Tensor> row_seeds) {
   TORCH_CHECK(logits.is_cuda() && logits.dtype() == torch::kFloat32);
   TORCH_CHECK(inv_temp.dtype() == torch::kFloat32);
   TORCH_CHECK(top_k.dtype() == torch::kInt32);
@@ -232,6 +250,6 @@ void topkp_sample(torch::Tensor out, torch::Tensor logits,
   topkp_sample_kernel<<<B, kBlock, 0, stream>>>(
       out.data_ptr<int32_t>(), logits.data_ptr<float>(),
       inv_temp.data_ptr<float>(), top_k.data_ptr<int32_t>(),
-      top_p.data_ptr<float>(), (uint64_t)seed, V);
+      top_p.data_ptr<float>(), (uint64_t)seed, opt_seeds(row_seeds, B), V);
   HIP_CHECK_KERNEL();
 }

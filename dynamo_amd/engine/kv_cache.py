@@ -139,7 +139,20 @@ class PageAllocator:
         return ev
 
     def clear(self):
-        self.__init__(self.num_pages, self.page_size, self.enable_prefix)
+        """Reset all page state (clear_kv_blocks / update_weights).
+
+        Preserves the host_tier LINK (it is wired once at engine init) but
+        clears the tier maps too: post-clear lookups must not onboard pages
+        whose KV was computed before the reset (stale after update_weights).
+        """
+        self.free = deque(range(self.num_pages))
+        self.ref = [0] * self.num_pages
+        self.page_hash = [None] * self.num_pages
+        self.hash_to_page = {}
+        self.evictable = OrderedDict()
+        self.events = []
+        if self.host_tier is not None:
+            self.host_tier.clear()
         self.events.append(KvEvent("cleared"))
 
 

@@ -53,6 +53,32 @@ def compute_logprobs(logits: torch.Tensor, sampled: torch.Tensor,
     return out
 
 
+_GOLDEN = 0x9E3779B97F4A7C15
+
+
+def _row_seeds(reqs: List[Request], step_seed: int,
+               device) -> torch.Tensor | None:
+    """Per-row RNG seeds honoring client-supplied sampling seeds.
+
+    Seeded requests get a stream keyed by (user_seed, output position) —
+    reproducible regardless of batching/scheduling; unseeded rows keep the
+    engine-step stream (seed ^ b<<32, bit-identical to the scalar path).
+    Returns None when no request carries a seed (scalar fast path)."""
+    if not any(r.sampling.seed for r in reqs):
+        return None
+    mask = (1 << 64) - 1
+    rows = []
+    for b, r in enumerate(reqs):
+        if r.sampling.seed:
+            rows.append((r.sampling.seed * _GOLDEN
+                         + (len(r.output_tokens) + 1)) & mask)
+        else:
+            rows.append((step_seed ^ (b << 32)) & mask)
+    t = torch.tensor([x - (1 << 64) if x >= (1 << 63) else x
+                      for x in rows], dtype=torch.int64)
+    return t.to(device) if device.type == "cuda" else t
+
+
 def sample_tokens(logits: torch.Tensor, reqs: List[Request],
                   step_seed: int) -> torch.Tensor:
     """logits [n, V] fp32 -> token ids [n] int32."""
@@ -62,6 +88,7 @@ def sample_tokens(logits: torch.Tensor, reqs: List[Request],
     if all(t == 0.0 for t in temps):
         return ops.greedy_sample(logits)
 
+    row_seeds = _row_seeds(reqs, step_seed, logits.device)
     greedy_mask = torch.tensor([t == 0.0 for t in temps], device=logits.device)
     inv_t = torch.tensor([1.0 / t if t > 0 else 1.0 for t in temps],
                          dtype=torch.float32, device=logits.device)
@@ -73,7 +100,8 @@ def sample_tokens(logits: torch.Tensor, reqs: List[Request],
                           dtype=torch.int32, device=logits.device)
         tp = torch.tensor([r.sampling.top_p for r in reqs],
                           dtype=torch.float32, device=logits.device)
-        sampled = ops.topkp_sample(logits, inv_t, tk, tp, step_seed)
+        sampled = ops.topkp_sample(logits, inv_t, tk, tp, step_seed,
+                                   row_seeds)
     else:
         filt = logits
         if has_filter:
@@ -84,7 +112,7 @@ def sample_tokens(logits: torch.Tensor, reqs: List[Request],
                                               r.sampling.top_k,
                                               r.sampling.top_p))
             filt = torch.cat(rows, 0)
-        sampled = ops.gumbel_sample(filt, inv_t, step_seed)
+        sampled = ops.gumbel_sample(filt, inv_t, step_seed, row_seeds)
     if greedy_mask.any():
         greedy = ops.greedy_sample(logits)
         sampled = torch.where(greedy_mask, greedy, sampled)

@@ -432,27 +432,58 @@ def build_app(manager: ModelManager) -> FastAPI:
         t0 = time.time()
 
         if req.stream:
+            stops = _stop_list(req)
+
             async def sse():
                 produced: List[int] = []
+                acc = ""     # decoded text so far
+                sent = 0     # chars of acc already emitted
+                # stop-string enforcement (same discipline as streaming
+                # /v1/completions): a stop may span chunk boundaries, so
+                # withhold its max length minus one trailing chars
+                hold = max((len(s) for s in stops), default=1) - 1
                 first = True
+
+                def event(text, finish, role=False):
+                    delta = {}
+                    if role:
+                        delta["role"] = "assistant"
+                    delta["content"] = text
+                    data = {"id": rid, "object": "chat.completion.chunk",
+                            "model": entry.name, "choices": [{
+                                "index": 0, "delta": delta,
+                                "finish_reason": finish}]}
+                    return f"data: {json.dumps(data)}\n\n"
+
                 try:
+                    finish = None
                     async for chunk in _run(entry, token_ids, req, rid, _session_of(req, raw)):
                         if await raw.is_disconnected():
                             break
-                        delta = {}
+                        role = first
                         if first:
                             TTFT.labels(entry.name).observe(time.time() - t0)
-                            delta["role"] = "assistant"
                             first = False
                         prev = len(produced)
                         produced.extend(chunk.get("token_ids", []))
-                        delta["content"] = entry.tokenizer.decode_incremental(
-                            produced, prev)
-                        data = {"id": rid, "object": "chat.completion.chunk",
-                                "model": entry.name, "choices": [{
-                                    "index": 0, "delta": delta,
-                                    "finish_reason": chunk.get("finish_reason")}]}
-                        yield f"data: {json.dumps(data)}\n\n"
+                        acc += entry.tokenizer.decode_incremental(produced,
+                                                                  prev)
+                        finish = chunk.get("finish_reason")
+                        if stops:
+                            cut = _find_stop(acc, stops)
+                            if cut >= 0:
+                                yield event(acc[sent:cut], "stop", role)
+                                sent = cut
+                                finish = "stop"
+                                break
+                            safe = max(sent, len(acc) - hold)
+                            if safe > sent or finish or role:
+                                tail = len(acc) if finish else safe
+                                yield event(acc[sent:tail], finish, role)
+                                sent = tail
+                        else:
+                            yield event(acc[sent:], finish, role)
+                            sent = len(acc)
                     yield "data: [DONE]\n\n"
                 finally:
                     LATENCY.labels(entry.name).observe(time.time() - t0)

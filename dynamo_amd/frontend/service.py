@@ -238,5 +238,10 @@ class ModelManager:
             async for chunk in entry.router.client.generate(
                     payload, instance_id=iid):
                 yield chunk
+        except (EndpointError, ConnectionError, OSError):
+            # locally inhibit the failed instance so the migration retry
+            # cannot re-pick it during the discovery watch-lag window
+            entry.router.inhibit(iid)
+            raise
         finally:
             entry.router.end_request(iid, payload["token_ids"])

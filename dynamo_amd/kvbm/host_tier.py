@@ -176,3 +176,22 @@ class HostKVTier:
     def drain_events(self):
         ev, self.events = self.events, []
         return ev
+
+    def clear(self):
+        """Drop every cached page in G2/G3/G4: called from
+        PageAllocator.clear() so a KV reset (clear_kv_blocks /
+        update_weights) cannot onboard pages computed under old weights."""
+        if self.device.type == "cuda":
+            self.stream.synchronize()  # let in-flight offloads land first
+        self.map.clear()
+        self.free = list(range(self.num_host_pages))
+        self.events = []
+        if self.disk is not None:
+            for h in list(getattr(self.disk, "map", {})):
+                self.disk.remove(h)
+        if self.objects is not None:
+            # wait for queued publishes, then drop our published objects
+            self._obj_pool.shutdown(wait=True)
+            from concurrent.futures import ThreadPoolExecutor
+            self._obj_pool = ThreadPoolExecutor(max_workers=1)
+            self.objects.clear()

@@ -53,6 +53,15 @@ class ObjectKVTier:
         self.stats["put"] += 1
         return True
 
+    def clear(self):
+        """Remove every published object (KV reset: content computed under
+        old weights is invalid for every worker sharing this store)."""
+        import shutil
+        for sub in os.listdir(self.root):
+            p = os.path.join(self.root, sub)
+            if os.path.isdir(p) and len(sub) == 2:
+                shutil.rmtree(p, ignore_errors=True)
+
     def get(self, h: int) -> Optional[bytes]:
         try:
             with open(self._path(h), "rb") as fh:

@@ -184,19 +184,25 @@ def _splitmix64(x: torch.Tensor) -> torch.Tensor:
 
 
 def gumbel_sample(logits: torch.Tensor, inv_temp: torch.Tensor,
-                  seed: int) -> torch.Tensor:
+                  seed: int, row_seeds: torch.Tensor = None) -> torch.Tensor:
     if logits.is_cuda:
         out = torch.empty(logits.shape[0], dtype=torch.int32,
                           device=logits.device)
-        hip().gumbel_sample(out, logits.float().contiguous(), inv_temp, seed)
+        hip().gumbel_sample(out, logits.float().contiguous(), inv_temp, seed,
+                            row_seeds)
         return out
     # CPU: DETERMINISTIC counter-based Gumbel noise — the same
-    # splitmix64(seed ^ b<<32 ^ v) stream as the HIP kernel, so sampling
-    # is reproducible across engines/processes (TP lockstep needs this)
+    # splitmix64(row_seed ^ v) stream as the HIP kernel (row_seed defaults
+    # to seed ^ b<<32), so sampling is reproducible across
+    # engines/processes (TP lockstep needs this)
     B, V = logits.shape
-    b = torch.arange(B, dtype=torch.int64).unsqueeze(1) << 32
+    if row_seeds is None:
+        rs = (torch.tensor(seed, dtype=torch.int64)
+              ^ (torch.arange(B, dtype=torch.int64) << 32))
+    else:
+        rs = row_seeds.to(torch.int64).cpu()
     v = torch.arange(V, dtype=torch.int64).unsqueeze(0)
-    h = _splitmix64(torch.tensor(seed, dtype=torch.int64) ^ b ^ v)
+    h = _splitmix64(rs.unsqueeze(1) ^ v)
     u = (((h >> 11) & ((1 << 53) - 1)) + 1).double() * (2.0 ** -53)
     g = -torch.log(-torch.log(u.float()))
     return (logits.float() * inv_temp.unsqueeze(-1) + g).argmax(-1).to(torch.int32)
@@ -204,12 +210,12 @@ def gumbel_sample(logits: torch.Tensor, inv_temp: torch.Tensor,
 
 def topkp_sample(logits: torch.Tensor, inv_temp: torch.Tensor,
                  top_k: torch.Tensor, top_p: torch.Tensor,
-                 seed: int) -> torch.Tensor:
+                 seed: int, row_seeds: torch.Tensor = None) -> torch.Tensor:
     """Fused top-k/top-p Gumbel sampling (GPU only; per-row params)."""
     out = torch.empty(logits.shape[0], dtype=torch.int32,
                       device=logits.device)
     hip().topkp_sample(out, logits.float().contiguous(), inv_temp,
-                       top_k, top_p, seed)
+                       top_k, top_p, seed, row_seeds)
     return out
 
 

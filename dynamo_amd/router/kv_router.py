@@ -81,6 +81,13 @@ class KvRouter:
         self.indexer = _core.KvIndexer()
         self.workers: Dict[str, WorkerState] = {}
         self._sessions: "OrderedDict[str, str]" = OrderedDict()
+        # local worker inhibition (reference: distributed-runtime.md "Local
+        # Worker Inhibition", DYN_RUNTIME_INHIBITED_DURATION_SECS): after a
+        # failed send, exclude the instance locally until the inhibition
+        # expires — discovery-lease expiry alone leaves a window where
+        # migration retries re-pick the same dead worker.
+        self._inhibited: Dict[str, float] = {}   # iid -> expiry monotonic
+        self.inhibit_duration = 30.0
         self._rr = 0
         self._tasks: List[asyncio.Task] = []
         self._started = False
@@ -126,7 +133,12 @@ class KvRouter:
 
     @staticmethod
     def _wid(instance_id: str) -> int:
-        return int(instance_id[:12], 16)
+        # hash the FULL instance id: a prefix slice would alias two
+        # instances sharing the first hex chars into one indexer worker
+        import hashlib
+        return int.from_bytes(
+            hashlib.blake2b(instance_id.encode(), digest_size=8).digest(),
+            "big")
 
     async def _consume_events(self, address: str, iid: str):
         wid = self._wid(iid)
@@ -173,6 +185,15 @@ class KvRouter:
         insts = self.client.instances()
         if not insts:
             return None
+        # drop locally-inhibited instances (dead-worker watch-lag window);
+        # if that would leave nothing, fall back to the full list
+        if self._inhibited:
+            now = time.monotonic()
+            self._inhibited = {i: t for i, t in self._inhibited.items()
+                               if t > now}
+            ok = [i for i in insts if i.instance_id not in self._inhibited]
+            if ok:
+                insts = ok
         if self.cfg.busy_threshold > 0:
             if all(self._load(i.instance_id) > self.cfg.busy_threshold
                    for i in insts):
@@ -250,6 +271,11 @@ class KvRouter:
         if not ws:
             return 0.0
         return ws.active_requests + ws.kv_usage
+
+    def inhibit(self, iid: str, duration: Optional[float] = None):
+        """Locally exclude `iid` from selection after a failed send."""
+        self._inhibited[iid] = time.monotonic() + (
+            duration if duration is not None else self.inhibit_duration)
 
     # -- request accounting ---------------------------------------------
     def begin_request(self, iid: str, token_ids: List[int]):

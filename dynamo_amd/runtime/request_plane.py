@@ -29,13 +29,28 @@ class RequestContext:
     def __init__(self, rid: int):
         self.rid = rid
         self._cancelled = asyncio.Event()
+        self._callbacks = []
 
     @property
     def cancelled(self) -> bool:
         return self._cancelled.is_set()
 
+    def on_cancel(self, cb):
+        """Register a callback fired when a cancel frame arrives (lets
+        handlers wake event-driven waits instead of polling `cancelled`)."""
+        if self._cancelled.is_set():
+            cb()
+        else:
+            self._callbacks.append(cb)
+
     def cancel(self):
         self._cancelled.set()
+        cbs, self._callbacks = self._callbacks, []
+        for cb in cbs:
+            try:
+                cb()
+            except Exception:
+                pass
 
 
 class RequestPlaneServer:
@@ -198,9 +213,19 @@ class _Conn:
                                "error": "connection lost"}, None))
 
     async def send(self, header, body):
-        async with self.wlock:
+        # Explicit acquire/release: if this coroutine is GC'd mid-drain at
+        # interpreter teardown (GeneratorExit), a plain `async with` would
+        # release the lock via call_soon on a CLOSED loop and raise an
+        # unraisable RuntimeError; guard the release instead.
+        await self.wlock.acquire()
+        try:
             self.writer.write(encode_frame(header, body))
             await self.writer.drain()
+        finally:
+            try:
+                self.wlock.release()
+            except RuntimeError:
+                pass  # event loop already closed
 
 
 class RequestPlaneClient:

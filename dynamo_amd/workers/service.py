@@ -141,20 +141,51 @@ class WorkerService:
 
     # ------------------------------------------------------------------
     async def _engine_loop(self):
+        from dynamo_amd.engine.engine import StepOutput
         while True:
             await self._paused.wait()  # snapshot lifecycle pause
             if not self.engine.has_work():
                 self._work.clear()
                 await self._work.wait()
                 await self._paused.wait()
-            async with self._engine_lock:
-                outputs = await asyncio.to_thread(self.engine.step)
+            try:
+                async with self._engine_lock:
+                    outputs = await asyncio.to_thread(self.engine.step)
+            except Exception:
+                # a step exception must not kill the loop: abort the
+                # requests that were in flight (clients get a clean
+                # error-finish) and keep serving
+                log.exception("engine step failed; aborting in-flight work")
+                async with self._engine_lock:
+                    self._abort_inflight("error")
+                continue
             for so in outputs:
                 q = self.queues.get(so.req_id)
                 if q is not None:
                     q.put_nowait(so)
             self._fan_kv_events()
             await asyncio.sleep(0)
+
+    def _abort_inflight(self, reason: str):
+        """Abort every scheduled request, pushing a finished StepOutput so
+        waiting generate() handlers terminate their streams."""
+        from dynamo_amd.engine.engine import StepOutput
+        for rid in list(self.engine.requests):
+            try:
+                self.engine.abort(rid)
+            except Exception:
+                log.exception("abort %s failed", rid)
+            q = self.queues.get(rid)
+            if q is not None:
+                q.put_nowait(StepOutput(rid, None, True, reason))
+
+    async def _drain_idle(self, timeout: float) -> bool:
+        """Wait until the engine has no scheduled work. True if idle."""
+        deadline = asyncio.get_event_loop().time() + timeout
+        while (self.engine.has_work()
+               and asyncio.get_event_loop().time() < deadline):
+            await asyncio.sleep(0.01)
+        return not self.engine.has_work()
 
     def _fan_kv_events(self):
         """Drain engine KV events to subscribers — called from the engine
@@ -224,19 +255,17 @@ class WorkerService:
                 if pr is not None:
                     await self._attach_remote_kv(req, pr)
             self._work.set()
+            # event-driven: a cancel frame wakes the queue wait via a
+            # sentinel — no polling timeouts (reference behavior:
+            # push_endpoint responds to cancellation immediately)
+            ctx.on_cancel(lambda: q.put_nowait(None))
 
             while True:
-                if ctx.cancelled:   # checked EVERY iteration — a steadily
-                    # producing stream must still notice a cancel frame
+                so = await q.get()
+                if so is None or ctx.cancelled:
                     async with self._engine_lock:
                         self.engine.abort(req_id)
                     return
-                get = asyncio.create_task(q.get())
-                done, _ = await asyncio.wait({get}, timeout=0.05)
-                if not done:
-                    get.cancel()
-                    continue
-                so = get.result()
                 chunk: dict = {"token_ids": ([so.new_token]
                                              if so.new_token is not None
                                              else [])}
@@ -244,6 +273,23 @@ class WorkerService:
                     chunk["embedding"] = so.embedding
                 if so.logprobs is not None:
                     chunk["logprobs"] = [so.logprobs]
+                # coalesce: if the consumer fell behind the engine, fold
+                # every already-queued plain token into ONE frame instead
+                # of sending one frame per token
+                while (not so.finished and so.logprobs is None
+                       and so.embedding is None and not q.empty()):
+                    nxt = q.get_nowait()
+                    if nxt is None:
+                        q.put_nowait(None)   # re-deliver cancel sentinel
+                        break
+                    if nxt.embedding is not None or nxt.logprobs is not None:
+                        chunk["embedding"] = nxt.embedding
+                        if nxt.logprobs is not None:
+                            chunk.setdefault("logprobs", []).append(
+                                nxt.logprobs)
+                    if nxt.new_token is not None:
+                        chunk["token_ids"].append(nxt.new_token)
+                    so = nxt
                 if so.finished:
                     chunk["finish_reason"] = so.finish_reason
                     if is_prefill_role:
@@ -306,6 +352,17 @@ class WorkerService:
 
     # ------------------------------------------------------------------
     async def clear_kv_blocks(self, payload, ctx):
+        # the allocator reset is only safe with no in-flight sequences
+        # (their SequenceKV.release would decref fresh refcounts); drain
+        # first and REFUSE if live traffic keeps the engine busy
+        if not await self._drain_idle(float(payload.get("drain_timeout", 5.0))
+                                      if isinstance(payload, dict) else 5.0):
+            if not (isinstance(payload, dict) and payload.get("force")):
+                yield {"status": "busy",
+                       "num_running": self.engine.scheduler.num_running()}
+                return
+            async with self._engine_lock:
+                self._abort_inflight("abort")
         async with self._engine_lock:
             self.engine.clear_kv()
         self._fan_kv_events()
@@ -340,14 +397,20 @@ class WorkerService:
             self.kv_event_subs.append(q)
         if snapshot:
             yield [{"kind": "stored", "hashes": snapshot, "parent": None}]
+        ctx.on_cancel(lambda: q.put_nowait(None))
         try:
             while not ctx.cancelled:
-                get = asyncio.create_task(q.get())
-                done, _ = await asyncio.wait({get}, timeout=0.1)
-                if not done:
-                    get.cancel()
-                    continue
-                yield get.result()
+                batch = await q.get()
+                if batch is None:
+                    break
+                # coalesce any backlog into one frame
+                while not q.empty():
+                    more = q.get_nowait()
+                    if more is None:
+                        q.put_nowait(None)
+                        break
+                    batch.extend(more)
+                yield batch
         finally:
             self.kv_event_subs.remove(q)
 
@@ -382,6 +445,13 @@ class WorkerService:
     async def update_weights(self, payload, ctx):
         seed = int(payload.get("seed", 0))
         scale = float(payload.get("scale", 0.01))
+        # a weight push invalidates in-flight work: drain briefly, then
+        # abort whatever is still running (old-weight KV must not mix with
+        # new-weight decode steps)
+        if not await self._drain_idle(float(payload.get("drain_timeout",
+                                                        5.0))):
+            async with self._engine_lock:
+                self._abort_inflight("abort")
         async with self._engine_lock:
             n = await asyncio.to_thread(self._apply_weight_delta, seed, scale)
             self.engine.clear_kv()
