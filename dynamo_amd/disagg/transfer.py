@@ -35,6 +35,39 @@ def pool_transfer_metadata(instance_id: str, kv_pool) -> dict:
     }
     if kv_pool.device.type == "cuda":
         meta["ipc_handle"] = binascii.hexlify(kv_pool.ipc_export()).decode()
+    elif getattr(kv_pool, "shm_path", None):
+        meta["shm_path"] = kv_pool.shm_path
+    return meta
+
+
+def tp_transfer_metadata(instance_id: str, kv_pool, tp) -> dict:
+    """Gather per-rank pool metadata across a TP group (rank0 publishes).
+
+    The deterministic lockstep engines use identical page ids on every
+    rank, so a decode TP group pulls pages rank-to-rank: decode rank r
+    opens prefill rank r's pool (equal TP degrees; the reference reshardes
+    via permute_scatter when degrees differ — out of scope for the target
+    configs)."""
+    import torch.distributed as dist
+    local = pool_transfer_metadata(f"{instance_id}#r{tp.rank}", kv_pool)
+    box = [None] * tp.size
+    dist.all_gather_object(box, local, group=tp.control_group)
+    return {"instance_id": instance_id, "tp_size": tp.size, "ranks": box}
+
+
+def rank_meta(meta: dict, tp_rank: int) -> dict:
+    """Select this rank's slice of (possibly TP-grouped) pool metadata."""
+    if "ranks" in meta:
+        ranks = meta["ranks"]
+        if tp_rank >= len(ranks):
+            raise RuntimeError(
+                f"KV transfer between unequal TP degrees unsupported: "
+                f"need rank {tp_rank}, peer has {len(ranks)}")
+        return ranks[tp_rank]
+    if tp_rank != 0:
+        raise RuntimeError(
+            "KV transfer between unequal TP degrees unsupported: "
+            f"peer is TP1, local rank {tp_rank}")
     return meta
 
 
@@ -55,13 +88,25 @@ class KvPuller:
         pool = _LOCAL_POOLS.get(iid)
         if pool is not None:  # same process
             buf = pool.buffer
-        else:
+        elif "ipc_handle" in meta:
             from dynamo_amd import ops
             handle = binascii.unhexlify(meta["ipc_handle"])
             raw = ops.hip().ipc_open(handle, meta["pool_nbytes"],
                                      self.local.device.index or 0)
             dtype = getattr(torch, meta["dtype"])
             buf = raw.view(dtype).view(meta["pool_shape"])
+        elif "shm_path" in meta:  # CPU cross-process (shared file mapping)
+            dtype = getattr(torch, meta["dtype"])
+            shape = meta["pool_shape"]
+            numel = 1
+            for s in shape:
+                numel *= s
+            buf = torch.from_file(meta["shm_path"], shared=True,
+                                  size=numel, dtype=dtype).view(shape)
+        else:
+            raise RuntimeError(
+                f"no transfer path to pool {iid}: not local, no ipc_handle, "
+                "no shm_path")
         self._mapped[iid] = buf
         return buf
 

@@ -136,6 +136,10 @@ class EngineConfig:
     block_salt: int = 0
     # disaggregation
     worker_type: str = "aggregated"     # aggregated | prefill | decode
+    # CPU-only: back the KV pool with a /dev/shm file so a peer PROCESS can
+    # map it by path (the CPU stand-in for the GPU pool's hipIpc handle;
+    # lets cross-process disagg pulls run in CPU/gloo tests)
+    cpu_shm_pool: bool = False
     # tensor parallelism (process group wired by the worker)
     tp_size: int = 1
     tp_rank: int = 0

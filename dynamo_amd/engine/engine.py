@@ -85,6 +85,8 @@ class LLMEngine:
                                             use_graphs=use_graphs)
         self._last_sampled = None
         self._lora = None
+        self._puller = None   # disagg decode-side KvPuller (lazy)
+        self._last_weight_delta_count = 0
 
     # -- LoRA (engine-level activation; see dynamo_amd/lora) -----------
     @property
@@ -318,6 +320,43 @@ class LLMEngine:
             req.kv.commit_full_pages(req.all_tokens, req.num_computed)
             req.kv.release()
             req.kv = None
+
+    def attach_request(self, spec: dict):
+        """Disagg decode-side handoff: add a request whose prefill ran on a
+        peer worker and pull its KV pages (see parallel/tp.attach_remote;
+        TP groups broadcast the same spec to every rank)."""
+        from dynamo_amd.parallel.tp import attach_remote
+        return attach_remote(self, spec, tp_rank=self.cfg.tp_rank)
+
+    def apply_weight_delta(self, seed: int, scale: float) -> int:
+        """RL weight-update surface (reference: lib/rl/src/lib.rs:4-16):
+        apply a deterministic seeded in-place delta to every weight tensor
+        (modeling a policy push without checkpoint files), then invalidate
+        captured graphs and flush the KV/prefix cache — cached KV computed
+        under the old weights is invalid."""
+        model = getattr(self.runner, "model", None)
+        count = 0
+        if model is not None:
+            seen = set()
+            with torch.no_grad():
+                for mod in model.modules():
+                    for name, val in sorted(vars(mod).items()):
+                        if (isinstance(val, torch.Tensor)
+                                and val.is_floating_point() and val.numel()
+                                and name != "cos_sin"  # rope table
+                                and id(val) not in seen):  # tied tensors once
+                            seen.add(id(val))
+                            g = torch.Generator(device="cpu")
+                            g.manual_seed(seed * 1000003 + count)
+                            noise = torch.randn(val.shape, generator=g,
+                                                dtype=torch.float32)
+                            val.add_(noise.to(val.device, val.dtype),
+                                     alpha=scale)
+                            count += 1
+        self.clear_kv()
+        self._invalidate_graphs()
+        self._last_weight_delta_count = count
+        return count
 
     def drain_kv_events(self) -> List[KvEvent]:
         if self.host_tier is not None:

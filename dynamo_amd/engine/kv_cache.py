@@ -232,12 +232,13 @@ class KVCachePool:
 
     def __init__(self, num_layers: int, num_pages: int, num_kv_heads: int,
                  page_size: int, head_dim: int, device: str,
-                 dtype=torch.bfloat16):
+                 dtype=torch.bfloat16, shm_export: bool = False):
         self.shape = (num_layers, 2, num_pages, num_kv_heads, page_size, head_dim)
         self.num_pages = num_pages
         self.page_size = page_size
         self.device = torch.device(device)
         self.dtype = dtype
+        self.shm_path = None
         numel = 1
         for s in self.shape:
             numel *= s
@@ -246,9 +247,31 @@ class KVCachePool:
             from dynamo_amd import ops
             self._raw = ops.hip().ipc_alloc(nbytes, self.device.index or 0)
             self.buffer = self._raw.view(dtype).view(self.shape)
+        elif shm_export:
+            # CPU stand-in for hipIpc: a file-backed shared mapping a peer
+            # process can open by path (cross-process disagg pulls in
+            # CPU/gloo tests mirror the GPU hipIpc path)
+            import os
+            import tempfile
+            d = "/dev/shm" if os.path.isdir("/dev/shm") else None
+            fd, self.shm_path = tempfile.mkstemp(prefix="dynamo_kvpool_",
+                                                 dir=d)
+            os.close(fd)
+            self.buffer = torch.from_file(self.shm_path, shared=True,
+                                          size=numel, dtype=dtype).view(
+                                              self.shape)
+            self.buffer.zero_()
         else:
             self.buffer = torch.zeros(self.shape, dtype=dtype, device=device)
         self.nbytes = nbytes
+
+    def __del__(self):  # pragma: no cover
+        if getattr(self, "shm_path", None):
+            import os
+            try:
+                os.unlink(self.shm_path)
+            except OSError:
+                pass
 
     def kcache(self, layer: int) -> torch.Tensor:
         return self.buffer[layer, 0]

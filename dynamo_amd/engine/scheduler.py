@@ -175,6 +175,8 @@ class Scheduler:
         for req in list(self.running):
             if budget <= 0:
                 break
+            if req.state is not ReqState.RUNNING:
+                continue  # preempted by an earlier iteration of this loop
             remaining = req.total_len - req.num_computed
             if remaining <= 0:
                 continue

@@ -23,6 +23,9 @@ ENV_VARS = {
     "DYN_HOST_CACHE_PAGES": "KVBM G2 pinned-host tier size (pages)",
     "DYN_TUNABLEOP": "0 disables the pre-tuned hipBLASLt table",
     "DYN_WORKER_TYPE": "aggregated | prefill | decode",
+    "DYN_BYPASS_TOKEN_THRESHOLD":
+        "conditional disagg: net-new prefill tokens below this bypass the "
+        "prefill pool (reference conditional_disagg.rs:15; default 2048)",
 }
 
 

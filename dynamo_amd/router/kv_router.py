@@ -134,11 +134,13 @@ class KvRouter:
     @staticmethod
     def _wid(instance_id: str) -> int:
         # hash the FULL instance id: a prefix slice would alias two
-        # instances sharing the first hex chars into one indexer worker
+        # instances sharing the first hex chars into one indexer worker.
+        # Signed 64-bit (the C++ indexer takes int64).
         import hashlib
-        return int.from_bytes(
+        v = int.from_bytes(
             hashlib.blake2b(instance_id.encode(), digest_size=8).digest(),
             "big")
+        return v - (1 << 64) if v >= (1 << 63) else v
 
     async def _consume_events(self, address: str, iid: str):
         wid = self._wid(iid)
@@ -152,8 +154,10 @@ class KvRouter:
                         self.indexer.apply_removed(wid, ev["hashes"])
                     elif ev["kind"] == "cleared":
                         self.indexer.clear_worker(wid)
-        except Exception:
+        except (ConnectionError, OSError, asyncio.CancelledError):
             pass  # worker died; watch loop will clean up
+        except Exception:
+            log.exception("kv_events consumer for %s failed", iid)
 
     async def _poll_metrics(self, address: str, iid: str):
         while iid in self.workers:

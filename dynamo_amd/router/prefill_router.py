@@ -27,7 +27,7 @@ class PrefillRouter:
                  prefill_component: str = "prefill",
                  decode_component: str = "backend",
                  cfg: RouterConfig | None = None,
-                 bypass_token_threshold: int = 2048,
+                 bypass_token_threshold: int | None = None,
                  bypass_overlap_ratio: float = 0.7,
                  bypass_decode_busy_waiting: int = 4):
         self.runtime = runtime
@@ -37,6 +37,10 @@ class PrefillRouter:
                                        self.cfg)
         self.decode_router = KvRouter(runtime, namespace, decode_component,
                                       self.cfg)
+        if bypass_token_threshold is None:
+            from dynamo_amd import envs
+            bypass_token_threshold = envs.get("DYN_BYPASS_TOKEN_THRESHOLD",
+                                              2048, int)
         self.bypass_token_threshold = bypass_token_threshold
         self.bypass_overlap_ratio = bypass_overlap_ratio
         # busy gating (conditional_disagg.rs:111 parity): never bypass onto

@@ -41,6 +41,9 @@ def build_parser():
     p.add_argument("--no-prefix-caching", action="store_true")
     p.add_argument("--no-hip-graphs", action="store_true")
     p.add_argument("--seed", type=int, default=0)
+    p.add_argument("--dtype", default=None,
+                   help="bfloat16 | float32 (default: bf16 on GPU, "
+                        "fp32 on CPU)")
     p.add_argument("--host", default="127.0.0.1")
     p.add_argument("--host-cache-pages", type=int, default=0,
                    help="KVBM G2 pinned-host tier size in pages")
@@ -56,10 +59,17 @@ def build_parser():
     p.add_argument("--status-port", type=int, default=0,
                    help="serve a JSON system-status HTTP endpoint on this "
                         "port (reference DYN_SYSTEM_PORT parity; 0 = off)")
+    p.add_argument("--tp-size", type=int, default=1,
+                   help="tensor parallelism: launch via torchrun "
+                        "--nproc-per-node N, one rank per GPU over RCCL; "
+                        "rank 0 serves the request plane, followers run "
+                        "the lockstep loop (reference parity: engine "
+                        "--tensor-parallel-size, recipes/llama-3-70b/"
+                        "vllm/disagg-single-node/deploy.yaml:57,100)")
     return p
 
 
-def make_engine_from_args(args) -> LLMEngine:
+def make_engine_from_args(args, tp=None) -> LLMEngine:
     mc = resolve_model_config(args.model)
     if args.mock:
         from dynamo_amd.mocker import make_mock_engine
@@ -70,7 +80,12 @@ def make_engine_from_args(args) -> LLMEngine:
             num_pages=args.kv_pool_pages or 1024,
             worker_type=args.worker_type)
     import torch
-    device = args.device or ("cuda:0" if torch.cuda.is_available() else "cpu")
+    if tp is not None and torch.cuda.is_available():
+        device = f"cuda:{int(os.environ.get('LOCAL_RANK', tp.rank))}"
+        torch.cuda.set_device(device)
+    else:
+        device = args.device or ("cuda:0" if torch.cuda.is_available()
+                                 else "cpu")
     if device.startswith("cuda"):
         from dynamo_amd.utils import enable_tunableop
         enable_tunableop(tuning=False)
@@ -87,7 +102,10 @@ def make_engine_from_args(args) -> LLMEngine:
         logging.info("imported %d weight bytes zero-copy from GMS %s",
                      weight_pool.buffer.numel(), metas[0].instance_id)
     cfg = EngineConfig(
-        model=mc, device=device, page_size=args.page_size,
+        model=mc, device=device,
+        dtype=(args.dtype or
+               ("bfloat16" if device.startswith("cuda") else "float32")),
+        page_size=args.page_size,
         max_num_seqs=args.max_num_seqs,
         max_batched_tokens=args.max_batched_tokens,
         max_model_len=args.max_model_len,
@@ -96,6 +114,12 @@ def make_engine_from_args(args) -> LLMEngine:
         enable_prefix_caching=not args.no_prefix_caching,
         enable_hip_graphs=not args.no_hip_graphs,
         worker_type=args.worker_type,
+        tp_size=tp.size if tp is not None else 1,
+        tp_rank=tp.rank if tp is not None else 0,
+        # CPU workers in a P/D split export the pool as a shared mapping
+        # so decode processes can pull (GPU uses hipIpc)
+        cpu_shm_pool=(device == "cpu"
+                      and args.worker_type in ("prefill", "decode")),
         host_cache_pages=args.host_cache_pages,
         disk_cache_pages=args.disk_cache_pages,
         object_cache_dir=args.object_cache_dir,
@@ -103,18 +127,19 @@ def make_engine_from_args(args) -> LLMEngine:
                          (os.path.join(tempfile.gettempdir(),
                                        f"dynamo_kv_g3_{os.getpid()}.bin")
                           if args.disk_cache_pages else "")))
-    return LLMEngine(cfg, seed=args.seed, weight_pool=weight_pool)
+    return LLMEngine(cfg, tp=tp, seed=args.seed, weight_pool=weight_pool)
 
 
-async def async_main(args):
+async def async_main(args, engine=None, kv_meta=None):
     logging.basicConfig(level=logging.INFO,
                         format="%(asctime)s %(name)s %(levelname)s %(message)s")
-    engine = make_engine_from_args(args)
+    if engine is None:
+        engine = make_engine_from_args(args)
     rt = DistributedRuntime(args.discovery, host=args.host)
     component = args.component or (
         "prefill" if args.worker_type == "prefill" else "backend")
     ws = WorkerService(engine, rt, namespace=args.namespace,
-                       component=component)
+                       component=component, kv_transfer_meta=kv_meta)
     await ws.start()
     status_runner = None
     if args.status_port:
@@ -138,9 +163,45 @@ async def async_main(args):
     await rt.shutdown()
 
 
+def run_tp(args):
+    """TP>1 worker: torchrun launches one process per GPU; rank 0 serves
+    the request plane behind a TPEngineGroup facade, ranks 1..N-1 apply
+    the lockstep command broadcast (parallel/tp.follower_loop)."""
+    import uuid
+
+    import torch.distributed as dist
+
+    from dynamo_amd.disagg.transfer import tp_transfer_metadata
+    from dynamo_amd.parallel.tp import (TPEngineGroup, follower_loop,
+                                        init_tp)
+    tp = init_tp()
+    if tp.size != args.tp_size:
+        raise RuntimeError(f"torchrun world size {tp.size} != "
+                           f"--tp-size {args.tp_size}")
+    engine = make_engine_from_args(args, tp=tp)
+    kv_meta = None
+    if engine.runner.kv_pool is not None:
+        box = [uuid.uuid4().hex if tp.rank == 0 else None]
+        dist.broadcast_object_list(box, src=0, group=tp.control_group)
+        kv_meta = tp_transfer_metadata(box[0], engine.runner.kv_pool, tp)
+    if tp.rank == 0:
+        group = TPEngineGroup(engine, tp)
+        try:
+            asyncio.run(async_main(args, engine=group, kv_meta=kv_meta))
+        finally:
+            group.shutdown()
+    else:
+        logging.basicConfig(level=logging.INFO)
+        follower_loop(engine, tp)
+    dist.destroy_process_group()
+
+
 def main():
     args = build_parser().parse_args()
-    asyncio.run(async_main(args))
+    if args.tp_size > 1:
+        run_tp(args)
+    else:
+        asyncio.run(async_main(args))
 
 
 if __name__ == "__main__":

@@ -52,7 +52,11 @@ def make_sampling(payload: dict) -> SamplingParams:
 class WorkerService:
     def __init__(self, engine: LLMEngine, runtime: DistributedRuntime,
                  namespace: str = "dynamo", component: str = "backend",
-                 model_name: Optional[str] = None):
+                 model_name: Optional[str] = None,
+                 kv_transfer_meta: Optional[dict] = None):
+        # kv_transfer_meta: pre-gathered TP-group pool metadata (built by
+        # the worker CLI with a collective BEFORE the event loop starts)
+        self._kv_transfer_meta = kv_transfer_meta
         self.engine = engine
         self.runtime = runtime
         self.namespace = namespace
@@ -113,7 +117,13 @@ class WorkerService:
         self.comp.serve_endpoint("list_loras", self.list_loras)
         self.comp.serve_endpoint("update_weights", self.update_weights)
         metadata = {"worker_type": self.worker_type}
-        if self.engine.runner.kv_pool is not None:
+        if self._kv_transfer_meta is not None:
+            # TP group: every rank's pool handle, gathered by the CLI
+            # (decode rank r pulls from prefill rank r; page ids agree via
+            # lockstep determinism)
+            metadata["kv_transfer"] = self._kv_transfer_meta
+            metadata["tp_size"] = self._kv_transfer_meta.get("tp_size", 1)
+        elif self.engine.runner.kv_pool is not None:
             metadata["kv_transfer"] = pool_transfer_metadata(
                 self.comp.instance_id, self.engine.runner.kv_pool)
         await self.comp.register(model_card=self.model_card(),
@@ -246,14 +256,18 @@ class WorkerService:
             prompt_embeds = torch.from_numpy(
                 arr.reshape(pe["shape"]).copy()).float()
         try:
+            pr = payload.get("prefill_result")
             async with self._engine_lock:
-                req = self.engine.add_request(req_id, tokens, sp,
-                                              prompt_embeds=prompt_embeds)
-                if is_prefill_role:
-                    req.hold_kv = True
-                pr = payload.get("prefill_result")
                 if pr is not None:
-                    await self._attach_remote_kv(req, pr)
+                    # disagg decode side: attach + pull KV (TP engines
+                    # broadcast the same spec to every rank)
+                    req = await self._attach_remote_kv(req_id, tokens, sp,
+                                                       pr)
+                else:
+                    req = self.engine.add_request(
+                        req_id, tokens, sp, prompt_embeds=prompt_embeds)
+                    if is_prefill_role:
+                        req.hold_kv = True
             self._work.set()
             # event-driven: a cancel frame wakes the queue wait via a
             # sentinel — no polling timeouts (reference behavior:
@@ -310,13 +324,17 @@ class WorkerService:
     def _disagg_params(self, req: Request) -> dict:
         return {
             "prefill_instance_id": self.instance_id,
+            "prefill_component": self.component_name,
             "page_ids": list(req.kv.pages),
             "num_tokens": req.num_computed,
             "first_token": req.output_tokens[0] if req.output_tokens else None,
         }
 
-    async def _attach_remote_kv(self, req: Request, pr: dict):
-        """Decode side of the disagg handoff: pull KV pages over xGMI."""
+    async def _attach_remote_kv(self, req_id: str, tokens, sp, pr: dict
+                                ) -> Request:
+        """Decode side of the disagg handoff: add the request and pull its
+        KV pages over xGMI (every TP rank pulls its own shard from the
+        matching prefill rank — parallel/tp.attach_remote)."""
         insts = self.runtime.discovery.list(self.namespace)
         src_meta = None
         src_addr = None
@@ -328,27 +346,26 @@ class WorkerService:
         if src_meta is None:
             raise RuntimeError(
                 f"prefill instance {pr['prefill_instance_id']} not found")
-        if self._puller is None:
-            self._puller = KvPuller(self.engine.runner.kv_pool)
-        num_tokens = int(pr["num_tokens"])
-        kv = SequenceKV(self.engine.alloc, self.engine.cfg.block_salt)
-        kv.ensure_capacity(num_tokens)
-        src_pages = [int(p) for p in pr["page_ids"]]
-        npages = (num_tokens + self.engine.cfg.page_size - 1) // self.engine.cfg.page_size
-        await asyncio.to_thread(self._puller.pull, src_meta,
-                                src_pages[:npages], kv.pages[:npages])
-        req.kv = kv
-        req.num_computed = num_tokens
-        if pr.get("first_token") is not None:
-            req.output_tokens.append(int(pr["first_token"]))
+        spec = {
+            "request_id": req_id,
+            "token_ids": list(tokens),
+            "sampling": sp.__dict__.copy(),
+            "num_tokens": int(pr["num_tokens"]),
+            "page_ids": [int(p) for p in pr["page_ids"]],
+            "first_token": pr.get("first_token"),
+            "src_meta": src_meta,
+            "arrival": time.monotonic(),
+        }
+        req = await asyncio.to_thread(self.engine.attach_request, spec)
         # release the prefill side's hold
-        comp = pr["prefill_instance_id"]
         try:
             await self.runtime.client.call(
-                src_addr, f"{self.component_name}.release_kv",
-                {"request_id": req.req_id})
+                src_addr, f"{pr.get('prefill_component', 'prefill')}."
+                "release_kv", {"request_id": req_id})
         except Exception:
-            log.warning("release_kv to %s failed", comp)
+            log.warning("release_kv to %s failed",
+                        pr["prefill_instance_id"])
+        return req
 
     # ------------------------------------------------------------------
     async def clear_kv_blocks(self, payload, ctx):
@@ -453,33 +470,9 @@ class WorkerService:
             async with self._engine_lock:
                 self._abort_inflight("abort")
         async with self._engine_lock:
-            n = await asyncio.to_thread(self._apply_weight_delta, seed, scale)
-            self.engine.clear_kv()
-            self.engine._invalidate_graphs()
+            n = await asyncio.to_thread(self.engine.apply_weight_delta,
+                                        seed, scale)
         yield {"status": "ok", "tensors_updated": n, "seed": seed}
-
-    def _apply_weight_delta(self, seed: int, scale: float) -> int:
-        import torch
-        model = getattr(self.engine.runner, "model", None)
-        if model is None:
-            return 0  # mock engine: nothing to update
-        count = 0
-        seen = set()
-        with torch.no_grad():
-            for mod in model.modules():
-                for name, val in sorted(vars(mod).items()):
-                    if (isinstance(val, torch.Tensor)
-                            and val.is_floating_point() and val.numel()
-                            and name != "cos_sin"       # rope table, not a weight
-                            and id(val) not in seen):   # tied tensors once
-                        seen.add(id(val))
-                        g = torch.Generator(device="cpu")
-                        g.manual_seed(seed * 1000003 + count)
-                        noise = torch.randn(val.shape, generator=g,
-                                            dtype=torch.float32)
-                        val.add_(noise.to(val.device, val.dtype), alpha=scale)
-                        count += 1
-        return count
 
     # -- LoRA endpoints (reference: vllm/worker_factory.py:1378-1413) ----
     async def load_lora(self, payload, ctx):

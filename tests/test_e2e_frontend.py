@@ -418,7 +418,7 @@ def test_rl_update_weights():
         # determinism: a twin engine given the same update matches
         eng2 = LLMEngine(cfg, seed=7)
         ws2 = WorkerService(eng2, rt, component="backend2")
-        n = ws2._apply_weight_delta(42, 0.5)
+        n = eng2.apply_weight_delta(42, 0.5)
         assert n == resp["tensors_updated"]
         import torch
         assert torch.equal(eng.runner.model.embed, eng2.runner.model.embed)

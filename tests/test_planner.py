@@ -96,8 +96,12 @@ def test_planner_service_with_mock_workers():
 
         shared = MemoryDiscovery()
         rt = DistributedRuntime(shared)
+        # decode_step_ms keeps the pool saturated long enough for the
+        # planner's 0.1 s polling to observe it (a healthy engine on a CPU
+        # mock otherwise finishes the burst in milliseconds)
         eng = make_mock_engine(model=ModelConfig(name="m", vocab_size=512),
-                               num_pages=32, page_size=16)
+                               num_pages=32, page_size=16,
+                               decode_step_ms=50.0)
         ws = WorkerService(eng, rt)
         await ws.start()
         # saturate the tiny KV pool
