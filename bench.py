@@ -368,6 +368,14 @@ def run_disagg(args, world, rank, local_rank, device):
         dev_sync(device)
         dist.barrier()
         dist.barrier()
+        if is_leader:
+            # rank 0 prints the result line (driver contract); the decode
+            # leader measured it and ships it over the leaders ctl group
+            box = [None]
+            dist.recv_object_list(box, src=PN, group=leaders_ctl)
+            r = box[0]
+            emit(args, world, r["value"], r["ms_per_step"], r["ttft_p50"],
+                 r["prefill_time"], r["mode"], r["conc"])
         return
 
     # ---- decode side ----
@@ -399,9 +407,13 @@ def run_disagg(args, world, rank, local_rank, device):
         dist.barrier()
         elapsed = time.monotonic() - t0
         side_cmd({"shutdown": True})
-        emit(args, world, gen / elapsed, elapsed / args.steps * 1000,
-             statistics.median(ttfts) if ttfts else None, prefill_time,
-             f"disagg_p{PN}tp{PN}_d{world - PN}tp{world - PN}", conc_total)
+        dist.send_object_list([{
+            "value": gen / elapsed,
+            "ms_per_step": elapsed / args.steps * 1000,
+            "ttft_p50": statistics.median(ttfts) if ttfts else None,
+            "prefill_time": prefill_time,
+            "mode": f"disagg_p{PN}tp{PN}_d{world - PN}tp{world - PN}",
+            "conc": conc_total}], dst=0, group=leaders_ctl)
     else:
         while True:
             box = [None]

@@ -46,3 +46,28 @@ def test_bench_disagg_world2():
     assert r["config"]["parallelism"] == "disagg_p1tp1_d1tp1"
     assert r["value"] > 0
     assert r["ms_per_step"] > 0
+
+
+@pytest.mark.timeout(300)
+def test_bench_disagg_world4_tp2():
+    """The config-#4 shape at CPU scale: TP2 prefill group + TP2 decode
+    group with cross-group page transfer (the driver's SCALE run shape)."""
+    r = run_bench(4, [])
+    assert r["n_gpus"] == 4
+    assert r["config"]["parallelism"] == "disagg_p2tp2_d2tp2"
+    assert r["value"] > 0
+
+
+@pytest.mark.timeout(300)
+def test_bench_dp_world2():
+    r = run_bench(2, ["--mode", "dp"])
+    assert r["config"]["parallelism"] == "dp2"
+    assert r["value"] > 0
+
+
+@pytest.mark.timeout(300)
+def test_bench_moe_ep_world2():
+    """config #5 path: Mixtral disagg with expert parallelism."""
+    r = run_bench(2, ["--model", "tiny-mixtral"])
+    assert r["n_gpus"] == 2
+    assert r["value"] > 0
