@@ -216,6 +216,33 @@ def test_gumbel_sample_distribution():
     assert (freq - ref).abs().max() < 0.05
 
 
+def test_gumbel_sample_row_seeds_match_cpu():
+    """Per-row seed path (client sampling seeds): GPU kernel stream must
+    bit-match the CPU splitmix64 reference, and default to the scalar
+    stream when row_seeds[b] == seed ^ b<<32."""
+    torch.manual_seed(3)
+    B, V = 8, 5000
+    logits = torch.randn(B, V, device=DEV)
+    inv_t = torch.full((B,), 1.25, device=DEV)
+    seed = 987654321
+    scalar = ops.gumbel_sample(logits, inv_t, seed)
+    rows = torch.tensor([seed ^ (b << 32) for b in range(B)],
+                        dtype=torch.int64, device=DEV)
+    via_rows = ops.gumbel_sample(logits, inv_t, 0, rows)
+    assert torch.equal(scalar.cpu(), via_rows.cpu())
+    # arbitrary per-row seeds: GPU == CPU reference stream
+    mixed = torch.tensor([11, -5, 1 << 62, 0, 42, 42, 7, -(1 << 60)],
+                         dtype=torch.int64)
+    gpu = ops.gumbel_sample(logits, inv_t, 0, mixed.to(DEV))
+    cpu = ops.gumbel_sample(logits.cpu(), inv_t.cpu(), 0, mixed)
+    assert torch.equal(gpu.cpu(), cpu)
+    # identical seeds + identical rows sample identically
+    same = torch.full((B,), 1234, dtype=torch.int64, device=DEV)
+    eq_logits = logits[:1].repeat(B, 1)
+    out = ops.gumbel_sample(eq_logits, inv_t, 0, same)
+    assert (out == out[0]).all()
+
+
 def test_topk_gating():
     torch.manual_seed(0)
     T, E, K = 33, 8, 2
