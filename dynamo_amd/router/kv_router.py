@@ -65,6 +65,35 @@ class WorkerState:
     total_kv_pages: int = 1
     last_metrics_ts: float = 0.0
     event_task: Optional[asyncio.Task] = None
+    metrics_task: Optional[asyncio.Task] = None
+    # event-driven global load (reference selector/default.rs:217 uses
+    # event-derived state, not per-frontend counters): the worker reports
+    # running/waiting/blocks after every step; `sent_since_report` counts
+    # requests THIS frontend dispatched since the last report (covers the
+    # send->schedule window and multi-frontend blindness)
+    reported_running: int = -1      # -1 = no report yet (fall back local)
+    reported_waiting: int = 0
+    reported_blocks: int = 0
+    # deltas THIS frontend caused since the last report (sends not yet
+    # visible in the report; completions the report predates)
+    sent_since_report: int = 0
+    sent_blocks_since_report: int = 0
+    done_since_report: int = 0
+    done_blocks_since_report: int = 0
+
+    @property
+    def eff_active_requests(self) -> float:
+        if self.reported_running < 0:
+            return self.active_requests
+        return max(0, self.reported_running + self.reported_waiting
+                   + self.sent_since_report - self.done_since_report)
+
+    @property
+    def eff_active_blocks(self) -> float:
+        if self.reported_running < 0:
+            return self.active_blocks
+        return max(0, self.reported_blocks + self.sent_blocks_since_report
+                   - self.done_blocks_since_report)
 
 
 class KvRouter:
@@ -104,6 +133,8 @@ class KvRouter:
         for w in self.workers.values():
             if w.event_task:
                 w.event_task.cancel()
+            if w.metrics_task:
+                w.metrics_task.cancel()
 
     # -- discovery + event/metrics ingestion ---------------------------
     async def _watch_loop(self):
@@ -122,13 +153,15 @@ class KvRouter:
                 # pre-created by begin_request accounting)
                 ws.event_task = asyncio.create_task(
                     self._consume_events(inst.address, iid))
-                self._tasks.append(asyncio.create_task(
-                    self._poll_metrics(inst.address, iid)))
+                ws.metrics_task = asyncio.create_task(
+                    self._consume_metrics(inst.address, iid))
         for iid in list(self.workers):
             if iid not in live:
                 ws = self.workers.pop(iid)
                 if ws.event_task:
                     ws.event_task.cancel()
+                if ws.metrics_task:
+                    ws.metrics_task.cancel()
                 self.indexer.remove_worker(self._wid(iid))
 
     @staticmethod
@@ -159,20 +192,44 @@ class KvRouter:
         except Exception:
             log.exception("kv_events consumer for %s failed", iid)
 
-    async def _poll_metrics(self, address: str, iid: str):
-        while iid in self.workers:
-            try:
-                m = await self.runtime.client.call(
-                    address, f"{self.component}.get_perf_metrics", {})
-                ws = self.workers.get(iid)
-                if ws and m:
-                    ws.kv_usage = m.get("kv_usage", 0.0)
-                    ws.num_waiting = m.get("num_waiting", 0)
-                    ws.total_kv_pages = max(1, m.get("total_kv_pages", 1))
-                    ws.last_metrics_ts = time.time()
-            except Exception:
-                pass
-            await asyncio.sleep(self.cfg.metrics_poll_interval)
+    def _ingest_metrics(self, iid: str, m: dict):
+        ws = self.workers.get(iid)
+        if not ws or not m:
+            return
+        ws.kv_usage = m.get("kv_usage", 0.0)
+        ws.num_waiting = m.get("num_waiting", 0)
+        ws.total_kv_pages = max(1, m.get("total_kv_pages", 1))
+        ws.reported_running = m.get("num_running", -1)
+        ws.reported_waiting = m.get("num_waiting", 0)
+        ws.reported_blocks = m.get("active_blocks", 0)
+        ws.sent_since_report = 0
+        ws.sent_blocks_since_report = 0
+        ws.done_since_report = 0
+        ws.done_blocks_since_report = 0
+        ws.last_metrics_ts = time.time()
+
+    async def _consume_metrics(self, address: str, iid: str):
+        """Event-driven load state: one snapshot per engine step pushed by
+        the worker (reference parity: FPM over the event plane instead of
+        router polling). Falls back to 1 Hz get_perf_metrics polling for
+        workers without the streaming endpoint."""
+        try:
+            async for m in self.runtime.client.call_stream(
+                    address, f"{self.component}.metrics_events", {}):
+                self._ingest_metrics(iid, m)
+        except asyncio.CancelledError:
+            raise
+        except Exception:
+            while iid in self.workers:   # polling fallback
+                try:
+                    m = await self.runtime.client.call(
+                        address, f"{self.component}.get_perf_metrics", {})
+                    self._ingest_metrics(iid, m)
+                except asyncio.CancelledError:
+                    raise
+                except Exception:
+                    pass
+                await asyncio.sleep(self.cfg.metrics_poll_interval)
 
     # -- selection ------------------------------------------------------
     def select(self, token_ids: List[int],
@@ -252,8 +309,9 @@ class KvRouter:
             cost = (cfg.prefill_load_scale
                     * max(0.0, prefill_blocks
                           - cfg.overlap_score_weight * overlap)
-                    + ws.active_blocks
-                    + cfg.decode_active_request_weight * ws.active_requests)
+                    + ws.eff_active_blocks
+                    + cfg.decode_active_request_weight
+                    * ws.eff_active_requests)
             logits.append(cost)
         T = cfg.router_temperature
         if T <= 0:
@@ -274,7 +332,7 @@ class KvRouter:
         ws = self.workers.get(iid)
         if not ws:
             return 0.0
-        return ws.active_requests + ws.kv_usage
+        return ws.eff_active_requests + ws.kv_usage
 
     def inhibit(self, iid: str, duration: Optional[float] = None):
         """Locally exclude `iid` from selection after a failed send."""
@@ -284,13 +342,17 @@ class KvRouter:
     # -- request accounting ---------------------------------------------
     def begin_request(self, iid: str, token_ids: List[int]):
         ws = self.workers.setdefault(iid, WorkerState(iid))
+        blocks = (len(token_ids) + self.cfg.block_size - 1) // self.cfg.block_size
         ws.active_requests += 1
-        ws.active_blocks += (len(token_ids) + self.cfg.block_size - 1) // self.cfg.block_size
+        ws.active_blocks += blocks
+        ws.sent_since_report += 1
+        ws.sent_blocks_since_report += blocks
 
     def end_request(self, iid: str, token_ids: List[int]):
         ws = self.workers.get(iid)
         if ws:
+            blocks = (len(token_ids) + self.cfg.block_size - 1) // self.cfg.block_size
             ws.active_requests = max(0, ws.active_requests - 1)
-            ws.active_blocks = max(
-                0, ws.active_blocks
-                - (len(token_ids) + self.cfg.block_size - 1) // self.cfg.block_size)
+            ws.active_blocks = max(0, ws.active_blocks - blocks)
+            ws.done_since_report += 1
+            ws.done_blocks_since_report += blocks

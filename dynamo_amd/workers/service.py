@@ -65,6 +65,7 @@ class WorkerService:
         self.comp = runtime.namespace(namespace).component(component)
         self.queues: Dict[str, asyncio.Queue] = {}
         self.kv_event_subs: list = []
+        self.metrics_subs: list = []
         self._work = asyncio.Event()
         self._loop_task: Optional[asyncio.Task] = None
         self._engine_lock = asyncio.Lock()
@@ -108,6 +109,7 @@ class WorkerService:
         self.comp.serve_endpoint("generate", self.generate)
         self.comp.serve_endpoint("clear_kv_blocks", self.clear_kv_blocks)
         self.comp.serve_endpoint("get_perf_metrics", self.get_perf_metrics)
+        self.comp.serve_endpoint("metrics_events", self.metrics_events)
         self.comp.serve_endpoint("kv_events", self.kv_events)
         self.comp.serve_endpoint("release_kv", self.release_kv)
         self.comp.serve_endpoint("pause", self.pause)
@@ -174,6 +176,7 @@ class WorkerService:
                 if q is not None:
                     q.put_nowait(so)
             self._fan_kv_events()
+            self._fan_metrics()
             await asyncio.sleep(0)
 
     def _abort_inflight(self, reason: str):
@@ -208,6 +211,59 @@ class WorkerService:
                       "parent": e.parent} for e in events]
             for q in self.kv_event_subs:
                 q.put_nowait(batch)
+
+    def _metrics_snapshot(self) -> dict:
+        m = self.engine.last_metrics
+        return {
+            "worker_id": self.instance_id,
+            "worker_type": self.worker_type,
+            "step": m.step,
+            "num_running": self.engine.scheduler.num_running(),
+            "num_waiting": self.engine.scheduler.num_waiting(),
+            "kv_usage": self.engine.alloc.usage,
+            "total_kv_pages": self.engine.alloc.num_pages,
+            "active_blocks": (self.engine.alloc.num_pages
+                              - len(self.engine.alloc.free)
+                              - len(self.engine.alloc.evictable)),
+            "num_tokens_step": m.num_tokens_step,
+            "prefill_tokens_step": m.prefill_tokens_step,
+            "decode_tokens_step": m.decode_tokens_step,
+            "step_time_ms": m.step_time_ms,
+            "ts": time.time(),
+        }
+
+    def _fan_metrics(self):
+        """Push a post-step metrics snapshot to subscribers (event-driven
+        load state for routers — the reference fans ForwardPassMetrics
+        through its event plane instead of letting routers poll,
+        components/src/dynamo/common/forward_pass_metrics.py:14-28).
+        Replace-don't-queue: only the latest snapshot matters."""
+        if not self.metrics_subs:
+            return
+        snap = self._metrics_snapshot()
+        for q in self.metrics_subs:
+            while not q.empty():   # drop stale unconsumed snapshots
+                try:
+                    q.get_nowait()
+                except asyncio.QueueEmpty:
+                    break
+            q.put_nowait(snap)
+
+    async def metrics_events(self, payload, ctx):
+        """Streaming FPM subscription (one snapshot per engine step,
+        coalesced to the latest while the subscriber is slow)."""
+        q: asyncio.Queue = asyncio.Queue()
+        self.metrics_subs.append(q)
+        ctx.on_cancel(lambda: q.put_nowait(None))
+        try:
+            yield self._metrics_snapshot()   # immediate state on subscribe
+            while not ctx.cancelled:
+                snap = await q.get()
+                if snap is None:
+                    break
+                yield snap
+        finally:
+            self.metrics_subs.remove(q)
 
     # ------------------------------------------------------------------
     async def generate(self, payload: dict, ctx: RequestContext
@@ -386,21 +442,7 @@ class WorkerService:
         yield {"status": "ok"}
 
     async def get_perf_metrics(self, payload, ctx):
-        m = self.engine.last_metrics
-        yield {
-            "worker_id": self.instance_id,
-            "worker_type": self.worker_type,
-            "step": m.step,
-            "num_running": self.engine.scheduler.num_running(),
-            "num_waiting": self.engine.scheduler.num_waiting(),
-            "kv_usage": self.engine.alloc.usage,
-            "total_kv_pages": self.engine.alloc.num_pages,
-            "num_tokens_step": m.num_tokens_step,
-            "prefill_tokens_step": m.prefill_tokens_step,
-            "decode_tokens_step": m.decode_tokens_step,
-            "step_time_ms": m.step_time_ms,
-            "ts": time.time(),
-        }
+        yield self._metrics_snapshot()
 
     async def kv_events(self, payload, ctx):
         """Streaming subscription: batches of KV events (router feed).
