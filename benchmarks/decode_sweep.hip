@@ -48,13 +48,17 @@ static Bufs make(int G) {
   return bf;
 }
 
-template <int G>
+template <int G, int DEFER = 0, int PRIO = 0>
 static void run_mfma(const Bufs& bf) {
   dim3 grid(B, Hkv, bf.C);
   const int lds = mfma_lds_bytes(G, HD);
   const int iters = 30;
+  if (lds > 65536)
+    (void)hipFuncSetAttribute(
+        reinterpret_cast<const void*>(&paged_decode_mfma<DEFER, PRIO>),
+        hipFuncAttributeMaxDynamicSharedMemorySize, lds);
   for (int i = 0; i < 5; i++)
-    paged_decode_mfma<<<grid, kBlock, lds>>>(
+    paged_decode_mfma<DEFER, PRIO><<<grid, kBlock, lds>>>(
         bf.partial, bf.ml, bf.out, bf.q, bf.kc, bf.vc, bf.pt, bf.ctx,
         0.0883883f, kChunk, G, B, Hkv, bf.C, CTX / PS, 6, HD);
   CK(hipDeviceSynchronize());
@@ -62,7 +66,7 @@ static void run_mfma(const Bufs& bf) {
   CK(hipEventCreate(&e0)); CK(hipEventCreate(&e1));
   CK(hipEventRecord(e0));
   for (int i = 0; i < iters; i++) {
-    paged_decode_mfma<<<grid, kBlock, lds>>>(
+    paged_decode_mfma<DEFER, PRIO><<<grid, kBlock, lds>>>(
         bf.partial, bf.ml, bf.out, bf.q, bf.kc, bf.vc, bf.pt, bf.ctx,
         0.0883883f, kChunk, G, B, Hkv, bf.C, CTX / PS, 6, HD);
     paged_decode_phase2<<<dim3(B, G * Hkv), 128>>>(
@@ -74,7 +78,8 @@ static void run_mfma(const Bufs& bf) {
   CK(hipEventElapsedTime(&ms, e0, e1));
   double t = ms / 1000.0 / iters;
   double gb = 2.0 * B * CTX * Hkv * HD * 2 / 1e9;
-  printf("G%d MFMA            %8.1f us  %7.0f GB/s\n", G, t * 1e6, gb / t);
+  printf("G%d MFMA DF%d PR%d   %8.1f us  %7.0f GB/s\n", G, DEFER, PRIO,
+         t * 1e6, gb / t);
   fflush(stdout);
 }
 
@@ -122,7 +127,10 @@ int main() {
     run<8, 8, 4, 3>(bf, "");
     run<8, 16, 1, 2>(bf, "");
     run<8, 8, 2, 2>(bf, "");
-    run_mfma<8>(bf);
+    run_mfma<8, 0, 0>(bf);
+    run_mfma<8, 1, 0>(bf);
+    run_mfma<8, 0, 1>(bf);
+    run_mfma<8, 1, 1>(bf);
   }
   {
     Bufs bf = make(4);
@@ -134,7 +142,7 @@ int main() {
     run<4, 16, 4, 2>(bf, "");
     run<4, 8, 2, 2>(bf, "");
     run<4, 8, 1, 2>(bf, "");
-    run_mfma<4>(bf);
+    run_mfma<4, 1, 0>(bf);
   }
   {
     Bufs bf = make(1);

@@ -133,7 +133,7 @@ __device__ __forceinline__ float xor32_swap(float x, int hi) {
 //   per write) — 4x fewer LDS write ops than scalar V^T stores.
 // DEFER: skip the O-rescale when the tile max is within 8 of the running
 //   max (P bounded by e^8; guide "defer-max RESCALE_THRESHOLD").
-template <int GW, int ASTAGE, int VSTAGE = 0, int DEFER = 0>
+template <int GW, int ASTAGE, int VSTAGE = 0, int DEFER = 0, int PRIO = 0>
 __global__ __launch_bounds__(GW * 64) void prefill32_kernel(
     short* __restrict__ out,        // [B, S, Hq, 128]
     const short* __restrict__ q,    // [B, S, Hq, 128]
@@ -271,6 +271,7 @@ __global__ __launch_bounds__(GW * 64) void prefill32_kernel(
 
     // ---- QK^T: S^T tiles [32tok x 32q], toks 0-31 and 32-63 -------------
     f32x16 s0 = {}, s1 = {};
+    if constexpr (PRIO) __builtin_amdgcn_s_setprio(1);
 #pragma unroll
     for (int ds = 0; ds < 8; ds++) {
       const int koff = (ds * 32 + hi * 16);
@@ -281,6 +282,7 @@ __global__ __launch_bounds__(GW * 64) void prefill32_kernel(
       s0 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a0, qreg[ds], s0, 0, 0, 0);
       s1 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a1, qreg[ds], s1, 0, 0, 0);
     }
+    if constexpr (PRIO) __builtin_amdgcn_s_setprio(0);
 
     // ---- mask + scale + in-register softmax -----------------------------
     const int qpos = q0 + lo;
@@ -352,6 +354,7 @@ __global__ __launch_bounds__(GW * 64) void prefill32_kernel(
     }
 
     // ---- PV: O[q][d] += P[q][k] V[k][d] ---------------------------------
+    if constexpr (PRIO) __builtin_amdgcn_s_setprio(1);
 #pragma unroll
     for (int dt = 0; dt < 4; dt++) {
       const int drow = dt * 32 + lo;
@@ -363,6 +366,7 @@ __global__ __launch_bounds__(GW * 64) void prefill32_kernel(
         o[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa[ks], vb, o[dt], 0, 0, 0);
       }
     }
+    if constexpr (PRIO) __builtin_amdgcn_s_setprio(0);
 
     if constexpr (ASTAGE == 2) {
       // stage the (already loaded) next tile into the other buffer while
@@ -424,7 +428,7 @@ static void cpu_ref(std::vector<float>& o, const std::vector<short>& q,
     }
 }
 
-template <int GW, int ASTAGE, int VSTAGE = 0, int DEFER = 0>
+template <int GW, int ASTAGE, int VSTAGE = 0, int DEFER = 0, int PRIO = 0>
 static void run(Cfg c, bool check, int iters) {
   const int D = 128;
   const float scale = 1.f / sqrtf((float)D);
@@ -444,7 +448,7 @@ static void run(Cfg c, bool check, int iters) {
   int lds = KVBLK * 256 + 128 * 128;          // K + V^T = 32 KB
   if (ASTAGE == 2) lds *= 2;                  // double-buffered
   auto launch = [&] {
-    prefill32_kernel<GW, ASTAGE, VSTAGE, DEFER><<<grid, GW * 64, lds>>>(
+    prefill32_kernel<GW, ASTAGE, VSTAGE, DEFER, PRIO><<<grid, GW * 64, lds>>>(
         dout, dq, dk, dv, c.S, c.Hq, c.Hkv, scale);
   };
   launch();
@@ -461,8 +465,8 @@ static void run(Cfg c, bool check, int iters) {
       if (err > 0.05 && bad++ < 8)
         printf("  mismatch [%zu] got %f want %f\n", i, bf2f(got[i]), ref[i]);
     }
-    printf("check GW=%d AS=%d VS=%d DF=%d B=%d S=%d Hq=%d: maxerr=%.4f %s\n",
-           GW, ASTAGE, VSTAGE, DEFER, c.B, c.S, c.Hq, maxerr,
+    printf("check GW=%d AS=%d VS=%d DF=%d PR=%d B=%d S=%d Hq=%d: maxerr=%.4f %s\n",
+           GW, ASTAGE, VSTAGE, DEFER, PRIO, c.B, c.S, c.Hq, maxerr,
            bad ? "FAIL" : "PASS");
   }
   if (iters > 0) {
@@ -476,8 +480,8 @@ static void run(Cfg c, bool check, int iters) {
     float ms; hipEventElapsedTime(&ms, e0, e1);
     ms /= iters;
     const double fl = 2.0 * c.B * c.Hq * (double)c.S * c.S * D;
-    printf("bench GW=%d AS=%d VS=%d DF=%d B=%d S=%d Hq=%d Hkv=%d: %.3f ms  %.1f TF\n",
-           GW, ASTAGE, VSTAGE, DEFER, c.B, c.S, c.Hq, c.Hkv, ms,
+    printf("bench GW=%d AS=%d VS=%d DF=%d PR=%d B=%d S=%d Hq=%d Hkv=%d: %.3f ms  %.1f TF\n",
+           GW, ASTAGE, VSTAGE, DEFER, PRIO, c.B, c.S, c.Hq, c.Hkv, ms,
            fl / (ms * 1e-3) / 1e12);
   }
   hipFree(dq); hipFree(dk); hipFree(dv); hipFree(dout);
@@ -488,10 +492,12 @@ int main() {
   permprobe();
   run<8, 2, 1, 1>({1, 256, 16, 2}, true, 0);
   run<4, 2>({1, 256, 8, 2}, true, 0);
+  run<8, 2, 1, 1, 1>({1, 256, 16, 2}, true, 0);
   run<8, 1, 1, 1>({1, 8192, 64, 8}, false, 20);
-  run<8, 2, 1, 0>({1, 8192, 64, 8}, false, 20);
   run<8, 2, 1, 1>({1, 8192, 64, 8}, false, 20);
-  run<8, 2, 1, 1>({2, 4096, 64, 8}, false, 20);
+  run<8, 1, 1, 1, 1>({1, 8192, 64, 8}, false, 20);
+  run<8, 2, 1, 1, 1>({1, 8192, 64, 8}, false, 20);
+  run<8, 2, 1, 1, 1>({2, 4096, 64, 8}, false, 20);
   run<4, 2>({2, 4096, 32, 8}, false, 20);
   return 0;
 }

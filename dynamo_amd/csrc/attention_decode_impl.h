@@ -299,6 +299,11 @@ __global__ __launch_bounds__(kBlock) void paged_decode_phase1(
 // Requires page_size % 32 == 0 and hd == 128. G (<= 16) is RUNTIME — the
 // group only appears in bounds/guards, never in register shapes, so one
 // kernel serves every GQA group (e.g. qwen2's G=7).
+//   DEFER: defer-max rescale skip (guide T13): while the tile max stays
+//          within 8 of the running max, keep m_old and skip the
+//          8xf32x4 acc rescale (P bounded by e^8 — fp32 accum headroom).
+//   PRIO:  s_setprio(1) around the MFMA clusters (guide T5).
+template <int DEFER = 0, int PRIO = 0>
 __global__ __launch_bounds__(kBlock) void paged_decode_mfma(
     float* __restrict__ partial, float* __restrict__ ml,
     short* __restrict__ out, const short* __restrict__ q,
@@ -393,6 +398,7 @@ __global__ __launch_bounds__(kBlock) void paged_decode_mfma(
       const short* krA = kcache + pbase + (int64_t)(tA & (ps - 1)) * hd;
       const short* krB = kcache + pbase + (int64_t)(tB & (ps - 1)) * hd;
       const bool vA = tA < slab_end, vB = tB < slab_end;
+      if constexpr (PRIO) __builtin_amdgcn_s_setprio(1);
 #pragma unroll
       for (int kc = 0; kc < 4; kc++) {
         short8 ka = vA ? *reinterpret_cast<const short8*>(krA + kc * 32 + lg * 8)
@@ -409,6 +415,7 @@ __global__ __launch_bounds__(kBlock) void paged_decode_mfma(
               q_frag[kc], *reinterpret_cast<bf16x8_t*>(&kb2), sB, 0, 0, 0);
         }
       }
+      if constexpr (PRIO) __builtin_amdgcn_s_setprio(0);
     }
     // ---- stage V^T tile (128 dims x 32 toks, 80B row stride) ----
     // unit = (8-dim chunk, 4-token group): 4 b128 global loads, then 8
@@ -448,7 +455,10 @@ __global__ __launch_bounds__(kBlock) void paged_decode_mfma(
       float a = vA ? sA[r] * scale : kNegInf;
       float bb = vB ? sB[r] * scale : kNegInf;
       float mt = row16_reduce_max(fmaxf(a, bb));
-      if (mt > m[r]) {
+      // defer-max: mt is row-uniform after the reduce, so the skip
+      // condition is uniform across the row's 16 lanes (no vote needed)
+      const float thr = DEFER ? 8.0f : 0.0f;
+      if (mt > m[r] + thr) {
         const float corr = (m[r] <= kNegInf * 0.5f) ? 0.f : __expf(m[r] - mt);
         l[r] *= corr;
 #pragma unroll
@@ -485,6 +495,7 @@ __global__ __launch_bounds__(kBlock) void paged_decode_mfma(
           (char*)p_lds + row * 64 + ((lg * 16) ^ x));
     }
     bf16x8_t pa = *reinterpret_cast<bf16x8_t*>(&pa_s);
+    if constexpr (PRIO) __builtin_amdgcn_s_setprio(1);
 #pragma unroll
     for (int db = 0; db < 8; db++) {
       // B-frag: lane holds V[tok = 8*lg + j][dim = db*16 + lr], j = 0..7 —
@@ -495,6 +506,7 @@ __global__ __launch_bounds__(kBlock) void paged_decode_mfma(
       acc[db] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pa, vbf, acc[db],
                                                         0, 0, 0);
     }
+    if constexpr (PRIO) __builtin_amdgcn_s_setprio(0);
     }  // active
     __builtin_amdgcn_wave_barrier();
   }

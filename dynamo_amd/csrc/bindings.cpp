@@ -9,6 +9,11 @@ void rmsnorm(torch::Tensor out, torch::Tensor input, torch::Tensor weight, doubl
 void fused_add_rmsnorm(torch::Tensor input, torch::Tensor residual,
                        torch::Tensor weight, double eps);
 // rope.hip
+void rope_append_qkv(torch::Tensor q_out, torch::Tensor kcache,
+                     torch::Tensor vcache, torch::Tensor qkv,
+                     c10::optional<torch::Tensor> bias,
+                     torch::Tensor positions, torch::Tensor slot_mapping,
+                     torch::Tensor cos_sin_cache);
 void rope_inplace(torch::Tensor q, torch::Tensor k, torch::Tensor positions,
                   torch::Tensor cos_sin_cache, int64_t num_q_heads,
                   int64_t num_k_heads, int64_t head_dim);
@@ -66,6 +71,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rmsnorm", &rmsnorm);
   m.def("fused_add_rmsnorm", &fused_add_rmsnorm);
   m.def("rope_inplace", &rope_inplace);
+  m.def("rope_append_qkv", &rope_append_qkv,
+        py::arg("q_out"), py::arg("kcache"), py::arg("vcache"),
+        py::arg("qkv"), py::arg("bias"), py::arg("positions"),
+        py::arg("slot_mapping"), py::arg("cos_sin"));
   m.def("silu_mul", &silu_mul);
   m.def("gelu", &gelu);
   m.def("kv_cache_append", &kv_cache_append);

@@ -50,7 +50,124 @@ __global__ void rope_kernel(short* __restrict__ q,    // [T, Hq*hd]
   }
 }
 
+// Fused QKV epilogue: reads the QKV-GEMM output [T, (Hq+2Hkv)*hd] STRIDED
+// (no q/k .contiguous() copies), optionally adds the qkv bias, applies
+// NeoX rope to q and k, writes q to a contiguous [T, Hq*hd] buffer and
+// scatters k/v straight into the paged cache. Replaces 4 kernels
+// (q copy, k copy, rope, kv_append) with one launch — on Llama-70B decode
+// that is 3 fewer hipGraph nodes per layer x 80 layers.
+__global__ void rope_append_kernel(
+    short* __restrict__ q_out,          // [T, Hq*hd] contiguous
+    short* __restrict__ kcache,         // [P, Hkv, ps, hd]
+    short* __restrict__ vcache,
+    const short* __restrict__ qkv,      // [T, row_stride] (q|k|v packed)
+    const short* __restrict__ bias,     // [(Hq+2Hkv)*hd] or null
+    const int32_t* __restrict__ positions,  // [T]
+    const int64_t* __restrict__ slots,      // [T]
+    const float* __restrict__ cos_sin,      // [max_pos, hd]
+    int T, int Hq, int Hkv, int page_size, int hd, int row_stride) {
+  const int t = blockIdx.x;
+  if (t >= T) return;
+  const int half = hd / 2;
+  const int pos = positions[t];
+  const float* cs = cos_sin + (int64_t)pos * hd;
+  const short* row = qkv + (int64_t)t * row_stride;
+  const int64_t slot = slots[t];
+  const int64_t page = slot >= 0 ? slot / page_size : 0;
+  const int off = slot >= 0 ? (int)(slot % page_size) : 0;
+
+  // part 1: rope heads (q then k), 8 rotary indices per thread
+  const int rope_units = (Hq + Hkv) * (half / 8);
+  for (int idx = threadIdx.x; idx < rope_units; idx += kBlock) {
+    const int h = idx / (half / 8);
+    const int i0 = (idx % (half / 8)) * 8;
+    const int src_off = h * hd;  // q heads first, then k heads
+    short8 x1 = *reinterpret_cast<const short8*>(row + src_off + i0);
+    short8 x2 = *reinterpret_cast<const short8*>(row + src_off + half + i0);
+    float4v c0 = *reinterpret_cast<const float4v*>(cs + i0);
+    float4v c1 = *reinterpret_cast<const float4v*>(cs + i0 + 4);
+    float4v s0 = *reinterpret_cast<const float4v*>(cs + half + i0);
+    float4v s1 = *reinterpret_cast<const float4v*>(cs + half + i0 + 4);
+    short8 o1, o2;
+#pragma unroll
+    for (int i = 0; i < 8; i++) {
+      const float c = (i < 4) ? c0[i] : c1[i - 4];
+      const float s = (i < 4) ? s0[i] : s1[i - 4];
+      float a = bf16_to_f32(x1[i]);
+      float b = bf16_to_f32(x2[i]);
+      if (bias) {
+        a += bf16_to_f32(bias[src_off + i0 + i]);
+        b += bf16_to_f32(bias[src_off + half + i0 + i]);
+      }
+      o1[i] = f32_to_bf16(a * c - b * s);
+      o2[i] = f32_to_bf16(b * c + a * s);
+    }
+    if (h < Hq) {
+      short* qb = q_out + ((int64_t)t * Hq + h) * hd;
+      *reinterpret_cast<short8*>(qb + i0) = o1;
+      *reinterpret_cast<short8*>(qb + half + i0) = o2;
+    } else if (slot >= 0) {
+      const int kh = h - Hq;
+      short* kb = kcache + (((page * Hkv + kh) * page_size + off) * hd);
+      *reinterpret_cast<short8*>(kb + i0) = o1;
+      *reinterpret_cast<short8*>(kb + half + i0) = o2;
+    }
+  }
+  // part 2: v heads (plain copy into the cache)
+  if (slot >= 0) {
+    const int voff = (Hq + Hkv) * hd;
+    const int nvec = Hkv * hd / 8;
+    for (int i = threadIdx.x; i < nvec; i += kBlock) {
+      const int h = (i * 8) / hd;
+      const int d = (i * 8) % hd;
+      short8 v = *reinterpret_cast<const short8*>(row + voff + h * hd + d);
+      if (bias) {
+#pragma unroll
+        for (int e = 0; e < 8; e++)
+          v[e] = f32_to_bf16(bf16_to_f32(v[e]) +
+                             bf16_to_f32(bias[voff + h * hd + d + e]));
+      }
+      *reinterpret_cast<short8*>(
+          vcache + (((page * Hkv + h) * page_size + off) * hd + d)) = v;
+    }
+  }
+}
+
 }  // namespace
+
+void rope_append_qkv(torch::Tensor q_out, torch::Tensor kcache,
+                     torch::Tensor vcache, torch::Tensor qkv,
+                     c10::optional<torch::Tensor> bias,
+                     torch::Tensor positions, torch::Tensor slot_mapping,
+                     torch::Tensor cos_sin_cache) {
+  TORCH_CHECK(qkv.is_cuda() && qkv.dtype() == torch::kBFloat16);
+  TORCH_CHECK(qkv.dim() == 2 && qkv.stride(1) == 1);
+  TORCH_CHECK(positions.dtype() == torch::kInt32);
+  TORCH_CHECK(slot_mapping.dtype() == torch::kInt64);
+  TORCH_CHECK(cos_sin_cache.dtype() == torch::kFloat32);
+  const int T = qkv.size(0);
+  const int Hkv = kcache.size(1);
+  const int page_size = kcache.size(2);
+  const int hd = kcache.size(3);
+  const int Hq = q_out.numel() / (T ? (int64_t)T * hd : hd);
+  TORCH_CHECK(hd % 16 == 0 && (hd / 2) % 8 == 0);
+  TORCH_CHECK(qkv.size(1) >= (Hq + 2 * Hkv) * hd);
+  const short* bp = nullptr;
+  if (bias.has_value()) {
+    TORCH_CHECK(bias->numel() == (Hq + 2 * Hkv) * hd &&
+                bias->dtype() == torch::kBFloat16 && bias->is_contiguous());
+    bp = (const short*)bias->data_ptr();
+  }
+  if (T == 0) return;
+  auto stream = at::cuda::getCurrentHIPStream();
+  rope_append_kernel<<<T, kBlock, 0, stream>>>(
+      (short*)q_out.data_ptr(), (short*)kcache.data_ptr(),
+      (short*)vcache.data_ptr(), (const short*)qkv.data_ptr(), bp,
+      positions.data_ptr<int32_t>(), slot_mapping.data_ptr<int64_t>(),
+      cos_sin_cache.data_ptr<float>(), T, Hq, Hkv, page_size, hd,
+      (int)qkv.stride(0));
+  HIP_CHECK_KERNEL();
+}
 
 void rope_inplace(torch::Tensor q, torch::Tensor k, torch::Tensor positions,
                   torch::Tensor cos_sin_cache, int64_t num_q_heads,

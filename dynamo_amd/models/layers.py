@@ -195,25 +195,19 @@ class Attention(torch.nn.Module):
     def forward(self, x, cos_sin, kcache, vcache, meta: AttnMetadata):
         T = x.shape[0]
         qkv = linear_lora(x, self.wqkv, self.lora, "qkv")
-        if self.bqkv is not None:
-            qkv = qkv + self.bqkv
-        q, k, v = qkv.split([self.hq * self.hd, self.hkv * self.hd,
-                             self.hkv * self.hd], dim=-1)
-        q = q.contiguous()
-        k = k.contiguous()
-        q, k = ops.rope_inplace(q, k, meta.positions, cos_sin, self.hq,
-                                self.hkv, self.hd)
-        kh = k.view(T, self.hkv, self.hd)
-        vh = v.view(T, self.hkv, self.hd)
-        ops.kv_cache_append(kcache, vcache, kh, vh, meta.slot_mapping)
-
+        # fused epilogue: strided qkv read (+bias) -> rope -> q contiguous,
+        # k/v scattered into the cache (one kernel vs four)
+        q = ops.rope_append_qkv(qkv, self.bqkv, meta.positions,
+                                meta.slot_mapping, cos_sin, kcache, vcache,
+                                self.hq, self.hkv, self.hd)
         qh = q.view(T, self.hq, self.hd)
         out = torch.empty_like(qh)
         nd = meta.num_decode
         if nd:
-            out[:nd] = ops.paged_attention_decode(
+            ops.paged_attention_decode(
                 qh[:nd], kcache, vcache, meta.decode_page_table,
-                meta.decode_ctx_lens, self.scale, meta.decode_scratch)
+                meta.decode_ctx_lens, self.scale, meta.decode_scratch,
+                out=out[:nd])
         if meta.num_prefill_tokens:
             out[nd:] = ops.attention_prefill_paged(
                 qh[nd:].contiguous(), kcache, vcache, meta.prefill_page_table,

@@ -92,6 +92,32 @@ def kv_cache_append(kcache, vcache, k, v, slot_mapping):
         torch_ref.kv_cache_append(kcache, vcache, k, v, slot_mapping)
 
 
+def rope_append_qkv(qkv, bias, positions, slot_mapping, cos_sin,
+                    kcache, vcache, num_q_heads, num_kv_heads, head_dim):
+    """Fused QKV epilogue: strided qkv read (+bias) -> rope(q,k) ->
+    q contiguous out, k/v scattered into the paged cache. One kernel
+    instead of {q copy, k copy, rope, kv_append}. Returns q [T, Hq*hd]."""
+    T = qkv.shape[0]
+    if qkv.is_cuda:
+        q_out = torch.empty(T, num_q_heads * head_dim, dtype=qkv.dtype,
+                            device=qkv.device)
+        hip().rope_append_qkv(q_out, kcache, vcache, qkv, bias, positions,
+                              slot_mapping, cos_sin)
+        return q_out
+    # CPU reference: the unfused sequence
+    if bias is not None:
+        qkv = qkv + bias
+    q, k, v = qkv.split([num_q_heads * head_dim, num_kv_heads * head_dim,
+                         num_kv_heads * head_dim], dim=-1)
+    q = q.contiguous()
+    k = k.contiguous()
+    q, k = torch_ref.rope(q, k, positions, cos_sin, num_q_heads,
+                          num_kv_heads, head_dim)
+    torch_ref.kv_cache_append(kcache, vcache, k.view(T, num_kv_heads, head_dim),
+                              v.view(T, num_kv_heads, head_dim), slot_mapping)
+    return q
+
+
 class DecodeScratch:
     """Reusable scratch for two-phase flash-decode (sized once per engine)."""
 
@@ -109,17 +135,22 @@ class DecodeScratch:
 
 
 def paged_attention_decode(q, kcache, vcache, page_table, ctx_lens, scale,
-                           scratch: "DecodeScratch | None" = None):
+                           scratch: "DecodeScratch | None" = None, out=None):
     if q.is_cuda:
-        out = torch.empty_like(q)
+        if out is None:
+            out = torch.empty_like(q)
         assert scratch is not None, "GPU decode needs a DecodeScratch"
         partial, ml = scratch.view(q.shape[0])
         hip().paged_attention_decode(out, q, kcache, vcache, page_table,
                                      ctx_lens, partial, ml, scale,
                                      scratch.chunk_tokens)
         return out
-    return torch_ref.paged_attention_decode(q, kcache, vcache, page_table,
-                                            ctx_lens, scale)
+    r = torch_ref.paged_attention_decode(q, kcache, vcache, page_table,
+                                         ctx_lens, scale)
+    if out is not None:
+        out.copy_(r)
+        return out
+    return r
 
 
 def prefill_tile_rows(num_q_heads: int, num_kv_heads: int) -> int:

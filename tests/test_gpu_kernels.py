@@ -90,6 +90,37 @@ def test_kv_cache_append():
     assert_close(vc, vc2, rtol=0, atol=0)
 
 
+@pytest.mark.parametrize("bias", [False, True])
+def test_rope_append_qkv_fused(bias):
+    """Fused qkv epilogue == {bias add, split, rope, kv_append} sequence."""
+    torch.manual_seed(1)
+    Hq, Hkv, hd, P, ps, T = 8, 2, 128, 16, 64, 37
+    qkv = torch.randn(T, (Hq + 2 * Hkv) * hd, dtype=torch.bfloat16,
+                      device=DEV)
+    b = (torch.randn((Hq + 2 * Hkv) * hd, dtype=torch.bfloat16, device=DEV)
+         if bias else None)
+    pos = torch.randint(0, 500, (T,), dtype=torch.int32, device=DEV)
+    slots = torch.randperm(P * ps, device=DEV)[:T].to(torch.int64)
+    cos_sin = torch_ref.make_cos_sin_cache(512, hd, 10000.0, device=DEV)
+    kc = torch.zeros(P, Hkv, ps, hd, dtype=torch.bfloat16, device=DEV)
+    vc = torch.zeros_like(kc)
+    q_fused = ops.rope_append_qkv(qkv, b, pos, slots, cos_sin, kc, vc,
+                                  Hq, Hkv, hd)
+    # unfused reference
+    kc2 = torch.zeros_like(kc)
+    vc2 = torch.zeros_like(vc)
+    x = qkv + b if bias else qkv
+    q, k, v = x.split([Hq * hd, Hkv * hd, Hkv * hd], dim=-1)
+    q, k = ops.rope_inplace(q.contiguous(), k.contiguous(), pos, cos_sin,
+                            Hq, Hkv, hd)
+    ops.kv_cache_append(kc2, vc2, k.view(T, Hkv, hd), v.view(T, Hkv, hd),
+                        slots)
+    tol = dict(rtol=0.02, atol=0.02) if bias else dict(rtol=0, atol=0)
+    assert_close(q_fused, q, **tol)
+    assert_close(kc, kc2, **tol)
+    assert_close(vc, vc2, **tol)
+
+
 @pytest.mark.parametrize("G,ctxs", [
     (4, [1, 5, 64]),
     (4, [1000, 513, 2048, 7]),
