@@ -1,9 +1,13 @@
-from .planner import (Connector, ConstantPredictor, LoadPlanner,
+from .planner import (ARIMAPredictor, Connector, ConstantPredictor,
+                      CorrectionFactors, InterpolatedPerfModel, LoadPlanner,
                       MovingAveragePredictor, PerfModel, PlannerService,
-                      PoolObservation, PoolPolicy, SLATargets,
-                      ThroughputPlanner, TrendPredictor, VirtualConnector)
+                      PoolObservation, PoolPolicy, ScalingState, SLAPlanner,
+                      SLATargets, ThroughputPlanner, TrendPredictor,
+                      VirtualConnector)
 
-__all__ = ["Connector", "ConstantPredictor", "LoadPlanner",
+__all__ = ["ARIMAPredictor", "Connector", "ConstantPredictor",
+           "CorrectionFactors", "InterpolatedPerfModel", "LoadPlanner",
            "MovingAveragePredictor", "PerfModel", "PlannerService",
-           "PoolObservation", "PoolPolicy", "SLATargets",
-           "ThroughputPlanner", "TrendPredictor", "VirtualConnector"]
+           "PoolObservation", "PoolPolicy", "ScalingState", "SLAPlanner",
+           "SLATargets", "ThroughputPlanner", "TrendPredictor",
+           "VirtualConnector"]

@@ -80,8 +80,43 @@ class TrendPredictor(Predictor):
         return max(0.0, my + slope * ((n - 1) - mx + 1))
 
 
+class ARIMAPredictor(Predictor):
+    """AR(p) on first differences (ARIMA(p,1,0)) fitted by least squares —
+    the ARIMA-class predictor of the reference's planner family
+    (components/src/dynamo/planner core/load/predictors.py,
+    simulation/load_predictor.py) without a stats dependency. Falls back
+    to a linear trend until enough history accumulates."""
+
+    def __init__(self, window: int = 48, p: int = 4):
+        self.buf: Deque[float] = deque(maxlen=window)
+        self.p = p
+        self._trend = TrendPredictor(window=min(window, 12))
+
+    def observe(self, v):
+        self.buf.append(v)
+        self._trend.observe(v)
+
+    def predict(self):
+        n = len(self.buf)
+        if n < self.p + 4:
+            return self._trend.predict()
+        import numpy as np
+        y = np.asarray(self.buf, dtype=np.float64)
+        d = np.diff(y)                       # first differences
+        p = self.p
+        X = np.column_stack([d[i:len(d) - p + i] for i in range(p)])
+        t = d[p:]
+        try:
+            coef, *_ = np.linalg.lstsq(X, t, rcond=None)
+        except Exception:
+            return self._trend.predict()
+        nxt = float(np.dot(coef, d[-p:]))    # next difference
+        return max(0.0, y[-1] + nxt)
+
+
 PREDICTORS = {"constant": ConstantPredictor, "moving_average":
-              MovingAveragePredictor, "trend": TrendPredictor}
+              MovingAveragePredictor, "trend": TrendPredictor,
+              "arima": ARIMAPredictor}
 
 
 # -- connectors -------------------------------------------------------------
@@ -183,6 +218,172 @@ class PerfModel:
     prefill_tokens_per_s: float = 100_000.0   # per prefill replica
     decode_tokens_per_s_at_itl: float = 600.0  # per decode replica at SLA ITL
     max_conc_at_itl: int = 16
+
+
+def _interp(x: float, xs: List[float], ys: List[float]) -> float:
+    """Piecewise-linear interpolation with flat extrapolation."""
+    if not xs:
+        return 0.0
+    if x <= xs[0]:
+        return ys[0]
+    if x >= xs[-1]:
+        return ys[-1]
+    for i in range(1, len(xs)):
+        if x <= xs[i]:
+            f = (x - xs[i - 1]) / (xs[i] - xs[i - 1])
+            return ys[i - 1] + f * (ys[i] - ys[i - 1])
+    return ys[-1]
+
+
+class InterpolatedPerfModel:
+    """Perf model interpolated from a profiler concurrency sweep
+    (reference parity: planner core/perf_model/{prefill,decode}.py + AIC
+    interpolation monitoring/aic_*.py — measured (concurrency, ITL,
+    throughput) points instead of a single operating point)."""
+
+    def __init__(self, sweep: List[dict], isl: int):
+        pts = sorted((r for r in sweep if r.get("itl_p50_ms")),
+                     key=lambda r: r["concurrency"])
+        self.conc = [float(r["concurrency"]) for r in pts]
+        self.itl = [float(r["itl_p50_ms"]) for r in pts]
+        self.tps = [float(r.get("output_tok_s", 0.0)) for r in pts]
+        ttfts = [r.get("ttft_p50_s") for r in pts if r.get("ttft_p50_s")]
+        # prefill rate from the least-loaded level's TTFT
+        self.prefill_tokens_per_s = (isl / ttfts[0]) if ttfts else 1e9
+        self.isl = isl
+
+    @classmethod
+    def from_profile(cls, profile: dict) -> "InterpolatedPerfModel":
+        """Build from `python -m dynamo_amd.profiler` output JSON."""
+        return cls(profile["sweep"], int(profile.get("isl", 8192)))
+
+    def max_conc_at_itl(self, itl_slo_ms: float) -> float:
+        """Largest per-replica concurrency whose interpolated ITL meets
+        the SLO (ITL grows monotonically with concurrency)."""
+        if not self.conc:
+            return 1.0
+        if self.itl[-1] <= itl_slo_ms:
+            return self.conc[-1]
+        if self.itl[0] > itl_slo_ms:
+            return max(1.0, self.conc[0] * itl_slo_ms / self.itl[0])
+        # invert the (conc -> itl) curve at the SLO
+        return _interp(itl_slo_ms, self.itl, self.conc)
+
+    def decode_tps_at(self, conc: float) -> float:
+        return _interp(conc, self.conc, self.tps)
+
+
+class CorrectionFactors:
+    """EWMA observed/expected ratios applied to the replica computation
+    (reference: throughput_scaling.py correction factors): if measured
+    ITL or TTFT runs hotter than the perf model predicts at the current
+    load, requirements inflate proportionally."""
+
+    def __init__(self, alpha: float = 0.3, lo: float = 0.25, hi: float = 4.0):
+        self.alpha = alpha
+        self.lo, self.hi = lo, hi
+        self.prefill = 1.0
+        self.decode = 1.0
+
+    def _upd(self, cur: float, ratio: float) -> float:
+        ratio = min(self.hi, max(self.lo, ratio))
+        return (1 - self.alpha) * cur + self.alpha * ratio
+
+    def observe(self, expected_itl_ms: float, actual_itl_ms: Optional[float],
+                expected_ttft_s: float, actual_ttft_s: Optional[float]):
+        if actual_itl_ms and expected_itl_ms > 0:
+            self.decode = self._upd(self.decode,
+                                    actual_itl_ms / expected_itl_ms)
+        if actual_ttft_s and expected_ttft_s > 0:
+            self.prefill = self._upd(self.prefill,
+                                     actual_ttft_s / expected_ttft_s)
+
+
+class ScalingState:
+    STEADY = "steady"
+    SCALE_UP = "scale_up"
+    SCALE_DOWN = "scale_down"
+    COOLDOWN = "cooldown"
+
+
+class SLAPlanner:
+    """Closed-loop SLA planner: predicted request rate (ARIMA-class) ->
+    interpolated perf model + correction factors -> replica targets, with
+    an explicit scaling state machine (reference core/state_machine.py):
+    scale-up is immediate, scale-down requires `down_stable` consecutive
+    under-loaded intervals, and every action enters COOLDOWN."""
+
+    def __init__(self, sla: SLATargets, perf: InterpolatedPerfModel,
+                 connector: Connector, predictor: str = "arima",
+                 prefill_component: str = "prefill",
+                 decode_component: str = "backend",
+                 max_replicas: int = 64, cooldown_s: float = 10.0,
+                 down_stable: int = 3):
+        self.sla = sla
+        self.perf = perf
+        self.connector = connector
+        self.rate_pred = PREDICTORS[predictor]()
+        self.corrections = CorrectionFactors()
+        self.prefill_component = prefill_component
+        self.decode_component = decode_component
+        self.max_replicas = max_replicas
+        self.cooldown_s = cooldown_s
+        self.down_stable = down_stable
+        self.state = ScalingState.STEADY
+        self._last_action = 0.0
+        self._down_count: Dict[str, int] = {}
+
+    def required_replicas(self, req_per_s: float) -> Dict[str, int]:
+        c = self.corrections
+        prefill_load = req_per_s * self.sla.isl * c.prefill
+        n_prefill = max(1, math.ceil(
+            prefill_load / max(1.0, self.perf.prefill_tokens_per_s)))
+        # per-replica concurrency the ITL SLO allows, deflated by the
+        # decode correction (hotter-than-modeled => fewer slots/replica)
+        conc = max(1.0, self.perf.max_conc_at_itl(self.sla.itl_ms) / c.decode)
+        tps = max(1.0, self.perf.decode_tps_at(conc))
+        # steady-state decode demand: each request needs osl tokens over
+        # its lifetime; a replica sustains `tps` output tokens/s at SLA
+        n_decode = max(1, math.ceil(req_per_s * self.sla.osl / tps))
+        return {self.prefill_component: min(self.max_replicas, n_prefill),
+                self.decode_component: min(self.max_replicas, n_decode)}
+
+    async def observe_and_plan(self, req_per_s: float,
+                               actual_itl_ms: Optional[float] = None,
+                               actual_ttft_s: Optional[float] = None
+                               ) -> Dict[str, int]:
+        self.rate_pred.observe(req_per_s)
+        self.corrections.observe(self.sla.itl_ms, actual_itl_ms,
+                                 self.sla.ttft_s, actual_ttft_s)
+        targets = self.required_replicas(self.rate_pred.predict())
+        now = time.time()
+        if now - self._last_action < self.cooldown_s:
+            self.state = ScalingState.COOLDOWN
+            return targets
+        acted = False
+        for comp, n in targets.items():
+            cur = self.connector.current(comp)
+            if n > cur:
+                self._down_count[comp] = 0
+                await self.connector.scale(comp, n)
+                self.state = ScalingState.SCALE_UP
+                acted = True
+            elif n < cur:
+                # scale-down needs sustained under-load (state machine:
+                # avoid flapping on a transient dip)
+                self._down_count[comp] = self._down_count.get(comp, 0) + 1
+                if self._down_count[comp] >= self.down_stable:
+                    await self.connector.scale(comp, n)
+                    self.state = ScalingState.SCALE_DOWN
+                    self._down_count[comp] = 0
+                    acted = True
+            else:
+                self._down_count[comp] = 0
+        if acted:
+            self._last_action = now
+        elif self.state != ScalingState.COOLDOWN:
+            self.state = ScalingState.STEADY
+        return targets
 
 
 class ThroughputPlanner:

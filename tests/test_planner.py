@@ -125,3 +125,113 @@ def test_planner_service_with_mock_workers():
         await rt.shutdown(drain=False)
         await prt.shutdown(drain=False)
     run(main())
+
+
+# ---- round-2 planner depth: ARIMA, interpolation, corrections, SLA loop ----
+
+def test_arima_predictor_tracks_ramp():
+    from dynamo_amd.planner import ARIMAPredictor
+    p = ARIMAPredictor(window=48, p=4)
+    for i in range(30):
+        p.observe(10.0 + 2.0 * i)       # steady ramp: +2/step
+    pred = p.predict()
+    assert 68.0 <= pred <= 74.0, pred   # next value ~70 (68 observed last)
+    # a flat series predicts itself
+    p2 = ARIMAPredictor()
+    for _ in range(30):
+        p2.observe(5.0)
+    assert abs(p2.predict() - 5.0) < 0.5
+
+
+SWEEP = [
+    {"concurrency": 1, "itl_p50_ms": 10.0, "ttft_p50_s": 0.5,
+     "output_tok_s": 95.0},
+    {"concurrency": 4, "itl_p50_ms": 14.0, "ttft_p50_s": 0.6,
+     "output_tok_s": 280.0},
+    {"concurrency": 8, "itl_p50_ms": 20.0, "ttft_p50_s": 0.8,
+     "output_tok_s": 400.0},
+    {"concurrency": 16, "itl_p50_ms": 36.0, "ttft_p50_s": 1.2,
+     "output_tok_s": 444.0},
+]
+
+
+def test_interpolated_perf_model():
+    from dynamo_amd.planner import InterpolatedPerfModel
+    m = InterpolatedPerfModel(SWEEP, isl=2048)
+    # ITL 25ms sits between conc 8 (20ms) and 16 (36ms)
+    c = m.max_conc_at_itl(25.0)
+    assert 8.0 < c < 16.0
+    assert m.decode_tps_at(8) == 400.0
+    assert 400.0 < m.decode_tps_at(c) < 444.0
+    assert m.max_conc_at_itl(100.0) == 16.0   # beyond sweep: flat
+    assert m.prefill_tokens_per_s == 2048 / 0.5
+
+
+def test_correction_factors_inflate_replicas():
+    from dynamo_amd.planner import (InterpolatedPerfModel, SLAPlanner,
+                                    SLATargets, VirtualConnector)
+    sla = SLATargets(ttft_s=2.0, itl_ms=25.0, isl=2048, osl=256)
+    m = InterpolatedPerfModel(SWEEP, isl=2048)
+    pl = SLAPlanner(sla, m, VirtualConnector({"prefill": 1, "backend": 1}))
+    base = pl.required_replicas(10.0)
+    # observed ITL 2x hotter than SLO -> decode correction inflates
+    for _ in range(10):
+        pl.corrections.observe(25.0, 50.0, 2.0, 4.0)
+    hot = pl.required_replicas(10.0)
+    assert hot["backend"] >= base["backend"]
+    assert hot["prefill"] >= base["prefill"]
+
+
+def _simulate(planner_cls_kwargs, ramp, sla, model, interval_s=1.0):
+    """Closed-loop simulation: the perf model doubles as the 'cluster'.
+    Per tick: inflight = rate*osl/ (1000/itl) solved by fixed point at the
+    current replica count; observed ITL feeds back into the planner."""
+    from dynamo_amd.planner import SLAPlanner, VirtualConnector
+
+    async def run_loop():
+        conn = VirtualConnector({"prefill": 1, "backend": 1})
+        pl = SLAPlanner(sla, model, conn, cooldown_s=0.0, down_stable=2,
+                        **planner_cls_kwargs)
+        itl_hist, repl_hist = [], []
+        for rate in ramp:
+            n = max(1, conn.current("backend"))
+            # fixed point: conc/replica -> itl -> inflight -> conc
+            conc = 4.0
+            for _ in range(20):
+                itl = model.itl_at(conc)
+                inflight = rate * sla.osl * (itl / 1000.0)
+                conc = max(0.1, inflight / n)
+            itl_obs = model.itl_at(conc)
+            itl_hist.append(itl_obs)
+            repl_hist.append(n)
+            await pl.observe_and_plan(rate, actual_itl_ms=itl_obs)
+        return itl_hist, repl_hist
+
+    return asyncio.new_event_loop().run_until_complete(run_loop())
+
+
+def test_sla_planner_closed_loop_holds_slo():
+    """Ramp 1->30 req/s: the SLA planner keeps ITL at/under the SLO after
+    convergence while averaging fewer replicas than static peak
+    provisioning (the reference planner's headline claim)."""
+    from dynamo_amd.planner import InterpolatedPerfModel, SLATargets
+    sla = SLATargets(ttft_s=2.0, itl_ms=25.0, isl=2048, osl=64)
+    model = InterpolatedPerfModel(SWEEP, isl=2048)
+    # give the model an itl_at helper for the simulator
+    model.itl_at = lambda c: _interp_for_test(c, model)
+    ramp = ([1.0] * 5 + [5.0] * 5 + [12.0] * 8 + [30.0] * 12 + [8.0] * 10)
+    itl_hist, repl_hist = _simulate({}, ramp, sla, model)
+    # SLO held in steady phases (skip the 2-tick reaction window after
+    # each rate step)
+    steady = itl_hist[8:10] + itl_hist[16:18] + itl_hist[26:30]
+    assert all(v <= sla.itl_ms * 1.15 for v in steady), itl_hist
+    # fewer replicas on average than static peak provisioning
+    peak_static = max(repl_hist)
+    assert sum(repl_hist) / len(repl_hist) < peak_static
+    # scale-down happened after the ramp dropped
+    assert repl_hist[-1] < peak_static
+
+
+def _interp_for_test(c, model):
+    from dynamo_amd.planner.planner import _interp
+    return _interp(c, model.conc, model.itl)
