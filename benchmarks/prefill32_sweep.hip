@@ -396,6 +396,213 @@ __global__ __launch_bounds__(GW * 64) void prefill32_kernel(
 }
 
 // ---------------------------------------------------------------------------
+// T15 "att[2]" double-pipeline variant (guide: +7-11% attn): iteration t
+// issues QK^T MFMAs for tile t, then finishes softmax + PV of tile t-1 —
+// the softmax VALU chain of the PREVIOUS tile overlaps the current tile's
+// MFMA cluster (separate pipes). Two LDS buffers as before; two barriers
+// per tile (PV(t-1) reads vl[cur^1] which store(t+1) then overwrites).
+// Requires VSTAGE-style role-split staging (GW=8 only).
+template <int GW, int DEFER, int PRIO>
+__global__ __launch_bounds__(GW * 64) void prefill32_pipe_kernel(
+    short* __restrict__ out, const short* __restrict__ q,
+    const short* __restrict__ kc, const short* __restrict__ vc,
+    int S, int Hq, int Hkv, float scale) {
+  static_assert(GW == 8, "pipe variant: role-split staging needs 8 waves");
+  const int qt = blockIdx.x;
+  const int kvh = blockIdx.y;
+  const int bb = blockIdx.z;
+  const int w = threadIdx.x / 64;
+  const int l = threadIdx.x % 64;
+  const int h = kvh * GW + w;
+  const int q0 = qt * QBLK;
+  const int lo = l & 31, hi = l >> 5;
+
+  extern __shared__ char lds[];
+  constexpr int kHalf = KVBLK * 256 + 128 * 128;   // K + V^T per buffer
+
+  const short* qrow = q + (((int64_t)bb * S + q0 + lo) * Hq + h) * 128;
+  bf16x8 qreg[8];
+#pragma unroll
+  for (int ds = 0; ds < 8; ds++)
+    qreg[ds] = *reinterpret_cast<const bf16x8*>(qrow + ds * 16 + 8 * hi);
+
+  f32x16 o[4] = {};
+  float m_run = -1e30f, l_run = 0.f;
+  const int kv_end = q0 + QBLK;
+  const short* kbase = kc + (((int64_t)bb * Hkv + kvh) * S) * 128;
+  const short* vbase = vc + (((int64_t)bb * Hkv + kvh) * S) * 128;
+
+  const int vrole = threadIdx.x < 256;
+  short8 kreg[4];
+  auto load_tile = [&](int t0) {
+    if (vrole) {
+      const int unit = threadIdx.x;
+      const int row0 = (unit >> 4) * 4, d0 = (unit & 15) * 8;
+#pragma unroll
+      for (int j = 0; j < 4; j++)
+        kreg[j] = *reinterpret_cast<const short8*>(
+            vbase + (int64_t)(t0 + row0 + j) * 128 + d0);
+    } else {
+      const int idx = threadIdx.x - 256;
+#pragma unroll
+      for (int u = 0; u < 4; u++) {
+        const int c = idx + u * 256;
+        const int row = c >> 4, col8 = (c & 15) * 8;
+        kreg[u] = *reinterpret_cast<const short8*>(
+            kbase + (int64_t)(t0 + row) * 128 + col8);
+      }
+    }
+  };
+  auto store_tile = [&](int buf) {
+    char* kl = lds + buf * kHalf;
+    char* vl = kl + KVBLK * 256;
+    if (vrole) {
+      typedef __attribute__((ext_vector_type(4))) short short4_t;
+      const int unit = threadIdx.x;
+      const int row0 = (unit >> 4) * 4, d0 = (unit & 15) * 8;
+#pragma unroll
+      for (int i = 0; i < 8; i++) {
+        const int d = d0 + i;
+        short4_t pk = {kreg[0][i], kreg[1][i], kreg[2][i], kreg[3][i]};
+        *(short4_t*)(vl + d * 128 + ((row0 * 2) ^ ((d & 7) << 4))) = pk;
+      }
+    } else {
+      const int idx = threadIdx.x - 256;
+#pragma unroll
+      for (int u = 0; u < 4; u++) {
+        const int c = idx + u * 256;
+        const int row = c >> 4, col8 = (c & 15) * 8;
+        *reinterpret_cast<short8*>(
+            kl + row * 256 + ((col8 * 2) ^ ((row & 7) << 4))) = kreg[u];
+      }
+    }
+  };
+
+  auto qk = [&](int t0, f32x16& s0, f32x16& s1) {
+    const char* kl = lds + ((t0 / KVBLK) & 1) * kHalf;
+    if constexpr (PRIO) __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int ds = 0; ds < 8; ds++) {
+      const int koff = (ds * 32 + hi * 16);
+      const bf16x8 a0 = *reinterpret_cast<const bf16x8*>(
+          kl + lo * 256 + (koff ^ ((lo & 7) << 4)));
+      const bf16x8 a1 = *reinterpret_cast<const bf16x8*>(
+          kl + (lo + 32) * 256 + (koff ^ ((lo & 7) << 4)));
+      s0 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a0, qreg[ds], s0, 0, 0, 0);
+      s1 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a1, qreg[ds], s1, 0, 0, 0);
+    }
+    if constexpr (PRIO) __builtin_amdgcn_s_setprio(0);
+  };
+
+  auto finish = [&](int t0, const f32x16& s0, const f32x16& s1) {
+    const char* vl = lds + ((t0 / KVBLK) & 1) * kHalf + KVBLK * 256;
+    const int qpos = q0 + lo;
+    float p[32];
+    float mt = -1e30f;
+#pragma unroll
+    for (int r = 0; r < 16; r++) {
+      const int trow = (r & 3) + 8 * (r >> 2) + 4 * hi;
+      p[r] = (t0 + trow <= qpos) ? s0[r] * scale : -1e30f;
+      p[16 + r] = (t0 + 32 + trow <= qpos) ? s1[r] * scale : -1e30f;
+      mt = fmaxf(mt, fmaxf(p[r], p[16 + r]));
+    }
+    mt = fmaxf(mt, xor32_swap(mt, hi));
+    float m_new = fmaxf(m_run, mt);
+    bool skip_rescale = false;
+    if constexpr (DEFER) {
+      if (__all(mt - m_run <= 8.0f)) { m_new = m_run; skip_rescale = true; }
+    }
+    const float alpha = skip_rescale ? 1.f : __expf(m_run - m_new);
+    float ls = 0.f;
+#pragma unroll
+    for (int r = 0; r < 32; r++) {
+      p[r] = __expf(p[r] - m_new);
+      ls += p[r];
+    }
+    ls += xor32_swap(ls, hi);
+    l_run = l_run * alpha + ls;
+    m_run = m_new;
+    if (!skip_rescale) {
+      float arow[16];
+#pragma unroll
+      for (int r = 0; r < 16; r++)
+        arow[r] = __shfl(alpha, (r & 3) + 8 * (r >> 2) + 4 * hi, 64);
+#pragma unroll
+      for (int dt = 0; dt < 4; dt++)
+#pragma unroll
+        for (int r = 0; r < 16; r++) o[dt][r] *= arow[r];
+    }
+    auto cvtpk = [](float a, float b) {
+      unsigned int r;
+      asm("v_cvt_pk_bf16_f32 %0, %1, %2" : "=v"(r) : "v"(a), "v"(b));
+      return r;
+    };
+    bf16x8 pa[4];
+#pragma unroll
+    for (int ks = 0; ks < 4; ks++) {
+      const int b0 = 8 * (ks & 1) + 16 * (ks >> 1);
+      uint2_t rA = __builtin_amdgcn_permlane32_swap(
+          cvtpk(p[b0], p[b0 + 1]), cvtpk(p[b0 + 4], p[b0 + 5]), false, false);
+      uint2_t rB = __builtin_amdgcn_permlane32_swap(
+          cvtpk(p[b0 + 2], p[b0 + 3]), cvtpk(p[b0 + 6], p[b0 + 7]),
+          false, false);
+      unsigned int wds[4] = {rA.x, rB.x, rA.y, rB.y};
+      pa[ks] = *reinterpret_cast<bf16x8*>(wds);
+    }
+    if constexpr (PRIO) __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int dt = 0; dt < 4; dt++) {
+      const int drow = dt * 32 + lo;
+#pragma unroll
+      for (int ks = 0; ks < 4; ks++) {
+        const int toff = (ks * 16 + 8 * hi) * 2;
+        const bf16x8 vb = *reinterpret_cast<const bf16x8*>(
+            vl + drow * 128 + (toff ^ ((drow & 7) << 4)));
+        o[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa[ks], vb, o[dt],
+                                                        0, 0, 0);
+      }
+    }
+    if constexpr (PRIO) __builtin_amdgcn_s_setprio(0);
+  };
+
+  // prologue: stage tile 0, prefetch tile 1 into regs
+  load_tile(0);
+  store_tile(0);
+  __syncthreads();
+  if (KVBLK < kv_end) load_tile(KVBLK);
+
+  f32x16 sp0 = {}, sp1 = {};
+  int prev = -1;
+  for (int t0 = 0; t0 < kv_end; t0 += KVBLK) {
+    f32x16 s0 = {}, s1 = {};
+    qk(t0, s0, s1);                 // MFMA cluster for tile t ...
+    if (prev >= 0) finish(prev, sp0, sp1);   // ... overlaps VALU of t-1
+    sp0 = s0; sp1 = s1; prev = t0;
+    __syncthreads();                // PV(t-1) readers done with buf cur^1
+    if (t0 + KVBLK < kv_end) {
+      store_tile(((t0 / KVBLK) & 1) ^ 1);    // tile t+1 -> buf cur^1
+      if (t0 + 2 * KVBLK < kv_end) load_tile(t0 + 2 * KVBLK);
+    }
+    __syncthreads();                // store(t+1) visible before QK(t+1)
+  }
+  if (prev >= 0) finish(prev, sp0, sp1);     // drain the pipeline
+
+  float lrow[16];
+#pragma unroll
+  for (int r = 0; r < 16; r++)
+    lrow[r] = __shfl(l_run, (r & 3) + 8 * (r >> 2) + 4 * hi, 64);
+#pragma unroll
+  for (int dt = 0; dt < 4; dt++)
+#pragma unroll
+    for (int r = 0; r < 16; r++) {
+      const int qrow_i = (r & 3) + 8 * (r >> 2) + 4 * hi;
+      const int d = dt * 32 + lo;
+      out[(((int64_t)bb * S + q0 + qrow_i) * Hq + h) * 128 + d] =
+          f2bf(o[dt][r] / lrow[r]);
+    }
+}
+
+// ---------------------------------------------------------------------------
 struct Cfg { int B, S, Hq, Hkv; };
 
 static void cpu_ref(std::vector<float>& o, const std::vector<short>& q,
@@ -487,9 +694,70 @@ static void run(Cfg c, bool check, int iters) {
   hipFree(dq); hipFree(dk); hipFree(dv); hipFree(dout);
 }
 
+template <int GW, int DEFER, int PRIO>
+static void run_pipe(Cfg c, bool check, int iters) {
+  const int D = 128;
+  const float scale = 1.f / sqrtf((float)D);
+  size_t qe = (size_t)c.B * c.S * c.Hq * D, ke = (size_t)c.B * c.Hkv * c.S * D;
+  std::vector<short> hq(qe), hk(ke), hv(ke);
+  srand(42);
+  for (auto& x : hq) x = f2bf((rand() % 2001 - 1000) / 1000.f);
+  for (auto& x : hk) x = f2bf((rand() % 2001 - 1000) / 1000.f);
+  for (auto& x : hv) x = f2bf((rand() % 2001 - 1000) / 1000.f);
+  short *dq, *dk, *dv, *dout;
+  CK(hipMalloc(&dq, qe * 2)); CK(hipMalloc(&dk, ke * 2));
+  CK(hipMalloc(&dv, ke * 2)); CK(hipMalloc(&dout, qe * 2));
+  CK(hipMemcpy(dq, hq.data(), qe * 2, hipMemcpyHostToDevice));
+  CK(hipMemcpy(dk, hk.data(), ke * 2, hipMemcpyHostToDevice));
+  CK(hipMemcpy(dv, hv.data(), ke * 2, hipMemcpyHostToDevice));
+  dim3 grid(c.S / QBLK, c.Hkv, c.B);
+  int lds = 2 * (KVBLK * 256 + 128 * 128);
+  auto launch = [&] {
+    prefill32_pipe_kernel<GW, DEFER, PRIO><<<grid, GW * 64, lds>>>(
+        dout, dq, dk, dv, c.S, c.Hq, c.Hkv, scale);
+  };
+  launch();
+  CK(hipDeviceSynchronize());
+  if (check) {
+    std::vector<short> got(qe);
+    CK(hipMemcpy(got.data(), dout, qe * 2, hipMemcpyDeviceToHost));
+    std::vector<float> ref(qe);
+    cpu_ref(ref, hq, hk, hv, c, scale);
+    double maxerr = 0; int bad = 0;
+    for (size_t i = 0; i < qe; i++) {
+      const double err = fabs(bf2f(got[i]) - ref[i]);
+      if (err > maxerr) maxerr = err;
+      if (err > 0.05 && bad++ < 8)
+        printf("  mismatch [%zu] got %f want %f\n", i, bf2f(got[i]), ref[i]);
+    }
+    printf("check PIPE GW=%d DF=%d PR=%d B=%d S=%d Hq=%d: maxerr=%.4f %s\n",
+           GW, DEFER, PRIO, c.B, c.S, c.Hq, maxerr, bad ? "FAIL" : "PASS");
+  }
+  if (iters > 0) {
+    for (int i = 0; i < 3; i++) launch();
+    CK(hipDeviceSynchronize());
+    hipEvent_t e0, e1; hipEventCreate(&e0); hipEventCreate(&e1);
+    hipEventRecord(e0);
+    for (int i = 0; i < iters; i++) launch();
+    hipEventRecord(e1);
+    CK(hipDeviceSynchronize());
+    float ms; hipEventElapsedTime(&ms, e0, e1);
+    ms /= iters;
+    const double fl = 2.0 * c.B * c.Hq * (double)c.S * c.S * D;
+    printf("bench PIPE GW=%d DF=%d PR=%d B=%d S=%d Hq=%d Hkv=%d: %.3f ms  %.1f TF\n",
+           GW, DEFER, PRIO, c.B, c.S, c.Hq, c.Hkv, ms,
+           fl / (ms * 1e-3) / 1e12);
+  }
+  (void)hipFree(dq); (void)hipFree(dk); (void)hipFree(dv); (void)hipFree(dout);
+}
+
 int main() {
   if (!probe32()) return 1;
   permprobe();
+  run_pipe<8, 1, 0>({1, 256, 16, 2}, true, 0);
+  run_pipe<8, 1, 0>({1, 8192, 64, 8}, false, 20);
+  run_pipe<8, 1, 1>({1, 8192, 64, 8}, false, 20);
+  run_pipe<8, 1, 1>({2, 4096, 64, 8}, false, 20);
   run<8, 2, 1, 1>({1, 256, 16, 2}, true, 0);
   run<4, 2>({1, 256, 8, 2}, true, 0);
   run<8, 2, 1, 1, 1>({1, 256, 16, 2}, true, 0);
