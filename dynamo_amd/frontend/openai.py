@@ -87,6 +87,19 @@ class AnthropicMessagesRequest(BaseModel):
     user: Optional[str] = None
 
 
+class ResponsesRequest(BaseModel):
+    """OpenAI Responses API shape (reference parity:
+    lib/llm/src/http/service/openai.rs:4158 handler_responses)."""
+    model: str = ""
+    input: Union[str, List[dict]] = ""
+    instructions: Optional[str] = None
+    max_output_tokens: int = 128
+    temperature: float = 0.0
+    top_p: float = 1.0
+    stream: bool = False
+    user: Optional[str] = None
+
+
 def _session_of(req, raw: Request) -> Optional[str]:
     return raw.headers.get("x-session-id") or getattr(req, "user", None)
 
@@ -198,6 +211,200 @@ def build_app(manager: ModelManager) -> FastAPI:
             return manager.get(model)
         except KeyError as e:
             raise HTTPException(404, str(e))
+
+    # -- OpenAI Responses API (openai.rs:4158 handler_responses) --------
+    @app.post("/v1/responses")
+    async def responses(req: ResponsesRequest, raw: Request):
+        entry = _entry_or_404(req.model)
+        REQS.labels(entry.name, "responses").inc()
+        if isinstance(req.input, str):
+            msgs = [{"role": "user", "content": req.input}]
+        else:
+            msgs = [{"role": m.get("role", "user"),
+                     "content": (m.get("content") if isinstance(
+                         m.get("content"), str) else "".join(
+                             c.get("text", "") for c in m.get("content", [])
+                             if isinstance(c, dict)))}
+                    for m in req.input]
+        if req.instructions:
+            msgs = [{"role": "system", "content": req.instructions}] + msgs
+        prompt = entry.templater.render(msgs)
+        token_ids = entry.tokenizer.encode(prompt)
+        rid = f"resp_{uuid.uuid4().hex[:24]}"
+        t0 = time.time()
+        shim = CompletionRequest(model=req.model, max_tokens=req.max_output_tokens,
+                                 temperature=req.temperature, top_p=req.top_p)
+
+        if req.stream:
+            async def sse():
+                produced: List[int] = []
+                first = True
+                try:
+                    yield ("event: response.created\ndata: " + json.dumps(
+                        {"type": "response.created",
+                         "response": {"id": rid, "object": "response",
+                                      "status": "in_progress",
+                                      "model": entry.name}}) + "\n\n")
+                    async for chunk in _run(entry, token_ids, shim, rid,
+                                            _session_of(req, raw)):
+                        if await raw.is_disconnected():
+                            break
+                        if first:
+                            TTFT.labels(entry.name).observe(time.time() - t0)
+                            first = False
+                        prev = len(produced)
+                        produced.extend(chunk.get("token_ids", []))
+                        delta = entry.tokenizer.decode_incremental(produced,
+                                                                   prev)
+                        if delta:
+                            yield ("event: response.output_text.delta\n"
+                                   "data: " + json.dumps(
+                                       {"type": "response.output_text.delta",
+                                        "delta": delta}) + "\n\n")
+                    text = entry.tokenizer.decode(produced)
+                    yield ("event: response.completed\ndata: " + json.dumps(
+                        {"type": "response.completed",
+                         "response": _response_body(rid, entry.name, text,
+                                                    len(token_ids),
+                                                    len(produced))}) + "\n\n")
+                finally:
+                    LATENCY.labels(entry.name).observe(time.time() - t0)
+            return StreamingResponse(sse(), media_type="text/event-stream")
+
+        produced: List[int] = []
+        async for chunk in _run(entry, token_ids, shim, rid,
+                                _session_of(req, raw)):
+            produced.extend(chunk.get("token_ids", []))
+        LATENCY.labels(entry.name).observe(time.time() - t0)
+        return _response_body(rid, entry.name, entry.tokenizer.decode(produced),
+                              len(token_ids), len(produced))
+
+    def _response_body(rid, model, text, in_toks, out_toks):
+        return {
+            "id": rid, "object": "response", "status": "completed",
+            "created_at": int(time.time()), "model": model,
+            "output": [{"type": "message", "id": f"msg_{rid[5:]}",
+                        "role": "assistant", "status": "completed",
+                        "content": [{"type": "output_text", "text": text,
+                                     "annotations": []}]}],
+            "output_text": text,
+            "usage": {"input_tokens": in_toks, "output_tokens": out_toks,
+                      "total_tokens": in_toks + out_toks},
+        }
+
+    # -- Files + Batches (openai.rs:3984-3987 batch routes) --------------
+    files: dict = {}     # file_id -> {"bytes": b..., "filename": ...}
+    batches: dict = {}   # batch_id -> status dict
+
+    @app.post("/v1/files")
+    async def create_file(raw: Request):
+        # multipart when python-multipart is installed; otherwise accept
+        # the raw body as the file content (purpose/filename via query)
+        filename = raw.query_params.get("filename", "upload.jsonl")
+        purpose = raw.query_params.get("purpose", "batch")
+        data = None
+        ctype = raw.headers.get("content-type", "")
+        if ctype.startswith("multipart/"):
+            try:
+                form = await raw.form()
+                up = form.get("file")
+                if up is not None:
+                    data = await up.read()
+                    filename = up.filename or filename
+                    purpose = form.get("purpose", purpose)
+            except Exception:
+                pass
+        if data is None:
+            data = await raw.body()
+        if not data:
+            raise HTTPException(400, "empty file upload")
+        fid = f"file-{uuid.uuid4().hex[:24]}"
+        files[fid] = {"bytes": data, "filename": filename,
+                      "purpose": purpose}
+        return {"id": fid, "object": "file", "bytes": len(data),
+                "filename": filename, "purpose": purpose,
+                "created_at": int(time.time())}
+
+    @app.get("/v1/files/{fid}/content")
+    async def file_content(fid: str):
+        if fid not in files:
+            raise HTTPException(404, f"file {fid} not found")
+        from fastapi import Response
+        return Response(files[fid]["bytes"],
+                        media_type="application/octet-stream")
+
+    async def _run_batch(bid: str):
+        b = batches[bid]
+        b["status"] = "in_progress"
+        out_lines = []
+        nerr = 0
+        for line in files[b["input_file_id"]]["bytes"].splitlines():
+            if not line.strip():
+                continue
+            item = None
+            try:
+                item = json.loads(line)
+                body = item.get("body", {})
+                entry = manager.get(body.get("model", ""))
+                if b["endpoint"] == "/v1/chat/completions":
+                    prompt = entry.templater.render(body.get("messages", []))
+                else:
+                    prompt = body.get("prompt", "")
+                toks = entry.tokenizer.encode(prompt)
+                produced = []
+                async for ch in manager.generate_tokens(
+                        entry, toks,
+                        {"temperature": body.get("temperature", 0.0)},
+                        {"max_tokens": body.get("max_tokens", 128)}):
+                    produced.extend(ch.get("token_ids", []))
+                text = entry.tokenizer.decode(produced)
+                out_lines.append(json.dumps({
+                    "id": f"batch_req_{uuid.uuid4().hex[:16]}",
+                    "custom_id": item.get("custom_id"),
+                    "response": {"status_code": 200, "body": {
+                        "choices": [{"index": 0, "message": {
+                            "role": "assistant", "content": text},
+                            "finish_reason": "stop"}]}},
+                    "error": None}))
+                b["request_counts"]["completed"] += 1
+            except Exception as e:
+                nerr += 1
+                b["request_counts"]["failed"] += 1
+                out_lines.append(json.dumps({
+                    "custom_id": item.get("custom_id") if item else None,
+                    "response": None,
+                    "error": {"message": str(e)}}))
+        ofid = f"file-{uuid.uuid4().hex[:24]}"
+        files[ofid] = {"bytes": "\n".join(out_lines).encode(),
+                       "filename": f"{bid}_output.jsonl", "purpose": "batch_output"}
+        b["output_file_id"] = ofid
+        b["status"] = "completed" if nerr == 0 else "completed"
+        b["completed_at"] = int(time.time())
+
+    @app.post("/v1/batches")
+    async def create_batch(raw: Request):
+        body = await raw.json()
+        fid = body.get("input_file_id")
+        if fid not in files:
+            raise HTTPException(404, f"input file {fid} not found")
+        bid = f"batch_{uuid.uuid4().hex[:24]}"
+        batches[bid] = {
+            "id": bid, "object": "batch", "status": "validating",
+            "endpoint": body.get("endpoint", "/v1/chat/completions"),
+            "input_file_id": fid, "output_file_id": None,
+            "created_at": int(time.time()), "completed_at": None,
+            "request_counts": {"total": sum(
+                1 for ln in files[fid]["bytes"].splitlines() if ln.strip()),
+                "completed": 0, "failed": 0},
+        }
+        asyncio.get_running_loop().create_task(_run_batch(bid))
+        return batches[bid]
+
+    @app.get("/v1/batches/{bid}")
+    async def get_batch(bid: str):
+        if bid not in batches:
+            raise HTTPException(404, f"batch {bid} not found")
+        return batches[bid]
 
     @app.post("/v1/completions")
     async def completions(req: CompletionRequest, raw: Request):

@@ -902,3 +902,71 @@ def test_kv_events_snapshot_on_subscribe():
         await late_rt.shutdown(drain=False)
         await teardown(services, mgr, client)
     run(main())
+
+
+def test_responses_api_unary_and_stream():
+    """OpenAI Responses API surface (reference openai.rs:4158)."""
+    async def main():
+        shared, services, mgr, client = await with_stack(nworkers=1)
+        r = await client.post("/v1/responses", json={
+            "model": "mock-model", "input": "hello there",
+            "max_output_tokens": 6})
+        assert r.status_code == 200, r.text
+        body = r.json()
+        assert body["object"] == "response"
+        assert body["status"] == "completed"
+        assert body["usage"]["output_tokens"] == 6
+        assert body["output"][0]["content"][0]["type"] == "output_text"
+        # structured message-list input + instructions + streaming
+        async with client.stream("POST", "/v1/responses", json={
+                "model": "mock-model", "instructions": "be brief",
+                "input": [{"role": "user", "content": "hi"}],
+                "max_output_tokens": 5, "stream": True}) as resp:
+            assert resp.status_code == 200
+            events = []
+            async for line in resp.aiter_lines():
+                if line.startswith("event: "):
+                    events.append(line.split(" ", 1)[1])
+        assert events[0] == "response.created"
+        assert "response.output_text.delta" in events
+        assert events[-1] == "response.completed"
+        await teardown(services, mgr, client)
+    run(main())
+
+
+def test_files_and_batches_api():
+    """Batch surface (reference openai.rs:3984-3987): upload a JSONL of
+    chat requests, create a batch, poll to completion, fetch output."""
+    async def main():
+        shared, services, mgr, client = await with_stack(nworkers=1)
+        import json as js
+        lines = [js.dumps({
+            "custom_id": f"c{i}",
+            "method": "POST", "url": "/v1/chat/completions",
+            "body": {"model": "mock-model",
+                     "messages": [{"role": "user", "content": f"q {i}"}],
+                     "max_tokens": 4}}) for i in range(3)]
+        r = await client.post("/v1/files?purpose=batch&filename=b.jsonl",
+                              content="\n".join(lines).encode())
+        assert r.status_code == 200, r.text
+        fid = r.json()["id"]
+        r = await client.post("/v1/batches", json={
+            "input_file_id": fid, "endpoint": "/v1/chat/completions"})
+        assert r.status_code == 200, r.text
+        bid = r.json()["id"]
+        for _ in range(100):
+            r = await client.get(f"/v1/batches/{bid}")
+            if r.json()["status"] == "completed":
+                break
+            await asyncio.sleep(0.1)
+        b = r.json()
+        assert b["status"] == "completed", b
+        assert b["request_counts"]["completed"] == 3
+        assert b["request_counts"]["failed"] == 0
+        r = await client.get(f"/v1/files/{b['output_file_id']}/content")
+        out = [js.loads(ln) for ln in r.content.splitlines() if ln.strip()]
+        assert len(out) == 3
+        assert {o["custom_id"] for o in out} == {"c0", "c1", "c2"}
+        assert all(o["response"]["status_code"] == 200 for o in out)
+        await teardown(services, mgr, client)
+    run(main())
