@@ -268,3 +268,56 @@ def test_file_discovery_ttl_expiry(tmp_path):
             old = time.time() - 5
             os.utime(p, (old, old))
     assert d.list("dynamo") == []
+
+
+def test_otlp_exporter_posts_spans():
+    """OTLP/HTTP export hook: spans emitted through RequestTracer reach a
+    local collector endpoint as OTLP JSON (reference parity:
+    observability-architecture.md OTEL_EXPORT_ENABLED + request_trace
+    OTLP sink)."""
+    import http.server
+    import json as js
+    import threading
+
+    got = []
+
+    class H(http.server.BaseHTTPRequestHandler):
+        def do_POST(self):
+            n = int(self.headers.get("Content-Length", 0))
+            got.append((self.path, js.loads(self.rfile.read(n))))
+            self.send_response(200)
+            self.end_headers()
+            self.wfile.write(b"{}")
+
+        def log_message(self, *a):
+            pass
+
+    srv = http.server.HTTPServer(("127.0.0.1", 0), H)
+    t = threading.Thread(target=srv.serve_forever, daemon=True)
+    t.start()
+    try:
+        from dynamo_amd.observability import OtlpExporter, RequestTracer
+        from dynamo_amd.observability import set_tracer, span, trace_event
+        exp = OtlpExporter(
+            endpoint=f"http://127.0.0.1:{srv.server_port}",
+            flush_interval=30.0)
+        tracer = RequestTracer(path=None, otlp=exp)
+        set_tracer(tracer)
+        with span("http-request", request_id="req-42", model="m"):
+            trace_event("handle_payload", worker_id="w1")
+        set_tracer(None)
+        exp.flush()
+        assert exp.exported >= 2 and exp.errors == 0
+        path, body = got[0]
+        assert path == "/v1/traces"
+        spans = body["resourceSpans"][0]["scopeSpans"][0]["spans"]
+        names = {s["name"] for s in spans}
+        assert "http-request" in names and "handle_payload" in names
+        rs = body["resourceSpans"][0]["resource"]["attributes"]
+        assert any(a["key"] == "service.name" for a in rs)
+        # parent/child share the trace id
+        tid = {s["traceId"] for s in spans}
+        assert len(tid) == 1
+        tracer.close()
+    finally:
+        srv.shutdown()
