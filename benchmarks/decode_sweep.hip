@@ -15,6 +15,29 @@ using namespace decode_attn;
 
 constexpr int B = 16, Hkv = 8, PS = 64, HD = 128, CTX = 8192;
 
+// probe: permlane16_swap semantics (the swapped-MFMA P-exchange relies on
+// x = {l%32<16: v0(l), else: v1(l-16)}, y = {l%32<16: v0(l+16), else: v1(l)})
+__global__ void perm16_probe_kernel(unsigned int* out) {
+  typedef __attribute__((ext_vector_type(2))) unsigned int uint2_t;
+  unsigned int v0 = threadIdx.x, v1 = 1000 + threadIdx.x;
+  uint2_t r = __builtin_amdgcn_permlane16_swap(v0, v1, false, false);
+  out[threadIdx.x * 2] = r.x;
+  out[threadIdx.x * 2 + 1] = r.y;
+}
+
+static void perm16_probe() {
+  unsigned int* d; CK(hipMalloc(&d, 128 * 4));
+  perm16_probe_kernel<<<1, 64>>>(d);
+  CK(hipDeviceSynchronize());
+  std::vector<unsigned int> h(128);
+  CK(hipMemcpy(h.data(), d, 128 * 4, hipMemcpyDeviceToHost));
+  printf("permlane16_swap l0:(%u,%u) l16:(%u,%u) l5:(%u,%u) l21:(%u,%u) "
+         "l37:(%u,%u) l53:(%u,%u)\n",
+         h[0], h[1], h[32], h[33], h[10], h[11], h[42], h[43],
+         h[74], h[75], h[106], h[107]);
+  (void)hipFree(d);
+}
+
 struct Bufs {
   short *kc, *vc, *q, *out;
   float *partial, *ml;
@@ -41,11 +64,93 @@ static Bufs make(int G) {
   std::vector<int32_t> cl(B, CTX);
   CK(hipMalloc(&bf.ctx, B * 4));
   CK(hipMemcpy(bf.ctx, cl.data(), B * 4, hipMemcpyHostToDevice));
-  // fill kv with a pattern (values don't matter for timing)
-  CK(hipMemset(bf.kc, 0x3c, cache_e * 2));
-  CK(hipMemset(bf.vc, 0x3c, cache_e * 2));
-  CK(hipMemset(bf.q, 0x3c, (size_t)B * Hq * HD * 2));
+  // fill kv with RANDOM bf16 in [-1,1] (numerics cross-checks need it;
+  // bandwidth timing is data-independent)
+  auto fill = [](short* dptr, size_t n) {
+    std::vector<short> h(n);
+    for (auto& x : h) {
+      float f = (rand() % 2001 - 1000) / 1000.f;
+      union { float ff; unsigned int i; } c; c.ff = f;
+      unsigned int r = c.i + 0x7fff + ((c.i >> 16) & 1);
+      h[&x - h.data()] = (short)(r >> 16);
+    }
+    CK(hipMemcpy(dptr, h.data(), n * 2, hipMemcpyHostToDevice));
+  };
+  srand(7);
+  fill(bf.kc, cache_e);
+  fill(bf.vc, cache_e);
+  fill(bf.q, (size_t)B * Hq * HD);
   return bf;
+}
+
+// swapped-operand MFMA variant: bench + elementwise compare vs the proven
+// paged_decode_mfma output
+template <int G, int DEFER = 1, int PRIO = 1>
+static void run_mfma_swapped(const Bufs& bf, bool check) {
+  dim3 grid(B, Hkv, bf.C);
+  const int lds = mfma_swapped_lds_bytes(G, HD);
+  const int iters = 30;
+  if (lds > 65536)
+    (void)hipFuncSetAttribute(
+        reinterpret_cast<const void*>(&paged_decode_mfma_swapped<DEFER, PRIO>),
+        hipFuncAttributeMaxDynamicSharedMemorySize, lds);
+  auto launch = [&] {
+    paged_decode_mfma_swapped<DEFER, PRIO><<<grid, kBlock, lds>>>(
+        bf.partial, bf.ml, bf.out, bf.q, bf.kc, bf.vc, bf.pt, bf.ctx,
+        0.0883883f, kChunk, G, B, Hkv, bf.C, CTX / PS, 6, HD);
+    paged_decode_phase2<<<dim3(B, G * Hkv), 128>>>(
+        bf.out, bf.partial, bf.ml, bf.ctx, kChunk, G * Hkv, bf.C, HD);
+  };
+  if (check) {
+    const size_t n = (size_t)B * G * Hkv * HD;
+    launch();
+    CK(hipDeviceSynchronize());
+    std::vector<short> got(n);
+    CK(hipMemcpy(got.data(), bf.out, n * 2, hipMemcpyDeviceToHost));
+    // reference: the existing (validated) kernel
+    const int lds0 = mfma_lds_bytes(G, HD);
+    if (lds0 > 65536)
+      (void)hipFuncSetAttribute(
+          reinterpret_cast<const void*>(&paged_decode_mfma<0, 0>),
+          hipFuncAttributeMaxDynamicSharedMemorySize, lds0);
+    paged_decode_mfma<0, 0><<<grid, kBlock, lds0>>>(
+        bf.partial, bf.ml, bf.out, bf.q, bf.kc, bf.vc, bf.pt, bf.ctx,
+        0.0883883f, kChunk, G, B, Hkv, bf.C, CTX / PS, 6, HD);
+    paged_decode_phase2<<<dim3(B, G * Hkv), 128>>>(
+        bf.out, bf.partial, bf.ml, bf.ctx, kChunk, G * Hkv, bf.C, HD);
+    CK(hipDeviceSynchronize());
+    std::vector<short> ref(n);
+    CK(hipMemcpy(ref.data(), bf.out, n * 2, hipMemcpyDeviceToHost));
+    auto b2f = [](short u) {
+      union { float f; unsigned int i; } c;
+      c.i = ((unsigned int)(unsigned short)u) << 16; return c.f;
+    };
+    double maxerr = 0; int bad = 0;
+    for (size_t i = 0; i < n; i++) {
+      double e = fabs(b2f(got[i]) - b2f(ref[i]));
+      if (e > maxerr) maxerr = e;
+      if (e > 0.03 && bad++ < 6)
+        printf("  sw mismatch [%zu] got %f want %f\n", i,
+               b2f(got[i]), b2f(ref[i]));
+    }
+    printf("check G%d MFMA_SW vs MFMA: maxerr=%.4f %s\n", G, maxerr,
+           bad ? "FAIL" : "PASS");
+  }
+  for (int i = 0; i < 5; i++) launch();
+  CK(hipDeviceSynchronize());
+  hipEvent_t e0, e1;
+  CK(hipEventCreate(&e0)); CK(hipEventCreate(&e1));
+  CK(hipEventRecord(e0));
+  for (int i = 0; i < iters; i++) launch();
+  CK(hipEventRecord(e1));
+  CK(hipEventSynchronize(e1));
+  float ms;
+  CK(hipEventElapsedTime(&ms, e0, e1));
+  double t = ms / 1000.0 / iters;
+  double gb = 2.0 * B * CTX * Hkv * HD * 2 / 1e9;
+  printf("G%d MFMA_SW DF%d PR%d %8.1f us  %7.0f GB/s\n", G, DEFER, PRIO,
+         t * 1e6, gb / t);
+  fflush(stdout);
 }
 
 template <int G, int DEFER = 0, int PRIO = 0>
@@ -116,8 +221,11 @@ static void run(const Bufs& bf, const char* tag) {
 }
 
 int main() {
+  perm16_probe();
   {
     Bufs bf = make(8);
+    run_mfma_swapped<8, 0, 0>(bf, true);
+    run_mfma_swapped<8, 1, 1>(bf, false);
     run<8, 16, 4, 2>(bf, "");
     run<8, 16, 4, 3>(bf, "");
     run<8, 16, 4, 4>(bf, "");
@@ -143,6 +251,8 @@ int main() {
     run<4, 8, 2, 2>(bf, "");
     run<4, 8, 1, 2>(bf, "");
     run_mfma<4, 1, 0>(bf);
+    run_mfma_swapped<4, 0, 0>(bf, true);
+    run_mfma_swapped<4, 1, 1>(bf, false);
   }
   {
     Bufs bf = make(1);

@@ -57,11 +57,11 @@ void paged_attention_decode(torch::Tensor out, torch::Tensor q,
   do {                                                                        \
     if (mfma_lds_bytes(GG, hd) > 65536)                                       \
       (void)hipFuncSetAttribute(                                              \
-          reinterpret_cast<const void*>(&paged_decode_mfma<1, 0>),            \
+          reinterpret_cast<const void*>(&paged_decode_mfma<1, 1>),            \
           hipFuncAttributeMaxDynamicSharedMemorySize,                         \
           mfma_lds_bytes(GG, hd));                                            \
   } while (0);                                                                \
-  paged_decode_mfma<1, 0><<<grid, kBlock, mfma_lds_bytes(GG, hd), stream>>>(  \
+  paged_decode_mfma<1, 1><<<grid, kBlock, mfma_lds_bytes(GG, hd), stream>>>(  \
       partial.data_ptr<float>(), ml.data_ptr<float>(), (short*)out.data_ptr(),\
       (const short*)q.data_ptr(), (const short*)kcache.data_ptr(),            \
       (const short*)vcache.data_ptr(), page_table.data_ptr<int32_t>(),        \

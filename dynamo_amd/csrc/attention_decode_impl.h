@@ -565,6 +565,266 @@ inline int mfma_lds_bytes(int G, int hd) {
   return 4 * G * (hd + 2) * 4 + G * hd * 2 + 4 * 5120 * 2 + 4 * 512 * 2;
 }
 
+// ---------------------------------------------------------------------------
+// SWAPPED-operand MFMA decode: S^T = mfma(K, Q) puts TOKENS on the MFMA M
+// rows and HEADS on the columns, so each lane's softmax state is ONE
+// head's (m, l) scalar and — crucially — the P -> PV-B-fragment
+// re-layout becomes 4 v_cvt_pk_bf16_f32 + 4 permlane swaps IN REGISTERS
+// (verified swap semantics: permlane32_swap(v0,v1) -> x = {lo: v0(l),
+// hi: v1(l-32)}, y = {lo: v0(l+32), hi: v1(l)}; permlane16_swap the same
+// within each 32-half). This deletes the per-tile P LDS roundtrip and its
+// wave-level waitcnt fence of paged_decode_mfma, and the O rescale is one
+// scalar multiply (no per-row alpha shuffles). PV computes
+// O^T = mfma(V^T, P^T); K loads, V^T staging and the cross-wave merge
+// keep the same shapes. Derivation of the swap network:
+//   after s1 = permlane32_swap(A0, B0):
+//     x1@lg{0,1} = A0(lg),   x1@lg{2,3} = B0(lg-2)
+//     y1@lg{0,1} = A0(lg+2), y1@lg{2,3} = B0(lg)
+//   after s2 = permlane16_swap(x1, y1):
+//     x2@lg = (correct tile)(2*(lg&1)),  y2@lg = (correct tile)(2*(lg&1)+1)
+// which is exactly B-frag words {tok 8lg+0..1} and {tok 8lg+4..5}; the
+// (A1, B1) pair gives {8lg+2..3} and {8lg+6..7}.
+template <int DEFER = 1, int PRIO = 1>
+__global__ __launch_bounds__(kBlock) void paged_decode_mfma_swapped(
+    float* __restrict__ partial, float* __restrict__ ml,
+    short* __restrict__ out, const short* __restrict__ q,
+    const short* __restrict__ kcache, const short* __restrict__ vcache,
+    const int32_t* __restrict__ page_table, const int32_t* __restrict__ ctx_lens,
+    float scale, int chunk, int G, int B, int Hkv, int C, int max_pages,
+    int log2_ps, int hd) {
+  typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8_t;
+  typedef __attribute__((ext_vector_type(2))) unsigned int uint2_t;
+  const int kSlab = chunk / 4;
+  const int b = blockIdx.x;
+  const int h = blockIdx.y;
+  const int c = blockIdx.z;
+  const int Hq = Hkv * G;
+  const int ctx = ctx_lens[b];
+  const int chunk_start = c * chunk;
+  const int lane = threadIdx.x & 63;
+  const int wid = threadIdx.x >> 6;
+  const int lr = lane & 15;
+  const int lg = lane >> 4;
+  const int ps = 1 << log2_ps;
+
+  extern __shared__ float lds[];
+  float* merge = lds;                                   // 4*G*(hd+2)
+  short* q_lds_s = reinterpret_cast<short*>(merge + 4 * G * (hd + 2));
+  short* v_lds = q_lds_s + G * hd + wid * 5120;         // per-wave 10KB
+
+  if (chunk_start >= ctx) {
+    if (C > 1) {
+      for (int i = threadIdx.x; i < G; i += kBlock) {
+        const int qh = h * G + i;
+        float* mlp = ml + (((int64_t)b * Hq + qh) * C + c) * 2;
+        mlp[0] = kNegInf; mlp[1] = 0.f;
+      }
+    }
+    return;
+  }
+
+  for (int i = threadIdx.x; i < G * hd; i += kBlock) {
+    const int g = i / hd;
+    q_lds_s[i] = q[((int64_t)b * Hq + h * G + g) * hd + i % hd];
+  }
+  __syncthreads();
+
+  // Q as the QK B-operand: same fragment registers as the A-frag variant
+  bf16x8_t q_frag[4];
+#pragma unroll
+  for (int kc = 0; kc < 4; kc++) {
+    short8 v{};
+    if (lr < G)
+      v = *reinterpret_cast<const short8*>(q_lds_s + lr * hd + kc * 32 + lg * 8);
+    q_frag[kc] = *reinterpret_cast<bf16x8_t*>(&v);
+  }
+
+  // per-lane softmax state for head `lr` (replicated across the 4 lg
+  // groups); acc[da][r] = O^T[dim da*16 + lg*4 + r][head lr]
+  float m_run = kNegInf, l_run = 0.f;
+  f32x4 acc[8];
+#pragma unroll
+  for (int d = 0; d < 8; d++) acc[d] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+  const int slab_start = chunk_start + wid * kSlab;
+  const int slab_end = min(slab_start + kSlab, ctx);
+  const int32_t* pt = page_table + (int64_t)b * max_pages;
+
+  for (int ti = 0; ti < kSlab / 32; ti++) {
+    const int t0 = slab_start + ti * 32;
+    const bool active = t0 < slab_end;
+    const int64_t pbase = active
+        ? (((int64_t)pt[t0 >> log2_ps] * Hkv + h) * ps) * hd : 0;
+    f32x4 sA{0.f, 0.f, 0.f, 0.f}, sB{0.f, 0.f, 0.f, 0.f};
+    if (active) {
+      const int tA = t0 + lr, tB = t0 + 16 + lr;
+      const short* krA = kcache + pbase + (int64_t)(tA & (ps - 1)) * hd;
+      const short* krB = kcache + pbase + (int64_t)(tB & (ps - 1)) * hd;
+      const bool vA = tA < slab_end, vB = tB < slab_end;
+      if constexpr (PRIO) __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int kc = 0; kc < 4; kc++) {
+        short8 ka = vA ? *reinterpret_cast<const short8*>(krA + kc * 32 + lg * 8)
+                       : short8{};
+        sA = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            *reinterpret_cast<bf16x8_t*>(&ka), q_frag[kc], sA, 0, 0, 0);
+      }
+      if (t0 + 16 < slab_end) {
+#pragma unroll
+        for (int kc = 0; kc < 4; kc++) {
+          short8 kb2 = vB ? *reinterpret_cast<const short8*>(krB + kc * 32 + lg * 8)
+                          : short8{};
+          sB = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              *reinterpret_cast<bf16x8_t*>(&kb2), q_frag[kc], sB, 0, 0, 0);
+        }
+      }
+      if constexpr (PRIO) __builtin_amdgcn_s_setprio(0);
+    }
+    // ---- stage V^T tile (identical to the A-variant) ----
+    if (active) {
+      typedef __attribute__((ext_vector_type(4))) short short4v;
+#pragma unroll
+      for (int u = 0; u < 2; u++) {
+        const int unit = lane + u * 64;
+        const int d8 = unit & 15;
+        const int tg = unit >> 4;
+        short8 rows[4];
+#pragma unroll
+        for (int j = 0; j < 4; j++) {
+          const int t = t0 + tg * 4 + j;
+          rows[j] = (t < slab_end)
+              ? *reinterpret_cast<const short8*>(
+                    vcache + pbase + (int64_t)(t & (ps - 1)) * hd + d8 * 8)
+              : short8{};
+        }
+#pragma unroll
+        for (int i = 0; i < 8; i++) {
+          const int d = d8 * 8 + i;
+          short4v pk = {rows[0][i], rows[1][i], rows[2][i], rows[3][i]};
+          *reinterpret_cast<short4v*>((char*)v_lds + d * 80 + tg * 8) = pk;
+        }
+      }
+    }
+
+    if (active) {
+      // ---- softmax: rows are TOKENS (lg*4+r, +16 for the B tile), the
+      // lane's column is head lr; token-reduction = 8-reg fold + 2 shfls
+      float pA[4], pB[4];
+      float mt = kNegInf;
+#pragma unroll
+      for (int r = 0; r < 4; r++) {
+        const int tokA = t0 + lg * 4 + r, tokB = t0 + 16 + lg * 4 + r;
+        pA[r] = (tokA < slab_end) ? sA[r] * scale : kNegInf;
+        pB[r] = (tokB < slab_end) ? sB[r] * scale : kNegInf;
+        mt = fmaxf(mt, fmaxf(pA[r], pB[r]));
+      }
+      mt = fmaxf(mt, __shfl_xor(mt, 16, WAVE_SIZE));
+      mt = fmaxf(mt, __shfl_xor(mt, 32, WAVE_SIZE));
+      const float thr = DEFER ? 8.0f : 0.0f;
+      if (mt > m_run + thr) {
+        const float corr = (m_run <= kNegInf * 0.5f) ? 0.f
+                                                     : __expf(m_run - mt);
+        l_run *= corr;
+#pragma unroll
+        for (int d = 0; d < 8; d++) acc[d] *= corr;
+        m_run = mt;
+      }
+      float psum = 0.f;
+#pragma unroll
+      for (int r = 0; r < 4; r++) {
+        pA[r] = (pA[r] > kNegInf * 0.5f) ? __expf(pA[r] - m_run) : 0.f;
+        pB[r] = (pB[r] > kNegInf * 0.5f) ? __expf(pB[r] - m_run) : 0.f;
+        psum += pA[r] + pB[r];
+      }
+      psum += __shfl_xor(psum, 16, WAVE_SIZE);
+      psum += __shfl_xor(psum, 32, WAVE_SIZE);
+      l_run += psum;
+
+      // ---- P^T -> PV B-fragment in registers (4 cvt_pk + 4 swaps) ----
+      auto cvtpk = [](float a, float bb) {
+        unsigned int r;
+        asm("v_cvt_pk_bf16_f32 %0, %1, %2" : "=v"(r) : "v"(a), "v"(bb));
+        return r;
+      };
+      const unsigned int A0 = cvtpk(pA[0], pA[1]), A1 = cvtpk(pA[2], pA[3]);
+      const unsigned int B0 = cvtpk(pB[0], pB[1]), B1 = cvtpk(pB[2], pB[3]);
+      uint2_t s1 = __builtin_amdgcn_permlane32_swap(A0, B0, false, false);
+      uint2_t s2 = __builtin_amdgcn_permlane32_swap(A1, B1, false, false);
+      uint2_t f02 = __builtin_amdgcn_permlane16_swap(s1.x, s1.y, false, false);
+      uint2_t f13 = __builtin_amdgcn_permlane16_swap(s2.x, s2.y, false, false);
+      unsigned int w[4] = {f02.x, f13.x, f02.y, f13.y};
+      bf16x8_t p_frag = *reinterpret_cast<bf16x8_t*>(w);
+
+      // V staging for THIS tile must be visible (same wave, in-order LDS)
+      __builtin_amdgcn_wave_barrier();
+      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+
+      // ---- PV: O^T[dim][head] += V^T[dim][tok] P^T[tok][head] ----
+      if constexpr (PRIO) __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int da = 0; da < 8; da++) {
+        short8 va_s = *reinterpret_cast<const short8*>(
+            (const char*)v_lds + (da * 16 + lr) * 80 + lg * 16);
+        acc[da] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            *reinterpret_cast<bf16x8_t*>(&va_s), p_frag, acc[da], 0, 0, 0);
+      }
+      if constexpr (PRIO) __builtin_amdgcn_s_setprio(0);
+    }
+    __builtin_amdgcn_wave_barrier();
+  }
+
+  // ---- cross-wave merge: acc[da][r] is O^T[dim da*16+lg*4+r][head lr]
+  __syncthreads();
+  float* my = merge + wid * G * (hd + 2);
+  if (lr < G) {
+#pragma unroll
+    for (int da = 0; da < 8; da++)
+#pragma unroll
+      for (int r = 0; r < 4; r++)
+        my[lr * (hd + 2) + da * 16 + lg * 4 + r] = acc[da][r];
+    if (lg == 0) {
+      my[lr * (hd + 2) + hd] = m_run;
+      my[lr * (hd + 2) + hd + 1] = l_run;
+    }
+  }
+  __syncthreads();
+
+  for (int i = threadIdx.x; i < G * hd; i += kBlock) {
+    const int g = i / hd;
+    const int d = i % hd;
+    float mw[4], lw[4];
+    float mstar = kNegInf;
+#pragma unroll
+    for (int w2 = 0; w2 < 4; w2++) {
+      mw[w2] = merge[w2 * G * (hd + 2) + g * (hd + 2) + hd];
+      lw[w2] = merge[w2 * G * (hd + 2) + g * (hd + 2) + hd + 1];
+      mstar = fmaxf(mstar, mw[w2]);
+    }
+    float lsum = 0.f, asum = 0.f;
+#pragma unroll
+    for (int w2 = 0; w2 < 4; w2++) {
+      const float corr = (lw[w2] > 0.f) ? __expf(mw[w2] - mstar) : 0.f;
+      lsum += lw[w2] * corr;
+      asum += merge[w2 * G * (hd + 2) + g * (hd + 2) + d] * corr;
+    }
+    const int qh = h * G + g;
+    if (C == 1) {
+      out[((int64_t)b * Hq + qh) * hd + d] =
+          f32_to_bf16(lsum > 0.f ? asum / lsum : 0.f);
+    } else {
+      partial[(((int64_t)b * Hq + qh) * C + c) * hd + d] = asum;
+      if (d == 0) {
+        float* mlp = ml + (((int64_t)b * Hq + qh) * C + c) * 2;
+        mlp[0] = mstar; mlp[1] = lsum;
+      }
+    }
+  }
+}
+
+inline int mfma_swapped_lds_bytes(int G, int hd) {
+  return 4 * G * (hd + 2) * 4 + G * hd * 2 + 4 * 5120 * 2;  // no P tile
+}
+
 // Phase 2: merge chunk partials. grid (B, Hq), block = 128.
 __global__ inline void paged_decode_phase2(short* __restrict__ out,
                                            const float* __restrict__ partial,
