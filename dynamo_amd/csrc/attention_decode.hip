@@ -53,20 +53,44 @@ void paged_attention_decode(torch::Tensor out, torch::Tensor q,
     const char* e = getenv("DYNAMO_DECODE_MFMA");
     return e == nullptr || e[0] != '0';  // default ON (sweep-verified)
   }();
+  // swapped-operand variant (tokens on MFMA M, in-register P exchange via
+  // permlane swaps): sweep-verified bit-match + faster at every measured
+  // (G, chunk); DYNAMO_DECODE_SWAPPED=0 falls back to the A-variant
+  static const bool use_swapped = [] {
+    const char* e = getenv("DYNAMO_DECODE_SWAPPED");
+    return e == nullptr || e[0] != '0';
+  }();
 #define LAUNCH_MFMA(GG)                                                       \
   do {                                                                        \
+    if (use_swapped) {                                                        \
+      if (mfma_swapped_lds_bytes(GG, hd) > 65536)                             \
+        (void)hipFuncSetAttribute(                                            \
+            reinterpret_cast<const void*>(                                    \
+                &paged_decode_mfma_swapped<1, 1, 1>),                         \
+            hipFuncAttributeMaxDynamicSharedMemorySize,                       \
+            mfma_swapped_lds_bytes(GG, hd));                                  \
+      paged_decode_mfma_swapped<1, 1, 1>                                      \
+          <<<grid, kBlock, mfma_swapped_lds_bytes(GG, hd), stream>>>(         \
+          partial.data_ptr<float>(), ml.data_ptr<float>(),                    \
+          (short*)out.data_ptr(), (const short*)q.data_ptr(),                 \
+          (const short*)kcache.data_ptr(),                                    \
+          (const short*)vcache.data_ptr(), page_table.data_ptr<int32_t>(),    \
+          ctx_lens.data_ptr<int32_t>(), (float)scale, chunk, GG, B, Hkv, C,    \
+          max_pages, log2_ps, hd);                                            \
+      break;                                                                  \
+    }                                                                         \
     if (mfma_lds_bytes(GG, hd) > 65536)                                       \
       (void)hipFuncSetAttribute(                                              \
           reinterpret_cast<const void*>(&paged_decode_mfma<1, 1>),            \
           hipFuncAttributeMaxDynamicSharedMemorySize,                         \
           mfma_lds_bytes(GG, hd));                                            \
-  } while (0);                                                                \
-  paged_decode_mfma<1, 1><<<grid, kBlock, mfma_lds_bytes(GG, hd), stream>>>(  \
+    paged_decode_mfma<1, 1><<<grid, kBlock, mfma_lds_bytes(GG, hd), stream>>>(\
       partial.data_ptr<float>(), ml.data_ptr<float>(), (short*)out.data_ptr(),\
       (const short*)q.data_ptr(), (const short*)kcache.data_ptr(),            \
       (const short*)vcache.data_ptr(), page_table.data_ptr<int32_t>(),        \
       ctx_lens.data_ptr<int32_t>(), (float)scale, chunk, GG, B, Hkv, C,        \
-      max_pages, log2_ps, hd)
+      max_pages, log2_ps, hd);                                                \
+  } while (0)
   const bool mfma_ok = use_mfma && ps % 32 == 0 && hd == 128;
   switch (G) {  // combos picked by benchmarks/decode_sweep on MI355X
     case 1: LAUNCH_G(1, 8, 1, 2); break;

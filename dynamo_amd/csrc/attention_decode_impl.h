@@ -584,7 +584,13 @@ inline int mfma_lds_bytes(int G, int hd) {
 //     x2@lg = (correct tile)(2*(lg&1)),  y2@lg = (correct tile)(2*(lg&1)+1)
 // which is exactly B-frag words {tok 8lg+0..1} and {tok 8lg+4..5}; the
 // (A1, B1) pair gives {8lg+2..3} and {8lg+6..7}.
-template <int DEFER = 1, int PRIO = 1>
+// KPF: cross-tile register prefetch (T14 async-stage): K fragments and V
+// staging rows for tile t+1 are LOADED during tile t's softmax+PV, so the
+// HBM latency of the next tile hides under this tile's compute. The
+// simpler swapped-softmax leaves ~150 spare VGPRs (101 base measured), so
+// the +64 VGPR double-buffer costs no occupancy (LDS-bound at 2
+// blocks/CU).
+template <int DEFER = 1, int PRIO = 1, int KPF = 0>
 __global__ __launch_bounds__(kBlock) void paged_decode_mfma_swapped(
     float* __restrict__ partial, float* __restrict__ ml,
     short* __restrict__ out, const short* __restrict__ q,
@@ -650,59 +656,102 @@ __global__ __launch_bounds__(kBlock) void paged_decode_mfma_swapped(
   const int slab_end = min(slab_start + kSlab, ctx);
   const int32_t* pt = page_table + (int64_t)b * max_pages;
 
+  // prefetchable loads (KPF): K fragments (8x short8) + V staging rows
+  // (8x short8) per tile, double-buffered across iterations
+  typedef __attribute__((ext_vector_type(4))) short short4v;
+  auto load_k = [&](int t0_, short8 (&kf)[8]) {
+    if (t0_ >= slab_end) return;
+    const int64_t pb = (((int64_t)pt[t0_ >> log2_ps] * Hkv + h) * ps) * hd;
+    const int tA = t0_ + lr, tB = t0_ + 16 + lr;
+    const short* krA = kcache + pb + (int64_t)(tA & (ps - 1)) * hd;
+    const short* krB = kcache + pb + (int64_t)(tB & (ps - 1)) * hd;
+    const bool vA = tA < slab_end, vB = tB < slab_end;
+#pragma unroll
+    for (int kc = 0; kc < 4; kc++) {
+      kf[kc] = vA ? *reinterpret_cast<const short8*>(krA + kc * 32 + lg * 8)
+                  : short8{};
+      kf[4 + kc] = vB
+          ? *reinterpret_cast<const short8*>(krB + kc * 32 + lg * 8)
+          : short8{};
+    }
+  };
+  auto load_v = [&](int t0_, short8 (&vf)[8]) {
+    if (t0_ >= slab_end) return;
+    const int64_t pb = (((int64_t)pt[t0_ >> log2_ps] * Hkv + h) * ps) * hd;
+#pragma unroll
+    for (int u = 0; u < 2; u++) {
+      const int unit = lane + u * 64;
+      const int d8 = unit & 15;
+      const int tg = unit >> 4;
+#pragma unroll
+      for (int j = 0; j < 4; j++) {
+        const int t = t0_ + tg * 4 + j;
+        vf[u * 4 + j] = (t < slab_end)
+            ? *reinterpret_cast<const short8*>(
+                  vcache + pb + (int64_t)(t & (ps - 1)) * hd + d8 * 8)
+            : short8{};
+      }
+    }
+  };
+  auto store_v = [&](const short8 (&vf)[8]) {
+#pragma unroll
+    for (int u = 0; u < 2; u++) {
+      const int unit = lane + u * 64;
+      const int d8 = unit & 15;
+      const int tg = unit >> 4;
+#pragma unroll
+      for (int i = 0; i < 8; i++) {
+        const int d = d8 * 8 + i;
+        short4v pk = {vf[u * 4 + 0][i], vf[u * 4 + 1][i],
+                      vf[u * 4 + 2][i], vf[u * 4 + 3][i]};
+        *reinterpret_cast<short4v*>((char*)v_lds + d * 80 + tg * 8) = pk;
+      }
+    }
+  };
+
+  short8 kfr[2][8], vfr[2][8];
+  if constexpr (KPF) {
+    load_k(slab_start, kfr[0]);
+    load_v(slab_start, vfr[0]);
+  }
+
   for (int ti = 0; ti < kSlab / 32; ti++) {
     const int t0 = slab_start + ti * 32;
     const bool active = t0 < slab_end;
-    const int64_t pbase = active
-        ? (((int64_t)pt[t0 >> log2_ps] * Hkv + h) * ps) * hd : 0;
+    const int cur = ti & 1;
     f32x4 sA{0.f, 0.f, 0.f, 0.f}, sB{0.f, 0.f, 0.f, 0.f};
+    if constexpr (!KPF) {
+      if (active) {
+        load_k(t0, kfr[0]);
+        load_v(t0, vfr[0]);
+      }
+    }
+    const short8(&kf)[8] = KPF ? kfr[cur] : kfr[0];
+    const short8(&vf)[8] = KPF ? vfr[cur] : vfr[0];
     if (active) {
-      const int tA = t0 + lr, tB = t0 + 16 + lr;
-      const short* krA = kcache + pbase + (int64_t)(tA & (ps - 1)) * hd;
-      const short* krB = kcache + pbase + (int64_t)(tB & (ps - 1)) * hd;
-      const bool vA = tA < slab_end, vB = tB < slab_end;
       if constexpr (PRIO) __builtin_amdgcn_s_setprio(1);
 #pragma unroll
       for (int kc = 0; kc < 4; kc++) {
-        short8 ka = vA ? *reinterpret_cast<const short8*>(krA + kc * 32 + lg * 8)
-                       : short8{};
+        short8 ka = kf[kc];
         sA = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
             *reinterpret_cast<bf16x8_t*>(&ka), q_frag[kc], sA, 0, 0, 0);
       }
       if (t0 + 16 < slab_end) {
 #pragma unroll
         for (int kc = 0; kc < 4; kc++) {
-          short8 kb2 = vB ? *reinterpret_cast<const short8*>(krB + kc * 32 + lg * 8)
-                          : short8{};
+          short8 kb2 = kf[4 + kc];
           sB = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               *reinterpret_cast<bf16x8_t*>(&kb2), q_frag[kc], sB, 0, 0, 0);
         }
       }
       if constexpr (PRIO) __builtin_amdgcn_s_setprio(0);
-    }
-    // ---- stage V^T tile (identical to the A-variant) ----
-    if (active) {
-      typedef __attribute__((ext_vector_type(4))) short short4v;
-#pragma unroll
-      for (int u = 0; u < 2; u++) {
-        const int unit = lane + u * 64;
-        const int d8 = unit & 15;
-        const int tg = unit >> 4;
-        short8 rows[4];
-#pragma unroll
-        for (int j = 0; j < 4; j++) {
-          const int t = t0 + tg * 4 + j;
-          rows[j] = (t < slab_end)
-              ? *reinterpret_cast<const short8*>(
-                    vcache + pbase + (int64_t)(t & (ps - 1)) * hd + d8 * 8)
-              : short8{};
-        }
-#pragma unroll
-        for (int i = 0; i < 8; i++) {
-          const int d = d8 * 8 + i;
-          short4v pk = {rows[0][i], rows[1][i], rows[2][i], rows[3][i]};
-          *reinterpret_cast<short4v*>((char*)v_lds + d * 80 + tg * 8) = pk;
-        }
+      // stage THIS tile's V rows (loaded last iteration under KPF)
+      store_v(vf);
+      if constexpr (KPF) {
+        // issue next tile's K/V global loads now: their latency hides
+        // under this tile's softmax + PV
+        load_k(t0 + 32, kfr[cur ^ 1]);
+        load_v(t0 + 32, vfr[cur ^ 1]);
       }
     }
 
