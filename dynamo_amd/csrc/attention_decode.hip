@@ -66,10 +66,10 @@ void paged_attention_decode(torch::Tensor out, torch::Tensor q,
       if (mfma_swapped_lds_bytes(GG, hd) > 65536)                             \
         (void)hipFuncSetAttribute(                                            \
             reinterpret_cast<const void*>(                                    \
-                &paged_decode_mfma_swapped<1, 1, 1>),                         \
+                &paged_decode_mfma_swapped<1, 1, 0>),                         \
             hipFuncAttributeMaxDynamicSharedMemorySize,                       \
             mfma_swapped_lds_bytes(GG, hd));                                  \
-      paged_decode_mfma_swapped<1, 1, 1>                                      \
+      paged_decode_mfma_swapped<1, 1, 0>                                      \
           <<<grid, kBlock, mfma_swapped_lds_bytes(GG, hd), stream>>>(         \
           partial.data_ptr<float>(), ml.data_ptr<float>(),                    \
           (short*)out.data_ptr(), (const short*)q.data_ptr(),                 \

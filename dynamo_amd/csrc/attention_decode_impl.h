@@ -584,12 +584,11 @@ inline int mfma_lds_bytes(int G, int hd) {
 //     x2@lg = (correct tile)(2*(lg&1)),  y2@lg = (correct tile)(2*(lg&1)+1)
 // which is exactly B-frag words {tok 8lg+0..1} and {tok 8lg+4..5}; the
 // (A1, B1) pair gives {8lg+2..3} and {8lg+6..7}.
-// KPF: cross-tile register prefetch (T14 async-stage): K fragments and V
-// staging rows for tile t+1 are LOADED during tile t's softmax+PV, so the
-// HBM latency of the next tile hides under this tile's compute. The
-// simpler swapped-softmax leaves ~150 spare VGPRs (101 base measured), so
-// the +64 VGPR double-buffer costs no occupancy (LDS-bound at 2
-// blocks/CU).
+// KPF: cross-tile register prefetch (T14 async-stage) — MEASURED 2.5x
+// WORSE (1.4 vs 3.5 TB/s, k2048 sweep): the `kfr[cur]` runtime-indexed
+// register double-buffer goes to scratch (guide rule #20), exactly the
+// round-1 K-prefetch regression repeated. Kept compiled for the sweep's
+// record; production uses KPF=0.
 template <int DEFER = 1, int PRIO = 1, int KPF = 0>
 __global__ __launch_bounds__(kBlock) void paged_decode_mfma_swapped(
     float* __restrict__ partial, float* __restrict__ ml,

@@ -207,7 +207,7 @@ class LLMEngine:
                 else:
                     self._finish(req, reason)
             elif (self.cfg.kv_events or self.cfg.enable_prefix_caching) and \
-                    req.prompt_embeds is None and \
+                    not req.has_embeds and \
                     req.num_computed % self.cfg.page_size == 0:
                 req.kv.commit_full_pages(req.all_tokens, req.num_computed)
             outputs.append(StepOutput(req.req_id, int(tok), finished, reason,
@@ -289,7 +289,7 @@ class LLMEngine:
         if self.cfg.kv_events or self.cfg.enable_prefix_caching:
             for ss in sched.seqs:
                 r = ss.req
-                if r.kv is not None and r.prompt_embeds is None:
+                if r.kv is not None and not r.has_embeds:
                     r.kv.commit_full_pages(r.all_tokens, r.num_computed)
         self.kv_events.extend(self.alloc.drain_events())
 
@@ -309,7 +309,7 @@ class LLMEngine:
 
     def _finish(self, req: Request, reason: str):
         # commit full pages before release so prefix cache retains them
-        if req.kv is not None and req.prompt_embeds is None:
+        if req.kv is not None and not req.has_embeds:
             req.kv.commit_full_pages(req.all_tokens, req.num_computed)
         self.scheduler.finish(req, reason)
 

@@ -105,6 +105,16 @@ class ModelRunner:
                     base = nd + qpos
                     embeds_rows.extend(range(base, base + pe))
                     embeds_parts.append(r.prompt_embeds[nc:nc + pe])
+            if r.embed_spans:
+                # multimodal: sparse spans (image placeholder rows) that
+                # overlap this prefill chunk take encoder embeddings
+                for off, emb in r.embed_spans:
+                    lo = max(nc, off)
+                    hi = min(nc + n, off + emb.shape[0])
+                    if lo < hi:
+                        base = nd + qpos + (lo - nc)
+                        embeds_rows.extend(range(base, base + hi - lo))
+                        embeds_parts.append(emb[lo - off:hi - off])
             pf_tables.append(r.kv.pages)
             q_start.append(qpos)
             q_len.append(n)

@@ -69,6 +69,15 @@ class Request:
         # prefix caching is disabled for these requests (placeholder ids
         # must not produce hash hits)
         self.prompt_embeds = None
+        # multimodal: sparse embedding spans [(offset, tensor), ...] —
+        # image-placeholder rows of the prompt get encoder output instead
+        # of token-table embeddings (reference: preprocessor.rs:2248 media
+        # + encode-worker embeddings spliced into the P/D request)
+        self.embed_spans = None
+
+    @property
+    def has_embeds(self) -> bool:
+        return self.prompt_embeds is not None or bool(self.embed_spans)
 
     @property
     def all_tokens(self) -> List[int]:
@@ -205,7 +214,7 @@ class Scheduler:
             if req.kv is None:
                 req.kv = SequenceKV(self.alloc, self.cfg.block_salt)
                 if (self.cfg.enable_prefix_caching and not req.prefill_result
-                        and req.prompt_embeds is None):
+                        and not req.has_embeds):
                     req.num_computed = req.kv.match_prefix(req.all_tokens)
             remaining = req.total_len - req.num_computed
             n = min(remaining, budget)

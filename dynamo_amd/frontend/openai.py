@@ -47,7 +47,10 @@ class CompletionRequest(BaseModel):
 
 class ChatMessage(BaseModel):
     role: str
-    content: str
+    # plain text, or OpenAI content parts:
+    # [{"type": "text", "text": ...},
+    #  {"type": "image_url", "image_url": {"url": "data:image/png;base64,..."}}]
+    content: Union[str, List[dict]] = ""
 
 
 class EmbeddingRequest(BaseModel):
@@ -146,19 +149,19 @@ def build_app(manager: ModelManager) -> FastAPI:
             for name, e in manager.models.items()]}
 
     async def _run(entry, token_ids, req, rid,
-                   session_id=None) -> AsyncIterator[dict]:
+                   session_id=None, extra=None) -> AsyncIterator[dict]:
         sampling = {"temperature": req.temperature, "top_p": req.top_p,
-                    "top_k": req.top_k, "seed": req.seed,
+                    "top_k": getattr(req, "top_k", 0),
+                    "seed": getattr(req, "seed", 0),
                     "logprobs": getattr(req, "logprobs", None) or 0}
         eos = getattr(entry.tokenizer, "eos_id", None)
         stop = {"max_tokens": req.max_tokens,
-                "ignore_eos": req.ignore_eos,
+                "ignore_eos": getattr(req, "ignore_eos", False),
                 "stop_token_ids": [eos] if eos is not None else []}
         from dynamo_amd.router.kv_router import AllWorkersBusy
-        extra = None
         pe = getattr(req, "prompt_embeds", None)
         if pe:
-            extra = {"prompt_embeds": pe}
+            extra = dict(extra or {}, prompt_embeds=pe)
         try:
             async for chunk in manager.generate_tokens(
                     entry, token_ids, sampling, stop, request_id=rid,
@@ -627,14 +630,73 @@ def build_app(manager: ModelManager) -> FastAPI:
                 "usage": {"prompt_tokens": total_tokens,
                           "total_tokens": total_tokens}}
 
+    _IMG_SENT = "\x00img\x00"
+
+    def _image_payload(part) -> dict:
+        """Media fetch/decode (reference preprocessor.rs:2248): data: URIs
+        only — there is no egress for http(s) media in this environment."""
+        url = part.get("image_url")
+        if isinstance(url, dict):
+            url = url.get("url", "")
+        if not isinstance(url, str) or not url.startswith("data:"):
+            raise HTTPException(
+                400, "only data: image URLs are supported (no egress)")
+        head, _, b64 = url.partition(",")
+        if "tensor" in head:   # data:application/x-tensor;shape=3x32x32;base64,
+            shape = [int(x) for x in
+                     head.split("shape=")[1].split(";")[0].split("x")]
+            return {"b64_pixels": b64, "shape": shape, "dtype": "float32"}
+        return {"b64_image": b64}
+
+    async def _mm_prepare(entry, req):
+        """Multimodal chat: flatten content parts, route images to an
+        encode worker, splice embeddings over placeholder tokens.
+        Returns (token_ids, extra) or None for text-only requests."""
+        if not any(isinstance(m.content, list) for m in req.messages):
+            return None
+        msgs, images = [], []
+        for m in req.messages:
+            if isinstance(m.content, str):
+                msgs.append({"role": m.role, "content": m.content})
+                continue
+            text = []
+            for p in m.content:
+                if p.get("type") == "image_url":
+                    images.append(_image_payload(p))
+                    text.append(_IMG_SENT)
+                else:
+                    text.append(p.get("text", ""))
+            msgs.append({"role": m.role, "content": "".join(text)})
+        prompt = entry.templater.render(msgs)
+        if not images:
+            return entry.tokenizer.encode(prompt), None
+        embs = await manager.encode_images(entry.name, images)
+        segs = prompt.split(_IMG_SENT)
+        token_ids: List[int] = []
+        mm = []
+        for i, seg in enumerate(segs):
+            token_ids.extend(entry.tokenizer.encode(seg))
+            if i < len(segs) - 1:
+                emb = embs[i]
+                mm.append({"offset": len(token_ids), **emb})
+                token_ids.extend([0] * emb["shape"][0])
+        return token_ids, {"mm_embeds": mm}
+
     @app.post("/v1/chat/completions")
     async def chat(req: ChatRequest, raw: Request):
         entry = _entry_or_404(req.model)
         REQS.labels(entry.name, "chat").inc()
-        prompt = entry.templater.render([m.model_dump() for m in req.messages])
-        token_ids = (await asyncio.to_thread(entry.tokenizer.encode, prompt)
-                     if len(prompt) > 4096 else
-                     entry.tokenizer.encode(prompt))
+        mm_extra = None
+        mm = await _mm_prepare(entry, req)
+        if mm is not None:
+            token_ids, mm_extra = mm
+        else:
+            prompt = entry.templater.render(
+                [m.model_dump() for m in req.messages])
+            token_ids = (await asyncio.to_thread(entry.tokenizer.encode,
+                                                 prompt)
+                         if len(prompt) > 4096 else
+                         entry.tokenizer.encode(prompt))
         rid = f"chatcmpl-{uuid.uuid4().hex[:24]}"
         t0 = time.time()
 
@@ -664,7 +726,7 @@ def build_app(manager: ModelManager) -> FastAPI:
 
                 try:
                     finish = None
-                    async for chunk in _run(entry, token_ids, req, rid, _session_of(req, raw)):
+                    async for chunk in _run(entry, token_ids, req, rid, _session_of(req, raw), extra=mm_extra):
                         if await raw.is_disconnected():
                             break
                         role = first
@@ -701,7 +763,7 @@ def build_app(manager: ModelManager) -> FastAPI:
         stops = _stop_list(req)
         text_cut = None
         acc = ""
-        async for chunk in _run(entry, token_ids, req, rid, _session_of(req, raw)):
+        async for chunk in _run(entry, token_ids, req, rid, _session_of(req, raw), extra=mm_extra):
             prev = len(produced)
             produced.extend(chunk.get("token_ids", []))
             finish = chunk.get("finish_reason") or finish

@@ -147,6 +147,34 @@ class ModelManager:
                 e = self.models.pop(name)
                 await e.router.stop()
 
+    # -- multimodal: encode workers (E/PD disaggregation) ---------------
+    def encoder_instances(self, model: str):
+        """Live encode workers feeding `model` (reference parity:
+        EncoderRouter, lib/llm/src/kv_router/encoder_router.rs)."""
+        out = []
+        for inst in self.runtime.discovery.list(self.namespace, "encoder"):
+            m = inst.metadata or {}
+            if m.get("worker_type") == "encoder" and (
+                    not m.get("model") or m["model"] == model):
+                out.append(inst)
+        return out
+
+    async def encode_images(self, model: str, images: list) -> list:
+        """Round-robin an encode worker; returns embedding specs
+        [{"b64","shape","dtype"}]. Raises if no encoder is registered."""
+        insts = self.encoder_instances(model)
+        if not insts:
+            raise RuntimeError(f"no encode workers for model {model!r}")
+        self._enc_rr = getattr(self, "_enc_rr", 0) + 1
+        inst = insts[self._enc_rr % len(insts)]
+        r = await self.runtime.client.call(
+            inst.address, "encoder.encode", {"images": images})
+        return r["embeddings"]
+
+    def encoder_tokens_per_image(self, model: str) -> int:
+        insts = self.encoder_instances(model)
+        return int(insts[0].metadata.get("tokens_per_image", 0)) if insts else 0
+
     # ------------------------------------------------------------------
     def get(self, model: str) -> ModelEntry:
         if model in self.models:

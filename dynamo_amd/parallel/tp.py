@@ -81,6 +81,8 @@ def apply_command(engine: LLMEngine, cmd: dict, tp_rank: int = 0):
             # rank0's arrival stamp: queue-policy tie-breaks must order
             # identically on every rank
             req.arrival = r["arrival"]
+        if r.get("embed_spans"):
+            req.embed_spans = r["embed_spans"]
     for rid in cmd.get("aborts", []):
         engine.abort(rid)
     for rid in cmd.get("release", []):
@@ -172,6 +174,9 @@ class TPEngineGroup:
                         "token_ids": list(r.prompt_tokens),
                         "sampling": r.sampling.__dict__.copy(),
                         "hold_kv": bool(getattr(r, "hold_kv", False)),
+                        # (offset, tensor) pairs pickle through the gloo
+                        # object broadcast; identical on every rank
+                        "embed_spans": r.embed_spans,
                         "arrival": r.arrival}
                        for r in self._pending_new],
                "aborts": self._pending_aborts, "step": step}

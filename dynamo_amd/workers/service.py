@@ -300,17 +300,22 @@ class WorkerService:
         q: asyncio.Queue = asyncio.Queue()
         self.queues[req_id] = q
         completed = False
+        def _decode_tensor(spec):
+            import base64
+            import numpy as np
+            import torch
+            arr = np.frombuffer(base64.b64decode(spec["b64"]),
+                                dtype=np.dtype(spec.get("dtype", "float16")))
+            return torch.from_numpy(arr.reshape(spec["shape"]).copy()).float()
+
         pe = payload.get("prompt_embeds")
         prompt_embeds = None
         if pe is not None:
             # PreprocessedRequest prompt_embeds (b64) parity
-            import base64
-            import numpy as np
-            import torch
-            arr = np.frombuffer(base64.b64decode(pe["b64"]),
-                                dtype=np.dtype(pe.get("dtype", "float16")))
-            prompt_embeds = torch.from_numpy(
-                arr.reshape(pe["shape"]).copy()).float()
+            prompt_embeds = _decode_tensor(pe)
+        # multimodal embedding spans: [{"offset", "b64", "shape", "dtype"}]
+        mm = payload.get("mm_embeds") or []
+        embed_spans = [(int(s["offset"]), _decode_tensor(s)) for s in mm]
         try:
             pr = payload.get("prefill_result")
             async with self._engine_lock:
@@ -322,6 +327,8 @@ class WorkerService:
                 else:
                     req = self.engine.add_request(
                         req_id, tokens, sp, prompt_embeds=prompt_embeds)
+                    if embed_spans:
+                        req.embed_spans = embed_spans
                     if is_prefill_role:
                         req.hold_kv = True
             self._work.set()
