@@ -54,7 +54,7 @@ class VisionEncoder(torch.nn.Module):
         g = torch.Generator(device="cpu").manual_seed(seed)
 
         def w(*shape, std=0.02):
-            t = torch.empty(*shape, generator=g).normal_(0, std)
+            t = torch.empty(shape).normal_(0, std, generator=g)
             return torch.nn.Parameter(t.to(self.device, dtype),
                                       requires_grad=False)
 
