@@ -12,6 +12,7 @@
 //           spill to scratch).
 #pragma once
 #include "common.h"
+#include <cstdlib>
 
 namespace decode_attn {
 
@@ -23,6 +24,12 @@ constexpr int kBlock = 256;        // 4 waves
 // ctx 8192), but the grid is B*Hkv*C blocks, so short contexts need small
 // chunks to fill 256 CUs. Python mirrors this via the binding.
 inline int decode_chunk_tokens(int max_ctx) {
+#ifndef __HIP_DEVICE_COMPILE__
+  if (const char* e = getenv("DYNAMO_DECODE_CHUNK")) {
+    const int v = atoi(e);
+    if (v >= 128 && v % 128 == 0) return v;   // A/B sweeps on real serving
+  }
+#endif
   if (max_ctx >= 8192) return 2048;
   if (max_ctx >= 2048) return 1024;
   return 512;

@@ -59,6 +59,12 @@ def build_parser():
     p.add_argument("--status-port", type=int, default=0,
                    help="serve a JSON system-status HTTP endpoint on this "
                         "port (reference DYN_SYSTEM_PORT parity; 0 = off)")
+    p.add_argument("--moe-ep", action="store_true",
+                   help="MoE models at tp>1: expert parallelism over the "
+                        "TP group (default: auto-on for MoE at tp>1; "
+                        "reference flag moe_ep_size)")
+    p.add_argument("--no-moe-ep", action="store_true",
+                   help="force MoE tensor-parallel sharding instead of EP")
     p.add_argument("--tp-size", type=int, default=1,
                    help="tensor parallelism: launch via torchrun "
                         "--nproc-per-node N, one rank per GPU over RCCL; "
@@ -71,6 +77,11 @@ def build_parser():
 
 def make_engine_from_args(args, tp=None) -> LLMEngine:
     mc = resolve_model_config(args.model)
+    if mc.num_experts and tp is not None and tp.size > 1:
+        # expert parallelism over the TP group (config #5 deployment);
+        # --no-moe-ep keeps MoE-TP sharding instead
+        import dataclasses
+        mc = dataclasses.replace(mc, moe_ep=not args.no_moe_ep)
     if args.mock:
         from dynamo_amd.mocker import make_mock_engine
         return make_mock_engine(

@@ -178,3 +178,25 @@ def test_disagg_tp1_processes_through_framework():
             wp.stop()
             wd.stop()
         assert outs == _baseline_tp1(), f"disagg-1p1d {outs}"
+
+
+@pytest.mark.timeout(420)
+def test_moe_ep_tp2_worker_through_framework():
+    """config #5 shape: a Mixtral TP2 worker (EP auto-on: experts sharded,
+    all-to-all-free partial-sum design) served through the framework;
+    output == single-rank engine."""
+    global MODEL
+    saved = MODEL
+    MODEL = "tiny-mixtral"
+    try:
+        with tempfile.TemporaryDirectory() as d:
+            disc = f"file:{d}/disc"
+            w = torchrun_worker(2, disc)
+            w.start()
+            try:
+                outs = run(_generate_via_stack(disc))
+            finally:
+                w.stop()
+            assert outs == _baseline_tp1(), f"moe-ep-serve {outs}"
+    finally:
+        MODEL = saved
