@@ -1,3 +1,5 @@
+from .global_planner import (GlobalPlanner, GlobalPlannerConnector,
+                             GlobalPlannerService, PoolBudgetPolicy)
 from .planner import (ARIMAPredictor, Connector, ConstantPredictor,
                       CorrectionFactors, InterpolatedPerfModel, LoadPlanner,
                       MovingAveragePredictor, PerfModel, PlannerService,
@@ -5,7 +7,8 @@ from .planner import (ARIMAPredictor, Connector, ConstantPredictor,
                       SLATargets, ThroughputPlanner, TrendPredictor,
                       VirtualConnector)
 
-__all__ = ["ARIMAPredictor", "Connector", "ConstantPredictor",
+__all__ = ["ARIMAPredictor", "GlobalPlanner", "GlobalPlannerConnector",
+           "GlobalPlannerService", "PoolBudgetPolicy", "Connector", "ConstantPredictor",
            "CorrectionFactors", "InterpolatedPerfModel", "LoadPlanner",
            "MovingAveragePredictor", "PerfModel", "PlannerService",
            "PoolObservation", "PoolPolicy", "ScalingState", "SLAPlanner",

@@ -235,3 +235,74 @@ def test_sla_planner_closed_loop_holds_slo():
 def _interp_for_test(c, model):
     from dynamo_amd.planner.planner import _interp
     return _interp(c, model.conc, model.itl)
+
+
+# ---- global planner: centralized scale execution under a GPU budget ----
+
+def test_global_planner_budget_arbitration():
+    from dynamo_amd.planner import (GlobalPlanner, PoolBudgetPolicy,
+                                    VirtualConnector)
+
+    async def main():
+        ex_a, ex_b = VirtualConnector(), VirtualConnector()
+        gp = GlobalPlanner(
+            total_budget=8,
+            policies=[PoolBudgetPolicy("a", weight=2.0, min_replicas=1),
+                      PoolBudgetPolicy("b", weight=1.0, min_replicas=1)],
+            executors={"a": ex_a, "b": ex_b})
+        # under budget: grants == requests
+        g = await gp.request_scale("a", "backend", 3)
+        assert g == 3 and ex_a.current("backend") == 3
+        g = await gp.request_scale("b", "backend", 4)
+        assert g == 4
+        assert gp.used == 7 <= 8
+        # contention: a wants 10, b wants 6 -> 16 > 8; weighted shares
+        ga = await gp.request_scale("a", "backend", 10)
+        gb = await gp.request_scale("b", "backend", 6)
+        assert ga + gb <= 8
+        assert ga >= 1 and gb >= 1          # floors respected
+        assert ga > gb                      # weight 2:1 favors pool a
+        assert ex_a.current("backend") == gp.granted[("a", "backend")]
+        # relaxing demand returns budget
+        ga2 = await gp.request_scale("a", "backend", 2)
+        gb2 = await gp.request_scale("b", "backend", 6)
+        assert ga2 == 2 and gb2 == 6
+
+    run(main())
+
+
+def test_global_planner_service_delegation():
+    """A local SLA/load planner using GlobalPlannerConnector gets its
+    scale decisions arbitrated by the central service over the request
+    plane (reference: planner connectors/global_planner.py)."""
+    from dynamo_amd.planner import (GlobalPlanner, GlobalPlannerConnector,
+                                    GlobalPlannerService, PoolBudgetPolicy,
+                                    VirtualConnector)
+    from dynamo_amd.runtime import DistributedRuntime, MemoryDiscovery
+
+    async def main():
+        shared = MemoryDiscovery()
+        rt = DistributedRuntime(shared)
+        gp = GlobalPlanner(total_budget=4,
+                           policies=[PoolBudgetPolicy("east"),
+                                     PoolBudgetPolicy("west")])
+        svc = GlobalPlannerService(rt, gp)
+        await svc.start()
+
+        rt2 = DistributedRuntime(shared)
+        local = VirtualConnector({"backend": 1})
+        conn = GlobalPlannerConnector(rt2, pool="east", local=local)
+        granted = await conn.scale("backend", 3)
+        assert granted == 3
+        assert local.current("backend") == 3
+        # second pool contends: total capped at 4
+        rt3 = DistributedRuntime(shared)
+        conn_w = GlobalPlannerConnector(rt3, pool="west")
+        gw = await conn_w.scale("backend", 3)
+        assert gw >= 1 and gp.used <= 4
+        await svc.stop()
+        await rt.shutdown(drain=False)
+        await rt2.shutdown(drain=False)
+        await rt3.shutdown(drain=False)
+
+    run(main())
