@@ -143,6 +143,15 @@ class KvIndexer {
     return it == worker_blocks_.end() ? 0 : (int64_t)it->second.size();
   }
 
+  // snapshot of every indexed block hash (digest building)
+  std::vector<uint64_t> all_hashes() const {
+    std::lock_guard<std::mutex> g(mu_);
+    std::vector<uint64_t> out;
+    out.reserve(index_.size());
+    for (const auto& kv : index_) out.push_back(kv.first);
+    return out;
+  }
+
   int64_t size() const {
     std::lock_guard<std::mutex> g(mu_);
     return (int64_t)index_.size();
@@ -208,6 +217,24 @@ class CuckooFilter {
   size_t count() const { return count_; }
   size_t memory_bytes() const { return buckets_.size() * sizeof(uint16_t); }
 
+  // digest wire form (kv_dc_relay parity: pools publish their KV-block
+  // membership as a compact filter; a global router ranks pools by
+  // max_prefix over a request's hash chain WITHOUT a per-request RPC)
+  py::bytes to_bytes() const {
+    return py::bytes(reinterpret_cast<const char*>(buckets_.data()),
+                     buckets_.size() * sizeof(uint16_t));
+  }
+  static CuckooFilter from_bytes(const py::bytes& data, size_t count) {
+    std::string s = data;
+    CuckooFilter cf(1);
+    cf.buckets_.assign(
+        reinterpret_cast<const uint16_t*>(s.data()),
+        reinterpret_cast<const uint16_t*>(s.data() + s.size()));
+    cf.nbuckets_ = cf.buckets_.size() / 4;
+    cf.count_ = count;
+    return cf;
+  }
+
   int64_t max_prefix(const std::vector<uint64_t>& chain) const {
     int64_t n = 0;
     for (uint64_t h : chain) {
@@ -253,7 +280,10 @@ PYBIND11_MODULE(_core, m) {
       .def("contains", &CuckooFilter::contains)
       .def("count", &CuckooFilter::count)
       .def("memory_bytes", &CuckooFilter::memory_bytes)
-      .def("max_prefix", &CuckooFilter::max_prefix);
+      .def("max_prefix", &CuckooFilter::max_prefix)
+      .def("to_bytes", &CuckooFilter::to_bytes)
+      .def_static("from_bytes", &CuckooFilter::from_bytes,
+                  py::arg("data"), py::arg("count") = 0);
   py::class_<KvIndexer>(m, "KvIndexer")
       .def(py::init<>())
       .def("apply_stored", &KvIndexer::apply_stored)
@@ -262,5 +292,6 @@ PYBIND11_MODULE(_core, m) {
       .def("clear_worker", &KvIndexer::clear_worker)
       .def("find_matches", &KvIndexer::find_matches)
       .def("worker_block_count", &KvIndexer::worker_block_count)
+      .def("all_hashes", &KvIndexer::all_hashes)
       .def("size", &KvIndexer::size);
 }

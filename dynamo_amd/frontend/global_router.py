@@ -35,6 +35,10 @@ class PoolState:
     healthy: bool = False
     inflight: int = 0
     last_check: float = 0.0
+    # cuckoo digests of the pool's cached KV blocks, per model
+    # (kv_dc_relay parity: digest stream instead of per-request RPC)
+    digests: Dict[str, object] = field(default_factory=dict)
+    digest_cfg: Dict[str, tuple] = field(default_factory=dict)  # (bs, salt)
 
 
 class GlobalRouter:
@@ -56,6 +60,7 @@ class GlobalRouter:
         await self.client.aclose()
 
     async def _refresh(self):
+        import base64 as _b64
         for p in self.pools.values():
             try:
                 r = await self.client.get(p.url + "/health", timeout=3.0)
@@ -65,6 +70,25 @@ class GlobalRouter:
             except Exception:
                 p.healthy = False
             p.last_check = time.time()
+            if not p.healthy:
+                continue
+            # pull each model's KV digest (best-effort; stale-tolerant)
+            for model in p.models:
+                try:
+                    r = await self.client.post(
+                        p.url + "/internal/kv_digest",
+                        json={"model": model}, timeout=1.0)
+                    d = r.json()
+                    if d.get("count", 0) and d.get("b64"):
+                        from dynamo_amd import _core
+                        p.digests[model] = _core.CuckooFilter.from_bytes(
+                            _b64.b64decode(d["b64"]), d["count"])
+                        p.digest_cfg[model] = (d["block_size"],
+                                               d.get("salt", 0))
+                    else:
+                        p.digests.pop(model, None)
+                except Exception:
+                    pass
 
     async def _watch(self):
         while True:
@@ -87,6 +111,19 @@ class GlobalRouter:
         by (-overlap_blocks, inflight)."""
         if not token_ids or len(pools) < 2:
             return pools
+
+        # fast path: rank from the polled cuckoo digests (no RPC) when
+        # every candidate pool has one
+        if all(model in p.digests for p in pools):
+            from dynamo_amd import _core
+            scores = []
+            for p in pools:
+                bs, salt = p.digest_cfg[model]
+                chain = _core.chain_hashes(list(token_ids), bs, salt)
+                scores.append(int(p.digests[model].max_prefix(chain)))
+            ranked = sorted(zip(pools, scores),
+                            key=lambda po: (-po[1], po[0].inflight))
+            return [p for p, _ in ranked]
 
         async def ask(p: PoolState) -> int:
             try:

@@ -194,6 +194,32 @@ def build_app(manager: ModelManager) -> FastAPI:
                 "block_size": r.cfg.block_size,
                 "total_blocks": len(hashes)}
 
+    @app.post("/internal/kv_digest")
+    async def kv_digest(raw: Request):
+        """Cuckoo-filter digest of this pool's cached KV blocks
+        (kv_dc_relay parity: lib/llm/src/kv_dc_relay builds cuckoo digests
+        pools publish for cross-DC routing). A global router polls this
+        periodically and ranks pools per request LOCALLY via max_prefix —
+        no per-request overlap RPC."""
+        import base64 as _b64
+        payload = await raw.json()
+        model = payload.get("model", "")
+        try:
+            entry = manager.get(model)
+        except KeyError:
+            return {"count": 0}
+        r = entry.router
+        if r is None:
+            return {"count": 0}
+        from dynamo_amd import _core
+        hashes = r.indexer.all_hashes()
+        cf = _core.CuckooFilter(max(1024, len(hashes) * 2))
+        for h in hashes:
+            cf.insert(h)
+        return {"b64": _b64.b64encode(cf.to_bytes()).decode(),
+                "count": cf.count(), "block_size": r.cfg.block_size,
+                "salt": r.cfg.block_salt}
+
     @app.get("/config")
     async def config_dump():
         """Reproducibility config dump (reference parity:

