@@ -24,6 +24,9 @@ void paged_attention_decode(torch::Tensor out, torch::Tensor q,
                             double scale, int64_t chunk_tokens) {
   TORCH_CHECK(q.is_cuda() && q.dtype() == torch::kBFloat16);
   TORCH_CHECK(page_table.dtype() == torch::kInt32 && ctx_lens.dtype() == torch::kInt32);
+  const bool fp8 = kcache.dtype() == torch::kFloat8_e4m3fn;
+  TORCH_CHECK(fp8 || kcache.dtype() == torch::kBFloat16,
+              "kv cache must be bf16 or float8_e4m3fn");
   const int B = q.size(0);
   const int Hq = q.size(1);
   const int hd = q.size(2);
@@ -92,6 +95,33 @@ void paged_attention_decode(torch::Tensor out, torch::Tensor q,
       max_pages, log2_ps, hd);                                                \
   } while (0)
   const bool mfma_ok = use_mfma && ps % 32 == 0 && hd == 128;
+  if (fp8) {
+    // fp8 KV: the runtime-G swapped kernel is the ONLY fp8 consumer
+    TORCH_CHECK(ps % 32 == 0 && hd == 128 && G <= 16,
+                "fp8 KV cache needs page_size%32==0, head_dim==128, G<=16");
+    if (mfma_swapped_lds_bytes(G, hd) > 65536)
+      (void)hipFuncSetAttribute(
+          reinterpret_cast<const void*>(&paged_decode_mfma_swapped<1, 1, 0, 1>),
+          hipFuncAttributeMaxDynamicSharedMemorySize,
+          mfma_swapped_lds_bytes(G, hd));
+    paged_decode_mfma_swapped<1, 1, 0, 1>
+        <<<grid, kBlock, mfma_swapped_lds_bytes(G, hd), stream>>>(
+        partial.data_ptr<float>(), ml.data_ptr<float>(),
+        (short*)out.data_ptr(), (const short*)q.data_ptr(),
+        (const short*)kcache.data_ptr(), (const short*)vcache.data_ptr(),
+        page_table.data_ptr<int32_t>(), ctx_lens.data_ptr<int32_t>(),
+        (float)scale, chunk, G, B, Hkv, C, max_pages, log2_ps, hd);
+    HIP_CHECK_KERNEL();
+    if (C > 1) {
+      dim3 grid2(B, Hq);
+      paged_decode_phase2<<<grid2, 128, 0, stream>>>(
+          (short*)out.data_ptr(), partial.data_ptr<float>(),
+          ml.data_ptr<float>(), ctx_lens.data_ptr<int32_t>(), chunk, Hq, C,
+          hd);
+      HIP_CHECK_KERNEL();
+    }
+    return;
+  }
   switch (G) {  // combos picked by benchmarks/decode_sweep on MI355X
     case 1: LAUNCH_G(1, 8, 1, 2); break;
     case 2: LAUNCH_G(2, 8, 2, 2); break;

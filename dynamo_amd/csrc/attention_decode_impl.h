@@ -596,7 +596,10 @@ inline int mfma_lds_bytes(int G, int hd) {
 // register double-buffer goes to scratch (guide rule #20), exactly the
 // round-1 K-prefetch regression repeated. Kept compiled for the sweep's
 // record; production uses KPF=0.
-template <int DEFER = 1, int PRIO = 1, int KPF = 0>
+// FP8: the paged cache stores OCP e4m3 bytes (halves decode HBM traffic);
+// fragments convert to bf16 in-kernel via packed v_cvt (guide: decode is
+// HBM-bound, VALU converts are free under the MFMA/mem overlap).
+template <int DEFER = 1, int PRIO = 1, int KPF = 0, int FP8 = 0>
 __global__ __launch_bounds__(kBlock) void paged_decode_mfma_swapped(
     float* __restrict__ partial, float* __restrict__ ml,
     short* __restrict__ out, const short* __restrict__ q,
@@ -665,20 +668,30 @@ __global__ __launch_bounds__(kBlock) void paged_decode_mfma_swapped(
   // prefetchable loads (KPF): K fragments (8x short8) + V staging rows
   // (8x short8) per tile, double-buffered across iterations
   typedef __attribute__((ext_vector_type(4))) short short4v;
+  auto ld8 = [&](const short* base, int64_t elem_off) -> short8 {
+    // one 8-element K/V slice as a bf16x8-compatible short8; FP8 caches
+    // hold bytes at the same ELEMENT offsets
+    if constexpr (FP8) {
+      const unsigned char* b =
+          reinterpret_cast<const unsigned char*>(base) + elem_off;
+      uchar8 raw = *reinterpret_cast<const uchar8*>(b);
+      bf16x8 cv = fp8x8_to_bf16x8(raw);
+      return *reinterpret_cast<short8*>(&cv);
+    } else {
+      return *reinterpret_cast<const short8*>(base + elem_off);
+    }
+  };
   auto load_k = [&](int t0_, short8 (&kf)[8]) {
     if (t0_ >= slab_end) return;
     const int64_t pb = (((int64_t)pt[t0_ >> log2_ps] * Hkv + h) * ps) * hd;
     const int tA = t0_ + lr, tB = t0_ + 16 + lr;
-    const short* krA = kcache + pb + (int64_t)(tA & (ps - 1)) * hd;
-    const short* krB = kcache + pb + (int64_t)(tB & (ps - 1)) * hd;
+    const int64_t oA = pb + (int64_t)(tA & (ps - 1)) * hd;
+    const int64_t oB = pb + (int64_t)(tB & (ps - 1)) * hd;
     const bool vA = tA < slab_end, vB = tB < slab_end;
 #pragma unroll
     for (int kc = 0; kc < 4; kc++) {
-      kf[kc] = vA ? *reinterpret_cast<const short8*>(krA + kc * 32 + lg * 8)
-                  : short8{};
-      kf[4 + kc] = vB
-          ? *reinterpret_cast<const short8*>(krB + kc * 32 + lg * 8)
-          : short8{};
+      kf[kc] = vA ? ld8(kcache, oA + kc * 32 + lg * 8) : short8{};
+      kf[4 + kc] = vB ? ld8(kcache, oB + kc * 32 + lg * 8) : short8{};
     }
   };
   auto load_v = [&](int t0_, short8 (&vf)[8]) {
@@ -693,8 +706,7 @@ __global__ __launch_bounds__(kBlock) void paged_decode_mfma_swapped(
       for (int j = 0; j < 4; j++) {
         const int t = t0_ + tg * 4 + j;
         vf[u * 4 + j] = (t < slab_end)
-            ? *reinterpret_cast<const short8*>(
-                  vcache + pb + (int64_t)(t & (ps - 1)) * hd + d8 * 8)
+            ? ld8(vcache, pb + (int64_t)(t & (ps - 1)) * hd + d8 * 8)
             : short8{};
       }
     }

@@ -272,7 +272,9 @@ constexpr int kLdsHalf32 = kKB32 * 256 + 128 * 128;  // K + V^T = 32 KB
 // GSPLIT generalizes to GQA groups G = 8/GSPLIT: the 8 warps are G heads
 // x GSPLIT 32-row q-subtiles, all sharing one staged K/V tile (block tile
 // is GSPLIT*32 q rows). G=8 -> 1 subtile; G=4 -> 2; G=2 -> 4.
-template <int GSPLIT>
+// FP8: paged cache stores OCP e4m3 (converted to bf16 while staging to
+// LDS; compute unchanged)
+template <int GSPLIT, int FP8 = 0>
 __global__ __launch_bounds__(512) void prefill32_kernel(
     short* __restrict__ out,            // [Tq, Hq, 128]
     const short* __restrict__ q,        // [Tq, Hq, 128]
@@ -328,6 +330,16 @@ __global__ __launch_bounds__(512) void prefill32_kernel(
   // threads 256-511 stage K (b128 writes); register prefetch double-buffer
   const int vrole = threadIdx.x < 256;
   short8 sreg[4];
+  auto cache8 = [&](const short* base, int64_t elem_off) -> short8 {
+    if constexpr (FP8) {
+      uchar8 raw = *reinterpret_cast<const uchar8*>(
+          reinterpret_cast<const unsigned char*>(base) + elem_off);
+      bf16x8 cv = fp8x8_to_bf16x8(raw);
+      return *reinterpret_cast<short8*>(&cv);
+    } else {
+      return *reinterpret_cast<const short8*>(base + elem_off);
+    }
+  };
   auto load_tile = [&](int t0) {
     if (vrole) {
       const int unit = threadIdx.x;
@@ -336,8 +348,8 @@ __global__ __launch_bounds__(512) void prefill32_kernel(
       for (int j = 0; j < 4; j++) {
         const int t = min(t0 + row0 + j, ctx - 1);
         const int64_t page = pt[t >> log2_ps];
-        sreg[j] = *reinterpret_cast<const short8*>(
-            vcache + ((page * Hkv + kvh) * ps + (t & (ps - 1))) * HD + d0);
+        sreg[j] = cache8(vcache,
+            ((page * Hkv + kvh) * ps + (t & (ps - 1))) * HD + d0);
       }
     } else {
       const int idx = threadIdx.x - 256;
@@ -347,8 +359,8 @@ __global__ __launch_bounds__(512) void prefill32_kernel(
         const int row = c >> 4, col8 = (c & 15) * 8;
         const int t = min(t0 + row, ctx - 1);
         const int64_t page = pt[t >> log2_ps];
-        sreg[u] = *reinterpret_cast<const short8*>(
-            kcache + ((page * Hkv + kvh) * ps + (t & (ps - 1))) * HD + col8);
+        sreg[u] = cache8(kcache,
+            ((page * Hkv + kvh) * ps + (t & (ps - 1))) * HD + col8);
       }
     }
   };
@@ -520,6 +532,9 @@ void attention_prefill_paged(torch::Tensor out, torch::Tensor q,
   TORCH_CHECK(q.is_cuda() && q.dtype() == torch::kBFloat16);
   TORCH_CHECK(q.size(-1) == 128, "only head_dim=128 supported natively");
   TORCH_CHECK(page_table.dtype() == torch::kInt32);
+  const bool fp8 = kcache.dtype() == torch::kFloat8_e4m3fn;
+  TORCH_CHECK(fp8 || kcache.dtype() == torch::kBFloat16,
+              "kv cache must be bf16 or float8_e4m3fn");
   const int Hq = q.size(1);
   const int Hkv = kcache.size(1);
   const int ps = kcache.size(2);
@@ -544,12 +559,17 @@ void attention_prefill_paged(torch::Tensor out, torch::Tensor q,
           seq_q_len.data_ptr<int32_t>(), seq_ctx_len.data_ptr<int32_t>(),
           (float)scale, Hq, Hkv, max_pages, log2_ps);
     };
-    if (G == 8) launch(prefill32_kernel<1>);
+    if (fp8) {
+      if (G == 8) launch(prefill32_kernel<1, 1>);
+      else if (G == 4) launch(prefill32_kernel<2, 1>);
+      else launch(prefill32_kernel<4, 1>);
+    } else if (G == 8) launch(prefill32_kernel<1>);
     else if (G == 4) launch(prefill32_kernel<2>);
     else launch(prefill32_kernel<4>);
     HIP_CHECK_KERNEL();
     return;
   }
+  TORCH_CHECK(!fp8, "fp8 KV prefill needs GQA group in {2,4,8}");
   dim3 grid(ntiles, Hq);
   prefill_kernel<<<grid, kBlock, 0, stream>>>(
       (short*)out.data_ptr(), (const short*)q.data_ptr(),

@@ -110,9 +110,12 @@ static void pages_launch_dims(int64_t page_elems, int n, dim3& grid) {
 void gather_pages(torch::Tensor staging, torch::Tensor cache, torch::Tensor page_ids) {
   TORCH_CHECK(staging.is_cuda() && cache.is_cuda());
   TORCH_CHECK(page_ids.dtype() == torch::kInt32);
-  const int64_t page_elems = cache.numel() / cache.size(0);
+  // elem-size-agnostic: kernels copy in 2-byte units (fp8 caches pass
+  // half as many "short elems" per page)
+  const int64_t page_elems =
+      cache.numel() / cache.size(0) * cache.element_size() / 2;
   const int n = page_ids.size(0);
-  TORCH_CHECK(staging.numel() >= n * page_elems);
+  TORCH_CHECK(staging.numel() * staging.element_size() >= n * page_elems * 2);
   if (n == 0) return;
   dim3 grid; pages_launch_dims(page_elems, n, grid);
   auto stream = at::cuda::getCurrentHIPStream();
@@ -125,7 +128,10 @@ void gather_pages(torch::Tensor staging, torch::Tensor cache, torch::Tensor page
 void scatter_pages(torch::Tensor staging, torch::Tensor cache, torch::Tensor page_ids) {
   TORCH_CHECK(staging.is_cuda() && cache.is_cuda());
   TORCH_CHECK(page_ids.dtype() == torch::kInt32);
-  const int64_t page_elems = cache.numel() / cache.size(0);
+  // elem-size-agnostic: kernels copy in 2-byte units (fp8 caches pass
+  // half as many "short elems" per page)
+  const int64_t page_elems =
+      cache.numel() / cache.size(0) * cache.element_size() / 2;
   const int n = page_ids.size(0);
   if (n == 0) return;
   dim3 grid; pages_launch_dims(page_elems, n, grid);
@@ -138,7 +144,8 @@ void scatter_pages(torch::Tensor staging, torch::Tensor cache, torch::Tensor pag
 
 void copy_pages(torch::Tensor dst_cache, torch::Tensor src_cache, torch::Tensor pairs) {
   TORCH_CHECK(pairs.dtype() == torch::kInt32 && pairs.dim() == 2 && pairs.size(1) == 2);
-  const int64_t page_elems = src_cache.numel() / src_cache.size(0);
+  const int64_t page_elems =
+      src_cache.numel() / src_cache.size(0) * src_cache.element_size() / 2;
   const int n = pairs.size(0);
   if (n == 0) return;
   dim3 grid; pages_launch_dims(page_elems, n, grid);

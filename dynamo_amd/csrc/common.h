@@ -88,6 +88,33 @@ DEVINL float block_reduce_sum(float x, float* lds_scratch) {
 // ---- misc ---------------------------------------------------------------
 DEVINL int64_t cdiv64(int64_t a, int64_t b) { return (a + b - 1) / b; }
 
+// ---- fp8 (OCP e4m3) KV cache support --------------------------------------
+typedef __attribute__((ext_vector_type(8))) unsigned char uchar8;
+
+#if !defined(__HIP_DEVICE_COMPILE__)
+// host pass only parses kernel bodies; device builtins are unavailable
+DEVINL bf16x8 fp8x8_to_bf16x8(uchar8) { return bf16x8{}; }
+#else
+// 8 packed e4m3 bytes -> 8 bf16 (MFMA A/B fragment shape): 4x native
+// v_cvt_pk_f32_fp8 + 4x v_cvt_pk_bf16_f32
+DEVINL bf16x8 fp8x8_to_bf16x8(uchar8 v) {
+  union { uchar8 u8; unsigned int w[2]; } in;
+  in.u8 = v;
+  unsigned int out[4];
+  // word-select operand must be a compile-time constant
+  float2v f0 = __builtin_amdgcn_cvt_pk_f32_fp8((int)in.w[0], false);
+  float2v f1 = __builtin_amdgcn_cvt_pk_f32_fp8((int)in.w[0], true);
+  float2v f2 = __builtin_amdgcn_cvt_pk_f32_fp8((int)in.w[1], false);
+  float2v f3 = __builtin_amdgcn_cvt_pk_f32_fp8((int)in.w[1], true);
+  asm("v_cvt_pk_bf16_f32 %0, %1, %2" : "=v"(out[0]) : "v"(f0.x), "v"(f0.y));
+  asm("v_cvt_pk_bf16_f32 %0, %1, %2" : "=v"(out[1]) : "v"(f1.x), "v"(f1.y));
+  asm("v_cvt_pk_bf16_f32 %0, %1, %2" : "=v"(out[2]) : "v"(f2.x), "v"(f2.y));
+  asm("v_cvt_pk_bf16_f32 %0, %1, %2" : "=v"(out[3]) : "v"(f3.x), "v"(f3.y));
+  return *reinterpret_cast<bf16x8*>(out);
+}
+#endif
+
+
 // 16-lane (DPP row) reductions via row_ror rotations: pure VALU, no
 // ds_bpermute — cross-lane LDS-unit shuffles have ~60-cycle latency and
 // serialize softmax dependency chains (measured: replacing them in the

@@ -4,6 +4,7 @@
 // (first half cos, second half sin) per guide Appendix B: on-device trig
 // turns a memory-bound op VALU-bound. Vectorized 8-wide bf16 loads.
 #include "common.h"
+#include <hip/hip_fp8.h>
 #include <torch/extension.h>
 #include <ATen/hip/HIPContext.h>
 
@@ -56,6 +57,7 @@ __global__ void rope_kernel(short* __restrict__ q,    // [T, Hq*hd]
 // scatters k/v straight into the paged cache. Replaces 4 kernels
 // (q copy, k copy, rope, kv_append) with one launch — on Llama-70B decode
 // that is 3 fewer hipGraph nodes per layer x 80 layers.
+template <int FP8 = 0>
 __global__ void rope_append_kernel(
     short* __restrict__ q_out,          // [T, Hq*hd] contiguous
     short* __restrict__ kcache,         // [P, Hkv, ps, hd]
@@ -108,9 +110,23 @@ __global__ void rope_append_kernel(
       *reinterpret_cast<short8*>(qb + half + i0) = o2;
     } else if (slot >= 0) {
       const int kh = h - Hq;
-      short* kb = kcache + (((page * Hkv + kh) * page_size + off) * hd);
-      *reinterpret_cast<short8*>(kb + i0) = o1;
-      *reinterpret_cast<short8*>(kb + half + i0) = o2;
+      const int64_t kb = ((page * Hkv + kh) * page_size + off) * hd;
+      if constexpr (FP8) {
+        unsigned char* kc8 = reinterpret_cast<unsigned char*>(kcache);
+        uchar8 p1, p2;
+#pragma unroll
+        for (int e = 0; e < 8; e++) {
+          p1[e] = __hip_cvt_float_to_fp8(bf16_to_f32(o1[e]),
+                                         __HIP_SATFINITE, __HIP_E4M3);
+          p2[e] = __hip_cvt_float_to_fp8(bf16_to_f32(o2[e]),
+                                         __HIP_SATFINITE, __HIP_E4M3);
+        }
+        *reinterpret_cast<uchar8*>(kc8 + kb + i0) = p1;
+        *reinterpret_cast<uchar8*>(kc8 + kb + half + i0) = p2;
+      } else {
+        *reinterpret_cast<short8*>(kcache + kb + i0) = o1;
+        *reinterpret_cast<short8*>(kcache + kb + half + i0) = o2;
+      }
     }
   }
   // part 2: v heads (plain copy into the cache)
@@ -127,8 +143,18 @@ __global__ void rope_append_kernel(
           v[e] = f32_to_bf16(bf16_to_f32(v[e]) +
                              bf16_to_f32(bias[voff + h * hd + d + e]));
       }
-      *reinterpret_cast<short8*>(
-          vcache + (((page * Hkv + h) * page_size + off) * hd + d)) = v;
+      const int64_t vb = ((page * Hkv + h) * page_size + off) * hd + d;
+      if constexpr (FP8) {
+        uchar8 pv;
+#pragma unroll
+        for (int e = 0; e < 8; e++)
+          pv[e] = __hip_cvt_float_to_fp8(bf16_to_f32(v[e]),
+                                         __HIP_SATFINITE, __HIP_E4M3);
+        *reinterpret_cast<uchar8*>(
+            reinterpret_cast<unsigned char*>(vcache) + vb) = pv;
+      } else {
+        *reinterpret_cast<short8*>(vcache + vb) = v;
+      }
     }
   }
 }
@@ -159,13 +185,20 @@ void rope_append_qkv(torch::Tensor q_out, torch::Tensor kcache,
     bp = (const short*)bias->data_ptr();
   }
   if (T == 0) return;
+  const bool fp8 = kcache.dtype() == torch::kFloat8_e4m3fn;
+  TORCH_CHECK(fp8 || kcache.dtype() == torch::kBFloat16,
+              "kv cache must be bf16 or float8_e4m3fn");
   auto stream = at::cuda::getCurrentHIPStream();
-  rope_append_kernel<<<T, kBlock, 0, stream>>>(
-      (short*)q_out.data_ptr(), (short*)kcache.data_ptr(),
-      (short*)vcache.data_ptr(), (const short*)qkv.data_ptr(), bp,
-      positions.data_ptr<int32_t>(), slot_mapping.data_ptr<int64_t>(),
-      cos_sin_cache.data_ptr<float>(), T, Hq, Hkv, page_size, hd,
-      (int)qkv.stride(0));
+  auto launch = [&](auto kern) {
+    kern<<<T, kBlock, 0, stream>>>(
+        (short*)q_out.data_ptr(), (short*)kcache.data_ptr(),
+        (short*)vcache.data_ptr(), (const short*)qkv.data_ptr(), bp,
+        positions.data_ptr<int32_t>(), slot_mapping.data_ptr<int64_t>(),
+        cos_sin_cache.data_ptr<float>(), T, Hq, Hkv, page_size, hd,
+        (int)qkv.stride(0));
+  };
+  if (fp8) launch(rope_append_kernel<1>);
+  else launch(rope_append_kernel<0>);
   HIP_CHECK_KERNEL();
 }
 

@@ -128,6 +128,11 @@ class EngineConfig:
     disk_cache_path: str = ""           # G3 backing file (required if G3 on)
     object_cache_dir: str = ""          # G4 shared object store (disabled="")
     kv_events: bool = True              # emit stored/removed block events
+    # KV cache storage dtype: "auto" matches the compute dtype; "fp8"
+    # stores OCP e4m3 (halves decode-attention HBM traffic; GPU-only,
+    # requires GQA group in 2/4/8/16 and head_dim 128 - reference parity
+    # with its engines' --kv-cache-dtype fp8)
+    kv_cache_dtype: str = "auto"
     # waiting-queue admission policy (reference parity: kv-router
     # scheduling/policy.rs SchedulingPolicy FCFS/LCFS/WSPT):
     #   fcfs = arrival order; lcfs = newest first; wspt = shortest
@@ -148,3 +153,12 @@ class EngineConfig:
     def torch_dtype(self):
         import torch
         return {"bfloat16": torch.bfloat16, "float32": torch.float32}[self.dtype]
+
+    @property
+    def kv_torch_dtype(self):
+        import torch
+        if self.kv_cache_dtype == "fp8":
+            if not self.device.startswith("cuda"):
+                raise ValueError("fp8 KV cache is GPU-only")
+            return torch.float8_e4m3fn
+        return self.torch_dtype

@@ -41,7 +41,8 @@ class ModelRunner:
         num_pages = cfg.kv_pool_pages or self._auto_pages()
         self.kv_pool = KVCachePool(m.num_layers, num_pages, self.hkv_local,
                                    cfg.page_size, m.head_dim, cfg.device,
-                                   self.dtype, shm_export=cfg.cpu_shm_pool)
+                                   cfg.kv_torch_dtype,
+                                   shm_export=cfg.cpu_shm_pool)
         self.num_pages = num_pages
         self.max_pages_per_seq = (cfg.max_model_len + cfg.page_size - 1) // cfg.page_size
         self.decode_scratch = None

@@ -41,6 +41,10 @@ def build_parser():
     p.add_argument("--no-prefix-caching", action="store_true")
     p.add_argument("--no-hip-graphs", action="store_true")
     p.add_argument("--seed", type=int, default=0)
+    p.add_argument("--kv-cache-dtype", default="auto",
+                   choices=["auto", "fp8"],
+                   help="fp8 stores the paged KV cache as OCP e4m3 "
+                        "(half the decode HBM bytes; GPU only)")
     p.add_argument("--dtype", default=None,
                    help="bfloat16 | float32 (default: bf16 on GPU, "
                         "fp32 on CPU)")
@@ -122,6 +126,7 @@ def make_engine_from_args(args, tp=None) -> LLMEngine:
         max_model_len=args.max_model_len,
         kv_pool_pages=args.kv_pool_pages,
         gpu_mem_fraction=args.gpu_mem_fraction,
+        kv_cache_dtype=args.kv_cache_dtype,
         enable_prefix_caching=not args.no_prefix_caching,
         enable_hip_graphs=not args.no_hip_graphs,
         worker_type=args.worker_type,

@@ -153,6 +153,74 @@ def test_paged_attention_decode(G, ctxs):
     assert_close(out, ref)
 
 
+@pytest.mark.parametrize("G", [8, 4, 7])
+def test_paged_attention_decode_fp8_kv(G):
+    """fp8 (e4m3) KV cache: decode over a quantized cache matches the
+    bf16-cache result within e4m3 quantization noise (values in [-1,1]:
+    ~2 mantissa-digit relative error through the softmax-weighted sum)."""
+    torch.manual_seed(2)
+    Hkv, ps, hd = 8, 64, 128
+    Hq = G * Hkv
+    ctxs = [900, 64, 2048]
+    B = len(ctxs)
+    max_pages_seq = (max(ctxs) + ps - 1) // ps
+    P = sum((c + ps - 1) // ps for c in ctxs) + 1
+    kc = torch.randn(P, Hkv, ps, hd, dtype=torch.bfloat16, device=DEV)
+    vc = torch.randn(P, Hkv, ps, hd, dtype=torch.bfloat16, device=DEV)
+    kc8 = kc.to(torch.float8_e4m3fn)
+    vc8 = vc.to(torch.float8_e4m3fn)
+    pt = torch.zeros(B, max_pages_seq, dtype=torch.int32, device=DEV)
+    nxt = 1
+    for b, c in enumerate(ctxs):
+        n = (c + ps - 1) // ps
+        pt[b, :n] = torch.arange(nxt, nxt + n, dtype=torch.int32)
+        nxt += n
+    q = torch.randn(B, Hq, hd, dtype=torch.bfloat16, device=DEV)
+    ctx_lens = torch.tensor(ctxs, dtype=torch.int32, device=DEV)
+    scale = hd ** -0.5
+    scratch = ops.DecodeScratch(B, Hq, hd, max(ctxs), DEV)
+    out8 = ops.paged_attention_decode(q, kc8, vc8, pt, ctx_lens, scale,
+                                      scratch)
+    # reference: bf16 path over the DEQUANTIZED cache (isolates the
+    # kernel's fp8 read path from quantization error)
+    ref = ops.paged_attention_decode(q, kc8.to(torch.bfloat16),
+                                     vc8.to(torch.bfloat16), pt, ctx_lens,
+                                     scale, scratch)
+    assert_close(out8, ref, rtol=0.02, atol=0.02)
+    # and against the true bf16 cache within quantization noise
+    full = ops.paged_attention_decode(q, kc, vc, pt, ctx_lens, scale,
+                                      scratch)
+    assert_close(out8, full, rtol=0.12, atol=0.12)
+
+
+def test_rope_append_and_prefill_fp8_kv():
+    """fp8 end-to-end write+read: rope_append_qkv writes an e4m3 cache;
+    prefill attention over it matches the bf16-cache prefill within
+    quantization noise."""
+    torch.manual_seed(4)
+    Hq, Hkv, hd, ps = 16, 2, 128, 64
+    S = 192
+    P = S // ps + 1
+    qkv = torch.randn(S, (Hq + 2 * Hkv) * hd, dtype=torch.bfloat16,
+                      device=DEV) * 0.5
+    pos = torch.arange(S, dtype=torch.int32, device=DEV)
+    slots = torch.arange(S, dtype=torch.int64, device=DEV)
+    cos_sin = torch_ref.make_cos_sin_cache(512, hd, 10000.0, device=DEV)
+    outs = {}
+    for tag, dt in (("bf16", torch.bfloat16), ("fp8", torch.float8_e4m3fn)):
+        kc = torch.zeros(P, Hkv, ps, hd, dtype=dt, device=DEV)
+        vc = torch.zeros(P, Hkv, ps, hd, dtype=dt, device=DEV)
+        qq = ops.rope_append_qkv(qkv, None, pos, slots, cos_sin, kc, vc,
+                                 Hq, Hkv, hd)
+        qh = qq.view(S, Hq, hd)
+        pt = torch.arange(P, dtype=torch.int32, device=DEV).unsqueeze(0)
+        st = torch.tensor([0], dtype=torch.int32, device=DEV)
+        ln = torch.tensor([S], dtype=torch.int32, device=DEV)
+        outs[tag] = ops.attention_prefill_paged(
+            qh, kc, vc, pt, st, ln, ln, hd ** -0.5)
+    assert_close(outs["fp8"], outs["bf16"], rtol=0.12, atol=0.12)
+
+
 @pytest.mark.parametrize("G,spec", [
     # (q_len, ctx_len) pairs; ctx_len >= q_len (chunked prefill / prefix hit)
     (4, [(64, 64)]),
