@@ -44,6 +44,10 @@ def parse_args():
     p.add_argument("--page-size", type=int, default=64)
     p.add_argument("--mode", choices=["auto", "dp", "disagg"], default="auto")
     p.add_argument("--kv-pool-pages", type=int, default=0)
+    p.add_argument("--kv-cache-dtype", default="auto",
+                   choices=["auto", "fp8"],
+                   help="fp8 KV measurement runs are SUPPLEMENTARY: the "
+                        "flagship headline stays bf16 (BASELINE config)")
     p.add_argument("--device", default=None, help="cpu for gloo testing")
     p.add_argument("--max-batched-tokens", type=int, default=0)
     p.add_argument("--moe-ep", action="store_true",
@@ -64,6 +68,7 @@ def make_cfg(args, mc, device, world=1, rank=0, worker_type="aggregated",
         max_model_len=args.isl + args.osl + 64,
         kv_pool_pages=args.kv_pool_pages,
         enable_prefix_caching=False,  # synthetic distinct prompts
+        kv_cache_dtype=args.kv_cache_dtype,
         dtype="bfloat16" if device.startswith("cuda") else "float32",
         worker_type=worker_type, tp_size=world, tp_rank=rank)
 
@@ -100,6 +105,7 @@ def emit(args, world, value, ms_per_step, ttft_p50, prefill_time, mode, conc):
             "seq_len": args.isl,
             "osl": args.osl,
             "parallelism": mode,
+            "kv_cache_dtype": args.kv_cache_dtype,
             "ttft_p50_s": round(ttft_p50, 3) if ttft_p50 else None,
             "itl_p50_ms": round(ms_per_step, 3),
             "prefill_time_s": round(prefill_time, 2),
