@@ -35,14 +35,15 @@ def pct(xs, p):
 
 
 async def one_request(client, args, prompt_tokens, stats, session=None,
-                      collect=None):
+                      collect=None, osl=None):
     t0 = time.monotonic()
     first = None
     ntok = 0
     last = t0
     itls = []
     body = {"model": args.model, "prompt": prompt_tokens,
-            "max_tokens": args.osl, "stream": True, "ignore_eos": True}
+            "max_tokens": osl or args.osl, "stream": True,
+            "ignore_eos": True}
     if session:
         body["user"] = session    # sticky-session pin
     try:
@@ -113,6 +114,62 @@ async def run_multiturn(args):
         "itl_p50_ms": round(pct(stats["itl"], 50), 3) if stats["itl"] else None,
         "config": {"isl_chunk": chunk, "osl": args.osl,
                    "concurrency": args.concurrency},
+    }
+    print(json.dumps(out))
+    return out
+
+
+async def run_trace(args):
+    """Replay a converted trace (tools/trace_convert.py output): each
+    record fires at ts_s with its own isl/osl; records sharing a
+    prefix_group share a prompt prefix of prefix_len tokens (the
+    reference's mooncake-trace replay discipline)."""
+    import json as js
+    rng = random.Random(args.seed)
+    vocab = args.vocab
+    recs = [js.loads(ln) for ln in open(args.trace) if ln.strip()]
+    prefixes = {}
+
+    def make_prompt(rec):
+        isl = max(1, rec.get("isl") or args.isl)
+        plen = min(int(rec.get("prefix_len") or 0), isl - 1)
+        g = rec.get("prefix_group")
+        pre = []
+        if g is not None and plen > 0:
+            key = (g, plen)
+            if key not in prefixes:
+                grng = random.Random(hash((args.seed, g)) & 0xFFFFFFFF)
+                prefixes[key] = [grng.randrange(vocab) for _ in range(plen)]
+            pre = prefixes[key]
+        return pre + [rng.randrange(vocab) for _ in range(isl - len(pre))]
+
+    stats = {"ttft": [], "itl": [], "latency": [], "tokens": 0, "errors": 0}
+    limits = httpx.Limits(max_connections=args.concurrency + 8)
+    async with httpx.AsyncClient(timeout=None, limits=limits) as client:
+        sem = asyncio.Semaphore(args.concurrency)
+        t0 = time.monotonic()
+
+        async def fire(rec):
+            delay = rec["ts_s"] - (time.monotonic() - t0)
+            if delay > 0:
+                await asyncio.sleep(delay)
+            async with sem:
+                await one_request(client, args, make_prompt(rec), stats,
+                                  osl=max(1, rec.get("osl") or args.osl))
+
+        await asyncio.gather(*[fire(r) for r in recs])
+        wall = time.monotonic() - t0
+
+    ttfts = [t for t in stats["ttft"] if t is not None]
+    out = {
+        "requests": len(recs),
+        "errors": stats["errors"],
+        "wall_s": round(wall, 3),
+        "output_tok_s": round(stats["tokens"] / wall, 2),
+        "ttft_p50_s": round(pct(ttfts, 50), 4) if ttfts else None,
+        "ttft_p95_s": round(pct(ttfts, 95), 4) if ttfts else None,
+        "itl_p50_ms": round(pct(stats["itl"], 50), 3) if stats["itl"] else None,
+        "config": {"trace": args.trace, "concurrency": args.concurrency},
     }
     print(json.dumps(out))
     return out
@@ -191,8 +248,16 @@ def main():
                    help=">0: open-loop Poisson req/s; 0: closed loop")
     p.add_argument("--vocab", type=int, default=512)
     p.add_argument("--seed", type=int, default=0)
+    p.add_argument("--trace", default=None,
+                   help="replay a converted trace file "
+                        "(tools/trace_convert.py output)")
     args = p.parse_args()
-    asyncio.run(run_multiturn(args) if args.turns > 0 else run(args))
+    if args.trace:
+        asyncio.run(run_trace(args))
+    elif args.turns > 0:
+        asyncio.run(run_multiturn(args))
+    else:
+        asyncio.run(run(args))
 
 
 if __name__ == "__main__":

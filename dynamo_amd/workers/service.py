@@ -133,13 +133,15 @@ class WorkerService:
         self._loop_task = asyncio.create_task(self._engine_loop())
         return self
 
-    async def stop(self, drain: bool = False, drain_timeout: float = 30.0):
-        """Stop serving. drain=True: deregister from discovery first (no
-        new routing), then let in-flight requests finish before killing
-        the engine loop (reference parity: push_endpoint.rs:46-56 inflight
-        counter + graceful-shutdown-architecture.md)."""
+    async def stop(self, drain: bool = True, drain_timeout: float = 10.0):
+        """Stop serving. Default is GRACEFUL (reference parity:
+        push_endpoint.rs:46-56 inflight counter +
+        graceful-shutdown-architecture.md): deregister from discovery
+        first (no new routing), then let in-flight requests finish before
+        killing the engine loop. drain=False aborts immediately (tests /
+        fault injection)."""
         self.comp.deregister()
-        if drain:
+        if drain and self.engine.has_work():
             deadline = asyncio.get_event_loop().time() + drain_timeout
             while (self.engine.has_work()
                    and asyncio.get_event_loop().time() < deadline):

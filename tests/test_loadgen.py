@@ -136,3 +136,65 @@ def test_profiler_sweep(tmp_path):
     finally:
         front.stop()
         w.stop()
+
+
+@pytest.mark.timeout(180)
+def test_trace_convert_and_replay(tmp_path):
+    """Mooncake-format trace -> converter -> loadgen --trace replay
+    against a live mock stack (reference parity: lib/bench trace
+    converters + offline_replay_bench)."""
+    import os
+    ROOT = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    raw = tmp_path / "mooncake.jsonl"
+    recs = [
+        {"timestamp": 1000, "input_length": 40, "output_length": 3,
+         "hash_ids": [7, 8], "block_size": 8},
+        {"timestamp": 1050, "input_length": 40, "output_length": 3,
+         "hash_ids": [7, 8], "block_size": 8},     # same prefix group
+        {"timestamp": 1200, "input_length": 24, "output_length": 2,
+         "hash_ids": [], "block_size": 8},
+    ]
+    raw.write_text("\n".join(json.dumps(r) for r in recs))
+    conv = tmp_path / "replay.jsonl"
+    r = subprocess.run(
+        [sys.executable, "-m", "dynamo_amd.tools.trace_convert",
+         "--format", "mooncake", "--in", str(raw), "--out", str(conv),
+         "--speed", "10.0"],
+        capture_output=True, text=True, cwd=ROOT)
+    assert r.returncode == 0, r.stderr
+    lines = [json.loads(ln) for ln in conv.read_text().splitlines()]
+    assert len(lines) == 3
+    assert lines[0]["ts_s"] == 0.0
+    assert lines[0]["prefix_group"] == 7 and lines[0]["prefix_len"] == 16
+    assert lines[2]["prefix_group"] is None
+
+    disc = f"file:{tmp_path}/disc"
+    w = ManagedProcess(worker_cmd(mock=True, model="tiny-llama",
+                                  discovery=disc, page_size=16),
+                       ready_marker="WORKER_READY").start()
+    front = ManagedProcess(
+        [sys.executable, "-m", "dynamo_amd.frontend", "--discovery", disc,
+         "--port", "18237"], ready_marker="FRONTEND_READY").start()
+    try:
+        base = "http://127.0.0.1:18237"
+        deadline = time.time() + 60
+        with httpx.Client(timeout=10) as c:
+            while time.time() < deadline:
+                try:
+                    if c.get(base + "/health").json()["models"]:
+                        break
+                except httpx.TransportError:
+                    pass
+                time.sleep(0.3)
+        out = subprocess.run(
+            [sys.executable, "benchmarks/loadgen.py", "--url", base,
+             "--model", "tiny-llama", "--trace", str(conv),
+             "--concurrency", "4", "--vocab", "512"],
+            capture_output=True, text=True, cwd=ROOT, timeout=120)
+        assert out.returncode == 0, out.stdout + out.stderr
+        rep = json.loads(out.stdout.strip().splitlines()[-1])
+        assert rep["requests"] == 3 and rep["errors"] == 0
+        assert rep["output_tok_s"] > 0
+    finally:
+        front.stop()
+        w.stop()
