@@ -140,14 +140,17 @@ __global__ __launch_bounds__(kBlock) void moe_gemm_mfma_kernel(
   const int wid = threadIdx.x >> 6;
   const int lr = lane & 15;
   const int lg = lane >> 4;
-  // TWO 16-column N-tiles per wave: doubles the independent B loads in
-  // flight per lane (deeper W-stream pipeline) and reuses the A fragment
-  const int n0 = (blockIdx.y * 4 + wid) * 32;
-  const int n1 = n0 + 16;
-  const short* wbase0 = w + ((int64_t)e * N + n0 + lr) * D;
-  const short* wbase1 = w + ((int64_t)e * N + n1 + lr) * D;
-  const bool ok0 = (n0 + lr) < N;
-  const bool ok1 = (n1 + lr) < N;
+  // FOUR 16-column N-tiles per wave: 4 independent B loads in flight per
+  // kc per lane (the VALU kernel's lesson: in-flight depth beats
+  // everything on the W stream), all sharing one A fragment
+  const int n0 = (blockIdx.y * 4 + wid) * 64;
+  const short* wb[4];
+  bool ok[4];
+#pragma unroll
+  for (int t = 0; t < 4; t++) {
+    wb[t] = w + ((int64_t)e * N + n0 + t * 16 + lr) * D;
+    ok[t] = (n0 + t * 16 + lr) < N;
+  }
 
   // X tile [16][kXC] staged per 512-dim chunk (16 KB LDS), XOR-swizzled
   // rows; the wide chunk keeps 16 B-loads + MFMAs between barriers so the
@@ -155,7 +158,7 @@ __global__ __launch_bounds__(kBlock) void moe_gemm_mfma_kernel(
   constexpr int kXC = 512;
   __shared__ short x_lds[kMaxM * kXC];
 
-  f32x4 acc{0.f, 0.f, 0.f, 0.f}, acc1{0.f, 0.f, 0.f, 0.f};
+  f32x4 accs[4] = {};
   for (int dc = 0; dc < D; dc += kXC) {
     const int cw = min(kXC, D - dc);
     __syncthreads();
@@ -176,32 +179,30 @@ __global__ __launch_bounds__(kBlock) void moe_gemm_mfma_kernel(
       short8 a = *reinterpret_cast<const short8*>(
           (char*)x_lds + lr * (kXC * 2) +
           ((kc * 64 + lg * 16) ^ ((lr & 7) << 4)));
-      // B: W[n][dc + kc*32 + 8lg + j] — direct global b128, two tiles
-      short8 b0 = ok0
-          ? *reinterpret_cast<const short8*>(wbase0 + dc + kc * 32 + lg * 8)
-          : short8{};
-      short8 b1 = ok1
-          ? *reinterpret_cast<const short8*>(wbase1 + dc + kc * 32 + lg * 8)
-          : short8{};
-      acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-          *reinterpret_cast<bf16x8_t*>(&a), *reinterpret_cast<bf16x8_t*>(&b0),
-          acc, 0, 0, 0);
-      acc1 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-          *reinterpret_cast<bf16x8_t*>(&a), *reinterpret_cast<bf16x8_t*>(&b1),
-          acc1, 0, 0, 0);
+      // B: W[n][dc + kc*32 + 8lg + j] — direct global b128, four tiles
+      short8 bt[4];
+#pragma unroll
+      for (int t = 0; t < 4; t++)
+        bt[t] = ok[t]
+            ? *reinterpret_cast<const short8*>(wb[t] + dc + kc * 32 + lg * 8)
+            : short8{};
+#pragma unroll
+      for (int t = 0; t < 4; t++)
+        accs[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            *reinterpret_cast<bf16x8_t*>(&a),
+            *reinterpret_cast<bf16x8_t*>(&bt[t]), accs[t], 0, 0, 0);
     }
   }
   // D[row=token lg*4+r][col=n lr]
 #pragma unroll
-  for (int r = 0; r < 4; r++) {
-    const int mi = lg * 4 + r;
-    if (mi < m) {
-      if (ok0)
-        y[((int64_t)(r0 + mi)) * N + n0 + lr] = f32_to_bf16(acc[r]);
-      if (ok1)
-        y[((int64_t)(r0 + mi)) * N + n1 + lr] = f32_to_bf16(acc1[r]);
+  for (int t = 0; t < 4; t++)
+#pragma unroll
+    for (int r = 0; r < 4; r++) {
+      const int mi = lg * 4 + r;
+      if (mi < m && ok[t])
+        y[((int64_t)(r0 + mi)) * N + n0 + t * 16 + lr] =
+            f32_to_bf16(accs[t][r]);
     }
-  }
 }
 
 // top-k gating: softmax over E experts, renormalized top-k weights.
@@ -264,7 +265,7 @@ void moe_grouped_gemm(torch::Tensor y, torch::Tensor x, torch::Tensor w,
     return env == nullptr || env[0] != '0';
   }();
   if (use_mfma && D % 32 == 0) {
-    dim3 grid(ntiles, (N + 127) / 128);
+    dim3 grid(ntiles, (N + 255) / 256);
     moe_gemm_mfma_kernel<<<grid, kBlock, 0, stream>>>(
         (short*)y.data_ptr(), (const short*)x.data_ptr(),
         (const short*)w.data_ptr(), tiles.data_ptr<int32_t>(), nullptr,
@@ -297,7 +298,7 @@ void moe_grouped_gemm_seg(torch::Tensor y, torch::Tensor x, torch::Tensor w,
     return env == nullptr || env[0] != '0';
   }();
   if (use_mfma && D % 32 == 0) {
-    dim3 grid(E * max_mt, (N + 127) / 128);
+    dim3 grid(E * max_mt, (N + 255) / 256);
     moe_gemm_mfma_kernel<<<grid, kBlock, 0, stream>>>(
         (short*)y.data_ptr(), (const short*)x.data_ptr(),
         (const short*)w.data_ptr(), nullptr, seg_start.data_ptr<int32_t>(),
