@@ -113,6 +113,7 @@ __global__ __launch_bounds__(kBlock) void moe_gemm_kernel(
 // K chunk one v_mfma_f32_16x16x32_bf16 with B[k=8*lg+j][col=lr] =
 // W[n0+lr][kc*32+8lg+j] as a single b128 global load per lane. 4 waves
 // per block cover 64 N columns sharing the X tile.
+template <int NT>   // 16-column N-tiles per wave (grid fill vs depth)
 __global__ __launch_bounds__(kBlock) void moe_gemm_mfma_kernel(
     short* __restrict__ y,        // [T, N]
     const short* __restrict__ x,  // [T, D]
@@ -140,14 +141,16 @@ __global__ __launch_bounds__(kBlock) void moe_gemm_mfma_kernel(
   const int wid = threadIdx.x >> 6;
   const int lr = lane & 15;
   const int lg = lane >> 4;
-  // FOUR 16-column N-tiles per wave: 4 independent B loads in flight per
+  // NT 16-column N-tiles per wave: NT independent B loads in flight per
   // kc per lane (the VALU kernel's lesson: in-flight depth beats
-  // everything on the W stream), all sharing one A fragment
-  const int n0 = (blockIdx.y * 4 + wid) * 64;
-  const short* wb[4];
-  bool ok[4];
+  // everything on the W stream), all sharing one A fragment. NT=4 for
+  // wide N (gate_up), NT=2 for narrow N (down proj) to keep the grid
+  // filling 256 CUs.
+  const int n0 = (blockIdx.y * 4 + wid) * (NT * 16);
+  const short* wb[NT];
+  bool ok[NT];
 #pragma unroll
-  for (int t = 0; t < 4; t++) {
+  for (int t = 0; t < NT; t++) {
     wb[t] = w + ((int64_t)e * N + n0 + t * 16 + lr) * D;
     ok[t] = (n0 + t * 16 + lr) < N;
   }
@@ -158,7 +161,7 @@ __global__ __launch_bounds__(kBlock) void moe_gemm_mfma_kernel(
   constexpr int kXC = 512;
   __shared__ short x_lds[kMaxM * kXC];
 
-  f32x4 accs[4] = {};
+  f32x4 accs[NT] = {};
   for (int dc = 0; dc < D; dc += kXC) {
     const int cw = min(kXC, D - dc);
     __syncthreads();
@@ -180,14 +183,14 @@ __global__ __launch_bounds__(kBlock) void moe_gemm_mfma_kernel(
           (char*)x_lds + lr * (kXC * 2) +
           ((kc * 64 + lg * 16) ^ ((lr & 7) << 4)));
       // B: W[n][dc + kc*32 + 8lg + j] — direct global b128, four tiles
-      short8 bt[4];
+      short8 bt[NT];
 #pragma unroll
-      for (int t = 0; t < 4; t++)
+      for (int t = 0; t < NT; t++)
         bt[t] = ok[t]
             ? *reinterpret_cast<const short8*>(wb[t] + dc + kc * 32 + lg * 8)
             : short8{};
 #pragma unroll
-      for (int t = 0; t < 4; t++)
+      for (int t = 0; t < NT; t++)
         accs[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
             *reinterpret_cast<bf16x8_t*>(&a),
             *reinterpret_cast<bf16x8_t*>(&bt[t]), accs[t], 0, 0, 0);
@@ -195,7 +198,7 @@ __global__ __launch_bounds__(kBlock) void moe_gemm_mfma_kernel(
   }
   // D[row=token lg*4+r][col=n lr]
 #pragma unroll
-  for (int t = 0; t < 4; t++)
+  for (int t = 0; t < NT; t++)
 #pragma unroll
     for (int r = 0; r < 4; r++) {
       const int mi = lg * 4 + r;
@@ -265,11 +268,19 @@ void moe_grouped_gemm(torch::Tensor y, torch::Tensor x, torch::Tensor w,
     return env == nullptr || env[0] != '0';
   }();
   if (use_mfma && D % 32 == 0) {
-    dim3 grid(ntiles, (N + 255) / 256);
-    moe_gemm_mfma_kernel<<<grid, kBlock, 0, stream>>>(
-        (short*)y.data_ptr(), (const short*)x.data_ptr(),
-        (const short*)w.data_ptr(), tiles.data_ptr<int32_t>(), nullptr,
-        D, N, ntiles);
+    if (N >= 16384) {
+      dim3 grid(ntiles, (N + 255) / 256);
+      moe_gemm_mfma_kernel<4><<<grid, kBlock, 0, stream>>>(
+          (short*)y.data_ptr(), (const short*)x.data_ptr(),
+          (const short*)w.data_ptr(), tiles.data_ptr<int32_t>(), nullptr,
+          D, N, ntiles);
+    } else {
+      dim3 grid(ntiles, (N + 127) / 128);
+      moe_gemm_mfma_kernel<2><<<grid, kBlock, 0, stream>>>(
+          (short*)y.data_ptr(), (const short*)x.data_ptr(),
+          (const short*)w.data_ptr(), tiles.data_ptr<int32_t>(), nullptr,
+          D, N, ntiles);
+    }
   } else {
     dim3 grid(ntiles, (N + kBlock - 1) / kBlock);
     moe_gemm_kernel<<<grid, kBlock, 0, stream>>>(
@@ -298,11 +309,19 @@ void moe_grouped_gemm_seg(torch::Tensor y, torch::Tensor x, torch::Tensor w,
     return env == nullptr || env[0] != '0';
   }();
   if (use_mfma && D % 32 == 0) {
-    dim3 grid(E * max_mt, (N + 255) / 256);
-    moe_gemm_mfma_kernel<<<grid, kBlock, 0, stream>>>(
-        (short*)y.data_ptr(), (const short*)x.data_ptr(),
-        (const short*)w.data_ptr(), nullptr, seg_start.data_ptr<int32_t>(),
-        D, N, max_mt);
+    if (N >= 16384) {
+      dim3 grid(E * max_mt, (N + 255) / 256);
+      moe_gemm_mfma_kernel<4><<<grid, kBlock, 0, stream>>>(
+          (short*)y.data_ptr(), (const short*)x.data_ptr(),
+          (const short*)w.data_ptr(), nullptr, seg_start.data_ptr<int32_t>(),
+          D, N, max_mt);
+    } else {
+      dim3 grid(E * max_mt, (N + 127) / 128);
+      moe_gemm_mfma_kernel<2><<<grid, kBlock, 0, stream>>>(
+          (short*)y.data_ptr(), (const short*)x.data_ptr(),
+          (const short*)w.data_ptr(), nullptr, seg_start.data_ptr<int32_t>(),
+          D, N, max_mt);
+    }
   } else {
     dim3 grid(E * max_mt, (N + kBlock - 1) / kBlock);
     moe_gemm_kernel<<<grid, kBlock, 0, stream>>>(

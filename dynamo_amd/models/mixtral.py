@@ -99,12 +99,37 @@ class MoEMLP(torch.nn.Module):
                                          device=x.device)
             seg_start_full[1:] = torch.cumsum(counts, 0).to(torch.int32)
             if not self.ep:
-                xg = x[seg_token]
-                gu = ops.moe_grouped_gemm_seg(xg, self.w_gate_up,
-                                              seg_start_full, T * self.topk)
-                act = ops.silu_mul(gu)
-                yg = ops.moe_grouped_gemm_seg(act, self.w_down,
-                                              seg_start_full, T * self.topk)
+                import os as _os
+                use_bmm = (_os.environ.get("DYNAMO_MOE_BMM", "1") != "0"
+                           and T <= 64)
+                if use_bmm:
+                    # capacity-padded strided-batched GEMM: pad each
+                    # expert's tokens to cap=T rows and run ONE hipBLASLt
+                    # batched GEMM per projection — every expert's W
+                    # streams exactly once at dense-GEMM rates (~6 TB/s vs
+                    # ~3 for the grouped kernel). The E/topk flop padding
+                    # is free in the W-bandwidth-bound decode regime.
+                    # Fully device-side + static shapes: hipGraph-safe.
+                    cap = T
+                    slot = (torch.arange(T * self.topk, device=x.device)
+                            - seg_start_full[seg_expert].long())
+                    pad_idx = seg_expert * cap + slot
+                    xpad = torch.zeros(self.E * cap, D, dtype=x.dtype,
+                                       device=x.device)
+                    xpad[pad_idx] = x[seg_token]
+                    gu = torch.bmm(xpad.view(self.E, cap, D),
+                                   self.w_gate_up.transpose(1, 2))
+                    act = ops.silu_mul(gu.reshape(self.E * cap, -1))
+                    ypad = torch.bmm(act.view(self.E, cap, -1),
+                                     self.w_down.transpose(1, 2))
+                    yg = ypad.reshape(self.E * cap, D)[pad_idx]
+                else:
+                    xg = x[seg_token]
+                    gu = ops.moe_grouped_gemm_seg(
+                        xg, self.w_gate_up, seg_start_full, T * self.topk)
+                    act = ops.silu_mul(gu)
+                    yg = ops.moe_grouped_gemm_seg(
+                        act, self.w_down, seg_start_full, T * self.topk)
                 w = topw.reshape(-1)[order].unsqueeze(-1)
                 out.index_add_(0, seg_token, (yg * w).to(x.dtype))
                 return self.tp.all_reduce(out)
