@@ -26,6 +26,13 @@ from .service import ModelManager
 REQS = Counter("dynamo_amd_requests_total", "requests", ["model", "route"])
 TTFT = Histogram("dynamo_amd_ttft_seconds", "time to first token", ["model"])
 LATENCY = Histogram("dynamo_amd_request_seconds", "request latency", ["model"])
+# inter-token latency per streamed chunk gap (reference parity:
+# http/service/metrics.rs per-model latency depth)
+ITL = Histogram("dynamo_amd_itl_seconds", "inter-token latency", ["model"],
+                buckets=(.005, .01, .02, .03, .05, .075, .1, .15, .25, .5,
+                         1.0, 2.5))
+OUT_TOKENS = Counter("dynamo_amd_output_tokens_total", "output tokens",
+                     ["model"])
 
 
 class CompletionRequest(BaseModel):
@@ -472,14 +479,21 @@ def build_app(manager: ModelManager) -> FastAPI:
 
                 try:
                     finish = None
+                    last_t = None
                     async for chunk in _run(entry, token_ids, req, rid, _session_of(req, raw)):
                         if await raw.is_disconnected():
                             break
+                        now = time.time()
                         if first:
-                            TTFT.labels(entry.name).observe(time.time() - t0)
+                            TTFT.labels(entry.name).observe(now - t0)
                             first = False
+                        elif last_t is not None:
+                            ITL.labels(entry.name).observe(now - last_t)
+                        last_t = now
                         prev = len(produced)
                         produced.extend(chunk.get("token_ids", []))
+                        OUT_TOKENS.labels(entry.name).inc(
+                            len(produced) - prev)
                         acc += entry.tokenizer.decode_incremental(produced,
                                                                   prev)
                         finish = chunk.get("finish_reason")
