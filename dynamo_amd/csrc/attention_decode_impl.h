@@ -599,7 +599,14 @@ inline int mfma_lds_bytes(int G, int hd) {
 // FP8: the paged cache stores OCP e4m3 bytes (halves decode HBM traffic);
 // fragments convert to bf16 in-kernel via packed v_cvt (guide: decode is
 // HBM-bound, VALU converts are free under the MFMA/mem overlap).
-template <int DEFER = 1, int PRIO = 1, int KPF = 0, int FP8 = 0>
+// VS: V^T LDS row stride in BYTES. 80 (the original) makes the write
+// pattern collapse: rows step 8 apart start at the same bank
+// (8*20 banks % 32 == 0), so each b64 staging write is ~16-way
+// conflicted (PMC: 2.3e9 SQ_LDS_BANK_CONFLICT). An 88B stride gives
+// rows-step-8 a 16-bank offset (8*22 % 32 == 16), halving write
+// conflicts; 72 ditto with less LDS.
+template <int DEFER = 1, int PRIO = 1, int KPF = 0, int FP8 = 0,
+          int VS = 80>
 __global__ __launch_bounds__(kBlock) void paged_decode_mfma_swapped(
     float* __restrict__ partial, float* __restrict__ ml,
     short* __restrict__ out, const short* __restrict__ q,
@@ -625,7 +632,7 @@ __global__ __launch_bounds__(kBlock) void paged_decode_mfma_swapped(
   extern __shared__ float lds[];
   float* merge = lds;                                   // 4*G*(hd+2)
   short* q_lds_s = reinterpret_cast<short*>(merge + 4 * G * (hd + 2));
-  short* v_lds = q_lds_s + G * hd + wid * 5120;         // per-wave 10KB
+  short* v_lds = q_lds_s + G * hd + wid * (hd * VS / 2);  // per-wave
 
   if (chunk_start >= ctx) {
     if (C > 1) {
@@ -722,7 +729,7 @@ __global__ __launch_bounds__(kBlock) void paged_decode_mfma_swapped(
         const int d = d8 * 8 + i;
         short4v pk = {vf[u * 4 + 0][i], vf[u * 4 + 1][i],
                       vf[u * 4 + 2][i], vf[u * 4 + 3][i]};
-        *reinterpret_cast<short4v*>((char*)v_lds + d * 80 + tg * 8) = pk;
+        *reinterpret_cast<short4v*>((char*)v_lds + d * VS + tg * 8) = pk;
       }
     }
   };
@@ -831,7 +838,7 @@ __global__ __launch_bounds__(kBlock) void paged_decode_mfma_swapped(
 #pragma unroll
       for (int da = 0; da < 8; da++) {
         short8 va_s = *reinterpret_cast<const short8*>(
-            (const char*)v_lds + (da * 16 + lr) * 80 + lg * 16);
+            (const char*)v_lds + (da * 16 + lr) * VS + lg * 16);
         acc[da] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
             *reinterpret_cast<bf16x8_t*>(&va_s), p_frag, acc[da], 0, 0, 0);
       }
@@ -888,8 +895,8 @@ __global__ __launch_bounds__(kBlock) void paged_decode_mfma_swapped(
   }
 }
 
-inline int mfma_swapped_lds_bytes(int G, int hd) {
-  return 4 * G * (hd + 2) * 4 + G * hd * 2 + 4 * 5120 * 2;  // no P tile
+inline int mfma_swapped_lds_bytes(int G, int hd, int vs = 80) {
+  return 4 * G * (hd + 2) * 4 + G * hd * 2 + 4 * hd * vs;  // no P tile
 }
 
 // Phase 2: merge chunk partials. grid (B, Hq), block = 128.
