@@ -133,7 +133,7 @@ __device__ __forceinline__ float xor32_swap(float x, int hi) {
 //   per write) — 4x fewer LDS write ops than scalar V^T stores.
 // DEFER: skip the O-rescale when the tile max is within 8 of the running
 //   max (P bounded by e^8; guide "defer-max RESCALE_THRESHOLD").
-template <int GW, int ASTAGE, int VSTAGE = 0, int DEFER = 0, int PRIO = 0>
+template <int GW, int ASTAGE, int VSTAGE = 0, int DEFER = 0, int PRIO = 0, int VST = 128>
 __global__ __launch_bounds__(GW * 64) void prefill32_kernel(
     short* __restrict__ out,        // [B, S, Hq, 128]
     const short* __restrict__ q,    // [B, S, Hq, 128]
@@ -150,9 +150,9 @@ __global__ __launch_bounds__(GW * 64) void prefill32_kernel(
   const int lo = l & 31, hi = l >> 5;
 
   extern __shared__ char lds[];
-  // buffer 0 (and, for ASTAGE==2 double-buffering, buffer 1 at +32 KB)
+  constexpr int kHalfB = KVBLK * 256 + 128 * VST;
   short* k_lds = (short*)lds;                    // [64][128] swizzled
-  short* v_lds = (short*)(lds + KVBLK * 256);    // [128][64] swizzled (V^T)
+  short* v_lds = (short*)(lds + KVBLK * 256);    // [128][VST/2] (V^T)
 
   // Q B-fragments: qreg[ds] holds Q[q0+lo][ds*16 + 8*hi + i]
   const short* qrow = q + (((int64_t)bb * S + q0 + lo) * Hq + h) * 128;
@@ -205,7 +205,7 @@ __global__ __launch_bounds__(GW * 64) void prefill32_kernel(
     }
   };
   auto store_tile = [&](int buf) {
-    char* kl = (char*)lds + buf * (KVBLK * 256 + 128 * 128);
+    char* kl = (char*)lds + buf * kHalfB;
     char* vl = kl + KVBLK * 256;
     if constexpr (VSTAGE) {
       if (vrole) {
@@ -216,7 +216,7 @@ __global__ __launch_bounds__(GW * 64) void prefill32_kernel(
         for (int i = 0; i < 8; i++) {
           const int d = d0 + i;
           short4_t pk = {kreg[0][i], kreg[1][i], kreg[2][i], kreg[3][i]};
-          *(short4_t*)(vl + d * 128 + ((row0 * 2) ^ ((d & 7) << 4))) = pk;
+          *(short4_t*)(vl + d * VST + ((row0 * 2) ^ ((d & 7) << 4))) = pk;
         }
       } else {
         const int idx = threadIdx.x - 256;
@@ -239,7 +239,7 @@ __global__ __launch_bounds__(GW * 64) void prefill32_kernel(
 #pragma unroll
       for (int i = 0; i < 8; i++) {
         const int d = col8 + i;
-        *(short*)(vl + d * 128 + ((row * 2) ^ ((d & 7) << 4))) = vreg[u][i];
+        *(short*)(vl + d * VST + ((row * 2) ^ ((d & 7) << 4))) = vreg[u][i];
       }
     }
   };
@@ -362,7 +362,7 @@ __global__ __launch_bounds__(GW * 64) void prefill32_kernel(
       for (int ks = 0; ks < 4; ks++) {
         const int toff = (ks * 16 + 8 * hi) * 2;
         const bf16x8 vb = *reinterpret_cast<const bf16x8*>(
-            (char*)v_lds + drow * 128 + (toff ^ ((drow & 7) << 4)));
+            (char*)v_lds + drow * VST + (toff ^ ((drow & 7) << 4)));
         o[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa[ks], vb, o[dt], 0, 0, 0);
       }
     }
@@ -635,7 +635,7 @@ static void cpu_ref(std::vector<float>& o, const std::vector<short>& q,
     }
 }
 
-template <int GW, int ASTAGE, int VSTAGE = 0, int DEFER = 0, int PRIO = 0>
+template <int GW, int ASTAGE, int VSTAGE = 0, int DEFER = 0, int PRIO = 0, int VST = 128>
 static void run(Cfg c, bool check, int iters) {
   const int D = 128;
   const float scale = 1.f / sqrtf((float)D);
@@ -652,10 +652,15 @@ static void run(Cfg c, bool check, int iters) {
   CK(hipMemcpy(dk, hk.data(), ke * 2, hipMemcpyHostToDevice));
   CK(hipMemcpy(dv, hv.data(), ke * 2, hipMemcpyHostToDevice));
   dim3 grid(c.S / QBLK, c.Hkv, c.B);
-  int lds = KVBLK * 256 + 128 * 128;          // K + V^T = 32 KB
+  int lds = KVBLK * 256 + 128 * VST;
   if (ASTAGE == 2) lds *= 2;                  // double-buffered
+  if (lds > 65536)
+    (void)hipFuncSetAttribute(
+        reinterpret_cast<const void*>(
+            &prefill32_kernel<GW, ASTAGE, VSTAGE, DEFER, PRIO, VST>),
+        hipFuncAttributeMaxDynamicSharedMemorySize, lds);
   auto launch = [&] {
-    prefill32_kernel<GW, ASTAGE, VSTAGE, DEFER, PRIO><<<grid, GW * 64, lds>>>(
+    prefill32_kernel<GW, ASTAGE, VSTAGE, DEFER, PRIO, VST><<<grid, GW * 64, lds>>>(
         dout, dq, dk, dv, c.S, c.Hq, c.Hkv, scale);
   };
   launch();
@@ -687,8 +692,8 @@ static void run(Cfg c, bool check, int iters) {
     float ms; hipEventElapsedTime(&ms, e0, e1);
     ms /= iters;
     const double fl = 2.0 * c.B * c.Hq * (double)c.S * c.S * D;
-    printf("bench GW=%d AS=%d VS=%d DF=%d PR=%d B=%d S=%d Hq=%d Hkv=%d: %.3f ms  %.1f TF\n",
-           GW, ASTAGE, VSTAGE, DEFER, PRIO, c.B, c.S, c.Hq, c.Hkv, ms,
+    printf("bench GW=%d AS=%d VS=%d DF=%d PR=%d VT=%d B=%d S=%d Hq=%d Hkv=%d: %.3f ms  %.1f TF\n",
+           GW, ASTAGE, VSTAGE, DEFER, PRIO, VST, c.B, c.S, c.Hq, c.Hkv, ms,
            fl / (ms * 1e-3) / 1e12);
   }
   hipFree(dq); hipFree(dk); hipFree(dv); hipFree(dout);

@@ -66,14 +66,14 @@ void paged_attention_decode(torch::Tensor out, torch::Tensor q,
 #define LAUNCH_MFMA(GG)                                                       \
   do {                                                                        \
     if (use_swapped) {                                                        \
-      if (mfma_swapped_lds_bytes(GG, hd) > 65536)                             \
+      if (mfma_swapped_lds_bytes(GG, hd, 72) > 65536)                             \
         (void)hipFuncSetAttribute(                                            \
             reinterpret_cast<const void*>(                                    \
-                &paged_decode_mfma_swapped<1, 1, 0>),                         \
+                &paged_decode_mfma_swapped<1, 1, 0, 0, 72>),                         \
             hipFuncAttributeMaxDynamicSharedMemorySize,                       \
-            mfma_swapped_lds_bytes(GG, hd));                                  \
-      paged_decode_mfma_swapped<1, 1, 0>                                      \
-          <<<grid, kBlock, mfma_swapped_lds_bytes(GG, hd), stream>>>(         \
+            mfma_swapped_lds_bytes(GG, hd, 72));                                  \
+      paged_decode_mfma_swapped<1, 1, 0, 0, 72>                                      \
+          <<<grid, kBlock, mfma_swapped_lds_bytes(GG, hd, 72), stream>>>(         \
           partial.data_ptr<float>(), ml.data_ptr<float>(),                    \
           (short*)out.data_ptr(), (const short*)q.data_ptr(),                 \
           (const short*)kcache.data_ptr(),                                    \
@@ -99,13 +99,13 @@ void paged_attention_decode(torch::Tensor out, torch::Tensor q,
     // fp8 KV: the runtime-G swapped kernel is the ONLY fp8 consumer
     TORCH_CHECK(ps % 32 == 0 && hd == 128 && G <= 16,
                 "fp8 KV cache needs page_size%32==0, head_dim==128, G<=16");
-    if (mfma_swapped_lds_bytes(G, hd) > 65536)
+    if (mfma_swapped_lds_bytes(G, hd, 72) > 65536)
       (void)hipFuncSetAttribute(
-          reinterpret_cast<const void*>(&paged_decode_mfma_swapped<1, 1, 0, 1>),
+          reinterpret_cast<const void*>(&paged_decode_mfma_swapped<1, 1, 0, 1, 72>),
           hipFuncAttributeMaxDynamicSharedMemorySize,
-          mfma_swapped_lds_bytes(G, hd));
-    paged_decode_mfma_swapped<1, 1, 0, 1>
-        <<<grid, kBlock, mfma_swapped_lds_bytes(G, hd), stream>>>(
+          mfma_swapped_lds_bytes(G, hd, 72));
+    paged_decode_mfma_swapped<1, 1, 0, 1, 72>
+        <<<grid, kBlock, mfma_swapped_lds_bytes(G, hd, 72), stream>>>(
         partial.data_ptr<float>(), ml.data_ptr<float>(),
         (short*)out.data_ptr(), (const short*)q.data_ptr(),
         (const short*)kcache.data_ptr(), (const short*)vcache.data_ptr(),
