@@ -759,18 +759,12 @@ static void run_pipe(Cfg c, bool check, int iters) {
 int main() {
   if (!probe32()) return 1;
   permprobe();
-  run_pipe<8, 1, 0>({1, 256, 16, 2}, true, 0);
-  run_pipe<8, 1, 0>({1, 8192, 64, 8}, false, 20);
-  run_pipe<8, 1, 1>({1, 8192, 64, 8}, false, 20);
-  run_pipe<8, 1, 1>({2, 4096, 64, 8}, false, 20);
-  run<8, 2, 1, 1>({1, 256, 16, 2}, true, 0);
-  run<4, 2>({1, 256, 8, 2}, true, 0);
-  run<8, 2, 1, 1, 1>({1, 256, 16, 2}, true, 0);
-  run<8, 1, 1, 1>({1, 8192, 64, 8}, false, 20);
+  run<8, 2, 1, 1, 0, 136>({1, 256, 16, 2}, true, 0);
+  run<8, 2, 1, 1, 0, 144>({1, 256, 16, 2}, true, 0);
   run<8, 2, 1, 1>({1, 8192, 64, 8}, false, 20);
-  run<8, 1, 1, 1, 1>({1, 8192, 64, 8}, false, 20);
-  run<8, 2, 1, 1, 1>({1, 8192, 64, 8}, false, 20);
-  run<8, 2, 1, 1, 1>({2, 4096, 64, 8}, false, 20);
-  run<4, 2>({2, 4096, 32, 8}, false, 20);
+  run<8, 2, 1, 1, 0, 136>({1, 8192, 64, 8}, false, 20);
+  run<8, 2, 1, 1, 0, 144>({1, 8192, 64, 8}, false, 20);
+  run<8, 1, 1, 1, 0, 136>({1, 8192, 64, 8}, false, 20);
+  run<8, 2, 1, 1, 0, 136>({2, 4096, 64, 8}, false, 20);
   return 0;
 }
