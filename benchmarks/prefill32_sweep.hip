@@ -255,7 +255,7 @@ __global__ __launch_bounds__(GW * 64) void prefill32_kernel(
   for (int t0 = 0; t0 < kv_end; t0 += KVBLK) {
     if constexpr (ASTAGE == 2) {
       const int cur = (t0 / KVBLK) & 1;
-      k_lds = (short*)((char*)lds + cur * (KVBLK * 256 + 128 * 128));
+      k_lds = (short*)((char*)lds + cur * (kHalfB));
       v_lds = (short*)((char*)k_lds + KVBLK * 256);
     } else {
       __syncthreads();
