@@ -30,6 +30,12 @@ constexpr int kKvTile = 64;  // kv tokens per LDS tile
 constexpr float kNegInf = -1e30f;
 
 DEVINL int swz(int byte_in_row, int row) { return byte_in_row ^ ((row & 7) << 4); }
+// V^T variant: fold row>>3 into the key (write instructions hold row&7
+// constant across their 16 row-slices — see the XK note in the 32x32
+// kernel below)
+DEVINL int swzv(int byte_in_row, int row) {
+  return byte_in_row ^ (((row ^ (row >> 3)) & 7) << 4);
+}
 
 __global__ __launch_bounds__(kBlock) void prefill_kernel(
     short* __restrict__ out,            // [Tq, Hq, 128]
@@ -112,7 +118,7 @@ __global__ __launch_bounds__(kBlock) void prefill_kernel(
 #pragma unroll
       for (int e = 0; e < 8; e++) {
         const int dim = col8 * 8 + e;
-        *(short*)((char*)vt_lds + dim * 128 + swz(row * 2, dim)) = kv_v[e];
+        *(short*)((char*)vt_lds + dim * 128 + swzv(row * 2, dim)) = kv_v[e];
       }
     }
     __syncthreads();
@@ -211,7 +217,7 @@ __global__ __launch_bounds__(kBlock) void prefill_kernel(
         // B-frag: V[tok = kt*32 + lg*8 + j][dim = d*16 + lr] = V^T[dim][tok]
         const int dim = d * 16 + lr;
         short8 vb = *reinterpret_cast<const short8*>(
-            (char*)vt_lds + dim * 128 + swz(kt * 64 + lg * 16, dim));
+            (char*)vt_lds + dim * 128 + swzv(kt * 64 + lg * 16, dim));
         acc_o[d] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
             *reinterpret_cast<bf16x8_t*>(&pa), *reinterpret_cast<bf16x8_t*>(&vb),
             acc_o[d], 0, 0, 0);
