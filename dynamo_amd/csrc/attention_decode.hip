@@ -69,10 +69,10 @@ void paged_attention_decode(torch::Tensor out, torch::Tensor q,
       if (mfma_swapped_lds_bytes(GG, hd, 72) > 65536)                             \
         (void)hipFuncSetAttribute(                                            \
             reinterpret_cast<const void*>(                                    \
-                &paged_decode_mfma_swapped<1, 1, 0, 0, 72>),                         \
+                &paged_decode_mfma_swapped<1, 1, 0, 0, 72, 1>),                         \
             hipFuncAttributeMaxDynamicSharedMemorySize,                       \
             mfma_swapped_lds_bytes(GG, hd, 72));                                  \
-      paged_decode_mfma_swapped<1, 1, 0, 0, 72>                                      \
+      paged_decode_mfma_swapped<1, 1, 0, 0, 72, 1>                                      \
           <<<grid, kBlock, mfma_swapped_lds_bytes(GG, hd, 72), stream>>>(         \
           partial.data_ptr<float>(), ml.data_ptr<float>(),                    \
           (short*)out.data_ptr(), (const short*)q.data_ptr(),                 \
@@ -101,10 +101,10 @@ void paged_attention_decode(torch::Tensor out, torch::Tensor q,
                 "fp8 KV cache needs page_size%32==0, head_dim==128, G<=16");
     if (mfma_swapped_lds_bytes(G, hd, 72) > 65536)
       (void)hipFuncSetAttribute(
-          reinterpret_cast<const void*>(&paged_decode_mfma_swapped<1, 1, 0, 1, 72>),
+          reinterpret_cast<const void*>(&paged_decode_mfma_swapped<1, 1, 0, 1, 72, 1>),
           hipFuncAttributeMaxDynamicSharedMemorySize,
           mfma_swapped_lds_bytes(G, hd, 72));
-    paged_decode_mfma_swapped<1, 1, 0, 1, 72>
+    paged_decode_mfma_swapped<1, 1, 0, 1, 72, 1>
         <<<grid, kBlock, mfma_swapped_lds_bytes(G, hd, 72), stream>>>(
         partial.data_ptr<float>(), ml.data_ptr<float>(),
         (short*)out.data_ptr(), (const short*)q.data_ptr(),

@@ -605,8 +605,13 @@ inline int mfma_lds_bytes(int G, int hd) {
 // conflicted (PMC: 2.3e9 SQ_LDS_BANK_CONFLICT). An 88B stride gives
 // rows-step-8 a 16-bank offset (8*22 % 32 == 16), halving write
 // conflicts; 72 ditto with less LDS.
+// XK2: XOR the token-group slot with bit d>>4 (<<5 bytes): per write
+// instruction the (d step 8) row-groups collapse to 2 bank starts; the
+// extra key doubles the distinct byte-slots (8 -> 16, i.e. ~4-way
+// writes). Reads shift whole 16B pairs (key constant per row), so the
+// b128 PV reads stay aligned.
 template <int DEFER = 1, int PRIO = 1, int KPF = 0, int FP8 = 0,
-          int VS = 80>
+          int VS = 80, int XK2 = 0>
 __global__ __launch_bounds__(kBlock) void paged_decode_mfma_swapped(
     float* __restrict__ partial, float* __restrict__ ml,
     short* __restrict__ out, const short* __restrict__ q,
@@ -729,7 +734,8 @@ __global__ __launch_bounds__(kBlock) void paged_decode_mfma_swapped(
         const int d = d8 * 8 + i;
         short4v pk = {vf[u * 4 + 0][i], vf[u * 4 + 1][i],
                       vf[u * 4 + 2][i], vf[u * 4 + 3][i]};
-        *reinterpret_cast<short4v*>((char*)v_lds + d * VS + tg * 8) = pk;
+        const int toff = XK2 ? ((tg * 8) ^ (((d >> 4) & 1) << 5)) : (tg * 8);
+        *reinterpret_cast<short4v*>((char*)v_lds + d * VS + toff) = pk;
       }
     }
   };
@@ -837,8 +843,11 @@ __global__ __launch_bounds__(kBlock) void paged_decode_mfma_swapped(
       if constexpr (PRIO) __builtin_amdgcn_s_setprio(1);
 #pragma unroll
       for (int da = 0; da < 8; da++) {
+        const int vrow = da * 16 + lr;
+        const int vtoff = XK2 ? ((lg * 16) ^ (((vrow >> 4) & 1) << 5))
+                              : (lg * 16);
         short8 va_s = *reinterpret_cast<const short8*>(
-            (const char*)v_lds + (da * 16 + lr) * VS + lg * 16);
+            (const char*)v_lds + vrow * VS + vtoff);
         acc[da] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
             *reinterpret_cast<bf16x8_t*>(&va_s), p_frag, acc[da], 0, 0, 0);
       }
