@@ -39,7 +39,7 @@ static void perm16_probe() {
 }
 
 struct Bufs {
-  short *kc, *vc, *q, *out;
+  short *kc, *vc, *vct, *q, *out;
   float *partial, *ml;
   int32_t *pt, *ctx;
   int npages, C;
@@ -80,23 +80,40 @@ static Bufs make(int G) {
   fill(bf.kc, cache_e);
   fill(bf.vc, cache_e);
   fill(bf.q, (size_t)B * Hq * HD);
+  // transposed V cache (VT layout): per (page, head) block,
+  // vt[d * PS + t] = v[t * HD + d]
+  CK(hipMalloc(&bf.vct, cache_e * 2));
+  {
+    std::vector<short> hv(cache_e), ht(cache_e);
+    CK(hipMemcpy(hv.data(), bf.vc, cache_e * 2, hipMemcpyDeviceToHost));
+    const size_t nblk = (size_t)bf.npages * Hkv;
+    for (size_t blk = 0; blk < nblk; blk++) {
+      const size_t base = blk * PS * HD;
+      for (int t = 0; t < PS; t++)
+        for (int d = 0; d < HD; d++)
+          ht[base + (size_t)d * PS + t] = hv[base + (size_t)t * HD + d];
+    }
+    CK(hipMemcpy(bf.vct, ht.data(), cache_e * 2, hipMemcpyHostToDevice));
+  }
   return bf;
 }
 
 // swapped-operand MFMA variant: bench + elementwise compare vs the proven
 // paged_decode_mfma output
-template <int G, int DEFER = 1, int PRIO = 1, int KPF = 0, int VS = 80, int XK2 = 0>
+template <int G, int DEFER = 1, int PRIO = 1, int KPF = 0, int VS = 80, int XK2 = 0,
+          int VT = 0>
 static void run_mfma_swapped(const Bufs& bf, bool check) {
   dim3 grid(B, Hkv, bf.C);
-  const int lds = mfma_swapped_lds_bytes(G, HD, VS);
+  const int lds = VT ? mfma_swapped_vt_lds_bytes(G, HD)
+                     : mfma_swapped_lds_bytes(G, HD, VS);
   const int iters = 30;
   if (lds > 65536)
     (void)hipFuncSetAttribute(
-        reinterpret_cast<const void*>(&paged_decode_mfma_swapped<DEFER, PRIO, KPF, 0, VS, XK2>),
+        reinterpret_cast<const void*>(&paged_decode_mfma_swapped<DEFER, PRIO, KPF, 0, VS, XK2, VT>),
         hipFuncAttributeMaxDynamicSharedMemorySize, lds);
   auto launch = [&] {
-    paged_decode_mfma_swapped<DEFER, PRIO, KPF, 0, VS, XK2><<<grid, kBlock, lds>>>(
-        bf.partial, bf.ml, bf.out, bf.q, bf.kc, bf.vc, bf.pt, bf.ctx,
+    paged_decode_mfma_swapped<DEFER, PRIO, KPF, 0, VS, XK2, VT><<<grid, kBlock, lds>>>(
+        bf.partial, bf.ml, bf.out, bf.q, bf.kc, VT ? bf.vct : bf.vc, bf.pt, bf.ctx,
         0.0883883f, kChunk, G, B, Hkv, bf.C, CTX / PS, 6, HD);
     paged_decode_phase2<<<dim3(B, G * Hkv), 128>>>(
         bf.out, bf.partial, bf.ml, bf.ctx, kChunk, G * Hkv, bf.C, HD);
@@ -148,8 +165,8 @@ static void run_mfma_swapped(const Bufs& bf, bool check) {
   CK(hipEventElapsedTime(&ms, e0, e1));
   double t = ms / 1000.0 / iters;
   double gb = 2.0 * B * CTX * Hkv * HD * 2 / 1e9;
-  printf("G%d MFMA_SW DF%d PR%d KP%d VS%d X%d %8.1f us  %7.0f GB/s\n", G, DEFER, PRIO, KPF, VS, XK2,
-         t * 1e6, gb / t);
+  printf("G%d MFMA_SW DF%d PR%d KP%d VS%d X%d VT%d %8.1f us  %7.0f GB/s\n", G,
+         DEFER, PRIO, KPF, VS, XK2, VT, t * 1e6, gb / t);
   fflush(stdout);
 }
 
@@ -225,9 +242,13 @@ int main() {
   {
     Bufs bf = make(8);
     run_mfma_swapped<8, 0, 0, 0, 72, 1>(bf, true);
-    run_mfma_swapped<8, 1, 1, 0, 72>(bf, false);
+    run_mfma_swapped<8, 0, 0, 0, 72, 0, 1>(bf, true);   // VT check
     run_mfma_swapped<8, 1, 1, 0, 72, 1>(bf, false);
-    run_mfma_swapped<8, 1, 1, 0, 88, 1>(bf, false);
+    run_mfma_swapped<8, 1, 1, 0, 72, 0, 1>(bf, false);  // VT
+    run_mfma_swapped<8, 1, 0, 0, 72, 0, 1>(bf, false);  // VT no-prio
+    run_mfma_swapped<8, 0, 0, 0, 72, 0, 2>(bf, true);   // VT2 check
+    run_mfma_swapped<8, 1, 1, 0, 72, 0, 2>(bf, false);  // VT2 pipelined
+    run_mfma_swapped<8, 1, 0, 0, 72, 0, 2>(bf, false);  // VT2 no-prio
     run<8, 16, 4, 2>(bf, "");
     run<8, 16, 4, 3>(bf, "");
     run<8, 16, 4, 4>(bf, "");
@@ -254,6 +275,10 @@ int main() {
     run<4, 8, 1, 2>(bf, "");
     run_mfma<4, 1, 0>(bf);
     run_mfma_swapped<4, 1, 1, 0, 88>(bf, false);
+    run_mfma_swapped<4, 0, 0, 0, 72, 0, 2>(bf, true);   // G4 VT2 check
+    run_mfma_swapped<4, 1, 1, 0, 72, 0, 2>(bf, false);  // G4 VT2
+    run_mfma_swapped<2, 0, 0, 0, 72, 0, 2>(bf, true);   // G2 VT2 check
+    run_mfma_swapped<2, 1, 1, 0, 72, 0, 2>(bf, false);  // G2 VT2
   }
   {
     Bufs bf = make(1);

@@ -21,7 +21,8 @@ void paged_attention_decode(torch::Tensor out, torch::Tensor q,
                             torch::Tensor kcache, torch::Tensor vcache,
                             torch::Tensor page_table, torch::Tensor ctx_lens,
                             torch::Tensor partial, torch::Tensor ml,
-                            double scale, int64_t chunk_tokens) {
+                            double scale, int64_t chunk_tokens,
+                            bool v_transposed) {
   TORCH_CHECK(q.is_cuda() && q.dtype() == torch::kBFloat16);
   TORCH_CHECK(page_table.dtype() == torch::kInt32 && ctx_lens.dtype() == torch::kInt32);
   const bool fp8 = kcache.dtype() == torch::kFloat8_e4m3fn;
@@ -42,8 +43,44 @@ void paged_attention_decode(torch::Tensor out, torch::Tensor q,
   const int chunk = (int)chunk_tokens;
   TORCH_CHECK(chunk >= 128 && chunk % 128 == 0, "bad chunk_tokens ", chunk);
   auto stream = at::cuda::getCurrentHIPStream();
-
   dim3 grid(B, Hkv, C);
+
+  if (v_transposed) {
+    // d-major V pages: the software-pipelined swapped kernel (VT=2) is the
+    // only consumer — sweep: G8 4476, G4 4987, G2 5069 GB/s vs 3906-3951
+    // for the token-major paths (benchmarks/decode_sweep.hip)
+    TORCH_CHECK(vcache.size(2) == hd && vcache.size(3) == ps,
+                "v_transposed expects vcache [P, Hkv, hd, ps]");
+    TORCH_CHECK(ps % 32 == 0 && hd == 128 && G >= 2 && G <= 16,
+                "v_transposed decode needs page_size%32==0, head_dim==128, "
+                "2<=G<=16 (got G=", G, ")");
+    const int lds = mfma_swapped_vt_lds_bytes(G, hd);
+    auto launch_vt = [&](auto* kern) {
+      if (lds > 65536)
+        (void)hipFuncSetAttribute(reinterpret_cast<const void*>(kern),
+                                  hipFuncAttributeMaxDynamicSharedMemorySize,
+                                  lds);
+      kern<<<grid, kBlock, lds, stream>>>(
+          partial.data_ptr<float>(), ml.data_ptr<float>(),
+          (short*)out.data_ptr(), (const short*)q.data_ptr(),
+          (const short*)kcache.data_ptr(), (const short*)vcache.data_ptr(),
+          page_table.data_ptr<int32_t>(), ctx_lens.data_ptr<int32_t>(),
+          (float)scale, chunk, G, B, Hkv, C, max_pages, log2_ps, hd);
+    };
+    if (fp8) launch_vt(&paged_decode_mfma_swapped<1, 1, 0, 1, 72, 0, 2>);
+    else launch_vt(&paged_decode_mfma_swapped<1, 1, 0, 0, 72, 0, 2>);
+    HIP_CHECK_KERNEL();
+    if (C > 1) {
+      dim3 grid2(B, Hq);
+      paged_decode_phase2<<<grid2, 128, 0, stream>>>(
+          (short*)out.data_ptr(), partial.data_ptr<float>(),
+          ml.data_ptr<float>(), ctx_lens.data_ptr<int32_t>(), chunk, Hq, C,
+          hd);
+      HIP_CHECK_KERNEL();
+    }
+    return;
+  }
+
 #define LAUNCH_G(GG, DP, HS, DEPTH)                                           \
   paged_decode_phase1<GG, DP, HS, DEPTH>                                      \
       <<<grid, kBlock, phase1_lds_bytes(GG, HS, hd), stream>>>(               \

@@ -610,8 +610,17 @@ inline int mfma_lds_bytes(int G, int hd) {
 // extra key doubles the distinct byte-slots (8 -> 16, i.e. ~4-way
 // writes). Reads shift whole 16B pairs (key constant per row), so the
 // b128 PV reads stay aligned.
+// VT: the V cache pages are stored TRANSPOSED (d-major: elem offset
+// d * page_size + token_in_page instead of token_in_page * hd + d). The
+// PV A-fragment (V^T[dim][8 consecutive tokens]) is then ONE contiguous
+// b128 global load per da - the whole load_v/store_v/ds_read staging
+// pipeline (16 ds_writes + ~64 VALU packs + 8 ds_reads per tile, the
+// kernel's VALU bottleneck: PMC VALU:MFMA was 28:1) disappears. The
+// write side pays ~32x DRAM amplification ONLY for single-token decode
+// appends (~10 MB/step vs 22 GB/step of V reads); prefill appends cover
+// whole 64B lines in L2 before eviction.
 template <int DEFER = 1, int PRIO = 1, int KPF = 0, int FP8 = 0,
-          int VS = 80, int XK2 = 0>
+          int VS = 80, int XK2 = 0, int VT = 0>
 __global__ __launch_bounds__(kBlock) void paged_decode_mfma_swapped(
     float* __restrict__ partial, float* __restrict__ ml,
     short* __restrict__ out, const short* __restrict__ q,
@@ -723,6 +732,25 @@ __global__ __launch_bounds__(kBlock) void paged_decode_mfma_swapped(
       }
     }
   };
+  auto load_vt = [&](int t0_, short8 (&vf)[8]) {
+    const int64_t pb = (((int64_t)pt[t0_ >> log2_ps] * Hkv + h) * ps) * hd;
+    const int tin = t0_ & (ps - 1);
+#pragma unroll
+    for (int da = 0; da < 8; da++) {
+      const int vrow = da * 16 + lr;
+      vf[da] = ld8(vcache, pb + (int64_t)vrow * ps + tin + lg * 8);
+    }
+    // boundary tile: zero tokens past slab_end (page tails can hold
+    // stale NaN/Inf bits; P=0 alone does not mask NaN * 0)
+    const int rem = slab_end - t0_;
+    if (rem < 32) {
+#pragma unroll
+      for (int da = 0; da < 8; da++)
+#pragma unroll
+        for (int j = 0; j < 8; j++)
+          if (lg * 8 + j >= rem) vf[da][j] = 0;
+    }
+  };
   auto store_v = [&](const short8 (&vf)[8]) {
 #pragma unroll
     for (int u = 0; u < 2; u++) {
@@ -746,6 +774,104 @@ __global__ __launch_bounds__(kBlock) void paged_decode_mfma_swapped(
     load_v(slab_start, vfr[0]);
   }
 
+  // VT-only tile compute (QK -> softmax -> P pack -> PV from registers):
+  // a specialization of the main-loop body below; keep the two in sync.
+  auto tile_vt = [&](int t0, const short8 (&kf)[8], const short8 (&vf)[8]) {
+    f32x4 sA{0.f, 0.f, 0.f, 0.f}, sB{0.f, 0.f, 0.f, 0.f};
+    if constexpr (PRIO) __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int kc = 0; kc < 4; kc++) {
+      short8 ka = kf[kc];
+      sA = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+          *reinterpret_cast<bf16x8_t*>(&ka), q_frag[kc], sA, 0, 0, 0);
+    }
+    if (t0 + 16 < slab_end) {
+#pragma unroll
+      for (int kc = 0; kc < 4; kc++) {
+        short8 kb2 = kf[4 + kc];
+        sB = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            *reinterpret_cast<bf16x8_t*>(&kb2), q_frag[kc], sB, 0, 0, 0);
+      }
+    }
+    if constexpr (PRIO) __builtin_amdgcn_s_setprio(0);
+    float pA[4], pB[4];
+    float mt = kNegInf;
+#pragma unroll
+    for (int r = 0; r < 4; r++) {
+      const int tokA = t0 + lg * 4 + r, tokB = t0 + 16 + lg * 4 + r;
+      pA[r] = (tokA < slab_end) ? sA[r] * scale : kNegInf;
+      pB[r] = (tokB < slab_end) ? sB[r] * scale : kNegInf;
+      mt = fmaxf(mt, fmaxf(pA[r], pB[r]));
+    }
+    mt = fmaxf(mt, __shfl_xor(mt, 16, WAVE_SIZE));
+    mt = fmaxf(mt, __shfl_xor(mt, 32, WAVE_SIZE));
+    const float thr = DEFER ? 8.0f : 0.0f;
+    if (mt > m_run + thr) {
+      const float corr = (m_run <= kNegInf * 0.5f) ? 0.f : __expf(m_run - mt);
+      l_run *= corr;
+#pragma unroll
+      for (int d = 0; d < 8; d++) acc[d] *= corr;
+      m_run = mt;
+    }
+    float psum = 0.f;
+#pragma unroll
+    for (int r = 0; r < 4; r++) {
+      pA[r] = (pA[r] > kNegInf * 0.5f) ? __expf(pA[r] - m_run) : 0.f;
+      pB[r] = (pB[r] > kNegInf * 0.5f) ? __expf(pB[r] - m_run) : 0.f;
+      psum += pA[r] + pB[r];
+    }
+    psum += __shfl_xor(psum, 16, WAVE_SIZE);
+    psum += __shfl_xor(psum, 32, WAVE_SIZE);
+    l_run += psum;
+    auto cvtpk = [](float a, float bb) {
+      unsigned int r;
+      asm("v_cvt_pk_bf16_f32 %0, %1, %2" : "=v"(r) : "v"(a), "v"(bb));
+      return r;
+    };
+    const unsigned int A0 = cvtpk(pA[0], pA[1]), A1 = cvtpk(pA[2], pA[3]);
+    const unsigned int B0 = cvtpk(pB[0], pB[1]), B1 = cvtpk(pB[2], pB[3]);
+    uint2_t s1 = __builtin_amdgcn_permlane32_swap(A0, B0, false, false);
+    uint2_t s2 = __builtin_amdgcn_permlane32_swap(A1, B1, false, false);
+    uint2_t f02 = __builtin_amdgcn_permlane16_swap(s1.x, s1.y, false, false);
+    uint2_t f13 = __builtin_amdgcn_permlane16_swap(s2.x, s2.y, false, false);
+    unsigned int w[4] = {f02.x, f13.x, f02.y, f13.y};
+    bf16x8_t p_frag = *reinterpret_cast<bf16x8_t*>(w);
+    if constexpr (PRIO) __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int da = 0; da < 8; da++) {
+      short8 va_s = vf[da];
+      acc[da] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+          *reinterpret_cast<bf16x8_t*>(&va_s), p_frag, acc[da], 0, 0, 0);
+    }
+    if constexpr (PRIO) __builtin_amdgcn_s_setprio(0);
+  };
+
+  if constexpr (VT == 2) {
+    // software-pipelined: tile i+1's K/V global loads are in flight while
+    // tile i computes; two compile-time register sets (no runtime
+    // indexing -> no scratch spill, unlike KPF)
+    short8 kA[8], vA[8], kB[8], vB[8];
+    const int ntiles = kSlab / 32;
+    if (slab_start < slab_end) {
+      load_k(slab_start, kA);
+      load_vt(slab_start, vA);
+    }
+    for (int ti = 0; ti < ntiles; ti += 2) {
+      const int t0 = slab_start + ti * 32;
+      const int t1 = t0 + 32;
+      if (t1 < slab_end) {
+        load_k(t1, kB);
+        load_vt(t1, vB);
+      }
+      if (t0 < slab_end) tile_vt(t0, kA, vA);
+      const int t2 = t0 + 64;
+      if (t2 < slab_end) {
+        load_k(t2, kA);
+        load_vt(t2, vA);
+      }
+      if (t1 < slab_end) tile_vt(t1, kB, vB);
+    }
+  } else {
   for (int ti = 0; ti < kSlab / 32; ti++) {
     const int t0 = slab_start + ti * 32;
     const bool active = t0 < slab_end;
@@ -754,7 +880,8 @@ __global__ __launch_bounds__(kBlock) void paged_decode_mfma_swapped(
     if constexpr (!KPF) {
       if (active) {
         load_k(t0, kfr[0]);
-        load_v(t0, vfr[0]);
+        if constexpr (VT) load_vt(t0, vfr[0]);
+        else load_v(t0, vfr[0]);
       }
     }
     const short8(&kf)[8] = KPF ? kfr[cur] : kfr[0];
@@ -776,8 +903,9 @@ __global__ __launch_bounds__(kBlock) void paged_decode_mfma_swapped(
         }
       }
       if constexpr (PRIO) __builtin_amdgcn_s_setprio(0);
-      // stage THIS tile's V rows (loaded last iteration under KPF)
-      store_v(vf);
+      // stage THIS tile's V rows (loaded last iteration under KPF);
+      // VT reads V^T fragments straight from global - nothing to stage
+      if constexpr (!VT) store_v(vf);
       if constexpr (KPF) {
         // issue next tile's K/V global loads now: their latency hides
         // under this tile's softmax + PV
@@ -835,19 +963,26 @@ __global__ __launch_bounds__(kBlock) void paged_decode_mfma_swapped(
       unsigned int w[4] = {f02.x, f13.x, f02.y, f13.y};
       bf16x8_t p_frag = *reinterpret_cast<bf16x8_t*>(w);
 
-      // V staging for THIS tile must be visible (same wave, in-order LDS)
-      __builtin_amdgcn_wave_barrier();
-      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+      if constexpr (!VT) {
+        // V staging for THIS tile must be visible (same wave, in-order LDS)
+        __builtin_amdgcn_wave_barrier();
+        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+      }
 
       // ---- PV: O^T[dim][head] += V^T[dim][tok] P^T[tok][head] ----
       if constexpr (PRIO) __builtin_amdgcn_s_setprio(1);
 #pragma unroll
       for (int da = 0; da < 8; da++) {
-        const int vrow = da * 16 + lr;
-        const int vtoff = XK2 ? ((lg * 16) ^ (((vrow >> 4) & 1) << 5))
-                              : (lg * 16);
-        short8 va_s = *reinterpret_cast<const short8*>(
-            (const char*)v_lds + vrow * VS + vtoff);
+        short8 va_s;
+        if constexpr (VT) {
+          va_s = vf[da];
+        } else {
+          const int vrow = da * 16 + lr;
+          const int vtoff = XK2 ? ((lg * 16) ^ (((vrow >> 4) & 1) << 5))
+                                : (lg * 16);
+          va_s = *reinterpret_cast<const short8*>(
+              (const char*)v_lds + vrow * VS + vtoff);
+        }
         acc[da] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
             *reinterpret_cast<bf16x8_t*>(&va_s), p_frag, acc[da], 0, 0, 0);
       }
@@ -855,6 +990,7 @@ __global__ __launch_bounds__(kBlock) void paged_decode_mfma_swapped(
     }
     __builtin_amdgcn_wave_barrier();
   }
+  }  // VT != 2
 
   // ---- cross-wave merge: acc[da][r] is O^T[dim da*16+lg*4+r][head lr]
   __syncthreads();
@@ -902,6 +1038,11 @@ __global__ __launch_bounds__(kBlock) void paged_decode_mfma_swapped(
       }
     }
   }
+}
+
+// VT layout needs no V staging LDS: merge + q only
+inline int mfma_swapped_vt_lds_bytes(int G, int hd) {
+  return 4 * G * (hd + 2) * 4 + G * hd * 2;
 }
 
 inline int mfma_swapped_lds_bytes(int G, int hd, int vs = 80) {

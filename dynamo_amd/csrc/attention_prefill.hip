@@ -37,6 +37,7 @@ DEVINL int swzv(int byte_in_row, int row) {
   return byte_in_row ^ (((row ^ (row >> 3)) & 7) << 4);
 }
 
+template <int VT = 0>  // VT: vcache is d-major [P, Hkv, hd, ps]
 __global__ __launch_bounds__(kBlock) void prefill_kernel(
     short* __restrict__ out,            // [Tq, Hq, 128]
     const short* __restrict__ q,        // [Tq, Hq, 128]
@@ -112,13 +113,38 @@ __global__ __launch_bounds__(kBlock) void prefill_kernel(
         const int64_t page = pt[t >> log2_ps];
         const int64_t base = ((page * Hkv + kvh) * ps + (t & (ps - 1))) * HD + col8 * 8;
         kv_k = *reinterpret_cast<const short8*>(kcache + base);
-        kv_v = *reinterpret_cast<const short8*>(vcache + base);
+        if constexpr (!VT)
+          kv_v = *reinterpret_cast<const short8*>(vcache + base);
       }
       *reinterpret_cast<short8*>((char*)k_lds + row * 256 + swz(col8 * 16, row)) = kv_k;
+      if constexpr (!VT) {
 #pragma unroll
-      for (int e = 0; e < 8; e++) {
-        const int dim = col8 * 8 + e;
-        *(short*)((char*)vt_lds + dim * 128 + swzv(row * 2, dim)) = kv_v[e];
+        for (int e = 0; e < 8; e++) {
+          const int dim = col8 * 8 + e;
+          *(short*)((char*)vt_lds + dim * 128 + swzv(row * 2, dim)) = kv_v[e];
+        }
+      }
+    }
+    if constexpr (VT) {
+      // d-major pages: each slot stages one 8-token run of one dim row —
+      // contiguous global b128, b128 LDS store (swzv key is 16B-granular)
+#pragma unroll
+      for (int i = 0; i < 4; i++) {
+        const int slot = i * kBlock + threadIdx.x;  // 128 dims x 8 chunks
+        const int dim = slot >> 3, tc = slot & 7;
+        const int tcs = t0 + tc * 8;
+        const int tsafe = min(tcs, (ctx - 1) & ~7);
+        const int64_t page = pt[tsafe >> log2_ps];
+        short8 vv = *reinterpret_cast<const short8*>(
+            vcache + ((page * Hkv + kvh) * (int64_t)HD + dim) * ps +
+            (tsafe & (ps - 1)));
+        if (tsafe + 7 >= ctx) {
+#pragma unroll
+          for (int e = 0; e < 8; e++)
+            if (tsafe + e >= ctx) vv[e] = 0;
+        }
+        *reinterpret_cast<short8*>(
+            (char*)vt_lds + dim * 128 + swzv(tc * 16, dim)) = vv;
       }
     }
     __syncthreads();
@@ -280,7 +306,7 @@ constexpr int kLdsHalf32 = kKB32 * 256 + 128 * 128;  // K + V^T = 32 KB
 // is GSPLIT*32 q rows). G=8 -> 1 subtile; G=4 -> 2; G=2 -> 4.
 // FP8: paged cache stores OCP e4m3 (converted to bf16 while staging to
 // LDS; compute unchanged)
-template <int GSPLIT, int FP8 = 0>
+template <int GSPLIT, int FP8 = 0, int VT = 0>
 __global__ __launch_bounds__(512) void prefill32_kernel(
     short* __restrict__ out,            // [Tq, Hq, 128]
     const short* __restrict__ q,        // [Tq, Hq, 128]
@@ -348,6 +374,26 @@ __global__ __launch_bounds__(512) void prefill32_kernel(
   };
   auto load_tile = [&](int t0) {
     if (vrole) {
+      if constexpr (VT) {
+        // d-major pages: 256 v-threads x 4 slots = 128 dims x 8 chunks;
+        // each slot is one contiguous 8-token b128 run of one dim row
+#pragma unroll
+        for (int j = 0; j < 4; j++) {
+          const int slot = threadIdx.x + j * 256;
+          const int d = slot >> 3, tc = slot & 7;
+          const int tcs = t0 + tc * 8;
+          const int tsafe = min(tcs, (ctx - 1) & ~7);
+          const int64_t page = pt[tsafe >> log2_ps];
+          sreg[j] = cache8(vcache,
+              ((page * Hkv + kvh) * (int64_t)HD + d) * ps + (tsafe & (ps - 1)));
+          if (tsafe + 7 >= ctx) {
+#pragma unroll
+            for (int e = 0; e < 8; e++)
+              if (tsafe + e >= ctx) sreg[j][e] = 0;
+          }
+        }
+        return;
+      }
       const int unit = threadIdx.x;
       const int row0 = (unit >> 4) * 4, d0 = (unit & 15) * 8;
 #pragma unroll
@@ -374,6 +420,17 @@ __global__ __launch_bounds__(512) void prefill32_kernel(
     char* kl = lds32 + buf * kLdsHalf32;
     char* vl = kl + kKB32 * 256;
     if (vrole) {
+      if constexpr (VT) {
+#pragma unroll
+        for (int j = 0; j < 4; j++) {
+          const int slot = threadIdx.x + j * 256;
+          const int d = slot >> 3, tc = slot & 7;
+          const int key = (d ^ (d >> 3)) & 7;
+          *reinterpret_cast<short8*>(
+              vl + d * 128 + ((tc * 16) ^ (key << 4))) = sreg[j];
+        }
+        return;
+      }
       const int unit = threadIdx.x;
       const int row0 = (unit >> 4) * 4, d0 = (unit & 15) * 8;
 #pragma unroll
@@ -541,7 +598,7 @@ void attention_prefill_paged(torch::Tensor out, torch::Tensor q,
                              torch::Tensor page_table, torch::Tensor tile_seq,
                              torch::Tensor tile_q0, torch::Tensor seq_q_start,
                              torch::Tensor seq_q_len, torch::Tensor seq_ctx_len,
-                             double scale) {
+                             double scale, bool v_transposed) {
   TORCH_CHECK(q.is_cuda() && q.dtype() == torch::kBFloat16);
   TORCH_CHECK(q.size(-1) == 128, "only head_dim=128 supported natively");
   TORCH_CHECK(page_table.dtype() == torch::kInt32);
@@ -561,6 +618,12 @@ void attention_prefill_paged(torch::Tensor out, torch::Tensor q,
   // (tile rows = (8/G)*32, built by the Python layer via prefill_tile_rows —
   // keep the predicates in sync)
   const int G = (Hq % Hkv == 0) ? Hq / Hkv : 0;
+  if (v_transposed) {
+    TORCH_CHECK(vcache.dim() == 4 && vcache.size(2) == q.size(-1) &&
+                    vcache.size(3) == ps,
+                "v_transposed expects vcache [P, Hkv, hd, ps]");
+    TORCH_CHECK(ps % 8 == 0, "v_transposed prefill needs page_size%8==0");
+  }
   if (G == 8 || G == 4 || G == 2) {
     dim3 grid(ntiles, Hkv);
     auto launch = [&](auto kern) {
@@ -572,7 +635,15 @@ void attention_prefill_paged(torch::Tensor out, torch::Tensor q,
           seq_q_len.data_ptr<int32_t>(), seq_ctx_len.data_ptr<int32_t>(),
           (float)scale, Hq, Hkv, max_pages, log2_ps);
     };
-    if (fp8) {
+    if (v_transposed) {
+      if (fp8) {
+        if (G == 8) launch(prefill32_kernel<1, 1, 1>);
+        else if (G == 4) launch(prefill32_kernel<2, 1, 1>);
+        else launch(prefill32_kernel<4, 1, 1>);
+      } else if (G == 8) launch(prefill32_kernel<1, 0, 1>);
+      else if (G == 4) launch(prefill32_kernel<2, 0, 1>);
+      else launch(prefill32_kernel<4, 0, 1>);
+    } else if (fp8) {
       if (G == 8) launch(prefill32_kernel<1, 1>);
       else if (G == 4) launch(prefill32_kernel<2, 1>);
       else launch(prefill32_kernel<4, 1>);
@@ -584,12 +655,16 @@ void attention_prefill_paged(torch::Tensor out, torch::Tensor q,
   }
   TORCH_CHECK(!fp8, "fp8 KV prefill needs GQA group in {2,4,8}");
   dim3 grid(ntiles, Hq);
-  prefill_kernel<<<grid, kBlock, 0, stream>>>(
-      (short*)out.data_ptr(), (const short*)q.data_ptr(),
-      (const short*)kcache.data_ptr(), (const short*)vcache.data_ptr(),
-      page_table.data_ptr<int32_t>(), tile_seq.data_ptr<int32_t>(),
-      tile_q0.data_ptr<int32_t>(), seq_q_start.data_ptr<int32_t>(),
-      seq_q_len.data_ptr<int32_t>(), seq_ctx_len.data_ptr<int32_t>(),
-      (float)scale, Hq, Hkv, max_pages, log2_ps);
+  auto launch16 = [&](auto kern) {
+    kern<<<grid, kBlock, 0, stream>>>(
+        (short*)out.data_ptr(), (const short*)q.data_ptr(),
+        (const short*)kcache.data_ptr(), (const short*)vcache.data_ptr(),
+        page_table.data_ptr<int32_t>(), tile_seq.data_ptr<int32_t>(),
+        tile_q0.data_ptr<int32_t>(), seq_q_start.data_ptr<int32_t>(),
+        seq_q_len.data_ptr<int32_t>(), seq_ctx_len.data_ptr<int32_t>(),
+        (float)scale, Hq, Hkv, max_pages, log2_ps);
+  };
+  if (v_transposed) launch16(prefill_kernel<1>);
+  else launch16(prefill_kernel<0>);
   HIP_CHECK_KERNEL();
 }

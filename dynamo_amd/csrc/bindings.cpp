@@ -13,7 +13,7 @@ void rope_append_qkv(torch::Tensor q_out, torch::Tensor kcache,
                      torch::Tensor vcache, torch::Tensor qkv,
                      c10::optional<torch::Tensor> bias,
                      torch::Tensor positions, torch::Tensor slot_mapping,
-                     torch::Tensor cos_sin_cache);
+                     torch::Tensor cos_sin_cache, bool v_transposed);
 void rope_inplace(torch::Tensor q, torch::Tensor k, torch::Tensor positions,
                   torch::Tensor cos_sin_cache, int64_t num_q_heads,
                   int64_t num_k_heads, int64_t head_dim);
@@ -22,7 +22,8 @@ void silu_mul(torch::Tensor out, torch::Tensor gate_up);
 void gelu(torch::Tensor out, torch::Tensor input);
 // cache.hip
 void kv_cache_append(torch::Tensor kcache, torch::Tensor vcache, torch::Tensor k,
-                     torch::Tensor v, torch::Tensor slot_mapping);
+                     torch::Tensor v, torch::Tensor slot_mapping,
+                     bool v_transposed);
 void gather_pages(torch::Tensor staging, torch::Tensor cache, torch::Tensor page_ids);
 void scatter_pages(torch::Tensor staging, torch::Tensor cache, torch::Tensor page_ids);
 void copy_pages(torch::Tensor dst_cache, torch::Tensor src_cache, torch::Tensor pairs);
@@ -33,14 +34,14 @@ void paged_attention_decode(torch::Tensor out, torch::Tensor q, torch::Tensor kc
                             torch::Tensor vcache, torch::Tensor page_table,
                             torch::Tensor ctx_lens, torch::Tensor partial,
                             torch::Tensor ml, double scale,
-                            int64_t chunk_tokens);
+                            int64_t chunk_tokens, bool v_transposed);
 // attention_prefill.hip
 void attention_prefill_paged(torch::Tensor out, torch::Tensor q,
                              torch::Tensor kcache, torch::Tensor vcache,
                              torch::Tensor page_table, torch::Tensor tile_seq,
                              torch::Tensor tile_q0, torch::Tensor seq_q_start,
                              torch::Tensor seq_q_len, torch::Tensor seq_ctx_len,
-                             double scale);
+                             double scale, bool v_transposed);
 // sampling.hip
 void greedy_sample(torch::Tensor out, torch::Tensor logits);
 void topkp_sample(torch::Tensor out, torch::Tensor logits,
@@ -74,17 +75,28 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rope_append_qkv", &rope_append_qkv,
         py::arg("q_out"), py::arg("kcache"), py::arg("vcache"),
         py::arg("qkv"), py::arg("bias"), py::arg("positions"),
-        py::arg("slot_mapping"), py::arg("cos_sin"));
+        py::arg("slot_mapping"), py::arg("cos_sin"),
+        py::arg("v_transposed") = false);
   m.def("silu_mul", &silu_mul);
   m.def("gelu", &gelu);
-  m.def("kv_cache_append", &kv_cache_append);
+  m.def("kv_cache_append", &kv_cache_append, py::arg("kcache"),
+        py::arg("vcache"), py::arg("k"), py::arg("v"),
+        py::arg("slot_mapping"), py::arg("v_transposed") = false);
   m.def("gather_pages", &gather_pages);
   m.def("scatter_pages", &scatter_pages);
   m.def("copy_pages", &copy_pages);
   m.def("paged_decode_num_chunks", &paged_decode_num_chunks);
   m.def("decode_chunk_tokens", &decode_chunk_tokens_py);
-  m.def("paged_attention_decode", &paged_attention_decode);
-  m.def("attention_prefill_paged", &attention_prefill_paged);
+  m.def("paged_attention_decode", &paged_attention_decode,
+        py::arg("out"), py::arg("q"), py::arg("kcache"), py::arg("vcache"),
+        py::arg("page_table"), py::arg("ctx_lens"), py::arg("partial"),
+        py::arg("ml"), py::arg("scale"), py::arg("chunk_tokens"),
+        py::arg("v_transposed") = false);
+  m.def("attention_prefill_paged", &attention_prefill_paged,
+        py::arg("out"), py::arg("q"), py::arg("kcache"), py::arg("vcache"),
+        py::arg("page_table"), py::arg("tile_seq"), py::arg("tile_q0"),
+        py::arg("seq_q_start"), py::arg("seq_q_len"), py::arg("seq_ctx_len"),
+        py::arg("scale"), py::arg("v_transposed") = false);
   m.def("greedy_sample", &greedy_sample);
   m.def("gumbel_sample", &gumbel_sample, py::arg("out"), py::arg("logits"),
         py::arg("inv_temp"), py::arg("seed"),

@@ -16,6 +16,7 @@ namespace {
 
 constexpr int kBlock = 256;
 
+template <int VT = 0>  // VT: vcache is d-major [P, Hkv, hd, ps]
 __global__ void kv_append_kernel(short* __restrict__ kcache,
                                  short* __restrict__ vcache,
                                  const short* __restrict__ k,  // [T, Hkv*hd]
@@ -35,8 +36,16 @@ __global__ void kv_append_kernel(short* __restrict__ kcache,
     const int64_t dst = (((page * Hkv + h) * page_size + off) * hd + d);
     *reinterpret_cast<short8*>(kcache + dst) =
         *reinterpret_cast<const short8*>(k + (int64_t)t * Hkv * hd + i * 8);
-    *reinterpret_cast<short8*>(vcache + dst) =
+    const short8 vv =
         *reinterpret_cast<const short8*>(v + (int64_t)t * Hkv * hd + i * 8);
+    if constexpr (VT) {
+      const int64_t vb0 = ((page * Hkv + h) * hd + d) * page_size + off;
+#pragma unroll
+      for (int e = 0; e < 8; e++)
+        vcache[vb0 + (int64_t)e * page_size] = vv[e];
+    } else {
+      *reinterpret_cast<short8*>(vcache + dst) = vv;
+    }
   }
 }
 
@@ -84,7 +93,8 @@ __global__ void page_pair_copy_kernel(short* __restrict__ dst_cache,
 }  // namespace
 
 void kv_cache_append(torch::Tensor kcache, torch::Tensor vcache,
-                     torch::Tensor k, torch::Tensor v, torch::Tensor slot_mapping) {
+                     torch::Tensor k, torch::Tensor v,
+                     torch::Tensor slot_mapping, bool v_transposed) {
   TORCH_CHECK(kcache.is_cuda() && kcache.dtype() == torch::kBFloat16);
   TORCH_CHECK(slot_mapping.dtype() == torch::kInt64);
   const int Hkv = kcache.size(1);
@@ -94,10 +104,14 @@ void kv_cache_append(torch::Tensor kcache, torch::Tensor vcache,
   TORCH_CHECK(hd % 8 == 0);
   if (T == 0) return;
   auto stream = at::cuda::getCurrentHIPStream();
-  kv_append_kernel<<<T, kBlock, 0, stream>>>(
-      (short*)kcache.data_ptr(), (short*)vcache.data_ptr(),
-      (const short*)k.data_ptr(), (const short*)v.data_ptr(),
-      slot_mapping.data_ptr<int64_t>(), T, Hkv, page_size, hd);
+  auto launch = [&](auto kern) {
+    kern<<<T, kBlock, 0, stream>>>(
+        (short*)kcache.data_ptr(), (short*)vcache.data_ptr(),
+        (const short*)k.data_ptr(), (const short*)v.data_ptr(),
+        slot_mapping.data_ptr<int64_t>(), T, Hkv, page_size, hd);
+  };
+  if (v_transposed) launch(kv_append_kernel<1>);
+  else launch(kv_append_kernel<0>);
   HIP_CHECK_KERNEL();
 }
 

@@ -57,7 +57,12 @@ __global__ void rope_kernel(short* __restrict__ q,    // [T, Hq*hd]
 // scatters k/v straight into the paged cache. Replaces 4 kernels
 // (q copy, k copy, rope, kv_append) with one launch — on Llama-70B decode
 // that is 3 fewer hipGraph nodes per layer x 80 layers.
-template <int FP8 = 0>
+// VT: d-major (transposed) V pages — vcache [P, Hkv, hd, ps]; the decode
+// kernel then reads PV A-fragments as contiguous b128 token runs. The
+// per-token scatter below touches hd cache lines (~32x DRAM write
+// amplification at decode: ~10 MB/step, irrelevant next to the 20+ GB/step
+// of V reads it speeds up; prefill batches cover whole lines in L2).
+template <int FP8 = 0, int VT = 0>
 __global__ void rope_append_kernel(
     short* __restrict__ q_out,          // [T, Hq*hd] contiguous
     short* __restrict__ kcache,         // [P, Hkv, ps, hd]
@@ -143,6 +148,22 @@ __global__ void rope_append_kernel(
           v[e] = f32_to_bf16(bf16_to_f32(v[e]) +
                              bf16_to_f32(bias[voff + h * hd + d + e]));
       }
+      if constexpr (VT) {
+        // transposed page: elem offset d * ps + token_in_page
+        const int64_t vb0 = ((page * Hkv + h) * hd + d) * page_size + off;
+#pragma unroll
+        for (int e = 0; e < 8; e++) {
+          if constexpr (FP8) {
+            reinterpret_cast<unsigned char*>(vcache)[vb0 + (int64_t)e *
+                                                     page_size] =
+                __hip_cvt_float_to_fp8(bf16_to_f32(v[e]), __HIP_SATFINITE,
+                                       __HIP_E4M3);
+          } else {
+            vcache[vb0 + (int64_t)e * page_size] = v[e];
+          }
+        }
+        continue;
+      }
       const int64_t vb = ((page * Hkv + h) * page_size + off) * hd + d;
       if constexpr (FP8) {
         uchar8 pv;
@@ -165,7 +186,7 @@ void rope_append_qkv(torch::Tensor q_out, torch::Tensor kcache,
                      torch::Tensor vcache, torch::Tensor qkv,
                      c10::optional<torch::Tensor> bias,
                      torch::Tensor positions, torch::Tensor slot_mapping,
-                     torch::Tensor cos_sin_cache) {
+                     torch::Tensor cos_sin_cache, bool v_transposed) {
   TORCH_CHECK(qkv.is_cuda() && qkv.dtype() == torch::kBFloat16);
   TORCH_CHECK(qkv.dim() == 2 && qkv.stride(1) == 1);
   TORCH_CHECK(positions.dtype() == torch::kInt32);
@@ -197,8 +218,13 @@ void rope_append_qkv(torch::Tensor q_out, torch::Tensor kcache,
         cos_sin_cache.data_ptr<float>(), T, Hq, Hkv, page_size, hd,
         (int)qkv.stride(0));
   };
-  if (fp8) launch(rope_append_kernel<1>);
-  else launch(rope_append_kernel<0>);
+  if (v_transposed)
+    TORCH_CHECK(vcache.size(2) == hd && vcache.size(3) == page_size,
+                "v_transposed expects vcache [P, Hkv, hd, ps]");
+  if (fp8 && v_transposed) launch(rope_append_kernel<1, 1>);
+  else if (fp8) launch(rope_append_kernel<1, 0>);
+  else if (v_transposed) launch(rope_append_kernel<0, 1>);
+  else launch(rope_append_kernel<0, 0>);
   HIP_CHECK_KERNEL();
 }
 

@@ -148,11 +148,30 @@ class EngineConfig:
     # tensor parallelism (process group wired by the worker)
     tp_size: int = 1
     tp_rank: int = 0
+    # V-cache page layout: "auto" stores V pages d-major ([P, Hkv, hd, ps])
+    # on GPU whenever the native kernels support it - the decode PV
+    # A-fragment then reads as contiguous token runs, deleting the V^T
+    # staging transpose (decode sweep: G8 3911 -> 4476, G4 3951 -> 4987
+    # GB/s). "never" keeps token-major pages everywhere.
+    kv_v_layout: str = "auto"           # auto | never
 
     @property
     def torch_dtype(self):
         import torch
         return {"bfloat16": torch.bfloat16, "float32": torch.float32}[self.dtype]
+
+    @property
+    def v_transposed(self) -> bool:
+        """Resolved V-page layout: d-major needs the swapped decode kernel
+        (head_dim 128, page_size % 32 == 0, GQA group 2..16) and the
+        prefill32/fallback staging (page_size % 8 == 0)."""
+        if self.kv_v_layout == "never":
+            return False
+        m = self.model
+        g = (m.num_q_heads // m.num_kv_heads
+             if m.num_kv_heads and m.num_q_heads % m.num_kv_heads == 0 else 0)
+        return (self.device.startswith("cuda") and m.head_dim == 128
+                and self.page_size % 32 == 0 and 2 <= g <= 16)
 
     @property
     def kv_torch_dtype(self):

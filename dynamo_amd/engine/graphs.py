@@ -60,6 +60,7 @@ class GraphRunner:
             decode_ctx_lens=self.ctx_lens[:bc],
             decode_scratch=self.runner.decode_scratch,
             num_prefill_tokens=0,
+            v_transposed=getattr(self.runner, "v_transposed", False),
         )
 
     @torch.inference_mode()

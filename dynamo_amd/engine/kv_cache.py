@@ -232,7 +232,12 @@ class KVCachePool:
 
     def __init__(self, num_layers: int, num_pages: int, num_kv_heads: int,
                  page_size: int, head_dim: int, device: str,
-                 dtype=torch.bfloat16, shm_export: bool = False):
+                 dtype=torch.bfloat16, shm_export: bool = False,
+                 v_transposed: bool = False):
+        # v_transposed: V pages are stored d-major; vcache() views the same
+        # storage as [P, Hkv, hd, ps] (page bytes and transfer granularity
+        # are unchanged - copies and exports stay layout-agnostic)
+        self.v_transposed = v_transposed
         self.shape = (num_layers, 2, num_pages, num_kv_heads, page_size, head_dim)
         self.num_pages = num_pages
         self.page_size = page_size
@@ -277,7 +282,11 @@ class KVCachePool:
         return self.buffer[layer, 0]
 
     def vcache(self, layer: int) -> torch.Tensor:
-        return self.buffer[layer, 1]
+        v = self.buffer[layer, 1]
+        if self.v_transposed:
+            _, _, P, hkv, ps, hd = self.shape
+            return v.reshape(P, hkv, hd, ps)
+        return v
 
     def ipc_export(self) -> bytes:
         from dynamo_amd import ops

@@ -39,10 +39,16 @@ class ModelRunner:
         self.hq_local = m.num_q_heads // self.tp.size
 
         num_pages = cfg.kv_pool_pages or self._auto_pages()
+        # resolve the V-page layout against the LOCAL (TP-sharded) head
+        # counts - the swapped decode kernel needs 2 <= G_local <= 16
+        g_local = (self.hq_local // self.hkv_local
+                   if self.hq_local % self.hkv_local == 0 else 0)
+        self.v_transposed = cfg.v_transposed and 2 <= g_local <= 16
         self.kv_pool = KVCachePool(m.num_layers, num_pages, self.hkv_local,
                                    cfg.page_size, m.head_dim, cfg.device,
                                    cfg.kv_torch_dtype,
-                                   shm_export=cfg.cpu_shm_pool)
+                                   shm_export=cfg.cpu_shm_pool,
+                                   v_transposed=self.v_transposed)
         self.num_pages = num_pages
         self.max_pages_per_seq = (cfg.max_model_len + cfg.page_size - 1) // cfg.page_size
         self.decode_scratch = None
@@ -150,6 +156,7 @@ class ModelRunner:
             seq_ctx_len=(torch.tensor(ctx_len, dtype=torch.int32).to(dev, non_blocking=True)
                          if ctx_len else None),
             logits_rows=torch.tensor(logits_rows, dtype=torch.int64).to(dev, non_blocking=True),
+            v_transposed=self.v_transposed,
         )
         if embeds_rows:
             meta.embeds_rows = torch.tensor(embeds_rows, dtype=torch.int64,

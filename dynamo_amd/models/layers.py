@@ -45,6 +45,8 @@ class AttnMetadata:
     # request instead of the embedding table
     embeds_rows: Optional[torch.Tensor] = None         # [Te] int64
     inputs_embeds: Optional[torch.Tensor] = None       # [Te, hidden]
+    # V pages stored d-major (see EngineConfig.kv_v_layout)
+    v_transposed: bool = False
 
 
 def linear(x: torch.Tensor, w: torch.Tensor) -> torch.Tensor:
@@ -199,7 +201,8 @@ class Attention(torch.nn.Module):
         # k/v scattered into the cache (one kernel vs four)
         q = ops.rope_append_qkv(qkv, self.bqkv, meta.positions,
                                 meta.slot_mapping, cos_sin, kcache, vcache,
-                                self.hq, self.hkv, self.hd)
+                                self.hq, self.hkv, self.hd,
+                                v_transposed=meta.v_transposed)
         qh = q.view(T, self.hq, self.hd)
         out = torch.empty_like(qh)
         nd = meta.num_decode
@@ -207,12 +210,12 @@ class Attention(torch.nn.Module):
             ops.paged_attention_decode(
                 qh[:nd], kcache, vcache, meta.decode_page_table,
                 meta.decode_ctx_lens, self.scale, meta.decode_scratch,
-                out=out[:nd])
+                out=out[:nd], v_transposed=meta.v_transposed)
         if meta.num_prefill_tokens:
             out[nd:] = ops.attention_prefill_paged(
                 qh[nd:].contiguous(), kcache, vcache, meta.prefill_page_table,
                 meta.seq_q_start, meta.seq_q_len, meta.seq_ctx_len, self.scale,
-                meta.prefill_tiles)
+                meta.prefill_tiles, v_transposed=meta.v_transposed)
         o = linear_lora(out.view(T, self.hq * self.hd), self.wo, self.lora, "o")
         return self.tp.all_reduce(o)
 

@@ -84,16 +84,19 @@ def gelu(x: torch.Tensor) -> torch.Tensor:
     return torch_ref.gelu(x)
 
 
-def kv_cache_append(kcache, vcache, k, v, slot_mapping):
+def kv_cache_append(kcache, vcache, k, v, slot_mapping,
+                    v_transposed=False):
     if kcache.is_cuda:
         hip().kv_cache_append(kcache, vcache, k.contiguous(), v.contiguous(),
-                              slot_mapping)
+                              slot_mapping, v_transposed)
     else:
-        torch_ref.kv_cache_append(kcache, vcache, k, v, slot_mapping)
+        torch_ref.kv_cache_append(kcache, vcache, k, v, slot_mapping,
+                                  v_transposed)
 
 
 def rope_append_qkv(qkv, bias, positions, slot_mapping, cos_sin,
-                    kcache, vcache, num_q_heads, num_kv_heads, head_dim):
+                    kcache, vcache, num_q_heads, num_kv_heads, head_dim,
+                    v_transposed=False):
     """Fused QKV epilogue: strided qkv read (+bias) -> rope(q,k) ->
     q contiguous out, k/v scattered into the paged cache. One kernel
     instead of {q copy, k copy, rope, kv_append}. Returns q [T, Hq*hd]."""
@@ -102,7 +105,7 @@ def rope_append_qkv(qkv, bias, positions, slot_mapping, cos_sin,
         q_out = torch.empty(T, num_q_heads * head_dim, dtype=qkv.dtype,
                             device=qkv.device)
         hip().rope_append_qkv(q_out, kcache, vcache, qkv, bias, positions,
-                              slot_mapping, cos_sin)
+                              slot_mapping, cos_sin, v_transposed)
         return q_out
     # CPU reference: the unfused sequence
     if bias is not None:
@@ -114,7 +117,8 @@ def rope_append_qkv(qkv, bias, positions, slot_mapping, cos_sin,
     q, k = torch_ref.rope(q, k, positions, cos_sin, num_q_heads,
                           num_kv_heads, head_dim)
     torch_ref.kv_cache_append(kcache, vcache, k.view(T, num_kv_heads, head_dim),
-                              v.view(T, num_kv_heads, head_dim), slot_mapping)
+                              v.view(T, num_kv_heads, head_dim), slot_mapping,
+                              v_transposed)
     return q
 
 
@@ -135,7 +139,8 @@ class DecodeScratch:
 
 
 def paged_attention_decode(q, kcache, vcache, page_table, ctx_lens, scale,
-                           scratch: "DecodeScratch | None" = None, out=None):
+                           scratch: "DecodeScratch | None" = None, out=None,
+                           v_transposed=False):
     if q.is_cuda:
         if out is None:
             out = torch.empty_like(q)
@@ -143,10 +148,11 @@ def paged_attention_decode(q, kcache, vcache, page_table, ctx_lens, scale,
         partial, ml = scratch.view(q.shape[0])
         hip().paged_attention_decode(out, q, kcache, vcache, page_table,
                                      ctx_lens, partial, ml, scale,
-                                     scratch.chunk_tokens)
+                                     scratch.chunk_tokens, v_transposed)
         return out
     r = torch_ref.paged_attention_decode(q, kcache, vcache, page_table,
-                                         ctx_lens, scale)
+                                         ctx_lens, scale,
+                                         v_transposed=v_transposed)
     if out is not None:
         out.copy_(r)
         return out
@@ -177,7 +183,8 @@ def build_prefill_tiles(seq_q_lens, device, rows: int = 64):
 
 
 def attention_prefill_paged(q, kcache, vcache, page_table, seq_q_start,
-                            seq_q_len, seq_ctx_len, scale, tiles=None):
+                            seq_q_len, seq_ctx_len, scale, tiles=None,
+                            v_transposed=False):
     if q.is_cuda:
         out = torch.empty_like(q)
         if tiles is None:
@@ -186,11 +193,13 @@ def attention_prefill_paged(q, kcache, vcache, page_table, seq_q_start,
         tile_seq, tile_q0 = tiles
         hip().attention_prefill_paged(out, q, kcache, vcache, page_table,
                                       tile_seq, tile_q0, seq_q_start,
-                                      seq_q_len, seq_ctx_len, scale)
+                                      seq_q_len, seq_ctx_len, scale,
+                                      v_transposed)
         return out
     return torch_ref.attention_prefill_paged(q, kcache, vcache, page_table,
                                              seq_q_start, seq_q_len,
-                                             seq_ctx_len, scale)
+                                             seq_ctx_len, scale,
+                                             v_transposed=v_transposed)
 
 
 def greedy_sample(logits: torch.Tensor) -> torch.Tensor:

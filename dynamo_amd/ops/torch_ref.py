@@ -85,28 +85,41 @@ def gelu(x: torch.Tensor) -> torch.Tensor:
     return torch.nn.functional.gelu(x.float(), approximate="tanh").to(x.dtype)
 
 
-def kv_cache_append(kcache, vcache, k, v, slot_mapping):
-    """kcache/vcache: [P, Hkv, ps, hd]; k/v: [T, Hkv, hd]; slots: [T] int64."""
+def kv_cache_append(kcache, vcache, k, v, slot_mapping,
+                    v_transposed=False):
+    """kcache: [P, Hkv, ps, hd]; k/v: [T, Hkv, hd]; slots: [T] int64.
+    v_transposed: vcache is d-major [P, Hkv, hd, ps] (the MI355X decode
+    layout — PV fragments read as contiguous token runs)."""
     P, Hkv, ps, hd = kcache.shape
     valid = slot_mapping >= 0
     slots = slot_mapping[valid]
     pages = (slots // ps).long()
     offs = (slots % ps).long()
     kcache[pages, :, offs] = k[valid].to(kcache.dtype)
-    vcache[pages, :, offs] = v[valid].to(vcache.dtype)
+    if v_transposed:
+        vcache[pages, :, :, offs] = v[valid].to(vcache.dtype)
+    else:
+        vcache[pages, :, offs] = v[valid].to(vcache.dtype)
 
 
-def _gather_kv(kcache, page_table, ctx_len):
+def _gather_kv(kcache, page_table, ctx_len, v_transposed=False):
     """-> [ctx_len, Hkv, hd] for one sequence."""
-    P, Hkv, ps, hd = kcache.shape
+    if v_transposed:
+        P, Hkv, hd, ps = kcache.shape
+    else:
+        P, Hkv, ps, hd = kcache.shape
     npages = (ctx_len + ps - 1) // ps
     pages = page_table[:npages].long()
-    kv = kcache[pages]                      # [np, Hkv, ps, hd]
-    kv = kv.permute(0, 2, 1, 3).reshape(npages * ps, Hkv, hd)
+    kv = kcache[pages]
+    if v_transposed:                        # [np, Hkv, hd, ps]
+        kv = kv.permute(0, 3, 1, 2).reshape(npages * ps, Hkv, hd)
+    else:                                   # [np, Hkv, ps, hd]
+        kv = kv.permute(0, 2, 1, 3).reshape(npages * ps, Hkv, hd)
     return kv[:ctx_len]
 
 
-def paged_attention_decode(q, kcache, vcache, page_table, ctx_lens, scale):
+def paged_attention_decode(q, kcache, vcache, page_table, ctx_lens, scale,
+                           v_transposed=False):
     """q: [B, Hq, hd] -> out [B, Hq, hd]."""
     B, Hq, hd = q.shape
     Hkv = kcache.shape[1]
@@ -115,7 +128,7 @@ def paged_attention_decode(q, kcache, vcache, page_table, ctx_lens, scale):
     for b in range(B):
         ctx = int(ctx_lens[b])
         kk = _gather_kv(kcache, page_table[b], ctx).float()  # [ctx, Hkv, hd]
-        vv = _gather_kv(vcache, page_table[b], ctx).float()
+        vv = _gather_kv(vcache, page_table[b], ctx, v_transposed).float()
         qq = q[b].float()                                    # [Hq, hd]
         kk = kk.repeat_interleave(G, dim=1)                  # [ctx, Hq, hd]
         vv = vv.repeat_interleave(G, dim=1)
@@ -127,7 +140,8 @@ def paged_attention_decode(q, kcache, vcache, page_table, ctx_lens, scale):
 
 
 def attention_prefill_paged(q, kcache, vcache, page_table, seq_q_start,
-                            seq_q_len, seq_ctx_len, scale):
+                            seq_q_len, seq_ctx_len, scale,
+                            v_transposed=False):
     """Varlen causal attention where K/V come from the paged cache.
 
     q: [Tq, Hq, hd]; query token j of seq s sits at absolute position
@@ -143,7 +157,7 @@ def attention_prefill_paged(q, kcache, vcache, page_table, seq_q_start,
         if ql == 0:
             continue
         kk = _gather_kv(kcache, page_table[s], ctx).float()
-        vv = _gather_kv(vcache, page_table[s], ctx).float()
+        vv = _gather_kv(vcache, page_table[s], ctx, v_transposed).float()
         qq = q[qs:qs + ql].float()                           # [ql, Hq, hd]
         kk = kk.repeat_interleave(G, dim=1)
         vv = vv.repeat_interleave(G, dim=1)

@@ -411,3 +411,63 @@ def test_unseeded_rows_keep_engine_stream():
         steps += 1
         assert steps < 200
     assert [outs[f"r{i}"] for i in range(len(prompts))] == base
+
+
+def test_torch_ref_v_transposed_consistency():
+    """torch_ref append/decode/prefill over d-major V pages match the
+    token-major layout results exactly."""
+    import torch
+    from dynamo_amd.ops import torch_ref
+    torch.manual_seed(7)
+    P, Hkv, ps, hd, Hq = 8, 2, 16, 32, 4
+    kc = torch.zeros(P, Hkv, ps, hd, dtype=torch.bfloat16)
+    vc = torch.zeros_like(kc)
+    vct = torch.zeros(P, Hkv, hd, ps, dtype=torch.bfloat16)
+    T = 40
+    k = torch.randn(T, Hkv, hd, dtype=torch.bfloat16)
+    v = torch.randn(T, Hkv, hd, dtype=torch.bfloat16)
+    slots = torch.randperm(P * ps)[:T].to(torch.int64)
+    torch_ref.kv_cache_append(kc, vc, k, v, slots)
+    torch_ref.kv_cache_append(kc, vct, k, v, slots, v_transposed=True)
+    assert torch.equal(vct, vc.permute(0, 1, 3, 2))
+
+    pt = torch.tensor([[0, 1, 2, 3], [4, 5, 6, 7]], dtype=torch.int32)
+    ctx = torch.tensor([40, 23], dtype=torch.int32)
+    kc = torch.randn(P, Hkv, ps, hd, dtype=torch.bfloat16)
+    vc = torch.randn(P, Hkv, ps, hd, dtype=torch.bfloat16)
+    q = torch.randn(2, Hq, hd, dtype=torch.bfloat16)
+    a = torch_ref.paged_attention_decode(q, kc, vc, pt, ctx, 0.1)
+    b = torch_ref.paged_attention_decode(
+        q, kc, vc.permute(0, 1, 3, 2).contiguous(), pt, ctx, 0.1,
+        v_transposed=True)
+    assert torch.equal(a, b)
+
+    qp = torch.randn(12, Hq, hd, dtype=torch.bfloat16)
+    sqs = torch.tensor([0, 5], dtype=torch.int32)
+    sql = torch.tensor([5, 7], dtype=torch.int32)
+    scl = torch.tensor([30, 20], dtype=torch.int32)
+    a = torch_ref.attention_prefill_paged(qp, kc, vc, pt, sqs, sql, scl, 0.1)
+    b = torch_ref.attention_prefill_paged(
+        qp, kc, vc.permute(0, 1, 3, 2).contiguous(), pt, sqs, sql, scl, 0.1,
+        v_transposed=True)
+    assert torch.equal(a, b)
+
+
+def test_v_transposed_config_resolution():
+    """kv_v_layout auto: on for GPU hd-128 GQA models, off on CPU, off for
+    unsupported geometry, and forceable off."""
+    from dynamo_amd.engine.config import EngineConfig, PRESETS as MODEL_PRESETS
+    cfg = EngineConfig(model=MODEL_PRESETS["llama-3-70b"], device="cuda:0")
+    assert cfg.v_transposed
+    assert not EngineConfig(model=MODEL_PRESETS["llama-3-70b"],
+                            device="cpu").v_transposed
+    assert not EngineConfig(model=MODEL_PRESETS["llama-3-70b"],
+                            device="cuda:0", kv_v_layout="never").v_transposed
+    # head_dim != 128 -> unsupported
+    assert not EngineConfig(model=MODEL_PRESETS["tiny-llama"],
+                            device="cuda:0").v_transposed
+    # MHA (G == 1) -> unsupported
+    mha = MODEL_PRESETS["llama-3-8b"]
+    import dataclasses
+    mha = dataclasses.replace(mha, num_kv_heads=mha.num_q_heads)
+    assert not EngineConfig(model=mha, device="cuda:0").v_transposed

@@ -461,3 +461,143 @@ def test_topkp_sample_temperature_sharpening():
     for seed in range(10):
         out = ops.topkp_sample(logits, inv_t, tk, tp, seed + 1)
         assert torch.equal(out.long(), argmax)
+
+
+# ---------------------------------------------------------------------------
+# d-major (transposed) V-page layout: the decode PV A-fragment reads
+# contiguous token runs; staging transposes disappear (see
+# EngineConfig.kv_v_layout and attention_decode_impl.h VT)
+
+def _vt(vc):
+    """token-major [P, Hkv, ps, hd] -> d-major [P, Hkv, hd, ps] pages."""
+    return vc.permute(0, 1, 3, 2).contiguous()
+
+
+@pytest.mark.parametrize("G,ctxs", [
+    (8, [900, 1, 4096]),
+    (4, [1000, 513, 2048, 7]),
+    (2, [63, 65, 129]),
+    (7, [700, 45, 1025]),
+])
+def test_paged_attention_decode_v_transposed(G, ctxs):
+    torch.manual_seed(3)
+    Hkv, ps, hd = 8, 64, 128
+    Hq = G * Hkv
+    B = len(ctxs)
+    max_pages_seq = (max(ctxs) + ps - 1) // ps
+    P = sum((c + ps - 1) // ps for c in ctxs) + 1
+    kc = torch.randn(P, Hkv, ps, hd, dtype=torch.bfloat16, device=DEV)
+    vc = torch.randn(P, Hkv, ps, hd, dtype=torch.bfloat16, device=DEV)
+    pt = torch.zeros(B, max_pages_seq, dtype=torch.int32, device=DEV)
+    nxt = 1
+    for b, c in enumerate(ctxs):
+        n = (c + ps - 1) // ps
+        pt[b, :n] = torch.arange(nxt, nxt + n, dtype=torch.int32)
+        nxt += n
+    q = torch.randn(B, Hq, hd, dtype=torch.bfloat16, device=DEV)
+    ctx_lens = torch.tensor(ctxs, dtype=torch.int32, device=DEV)
+    scale = hd ** -0.5
+    scratch = ops.DecodeScratch(B, Hq, hd, max(ctxs), DEV)
+    out = ops.paged_attention_decode(q, kc, _vt(vc), pt, ctx_lens, scale,
+                                     scratch, v_transposed=True)
+    ref = torch_ref.paged_attention_decode(q, kc, vc, pt, ctx_lens, scale)
+    assert_close(out, ref)
+
+
+def test_paged_attention_decode_v_transposed_fp8():
+    torch.manual_seed(4)
+    Hkv, ps, hd, G = 8, 64, 128, 8
+    Hq = G * Hkv
+    ctxs = [900, 64, 2048]
+    B = len(ctxs)
+    P = sum((c + ps - 1) // ps for c in ctxs) + 1
+    kc = torch.randn(P, Hkv, ps, hd, dtype=torch.bfloat16, device=DEV)
+    vc = torch.randn(P, Hkv, ps, hd, dtype=torch.bfloat16, device=DEV)
+    pt = torch.zeros(B, (max(ctxs) + ps - 1) // ps, dtype=torch.int32,
+                     device=DEV)
+    nxt = 1
+    for b, c in enumerate(ctxs):
+        n = (c + ps - 1) // ps
+        pt[b, :n] = torch.arange(nxt, nxt + n, dtype=torch.int32)
+        nxt += n
+    q = torch.randn(B, Hq, hd, dtype=torch.bfloat16, device=DEV)
+    ctx_lens = torch.tensor(ctxs, dtype=torch.int32, device=DEV)
+    scale = hd ** -0.5
+    scratch = ops.DecodeScratch(B, Hq, hd, max(ctxs), DEV)
+    kc8 = kc.to(torch.float8_e4m3fn)
+    vc8t = _vt(vc).to(torch.float8_e4m3fn)
+    out8 = ops.paged_attention_decode(q, kc8, vc8t, pt, ctx_lens, scale,
+                                      scratch, v_transposed=True)
+    ref = ops.paged_attention_decode(
+        q, kc8.to(torch.bfloat16),
+        vc8t.to(torch.bfloat16).permute(0, 1, 3, 2).contiguous(),
+        pt, ctx_lens, scale, scratch)
+    assert_close(out8, ref, rtol=0.05, atol=0.05)
+
+
+@pytest.mark.parametrize("G,spec", [
+    (8, [(128, 128), (700, 1000)]),     # prefill32 GSPLIT=1
+    (4, [(257, 900), (64, 64)]),        # prefill32 GSPLIT=2
+    (2, [(100, 100)]),                  # prefill32 GSPLIT=4
+    (7, [(300, 500)]),                  # 16x16 fallback kernel
+])
+def test_attention_prefill_v_transposed(G, spec):
+    torch.manual_seed(5)
+    Hkv, ps, hd = 4, 64, 128
+    Hq = G * Hkv
+    qls = [s[0] for s in spec]
+    ctxs = [s[1] for s in spec]
+    P = sum((c + ps - 1) // ps for c in ctxs) + 1
+    kc = torch.randn(P, Hkv, ps, hd, dtype=torch.bfloat16, device=DEV)
+    vc = torch.randn(P, Hkv, ps, hd, dtype=torch.bfloat16, device=DEV)
+    pt = torch.zeros(len(spec), (max(ctxs) + ps - 1) // ps,
+                     dtype=torch.int32, device=DEV)
+    nxt = 1
+    for s, c in enumerate(ctxs):
+        n = (c + ps - 1) // ps
+        pt[s, :n] = torch.arange(nxt, nxt + n, dtype=torch.int32)
+        nxt += n
+    Tq = sum(qls)
+    q = torch.randn(Tq, Hq, hd, dtype=torch.bfloat16, device=DEV)
+    starts = [0]
+    for ql in qls[:-1]:
+        starts.append(starts[-1] + ql)
+    sqs = torch.tensor(starts, dtype=torch.int32, device=DEV)
+    sql = torch.tensor(qls, dtype=torch.int32, device=DEV)
+    scl = torch.tensor(ctxs, dtype=torch.int32, device=DEV)
+    scale = hd ** -0.5
+    out = ops.attention_prefill_paged(q, kc, _vt(vc), pt, sqs, sql, scl,
+                                      scale, v_transposed=True)
+    ref = torch_ref.attention_prefill_paged(q, kc, vc, pt, sqs, sql, scl,
+                                            scale)
+    assert_close(out, ref, rtol=0.05, atol=0.05)
+
+
+def test_append_roundtrip_v_transposed():
+    """rope_append + kv_cache_append into d-major pages == token-major
+    result transposed."""
+    torch.manual_seed(6)
+    Hq, Hkv, hd, P, ps, T = 8, 2, 128, 16, 64, 53
+    qkv = torch.randn(T, (Hq + 2 * Hkv) * hd, dtype=torch.bfloat16,
+                      device=DEV)
+    pos = torch.randint(0, 500, (T,), dtype=torch.int32, device=DEV)
+    slots = torch.randperm(P * ps, device=DEV)[:T].to(torch.int64)
+    cos_sin = torch_ref.make_cos_sin_cache(512, hd, 10000.0, device=DEV)
+    kc = torch.zeros(P, Hkv, ps, hd, dtype=torch.bfloat16, device=DEV)
+    vc = torch.zeros_like(kc)
+    kct = torch.zeros_like(kc)
+    vct = torch.zeros(P, Hkv, hd, ps, dtype=torch.bfloat16, device=DEV)
+    q0 = ops.rope_append_qkv(qkv, None, pos, slots, cos_sin, kc, vc,
+                             Hq, Hkv, hd)
+    q1 = ops.rope_append_qkv(qkv, None, pos, slots, cos_sin, kct, vct,
+                             Hq, Hkv, hd, v_transposed=True)
+    assert_close(q0, q1, rtol=0, atol=0)
+    assert_close(kct, kc, rtol=0, atol=0)
+    assert_close(vct, _vt(vc), rtol=0, atol=0)
+    # plain kv_cache_append
+    k = torch.randn(T, Hkv, hd, dtype=torch.bfloat16, device=DEV)
+    v = torch.randn(T, Hkv, hd, dtype=torch.bfloat16, device=DEV)
+    vc.zero_(); vct.zero_()
+    ops.kv_cache_append(kc, vc, k, v, slots)
+    ops.kv_cache_append(kct, vct, k, v, slots, v_transposed=True)
+    assert_close(vct, _vt(vc), rtol=0, atol=0)
