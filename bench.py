@@ -48,6 +48,8 @@ def parse_args():
                    choices=["auto", "fp8"],
                    help="fp8 KV measurement runs are SUPPLEMENTARY: the "
                         "flagship headline stays bf16 (BASELINE config)")
+    p.add_argument("--kv-v-layout", default="auto", choices=["auto", "never"],
+                   help="V-page layout: auto = d-major on GPU when supported")
     p.add_argument("--device", default=None, help="cpu for gloo testing")
     p.add_argument("--max-batched-tokens", type=int, default=0)
     p.add_argument("--moe-ep", action="store_true",
@@ -69,6 +71,7 @@ def make_cfg(args, mc, device, world=1, rank=0, worker_type="aggregated",
         kv_pool_pages=args.kv_pool_pages,
         enable_prefix_caching=False,  # synthetic distinct prompts
         kv_cache_dtype=args.kv_cache_dtype,
+        kv_v_layout=args.kv_v_layout,
         dtype="bfloat16" if device.startswith("cuda") else "float32",
         worker_type=worker_type, tp_size=world, tp_rank=rank)
 

@@ -45,6 +45,9 @@ def build_parser():
                    choices=["auto", "fp8"],
                    help="fp8 stores the paged KV cache as OCP e4m3 "
                         "(half the decode HBM bytes; GPU only)")
+    p.add_argument("--kv-v-layout", default="auto",
+                   choices=["auto", "never"],
+                   help="V-page layout: auto = d-major on GPU when supported")
     p.add_argument("--dtype", default=None,
                    help="bfloat16 | float32 (default: bf16 on GPU, "
                         "fp32 on CPU)")
@@ -127,6 +130,7 @@ def make_engine_from_args(args, tp=None) -> LLMEngine:
         kv_pool_pages=args.kv_pool_pages,
         gpu_mem_fraction=args.gpu_mem_fraction,
         kv_cache_dtype=args.kv_cache_dtype,
+        kv_v_layout=args.kv_v_layout,
         enable_prefix_caching=not args.no_prefix_caching,
         enable_hip_graphs=not args.no_hip_graphs,
         worker_type=args.worker_type,
