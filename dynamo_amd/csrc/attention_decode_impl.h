@@ -619,8 +619,13 @@ inline int mfma_lds_bytes(int G, int hd) {
 // write side pays ~32x DRAM amplification ONLY for single-token decode
 // appends (~10 MB/step vs 22 GB/step of V reads); prefill appends cover
 // whole 64B lines in L2 before eviction.
+// LG2: softmax in the log2 domain + interior-tile mask elision. The
+// per-element chain mul(scale)+sub+mul(log2e)+exp collapses to
+// fma(s, scale*log2e, -m') + v_exp (saves 16 VALU/tile), and interior
+// tiles (t0+32 <= slab_end) skip the 8 cmp+sel masks. m is converted
+// back to the natural-log domain at the merge so phase2 is unchanged.
 template <int DEFER = 1, int PRIO = 1, int KPF = 0, int FP8 = 0,
-          int VS = 80, int XK2 = 0, int VT = 0>
+          int VS = 80, int XK2 = 0, int VT = 0, int LG2 = 0>
 __global__ __launch_bounds__(kBlock) void paged_decode_mfma_swapped(
     float* __restrict__ partial, float* __restrict__ ml,
     short* __restrict__ out, const short* __restrict__ q,
@@ -776,6 +781,7 @@ __global__ __launch_bounds__(kBlock) void paged_decode_mfma_swapped(
 
   // VT-only tile compute (QK -> softmax -> P pack -> PV from registers):
   // a specialization of the main-loop body below; keep the two in sync.
+  const float scale2 = scale * 1.44269504f;  // scale * log2(e)
   auto tile_vt = [&](int t0, const short8 (&kf)[8], const short8 (&vf)[8]) {
     f32x4 sA{0.f, 0.f, 0.f, 0.f}, sB{0.f, 0.f, 0.f, 0.f};
     if constexpr (PRIO) __builtin_amdgcn_s_setprio(1);
@@ -796,33 +802,78 @@ __global__ __launch_bounds__(kBlock) void paged_decode_mfma_swapped(
     if constexpr (PRIO) __builtin_amdgcn_s_setprio(0);
     float pA[4], pB[4];
     float mt = kNegInf;
+    if constexpr (LG2) {
+      // raw (unscaled) values; max commutes with the positive scale
+      const bool tail = t0 + 32 > slab_end;
+      if (!tail) {
 #pragma unroll
-    for (int r = 0; r < 4; r++) {
-      const int tokA = t0 + lg * 4 + r, tokB = t0 + 16 + lg * 4 + r;
-      pA[r] = (tokA < slab_end) ? sA[r] * scale : kNegInf;
-      pB[r] = (tokB < slab_end) ? sB[r] * scale : kNegInf;
-      mt = fmaxf(mt, fmaxf(pA[r], pB[r]));
-    }
-    mt = fmaxf(mt, __shfl_xor(mt, 16, WAVE_SIZE));
-    mt = fmaxf(mt, __shfl_xor(mt, 32, WAVE_SIZE));
-    const float thr = DEFER ? 8.0f : 0.0f;
-    if (mt > m_run + thr) {
-      const float corr = (m_run <= kNegInf * 0.5f) ? 0.f : __expf(m_run - mt);
-      l_run *= corr;
+        for (int r = 0; r < 4; r++) {
+          pA[r] = sA[r];
+          pB[r] = sB[r];
+          mt = fmaxf(mt, fmaxf(pA[r], pB[r]));
+        }
+      } else {
 #pragma unroll
-      for (int d = 0; d < 8; d++) acc[d] *= corr;
-      m_run = mt;
-    }
-    float psum = 0.f;
+        for (int r = 0; r < 4; r++) {
+          const int tokA = t0 + lg * 4 + r, tokB = t0 + 16 + lg * 4 + r;
+          pA[r] = (tokA < slab_end) ? sA[r] : kNegInf;
+          pB[r] = (tokB < slab_end) ? sB[r] : kNegInf;
+          mt = fmaxf(mt, fmaxf(pA[r], pB[r]));
+        }
+      }
+      mt = fmaxf(mt, __shfl_xor(mt, 16, WAVE_SIZE));
+      mt = fmaxf(mt, __shfl_xor(mt, 32, WAVE_SIZE));
+      const float mts = (mt <= kNegInf * 0.5f) ? kNegInf : mt * scale2;
+      const float thr = DEFER ? 8.0f * 1.44269504f : 0.0f;
+      if (mts > m_run + thr) {
+        const float corr =
+            (m_run <= kNegInf * 0.5f) ? 0.f : __builtin_amdgcn_exp2f(m_run - mts);
+        l_run *= corr;
 #pragma unroll
-    for (int r = 0; r < 4; r++) {
-      pA[r] = (pA[r] > kNegInf * 0.5f) ? __expf(pA[r] - m_run) : 0.f;
-      pB[r] = (pB[r] > kNegInf * 0.5f) ? __expf(pB[r] - m_run) : 0.f;
-      psum += pA[r] + pB[r];
+        for (int d = 0; d < 8; d++) acc[d] *= corr;
+        m_run = mts;
+      }
+      float psum = 0.f;
+#pragma unroll
+      for (int r = 0; r < 4; r++) {
+        pA[r] = (pA[r] > kNegInf * 0.5f)
+                    ? __builtin_amdgcn_exp2f(fmaf(pA[r], scale2, -m_run)) : 0.f;
+        pB[r] = (pB[r] > kNegInf * 0.5f)
+                    ? __builtin_amdgcn_exp2f(fmaf(pB[r], scale2, -m_run)) : 0.f;
+        psum += pA[r] + pB[r];
+      }
+      psum += __shfl_xor(psum, 16, WAVE_SIZE);
+      psum += __shfl_xor(psum, 32, WAVE_SIZE);
+      l_run += psum;
+    } else {
+#pragma unroll
+      for (int r = 0; r < 4; r++) {
+        const int tokA = t0 + lg * 4 + r, tokB = t0 + 16 + lg * 4 + r;
+        pA[r] = (tokA < slab_end) ? sA[r] * scale : kNegInf;
+        pB[r] = (tokB < slab_end) ? sB[r] * scale : kNegInf;
+        mt = fmaxf(mt, fmaxf(pA[r], pB[r]));
+      }
+      mt = fmaxf(mt, __shfl_xor(mt, 16, WAVE_SIZE));
+      mt = fmaxf(mt, __shfl_xor(mt, 32, WAVE_SIZE));
+      const float thr = DEFER ? 8.0f : 0.0f;
+      if (mt > m_run + thr) {
+        const float corr = (m_run <= kNegInf * 0.5f) ? 0.f : __expf(m_run - mt);
+        l_run *= corr;
+#pragma unroll
+        for (int d = 0; d < 8; d++) acc[d] *= corr;
+        m_run = mt;
+      }
+      float psum = 0.f;
+#pragma unroll
+      for (int r = 0; r < 4; r++) {
+        pA[r] = (pA[r] > kNegInf * 0.5f) ? __expf(pA[r] - m_run) : 0.f;
+        pB[r] = (pB[r] > kNegInf * 0.5f) ? __expf(pB[r] - m_run) : 0.f;
+        psum += pA[r] + pB[r];
+      }
+      psum += __shfl_xor(psum, 16, WAVE_SIZE);
+      psum += __shfl_xor(psum, 32, WAVE_SIZE);
+      l_run += psum;
     }
-    psum += __shfl_xor(psum, 16, WAVE_SIZE);
-    psum += __shfl_xor(psum, 32, WAVE_SIZE);
-    l_run += psum;
     auto cvtpk = [](float a, float bb) {
       unsigned int r;
       asm("v_cvt_pk_bf16_f32 %0, %1, %2" : "=v"(r) : "v"(a), "v"(bb));
