@@ -101,7 +101,7 @@ static Bufs make(int G) {
 // swapped-operand MFMA variant: bench + elementwise compare vs the proven
 // paged_decode_mfma output
 template <int G, int DEFER = 1, int PRIO = 1, int KPF = 0, int VS = 80, int XK2 = 0,
-          int VT = 0, int LG2 = 0>
+          int VT = 0, int LG2 = 0, int MINW = 1>
 static void run_mfma_swapped(const Bufs& bf, bool check) {
   dim3 grid(B, Hkv, bf.C);
   const int lds = VT ? mfma_swapped_vt_lds_bytes(G, HD)
@@ -109,10 +109,10 @@ static void run_mfma_swapped(const Bufs& bf, bool check) {
   const int iters = 30;
   if (lds > 65536)
     (void)hipFuncSetAttribute(
-        reinterpret_cast<const void*>(&paged_decode_mfma_swapped<DEFER, PRIO, KPF, 0, VS, XK2, VT, LG2>),
+        reinterpret_cast<const void*>(&paged_decode_mfma_swapped<DEFER, PRIO, KPF, 0, VS, XK2, VT, LG2, MINW>),
         hipFuncAttributeMaxDynamicSharedMemorySize, lds);
   auto launch = [&] {
-    paged_decode_mfma_swapped<DEFER, PRIO, KPF, 0, VS, XK2, VT, LG2><<<grid, kBlock, lds>>>(
+    paged_decode_mfma_swapped<DEFER, PRIO, KPF, 0, VS, XK2, VT, LG2, MINW><<<grid, kBlock, lds>>>(
         bf.partial, bf.ml, bf.out, bf.q, bf.kc, VT ? bf.vct : bf.vc, bf.pt, bf.ctx,
         0.0883883f, kChunk, G, B, Hkv, bf.C, CTX / PS, 6, HD);
     paged_decode_phase2<<<dim3(B, G * Hkv), 128>>>(
@@ -165,8 +165,8 @@ static void run_mfma_swapped(const Bufs& bf, bool check) {
   CK(hipEventElapsedTime(&ms, e0, e1));
   double t = ms / 1000.0 / iters;
   double gb = 2.0 * B * CTX * Hkv * HD * 2 / 1e9;
-  printf("G%d MFMA_SW DF%d PR%d KP%d VS%d X%d VT%d L%d %8.1f us  %7.0f GB/s\n", G,
-         DEFER, PRIO, KPF, VS, XK2, VT, LG2, t * 1e6, gb / t);
+  printf("G%d MFMA_SW DF%d PR%d KP%d VS%d X%d VT%d L%d W%d %8.1f us  %7.0f GB/s\n", G,
+         DEFER, PRIO, KPF, VS, XK2, VT, LG2, MINW, t * 1e6, gb / t);
   fflush(stdout);
 }
 
@@ -249,7 +249,10 @@ int main() {
     run_mfma_swapped<8, 0, 0, 0, 72, 0, 2>(bf, true);   // VT2 check
     run_mfma_swapped<8, 0, 0, 0, 72, 0, 2, 1>(bf, true);  // VT2+LG2 check
     run_mfma_swapped<8, 1, 1, 0, 72, 0, 2>(bf, false);  // VT2 pipelined
-    run_mfma_swapped<8, 1, 1, 0, 72, 0, 2, 1>(bf, false);  // VT2+LG2
+    run_mfma_swapped<8, 0, 0, 0, 72, 0, 2, 0, 2>(bf, true);   // VT2 minw2 chk
+    run_mfma_swapped<8, 1, 1, 0, 72, 0, 2, 0, 2>(bf, false);  // VT2 minw2
+    run_mfma_swapped<8, 1, 1, 0, 72, 0, 1, 0, 3>(bf, false);  // VT1 minw3
+    run_mfma_swapped<8, 1, 1, 0, 72, 0, 1, 0, 4>(bf, false);  // VT1 minw4
     run<8, 16, 4, 2>(bf, "");
     run<8, 16, 4, 3>(bf, "");
     run<8, 16, 4, 4>(bf, "");
@@ -279,6 +282,8 @@ int main() {
     run_mfma_swapped<4, 0, 0, 0, 72, 0, 2>(bf, true);   // G4 VT2 check
     run_mfma_swapped<4, 1, 1, 0, 72, 0, 2>(bf, false);  // G4 VT2
     run_mfma_swapped<4, 1, 1, 0, 72, 0, 2, 1>(bf, false);  // G4 VT2+LG2
+    run_mfma_swapped<4, 1, 1, 0, 72, 0, 2, 0, 2>(bf, false);  // G4 VT2 minw2
+    run_mfma_swapped<4, 1, 1, 0, 72, 0, 1, 0, 3>(bf, false);  // G4 VT1 minw3
     run_mfma_swapped<2, 0, 0, 0, 72, 0, 2>(bf, true);   // G2 VT2 check
     run_mfma_swapped<2, 1, 1, 0, 72, 0, 2>(bf, false);  // G2 VT2
   }

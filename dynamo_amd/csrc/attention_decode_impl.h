@@ -624,9 +624,12 @@ inline int mfma_lds_bytes(int G, int hd) {
 // fma(s, scale*log2e, -m') + v_exp (saves 16 VALU/tile), and interior
 // tiles (t0+32 <= slab_end) skip the 8 cmp+sel masks. m is converted
 // back to the natural-log domain at the merge so phase2 is unchanged.
+// MINW: minimum waves/SIMD the register allocator must honor (the VT2
+// double-buffer costs 285 VGPR+AGPR = 1 wave/SIMD; forcing 2 trades
+// registers for occupancy - decide by measurement, watch ScratchSize)
 template <int DEFER = 1, int PRIO = 1, int KPF = 0, int FP8 = 0,
-          int VS = 80, int XK2 = 0, int VT = 0, int LG2 = 0>
-__global__ __launch_bounds__(kBlock) void paged_decode_mfma_swapped(
+          int VS = 80, int XK2 = 0, int VT = 0, int LG2 = 0, int MINW = 1>
+__global__ __launch_bounds__(kBlock, MINW) void paged_decode_mfma_swapped(
     float* __restrict__ partial, float* __restrict__ ml,
     short* __restrict__ out, const short* __restrict__ q,
     const short* __restrict__ kcache, const short* __restrict__ vcache,
