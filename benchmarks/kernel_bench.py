@@ -50,12 +50,15 @@ def bench_decode_attn(B=16, ctx=8192, Hq=64, Hkv=8, ps=64):
     scratch = ops.DecodeScratch(B, Hq, hd, ctx, "cuda")
     t = timeit(lambda: ops.paged_attention_decode(
         q, kc, vc, pt, ctxl, hd ** -0.5, scratch))
+    vct = vc.permute(0, 1, 3, 2).contiguous()
+    tv = timeit(lambda: ops.paged_attention_decode(
+        q, kc, vct, pt, ctxl, hd ** -0.5, scratch, v_transposed=True))
     gb = 2 * B * ctx * Hkv * hd * 2 / 1e9  # K+V bytes
     print(f"decode_attn  [B{B} ctx{ctx} Hq{Hq}/{Hkv}]: {t*1e6:8.1f} us  "
-          f"{gb/t:7.2f} GB/s (KV stream)")
+          f"{gb/t:7.2f} GB/s | vt {tv*1e6:8.1f} us {gb/tv:7.2f} GB/s")
 
 
-def bench_prefill_attn(S=8192, Hq=32, Hkv=8, ps=64):
+def bench_prefill_attn(S=8192, Hq=32, Hkv=8, ps=64, v_transposed=False):
     hd = 128
     npages = (S + ps - 1) // ps
     kc = torch.randn(npages, Hkv, ps, hd, dtype=torch.bfloat16, device="cuda")
@@ -67,12 +70,15 @@ def bench_prefill_attn(S=8192, Hq=32, Hkv=8, ps=64):
     ctxl = torch.tensor([S], dtype=torch.int32, device="cuda")
     tiles = ops.build_prefill_tiles([S], "cuda",
                                     ops.prefill_tile_rows(Hq, Hkv))
+    if v_transposed:
+        vc = vc.permute(0, 1, 3, 2).contiguous()
     t = timeit(lambda: ops.attention_prefill_paged(
-        q, kc, vc, pt, starts, qlen, ctxl, hd ** -0.5, tiles), iters=5)
+        q, kc, vc, pt, starts, qlen, ctxl, hd ** -0.5, tiles,
+        v_transposed=v_transposed), iters=5)
     # causal flops: 2 gemms * 2*S*S/2*hd per head
     fl = 2 * 2 * Hq * (S * S / 2) * hd
-    print(f"prefill_attn [S{S} Hq{Hq}/{Hkv}]: {t*1e3:8.2f} ms  "
-          f"{fl/t/1e12:7.1f} TFLOP/s (causal)")
+    print(f"prefill_attn [S{S} Hq{Hq}/{Hkv} vt{int(v_transposed)}]: "
+          f"{t*1e3:8.2f} ms  {fl/t/1e12:7.1f} TFLOP/s (causal)")
 
 
 def bench_kv_append(T=8192, Hkv=8, ps=64):
@@ -119,6 +125,9 @@ if __name__ == "__main__":
         bench_decode_attn(B=16, ctx=8192, Hq=32, Hkv=8)
     if w in ("all", "prefill"):
         bench_prefill_attn()
+        bench_prefill_attn(v_transposed=True)
+        bench_prefill_attn(S=8192, Hq=64)
+        bench_prefill_attn(S=8192, Hq=64, v_transposed=True)
         bench_prefill_attn(S=2048)
         bench_prefill_attn(Hq=64)          # GQA 8 (llama-70b TP1)
         bench_prefill_attn(S=2048, Hq=64)
