@@ -252,7 +252,9 @@ int main() {
     run_mfma_swapped<8, 0, 0, 0, 72, 0, 2, 0, 2>(bf, true);   // VT2 minw2 chk
     run_mfma_swapped<8, 1, 1, 0, 72, 0, 2, 0, 2>(bf, false);  // VT2 minw2
     run_mfma_swapped<8, 1, 1, 0, 72, 0, 1, 0, 3>(bf, false);  // VT1 minw3
-    run_mfma_swapped<8, 1, 1, 0, 72, 0, 1, 0, 4>(bf, false);  // VT1 minw4
+    run_mfma_swapped<8, 0, 0, 0, 72, 0, 3, 0, 1>(bf, true);   // VT3 64tok chk
+    run_mfma_swapped<8, 1, 1, 0, 72, 0, 3, 0, 1>(bf, false);  // VT3 64tok
+    run_mfma_swapped<8, 1, 1, 0, 72, 0, 3, 0, 2>(bf, false);  // VT3 minw2
     run<8, 16, 4, 2>(bf, "");
     run<8, 16, 4, 3>(bf, "");
     run<8, 16, 4, 4>(bf, "");
@@ -284,6 +286,8 @@ int main() {
     run_mfma_swapped<4, 1, 1, 0, 72, 0, 2, 1>(bf, false);  // G4 VT2+LG2
     run_mfma_swapped<4, 1, 1, 0, 72, 0, 2, 0, 2>(bf, false);  // G4 VT2 minw2
     run_mfma_swapped<4, 1, 1, 0, 72, 0, 1, 0, 3>(bf, false);  // G4 VT1 minw3
+    run_mfma_swapped<4, 1, 1, 0, 72, 0, 3, 0, 1>(bf, false);  // G4 VT3
+    run_mfma_swapped<4, 1, 1, 0, 72, 0, 3, 0, 2>(bf, false);  // G4 VT3 minw2
     run_mfma_swapped<2, 0, 0, 0, 72, 0, 2>(bf, true);   // G2 VT2 check
     run_mfma_swapped<2, 1, 1, 0, 72, 0, 2>(bf, false);  // G2 VT2
   }

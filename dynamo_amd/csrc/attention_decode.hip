@@ -46,9 +46,8 @@ void paged_attention_decode(torch::Tensor out, torch::Tensor q,
   dim3 grid(B, Hkv, C);
 
   if (v_transposed) {
-    // d-major V pages: the software-pipelined swapped kernel (VT=2) is the
-    // only consumer — sweep: G8 4476, G4 4987, G2 5069 GB/s vs 3906-3951
-    // for the token-major paths (benchmarks/decode_sweep.hip)
+    // d-major V pages: the 64-token-tile swapped kernel (VT=3) is the
+    // production consumer (benchmarks/decode_sweep.hip ladder)
     TORCH_CHECK(vcache.size(2) == hd && vcache.size(3) == ps,
                 "v_transposed expects vcache [P, Hkv, hd, ps]");
     TORCH_CHECK(ps % 32 == 0 && hd == 128 && G >= 2 && G <= 16,
@@ -67,8 +66,13 @@ void paged_attention_decode(torch::Tensor out, torch::Tensor q,
           page_table.data_ptr<int32_t>(), ctx_lens.data_ptr<int32_t>(),
           (float)scale, chunk, G, B, Hkv, C, max_pages, log2_ps, hd);
     };
-    if (fp8) launch_vt(&paged_decode_mfma_swapped<1, 1, 0, 1, 72, 0, 2>);
-    else launch_vt(&paged_decode_mfma_swapped<1, 1, 0, 0, 72, 0, 2>);
+    // VT3 = 64-token tiles (32 loads in flight per batch) at 2
+    // waves/SIMD (MINW=2, 232 VGPR no-spill): sweep G8 5131 / G4 4864
+    // GB/s vs 4922/4511 for the 32-token VT2 on the same box
+    if (fp8)
+      launch_vt(&paged_decode_mfma_swapped<1, 1, 0, 1, 72, 0, 3, 0, 2>);
+    else
+      launch_vt(&paged_decode_mfma_swapped<1, 1, 0, 0, 72, 0, 3, 0, 2>);
     HIP_CHECK_KERNEL();
     if (C > 1) {
       dim3 grid2(B, Hq);

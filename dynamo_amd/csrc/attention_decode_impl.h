@@ -900,7 +900,149 @@ __global__ __launch_bounds__(kBlock, MINW) void paged_decode_mfma_swapped(
     if constexpr (PRIO) __builtin_amdgcn_s_setprio(0);
   };
 
-  if constexpr (VT == 2) {
+  // ---- VT3: 64-token tiles ----------------------------------------
+  // Rationale (PMC): at 32-token tiles the wave issues only 16 b128
+  // loads per compute batch, so load-issue duty caps effective BW at
+  // ~77% of the pattern ceiling (WAIT_ANY-dominated). 64-token tiles
+  // issue 32 loads back-to-back (2 KB in flight per wave) before the
+  // QK/softmax/PV batch. Single-buffered; MINW=2 keeps 2 waves/SIMD.
+  auto tile_vt64 = [&](int t0, const short8 (&kf)[16],
+                       const short8 (&vf)[16]) {
+    f32x4 sT[4];
+    if constexpr (PRIO) __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int tg = 0; tg < 4; tg++) {
+      f32x4 acc_s{0.f, 0.f, 0.f, 0.f};
+      if (t0 + tg * 16 < slab_end) {
+#pragma unroll
+        for (int kc = 0; kc < 4; kc++) {
+          short8 ka = kf[tg * 4 + kc];
+          acc_s = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              *reinterpret_cast<bf16x8_t*>(&ka), q_frag[kc], acc_s, 0, 0, 0);
+        }
+      }
+      sT[tg] = acc_s;
+    }
+    if constexpr (PRIO) __builtin_amdgcn_s_setprio(0);
+    float pv[4][4];
+    float mt = kNegInf;
+    const bool tail = t0 + 64 > slab_end;
+    if (!tail) {
+#pragma unroll
+      for (int tg = 0; tg < 4; tg++)
+#pragma unroll
+        for (int r = 0; r < 4; r++) {
+          pv[tg][r] = sT[tg][r] * scale;
+          mt = fmaxf(mt, pv[tg][r]);
+        }
+    } else {
+#pragma unroll
+      for (int tg = 0; tg < 4; tg++)
+#pragma unroll
+        for (int r = 0; r < 4; r++) {
+          const int tok = t0 + tg * 16 + lg * 4 + r;
+          pv[tg][r] = (tok < slab_end) ? sT[tg][r] * scale : kNegInf;
+          mt = fmaxf(mt, pv[tg][r]);
+        }
+    }
+    mt = fmaxf(mt, __shfl_xor(mt, 16, WAVE_SIZE));
+    mt = fmaxf(mt, __shfl_xor(mt, 32, WAVE_SIZE));
+    const float thr = DEFER ? 8.0f : 0.0f;
+    if (mt > m_run + thr) {
+      const float corr = (m_run <= kNegInf * 0.5f) ? 0.f : __expf(m_run - mt);
+      l_run *= corr;
+#pragma unroll
+      for (int d = 0; d < 8; d++) acc[d] *= corr;
+      m_run = mt;
+    }
+    float psum = 0.f;
+#pragma unroll
+    for (int tg = 0; tg < 4; tg++)
+#pragma unroll
+      for (int r = 0; r < 4; r++) {
+        pv[tg][r] = (pv[tg][r] > kNegInf * 0.5f)
+                        ? __expf(pv[tg][r] - m_run) : 0.f;
+        psum += pv[tg][r];
+      }
+    psum += __shfl_xor(psum, 16, WAVE_SIZE);
+    psum += __shfl_xor(psum, 32, WAVE_SIZE);
+    l_run += psum;
+    auto cvtpk = [](float a, float bb) {
+      unsigned int r;
+      asm("v_cvt_pk_bf16_f32 %0, %1, %2" : "=v"(r) : "v"(a), "v"(bb));
+      return r;
+    };
+    // two 32-token PV B-frags, each combining two 16-token S tiles via
+    // the same cvt_pk + permlane swap network as the 32-token path
+    bf16x8_t p_frag[2];
+#pragma unroll
+    for (int tc = 0; tc < 2; tc++) {
+      const float* pa = pv[tc * 2];
+      const float* pb = pv[tc * 2 + 1];
+      const unsigned int A0 = cvtpk(pa[0], pa[1]), A1 = cvtpk(pa[2], pa[3]);
+      const unsigned int B0 = cvtpk(pb[0], pb[1]), B1 = cvtpk(pb[2], pb[3]);
+      uint2_t s1 = __builtin_amdgcn_permlane32_swap(A0, B0, false, false);
+      uint2_t s2 = __builtin_amdgcn_permlane32_swap(A1, B1, false, false);
+      uint2_t f02 = __builtin_amdgcn_permlane16_swap(s1.x, s1.y, false, false);
+      uint2_t f13 = __builtin_amdgcn_permlane16_swap(s2.x, s2.y, false, false);
+      unsigned int w[4] = {f02.x, f13.x, f02.y, f13.y};
+      p_frag[tc] = *reinterpret_cast<bf16x8_t*>(w);
+    }
+    if constexpr (PRIO) __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int da = 0; da < 8; da++)
+#pragma unroll
+      for (int tc = 0; tc < 2; tc++) {
+        short8 va_s = vf[da * 2 + tc];
+        acc[da] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            *reinterpret_cast<bf16x8_t*>(&va_s), p_frag[tc], acc[da], 0, 0, 0);
+      }
+    if constexpr (PRIO) __builtin_amdgcn_s_setprio(0);
+  };
+  auto load_k64 = [&](int t0_, short8 (&kf)[16]) {
+#pragma unroll
+    for (int tg = 0; tg < 4; tg++) {
+      const int tb = t0_ + tg * 16;
+      if (tb >= slab_end) break;
+      const int64_t pb = (((int64_t)pt[tb >> log2_ps] * Hkv + h) * ps) * hd;
+      const int tA = tb + lr;
+      const int64_t oA = pb + (int64_t)(tA & (ps - 1)) * hd;
+      const bool vA = tA < slab_end;
+#pragma unroll
+      for (int kc = 0; kc < 4; kc++)
+        kf[tg * 4 + kc] = vA ? ld8(kcache, oA + kc * 32 + lg * 8) : short8{};
+    }
+  };
+  auto load_v64 = [&](int t0_, short8 (&vf)[16]) {
+#pragma unroll
+    for (int tc = 0; tc < 2; tc++) {
+      const int tb = t0_ + tc * 32;
+      if (tb >= slab_end) break;
+      const int64_t pb = (((int64_t)pt[tb >> log2_ps] * Hkv + h) * ps) * hd;
+      const int tin = tb & (ps - 1);
+      const int rem = slab_end - tb;
+#pragma unroll
+      for (int da = 0; da < 8; da++) {
+        const int vrow = da * 16 + lr;
+        short8 v8 = ld8(vcache, pb + (int64_t)vrow * ps + tin + lg * 8);
+        if (rem < 32) {
+#pragma unroll
+          for (int j = 0; j < 8; j++)
+            if (lg * 8 + j >= rem) v8[j] = 0;
+        }
+        vf[da * 2 + tc] = v8;
+      }
+    }
+  };
+
+  if constexpr (VT == 3) {
+    short8 k64[16], v64[16];
+    for (int t0 = slab_start; t0 < slab_end; t0 += 64) {
+      load_k64(t0, k64);
+      load_v64(t0, v64);
+      tile_vt64(t0, k64, v64);
+    }
+  } else if constexpr (VT == 2) {
     // software-pipelined: tile i+1's K/V global loads are in flight while
     // tile i computes; two compile-time register sets (no runtime
     // indexing -> no scratch spill, unlike KPF)
