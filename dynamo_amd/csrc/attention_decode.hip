@@ -69,8 +69,18 @@ void paged_attention_decode(torch::Tensor out, torch::Tensor q,
     // VT3 = 64-token tiles (32 loads in flight per batch) at 2
     // waves/SIMD (MINW=2, 232 VGPR no-spill): sweep G8 5131 / G4 4864
     // GB/s vs 4922/4511 for the 32-token VT2 on the same box
+    static const int vt3v = [] {
+      const char* e = getenv("DYNAMO_VT3_VARIANT");
+      return e ? atoi(e) : 0;
+    }();
     if (fp8)
       launch_vt(&paged_decode_mfma_swapped<1, 1, 0, 1, 72, 0, 3, 0, 2>);
+    else if (vt3v == 1)
+      launch_vt(&paged_decode_mfma_swapped<0, 0, 0, 0, 72, 0, 3, 0, 2>);
+    else if (vt3v == 2)
+      launch_vt(&paged_decode_mfma_swapped<1, 1, 0, 0, 72, 0, 3, 0, 1>);
+    else if (vt3v == 3)
+      launch_vt(&paged_decode_mfma_swapped<1, 1, 0, 0, 72, 0, 2>);
     else
       launch_vt(&paged_decode_mfma_swapped<1, 1, 0, 0, 72, 0, 3, 0, 2>);
     HIP_CHECK_KERNEL();

@@ -1000,11 +1000,14 @@ __global__ __launch_bounds__(kBlock, MINW) void paged_decode_mfma_swapped(
     if constexpr (PRIO) __builtin_amdgcn_s_setprio(0);
   };
   auto load_k64 = [&](int t0_, short8 (&kf)[16]) {
+    // uniform (no break-in-unroll): groups past slab_end clamp their page
+    // lookup to the last valid token and mask values via vA
 #pragma unroll
     for (int tg = 0; tg < 4; tg++) {
       const int tb = t0_ + tg * 16;
-      if (tb >= slab_end) break;
-      const int64_t pb = (((int64_t)pt[tb >> log2_ps] * Hkv + h) * ps) * hd;
+      const int tsafe = tb < slab_end ? tb : slab_end - 1;
+      const int64_t pb =
+          (((int64_t)pt[tsafe >> log2_ps] * Hkv + h) * ps) * hd;
       const int tA = tb + lr;
       const int64_t oA = pb + (int64_t)(tA & (ps - 1)) * hd;
       const bool vA = tA < slab_end;
@@ -1014,12 +1017,16 @@ __global__ __launch_bounds__(kBlock, MINW) void paged_decode_mfma_swapped(
     }
   };
   auto load_v64 = [&](int t0_, short8 (&vf)[16]) {
+    // uniform: a 32-token group fully past slab_end clamps its page to the
+    // last valid 32-block (safe memory) and rem <= 0 zeroes every element
 #pragma unroll
     for (int tc = 0; tc < 2; tc++) {
       const int tb = t0_ + tc * 32;
-      if (tb >= slab_end) break;
-      const int64_t pb = (((int64_t)pt[tb >> log2_ps] * Hkv + h) * ps) * hd;
-      const int tin = tb & (ps - 1);
+      const int tlast = ((slab_end - 1) >> 5) << 5;  // last valid 32-block
+      const int tsafe = tb < slab_end ? tb : tlast;
+      const int64_t pb =
+          (((int64_t)pt[tsafe >> log2_ps] * Hkv + h) * ps) * hd;
+      const int tin = tsafe & (ps - 1);
       const int rem = slab_end - tb;
 #pragma unroll
       for (int da = 0; da < 8; da++) {
