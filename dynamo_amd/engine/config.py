@@ -124,6 +124,9 @@ class EngineConfig:
     enable_prefix_caching: bool = True
     enable_hip_graphs: bool = True
     host_cache_pages: int = 0           # KVBM G2 tier size (0 = disabled)
+    # G2 eviction policy: "lru", or "tinylfu" = LRU victim + TinyLFU
+    # admission filter (reference kvbm-logical tinylfu.rs parity)
+    host_cache_policy: str = "lru"
     disk_cache_pages: int = 0           # KVBM G3 tier size (0 = disabled)
     disk_cache_path: str = ""           # G3 backing file (required if G3 on)
     object_cache_dir: str = ""          # G4 shared object store (disabled="")

@@ -62,7 +62,8 @@ class LLMEngine:
                                         cfg.host_cache_pages,
                                         disk_path=cfg.disk_cache_path,
                                         num_disk_pages=cfg.disk_cache_pages,
-                                        object_dir=cfg.object_cache_dir)
+                                        object_dir=cfg.object_cache_dir,
+                                        policy=cfg.host_cache_policy)
             self.alloc.host_tier = self.host_tier
         self.scheduler = Scheduler(cfg, self.alloc)
         self.requests: Dict[str, Request] = {}

@@ -23,10 +23,53 @@ import torch
 log = logging.getLogger("dynamo_amd.kvbm")
 
 
+class TinyLFU:
+    """W-TinyLFU frequency sketch + admission filter (reference parity:
+    lib/kvbm-logical/src/tinylfu.rs). A 4-row count-min sketch of 8-bit
+    counters estimates page-hash access frequency; counters halve after a
+    sample window (aging). On a full cache the incoming page is ADMITTED
+    only if its estimated frequency beats the LRU victim's, which keeps
+    one-shot scans from flushing hot prefixes out of the host tier."""
+
+    ROWS = 4
+    CAP = 255
+
+    def __init__(self, size_hint: int):
+        self.width = max(64, 1 << (max(1, size_hint).bit_length() + 2))
+        self.mask = self.width - 1
+        self.counts = [bytearray(self.width) for _ in range(self.ROWS)]
+        self.sample = max(256, 8 * size_hint)
+        self.ops = 0
+
+    def _idx(self, h: int, i: int) -> int:
+        x = (h ^ (0x9e3779b97f4a7c15 * (i + 1))) & (1 << 64) - 1
+        x = (x ^ (x >> 33)) * 0xff51afd7ed558ccd & (1 << 64) - 1
+        return (x >> 32) & self.mask
+
+    def touch(self, h: int):
+        for i in range(self.ROWS):
+            row = self.counts[i]
+            j = self._idx(h, i)
+            if row[j] < self.CAP:
+                row[j] += 1
+        self.ops += 1
+        if self.ops >= self.sample:
+            self.ops = 0
+            for row in self.counts:
+                for j in range(self.width):
+                    row[j] >>= 1
+
+    def estimate(self, h: int) -> int:
+        return min(self.counts[i][self._idx(h, i)] for i in range(self.ROWS))
+
+    def admit(self, new_h: int, victim_h: int) -> bool:
+        return self.estimate(new_h) >= self.estimate(victim_h)
+
+
 class HostKVTier:
     def __init__(self, kv_pool, num_host_pages: int,
                  disk_path: str = "", num_disk_pages: int = 0,
-                 object_dir: str = ""):
+                 object_dir: str = "", policy: str = "lru"):
         self.pool = kv_pool
         self.device = kv_pool.device
         L, two, P, hkv, ps, hd = kv_pool.shape
@@ -42,6 +85,8 @@ class HostKVTier:
         # hash -> host page (insertion-ordered for LRU)
         self.map: "OrderedDict[int, int]" = OrderedDict()
         self.free = list(range(num_host_pages))
+        # eviction policy: plain LRU, or LRU victim + TinyLFU admission
+        self.lfu = TinyLFU(num_host_pages) if policy == "tinylfu" else None
         if self.device.type == "cuda":
             self.stream = torch.cuda.Stream(device=self.device)
             self.staging = torch.empty(self.page_elems, dtype=kv_pool.dtype,
@@ -85,11 +130,19 @@ class HostKVTier:
             self.stream.synchronize()
         return self.host[hp].view(torch.uint8).numpy().tobytes()
 
-    def _alloc_host(self) -> Optional[int]:
+    def _alloc_host(self, new_h: Optional[int] = None) -> Optional[int]:
         if self.free:
             return self.free.pop()
         if self.map:
-            old_h, hp = self.map.popitem(last=False)  # LRU
+            victim_h = next(iter(self.map))       # LRU victim
+            if (self.lfu is not None and new_h is not None
+                    and not self.lfu.admit(new_h, victim_h)):
+                # TinyLFU admission: the incoming page is colder than the
+                # coldest resident - refuse the insert instead of evicting
+                self.stats["admission_rejects"] = (
+                    self.stats.get("admission_rejects", 0) + 1)
+                return None
+            old_h, hp = self.map.popitem(last=False)
             self.stats["evicted_host"] += 1
             if self.disk is not None and self.disk.put(old_h,
                                                        self._page_bytes(hp)):
@@ -108,7 +161,7 @@ class HostKVTier:
                 self.stats["onboarded_object"] += 1
         if data is None:
             return None
-        hp = self._alloc_host()
+        hp = self._alloc_host(h)
         if hp is None:
             return None
         self.host[hp].view(torch.uint8).copy_(
@@ -120,9 +173,11 @@ class HostKVTier:
 
     # -- device -> host (called from PageAllocator eviction hook) --------
     def offload(self, pid: int, h: int):
+        if self.lfu is not None:
+            self.lfu.touch(h)
         if h in self.map:
             return
-        hp = self._alloc_host()
+        hp = self._alloc_host(h)
         if hp is None:
             return
         if self.device.type == "cuda":
@@ -146,6 +201,8 @@ class HostKVTier:
 
     # -- host -> device (prefix-cache onboard) ---------------------------
     def onboard(self, h: int, pid: int) -> bool:
+        if self.lfu is not None:
+            self.lfu.touch(h)
         hp = self.map.get(h)
         if hp is None:
             hp = self._onboard_from_disk(h)   # G3 -> G2 promote

@@ -140,3 +140,51 @@ def test_object_tier_unit(tmp_path):
     # same content-addressing across instances (shared store semantics)
     t2 = ObjectKVTier(str(tmp_path / "o"), page_bytes=16)
     assert t2.get(123) == b"a" * 16
+
+
+def test_tinylfu_sketch_basics():
+    from dynamo_amd.kvbm.host_tier import TinyLFU
+    lfu = TinyLFU(16)
+    for _ in range(5):
+        lfu.touch(111)
+    lfu.touch(222)
+    assert lfu.estimate(111) >= 5
+    assert lfu.estimate(222) <= 2
+    assert lfu.admit(111, 222)
+    assert not lfu.admit(333, 111)   # cold newcomer vs hot resident
+    # aging halves counters
+    lfu.sample = lfu.ops + 1
+    lfu.touch(222)
+    assert lfu.estimate(111) <= 3
+
+
+def test_tinylfu_admission_protects_hot_pages():
+    """With a scan workload, tinylfu keeps the hot page resident while
+    plain LRU evicts it."""
+    import torch
+    from dynamo_amd.engine.kv_cache import KVCachePool
+    from dynamo_amd.kvbm.host_tier import HostKVTier
+
+    def run(policy):
+        pool = KVCachePool(1, 8, 1, 4, 8, "cpu")
+        tier = HostKVTier(pool, num_host_pages=2, policy=policy)
+        HOT = 10_001
+        for _ in range(6):                      # make HOT clearly hot
+            tier.offload(0, HOT)
+            tier.onboard(HOT, 0)
+        for i, h in enumerate(range(20_000, 20_006)):  # one-shot scan
+            tier.offload((i % 7) + 1, h)
+        return tier.contains(HOT)
+
+    assert run("tinylfu") is True
+    assert run("lru") is False
+
+
+def test_tinylfu_through_engine_config():
+    from dynamo_amd.engine.config import EngineConfig, PRESETS
+    from dynamo_amd.engine.engine import LLMEngine
+    cfg = EngineConfig(model=PRESETS["tiny-llama"], device="cpu",
+                       kv_pool_pages=32, host_cache_pages=8,
+                       host_cache_policy="tinylfu", max_model_len=256)
+    eng = LLMEngine(cfg)
+    assert eng.host_tier is not None and eng.host_tier.lfu is not None
