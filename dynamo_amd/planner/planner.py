@@ -204,6 +204,58 @@ class LoadPlanner:
 
 
 @dataclass
+class ParallelizationConfig:
+    """Per-pool parallelism mapping (reference parity:
+    components/src/dynamo/planner/config/parallelization.py:99-126):
+    how many GPUs one replica of each worker type occupies. The planner
+    scales REPLICAS; deployments budget GPUS - this is the conversion,
+    plus the cluster-cap arbitration applied to scaling decisions."""
+    tp_size: int = 1
+    pp_size: int = 1
+    moe_tp_size: int = 0          # 0 = dense model (use tp_size)
+    moe_ep_size: int = 0
+    dp_size: int = 1
+
+    def __post_init__(self):
+        for f in ("tp_size", "pp_size", "dp_size"):
+            if getattr(self, f) < 1:
+                raise ValueError(f"{f} must be >= 1")
+        if (self.moe_tp_size > 0) != (self.moe_ep_size > 0):
+            raise ValueError("moe_tp_size and moe_ep_size go together")
+
+    @property
+    def gpus_per_replica(self) -> int:
+        attn = self.tp_size
+        moe = (self.moe_tp_size * self.moe_ep_size
+               if self.moe_tp_size else attn)
+        return max(attn, moe) * self.pp_size * self.dp_size
+
+
+def cap_to_gpu_budget(targets: Dict[str, int],
+                      parallel: "Dict[str, ParallelizationConfig]",
+                      total_gpus: int) -> Dict[str, int]:
+    """Scale replica targets down proportionally when the plan exceeds the
+    cluster GPU budget, never below 1 replica per pool (the reference
+    planner's budget arbitration step)."""
+    if total_gpus <= 0:
+        return dict(targets)
+    cost = {c: parallel.get(c, ParallelizationConfig()).gpus_per_replica
+            for c in targets}
+    need = sum(targets[c] * cost[c] for c in targets)
+    if need <= total_gpus:
+        return dict(targets)
+    scale = total_gpus / need
+    capped = {c: max(1, int(targets[c] * scale)) for c in targets}
+    while sum(capped[c] * cost[c] for c in capped) > total_gpus:
+        big = max((c for c in capped if capped[c] > 1),
+                  key=lambda c: capped[c] * cost[c], default=None)
+        if big is None:
+            break
+        capped[big] -= 1
+    return capped
+
+
+@dataclass
 class SLATargets:
     ttft_s: float = 2.0
     itl_ms: float = 25.0
@@ -318,7 +370,9 @@ class SLAPlanner:
                  prefill_component: str = "prefill",
                  decode_component: str = "backend",
                  max_replicas: int = 64, cooldown_s: float = 10.0,
-                 down_stable: int = 3):
+                 down_stable: int = 3,
+                 parallel: "Optional[Dict[str, ParallelizationConfig]]" = None,
+                 total_gpus: int = 0):
         self.sla = sla
         self.perf = perf
         self.connector = connector
@@ -329,6 +383,8 @@ class SLAPlanner:
         self.max_replicas = max_replicas
         self.cooldown_s = cooldown_s
         self.down_stable = down_stable
+        self.parallel = parallel
+        self.total_gpus = total_gpus
         self.state = ScalingState.STEADY
         self._last_action = 0.0
         self._down_count: Dict[str, int] = {}
@@ -345,8 +401,12 @@ class SLAPlanner:
         # steady-state decode demand: each request needs osl tokens over
         # its lifetime; a replica sustains `tps` output tokens/s at SLA
         n_decode = max(1, math.ceil(req_per_s * self.sla.osl / tps))
-        return {self.prefill_component: min(self.max_replicas, n_prefill),
-                self.decode_component: min(self.max_replicas, n_decode)}
+        targets = {self.prefill_component: min(self.max_replicas, n_prefill),
+                   self.decode_component: min(self.max_replicas, n_decode)}
+        if self.parallel is not None:
+            targets = cap_to_gpu_budget(targets, self.parallel,
+                                        self.total_gpus)
+        return targets
 
     async def observe_and_plan(self, req_per_s: float,
                                actual_itl_ms: Optional[float] = None,

@@ -306,3 +306,55 @@ def test_global_planner_service_delegation():
         await rt3.shutdown(drain=False)
 
     run(main())
+
+
+def test_parallelization_config_gpus_per_replica():
+    from dynamo_amd.planner.planner import ParallelizationConfig
+    assert ParallelizationConfig().gpus_per_replica == 1
+    assert ParallelizationConfig(tp_size=4).gpus_per_replica == 4
+    assert ParallelizationConfig(tp_size=4, pp_size=2).gpus_per_replica == 8
+    # MoE: attention TP 2 but experts on moe_tp2 x ep4 = 8 GPUs
+    p = ParallelizationConfig(tp_size=2, moe_tp_size=2, moe_ep_size=4)
+    assert p.gpus_per_replica == 8
+    import pytest
+    with pytest.raises(ValueError):
+        ParallelizationConfig(tp_size=0)
+    with pytest.raises(ValueError):
+        ParallelizationConfig(moe_tp_size=2)
+
+
+def test_cap_to_gpu_budget():
+    from dynamo_amd.planner.planner import (ParallelizationConfig,
+                                            cap_to_gpu_budget)
+    par = {"prefill": ParallelizationConfig(tp_size=4),
+           "backend": ParallelizationConfig(tp_size=4)}
+    t = cap_to_gpu_budget({"prefill": 2, "backend": 6}, par, 32)
+    assert t == {"prefill": 2, "backend": 6}       # 32 GPUs fits
+    t = cap_to_gpu_budget({"prefill": 2, "backend": 6}, par, 16)
+    assert sum(v * 4 for v in t.values()) <= 16
+    assert t["prefill"] >= 1 and t["backend"] >= 1
+    assert t["backend"] > t["prefill"] or t["prefill"] == 1
+    # budget smaller than 1+1 replicas: floors at 1 each
+    t = cap_to_gpu_budget({"prefill": 4, "backend": 4}, par, 4)
+    assert t == {"prefill": 1, "backend": 1}
+
+
+def test_sla_planner_respects_gpu_budget():
+    import asyncio
+    from dynamo_amd.planner.planner import (InterpolatedPerfModel,
+                                            ParallelizationConfig,
+                                            SLAPlanner, SLATargets,
+                                            VirtualConnector)
+    perf = InterpolatedPerfModel(
+        [{"concurrency": 1, "itl_ms": 10.0, "tokens_per_s": 100.0,
+          "prefill_tokens_per_s": 50_000.0},
+         {"concurrency": 32, "itl_ms": 40.0, "tokens_per_s": 1200.0,
+          "prefill_tokens_per_s": 50_000.0}], isl=8192)
+    par = {"prefill": ParallelizationConfig(tp_size=4),
+           "backend": ParallelizationConfig(tp_size=4)}
+    pl = SLAPlanner(SLATargets(itl_ms=40.0), perf, VirtualConnector(),
+                    parallel=par, total_gpus=8)
+    targets = pl.required_replicas(req_per_s=50.0)
+    gpus = sum(v * 4 for v in targets.values())
+    assert gpus <= 8
+    assert all(v >= 1 for v in targets.values())
