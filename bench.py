@@ -60,7 +60,8 @@ def parse_args():
 def make_cfg(args, mc, device, world=1, rank=0, worker_type="aggregated",
              max_seqs=None):
     from dynamo_amd.engine import EngineConfig
-    if mc.num_experts and world > 1 and (args.moe_ep or True):
+    if mc.num_experts and (world > 1 or args.moe_ep):
+        # MoE at N>1 defaults to expert parallelism (BASELINE config #5)
         import dataclasses
         mc = dataclasses.replace(mc, moe_ep=True)
     return EngineConfig(
