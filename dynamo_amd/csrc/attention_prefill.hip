@@ -335,6 +335,7 @@ __global__ __launch_bounds__(512) void prefill32_kernel(
   const int ps = 1 << log2_ps;
   const int32_t* pt = page_table + (int64_t)seq * max_pages;
   const int lo = l & 31, hi = l >> 5;
+  const float scale2 = scale * 1.44269504f;  // scale * log2(e)
 
   extern __shared__ char lds32[];
 
@@ -482,31 +483,61 @@ __global__ __launch_bounds__(512) void prefill32_kernel(
           *reinterpret_cast<bf16x8_t*>(&a1), qreg[ds], s1, 0, 0, 0);
     }
 
-    // ---- mask + scale + in-register softmax (lane owns q-row lo) ----
+    // ---- mask + softmax, log2 domain (lane owns q-row lo) ----
+    // PMC post-XK: VALU:MFMA 16.6:1 issue-bound, conflicts gone. Two
+    // VALU cuts (null on the wait-bound decode, real here):
+    //  - log2 domain: raw S max (scale folds into one mul of the max),
+    //    exp chain mul+sub+mul+exp -> fma+exp2. m/l never leave the
+    //    kernel, so the domain is fully internal.
+    //  - full-tile fast path: a kv tile entirely at/below every q row of
+    //    this wave (t0+63 <= qpos of row q0) skips all 64 causal
+    //    cmp+sels; ~half of all tiles in a long causal prefill qualify.
     float p[32];
     float mt = kNegInf;
+    const bool full = (t0 + kKB32 - 1) <= (ctx - qlen + q0);
+    if (full) {
 #pragma unroll
-    for (int r = 0; r < 16; r++) {
-      const int trow = (r & 3) + 8 * (r >> 2) + 4 * hi;
-      const bool ok0 = row_valid && t0 + trow <= my_qpos;
-      const bool ok1 = row_valid && t0 + 32 + trow <= my_qpos;
-      p[r] = ok0 ? s0[r] * scale : kNegInf;
-      p[16 + r] = ok1 ? s1[r] * scale : kNegInf;
-      mt = fmaxf(mt, fmaxf(p[r], p[16 + r]));
+      for (int r = 0; r < 16; r++) {
+        p[r] = s0[r];
+        p[16 + r] = s1[r];
+        mt = fmaxf(mt, fmaxf(p[r], p[16 + r]));
+      }
+    } else {
+#pragma unroll
+      for (int r = 0; r < 16; r++) {
+        const int trow = (r & 3) + 8 * (r >> 2) + 4 * hi;
+        const bool ok0 = row_valid && t0 + trow <= my_qpos;
+        const bool ok1 = row_valid && t0 + 32 + trow <= my_qpos;
+        p[r] = ok0 ? s0[r] : kNegInf;
+        p[16 + r] = ok1 ? s1[r] : kNegInf;
+        mt = fmaxf(mt, fmaxf(p[r], p[16 + r]));
+      }
     }
     mt = fmaxf(mt, xor32_swap(mt, hi));
-    float m_new = fmaxf(m_run, mt);
+    const float mts = (mt <= kNegInf * 0.5f) ? kNegInf : mt * scale2;
+    float m_new = fmaxf(m_run, mts);
     bool skip_rescale = false;
     // defer-max: skip the O-rescale while the tile max stays within 8
-    if (__all(mt - m_run <= 8.0f)) { m_new = m_run; skip_rescale = true; }
+    // (8 nats = 11.54 in the log2 domain)
+    if (__all(mts - m_run <= 11.5417f)) { m_new = m_run; skip_rescale = true; }
     const float alpha =
-        (skip_rescale || m_run <= kNegInf * 0.5f) ? 1.f : __expf(m_run - m_new);
+        (skip_rescale || m_run <= kNegInf * 0.5f)
+            ? 1.f : __builtin_amdgcn_exp2f(m_run - m_new);
     float ls = 0.f;
+    if (full) {
 #pragma unroll
-    for (int r = 0; r < 32; r++) {
-      p[r] = (p[r] <= kNegInf * 0.5f || m_new <= kNegInf * 0.5f)
-                 ? 0.f : __expf(p[r] - m_new);
-      ls += p[r];
+      for (int r = 0; r < 32; r++) {
+        p[r] = __builtin_amdgcn_exp2f(fmaf(p[r], scale2, -m_new));
+        ls += p[r];
+      }
+    } else {
+#pragma unroll
+      for (int r = 0; r < 32; r++) {
+        p[r] = (p[r] <= kNegInf * 0.5f || m_new <= kNegInf * 0.5f)
+                   ? 0.f
+                   : __builtin_amdgcn_exp2f(fmaf(p[r], scale2, -m_new));
+        ls += p[r];
+      }
     }
     ls += xor32_swap(ls, hi);
     l_run = l_run * alpha + ls;
