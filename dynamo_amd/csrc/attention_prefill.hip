@@ -61,6 +61,7 @@ __global__ __launch_bounds__(kBlock) void prefill_kernel(
   const int qstart = seq_q_start[seq];
   const int ps = 1 << log2_ps;
   const int32_t* pt = page_table + (int64_t)seq * max_pages;
+  const float scale2 = scale * 1.44269504f;  // scale * log2(e)
 
   const int lane = threadIdx.x & 63;
   const int wid = threadIdx.x >> 6;
@@ -166,26 +167,35 @@ __global__ __launch_bounds__(kBlock) void prefill_kernel(
       s[n] = accs;
     }
 
-    // ---- online softmax (per row; rows live in (lg, r); cols in lr) ----
+    // ---- online softmax, log2 domain (rows live in (lg, r); cols lr) ----
+    // Same two VALU cuts as prefill32 (log2 domain + full-visible-tile
+    // mask elision); m/lsum stay kernel-internal so the domain is too.
+    const bool fullt =
+        (t0 + kKvTile - 1) <= (ctx - qlen + q0 + wid * 16);
     float mt[4];
 #pragma unroll
     for (int r = 0; r < 4; r++) {
-      const int qrow = q0 + wid * 16 + lg * 4 + r;
-      const int qpos = ctx - qlen + qrow;
       float mx = kNegInf;
+      if (fullt) {
 #pragma unroll
-      for (int n = 0; n < 4; n++) {
-        const int kvpos = t0 + n * 16 + lr;
-        float sv = s[n][r] * scale;
-        sv = (qrow < qlen && kvpos <= qpos && kvpos < ctx) ? sv : kNegInf;
-        s[n][r] = sv;
-        mx = fmaxf(mx, sv);
+        for (int n = 0; n < 4; n++) mx = fmaxf(mx, s[n][r]);
+      } else {
+        const int qrow = q0 + wid * 16 + lg * 4 + r;
+        const int qpos = ctx - qlen + qrow;
+#pragma unroll
+        for (int n = 0; n < 4; n++) {
+          const int kvpos = t0 + n * 16 + lr;
+          float sv = s[n][r];
+          sv = (qrow < qlen && kvpos <= qpos && kvpos < ctx) ? sv : kNegInf;
+          s[n][r] = sv;
+          mx = fmaxf(mx, sv);
+        }
       }
       mx = fmaxf(mx, __shfl_xor(mx, 1, WAVE_SIZE));
       mx = fmaxf(mx, __shfl_xor(mx, 2, WAVE_SIZE));
       mx = fmaxf(mx, __shfl_xor(mx, 4, WAVE_SIZE));
       mx = fmaxf(mx, __shfl_xor(mx, 8, WAVE_SIZE));
-      mt[r] = mx;
+      mt[r] = (mx <= kNegInf * 0.5f) ? kNegInf : mx * scale2;
     }
 
     // NOTE: rows of S (and acc_o) map to (lg, r): row = lg*4 + r. The running
@@ -196,7 +206,9 @@ __global__ __launch_bounds__(kBlock) void prefill_kernel(
     for (int r = 0; r < 4; r++) {
       float mnew = m[r];
       if (mt[r] > mnew) {
-        const float corr = (mnew <= kNegInf * 0.5f) ? 0.f : __expf(mnew - mt[r]);
+        const float corr = (mnew <= kNegInf * 0.5f)
+                               ? 0.f
+                               : __builtin_amdgcn_exp2f(mnew - mt[r]);
         lsum[r] *= corr;
 #pragma unroll
         for (int d = 0; d < 8; d++) acc_o[d][r] *= corr;
@@ -204,12 +216,24 @@ __global__ __launch_bounds__(kBlock) void prefill_kernel(
         m[r] = mnew;
       }
       float rowsum = 0.f;
+      if (fullt) {
 #pragma unroll
-      for (int n = 0; n < 4; n++) {
-        const float pv = (s[n][r] <= kNegInf * 0.5f || mnew <= kNegInf * 0.5f)
-                             ? 0.f : __expf(s[n][r] - mnew);
-        p[n][r] = pv;
-        rowsum += pv;
+        for (int n = 0; n < 4; n++) {
+          const float pv =
+              __builtin_amdgcn_exp2f(fmaf(s[n][r], scale2, -mnew));
+          p[n][r] = pv;
+          rowsum += pv;
+        }
+      } else {
+#pragma unroll
+        for (int n = 0; n < 4; n++) {
+          const float pv =
+              (s[n][r] <= kNegInf * 0.5f || mnew <= kNegInf * 0.5f)
+                  ? 0.f
+                  : __builtin_amdgcn_exp2f(fmaf(s[n][r], scale2, -mnew));
+          p[n][r] = pv;
+          rowsum += pv;
+        }
       }
       rowsum += __shfl_xor(rowsum, 1, WAVE_SIZE);
       rowsum += __shfl_xor(rowsum, 2, WAVE_SIZE);
