@@ -114,7 +114,7 @@ static void run_mfma_swapped(const Bufs& bf, bool check) {
   auto launch = [&] {
     paged_decode_mfma_swapped<DEFER, PRIO, KPF, 0, VS, XK2, VT, LG2, MINW><<<grid, kBlock, lds>>>(
         bf.partial, bf.ml, bf.out, bf.q, bf.kc, VT ? bf.vct : bf.vc, bf.pt, bf.ctx,
-        0.0883883f, kChunk, G, B, Hkv, bf.C, CTX / PS, 6, HD);
+        0.0883883f, kChunk, G, B, Hkv, bf.C, CTX / PS, 6, HD, nullptr);
     paged_decode_phase2<<<dim3(B, G * Hkv), 128>>>(
         bf.out, bf.partial, bf.ml, bf.ctx, kChunk, G * Hkv, bf.C, HD);
   };

@@ -22,7 +22,8 @@ void paged_attention_decode(torch::Tensor out, torch::Tensor q,
                             torch::Tensor page_table, torch::Tensor ctx_lens,
                             torch::Tensor partial, torch::Tensor ml,
                             double scale, int64_t chunk_tokens,
-                            bool v_transposed) {
+                            bool v_transposed,
+                            c10::optional<torch::Tensor> chunk_cnt) {
   TORCH_CHECK(q.is_cuda() && q.dtype() == torch::kBFloat16);
   TORCH_CHECK(page_table.dtype() == torch::kInt32 && ctx_lens.dtype() == torch::kInt32);
   const bool fp8 = kcache.dtype() == torch::kFloat8_e4m3fn;
@@ -54,6 +55,19 @@ void paged_attention_decode(torch::Tensor out, torch::Tensor q,
                 "v_transposed decode needs page_size%32==0, head_dim==128, "
                 "2<=G<=16 (got G=", G, ")");
     const int lds = mfma_swapped_vt_lds_bytes(G, hd);
+    // fused chunk merge (DYNAMO_FUSED_MERGE=1): MEASURED NET NEGATIVE on
+    // the flagship (38.75 -> 47.35 ms/step): the two __threadfence()s per
+    // block cost ~6 ms/step across 640 blocks x 80 layers, an order of
+    // magnitude more than the 5.6 us phase2 launches they replace. Kept
+    // behind the env flag as a recorded negative result.
+    static const bool use_fused_merge = [] {
+      const char* e = getenv("DYNAMO_FUSED_MERGE");
+      return e != nullptr && e[0] == '1';
+    }();
+    int32_t* cnt = nullptr;
+    if (use_fused_merge && C > 1 && chunk_cnt.has_value() &&
+        chunk_cnt->numel() >= (int64_t)B * Hkv)
+      cnt = chunk_cnt->data_ptr<int32_t>();
     auto launch_vt = [&](auto* kern) {
       if (lds > 65536)
         (void)hipFuncSetAttribute(reinterpret_cast<const void*>(kern),
@@ -64,7 +78,7 @@ void paged_attention_decode(torch::Tensor out, torch::Tensor q,
           (short*)out.data_ptr(), (const short*)q.data_ptr(),
           (const short*)kcache.data_ptr(), (const short*)vcache.data_ptr(),
           page_table.data_ptr<int32_t>(), ctx_lens.data_ptr<int32_t>(),
-          (float)scale, chunk, G, B, Hkv, C, max_pages, log2_ps, hd);
+          (float)scale, chunk, G, B, Hkv, C, max_pages, log2_ps, hd, cnt);
     };
     // VT3 = 64-token tiles (32 loads in flight per batch) at 2
     // waves/SIMD (MINW=2, 232 VGPR no-spill): sweep G8 5131 / G4 4864
@@ -84,7 +98,7 @@ void paged_attention_decode(torch::Tensor out, torch::Tensor q,
     else
       launch_vt(&paged_decode_mfma_swapped<1, 1, 0, 0, 72, 0, 3, 0, 2>);
     HIP_CHECK_KERNEL();
-    if (C > 1) {
+    if (C > 1 && cnt == nullptr) {
       dim3 grid2(B, Hq);
       paged_decode_phase2<<<grid2, 128, 0, stream>>>(
           (short*)out.data_ptr(), partial.data_ptr<float>(),
@@ -130,7 +144,7 @@ void paged_attention_decode(torch::Tensor out, torch::Tensor q,
           (const short*)kcache.data_ptr(),                                    \
           (const short*)vcache.data_ptr(), page_table.data_ptr<int32_t>(),    \
           ctx_lens.data_ptr<int32_t>(), (float)scale, chunk, GG, B, Hkv, C,    \
-          max_pages, log2_ps, hd);                                            \
+          max_pages, log2_ps, hd, nullptr);                                   \
       break;                                                                  \
     }                                                                         \
     if (mfma_lds_bytes(GG, hd) > 65536)                                       \
@@ -161,7 +175,7 @@ void paged_attention_decode(torch::Tensor out, torch::Tensor q,
         (short*)out.data_ptr(), (const short*)q.data_ptr(),
         (const short*)kcache.data_ptr(), (const short*)vcache.data_ptr(),
         page_table.data_ptr<int32_t>(), ctx_lens.data_ptr<int32_t>(),
-        (float)scale, chunk, G, B, Hkv, C, max_pages, log2_ps, hd);
+        (float)scale, chunk, G, B, Hkv, C, max_pages, log2_ps, hd, nullptr);
     HIP_CHECK_KERNEL();
     if (C > 1) {
       dim3 grid2(B, Hq);

@@ -635,7 +635,12 @@ __global__ __launch_bounds__(kBlock, MINW) void paged_decode_mfma_swapped(
     const short* __restrict__ kcache, const short* __restrict__ vcache,
     const int32_t* __restrict__ page_table, const int32_t* __restrict__ ctx_lens,
     float scale, int chunk, int G, int B, int Hkv, int C, int max_pages,
-    int log2_ps, int hd) {
+    int log2_ps, int hd, int32_t* __restrict__ chunk_cnt) {
+  // chunk_cnt (optional, [B*Hkv] zero-initialized): fused chunk merge.
+  // When non-null and C > 1, every block of a (b, h) column counts down
+  // via atomicAdd after publishing its partial/ml; the LAST block inlines
+  // the phase2 merge and resets the counter - one kernel launch instead
+  // of two per decode layer (~0.45 ms/step on the 80-layer flagship).
   typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8_t;
   typedef __attribute__((ext_vector_type(2))) unsigned int uint2_t;
   const int kSlab = chunk / 4;
@@ -656,6 +661,39 @@ __global__ __launch_bounds__(kBlock, MINW) void paged_decode_mfma_swapped(
   short* q_lds_s = reinterpret_cast<short*>(merge + 4 * G * (hd + 2));
   short* v_lds = q_lds_s + G * hd + wid * (hd * VS / 2);  // per-wave
 
+  auto fused_merge = [&]() {
+    // countdown; the last-arriving block of this (b, h) column merges
+    __threadfence();
+    __shared__ int lastflag;
+    if (threadIdx.x == 0) {
+      const int prev = atomicAdd(&chunk_cnt[b * Hkv + h], 1);
+      lastflag = (prev == (int)gridDim.z - 1) ? 1 : 0;
+    }
+    __syncthreads();
+    if (!lastflag) return;
+    __threadfence();  // acquire the other blocks' partial/ml stores
+    if (threadIdx.x == 0) chunk_cnt[b * Hkv + h] = 0;  // next-launch reset
+    const int nc = min(C, (ctx + chunk - 1) / chunk);
+    for (int i = threadIdx.x; i < G * hd; i += kBlock) {
+      const int g = i / hd;
+      const int d = i % hd;
+      const int qh = h * G + g;
+      const float* mlp = ml + (((int64_t)b * Hq + qh) * C) * 2;
+      float mstar = kNegInf;
+      for (int cc = 0; cc < nc; cc++) mstar = fmaxf(mstar, mlp[2 * cc]);
+      float asum = 0.f, lsum2 = 0.f;
+      for (int cc = 0; cc < nc; cc++) {
+        const float lc = mlp[2 * cc + 1];
+        if (lc <= 0.f) continue;
+        const float corr = __expf(mlp[2 * cc] - mstar);
+        asum += partial[(((int64_t)b * Hq + qh) * C + cc) * hd + d] * corr;
+        lsum2 += lc * corr;
+      }
+      out[((int64_t)b * Hq + qh) * hd + d] =
+          f32_to_bf16(lsum2 > 0.f ? asum / lsum2 : 0.f);
+    }
+  };
+
   if (chunk_start >= ctx) {
     if (C > 1) {
       for (int i = threadIdx.x; i < G; i += kBlock) {
@@ -663,6 +701,7 @@ __global__ __launch_bounds__(kBlock, MINW) void paged_decode_mfma_swapped(
         float* mlp = ml + (((int64_t)b * Hq + qh) * C + c) * 2;
         mlp[0] = kNegInf; mlp[1] = 0.f;
       }
+      if (chunk_cnt != nullptr) fused_merge();
     }
     return;
   }
@@ -1240,6 +1279,10 @@ __global__ __launch_bounds__(kBlock, MINW) void paged_decode_mfma_swapped(
         mlp[0] = mstar; mlp[1] = lsum;
       }
     }
+  }
+  if (C > 1 && chunk_cnt != nullptr) {
+    __syncthreads();  // all merge-loop writes of this block done
+    fused_merge();
   }
 }
 

@@ -34,7 +34,8 @@ void paged_attention_decode(torch::Tensor out, torch::Tensor q, torch::Tensor kc
                             torch::Tensor vcache, torch::Tensor page_table,
                             torch::Tensor ctx_lens, torch::Tensor partial,
                             torch::Tensor ml, double scale,
-                            int64_t chunk_tokens, bool v_transposed);
+                            int64_t chunk_tokens, bool v_transposed,
+                            c10::optional<torch::Tensor> chunk_cnt);
 // attention_prefill.hip
 void attention_prefill_paged(torch::Tensor out, torch::Tensor q,
                              torch::Tensor kcache, torch::Tensor vcache,
@@ -91,7 +92,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("out"), py::arg("q"), py::arg("kcache"), py::arg("vcache"),
         py::arg("page_table"), py::arg("ctx_lens"), py::arg("partial"),
         py::arg("ml"), py::arg("scale"), py::arg("chunk_tokens"),
-        py::arg("v_transposed") = false);
+        py::arg("v_transposed") = false,
+        py::arg("chunk_cnt") = py::none());
   m.def("attention_prefill_paged", &attention_prefill_paged,
         py::arg("out"), py::arg("q"), py::arg("kcache"), py::arg("vcache"),
         py::arg("page_table"), py::arg("tile_seq"), py::arg("tile_q0"),

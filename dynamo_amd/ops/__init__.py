@@ -133,6 +133,9 @@ class DecodeScratch:
                                    dtype=torch.float32, device=device)
         self.ml = torch.empty(max_batch, num_q_heads, self.chunks, 2,
                               dtype=torch.float32, device=device)
+        # fused-merge countdown ([B * Hkv] upper bound; kernel resets it)
+        self.chunk_cnt = torch.zeros(max_batch * 128, dtype=torch.int32,
+                                     device=device)
 
     def view(self, batch: int):
         return self.partial[:batch], self.ml[:batch]
@@ -148,7 +151,8 @@ def paged_attention_decode(q, kcache, vcache, page_table, ctx_lens, scale,
         partial, ml = scratch.view(q.shape[0])
         hip().paged_attention_decode(out, q, kcache, vcache, page_table,
                                      ctx_lens, partial, ml, scale,
-                                     scratch.chunk_tokens, v_transposed)
+                                     scratch.chunk_tokens, v_transposed,
+                                     scratch.chunk_cnt)
         return out
     r = torch_ref.paged_attention_decode(q, kcache, vcache, page_table,
                                          ctx_lens, scale,
