@@ -253,8 +253,10 @@ int main() {
     run_mfma_swapped<8, 1, 1, 0, 72, 0, 2, 0, 2>(bf, false);  // VT2 minw2
     run_mfma_swapped<8, 1, 1, 0, 72, 0, 1, 0, 3>(bf, false);  // VT1 minw3
     run_mfma_swapped<8, 0, 0, 0, 72, 0, 3, 0, 1>(bf, true);   // VT3 64tok chk
-    run_mfma_swapped<8, 1, 1, 0, 72, 0, 3, 0, 1>(bf, false);  // VT3 64tok
     run_mfma_swapped<8, 1, 1, 0, 72, 0, 3, 0, 2>(bf, false);  // VT3 minw2
+    run_mfma_swapped<8, 0, 0, 0, 72, 0, 4, 0, 1>(bf, true);   // VT4 chk
+    run_mfma_swapped<8, 1, 1, 0, 72, 0, 4, 0, 1>(bf, false);  // VT4
+    run_mfma_swapped<8, 1, 1, 0, 72, 0, 4, 0, 2>(bf, false);  // VT4 minw2
     run<8, 16, 4, 2>(bf, "");
     run<8, 16, 4, 3>(bf, "");
     run<8, 16, 4, 4>(bf, "");

@@ -54,7 +54,9 @@ void paged_attention_decode(torch::Tensor out, torch::Tensor q,
     TORCH_CHECK(ps % 32 == 0 && hd == 128 && G >= 2 && G <= 16,
                 "v_transposed decode needs page_size%32==0, head_dim==128, "
                 "2<=G<=16 (got G=", G, ")");
-    const int lds = mfma_swapped_vt_lds_bytes(G, hd);
+    // VT4 stages K through the per-wave LDS region (reuses the V-staging
+    // space of the token-major layout: 8 KB/wave of the VS=72 slab)
+    const int lds = mfma_swapped_lds_bytes(G, hd, 72);
     // fused chunk merge (DYNAMO_FUSED_MERGE=1): MEASURED NET NEGATIVE on
     // the flagship (38.75 -> 47.35 ms/step): the two __threadfence()s per
     // block cost ~6 ms/step across 640 blocks x 80 layers, an order of
@@ -80,23 +82,21 @@ void paged_attention_decode(torch::Tensor out, torch::Tensor q,
           page_table.data_ptr<int32_t>(), ctx_lens.data_ptr<int32_t>(),
           (float)scale, chunk, G, B, Hkv, C, max_pages, log2_ps, hd, cnt);
     };
-    // VT3 = 64-token tiles (32 loads in flight per batch) at 2
-    // waves/SIMD (MINW=2, 232 VGPR no-spill): sweep G8 5131 / G4 4864
-    // GB/s vs 4922/4511 for the 32-token VT2 on the same box
+    // VT4 = 64-token tiles + K staged via per-wave LDS in contiguous
+    // 1KB bursts (sweep same-box: 5357 GB/s vs 5150 VT3 / 4922 VT2);
+    // DYNAMO_VT3_VARIANT selects the bring-up fallbacks
     static const int vt3v = [] {
       const char* e = getenv("DYNAMO_VT3_VARIANT");
       return e ? atoi(e) : 0;
     }();
     if (fp8)
-      launch_vt(&paged_decode_mfma_swapped<1, 1, 0, 1, 72, 0, 3, 0, 2>);
+      launch_vt(&paged_decode_mfma_swapped<1, 1, 0, 1, 72, 0, 4>);
     else if (vt3v == 1)
-      launch_vt(&paged_decode_mfma_swapped<0, 0, 0, 0, 72, 0, 3, 0, 2>);
+      launch_vt(&paged_decode_mfma_swapped<1, 1, 0, 0, 72, 0, 3, 0, 2>);
     else if (vt3v == 2)
-      launch_vt(&paged_decode_mfma_swapped<1, 1, 0, 0, 72, 0, 3, 0, 1>);
-    else if (vt3v == 3)
       launch_vt(&paged_decode_mfma_swapped<1, 1, 0, 0, 72, 0, 2>);
     else
-      launch_vt(&paged_decode_mfma_swapped<1, 1, 0, 0, 72, 0, 3, 0, 2>);
+      launch_vt(&paged_decode_mfma_swapped<1, 1, 0, 0, 72, 0, 4>);
     HIP_CHECK_KERNEL();
     if (C > 1 && cnt == nullptr) {
       dim3 grid2(B, Hq);

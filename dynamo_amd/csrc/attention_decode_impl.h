@@ -945,24 +945,8 @@ __global__ __launch_bounds__(kBlock, MINW) void paged_decode_mfma_swapped(
   // ~77% of the pattern ceiling (WAIT_ANY-dominated). 64-token tiles
   // issue 32 loads back-to-back (2 KB in flight per wave) before the
   // QK/softmax/PV batch. Single-buffered; MINW=2 keeps 2 waves/SIMD.
-  auto tile_vt64 = [&](int t0, const short8 (&kf)[16],
-                       const short8 (&vf)[16]) {
-    f32x4 sT[4];
-    if constexpr (PRIO) __builtin_amdgcn_s_setprio(1);
-#pragma unroll
-    for (int tg = 0; tg < 4; tg++) {
-      f32x4 acc_s{0.f, 0.f, 0.f, 0.f};
-      if (t0 + tg * 16 < slab_end) {
-#pragma unroll
-        for (int kc = 0; kc < 4; kc++) {
-          short8 ka = kf[tg * 4 + kc];
-          acc_s = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-              *reinterpret_cast<bf16x8_t*>(&ka), q_frag[kc], acc_s, 0, 0, 0);
-        }
-      }
-      sT[tg] = acc_s;
-    }
-    if constexpr (PRIO) __builtin_amdgcn_s_setprio(0);
+  auto tile_vt64_post = [&](int t0, f32x4 (&sT)[4],
+                            const short8 (&vf)[16]) {
     float pv[4][4];
     float mt = kNegInf;
     const bool tail = t0 + 64 > slab_end;
@@ -1038,6 +1022,26 @@ __global__ __launch_bounds__(kBlock, MINW) void paged_decode_mfma_swapped(
       }
     if constexpr (PRIO) __builtin_amdgcn_s_setprio(0);
   };
+  auto tile_vt64 = [&](int t0, const short8 (&kf)[16],
+                       const short8 (&vf)[16]) {
+    f32x4 sT[4];
+    if constexpr (PRIO) __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int tg = 0; tg < 4; tg++) {
+      f32x4 acc_s{0.f, 0.f, 0.f, 0.f};
+      if (t0 + tg * 16 < slab_end) {
+#pragma unroll
+        for (int kc = 0; kc < 4; kc++) {
+          short8 ka = kf[tg * 4 + kc];
+          acc_s = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              *reinterpret_cast<bf16x8_t*>(&ka), q_frag[kc], acc_s, 0, 0, 0);
+        }
+      }
+      sT[tg] = acc_s;
+    }
+    if constexpr (PRIO) __builtin_amdgcn_s_setprio(0);
+    tile_vt64_post(t0, sT, vf);
+  };
   auto load_k64 = [&](int t0_, short8 (&kf)[16]) {
     // uniform (no break-in-unroll): groups past slab_end clamp their page
     // lookup to the last valid token and mask values via vA
@@ -1081,7 +1085,75 @@ __global__ __launch_bounds__(kBlock, MINW) void paged_decode_mfma_swapped(
     }
   };
 
-  if constexpr (VT == 3) {
+  // ---- VT4: VT3 + K staged through per-wave LDS --------------------
+  // K's direct B-frag loads touch 16 rows x 64B per instruction (the
+  // 5.74 TB/s pattern); staging K cooperatively IN-WAVE with 4-row x
+  // 256B fully-contiguous 1KB bursts targets the 6.3 linear ceiling for
+  // half the kernel's bytes. Two 32-token halves reuse one 8 KB buffer
+  // per wave (block LDS stays under 2-resident at MINW=2); V loads stay
+  // direct (their rows are already 128B-contiguous).
+  auto stage_k32 = [&](int tb32, short* klds_w, short8 (&sreg)[8]) {
+    // 8 bursts: burst i covers rows {i*4 .. i*4+3} fully (lane>>4 = row,
+    // lane&15 = 16B chunk)
+#pragma unroll
+    for (int i = 0; i < 8; i++) {
+      const int tok = tb32 + i * 4 + (lane >> 4);
+      const int tsafe = tok < slab_end ? tok : slab_end - 1;
+      const int64_t pb =
+          (((int64_t)pt[tsafe >> log2_ps] * Hkv + h) * ps) * hd;
+      sreg[i] = ld8(kcache,
+                    pb + (int64_t)(tsafe & (ps - 1)) * hd + (lane & 15) * 8);
+    }
+#pragma unroll
+    for (int i = 0; i < 8; i++) {
+      const int row = i * 4 + (lane >> 4);
+      *reinterpret_cast<short8*>(
+          (char*)klds_w + row * 256 +
+          (((lane & 15) * 16) ^ ((row & 7) << 4))) = sreg[i];
+    }
+    __builtin_amdgcn_wave_barrier();
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+  };
+  auto qk_half = [&](int t0, int half, const short* klds_w, f32x4 (&sT)[4]) {
+#pragma unroll
+    for (int tg2 = 0; tg2 < 2; tg2++) {
+      const int tg = half * 2 + tg2;
+      f32x4 acc_s{0.f, 0.f, 0.f, 0.f};
+      if (t0 + tg * 16 < slab_end) {
+#pragma unroll
+        for (int kc = 0; kc < 4; kc++) {
+          const int row = tg2 * 16 + lr;
+          short8 ka = *reinterpret_cast<const short8*>(
+              (const char*)klds_w + row * 256 +
+              ((kc * 64 + lg * 16) ^ ((row & 7) << 4)));
+          const bool valid = t0 + tg * 16 + lr < slab_end;
+          if (!valid) ka = short8{};
+          acc_s = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              *reinterpret_cast<bf16x8_t*>(&ka), q_frag[kc], acc_s, 0, 0, 0);
+        }
+      }
+      sT[tg] = acc_s;
+    }
+  };
+
+  if constexpr (VT == 4) {
+    short* klds_w = v_lds;  // reuse the (otherwise unused) V-staging LDS
+    short8 v64[16], sreg[8];
+    for (int t0 = slab_start; t0 < slab_end; t0 += 64) {
+      f32x4 sT[4];
+      stage_k32(t0, klds_w, sreg);          // K half A (tokens t0..t0+31)
+      load_v64(t0, v64);                    // V loads overlap QK below
+      if constexpr (PRIO) __builtin_amdgcn_s_setprio(1);
+      qk_half(t0, 0, klds_w, sT);
+      if constexpr (PRIO) __builtin_amdgcn_s_setprio(0);
+      __builtin_amdgcn_wave_barrier();
+      stage_k32(t0 + 32, klds_w, sreg);     // K half B overwrites A
+      if constexpr (PRIO) __builtin_amdgcn_s_setprio(1);
+      qk_half(t0, 1, klds_w, sT);
+      if constexpr (PRIO) __builtin_amdgcn_s_setprio(0);
+      tile_vt64_post(t0, sT, v64);
+    }
+  } else if constexpr (VT == 3) {
     short8 k64[16], v64[16];
     for (int t0 = slab_start; t0 < slab_end; t0 += 64) {
       load_k64(t0, k64);
