@@ -54,6 +54,9 @@ def build_parser():
     p.add_argument("--host", default="127.0.0.1")
     p.add_argument("--host-cache-pages", type=int, default=0,
                    help="KVBM G2 pinned-host tier size in pages")
+    p.add_argument("--host-cache-policy", default="lru",
+                   choices=["lru", "tinylfu"],
+                   help="G2 eviction: lru, or tinylfu admission filter")
     p.add_argument("--disk-cache-pages", type=int, default=0,
                    help="KVBM G3 disk tier size in pages")
     p.add_argument("--disk-cache-path", default="",
@@ -141,6 +144,7 @@ def make_engine_from_args(args, tp=None) -> LLMEngine:
         cpu_shm_pool=(device == "cpu"
                       and args.worker_type in ("prefill", "decode")),
         host_cache_pages=args.host_cache_pages,
+        host_cache_policy=args.host_cache_policy,
         disk_cache_pages=args.disk_cache_pages,
         object_cache_dir=args.object_cache_dir,
         disk_cache_path=(args.disk_cache_path or
