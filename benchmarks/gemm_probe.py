@@ -22,6 +22,9 @@ SHAPES = [
 
 
 def main(M=16, iters=50):
+    if M >= 1024:
+        return main_big(M)
+
     torch.cuda.init()
     dev = "cuda:0"
     print(f"M={M} bf16, {iters} iters, W bytes / time")
@@ -39,6 +42,30 @@ def main(M=16, iters=50):
         gb = N * K * 2 / 1e9
         print(f"{name} K={K:6d} N={N:6d}  {dt*1e6:8.1f} us  "
               f"{gb/dt/1000:6.2f} TB/s")
+    del y  # noqa: F841
+
+
+
+
+def main_big(M):
+    """Prefill-regime GEMMs: report TFLOP/s vs the 2.5 PF bf16 dense peak."""
+    import time
+    dev = "cuda:0"
+    print(f"M={M} bf16 prefill shapes")
+    for name, K, N in SHAPES[:5]:
+        x = torch.randn(M, K, device=dev, dtype=torch.bfloat16)
+        w = torch.randn(N, K, device=dev, dtype=torch.bfloat16)
+        for _ in range(3):
+            y = x @ w.t()
+        torch.cuda.synchronize()
+        t0 = time.perf_counter()
+        for _ in range(10):
+            y = x @ w.t()
+        torch.cuda.synchronize()
+        dt = (time.perf_counter() - t0) / 10
+        fl = 2.0 * M * N * K
+        print(f"{name} K={K:6d} N={N:6d}  {dt*1e3:8.2f} ms  "
+              f"{fl/dt/1e12:7.1f} TF  ({fl/dt/25e12:4.1f}% of 2.5PF)")
     del y  # noqa: F841
 
 
