@@ -76,3 +76,83 @@ def test_cuckoo_filter():
     assert f.max_prefix(chain) >= 7
     empty = _core.CuckooFilter(128)
     assert empty.max_prefix(chain) == 0
+
+
+# ---------------------------------------------------------------------------
+# Property-based hardening (hypothesis): invariants that must hold for ANY
+# event interleaving, not just the scripted cases above.
+
+from hypothesis import given, settings, strategies as st
+
+
+@settings(max_examples=60, deadline=None)
+@given(st.lists(st.integers(0, 1023), min_size=0, max_size=200),
+       st.integers(1, 64), st.integers(0, 2**31))
+def test_chain_hashes_prefix_property(tokens, bs, salt):
+    """Chained hashes are a pure function of the token prefix: equal-prefix
+    sequences share exactly their common full-block hash prefix."""
+    h = _core.chain_hashes(tokens, bs, salt)
+    assert len(h) == len(tokens) // bs
+    # re-hash a mutated tail: hashes before the mutated block are unchanged
+    if len(tokens) >= bs:
+        mutated = list(tokens)
+        mutated[-1] ^= 1
+        h2 = _core.chain_hashes(mutated, bs, salt)
+        nfull = len(tokens) // bs
+        changed_block = (len(tokens) - 1) // bs
+        assert h2[:min(changed_block, nfull)] == h[:min(changed_block, nfull)]
+        if changed_block < nfull:
+            assert h2[changed_block] != h[changed_block]
+
+
+@settings(max_examples=40, deadline=None)
+@given(st.data())
+def test_indexer_matches_reference_model(data):
+    """KvIndexer under random stored/removed/remove_worker interleavings
+    agrees with a pure-python reference model on find_matches."""
+    seqs = [_core.chain_hashes(list(range(i, i + 96)), 16, 0)
+            for i in range(4)]
+    idx = _core.KvIndexer()
+    model = {}  # worker -> set(hash)
+    ops = data.draw(st.lists(st.tuples(
+        st.sampled_from(["store", "remove", "drop"]),
+        st.integers(0, 2), st.integers(0, 3), st.integers(0, 6)),
+        min_size=1, max_size=40))
+    for kind, w, si, n in ops:
+        h = seqs[si]
+        if kind == "store":
+            idx.apply_stored(w, h[:n])
+            model.setdefault(w, set()).update(h[:n])
+        elif kind == "remove":
+            idx.apply_removed(w, h[:n])
+            model.setdefault(w, set()).difference_update(h[:n])
+        else:
+            idx.remove_worker(w)
+            model.pop(w, None)
+    for q in seqs:
+        got = idx.find_matches(q)
+        want = {}
+        for w, hs in model.items():
+            d = 0
+            for hh in q:
+                if hh not in hs:
+                    break
+                d += 1
+            if d:
+                want[w] = d
+        assert got == want, (got, want)
+
+
+@settings(max_examples=30, deadline=None)
+@given(st.lists(st.integers(0, 2**62), min_size=0, max_size=500),
+       st.integers(0, 2**62))
+def test_cuckoo_no_false_negatives_and_roundtrip(items, probe):
+    f = _core.CuckooFilter(2048)
+    ok = [h for h in items if f.insert(h)]
+    for h in ok:
+        assert f.contains(h)
+    blob = f.to_bytes()
+    g = _core.CuckooFilter.from_bytes(blob, f.count())
+    for h in ok:
+        assert g.contains(h)
+    assert g.contains(probe) == f.contains(probe)
