@@ -104,8 +104,11 @@ template <int G, int DEFER = 1, int PRIO = 1, int KPF = 0, int VS = 80, int XK2 
           int VT = 0, int LG2 = 0, int MINW = 1>
 static void run_mfma_swapped(const Bufs& bf, bool check) {
   dim3 grid(B, Hkv, bf.C);
-  const int lds = VT ? mfma_swapped_vt_lds_bytes(G, HD)
-                     : mfma_swapped_lds_bytes(G, HD, VS);
+  // VT4/5 stage K (and V) through the per-wave v_lds region; VT1-3 skip
+  // all staging LDS
+  const int lds = (VT >= 4) ? mfma_swapped_lds_bytes(G, HD, VS)
+                : VT        ? mfma_swapped_vt_lds_bytes(G, HD)
+                            : mfma_swapped_lds_bytes(G, HD, VS);
   const int iters = 30;
   if (lds > 65536)
     (void)hipFuncSetAttribute(
@@ -256,7 +259,10 @@ int main() {
     run_mfma_swapped<8, 1, 1, 0, 72, 0, 3, 0, 2>(bf, false);  // VT3 minw2
     run_mfma_swapped<8, 0, 0, 0, 72, 0, 4, 0, 1>(bf, true);   // VT4 chk
     run_mfma_swapped<8, 1, 1, 0, 72, 0, 4, 0, 1>(bf, false);  // VT4
-    run_mfma_swapped<8, 1, 1, 0, 72, 0, 4, 0, 2>(bf, false);  // VT4 minw2
+    run_mfma_swapped<8, 0, 0, 0, 72, 0, 5, 0, 1>(bf, true);   // VT5 chk
+    run_mfma_swapped<8, 1, 1, 0, 72, 0, 5, 0, 1>(bf, false);  // VT5
+    run_mfma_swapped<8, 1, 1, 0, 72, 0, 5, 0, 2>(bf, false);  // VT5 minw2
+    run_mfma_swapped<8, 1, 1, 0, 72, 0, 5, 0, 3>(bf, false);  // VT5 minw3
     run<8, 16, 4, 2>(bf, "");
     run<8, 16, 4, 3>(bf, "");
     run<8, 16, 4, 4>(bf, "");

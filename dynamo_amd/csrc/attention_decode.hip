@@ -95,6 +95,11 @@ void paged_attention_decode(torch::Tensor out, torch::Tensor q,
       launch_vt(&paged_decode_mfma_swapped<1, 1, 0, 0, 72, 0, 3, 0, 2>);
     else if (vt3v == 2)
       launch_vt(&paged_decode_mfma_swapped<1, 1, 0, 0, 72, 0, 2>);
+    else if (vt3v == 5 && ps == 64)
+      // VT5 (V through LDS too): measured NEGATIVE once the sweep's LDS
+      // under-allocation was fixed - 4266 vs 5222 GB/s: PV serializes on
+      // the V stage. Kept selectable for the record.
+      launch_vt(&paged_decode_mfma_swapped<1, 1, 0, 0, 72, 0, 5>);
     else
       launch_vt(&paged_decode_mfma_swapped<1, 1, 0, 0, 72, 0, 4>);
     HIP_CHECK_KERNEL();

@@ -945,8 +945,7 @@ __global__ __launch_bounds__(kBlock, MINW) void paged_decode_mfma_swapped(
   // ~77% of the pattern ceiling (WAIT_ANY-dominated). 64-token tiles
   // issue 32 loads back-to-back (2 KB in flight per wave) before the
   // QK/softmax/PV batch. Single-buffered; MINW=2 keeps 2 waves/SIMD.
-  auto tile_vt64_post = [&](int t0, f32x4 (&sT)[4],
-                            const short8 (&vf)[16]) {
+  auto softmax_pack64 = [&](int t0, f32x4 (&sT)[4], bf16x8_t (&p_frag)[2]) {
     float pv[4][4];
     float mt = kNegInf;
     const bool tail = t0 + 64 > slab_end;
@@ -997,7 +996,6 @@ __global__ __launch_bounds__(kBlock, MINW) void paged_decode_mfma_swapped(
     };
     // two 32-token PV B-frags, each combining two 16-token S tiles via
     // the same cvt_pk + permlane swap network as the 32-token path
-    bf16x8_t p_frag[2];
 #pragma unroll
     for (int tc = 0; tc < 2; tc++) {
       const float* pa = pv[tc * 2];
@@ -1011,6 +1009,11 @@ __global__ __launch_bounds__(kBlock, MINW) void paged_decode_mfma_swapped(
       unsigned int w[4] = {f02.x, f13.x, f02.y, f13.y};
       p_frag[tc] = *reinterpret_cast<bf16x8_t*>(w);
     }
+  };
+  auto tile_vt64_post = [&](int t0, f32x4 (&sT)[4],
+                            const short8 (&vf)[16]) {
+    bf16x8_t p_frag[2];
+    softmax_pack64(t0, sT, p_frag);
     if constexpr (PRIO) __builtin_amdgcn_s_setprio(1);
 #pragma unroll
     for (int da = 0; da < 8; da++)
@@ -1136,7 +1139,82 @@ __global__ __launch_bounds__(kBlock, MINW) void paged_decode_mfma_swapped(
     }
   };
 
-  if constexpr (VT == 4) {
+  // ---- VT5: VT4 + V through the same LDS buffer, fully linear -------
+  // With ps == 64 every 64-token tile is page-aligned, so a (page, head)
+  // d-major V plane is 16 KB CONTIGUOUS. Stage it in two 8 KB dim-halves
+  // (dims [0,64) then [64,128)) with 1KB linear bursts, consuming each
+  // half with the matching PV da range (da 0-3 reads rows < 64).
+  auto stage_v_half = [&](int tb64, int dlo, short* klds_w) {
+    const int64_t pb =
+        (((int64_t)pt[tb64 >> log2_ps] * Hkv + h) * ps) * hd;
+    const int rem = slab_end - tb64;
+    short8 vreg[8];
+#pragma unroll
+    for (int i = 0; i < 8; i++)
+      vreg[i] = ld8(vcache, pb + (int64_t)dlo * ps + (i * 64 + lane) * 8);
+    if (rem < 64) {
+      const int tok0 = (lane & 7) * 8;
+#pragma unroll
+      for (int i = 0; i < 8; i++)
+#pragma unroll
+        for (int j = 0; j < 8; j++)
+          if (tok0 + j >= rem) vreg[i][j] = 0;
+    }
+#pragma unroll
+    for (int i = 0; i < 8; i++) {
+      const int row = i * 8 + (lane >> 3);
+      *reinterpret_cast<short8*>(
+          (char*)klds_w + row * 128 +
+          ((((lane & 7)) * 16) ^ (((row ^ (row >> 3)) & 7) << 4))) = vreg[i];
+    }
+    __builtin_amdgcn_wave_barrier();
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+  };
+  auto pv_half = [&](int dabase, const short* klds_w,
+                     const bf16x8_t (&p_frag)[2]) {
+    if constexpr (PRIO) __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int dl = 0; dl < 4; dl++)
+#pragma unroll
+      for (int tc = 0; tc < 2; tc++) {
+        const int row = dl * 16 + lr;
+        short8 va_s = *reinterpret_cast<const short8*>(
+            (const char*)klds_w + row * 128 +
+            ((tc * 64 + lg * 16) ^ (((row ^ (row >> 3)) & 7) << 4)));
+        acc[dabase + dl] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            *reinterpret_cast<bf16x8_t*>(&va_s), p_frag[tc],
+            acc[dabase + dl], 0, 0, 0);
+      }
+    if constexpr (PRIO) __builtin_amdgcn_s_setprio(0);
+  };
+
+  if constexpr (VT == 5) {
+    short* klds_w = v_lds;
+    short8 sreg[8];
+    for (int t0 = slab_start; t0 < slab_end; t0 += 64) {
+      f32x4 sT[4];
+      stage_k32(t0, klds_w, sreg);
+      if constexpr (PRIO) __builtin_amdgcn_s_setprio(1);
+      qk_half(t0, 0, klds_w, sT);
+      if constexpr (PRIO) __builtin_amdgcn_s_setprio(0);
+      __builtin_amdgcn_wave_barrier();
+      stage_k32(t0 + 32, klds_w, sreg);
+      if constexpr (PRIO) __builtin_amdgcn_s_setprio(1);
+      qk_half(t0, 1, klds_w, sT);
+      if constexpr (PRIO) __builtin_amdgcn_s_setprio(0);
+      bf16x8_t p_frag[2];
+      softmax_pack64(t0, sT, p_frag);
+      __builtin_amdgcn_wave_barrier();
+      stage_v_half(t0, 0, klds_w);
+      pv_half(0, klds_w, p_frag);
+      __builtin_amdgcn_wave_barrier();
+      stage_v_half(t0, 64, klds_w);
+      pv_half(4, klds_w, p_frag);
+      // back-edge fence: the NEXT tile's staging ds_writes must not be
+      // scheduled above this tile's outstanding LDS frag reads
+      __builtin_amdgcn_wave_barrier();
+    }
+  } else if constexpr (VT == 4) {
     short* klds_w = v_lds;  // reuse the (otherwise unused) V-staging LDS
     short8 v64[16], sreg[8];
     for (int t0 = slab_start; t0 < slab_end; t0 += 64) {
@@ -1152,6 +1230,7 @@ __global__ __launch_bounds__(kBlock, MINW) void paged_decode_mfma_swapped(
       qk_half(t0, 1, klds_w, sT);
       if constexpr (PRIO) __builtin_amdgcn_s_setprio(0);
       tile_vt64_post(t0, sT, v64);
+      __builtin_amdgcn_wave_barrier();  // back-edge fence (see VT5)
     }
   } else if constexpr (VT == 3) {
     short8 k64[16], v64[16];
