@@ -226,7 +226,17 @@ class Scheduler:
             req.state = ReqState.RUNNING
             self.running.append(req)
             sample = (req.num_computed + n == req.total_len)
-            out.prefills.append(ScheduledSeq(req, n, sample))
+            # an admitted request whose whole PROMPT is already computed
+            # (disagg decode-side attach) continues as a DECODE step, not a
+            # 1-token prefill chunk: the decode attention kernel is both
+            # faster and bit-aligned with the aggregated path - routing it
+            # through the prefill kernel produced last-ulp different
+            # layer-2+ KV at the boundary position and broke
+            # disagg==aggregated bit-equality at argmax knife-edges
+            if req.is_decode and n == 1:
+                out.decodes.append(ScheduledSeq(req, n, sample))
+            else:
+                out.prefills.append(ScheduledSeq(req, n, sample))
             budget -= n
 
         return out

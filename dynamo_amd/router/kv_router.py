@@ -49,6 +49,11 @@ class RouterConfig:
     # sticky sessions (reference parity: lib/llm session_affinity/): cap on
     # remembered session -> worker pins (LRU evicted beyond this).
     max_sessions: int = 4096
+    # lower-tier (G2/G3) overlap credit (reference parity: kv-router
+    # indexer/lower_tier.rs): a prefix resident in a worker's HOST tier
+    # onboards over PCIe instead of being recomputed, so it counts toward
+    # the overlap credit at a discount. 0 disables the host indexer.
+    host_overlap_weight: float = 0.8
 
 
 class AllWorkersBusy(Exception):
@@ -108,6 +113,8 @@ class KvRouter:
         self.cfg = cfg or RouterConfig()
         self.client = PushClient(runtime, namespace, component, endpoint)
         self.indexer = _core.KvIndexer()
+        # G2/G3 prefix digests per worker (stored_host/removed_host events)
+        self.host_indexer = _core.KvIndexer()
         self.workers: Dict[str, WorkerState] = {}
         self._sessions: "OrderedDict[str, str]" = OrderedDict()
         # local worker inhibition (reference: distributed-runtime.md "Local
@@ -163,6 +170,7 @@ class KvRouter:
                 if ws.metrics_task:
                     ws.metrics_task.cancel()
                 self.indexer.remove_worker(self._wid(iid))
+                self.host_indexer.remove_worker(self._wid(iid))
 
     @staticmethod
     def _wid(instance_id: str) -> int:
@@ -185,8 +193,13 @@ class KvRouter:
                         self.indexer.apply_stored(wid, ev["hashes"])
                     elif ev["kind"] == "removed":
                         self.indexer.apply_removed(wid, ev["hashes"])
+                    elif ev["kind"] == "stored_host":
+                        self.host_indexer.apply_stored(wid, ev["hashes"])
+                    elif ev["kind"] == "removed_host":
+                        self.host_indexer.apply_removed(wid, ev["hashes"])
                     elif ev["kind"] == "cleared":
                         self.indexer.clear_worker(wid)
+                        self.host_indexer.clear_worker(wid)
         except (ConnectionError, OSError, asyncio.CancelledError):
             pass  # worker died; watch loop will clean up
         except Exception:
@@ -282,7 +295,7 @@ class KvRouter:
             import dataclasses
             allowed = {"mode", "overlap_score_weight",
                        "decode_active_request_weight", "prefill_load_scale",
-                       "router_temperature"}
+                       "router_temperature", "host_overlap_weight"}
             cfg = dataclasses.replace(
                 cfg, **{k: v for k, v in override.items() if k in allowed})
         mode = cfg.mode
@@ -300,12 +313,19 @@ class KvRouter:
         bs = cfg.block_size
         hashes = _core.chain_hashes(token_ids, bs, cfg.block_salt)
         matches = self.indexer.find_matches(hashes)
+        host_matches = (self.host_indexer.find_matches(hashes)
+                        if cfg.host_overlap_weight > 0 else {})
         prefill_blocks = (len(token_ids) + bs - 1) // bs
         logits = []
         for inst in insts:
             ws = self.workers.get(inst.instance_id) or WorkerState(
                 inst.instance_id)
-            overlap = matches.get(self._wid(inst.instance_id), 0)
+            wid = self._wid(inst.instance_id)
+            overlap = matches.get(wid, 0)
+            # a deeper HOST-resident prefix beats a shallower device one at
+            # the onboard discount (lower_tier.rs semantics)
+            overlap = max(overlap,
+                          cfg.host_overlap_weight * host_matches.get(wid, 0))
             cost = (cfg.prefill_load_scale
                     * max(0.0, prefill_blocks
                           - cfg.overlap_score_weight * overlap)

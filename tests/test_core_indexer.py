@@ -156,3 +156,45 @@ def test_cuckoo_no_false_negatives_and_roundtrip(items, probe):
     for h in ok:
         assert g.contains(h)
     assert g.contains(probe) == f.contains(probe)
+
+
+def test_router_host_tier_overlap_routing():
+    """stored_host events steer routing at the onboard discount: a worker
+    with a DEEPER host-resident prefix beats a shallower device-only one,
+    but a device prefix of equal depth wins (lower_tier.rs parity)."""
+    from dynamo_amd.router.kv_router import KvRouter, RouterConfig
+
+    class _Inst:
+        def __init__(self, iid):
+            self.instance_id = iid
+            self.address = iid
+
+    class _Client:
+        def __init__(self, insts):
+            self._insts = insts
+
+        def instances(self):
+            return self._insts
+
+    r = KvRouter.__new__(KvRouter)
+    r.cfg = RouterConfig(block_size=16)
+    r.client = _Client([_Inst("A"), _Inst("B")])
+    r.indexer = _core.KvIndexer()
+    r.host_indexer = _core.KvIndexer()
+    r.workers = {}
+    r.sessions = {}
+    r._inhibited = {}
+    r._rr = 0
+    toks = list(range(96))                     # 6 blocks
+    h = _core.chain_hashes(toks, 16, 0)
+    wa, wb = r._wid("A"), r._wid("B")
+    # A: 2 blocks hot on device; B: 5 blocks on HOST tier only
+    r.indexer.apply_stored(wa, h[:2])
+    r.host_indexer.apply_stored(wb, h[:5])
+    assert r.select(toks) == "B"               # 0.8*5 = 4 > 2
+    # equal depth: device beats host (discount < 1)
+    r.indexer.apply_stored(wa, h[:5])
+    assert r.select(toks) == "A"
+    # host credit disabled -> B has no credit at all
+    r.cfg.host_overlap_weight = 0.0
+    assert r.select(toks) == "A"

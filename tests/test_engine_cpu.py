@@ -471,3 +471,27 @@ def test_v_transposed_config_resolution():
     import dataclasses
     mha = dataclasses.replace(mha, num_kv_heads=mha.num_q_heads)
     assert not EngineConfig(model=mha, device="cuda:0").v_transposed
+
+
+def test_attached_request_first_step_is_decode():
+    """Disagg decode-side attach: a request admitted with its whole prompt
+    already computed must continue as a DECODE step, not a 1-token prefill
+    chunk (the prefill kernel's different rounding at the boundary position
+    broke disagg == aggregated bit-equality; caught on GPU)."""
+    from dynamo_amd.engine import EngineConfig, LLMEngine, SamplingParams
+    from dynamo_amd.engine.config import PRESETS
+    cfg = EngineConfig(model=PRESETS["tiny-llama"], device="cpu",
+                       kv_pool_pages=64, max_model_len=512, page_size=16)
+    eng = LLMEngine(cfg)
+    prompt = list(range(100))
+    req = eng.add_request("a", prompt, SamplingParams(max_tokens=4,
+                                                      ignore_eos=True))
+    from dynamo_amd.engine.kv_cache import SequenceKV
+    kv = SequenceKV(eng.alloc, cfg.block_salt)
+    kv.ensure_capacity(len(prompt))
+    req.kv = kv
+    req.num_computed = len(prompt)
+    req.output_tokens.append(7)
+    sched = eng.scheduler.schedule()
+    assert len(sched.decodes) == 1 and not sched.prefills
+    assert sched.decodes[0].req is req and sched.decodes[0].n_new == 1
