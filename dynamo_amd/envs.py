@@ -26,6 +26,19 @@ ENV_VARS = {
     "DYN_BYPASS_TOKEN_THRESHOLD":
         "conditional disagg: net-new prefill tokens below this bypass the "
         "prefill pool (reference conditional_disagg.rs:15; default 2048)",
+    # kernel bring-up / A-B toggles (read by the HIP dispatch layer)
+    "DYNAMO_DECODE_CHUNK": "decode attention context-chunk tokens override",
+    "DYNAMO_DECODE_MFMA": "0 = force the VALU decode path (token-major only)",
+    "DYNAMO_DECODE_SWAPPED": "0 = A-operand MFMA decode (token-major only)",
+    "DYNAMO_VT3_VARIANT":
+        "d-major decode variant select: 0=VT4 (default), 1=VT3, 2=VT2, "
+        "5=VT5 (recorded-negative V-LDS variant)",
+    "DYNAMO_FUSED_MERGE":
+        "1 = fused decode chunk merge (recorded-negative; threadfence cost)",
+    "DYNAMO_MOE_BMM": "0 = grouped-kernel MoE decode instead of padded bmm",
+    "DYNAMO_MOE_MFMA": "0 = VALU grouped MoE GEMM",
+    "DYNAMO_MOE_GRAPHS": "0 = exclude MoE layers from hipGraph capture",
+    "DYNAMO_ROCTX": "1 = roctx range annotations for rocprofv3 traces",
 }
 
 
