@@ -67,3 +67,45 @@ def run_profile(url: str, model: str, isl: int, osl: int,
     if out:
         Path(out).write_text(json.dumps(result, indent=1))
     return result
+
+
+def choose_parallelization(profiles: dict, itl_slo_ms: float,
+                           ttft_slo_s: float, total_gpus: int,
+                           req_per_s: float, isl: int, osl: int) -> dict:
+    """TP-config search over per-config profiles (reference parity:
+    profiler/profile_sla.py:343's parallelism sweep selection).
+
+    `profiles` maps a config label -> {"parallel": ParallelizationConfig,
+    "profile": run_profile() result}. For each config that meets the SLOs,
+    compute the replicas needed for `req_per_s` and its GPU cost; pick the
+    config serving the load with the fewest GPUs (ties -> higher headroom
+    per GPU). Configs whose replica demand exceeds `total_gpus` are
+    reported as infeasible."""
+    import math
+
+    from dynamo_amd.planner.planner import ParallelizationConfig
+
+    scored = []
+    for label, ent in profiles.items():
+        par = ent.get("parallel") or ParallelizationConfig()
+        pm = ent["profile"]["perf_model"]
+        if not ent["profile"].get("meets_slo") or not pm["max_conc_at_itl"]:
+            scored.append({"config": label, "feasible": False,
+                           "reason": "SLO not met at any concurrency"})
+            continue
+        tps = pm["decode_tokens_per_s_at_itl"]
+        prefill_tps = pm["prefill_tokens_per_s"] or float("inf")
+        n_decode = max(1, math.ceil(req_per_s * osl / max(1.0, tps)))
+        n_prefill = max(1, math.ceil(req_per_s * isl / prefill_tps))
+        replicas = max(n_decode, n_prefill)
+        gpus = replicas * par.gpus_per_replica
+        scored.append({
+            "config": label, "feasible": gpus <= total_gpus,
+            "replicas": replicas, "gpus": gpus,
+            "tokens_per_gpu": tps / par.gpus_per_replica,
+        })
+    feasible = [s for s in scored if s.get("feasible")]
+    best = (min(feasible, key=lambda s: (s["gpus"], -s["tokens_per_gpu"]))
+            if feasible else None)
+    return {"candidates": scored,
+            "best": best["config"] if best else None}

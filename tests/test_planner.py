@@ -358,3 +358,45 @@ def test_sla_planner_respects_gpu_budget():
     gpus = sum(v * 4 for v in targets.values())
     assert gpus <= 8
     assert all(v >= 1 for v in targets.values())
+
+
+def test_profiler_choose_parallelization():
+    """TP-config search: picks the config that serves the load with the
+    fewest GPUs among SLO-meeting candidates."""
+    from dynamo_amd.planner.planner import ParallelizationConfig
+    from dynamo_amd.profiler.profile_sla import choose_parallelization
+
+    def prof(meets, tps, prefill_tps=1e9, conc=16):
+        return {"meets_slo": meets, "perf_model": {
+            "max_conc_at_itl": conc if meets else None,
+            "decode_tokens_per_s_at_itl": tps,
+            "prefill_tokens_per_s": prefill_tps}}
+
+    profiles = {
+        # TP1: cheap per replica but low throughput -> many replicas
+        "tp1": {"parallel": ParallelizationConfig(tp_size=1),
+                "profile": prof(True, tps=400.0)},
+        # TP4: 3x throughput for 4x GPUs
+        "tp4": {"parallel": ParallelizationConfig(tp_size=4),
+                "profile": prof(True, tps=1200.0)},
+        # TP8 misses SLO entirely
+        "tp8": {"parallel": ParallelizationConfig(tp_size=8),
+                "profile": prof(False, tps=0.0)},
+    }
+    r = choose_parallelization(profiles, itl_slo_ms=25, ttft_slo_s=2,
+                               total_gpus=8, req_per_s=2.0, isl=8192,
+                               osl=1024)
+    # demand: 2 req/s * 1024 osl = 2048 tok/s -> tp1 needs 6 replicas
+    # (6 GPUs), tp4 needs 2 replicas (8 GPUs) -> tp1 wins on GPU count
+    assert r["best"] == "tp1"
+    by = {c["config"]: c for c in r["candidates"]}
+    assert by["tp1"]["gpus"] == 6 and by["tp4"]["gpus"] == 8
+    assert by["tp8"]["feasible"] is False
+    # tighter budget: only tp1 fits
+    r = choose_parallelization(profiles, 25, 2, total_gpus=6,
+                               req_per_s=2.0, isl=8192, osl=1024)
+    assert r["best"] == "tp1"
+    # huge load: nothing fits -> best None, all reported
+    r = choose_parallelization(profiles, 25, 2, total_gpus=2,
+                               req_per_s=50.0, isl=8192, osl=1024)
+    assert r["best"] is None
