@@ -220,8 +220,14 @@ def build_app(manager: ModelManager) -> FastAPI:
             return {"count": 0}
         from dynamo_amd import _core
         hashes = r.indexer.all_hashes()
-        cf = _core.CuckooFilter(max(1024, len(hashes) * 2))
+        # host-tier (G2/G3) blocks are onboardable prefix warmth too -
+        # include them so cross-pool ranking sees the full cached set
+        host_hashes = r.host_indexer.all_hashes()
+        cf = _core.CuckooFilter(max(1024, (len(hashes)
+                                           + len(host_hashes)) * 2))
         for h in hashes:
+            cf.insert(h)
+        for h in host_hashes:
             cf.insert(h)
         return {"b64": _b64.b64encode(cf.to_bytes()).decode(),
                 "count": cf.count(), "block_size": r.cfg.block_size,
