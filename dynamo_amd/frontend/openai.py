@@ -197,7 +197,13 @@ def build_app(manager: ModelManager) -> FastAPI:
         hashes = _core.chain_hashes(list(token_ids), r.cfg.block_size,
                                     r.cfg.block_salt)
         matches = r.indexer.find_matches(hashes)
-        return {"overlap_blocks": max(matches.values()) if matches else 0,
+        best = max(matches.values()) if matches else 0
+        if r.cfg.host_overlap_weight > 0:
+            hm = r.host_indexer.find_matches(hashes)
+            if hm:
+                best = max(best, int(r.cfg.host_overlap_weight
+                                     * max(hm.values())))
+        return {"overlap_blocks": best,
                 "block_size": r.cfg.block_size,
                 "total_blocks": len(hashes)}
 
