@@ -970,3 +970,18 @@ def test_files_and_batches_api():
         assert all(o["response"]["status_code"] == 200 for o in out)
         await teardown(services, mgr, client)
     run(main())
+
+
+def test_recipe_cpu_check():
+    """recipes/opt125m-cpu-agg.sh --check: the BASELINE config #1 recipe
+    boots a CPU worker + frontend and answers /health."""
+    import pathlib
+    import subprocess
+    root = pathlib.Path(__file__).resolve().parents[1]
+    r = subprocess.run(["bash", str(root / "recipes" / "opt125m-cpu-agg.sh"),
+                        "--check"],
+                       capture_output=True, text=True, timeout=240,
+                       env={**__import__("os").environ,
+                            "DYN_HTTP_PORT": "8077"})
+    assert r.returncode == 0, r.stdout + r.stderr
+    assert "CHECK OK" in r.stdout
