@@ -10,6 +10,7 @@ from __future__ import annotations
 
 import asyncio
 import json
+import logging
 import time
 import uuid
 from typing import AsyncIterator, List, Optional, Union
@@ -22,6 +23,8 @@ from prometheus_client import (CONTENT_TYPE_LATEST, Counter, Histogram,
                                generate_latest)
 
 from .service import ModelManager
+
+log = logging.getLogger("dynamo_amd.frontend")
 
 REQS = Counter("dynamo_amd_requests_total", "requests", ["model", "route"])
 TTFT = Histogram("dynamo_amd_ttft_seconds", "time to first token", ["model"])
@@ -79,6 +82,13 @@ class ChatRequest(BaseModel):
     seed: int = 0
     ignore_eos: bool = False
     user: Optional[str] = None      # sticky-session key
+    # next-turn KV warming (reference parity: preprocessor/
+    # speculative_prefill.rs nvext.agent_hints.speculative_prefill): after
+    # the response completes, fire a background max_tokens=1 request with
+    # the re-rendered conversation (assistant turn included, no generation
+    # prompt) so the NEXT user turn hits a warm prefix.
+    speculative_prefill: Optional[bool] = None
+    nvext: Optional[dict] = None
 
 
 class AnthropicMessagesRequest(BaseModel):
@@ -734,6 +744,31 @@ def build_app(manager: ModelManager) -> FastAPI:
                 token_ids.extend([0] * emb["shape"][0])
         return token_ids, {"mm_embeds": mm}
 
+    def _spec_prefill_enabled(req) -> bool:
+        v = getattr(req, "speculative_prefill", None)
+        if v is not None:
+            return bool(v)
+        hints = (getattr(req, "nvext", None) or {}).get("agent_hints") or {}
+        return bool(hints.get("speculative_prefill"))
+
+    async def _spec_prefill(entry, messages: List[dict], reply_text: str):
+        """Warm the next turn's prefix: render the conversation WITH the
+        completed assistant message and no generation prompt (the exact
+        prefix the next user turn extends), then run a 1-token request
+        through the normal routed path - the KV router lands it on the
+        warmest worker and the prefix cache keeps the blocks."""
+        try:
+            convo = messages + [{"role": "assistant", "content": reply_text}]
+            prefix = entry.templater.render(convo,
+                                            add_generation_prompt=False)
+            toks = await asyncio.to_thread(entry.tokenizer.encode, prefix)
+            sp = {"temperature": 0.0}
+            opts = {"max_tokens": 1, "ignore_eos": True}
+            async for _ in manager.generate_tokens(entry, toks, sp, opts):
+                pass
+        except Exception:
+            log.debug("speculative next-turn prefill failed", exc_info=True)
+
     @app.post("/v1/chat/completions")
     async def chat(req: ChatRequest, raw: Request):
         entry = _entry_or_404(req.model)
@@ -806,6 +841,10 @@ def build_app(manager: ModelManager) -> FastAPI:
                             yield event(acc[sent:], finish, role)
                             sent = len(acc)
                     yield "data: [DONE]\n\n"
+                    if _spec_prefill_enabled(req):
+                        asyncio.get_running_loop().create_task(_spec_prefill(
+                            entry, [m.model_dump() for m in req.messages],
+                            acc))
                 finally:
                     LATENCY.labels(entry.name).observe(time.time() - t0)
             return StreamingResponse(sse(), media_type="text/event-stream")
@@ -841,6 +880,10 @@ def build_app(manager: ModelManager) -> FastAPI:
             message["reasoning_content"] = reasoning
         if tool_calls:
             message["tool_calls"] = tool_calls
+        if _spec_prefill_enabled(req):
+            asyncio.get_running_loop().create_task(_spec_prefill(
+                entry, [m.model_dump() for m in req.messages],
+                text_out or ""))
         return {
             "id": rid, "object": "chat.completion", "created": int(t0),
             "model": entry.name,

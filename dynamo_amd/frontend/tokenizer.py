@@ -17,7 +17,9 @@ DEFAULT_CHAT_TEMPLATE = (
     "{% for m in messages %}"
     "<|start_header_id|>{{ m.role }}<|end_header_id|>\n\n{{ m.content }}<|eot_id|>"
     "{% endfor %}"
+    "{% if add_generation_prompt %}"
     "<|start_header_id|>assistant<|end_header_id|>\n\n"
+    "{% endif %}"
 )
 
 
@@ -93,5 +95,7 @@ class ChatTemplater:
         self.env = jinja2.Environment()
         self.template = self.env.from_string(template or DEFAULT_CHAT_TEMPLATE)
 
-    def render(self, messages: List[dict]) -> str:
-        return self.template.render(messages=messages)
+    def render(self, messages: List[dict],
+               add_generation_prompt: bool = True) -> str:
+        return self.template.render(messages=messages,
+                                    add_generation_prompt=add_generation_prompt)

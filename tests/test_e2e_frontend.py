@@ -985,3 +985,78 @@ def test_recipe_cpu_check():
                             "DYN_HTTP_PORT": "8077"})
     assert r.returncode == 0, r.stdout + r.stderr
     assert "CHECK OK" in r.stdout
+
+
+def test_speculative_next_turn_prefill():
+    """speculative_prefill hint: after a chat turn completes, the frontend
+    fires a background 1-token request with the re-rendered next-turn
+    prefix, so the router's KV indexer learns the warmed blocks
+    (preprocessor/speculative_prefill.rs parity)."""
+    async def main():
+        # real CPU engine (the mock engine emits no KV events, so the
+        # router indexer would never learn the warmed blocks)
+        from dynamo_amd.engine import EngineConfig, LLMEngine
+        from dynamo_amd.engine.config import PRESETS
+        shared = MemoryDiscovery()
+        rt = DistributedRuntime(shared)
+        cfg = EngineConfig(model=PRESETS["tiny-llama"], device="cpu",
+                           kv_pool_pages=128, max_model_len=2048,
+                           page_size=16, max_batched_tokens=2048)
+        eng = LLMEngine(cfg)
+        eng.model_config = PRESETS["tiny-llama"]
+        ws = WorkerService(eng, rt, component="backend")
+        await ws.start()
+        services = [(ws, rt)]
+        mgr_rt = DistributedRuntime(shared)
+        mgr = ModelManager(mgr_rt)
+        await mgr.start(watch_interval=0.2)
+        app = build_app(mgr)
+        transport = httpx.ASGITransport(app=app)
+        client = httpx.AsyncClient(transport=transport, base_url="http://t")
+        for _ in range(50):
+            try:
+                mgr.get("tiny-llama")
+                break
+            except KeyError:
+                await asyncio.sleep(0.1)
+        msgs = [{"role": "user", "content": "tell me a story " * 40}]
+        r = await client.post("/v1/chat/completions", json={
+            "model": "tiny-llama", "messages": msgs, "max_tokens": 4,
+            "nvext": {"agent_hints": {"speculative_prefill": True}}})
+        assert r.status_code == 200, r.text
+        reply = r.json()["choices"][0]["message"]["content"] or ""
+        # the warm request is async; give it a few event-loop turns
+        entry = mgr.get("tiny-llama")
+        prefix = entry.templater.render(
+            msgs + [{"role": "assistant", "content": reply}],
+            add_generation_prompt=False)
+        toks = entry.tokenizer.encode(prefix)
+        overlap, total = 0, 1
+        for _ in range(50):
+            await asyncio.sleep(0.1)
+            rr = await client.post("/internal/kv_overlap", json={
+                "model": "tiny-llama", "token_ids": toks})
+            overlap = rr.json()["overlap_blocks"]
+            total = rr.json()["total_blocks"]
+            if overlap >= total:
+                break
+        # the warmed next-turn prefix is FULLY cached (every block,
+        # including the re-framed assistant turn past the prompt boundary)
+        assert overlap == total, f"warmed {overlap}/{total}"
+        # without the hint, a fresh conversation's prefix stays cold
+        msgs2 = [{"role": "user", "content": "completely different " * 40}]
+        r2 = await client.post("/v1/chat/completions", json={
+            "model": "tiny-llama", "messages": msgs2, "max_tokens": 4})
+        reply2 = r2.json()["choices"][0]["message"]["content"] or ""
+        prefix2 = entry.templater.render(
+            msgs2 + [{"role": "assistant", "content": reply2}],
+            add_generation_prompt=False)
+        toks2 = entry.tokenizer.encode(prefix2)
+        await asyncio.sleep(0.5)
+        rr2 = await client.post("/internal/kv_overlap", json={
+            "model": "tiny-llama", "token_ids": toks2})
+        # without the hint only the served-prompt prefix is cached: the
+        # re-framed assistant tail stays cold, so coverage is partial
+        assert rr2.json()["overlap_blocks"] < rr2.json()["total_blocks"]
+        await teardown(services, mgr, client)
+    run(main())
