@@ -34,6 +34,10 @@ def build_parser():
     p.add_argument("--record", default=None, metavar="FILE",
                    help="record requests + response chunks to a JSONL file "
                         "(replay with python -m dynamo_amd.tools.replay)")
+    p.add_argument("--request-template", default=None, metavar="FILE",
+                   help="JSON {model, temperature, max_completion_tokens} "
+                        "defaults applied when a request omits the field "
+                        "(reference request_template.rs parity)")
     p.add_argument("--grpc-port", type=int, default=0,
                    help="also serve the KServe v2 gRPC inference protocol "
                         "on this port (0 = disabled)")
@@ -50,7 +54,11 @@ async def async_main(args):
                            busy_threshold=args.busy_threshold),
                        record_path=args.record)
     await mgr.start()
-    app = build_app(mgr)
+    tmpl = None
+    if args.request_template:
+        import json as _json
+        tmpl = _json.loads(open(args.request_template).read())
+    app = build_app(mgr, request_template=tmpl)
     config = uvicorn.Config(app, host=args.host, port=args.port,
                             log_level="warning")
     server = uvicorn.Server(config)

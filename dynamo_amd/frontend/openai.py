@@ -41,8 +41,8 @@ OUT_TOKENS = Counter("dynamo_amd_output_tokens_total", "output tokens",
 class CompletionRequest(BaseModel):
     model: str = ""
     prompt: Union[str, List[int]] = ""
-    max_tokens: int = 128
-    temperature: float = 0.0
+    max_tokens: Optional[int] = None
+    temperature: Optional[float] = None
     top_p: float = 1.0
     top_k: int = 0
     stream: bool = False
@@ -73,8 +73,8 @@ class ChatRequest(BaseModel):
     model: str = ""
     messages: List[ChatMessage] = Field(default_factory=list)
     tools: Optional[List[dict]] = None   # enables tool-call parsing
-    max_tokens: int = 128
-    temperature: float = 0.0
+    max_tokens: Optional[int] = None
+    temperature: Optional[float] = None
     top_p: float = 1.0
     top_k: int = 0
     stream: bool = False
@@ -95,10 +95,10 @@ class AnthropicMessagesRequest(BaseModel):
     """Anthropic Messages API shape (reference parity:
     lib/llm/src/http/service/anthropic.rs /v1/messages)."""
     model: str = ""
-    max_tokens: int = 128
+    max_tokens: Optional[int] = None
     messages: List[ChatMessage] = Field(default_factory=list)
     system: Optional[str] = None
-    temperature: float = 0.0
+    temperature: Optional[float] = None
     top_p: float = 1.0
     top_k: int = 0
     stream: bool = False
@@ -114,7 +114,7 @@ class ResponsesRequest(BaseModel):
     input: Union[str, List[dict]] = ""
     instructions: Optional[str] = None
     max_output_tokens: int = 128
-    temperature: float = 0.0
+    temperature: Optional[float] = None
     top_p: float = 1.0
     stream: bool = False
     user: Optional[str] = None
@@ -144,9 +144,28 @@ def _find_stop(text: str, stops: List[str]) -> int:
     return best
 
 
-def build_app(manager: ModelManager) -> FastAPI:
+def build_app(manager: ModelManager,
+              request_template: Optional[dict] = None) -> FastAPI:
+    """request_template (reference parity: request_template.rs + the
+    frontend --request-template flag): {"model", "temperature",
+    "max_completion_tokens"} defaults applied when a request omits the
+    field; an explicit value always wins."""
     app = FastAPI(title="dynamo_amd", version="0.1.0")
     app.state.manager = manager
+    tmpl = request_template or {}
+
+    def _model_of(req) -> str:
+        return req.model or tmpl.get("model", "")
+
+    def _temp_of(req) -> float:
+        if getattr(req, "temperature", None) is not None:
+            return req.temperature
+        return float(tmpl.get("temperature", 0.0))
+
+    def _max_tokens_of(req) -> int:
+        if getattr(req, "max_tokens", None) is not None:
+            return req.max_tokens
+        return int(tmpl.get("max_completion_tokens", 128))
 
     @app.get("/health")
     async def health():
@@ -167,12 +186,12 @@ def build_app(manager: ModelManager) -> FastAPI:
 
     async def _run(entry, token_ids, req, rid,
                    session_id=None, extra=None) -> AsyncIterator[dict]:
-        sampling = {"temperature": req.temperature, "top_p": req.top_p,
+        sampling = {"temperature": _temp_of(req), "top_p": req.top_p,
                     "top_k": getattr(req, "top_k", 0),
                     "seed": getattr(req, "seed", 0),
                     "logprobs": getattr(req, "logprobs", None) or 0}
         eos = getattr(entry.tokenizer, "eos_id", None)
-        stop = {"max_tokens": req.max_tokens,
+        stop = {"max_tokens": _max_tokens_of(req),
                 "ignore_eos": getattr(req, "ignore_eos", False),
                 "stop_token_ids": [eos] if eos is not None else []}
         from dynamo_amd.router.kv_router import AllWorkersBusy
@@ -273,7 +292,7 @@ def build_app(manager: ModelManager) -> FastAPI:
     # -- OpenAI Responses API (openai.rs:4158 handler_responses) --------
     @app.post("/v1/responses")
     async def responses(req: ResponsesRequest, raw: Request):
-        entry = _entry_or_404(req.model)
+        entry = _entry_or_404(_model_of(req))
         REQS.labels(entry.name, "responses").inc()
         if isinstance(req.input, str):
             msgs = [{"role": "user", "content": req.input}]
@@ -466,7 +485,7 @@ def build_app(manager: ModelManager) -> FastAPI:
 
     @app.post("/v1/completions")
     async def completions(req: CompletionRequest, raw: Request):
-        entry = _entry_or_404(req.model)
+        entry = _entry_or_404(_model_of(req))
         REQS.labels(entry.name, "completions").inc()
         if isinstance(req.prompt, list):
             token_ids = list(req.prompt)
@@ -586,7 +605,7 @@ def build_app(manager: ModelManager) -> FastAPI:
     @app.post("/v1/messages")
     async def anthropic_messages(req: AnthropicMessagesRequest, raw: Request):
         """Anthropic Messages API over the same engine pipeline."""
-        entry = _entry_or_404(req.model)
+        entry = _entry_or_404(_model_of(req))
         REQS.labels(entry.name, "messages").inc()
         msgs = [m.model_dump() for m in req.messages]
         if req.system:
@@ -663,7 +682,7 @@ def build_app(manager: ModelManager) -> FastAPI:
     async def embeddings(req: EmbeddingRequest):
         """Mean-pooled last-hidden-state embeddings (reference parity:
         lib/llm/src/http embeddings route)."""
-        entry = _entry_or_404(req.model)
+        entry = _entry_or_404(_model_of(req))
         REQS.labels(entry.name, "embeddings").inc()
         raw_inputs = req.input
         if isinstance(raw_inputs, str):
@@ -771,7 +790,7 @@ def build_app(manager: ModelManager) -> FastAPI:
 
     @app.post("/v1/chat/completions")
     async def chat(req: ChatRequest, raw: Request):
-        entry = _entry_or_404(req.model)
+        entry = _entry_or_404(_model_of(req))
         REQS.labels(entry.name, "chat").inc()
         mm_extra = None
         mm = await _mm_prepare(entry, req)

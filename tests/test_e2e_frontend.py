@@ -1060,3 +1060,37 @@ def test_speculative_next_turn_prefill():
         assert rr2.json()["overlap_blocks"] < rr2.json()["total_blocks"]
         await teardown(services, mgr, client)
     run(main())
+
+
+def test_request_template_defaults():
+    """--request-template defaults: empty model resolves to the template's
+    model; omitted max_tokens takes max_completion_tokens; explicit values
+    win (request_template.rs parity)."""
+    async def main():
+        shared, services, mgr, client0 = await with_stack(nworkers=1)
+        await client0.aclose()
+        app = build_app(mgr, request_template={
+            "model": "mock-model", "temperature": 0.0,
+            "max_completion_tokens": 5})
+        transport = httpx.ASGITransport(app=app)
+        client = httpx.AsyncClient(transport=transport, base_url="http://t")
+        # omitted model + max_tokens -> template values
+        r = await client.post("/v1/completions", json={"prompt": "hi"})
+        assert r.status_code == 200, r.text
+        assert r.json()["usage"]["completion_tokens"] == 5
+        assert r.json()["model"] == "mock-model"
+        # explicit values win
+        r = await client.post("/v1/completions", json={
+            "model": "mock-model", "prompt": "hi", "max_tokens": 3})
+        assert r.json()["usage"]["completion_tokens"] == 3
+        # no template app: omitted max_tokens keeps the built-in default
+        # (empty model falls back to the single registered model)
+        app2 = build_app(mgr)
+        c2 = httpx.AsyncClient(transport=httpx.ASGITransport(app=app2),
+                               base_url="http://t")
+        r = await c2.post("/v1/completions", json={"prompt": "hi"})
+        assert r.status_code == 200
+        assert r.json()["usage"]["completion_tokens"] == 128
+        await c2.aclose()
+        await teardown(services, mgr, client)
+    run(main())
