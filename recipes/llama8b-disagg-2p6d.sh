@@ -6,18 +6,28 @@ source recipes/_lib.sh
 MODEL=${MODEL:-llama-3-8b}
 NP=${NP:-2}
 ND=${ND:-6}
+# GPUS: space-separated device list, one per worker (default 0..NP+ND-1);
+# e.g. GPUS="0 0" NP=1 ND=1 co-locates both workers on one GPU for smoke
+GPUS=${GPUS:-$(seq -s" " 0 $((NP + ND - 1)))}
+# per-worker HBM fraction (lower it when co-locating workers on one GPU)
+MEMFRAC=${MEMFRAC:-0.9}
+read -ra GL <<< "$GPUS"
 
 g=0
 for i in $(seq 1 "$NP"); do
-  python -m dynamo_amd.workers --model "$MODEL" --device "cuda:$g" \
+  : > "$NS/prefill$i.log"   # create before fork (the ready-wait globs)
+  python -m dynamo_amd.workers --model "$MODEL" --device "cuda:${GL[$g]}" \
       --worker-type prefill --discovery "$DISC" \
-      > "$NS/prefill$i.log" 2>&1 &
+      --gpu-mem-fraction "$MEMFRAC" \
+      >> "$NS/prefill$i.log" 2>&1 &
   PIDS+=($!); g=$((g+1))
 done
 for i in $(seq 1 "$ND"); do
-  python -m dynamo_amd.workers --model "$MODEL" --device "cuda:$g" \
+  : > "$NS/decode$i.log"
+  python -m dynamo_amd.workers --model "$MODEL" --device "cuda:${GL[$g]}" \
       --worker-type decode --discovery "$DISC" \
-      > "$NS/decode$i.log" 2>&1 &
+      --gpu-mem-fraction "$MEMFRAC" \
+      >> "$NS/decode$i.log" 2>&1 &
   PIDS+=($!); g=$((g+1))
 done
 for f in "$NS"/prefill*.log "$NS"/decode*.log; do
