@@ -39,6 +39,9 @@ class ModelConfig:
     learned_pos_emb: bool = False
     # qwen2-style QKV bias
     attn_bias: bool = False
+    # local HF checkpoint dir to load real weights from ("" = random init)
+    weights_path: str = ""
+
 
     def to_dict(self):
         return dataclasses.asdict(self)

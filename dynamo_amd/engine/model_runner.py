@@ -35,6 +35,12 @@ class ModelRunner:
                                          seed)
         else:
             self.model = build_model(m, self.device, self.dtype, self.tp, seed)
+        if getattr(m, "weights_path", ""):
+            from dynamo_amd.models.loader import load_weights
+            n = load_weights(self.model, m.weights_path)
+            import logging
+            logging.getLogger("dynamo_amd.engine").info(
+                "loaded %d checkpoint tensors from %s", n, m.weights_path)
         self.hkv_local = max(1, m.num_kv_heads // self.tp.size)
         self.hq_local = m.num_q_heads // self.tp.size
 
