@@ -141,3 +141,36 @@ def test_gpu_fp8_kv_cache_generation():
     match = sum(a == b for x, y in zip(bf, f8) for a, b in zip(x, y))
     total = sum(len(x) for x in bf)
     assert match >= total - 2, (bf, f8)
+
+
+def test_engine_from_checkpoint_dir(tmp_path):
+    """HF-checkpoint loading on GPU: an engine built from an exported
+    checkpoint dir generates the same tokens as the source-init engine."""
+    import dataclasses
+
+    from dynamo_amd.engine import EngineConfig, LLMEngine, SamplingParams
+    from dynamo_amd.engine.config import PRESETS
+    from dynamo_amd.models.loader import export_hf
+    from dynamo_amd.models.registry import build_model, resolve_model_config
+    cfg = dataclasses.replace(PRESETS["tiny-llama-gpu"])
+    src = build_model(cfg, "cuda:0", torch.bfloat16, None, seed=21)
+    export_hf(src, str(tmp_path))
+    del src
+    torch.cuda.empty_cache()
+
+    def gen(ref):
+        e = LLMEngine(EngineConfig(model=ref, device="cuda:0",
+                                   kv_pool_pages=128, max_model_len=1024,
+                                   page_size=64), seed=21)
+        e.add_request("x", list(range(200)),
+                      SamplingParams(max_tokens=8, ignore_eos=True))
+        toks = []
+        while e.has_work():
+            for so in e.step():
+                toks.append(so.new_token)
+        del e
+        torch.cuda.empty_cache()
+        return toks
+    a = gen(cfg)
+    b = gen(resolve_model_config(str(tmp_path)))
+    assert a == b
