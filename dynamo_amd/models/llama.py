@@ -1,8 +1,9 @@
 """Llama-family causal LM (Llama-3 8B/70B, Qwen2-style also fits) —
 MI355X-native forward over the paged KV cache.
 
-Weights are random-initialized (no network for checkpoints; BASELINE.md
-benches on synthetic data / random weights).
+Weights are random-initialized by default (BASELINE.md benches on
+synthetic data / random weights); local HF safetensors checkpoints load
+via models/loader.py (`--model /path/to/checkpoint`).
 """
 from __future__ import annotations
 
