@@ -91,7 +91,8 @@ class WorkerService:
 
     def _card_body(self) -> dict:
         cfg = self.engine.cfg
-        return {
+        body = self._card_extras()
+        body.update({
             "name": self.model_name,
             "model_config": cfg.model.to_dict(),
             "context_length": cfg.max_model_len,
@@ -103,7 +104,32 @@ class WorkerService:
                 "max_num_batched_tokens": cfg.max_batched_tokens,
                 "worker_type": self.worker_type,
             },
-        }
+        })
+        return body
+
+    def _card_extras(self) -> dict:
+        """Tokenizer + chat template advertised from a local HF checkpoint
+        dir (single-node: the frontend shares the filesystem). Falls back
+        to the byte tokenizer + default template when absent."""
+        import json as _json
+        import os as _os
+        wp = getattr(self.engine.cfg.model, "weights_path", "")
+        out: dict = {}
+        if not wp:
+            return out
+        tok = _os.path.join(wp, "tokenizer.json")
+        if _os.path.exists(tok):
+            out["tokenizer"] = {"type": "hf", "path": tok}
+        tcfg = _os.path.join(wp, "tokenizer_config.json")
+        if _os.path.exists(tcfg):
+            try:
+                with open(tcfg) as f:
+                    tmpl = _json.load(f).get("chat_template")
+                if tmpl:
+                    out["chat_template"] = tmpl
+            except (OSError, ValueError):
+                pass
+        return out
 
     async def start(self):
         self.comp.serve_endpoint("generate", self.generate)

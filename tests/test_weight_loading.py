@@ -118,3 +118,46 @@ def test_engine_with_checkpoint(tmp_path):
     a = gen(cfg)                                   # random init, seed 11
     b = gen(resolve_model_config(str(tmp_path)))   # loaded checkpoint
     assert a == b
+
+
+def test_worker_card_advertises_checkpoint_tokenizer(tmp_path):
+    """A worker built from a checkpoint dir with tokenizer.json publishes
+    an hf tokenizer spec + chat template in its model card, and the
+    frontend's make_tokenizer loads it."""
+    import asyncio
+    import json
+
+    from dynamo_amd.engine import EngineConfig, LLMEngine
+    from dynamo_amd.frontend.tokenizer import make_tokenizer
+    from dynamo_amd.runtime import DistributedRuntime, MemoryDiscovery
+    from dynamo_amd.workers import WorkerService
+
+    cfg = dataclasses.replace(PRESETS["tiny-llama"])
+    src = build_model(cfg, "cpu", torch.float32, None, seed=2)
+    export_hf(src, str(tmp_path))
+    # minimal real tokenizers JSON (WordLevel with a tiny vocab)
+    vocab = {"<unk>": 0, "</s>": 1, "hello": 2, "world": 3}
+    tok = {"version": "1.0", "truncation": None, "padding": None,
+           "added_tokens": [], "normalizer": None,
+           "pre_tokenizer": {"type": "Whitespace"},
+           "post_processor": None, "decoder": None,
+           "model": {"type": "WordLevel", "vocab": vocab, "unk_token": "<unk>"}}
+    (tmp_path / "tokenizer.json").write_text(json.dumps(tok))
+    (tmp_path / "tokenizer_config.json").write_text(json.dumps(
+        {"chat_template": "{% for m in messages %}[{{ m.role }}]{{ m.content }}{% endfor %}"}))
+
+    async def main():
+        rt = DistributedRuntime(MemoryDiscovery())
+        eng = LLMEngine(EngineConfig(
+            model=resolve_model_config(str(tmp_path)), device="cpu",
+            dtype="float32", kv_pool_pages=32, max_model_len=256,
+            page_size=16))
+        ws = WorkerService(eng, rt)
+        card = ws._card_body()
+        assert card["tokenizer"]["type"] == "hf"
+        assert "chat_template" in card
+        t = make_tokenizer(card["tokenizer"])
+        assert t.encode("hello world") == [2, 3]
+        assert t.eos_id == 1
+        await rt.shutdown(drain=False)
+    asyncio.new_event_loop().run_until_complete(main())
