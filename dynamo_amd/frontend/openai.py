@@ -190,10 +190,15 @@ def build_app(manager: ModelManager,
                     "top_k": getattr(req, "top_k", 0),
                     "seed": getattr(req, "seed", 0),
                     "logprobs": getattr(req, "logprobs", None) or 0}
-        eos = getattr(entry.tokenizer, "eos_id", None)
+        # checkpoint generation_config eos ids (possibly several, e.g.
+        # llama-3) win over the tokenizer's probed eos token
+        eos_ids = (entry.card or {}).get("eos_token_ids")
+        if not eos_ids:
+            eos = getattr(entry.tokenizer, "eos_id", None)
+            eos_ids = [eos] if eos is not None else []
         stop = {"max_tokens": _max_tokens_of(req),
                 "ignore_eos": getattr(req, "ignore_eos", False),
-                "stop_token_ids": [eos] if eos is not None else []}
+                "stop_token_ids": list(eos_ids)}
         from dynamo_amd.router.kv_router import AllWorkersBusy
         pe = getattr(req, "prompt_embeds", None)
         if pe:

@@ -120,6 +120,16 @@ class WorkerService:
         tok = _os.path.join(wp, "tokenizer.json")
         if _os.path.exists(tok):
             out["tokenizer"] = {"type": "hf", "path": tok}
+        gcfg = _os.path.join(wp, "generation_config.json")
+        if _os.path.exists(gcfg):
+            try:
+                with open(gcfg) as f:
+                    eos = _json.load(f).get("eos_token_id")
+                if eos is not None:
+                    out["eos_token_ids"] = (eos if isinstance(eos, list)
+                                            else [eos])
+            except (OSError, ValueError):
+                pass
         tcfg = _os.path.join(wp, "tokenizer_config.json")
         if _os.path.exists(tcfg):
             try:

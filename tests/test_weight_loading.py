@@ -145,6 +145,8 @@ def test_worker_card_advertises_checkpoint_tokenizer(tmp_path):
     (tmp_path / "tokenizer.json").write_text(json.dumps(tok))
     (tmp_path / "tokenizer_config.json").write_text(json.dumps(
         {"chat_template": "{% for m in messages %}[{{ m.role }}]{{ m.content }}{% endfor %}"}))
+    (tmp_path / "generation_config.json").write_text(json.dumps(
+        {"eos_token_id": [1, 3]}))
 
     async def main():
         rt = DistributedRuntime(MemoryDiscovery())
@@ -159,5 +161,6 @@ def test_worker_card_advertises_checkpoint_tokenizer(tmp_path):
         t = make_tokenizer(card["tokenizer"])
         assert t.encode("hello world") == [2, 3]
         assert t.eos_id == 1
+        assert card["eos_token_ids"] == [1, 3]
         await rt.shutdown(drain=False)
     asyncio.new_event_loop().run_until_complete(main())
