@@ -92,3 +92,83 @@ def test_lora_worker_endpoints():
         await ws.stop()
         await rt.shutdown(drain=False)
     asyncio.new_event_loop().run_until_complete(main())
+
+
+def test_peft_adapter_exact_vs_dense_delta(tmp_path):
+    """PEFT-format adapter (separate q/k/v/o + gate/up/down targets) loads
+    onto the fused projections as a block-form LoRA; activating it must
+    equal applying the dense deltas W += scale*B@A directly."""
+    import dataclasses
+    import json
+
+    import torch
+    from safetensors.torch import save_file
+
+    from dynamo_amd.engine.config import PRESETS
+    from dynamo_amd.lora.manager import LoRAManager
+    from dynamo_amd.models.registry import build_model
+
+    cfg = dataclasses.replace(PRESETS["tiny-llama"])
+    torch.manual_seed(0)
+    model = build_model(cfg, "cpu", torch.float32, None, seed=4)
+    r, alpha = 4, 8.0
+    g = torch.Generator().manual_seed(9)
+    sd = {}
+    hd = cfg.head_dim
+    D = cfg.hidden_size
+    dims = {"q_proj": cfg.num_q_heads * hd, "k_proj": cfg.num_kv_heads * hd,
+            "v_proj": cfg.num_kv_heads * hd, "o_proj": D,
+            "gate_proj": cfg.intermediate_size,
+            "up_proj": cfg.intermediate_size, "down_proj": D}
+    ins = {"q_proj": D, "k_proj": D, "v_proj": D,
+           "o_proj": cfg.num_q_heads * hd, "gate_proj": D, "up_proj": D,
+           "down_proj": cfg.intermediate_size}
+    for li in range(cfg.num_layers):
+        for proj, outf in dims.items():
+            mod = "self_attn" if "proj" in proj and proj[0] in "qkvo" else "mlp"
+            stem = f"base_model.model.model.layers.{li}.{mod}.{proj}"
+            sd[f"{stem}.lora_A.weight"] = torch.randn(r, ins[proj],
+                                                      generator=g) * 0.1
+            sd[f"{stem}.lora_B.weight"] = torch.randn(outf, r,
+                                                      generator=g) * 0.1
+    save_file(sd, str(tmp_path / "adapter_model.safetensors"))
+    (tmp_path / "adapter_config.json").write_text(json.dumps(
+        {"r": r, "lora_alpha": alpha,
+         "target_modules": list(dims)}))
+
+    x = torch.randn(5, D)
+    lm = LoRAManager(model)
+    lm.load("a", str(tmp_path))
+    lm.activate("a")
+    at = model.layers[0].attn
+    from dynamo_amd.models.layers import linear_lora
+    y_lora = linear_lora(x, at.wqkv, at.lora, "qkv")
+    # dense reference: per-projection deltas applied to the fused weight
+    scale = alpha / r
+    wq = at.wqkv.clone()
+    qr, kr = at.hq * hd, at.hkv * hd
+    pre = "base_model.model.model.layers.0.self_attn."
+    wq[:qr] += scale * (sd[pre + "q_proj.lora_B.weight"]
+                        @ sd[pre + "q_proj.lora_A.weight"])
+    wq[qr:qr + kr] += scale * (sd[pre + "k_proj.lora_B.weight"]
+                               @ sd[pre + "k_proj.lora_A.weight"])
+    wq[qr + kr:] += scale * (sd[pre + "v_proj.lora_B.weight"]
+                             @ sd[pre + "v_proj.lora_A.weight"])
+    y_ref = x @ wq.t()
+    assert torch.allclose(y_lora, y_ref, rtol=1e-4, atol=1e-4)
+    # mlp fused gate_up too
+    mlp = model.layers[0].mlp
+    y2 = linear_lora(x, mlp.w_gate_up, mlp.lora, "gate_up")
+    wg = mlp.w_gate_up.clone()
+    pm = "base_model.model.model.layers.0.mlp."
+    wg[:mlp.I] += scale * (sd[pm + "gate_proj.lora_B.weight"]
+                           @ sd[pm + "gate_proj.lora_A.weight"])
+    wg[mlp.I:] += scale * (sd[pm + "up_proj.lora_B.weight"]
+                           @ sd[pm + "up_proj.lora_A.weight"])
+    assert torch.allclose(y2, x @ wg.t(), rtol=1e-4, atol=1e-4)
+    # partial adapter (q only) activates without KeyError
+    sd2 = {k: v for k, v in sd.items() if "q_proj" in k}
+    save_file(sd2, str(tmp_path / "adapter_model.safetensors"))
+    lm.load("partial", str(tmp_path))
+    lm.activate("partial")
+    assert "o" not in (model.layers[0].attn.lora or {})
