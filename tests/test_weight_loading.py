@@ -164,3 +164,29 @@ def test_worker_card_advertises_checkpoint_tokenizer(tmp_path):
         assert card["eos_token_ids"] == [1, 3]
         await rt.shutdown(drain=False)
     asyncio.new_event_loop().run_until_complete(main())
+
+
+def test_multifile_checkpoint(tmp_path):
+    """Sharded checkpoints (model-00001-of-0000N.safetensors) load the
+    same as single-file ones."""
+    from safetensors import safe_open
+    from safetensors.torch import save_file
+    cfg = dataclasses.replace(PRESETS["tiny-llama"])
+    src = build_model(cfg, "cpu", torch.float32, None, seed=6)
+    export_hf(src, str(tmp_path))
+    # split model.safetensors into two shards
+    tensors = {}
+    with safe_open(str(tmp_path / "model.safetensors"),
+                   framework="pt") as f:
+        for k in f.keys():
+            tensors[k] = f.get_tensor(k)
+    (tmp_path / "model.safetensors").unlink()
+    keys = sorted(tensors)
+    mid = len(keys) // 2
+    save_file({k: tensors[k] for k in keys[:mid]},
+              str(tmp_path / "model-00001-of-00002.safetensors"))
+    save_file({k: tensors[k] for k in keys[mid:]},
+              str(tmp_path / "model-00002-of-00002.safetensors"))
+    dst = build_model(cfg, "cpu", torch.float32, None, seed=77)
+    assert load_weights(dst, str(tmp_path)) == len(keys)
+    assert torch.equal(_logits(src, cfg), _logits(dst, cfg))
