@@ -190,3 +190,25 @@ def test_multifile_checkpoint(tmp_path):
     dst = build_model(cfg, "cpu", torch.float32, None, seed=77)
     assert load_weights(dst, str(tmp_path)) == len(keys)
     assert torch.equal(_logits(src, cfg), _logits(dst, cfg))
+
+
+def test_mixtral_ep_expert_slicing(tmp_path):
+    """Under expert parallelism each rank loads only its expert window,
+    at full intermediate width."""
+    from dynamo_amd.models.layers import TPContext
+    cfg = dataclasses.replace(PRESETS["tiny-mixtral"], moe_ep=True)
+    full = build_model(dataclasses.replace(cfg, moe_ep=False),
+                       "cpu", torch.float32, None, seed=8)
+    export_hf(full, str(tmp_path))
+    E = cfg.num_experts
+    for rank in (0, 1):
+        tp = TPContext(2, rank, group=None)
+        m = build_model(cfg, "cpu", torch.float32, tp, seed=55)
+        load_weights(m, str(tmp_path))
+        moe = m.layers[0].moe
+        assert moe.ep and moe.El == E // 2
+        fmoe = full.layers[0].moe
+        for el in range(moe.El):
+            e = moe.e0 + el
+            assert torch.equal(moe.w_gate_up[el], fmoe.w_gate_up[e])
+            assert torch.equal(moe.w_down[el], fmoe.w_down[e])
