@@ -15,7 +15,7 @@ import time
 import uuid
 from typing import AsyncIterator, List, Optional, Union
 
-from fastapi import FastAPI, HTTPException, Request
+from fastapi import FastAPI, HTTPException, Request, WebSocket
 from fastapi.responses import JSONResponse, StreamingResponse
 from pydantic import BaseModel, Field
 
@@ -272,6 +272,86 @@ def build_app(manager: ModelManager,
         return {"b64": _b64.b64encode(cf.to_bytes()).decode(),
                 "count": cf.count(), "block_size": r.cfg.block_size,
                 "salt": r.cfg.block_salt}
+
+    @app.websocket("/v1/realtime")
+    async def realtime(ws: WebSocket):
+        """Realtime bidirectional session, TEXT modality (the reference's
+        realtime path is audio-centric via vllm/omni realtime_handler.py;
+        this build has no audio models, so the OpenAI realtime event
+        protocol is served for text): session.update ->
+        session.updated; conversation.item.create (ack'd) accumulates the
+        dialogue; response.create streams response.created ->
+        response.output_text.delta* -> response.output_text.done ->
+        response.done. Unlike the reference MVP the session is STATEFUL -
+        each response re-renders the full conversation (and appends the
+        assistant turn for later ones)."""
+        await ws.accept()
+        import uuid as _uuid
+        session = {"id": f"sess_{_uuid.uuid4().hex[:16]}",
+                   "model": ws.query_params.get("model", ""),
+                   "modalities": ["text"]}
+        conversation: List[dict] = []
+        await ws.send_json({"type": "session.created", "session": session})
+        try:
+            while True:
+                ev = await ws.receive_json()
+                t = ev.get("type")
+                if t == "session.update":
+                    patch = ev.get("session") or {}
+                    session.update({k: v for k, v in patch.items()
+                                    if k in ("model", "modalities",
+                                             "instructions")})
+                    await ws.send_json({"type": "session.updated",
+                                        "session": session})
+                elif t == "conversation.item.create":
+                    item = ev.get("item") or {}
+                    text = "".join(
+                        c.get("text", "") for c in item.get("content", [])
+                        if isinstance(c, dict))
+                    conversation.append({"role": item.get("role", "user"),
+                                         "content": text})
+                    await ws.send_json({"type": "conversation.item.created",
+                                        "item": {"id": f"item_{len(conversation)}",
+                                                 "role": item.get("role", "user")}})
+                elif t == "response.create":
+                    try:
+                        entry = manager.get(session.get("model") or "")
+                    except KeyError:
+                        await ws.send_json({"type": "error", "error": {
+                            "message": "unknown model"}})
+                        continue
+                    msgs = list(conversation)
+                    if session.get("instructions"):
+                        msgs = [{"role": "system",
+                                 "content": session["instructions"]}] + msgs
+                    prompt = entry.templater.render(msgs)
+                    toks = entry.tokenizer.encode(prompt)
+                    rid = f"resp_{_uuid.uuid4().hex[:16]}"
+                    await ws.send_json({"type": "response.created",
+                                        "response": {"id": rid}})
+                    opts = ((ev.get("response") or {}).get(
+                        "max_output_tokens") or 128)
+                    produced: List[int] = []
+                    acc = ""
+                    async for chunk in manager.generate_tokens(
+                            entry, toks, {"temperature": 0.0},
+                            {"max_tokens": int(opts)}):
+                        prev = len(produced)
+                        produced.extend(chunk.get("token_ids", []))
+                        d = entry.tokenizer.decode_incremental(produced, prev)
+                        if d:
+                            acc += d
+                            await ws.send_json({
+                                "type": "response.output_text.delta",
+                                "response_id": rid, "delta": d})
+                    conversation.append({"role": "assistant", "content": acc})
+                    await ws.send_json({"type": "response.output_text.done",
+                                        "response_id": rid, "text": acc})
+                    await ws.send_json({"type": "response.done",
+                                        "response": {"id": rid,
+                                                     "status": "completed"}})
+        except Exception:
+            return
 
     @app.get("/config")
     async def config_dump():

@@ -1094,3 +1094,85 @@ def test_request_template_defaults():
         await c2.aclose()
         await teardown(services, mgr, client)
     run(main())
+
+
+def test_realtime_websocket_text_session():
+    """/v1/realtime (text modality): session lifecycle, stateful
+    conversation, streamed response deltas - driven over raw ASGI on the
+    same event loop as the stack."""
+    async def main():
+        shared, services, mgr, client = await with_stack(nworkers=1)
+        app = client._transport.app
+
+        to_app: asyncio.Queue = asyncio.Queue()
+        from_app: asyncio.Queue = asyncio.Queue()
+
+        async def receive():
+            return await to_app.get()
+
+        async def send(msg):
+            await from_app.put(msg)
+
+        scope = {"type": "websocket", "path": "/v1/realtime",
+                 "query_string": b"model=mock-model", "headers": [],
+                 "subprotocols": []}
+        task = asyncio.ensure_future(app(scope, receive, send))
+        import json as _json
+
+        async def send_json(o):
+            await to_app.put({"type": "websocket.receive",
+                              "text": _json.dumps(o)})
+
+        async def recv_json():
+            while True:
+                m = await asyncio.wait_for(from_app.get(), 10)
+                if m["type"] == "websocket.send" and "text" in m:
+                    return _json.loads(m["text"])
+                if m["type"] == "websocket.accept":
+                    continue
+                raise AssertionError(m)
+
+        await to_app.put({"type": "websocket.connect"})
+        ev = await recv_json()
+        assert ev["type"] == "session.created"
+        await send_json({"type": "session.update",
+                         "session": {"instructions": "be brief"}})
+        assert (await recv_json())["type"] == "session.updated"
+        await send_json({"type": "conversation.item.create",
+                         "item": {"role": "user", "content": [
+                             {"type": "input_text", "text": "hello"}]}})
+        assert (await recv_json())["type"] == "conversation.item.created"
+        await send_json({"type": "response.create",
+                         "response": {"max_output_tokens": 6}})
+        assert (await recv_json())["type"] == "response.created"
+        deltas = []
+        while True:
+            ev = await recv_json()
+            if ev["type"] == "response.output_text.delta":
+                deltas.append(ev["delta"])
+            elif ev["type"] == "response.output_text.done":
+                assert ev["text"] == "".join(deltas)
+            elif ev["type"] == "response.done":
+                break
+        assert deltas
+        # second turn re-renders the (now longer) conversation
+        await send_json({"type": "conversation.item.create",
+                         "item": {"role": "user", "content": [
+                             {"type": "input_text", "text": "more"}]}})
+        await recv_json()
+        await send_json({"type": "response.create",
+                         "response": {"max_output_tokens": 4}})
+        types = []
+        while True:
+            ev = await recv_json()
+            types.append(ev["type"])
+            if ev["type"] == "response.done":
+                break
+        assert "response.output_text.delta" in types
+        await to_app.put({"type": "websocket.disconnect", "code": 1000})
+        try:
+            await asyncio.wait_for(task, 5)
+        except Exception:
+            task.cancel()
+        await teardown(services, mgr, client)
+    run(main())
