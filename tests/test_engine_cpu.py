@@ -495,3 +495,67 @@ def test_attached_request_first_step_is_decode():
     sched = eng.scheduler.schedule()
     assert len(sched.decodes) == 1 and not sched.prefills
     assert sched.decodes[0].req is req and sched.decodes[0].n_new == 1
+
+
+def test_engine_allocator_invariants_random_ops():
+    """Property test: under random add/step/abort/clear interleavings the
+    page allocator never leaks or double-frees - every page is exactly one
+    of {free, referenced, evictable(prefix-cached, ref 0)} and the three
+    partitions tile the pool."""
+    from hypothesis import HealthCheck, given, settings
+    from hypothesis import strategies as st
+
+    from dynamo_amd.engine import EngineConfig, LLMEngine, SamplingParams
+    from dynamo_amd.engine.config import PRESETS
+
+    def check(alloc):
+        free = set(alloc.free)
+        refed = {p for p, r in enumerate(alloc.ref) if r > 0}
+        evict = set(alloc.evictable)
+        assert not (free & refed), "page both free and referenced"
+        assert not (free & evict), "page both free and evictable"
+        assert not (refed & evict), "page referenced AND evictable"
+        assert len(free) + len(refed) + len(evict) == alloc.num_pages, (
+            len(free), len(refed), len(evict), alloc.num_pages)
+        for p, r in enumerate(alloc.ref):
+            assert r >= 0, f"negative refcount on page {p}"
+
+    @settings(max_examples=15, deadline=None,
+              suppress_health_check=[HealthCheck.too_slow])
+    @given(st.data())
+    def run(data):
+        cfg = EngineConfig(model=PRESETS["tiny-llama"], device="cpu",
+                           kv_pool_pages=24, max_model_len=256,
+                           page_size=16, max_num_seqs=4,
+                           max_batched_tokens=128)
+        eng = LLMEngine(cfg)
+        nreq = 0
+        ops = data.draw(st.lists(st.sampled_from(
+            ["add", "step", "step", "abort", "clear"]),
+            min_size=4, max_size=30))
+        for op in ops:
+            if op == "add":
+                nreq += 1
+                n = data.draw(st.integers(8, 120))
+                eng.add_request(f"r{nreq}", list(range(n)),
+                                SamplingParams(max_tokens=4,
+                                               ignore_eos=True))
+            elif op == "step":
+                eng.step()
+            elif op == "abort" and eng.requests:
+                rid = data.draw(st.sampled_from(sorted(eng.requests)))
+                eng.abort(rid)
+            elif op == "clear" and not eng.requests:
+                eng.clear_kv()
+            check(eng.alloc)
+        # drain everything
+        for _ in range(200):
+            if not eng.has_work():
+                break
+            eng.step()
+            check(eng.alloc)
+        assert not eng.has_work()
+        check(eng.alloc)
+        assert sum(1 for r in eng.alloc.ref if r > 0) == 0
+
+    run()
