@@ -58,3 +58,31 @@ def test_postprocess_combined():
     assert reasoning == "need the weather tool"
     assert len(calls) == 1 and calls[0]["function"]["name"] == "w"
     assert content == "Sure."
+
+
+def test_incremental_detok_property():
+    """Property: concatenated decode_incremental chunks == full decode,
+    for any token stream and any chunking (streaming-text correctness)."""
+    from hypothesis import given, settings
+    from hypothesis import strategies as st
+
+    from dynamo_amd.frontend.tokenizer import ByteTokenizer
+
+    tok = ByteTokenizer(512)
+
+    @settings(max_examples=60, deadline=None)
+    @given(st.lists(st.integers(0, 255), min_size=0, max_size=80),
+           st.lists(st.integers(1, 7), min_size=1, max_size=30))
+    def run(ids, cuts):
+        acc = ""
+        pos = 0
+        i = 0
+        while pos < len(ids):
+            step = cuts[i % len(cuts)]
+            i += 1
+            nxt = min(len(ids), pos + step)
+            acc += tok.decode_incremental(ids[:nxt], pos)
+            pos = nxt
+        assert acc == tok.decode(ids)
+
+    run()
